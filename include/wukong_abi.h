@@ -1,0 +1,163 @@
+/*
+ * wukong_abi.h — C-ABI drop-in boundary of the MI355X-native
+ * graph-exploration engine.
+ *
+ * Each entry point cites the reference interface it replaces
+ * (paths relative to the SJTU-IPADS/wukong tree).  All tables are
+ * row-major uint32 (sid_t, core/type.hpp:36); variables are negative
+ * ids with col = v2c_map[-(vid+1)] (core/query.hpp:352-374); constants
+ * are positive sids; index ids satisfy 1 < id < 2^17, normal vertex
+ * ids >= 2^17 (core/store/vertex.hpp:34-43).
+ */
+#ifndef WUKONG_ABI_H
+#define WUKONG_ABI_H
+
+#include <stdint.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+typedef uint32_t wk_sid_t;  /* string id            — core/type.hpp:36 */
+typedef int32_t  wk_ssid_t; /* signed string id     — core/type.hpp:37 */
+
+enum { WK_DIR_IN = 0, WK_DIR_OUT = 1 };      /* core/type.hpp dir_t    */
+enum { WK_PREDICATE_ID = 0, WK_TYPE_ID = 1 };/* core/store/vertex.hpp:39 */
+#define WK_BLANK_ID 0xFFFFFFFFu              /* core/type.hpp:38       */
+
+/* One triple pattern — core/query.hpp:95-116 (Pattern{s,p,o,dir}). */
+typedef struct {
+    wk_ssid_t subject;
+    wk_ssid_t predicate;
+    wk_ssid_t object;
+    int32_t   direction;  /* WK_DIR_IN / WK_DIR_OUT */
+} wk_pattern_t;
+
+/* A full query plan — the planner-ordered pattern list plus the final
+ * projection (core/query.hpp required_vars/distinct/limit,
+ * core/engine/sparql.hpp:1424-1551). */
+typedef struct {
+    const wk_pattern_t *patterns;
+    int32_t   npatterns;
+    int32_t   nvars;              /* variables are -1..-nvars          */
+    const wk_ssid_t *required_vars;
+    int32_t   nrequired;
+    int32_t   distinct;           /* 0/1 */
+    int64_t   limit;              /* -1 = none */
+    int64_t   offset;
+} wk_plan_t;
+
+/* A materialised binding table (SPARQLQuery::Result subset —
+ * core/query.hpp:312-334).  Owned by the library; free with
+ * wk_result_free. */
+typedef struct {
+    int32_t   col_num;
+    int64_t   row_num;
+    wk_sid_t *table;       /* row-major, row_num*col_num entries */
+    int32_t   status_code; /* 0 = SUCCESS (utils/errors.hpp)     */
+} wk_result_t;
+
+/* ---------- synthetic data ------------------------------------- */
+/* Seeded LUBM-shaped ID-triple generator (replaces datagen/ +
+ * the absent Java LUBM generator; ID scheme per
+ * datagen/generate_data.cpp:122-123: index ids from 2, normal ids
+ * from 2^17, fixed schema enumeration order).  Returns the triples
+ * of partition `sid` of `nsrv` (pso-side: s%nsrv==sid OR pos-side:
+ * o%nsrv==sid — core/loader/base_loader.hpp:344-352 keeps both).
+ * Caller frees with wk_free_triples. */
+int64_t wk_lubm_gen(int32_t nuniv, uint64_t seed, int32_t sid, int32_t nsrv,
+                    wk_sid_t **out_spo /* 3*n entries */);
+void    wk_free_triples(wk_sid_t *spo);
+
+/* ---------- store ---------------------------------------------- */
+/* Host-side store build from ID-triples (StaticGStore::init,
+ * core/store/static_gstore.hpp:383-454; loader sort/dedup,
+ * core/loader/base_loader.hpp:302-373). */
+typedef struct wk_store wk_store_t;
+wk_store_t *wk_store_build(const wk_sid_t *spo, int64_t ntriples,
+                           int32_t sid, int32_t nsrv);
+void        wk_store_free(wk_store_t *);
+
+/* DGraph::get_triples / get_index (core/dgraph.hpp:106-112;
+ * GStore::get_edges core/store/gstore.hpp:1043-1054).  Pointer into
+ * the store's host edge array; not owned by the caller; empty =
+ * (NULL, *sz==0).  Lock-free after build. */
+const wk_sid_t *wk_store_get_triples(const wk_store_t *, wk_sid_t vid,
+                                     wk_sid_t pid, int32_t dir, uint64_t *sz);
+const wk_sid_t *wk_store_get_index(const wk_store_t *, wk_sid_t pid,
+                                   int32_t dir, uint64_t *sz);
+
+/* Store introspection (for tests / upload verification — the `gsck`
+ * idea, core/store/gchecker.hpp:364-392). */
+uint64_t wk_store_num_slots(const wk_store_t *);
+uint64_t wk_store_num_edges(const wk_store_t *);
+uint64_t wk_store_checksum(const wk_store_t *); /* FNV over vertices+edges */
+
+/* ---------- GPU engine ------------------------------------------ */
+/* Mirrors the five-call GPU surface: load_result_buf
+ * (core/gpu/gpu_engine.hpp:253-261), execute_one_pattern (:263-336),
+ * generate_sub_query (:338-391), result fetch
+ * (core/gpu/gpu_engine_cuda.hpp:189-195), plus engine create =
+ * GPUMem+GPUCache init (core/wukong.cpp:235-241) with the cache
+ * replaced by a fully HBM-resident store. */
+typedef struct wk_engine wk_engine_t;
+
+wk_engine_t *wk_engine_create(const wk_store_t *, int32_t device);
+void         wk_engine_destroy(wk_engine_t *);
+
+/* Whole-query execution on one GPU (Engine::execute_sparql_query +
+ * SPARQLEngine::execute_patterns, core/engine/sparql.hpp:1113-1154,
+ * 1564-1672, single-server path).  Returns 0 on success. */
+int32_t wk_engine_run_query(wk_engine_t *, const wk_plan_t *, wk_result_t *out);
+
+/* Step-level API (multi-GPU driver / tests). The engine holds ONE
+ * current query state with a dual device rbuf (gpu_mem.hpp:116-124). */
+int32_t wk_engine_begin_query(wk_engine_t *, const wk_plan_t *);
+/* Upload/replace the current binding table (H2D load_result_buf). */
+int32_t wk_engine_load_rbuf(wk_engine_t *, const wk_sid_t *table,
+                            int64_t nrows, int32_t ncols,
+                            const int32_t *v2c_map, int32_t pattern_step);
+/* Device-pointer variant: table already in HBM (e.g. a torch tensor). */
+int32_t wk_engine_load_rbuf_device(wk_engine_t *, const wk_sid_t *dev_table,
+                                   int64_t nrows, int32_t ncols,
+                                   const int32_t *v2c_map, int32_t pattern_step);
+/* Run pattern [pattern_step]; advances the step.  Out: new row count. */
+int32_t wk_engine_execute_one_pattern(wk_engine_t *, int64_t *nrows_out);
+int32_t wk_engine_pattern_step(const wk_engine_t *);
+int32_t wk_engine_col_num(const wk_engine_t *);
+/* Split the current table by hash of the next pattern's start var
+ * (sparql.hpp:746-799; gpu_hash.cu:600-760): fills per-destination row
+ * counts and packs rows into dev_out (device buffer, row-major,
+ * chunks contiguous per destination, capacity cap_rows rows). */
+int32_t wk_engine_generate_sub_query(wk_engine_t *, int32_t ndst,
+                                     wk_sid_t *dev_out, int64_t cap_rows,
+                                     int64_t *rows_per_dst);
+/* Finish: run remaining host-side final ops and download. */
+int32_t wk_engine_fetch_result(wk_engine_t *, const wk_plan_t *, wk_result_t *out);
+/* Download the CURRENT binding table without final ops (tests/driver). */
+int32_t wk_engine_fetch_raw(wk_engine_t *, wk_result_t *out);
+
+void wk_result_free(wk_result_t *);
+
+/* Per-kernel timing of the last run_query (HIP events, engine stream):
+ * totals in microseconds + bytes of algorithmic traffic,
+ * slots: 0=probe,1=scan,2=expand,3=filter,4=i2u/c2u,5=split,6=other. */
+int32_t wk_engine_kernel_stats(wk_engine_t *, double *usec7, double *bytes7,
+                               int64_t *launches7);
+
+/* Library/build info: returns gfx arch string the .so was built for. */
+const char *wk_build_arch(void);
+int32_t     wk_device_count(void);
+
+/* Layout pinning (tests only): the reference's TomasWang hash
+ * (utils/math.hpp:58-66) and ikey_t/iptr_t bit packings
+ * (core/store/vertex.hpp:50-151), checked against golden vectors dumped
+ * from the reference's own headers (tests/golden/hash_golden.csv). */
+uint64_t wk_hash_u64(uint64_t x);
+uint64_t wk_key_pack(uint64_t vid, uint64_t pid, uint64_t dir);
+uint64_t wk_ptr_pack(uint64_t size, uint64_t off, uint64_t type);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* WUKONG_ABI_H */

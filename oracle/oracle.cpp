@@ -1,0 +1,434 @@
+/*
+ * ORACLE — TEST INFRASTRUCTURE ONLY.
+ *
+ * Dependency-free C++17 CPU restatement of the reference engine's hot
+ * path, used exclusively as the parity checker and as bench.py's
+ * `cpu_baseline` leg.  Only `tests/`, `__graft_entry__.smoke()` and
+ * `bench.py` may link/load/call this library.  The product path
+ * (wukong_amd/) must never import or fall back to it.
+ *
+ * What it restates (reference = SJTU-IPADS/wukong):
+ *  - loader partition/sort/dedup:  core/loader/base_loader.hpp:302-373
+ *  - store semantics (edge lists per [vid|pid|dir] key, predicate/type
+ *    indexes):                     core/store/gstore.hpp:55-120,858-888,
+ *                                  core/store/static_gstore.hpp:64-280
+ *  - key/ptr bit layouts + hash:   core/store/vertex.hpp:34-151,
+ *                                  utils/math.hpp:51-66
+ *  - operators:                    core/engine/sparql.hpp:80-549
+ *      index_to_unknown :194-231, const_to_unknown :238-285,
+ *      known_to_unknown :295-407, known_to_known :416-476,
+ *      known_to_const :484-549, const_to_known :138-186,
+ *      index_to_known :80-135
+ *  - dispatch:                     core/engine/sparql.hpp:1016-1058
+ *  - mt_factor index slicing:      core/engine/sparql.hpp:210-221
+ *  - final (DISTINCT/OFFSET/LIMIT/projection): sparql.hpp:1424-1551
+ *
+ * Parity pinning: the reference does not compile in this
+ * container (boost/TBB/MPI absent — SURVEY.md §8c), so the oracle is
+ * pinned by (a) golden hash/bit-layout vectors dumped from the
+ * reference's OWN standalone headers (oracle/ref_dump.cpp →
+ * tests/golden/hash_golden.csv), and (b) an algorithmically independent
+ * brute-force BGP evaluator in brute.cpp (hash-join over the raw triple
+ * list, no store, no exploration) that must agree on every query.
+ * Count-level parity vs the reference's published LUBM-2560 numbers is
+ * PARTIAL: real LUBM data is not generatable here (Java generator).
+ *
+ * The store here is deliberately a *different* data structure from the
+ * product's cluster-hash (std::unordered_map keyed by the packed 64-bit
+ * ikey) — same key encoding and edge-list semantics, independent layout.
+ */
+#include <cstdint>
+#include <cstring>
+#include <cstdlib>
+#include <vector>
+#include <unordered_map>
+#include <unordered_set>
+#include <algorithm>
+#include <omp.h>
+#include "ok_internal.h"
+
+namespace ok {
+
+// TomasWang hash — utils/math.hpp:58-66
+static inline uint64_t hash_u64(uint64_t key) {
+    key = (~key) + (key << 21);
+    key = key ^ (key >> 24);
+    key = (key + (key << 3)) + (key << 8);
+    key = key ^ (key >> 14);
+    key = (key + (key << 2)) + (key << 4);
+    key = key ^ (key >> 28);
+    key = key + (key << 31);
+    return key;
+}
+
+// ---- build (loader + static store semantics) -------------------------
+static ctx *build(const sid_t *spo, int64_t n, int sid, int nsrv) {
+    ctx *c = new ctx();
+    c->sid = sid; c->nsrv = nsrv;
+
+    std::vector<triple> pso, pos;
+    for (int64_t i = 0; i < n; i++) {
+        triple t{spo[3 * i], spo[3 * i + 1], spo[3 * i + 2]};
+        // partition — base_loader.hpp:344-352
+        if ((int)(t.s % (sid_t)nsrv) == sid) pso.push_back(t);
+        if ((int)(t.o % (sid_t)nsrv) == sid) pos.push_back(t);
+    }
+    // sort + dedup — base_loader.hpp:367-377
+    std::sort(pso.begin(), pso.end(), [](const triple &a, const triple &b) {
+        return a.p != b.p ? a.p < b.p : (a.s != b.s ? a.s < b.s : a.o < b.o);
+    });
+    std::sort(pos.begin(), pos.end(), [](const triple &a, const triple &b) {
+        return a.p != b.p ? a.p < b.p : (a.o != b.o ? a.o < b.o : a.s < b.s);
+    });
+    auto eq = [](const triple &a, const triple &b) {
+        return a.s == b.s && a.p == b.p && a.o == b.o;
+    };
+    pso.erase(std::unique(pso.begin(), pso.end(), eq), pso.end());
+    pos.erase(std::unique(pos.begin(), pos.end(), eq), pos.end());
+    c->triples = pso;  // brute.cpp uses the pso-side dedup'd set
+
+    auto put = [&](uint64_t vid, uint64_t pid, int dir, const std::vector<sid_t> &vals) {
+        c->kv[key_pack(vid, pid, (uint64_t)dir)] = {c->edges.size(), vals.size()};
+        c->edges.insert(c->edges.end(), vals.begin(), vals.end());
+    };
+
+    // OUT keys [s|p|OUT] from pso runs — static_gstore.hpp:95-117; also
+    // collect pidx_in (subjects) and tidx (type members) — gstore.hpp:858-888
+    std::unordered_map<sid_t, std::vector<sid_t>> pidx_in, pidx_out, tidx;
+    for (size_t i = 0; i < pso.size();) {
+        size_t j = i + 1;
+        while (j < pso.size() && pso[j].p == pso[i].p && pso[j].s == pso[i].s) j++;
+        std::vector<sid_t> vals;
+        for (size_t k = i; k < j; k++) vals.push_back(pso[k].o);
+        put(pso[i].s, pso[i].p, DIR_OUT, vals);
+        if (pso[i].p == TYPE_ID)
+            for (sid_t t : vals) tidx[t].push_back(pso[i].s);
+        else if (pso[i].p != PREDICATE_ID)
+            pidx_in[pso[i].p].push_back(pso[i].s);
+        i = j;
+    }
+    // IN keys [o|p|IN] from pos runs, skipping tpid objects —
+    // static_gstore.hpp:125-152; collect pidx_out (objects)
+    for (size_t i = 0; i < pos.size();) {
+        size_t j = i + 1;
+        while (j < pos.size() && pos[j].p == pos[i].p && pos[j].o == pos[i].o) j++;
+        if (!is_tpid(pos[i].o)) {
+            std::vector<sid_t> vals;
+            for (size_t k = i; k < j; k++) vals.push_back(pos[k].s);
+            put(pos[i].o, pos[i].p, DIR_IN, vals);
+            if (pos[i].p != PREDICATE_ID && pos[i].p != TYPE_ID)
+                pidx_out[pos[i].p].push_back(pos[i].o);
+        }
+        i = j;
+    }
+    // index keys — insert_idx, static_gstore.hpp:217-280:
+    //   [0|pid|IN] = subjects, [0|pid|OUT] = objects, [0|tid|IN] = members
+    for (auto &kvp : pidx_in) put(0, kvp.first, DIR_IN, kvp.second);
+    for (auto &kvp : pidx_out) put(0, kvp.first, DIR_OUT, kvp.second);
+    for (auto &kvp : tidx) put(0, kvp.first, DIR_IN, kvp.second);
+    return c;
+}
+
+// ---- engine ----------------------------------------------------------
+struct pattern { ssid_t s, p, o; int dir; };
+
+struct query {
+    std::vector<pattern> pats;
+    int nvars = 0;
+    std::vector<int> v2c;        // query.hpp:352-374
+    std::vector<sid_t> table;    // row-major
+    int col_num = 0;
+    int step = 0;
+    int mt_tid = 0, mt_factor = 1;  // sparql.hpp:210-221
+
+    int var2col(ssid_t v) const { return v < 0 ? v2c[-(v + 1)] : -1; }
+    int64_t nrows() const { return col_num ? (int64_t)table.size() / col_num : 0; }
+};
+
+// one pattern — dispatch per sparql.hpp:1016-1058
+static void exec_pattern(const ctx &c, query &q) {
+    const pattern pat = q.pats[q.step];
+    const ssid_t s = pat.s, p = pat.p, o = pat.o;
+    const int d = pat.dir;
+    uint64_t sz = 0;
+
+    // index_to_unknown — sparql.hpp:194-231 (+ mt slicing :210-221)
+    if (q.step == 0 && s >= 0 && is_tpid(s)) {
+        const sid_t *edges = c.get(0, (uint64_t)s, d, &sz);
+        std::vector<sid_t> out;
+        uint64_t start = (uint64_t)(q.mt_tid % q.mt_factor);
+        uint64_t len = sz / (uint64_t)q.mt_factor;
+        for (uint64_t k = start * len; k < (start + 1) * len; k++) out.push_back(edges[k]);
+        if (start == (uint64_t)q.mt_factor - 1)
+            for (uint64_t k = (start + 1) * len; k < sz; k++) out.push_back(edges[k]);
+        q.table.swap(out);
+        q.col_num = 1;
+        q.v2c[-(o + 1)] = 0;
+        q.step++;
+        return;
+    }
+
+    if (s >= 0) {
+        const sid_t *vids = c.get((uint64_t)s, (uint64_t)p, d, &sz);
+        if (o < 0 && q.var2col(o) < 0) {
+            // const_to_unknown — sparql.hpp:250-263 (first pattern only)
+            std::vector<sid_t> out(vids, vids + sz);
+            q.table.swap(out);
+            q.v2c[-(o + 1)] = q.col_num;
+            q.col_num += 1;
+        } else {
+            // const_to_known — sparql.hpp:138-186
+            std::unordered_set<sid_t> set(vids, vids + sz);
+            int col = q.var2col(o);
+            std::vector<sid_t> out;
+            int64_t R = q.nrows();
+            for (int64_t i = 0; i < R; i++)
+                if (set.count(q.table[i * q.col_num + col]))
+                    out.insert(out.end(), q.table.begin() + i * q.col_num,
+                               q.table.begin() + (i + 1) * q.col_num);
+            q.table.swap(out);
+        }
+        q.step++;
+        return;
+    }
+
+    int col = q.var2col(s);
+    int64_t R = q.nrows();
+    const int ostat = (o >= 0) ? 2 : (q.var2col(o) >= 0 ? 1 : 0);
+
+    // consecutive-dup memo — sparql.hpp:322-345
+    sid_t cached = 0xFFFFFFFFu;
+    const sid_t *vids = nullptr;
+    sz = 0;
+    std::vector<sid_t> out;
+
+    if (ostat == 0) {
+        // known_to_unknown — sparql.hpp:295-407
+        for (int64_t i = 0; i < R; i++) {
+            sid_t cur = q.table[i * q.col_num + col];
+            if (cur != cached) {
+                cached = cur;
+                if ((sid_t)p == TYPE_ID && d == DIR_IN)
+                    vids = c.get(0, cur, d, &sz);       // :340-341 get_index
+                else
+                    vids = c.get(cur, (uint64_t)p, d, &sz);
+            }
+            for (uint64_t k = 0; k < sz; k++) {
+                out.insert(out.end(), q.table.begin() + i * q.col_num,
+                           q.table.begin() + (i + 1) * q.col_num);
+                out.push_back(vids[k]);
+            }
+        }
+        q.table.swap(out);
+        q.v2c[-(o + 1)] = q.col_num;
+        q.col_num += 1;
+    } else if (ostat == 1) {
+        // known_to_known — sparql.hpp:416-476
+        int col2 = q.var2col(o);
+        for (int64_t i = 0; i < R; i++) {
+            sid_t cur = q.table[i * q.col_num + col];
+            if (cur != cached) {
+                cached = cur;
+                vids = c.get(cur, (uint64_t)p, d, &sz);
+            }
+            sid_t known = q.table[i * q.col_num + col2];
+            for (uint64_t k = 0; k < sz; k++)
+                if (vids[k] == known) {
+                    out.insert(out.end(), q.table.begin() + i * q.col_num,
+                               q.table.begin() + (i + 1) * q.col_num);
+                    break;
+                }
+        }
+        q.table.swap(out);
+    } else {
+        // known_to_const — sparql.hpp:484-549 (memoised exist flag)
+        bool exist = false;
+        for (int64_t i = 0; i < R; i++) {
+            sid_t cur = q.table[i * q.col_num + col];
+            if (cur != cached) {
+                cached = cur;
+                exist = false;
+                vids = c.get(cur, (uint64_t)p, d, &sz);
+                for (uint64_t k = 0; k < sz; k++)
+                    if (vids[k] == (sid_t)o) { exist = true; break; }
+            }
+            if (exist)
+                out.insert(out.end(), q.table.begin() + i * q.col_num,
+                           q.table.begin() + (i + 1) * q.col_num);
+        }
+        q.table.swap(out);
+    }
+    q.step++;
+}
+
+// final ops — sparql.hpp:1424-1551 (DISTINCT = full-row sort + adjacent
+// equal-on-required-vars removal, exactly the reference's algorithm)
+static void final_process(query &q, const ssid_t *req, int nreq, int distinct,
+                          int64_t limit, int64_t offset) {
+    int C = q.col_num;
+    int64_t R = q.nrows();
+    if (distinct && R > 0) {
+        std::vector<int64_t> idx(R);
+        for (int64_t i = 0; i < R; i++) idx[i] = i;
+        std::sort(idx.begin(), idx.end(), [&](int64_t a, int64_t b) {
+            for (int c2 = 0; c2 < C; c2++) {
+                sid_t x = q.table[a * C + c2], y = q.table[b * C + c2];
+                if (x != y) return x < y;
+            }
+            return false;
+        });
+        std::vector<sid_t> kept;
+        auto eq_req = [&](int64_t a, int64_t b) {
+            for (int i = 0; i < nreq; i++) {
+                int cc = q.var2col(req[i]);
+                if (q.table[a * C + cc] != q.table[b * C + cc]) return false;
+            }
+            return true;
+        };
+        for (int64_t i = 0; i < R; i++) {
+            if (i > 0 && eq_req(idx[i - 1], idx[i])) continue;
+            kept.insert(kept.end(), q.table.begin() + idx[i] * C,
+                        q.table.begin() + (idx[i] + 1) * C);
+        }
+        q.table.swap(kept);
+        R = q.nrows();
+    }
+    if (offset > 0) {
+        int64_t drop = std::min<int64_t>(offset, R);
+        q.table.erase(q.table.begin(), q.table.begin() + drop * C);
+        R -= drop;
+    }
+    if (limit >= 0 && R > limit) { q.table.resize((size_t)limit * C); R = limit; }
+    // projection — sparql.hpp:1510-1536
+    std::vector<sid_t> proj((size_t)R * nreq);
+    for (int64_t i = 0; i < R; i++)
+        for (int j = 0; j < nreq; j++)
+            proj[i * nreq + j] = q.table[i * C + q.var2col(req[j])];
+    q.table.swap(proj);
+    q.col_num = nreq;
+}
+
+}  // namespace ok
+
+// ---------------------------------------------------------------------
+// C ABI (for ctypes in tests and bench.py cpu_baseline)
+// ---------------------------------------------------------------------
+extern "C" {
+
+typedef struct {
+    int32_t subject, predicate, object, direction;
+} ok_pattern_t;  // matches wk_pattern_t layout
+
+void *ok_build(const uint32_t *spo, int64_t n, int32_t sid, int32_t nsrv) {
+    return ok::build(spo, n, sid, nsrv);
+}
+void ok_free(void *c) { delete (ok::ctx *)c; }
+
+const uint32_t *ok_get_triples(void *cv, uint32_t vid, uint32_t pid, int32_t dir,
+                               uint64_t *sz) {
+    return ((ok::ctx *)cv)->get(vid, pid, dir, sz);
+}
+const uint32_t *ok_get_index(void *cv, uint32_t pid, int32_t dir, uint64_t *sz) {
+    return ((ok::ctx *)cv)->get(0, pid, dir, sz);
+}
+
+uint64_t ok_hash_u64(uint64_t x) { return ok::hash_u64(x); }
+uint64_t ok_key_pack(uint64_t vid, uint64_t pid, uint64_t dir) {
+    return ok::key_pack(vid, pid, dir);
+}
+
+// Run a full plan.  mt = number of slice-threads for index-start queries
+// (the reference's num_servers*mt_factor dispatch, sparql.hpp:1064-1111;
+// sub-results merged in slice order).  Returns row count; caller frees
+// *out with ok_free_table.
+int64_t ok_run_query(void *cv, const ok_pattern_t *pats, int32_t npat,
+                     int32_t nvars, const int32_t *req, int32_t nreq,
+                     int32_t distinct, int64_t limit, int64_t offset,
+                     int32_t mt, uint32_t **out, int32_t *out_cols) {
+    ok::ctx *c = (ok::ctx *)cv;
+    bool index_start = npat > 0 && pats[0].subject >= 0 && ok::is_tpid(pats[0].subject);
+    int nslices = (index_start && mt > 1) ? mt : 1;
+
+    std::vector<ok::query> qs(nslices);
+    for (int t = 0; t < nslices; t++) {
+        ok::query &q = qs[t];
+        for (int i = 0; i < npat; i++)
+            q.pats.push_back({pats[i].subject, pats[i].predicate, pats[i].object,
+                              pats[i].direction});
+        q.nvars = nvars;
+        q.v2c.assign(nvars, -1);
+        q.mt_tid = t;
+        q.mt_factor = nslices;
+    }
+#pragma omp parallel for num_threads(nslices) schedule(static, 1)
+    for (int t = 0; t < nslices; t++) {
+        ok::query &q = qs[t];
+        while (q.step < (int)q.pats.size()) ok::exec_pattern(*c, q);
+    }
+    // merge (RMap::merge/append_result, rmap.hpp:57-87, query.hpp:697-718)
+    ok::query &q0 = qs[0];
+    for (int t = 1; t < nslices; t++)
+        q0.table.insert(q0.table.end(), qs[t].table.begin(), qs[t].table.end());
+
+    ok::final_process(q0, req, nreq, distinct, limit, offset);
+    int64_t R = q0.nrows();
+    *out_cols = q0.col_num;
+    *out = (uint32_t *)malloc(std::max<size_t>(q0.table.size() * 4, 4));
+    memcpy(*out, q0.table.data(), q0.table.size() * 4);
+    return R;
+}
+
+void ok_free_table(uint32_t *t) { free(t); }
+
+// ---- step-level API (mirrors the engine's five-call surface so the
+// distributed driver can be exercised on CPU with gloo) ----
+struct ok_query_handle {
+    ok::ctx *c;
+    ok::query q;
+};
+
+void *ok_query_begin(void *cv, const ok_pattern_t *pats, int32_t npat, int32_t nvars) {
+    ok_query_handle *h = new ok_query_handle();
+    h->c = (ok::ctx *)cv;
+    for (int i = 0; i < npat; i++)
+        h->q.pats.push_back({pats[i].subject, pats[i].predicate, pats[i].object,
+                             pats[i].direction});
+    h->q.nvars = nvars;
+    h->q.v2c.assign(nvars, -1);
+    return h;
+}
+
+void ok_query_load(void *hv, const uint32_t *table, int64_t nrows, int32_t ncols,
+                   const int32_t *v2c, int32_t step) {
+    ok_query_handle *h = (ok_query_handle *)hv;
+    h->q.table.assign(table, table + (size_t)nrows * ncols);
+    h->q.col_num = ncols;
+    h->q.step = step;
+    if (v2c) h->q.v2c.assign(v2c, v2c + h->q.nvars);
+}
+
+int64_t ok_query_step(void *hv) {
+    ok_query_handle *h = (ok_query_handle *)hv;
+    if (h->q.step >= (int)h->q.pats.size()) return -1;
+    ok::exec_pattern(*h->c, h->q);
+    return h->q.nrows();
+}
+
+int32_t ok_query_cols(void *hv) { return ((ok_query_handle *)hv)->q.col_num; }
+int32_t ok_query_stepno(void *hv) { return ((ok_query_handle *)hv)->q.step; }
+
+int64_t ok_query_table(void *hv, const uint32_t **data) {
+    ok_query_handle *h = (ok_query_handle *)hv;
+    *data = h->q.table.data();
+    return h->q.nrows();
+}
+
+void ok_query_finalize(void *hv, const int32_t *req, int32_t nreq,
+                       int32_t distinct, int64_t limit, int64_t offset) {
+    ok_query_handle *h = (ok_query_handle *)hv;
+    ok::final_process(h->q, req, nreq, distinct, limit, offset);
+}
+
+void ok_query_free(void *hv) { delete (ok_query_handle *)hv; }
+
+}  // extern "C"

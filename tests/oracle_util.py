@@ -1,0 +1,202 @@
+"""Test-side ctypes wrapper for oracle/liboracle.so (TEST INFRASTRUCTURE —
+the oracle may only be touched from tests/, smoke() and bench.py's
+cpu_baseline leg)."""
+import ctypes
+import os
+
+import numpy as np
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+SO = os.path.join(HERE, "..", "oracle", "liboracle.so")
+
+_lib = None
+
+
+def lib():
+    global _lib
+    if _lib is None:
+        if not os.path.exists(SO):
+            raise RuntimeError("oracle not built: make -C oracle")
+        _lib = ctypes.CDLL(SO)
+        u32p = ctypes.POINTER(ctypes.c_uint32)
+        i32, i64, u64, vp = (ctypes.c_int32, ctypes.c_int64,
+                             ctypes.c_uint64, ctypes.c_void_p)
+        _lib.ok_build.restype = vp
+        _lib.ok_build.argtypes = [u32p, i64, i32, i32]
+        _lib.ok_free.argtypes = [vp]
+        _lib.ok_get_triples.restype = u32p
+        _lib.ok_get_triples.argtypes = [vp, ctypes.c_uint32, ctypes.c_uint32,
+                                        i32, ctypes.POINTER(u64)]
+        _lib.ok_get_index.restype = u32p
+        _lib.ok_get_index.argtypes = [vp, ctypes.c_uint32, i32, ctypes.POINTER(u64)]
+        _lib.ok_hash_u64.restype = u64
+        _lib.ok_hash_u64.argtypes = [u64]
+        _lib.ok_key_pack.restype = u64
+        _lib.ok_key_pack.argtypes = [u64, u64, u64]
+        _lib.ok_run_query.restype = i64
+        _lib.ok_run_query.argtypes = [vp, ctypes.c_void_p, i32, i32,
+                                      ctypes.POINTER(i32), i32, i32, i64, i64,
+                                      i32, ctypes.POINTER(u32p),
+                                      ctypes.POINTER(i32)]
+        _lib.ok_brute_query.restype = i64
+        _lib.ok_brute_query.argtypes = [vp, ctypes.c_void_p, i32, i32,
+                                        ctypes.POINTER(i32), i32,
+                                        ctypes.POINTER(u32p),
+                                        ctypes.POINTER(i32)]
+        _lib.ok_free_table.argtypes = [u32p]
+        _lib.ok_query_begin.restype = vp
+        _lib.ok_query_begin.argtypes = [vp, ctypes.c_void_p, i32, i32]
+        _lib.ok_query_load.argtypes = [vp, u32p, i64, i32, ctypes.POINTER(i32), i32]
+        _lib.ok_query_step.restype = i64
+        _lib.ok_query_step.argtypes = [vp]
+        _lib.ok_query_cols.restype = i32
+        _lib.ok_query_cols.argtypes = [vp]
+        _lib.ok_query_stepno.restype = i32
+        _lib.ok_query_stepno.argtypes = [vp]
+        _lib.ok_query_table.restype = i64
+        _lib.ok_query_table.argtypes = [vp, ctypes.POINTER(u32p)]
+        _lib.ok_query_finalize.argtypes = [vp, ctypes.POINTER(i32), i32, i32, i64, i64]
+        _lib.ok_query_free.argtypes = [vp]
+    return _lib
+
+
+class OkPattern(ctypes.Structure):
+    _fields_ = [("subject", ctypes.c_int32), ("predicate", ctypes.c_int32),
+                ("object", ctypes.c_int32), ("direction", ctypes.c_int32)]
+
+
+def sort_rows(a):
+    """Lexicographic row sort — the multiset-equality canonical form."""
+    a = np.asarray(a)
+    if a.size == 0:
+        return a
+    return a[np.lexsort(a.T[::-1])]
+
+
+def _pats(plan):
+    arr = (OkPattern * len(plan.patterns))()
+    for i, (s, p, d, o) in enumerate(plan.patterns):
+        arr[i] = OkPattern(s, p, o, d)
+    return arr
+
+
+class OracleCtx:
+    def __init__(self, triples, sid=0, nsrv=1):
+        t = np.ascontiguousarray(triples, dtype=np.uint32)
+        self._h = lib().ok_build(
+            t.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)), t.shape[0], sid, nsrv)
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            lib().ok_free(self._h)
+            self._h = None
+
+    def get_triples(self, vid, pid, direction):
+        sz = ctypes.c_uint64()
+        p = lib().ok_get_triples(self._h, vid, pid, direction, ctypes.byref(sz))
+        if not p or sz.value == 0:
+            return np.empty(0, dtype=np.uint32)
+        return np.ctypeslib.as_array(p, shape=(sz.value,)).copy()
+
+    def get_index(self, pid, direction):
+        sz = ctypes.c_uint64()
+        p = lib().ok_get_index(self._h, pid, direction, ctypes.byref(sz))
+        if not p or sz.value == 0:
+            return np.empty(0, dtype=np.uint32)
+        return np.ctypeslib.as_array(p, shape=(sz.value,)).copy()
+
+    def run_query(self, plan, mt=1):
+        pats = _pats(plan)
+        req = (ctypes.c_int32 * len(plan.required_vars))(*plan.required_vars)
+        out = ctypes.POINTER(ctypes.c_uint32)()
+        cols = ctypes.c_int32()
+        n = lib().ok_run_query(self._h, ctypes.cast(pats, ctypes.c_void_p),
+                               len(plan.patterns), plan.nvars, req,
+                               len(plan.required_vars),
+                               1 if plan.distinct else 0, plan.limit,
+                               plan.offset, mt, ctypes.byref(out),
+                               ctypes.byref(cols))
+        if n and cols.value:
+            tbl = np.ctypeslib.as_array(out, shape=(n, cols.value)).copy()
+        else:
+            tbl = np.empty((0, cols.value), dtype=np.uint32)
+        lib().ok_free_table(out)
+        return tbl
+
+    def brute_query(self, plan):
+        pats = _pats(plan)
+        req = (ctypes.c_int32 * len(plan.required_vars))(*plan.required_vars)
+        out = ctypes.POINTER(ctypes.c_uint32)()
+        cols = ctypes.c_int32()
+        n = lib().ok_brute_query(self._h, ctypes.cast(pats, ctypes.c_void_p),
+                                 len(plan.patterns), plan.nvars, req,
+                                 len(plan.required_vars), ctypes.byref(out),
+                                 ctypes.byref(cols))
+        if n and cols.value:
+            tbl = np.ctypeslib.as_array(out, shape=(n, cols.value)).copy()
+        else:
+            tbl = np.empty((0, cols.value), dtype=np.uint32)
+        lib().ok_free_table(out)
+        return tbl
+
+
+class OracleExecutor:
+    """Step-level executor over the oracle (the dist-driver interface:
+    begin/load/step/table/col_num/step_no/finalize) — CPU stand-in for
+    wukong_amd.Engine in the gloo multi-process tests."""
+
+    def __init__(self, ctx, plan):
+        self.ctx = ctx
+        self.plan = plan
+        self.npat = len(plan.patterns)
+        pats = _pats(plan)
+        self._keep = pats
+        self._h = lib().ok_query_begin(ctx._h, ctypes.cast(pats, ctypes.c_void_p),
+                                       self.npat, plan.nvars)
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            lib().ok_query_free(self._h)
+            self._h = None
+
+    def load(self, table, v2c, step):
+        t = np.ascontiguousarray(table, dtype=np.uint32)
+        v = (ctypes.c_int32 * len(v2c))(*v2c)
+        lib().ok_query_load(self._h,
+                            t.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)),
+                            t.shape[0] if t.size else 0,
+                            t.shape[1] if t.ndim == 2 else 0, v, step)
+
+    def step(self):
+        return lib().ok_query_step(self._h)
+
+    @property
+    def col_num(self):
+        return lib().ok_query_cols(self._h)
+
+    @property
+    def step_no(self):
+        return lib().ok_query_stepno(self._h)
+
+    def table(self):
+        data = ctypes.POINTER(ctypes.c_uint32)()
+        n = lib().ok_query_table(self._h, ctypes.byref(data))
+        c = self.col_num
+        if n and c:
+            return np.ctypeslib.as_array(data, shape=(n, c)).copy()
+        return np.empty((0, c), dtype=np.uint32)
+
+    def finalize(self):
+        p = self.plan
+        req = (ctypes.c_int32 * len(p.required_vars))(*p.required_vars)
+        lib().ok_query_finalize(self._h, req, len(p.required_vars),
+                                1 if p.distinct else 0, p.limit, p.offset)
+        return self.table()
+
+
+def hash_u64(x):
+    return lib().ok_hash_u64(x)
+
+
+def key_pack(v, p, d):
+    return lib().ok_key_pack(v, p, d)

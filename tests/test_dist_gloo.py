@@ -1,0 +1,63 @@
+"""Multi-process (world_size=2, gloo) test of the distributed driver:
+2 partitioned oracle executors + all-to-all exchange must equal the
+1-partition oracle.  Runs on CPU."""
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+QUERIES = ["q1", "q2", "q3", "q4", "q5", "q6", "q7"]
+
+
+def _worker(rank, world, port, results):
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch.distributed as dist
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from wukong_amd.dist import DistQuery
+    from tests.oracle_util import OracleCtx, OracleExecutor, sort_rows
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        triples = wk.lubm_gen(2, seed=42, sid=rank, nsrv=world)
+        ctx = OracleCtx(triples, sid=rank, nsrv=world)
+        out = {}
+        for name in QUERIES:
+            plan = Q.ALL[name]
+            ex = OracleExecutor(ctx, plan)
+            dq = DistQuery(ex, plan, rank, world)
+            dq.run()
+            merged = dq.gather_result()
+            out[name] = sort_rows(merged)
+        if rank == 0:
+            results.put(out)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_dist_two_ranks_equal_single():
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    port = 29871
+    procs = [ctx.Process(target=_worker, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=540)
+    for p in procs:
+        p.join(timeout=60)
+
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from tests.oracle_util import OracleCtx, sort_rows
+    full = OracleCtx(wk.lubm_gen(2, seed=42))
+    for name in QUERIES:
+        want = sort_rows(full.run_query(Q.ALL[name]))
+        assert got[name].shape == want.shape, (name, got[name].shape, want.shape)
+        assert np.array_equal(got[name], want), name

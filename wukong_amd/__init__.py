@@ -1,0 +1,289 @@
+"""wukong_amd — MI355X-native SPARQL graph-exploration engine.
+
+Python is plumbing only: ctypes over the C-ABI boundary declared in
+include/wukong_abi.h.  The compute path is the HIP
+library; there is NO Python/CPU fallback — a missing .so raises.
+"""
+import ctypes
+import os
+
+import numpy as np
+
+_HERE = os.path.dirname(os.path.abspath(__file__))
+_SO = os.path.join(_HERE, "libwukong_hip.so")
+
+if not os.path.exists(_SO):
+    raise ImportError(
+        f"wukong_amd: native engine {_SO} not built. "
+        "Run `python -m wukong_amd.build` (or __graft_entry__.build())."
+    )
+_lib = ctypes.CDLL(_SO)
+
+
+class WkPattern(ctypes.Structure):
+    _fields_ = [("subject", ctypes.c_int32), ("predicate", ctypes.c_int32),
+                ("object", ctypes.c_int32), ("direction", ctypes.c_int32)]
+
+
+class WkPlan(ctypes.Structure):
+    _fields_ = [("patterns", ctypes.POINTER(WkPattern)),
+                ("npatterns", ctypes.c_int32),
+                ("nvars", ctypes.c_int32),
+                ("required_vars", ctypes.POINTER(ctypes.c_int32)),
+                ("nrequired", ctypes.c_int32),
+                ("distinct", ctypes.c_int32),
+                ("limit", ctypes.c_int64),
+                ("offset", ctypes.c_int64)]
+
+
+class WkResult(ctypes.Structure):
+    _fields_ = [("col_num", ctypes.c_int32), ("row_num", ctypes.c_int64),
+                ("table", ctypes.POINTER(ctypes.c_uint32)),
+                ("status_code", ctypes.c_int32)]
+
+
+def _sig(name, res, args):
+    f = getattr(_lib, name)
+    f.restype = res
+    f.argtypes = args
+    return f
+
+c_u32p = ctypes.POINTER(ctypes.c_uint32)
+c_i32 = ctypes.c_int32
+c_i64 = ctypes.c_int64
+c_u64 = ctypes.c_uint64
+c_vp = ctypes.c_void_p
+
+_lubm_gen = _sig("wk_lubm_gen", c_i64, [c_i32, c_u64, c_i32, c_i32, ctypes.POINTER(c_u32p)])
+_free_triples = _sig("wk_free_triples", None, [c_u32p])
+_store_build = _sig("wk_store_build", c_vp, [c_u32p, c_i64, c_i32, c_i32])
+_store_free = _sig("wk_store_free", None, [c_vp])
+_get_triples = _sig("wk_store_get_triples", c_u32p, [c_vp, ctypes.c_uint32, ctypes.c_uint32, c_i32, ctypes.POINTER(c_u64)])
+_get_index = _sig("wk_store_get_index", c_u32p, [c_vp, ctypes.c_uint32, c_i32, ctypes.POINTER(c_u64)])
+_num_slots = _sig("wk_store_num_slots", c_u64, [c_vp])
+_num_edges = _sig("wk_store_num_edges", c_u64, [c_vp])
+_checksum = _sig("wk_store_checksum", c_u64, [c_vp])
+_eng_create = _sig("wk_engine_create", c_vp, [c_vp, c_i32])
+_eng_destroy = _sig("wk_engine_destroy", None, [c_vp])
+_eng_run = _sig("wk_engine_run_query", c_i32, [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(WkResult)])
+_eng_begin = _sig("wk_engine_begin_query", c_i32, [c_vp, ctypes.POINTER(WkPlan)])
+_eng_load = _sig("wk_engine_load_rbuf", c_i32, [c_vp, c_u32p, c_i64, c_i32, ctypes.POINTER(c_i32), c_i32])
+_eng_load_dev = _sig("wk_engine_load_rbuf_device", c_i32, [c_vp, c_vp, c_i64, c_i32, ctypes.POINTER(c_i32), c_i32])
+_eng_step = _sig("wk_engine_execute_one_pattern", c_i32, [c_vp, ctypes.POINTER(c_i64)])
+_eng_pattern_step = _sig("wk_engine_pattern_step", c_i32, [c_vp])
+_eng_col_num = _sig("wk_engine_col_num", c_i32, [c_vp])
+_eng_subq = _sig("wk_engine_generate_sub_query", c_i32, [c_vp, c_i32, c_vp, c_i64, ctypes.POINTER(c_i64)])
+_eng_fetch = _sig("wk_engine_fetch_result", c_i32, [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(WkResult)])
+_eng_fetch_raw = _sig("wk_engine_fetch_raw", c_i32, [c_vp, ctypes.POINTER(WkResult)])
+_res_free = _sig("wk_result_free", None, [ctypes.POINTER(WkResult)])
+_kstats = _sig("wk_engine_kernel_stats", c_i32, [c_vp, ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_double), ctypes.POINTER(c_i64)])
+_arch = _sig("wk_build_arch", ctypes.c_char_p, [])
+_devcount = _sig("wk_device_count", c_i32, [])
+hash_u64 = _sig("wk_hash_u64", c_u64, [c_u64])
+key_pack = _sig("wk_key_pack", c_u64, [c_u64, c_u64, c_u64])
+ptr_pack = _sig("wk_ptr_pack", c_u64, [c_u64, c_u64, c_u64])
+
+DIR_IN, DIR_OUT = 0, 1
+KERNEL_CATS = ["probe", "scan", "expand", "filter", "copy", "split", "other"]
+
+
+def build_arch():
+    return _arch().decode()
+
+
+def device_count():
+    return _devcount()
+
+
+def lubm_gen(nuniv, seed=42, sid=0, nsrv=1):
+    """Seeded LUBM-shaped synthetic ID-triples, partition (sid of nsrv)."""
+    out = c_u32p()
+    n = _lubm_gen(nuniv, seed, sid, nsrv, ctypes.byref(out))
+    if n < 0:
+        raise RuntimeError("wk_lubm_gen failed")
+    arr = np.ctypeslib.as_array(out, shape=(n, 3)).copy()
+    _free_triples(out)
+    return arr
+
+
+class Plan:
+    """A planner-ordered pattern list (vars negative: -1..-nvars)."""
+
+    def __init__(self, patterns, nvars, required_vars, distinct=False,
+                 limit=-1, offset=0):
+        self.patterns = list(patterns)
+        self.nvars = nvars
+        self.required_vars = list(required_vars)
+        self.distinct = distinct
+        self.limit = limit
+        self.offset = offset
+
+    def to_c(self):
+        pats = (WkPattern * len(self.patterns))()
+        for i, (s, p, d, o) in enumerate(self.patterns):
+            pats[i] = WkPattern(s, p, o, d)  # note arg order (s,p,d,o) -> struct
+        req = (ctypes.c_int32 * len(self.required_vars))(*self.required_vars)
+        plan = WkPlan(ctypes.cast(pats, ctypes.POINTER(WkPattern)),
+                      len(self.patterns), self.nvars,
+                      ctypes.cast(req, ctypes.POINTER(ctypes.c_int32)),
+                      len(self.required_vars), 1 if self.distinct else 0,
+                      self.limit, self.offset)
+        plan._keepalive = (pats, req)
+        return plan
+
+
+class Store:
+    def __init__(self, triples, sid=0, nsrv=1):
+        t = np.ascontiguousarray(triples, dtype=np.uint32)
+        assert t.ndim == 2 and t.shape[1] == 3
+        self._h = _store_build(t.ctypes.data_as(c_u32p), t.shape[0], sid, nsrv)
+        if not self._h:
+            raise RuntimeError("wk_store_build failed")
+        self.sid, self.nsrv = sid, nsrv
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            _store_free(self._h)
+            self._h = None
+
+    def get_triples(self, vid, pid, direction):
+        sz = c_u64()
+        p = _get_triples(self._h, vid, pid, direction, ctypes.byref(sz))
+        if not p or sz.value == 0:
+            return np.empty(0, dtype=np.uint32)
+        return np.ctypeslib.as_array(p, shape=(sz.value,)).copy()
+
+    def get_index(self, pid, direction):
+        sz = c_u64()
+        p = _get_index(self._h, pid, direction, ctypes.byref(sz))
+        if not p or sz.value == 0:
+            return np.empty(0, dtype=np.uint32)
+        return np.ctypeslib.as_array(p, shape=(sz.value,)).copy()
+
+    @property
+    def num_slots(self):
+        return _num_slots(self._h)
+
+    @property
+    def num_edges(self):
+        return _num_edges(self._h)
+
+    def checksum(self):
+        return _checksum(self._h)
+
+
+class Engine:
+    """GPU engine bound to one device; requires a GPU (fails loudly)."""
+
+    def __init__(self, store, device=0):
+        self._store = store  # keepalive
+        self._h = _eng_create(store._h, device)
+        if not self._h:
+            raise RuntimeError(
+                "wk_engine_create failed (no MI355X visible, or HBM alloc failed)")
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            _eng_destroy(self._h)
+            self._h = None
+
+    def run_query(self, plan):
+        cplan = plan.to_c()
+        res = WkResult()
+        rc = _eng_run(self._h, ctypes.byref(cplan), ctypes.byref(res))
+        if rc != 0:
+            raise RuntimeError(f"wk_engine_run_query rc={rc}")
+        try:
+            if res.row_num and res.col_num:
+                tbl = np.ctypeslib.as_array(
+                    res.table, shape=(res.row_num, res.col_num)).copy()
+            else:
+                tbl = np.empty((0, res.col_num), dtype=np.uint32)
+        finally:
+            _res_free(ctypes.byref(res))
+        return tbl
+
+    # step-level API (multi-GPU driver)
+    def begin_query(self, plan):
+        cplan = plan.to_c()
+        rc = _eng_begin(self._h, ctypes.byref(cplan))
+        if rc != 0:
+            raise RuntimeError(f"begin_query rc={rc}")
+        self._cplan = cplan
+
+    def load_rbuf(self, table, v2c, step):
+        t = np.ascontiguousarray(table, dtype=np.uint32)
+        v = (ctypes.c_int32 * len(v2c))(*v2c)
+        rc = _eng_load(self._h, t.ctypes.data_as(c_u32p),
+                       t.shape[0] if t.size else 0,
+                       t.shape[1] if t.ndim == 2 else 0, v, step)
+        if rc != 0:
+            raise RuntimeError(f"load_rbuf rc={rc}")
+
+    def load_rbuf_device(self, dev_ptr, nrows, ncols, v2c, step):
+        v = (ctypes.c_int32 * len(v2c))(*v2c)
+        rc = _eng_load_dev(self._h, dev_ptr, nrows, ncols, v, step)
+        if rc != 0:
+            raise RuntimeError(f"load_rbuf_device rc={rc}")
+
+    def execute_one_pattern(self):
+        n = c_i64()
+        rc = _eng_step(self._h, ctypes.byref(n))
+        if rc != 0:
+            raise RuntimeError(f"execute_one_pattern rc={rc}")
+        return n.value
+
+    @property
+    def pattern_step(self):
+        return _eng_pattern_step(self._h)
+
+    @property
+    def col_num(self):
+        return _eng_col_num(self._h)
+
+    def generate_sub_query(self, ndst, dev_out_ptr, cap_rows):
+        rows = (c_i64 * ndst)()
+        rc = _eng_subq(self._h, ndst, dev_out_ptr, cap_rows, rows)
+        if rc != 0:
+            raise RuntimeError(f"generate_sub_query rc={rc}")
+        return list(rows)
+
+    def fetch_result(self, plan=None):
+        cplan = self._cplan if plan is None else plan.to_c()
+        res = WkResult()
+        rc = _eng_fetch(self._h, ctypes.byref(cplan), ctypes.byref(res))
+        if rc != 0:
+            raise RuntimeError(f"fetch_result rc={rc}")
+        try:
+            if res.row_num and res.col_num:
+                tbl = np.ctypeslib.as_array(
+                    res.table, shape=(res.row_num, res.col_num)).copy()
+            else:
+                tbl = np.empty((0, res.col_num), dtype=np.uint32)
+        finally:
+            _res_free(ctypes.byref(res))
+        return tbl
+
+    def fetch_raw(self):
+        """Current device table, no final ops (driver/tests)."""
+        res = WkResult()
+        rc = _eng_fetch_raw(self._h, ctypes.byref(res))
+        if rc != 0:
+            raise RuntimeError(f"fetch_raw rc={rc}")
+        try:
+            if res.row_num and res.col_num:
+                tbl = np.ctypeslib.as_array(
+                    res.table, shape=(res.row_num, res.col_num)).copy()
+            else:
+                tbl = np.empty((0, res.col_num), dtype=np.uint32)
+        finally:
+            _res_free(ctypes.byref(res))
+        return tbl
+
+    def kernel_stats(self):
+        us = (ctypes.c_double * 7)()
+        by = (ctypes.c_double * 7)()
+        ln = (c_i64 * 7)()
+        _kstats(self._h, us, by, ln)
+        return {KERNEL_CATS[i]: dict(usec=us[i], bytes=by[i], launches=ln[i])
+                for i in range(7)}

@@ -1,0 +1,347 @@
+/*
+ * store.cpp — host-side build of the HBM-resident predicate-segmented
+ * cluster-hash store.
+ *
+ * Restates (not copies) the reference build pipeline:
+ *  - partition + sort pso/pos + dedup: core/loader/base_loader.hpp:302-373
+ *  - per-segment key/edge insertion:   core/store/static_gstore.hpp:64-161
+ *  - index collection + insertion:     core/store/gstore.hpp:858-888,
+ *                                      static_gstore.hpp:217-280
+ *  - cluster-hash insert/probe:        core/store/gstore.hpp:789-856,341-361
+ * Bucket allocation policy (per-segment ceil(keys/5) main buckets + shared
+ * ext region) is ours — DESIGN.md §2.
+ */
+#include "wk_store.h"
+#include "../../include/wukong_abi.h"
+#include <algorithm>
+#include <atomic>
+#include <cstring>
+#include <cstdio>
+#include <omp.h>
+
+using namespace wk;
+
+namespace {
+
+struct triple { sid_t s, p, o; };
+
+// parallel counting sort into buckets keyed by (p, id>>17), then per-bucket
+// std::sort — preserves full (p, major, minor) order because id>>17 is the
+// major prefix of the secondary key.
+static void sort_triples(std::vector<triple> &t, bool by_s /* pso vs pos */,
+                         uint32_t max_pid, uint32_t max_blk) {
+    const uint64_t nb = (uint64_t)(max_pid + 1) * (max_blk + 2);
+    const int64_t n = (int64_t)t.size();
+    auto bucket_of = [&](const triple &x) -> uint64_t {
+        uint32_t major = by_s ? x.s : x.o;
+        return (uint64_t)x.p * (max_blk + 2) + (major >> NBITS_IDX);
+    };
+    int nthr = omp_get_max_threads();
+    std::vector<uint64_t> hist((size_t)nthr * nb, 0);
+#pragma omp parallel num_threads(nthr)
+    {
+        int tid = omp_get_thread_num();
+        uint64_t *h = hist.data() + (size_t)tid * nb;
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n; i++) h[bucket_of(t[i])]++;
+    }
+    // exclusive offsets: bucket-major, thread-minor
+    std::vector<uint64_t> bucket_off(nb + 1, 0);
+    {
+        uint64_t acc = 0;
+        for (uint64_t b = 0; b < nb; b++) {
+            bucket_off[b] = acc;
+            for (int k = 0; k < nthr; k++) {
+                uint64_t c = hist[(size_t)k * nb + b];
+                hist[(size_t)k * nb + b] = acc;
+                acc += c;
+            }
+        }
+        bucket_off[nb] = acc;
+    }
+    std::vector<triple> out(t.size());
+#pragma omp parallel num_threads(nthr)
+    {
+        int tid = omp_get_thread_num();
+        uint64_t *h = hist.data() + (size_t)tid * nb;
+#pragma omp for schedule(static)
+        for (int64_t i = 0; i < n; i++) out[h[bucket_of(t[i])]++] = t[i];
+    }
+    t.swap(out);
+    // per-bucket comparison sort on the minor keys
+    auto cmp_pso = [](const triple &a, const triple &b) {
+        return a.s != b.s ? a.s < b.s : a.o < b.o;
+    };
+    auto cmp_pos = [](const triple &a, const triple &b) {
+        return a.o != b.o ? a.o < b.o : a.s < b.s;
+    };
+#pragma omp parallel for schedule(dynamic, 16)
+    for (uint64_t b = 0; b < nb; b++) {
+        auto *lo = t.data() + bucket_off[b], *hi = t.data() + bucket_off[b + 1];
+        if (hi - lo > 1) {
+            if (by_s) std::sort(lo, hi, cmp_pso);
+            else      std::sort(lo, hi, cmp_pos);
+        }
+    }
+}
+
+static void dedup_triples(std::vector<triple> &t) {
+    auto eq = [](const triple &a, const triple &b) {
+        return a.s == b.s && a.p == b.p && a.o == b.o;
+    };
+    auto end = std::unique(t.begin(), t.end(), eq);
+    t.erase(end, t.end());
+}
+
+// cluster-hash inserter for one segment (single-threaded per segment;
+// segments insert in parallel).  Mirrors GStore::insert_key
+// (gstore.hpp:789-856) with ext buckets drawn from a shared atomic region.
+struct inserter {
+    wk_store &st;
+    const seg_t &seg;
+    std::atomic<uint64_t> &ext_next;
+
+    uint64_t insert(uint64_t key) {
+        uint64_t bucket = seg.bucket_start + hash_u64(key) % seg.num_buckets;
+        while (true) {
+            vertex_t *b = &st.vertices[bucket * ASSOC];
+            for (int i = 0; i < ASSOC - 1; i++) {
+                if (b[i].key == key) { fprintf(stderr, "dup key!\n"); abort(); }
+                if (b[i].key == KEY_EMPTY) { b[i].key = key; return bucket * ASSOC + i; }
+            }
+            if (b[ASSOC - 1].key != KEY_EMPTY) {
+                bucket = key_vid(b[ASSOC - 1].key);
+                continue;
+            }
+            uint64_t nb = ext_next.fetch_add(1);
+            if (nb >= st.nbuckets_main + st.nbuckets_ext) {
+                fprintf(stderr, "out of ext buckets\n"); abort();
+            }
+            b[ASSOC - 1].key = key_pack(nb, 0, 0);
+            bucket = nb;
+        }
+    }
+};
+
+}  // namespace
+
+extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
+                                      int32_t sid, int32_t nsrv) {
+    if (!spo || ntriples < 0 || nsrv <= 0 || sid < 0 || sid >= nsrv) return nullptr;
+    wk_store *st = new wk_store();
+    st->sid = sid; st->nsrv = nsrv;
+
+    // 1. partition into pso (s%n==sid) / pos (o%n==sid) — base_loader.hpp:344-352
+    std::vector<triple> pso, pos;
+    pso.reserve(ntriples); pos.reserve(ntriples);
+    uint32_t max_pid = 1, max_id = 0;
+    for (int64_t i = 0; i < ntriples; i++) {
+        triple t{spo[3 * i], spo[3 * i + 1], spo[3 * i + 2]};
+        max_pid = std::max(max_pid, t.p);
+        if (t.p == TYPE_ID) max_pid = std::max(max_pid, t.o);  // types are index ids
+        max_id = std::max({max_id, t.s, t.o});
+        if ((int32_t)(t.s % (sid_t)nsrv) == sid) pso.push_back(t);
+        if ((int32_t)(t.o % (sid_t)nsrv) == sid) pos.push_back(t);
+    }
+    if (max_pid >= (1u << NBITS_IDX)) { delete st; return nullptr; }
+    st->max_pid = max_pid;
+    uint32_t max_blk = max_id >> NBITS_IDX;
+
+    // 2. sort + dedup — base_loader.hpp:367-377
+    sort_triples(pso, true, max_pid, max_blk);
+    sort_triples(pos, false, max_pid, max_blk);
+    dedup_triples(pso);
+    dedup_triples(pos);
+
+    // 3. pid slice boundaries in each sorted array
+    const uint32_t NP = max_pid + 1;
+    std::vector<int64_t> pso_lo(NP + 1, 0), pos_lo(NP + 1, 0);
+    {
+        int64_t i = 0;
+        for (uint32_t p = 0; p <= max_pid; p++) {
+            pso_lo[p] = i;
+            while (i < (int64_t)pso.size() && pso[i].p == p) i++;
+        }
+        pso_lo[NP] = (int64_t)pso.size();
+        // pso is bucket-sorted by p ascending, so slices are contiguous —
+        // but only if every element was consumed:
+        if (i != (int64_t)pso.size()) { fprintf(stderr, "pso slice error\n"); abort(); }
+        i = 0;
+        for (uint32_t p = 0; p <= max_pid; p++) {
+            pos_lo[p] = i;
+            while (i < (int64_t)pos.size() && pos[i].p == p) i++;
+        }
+        pos_lo[NP] = (int64_t)pos.size();
+        if (i != (int64_t)pos.size()) { fprintf(stderr, "pos slice error\n"); abort(); }
+    }
+
+    // 4. count keys/edges per segment.
+    //    OUT seg of pid: (s,p) runs in pso slice.
+    //    IN  seg of pid: (o,p) runs in pos slice, skipping is_tpid(o) runs
+    //    (type triples are never normal IN keys — static_gstore.hpp:128-131).
+    std::vector<uint64_t> out_keys(NP, 0), out_edges(NP, 0), in_keys(NP, 0), in_edges(NP, 0);
+#pragma omp parallel for schedule(dynamic)
+    for (uint32_t p = 0; p <= max_pid; p++) {
+        for (int64_t i = pso_lo[p]; i < pso_lo[p + 1];) {
+            int64_t j = i + 1;
+            while (j < pso_lo[p + 1] && pso[j].s == pso[i].s) j++;
+            out_keys[p]++; out_edges[p] += (uint64_t)(j - i);
+            i = j;
+        }
+        for (int64_t i = pos_lo[p]; i < pos_lo[p + 1];) {
+            int64_t j = i + 1;
+            while (j < pos_lo[p + 1] && pos[j].o == pos[i].o) j++;
+            if (!is_tpid(pos[i].o)) { in_keys[p]++; in_edges[p] += (uint64_t)(j - i); }
+            i = j;
+        }
+    }
+
+    // index segment contents (gstore.hpp:858-888):
+    //   [0|pid|IN]  = subjects of pid (from OUT keys, pid != TYPE_ID)
+    //   [0|pid|OUT] = objects  of pid (from IN keys)
+    //   [0|tid|IN]  = type members (from [vid|TYPE_ID|OUT] edges)
+    uint64_t idx_in_keys = 0, idx_in_edges = 0, idx_out_keys = 0, idx_out_edges = 0;
+    std::vector<uint64_t> type_members(NP, 0);
+    for (int64_t i = pso_lo[TYPE_ID]; i < pso_lo[TYPE_ID + 1]; i++)
+        type_members[pso[i].o]++;
+    for (uint32_t p = 2; p <= max_pid; p++) {
+        if (out_keys[p]) { idx_in_keys++; idx_in_edges += out_keys[p]; }
+        if (in_keys[p])  { idx_out_keys++; idx_out_edges += in_keys[p]; }
+        if (type_members[p]) { idx_in_keys++; idx_in_edges += type_members[p]; }
+    }
+    // [vid|TYPE_ID|OUT] keys exist but contribute nothing to pidx maps
+    // (collect_idx_info, gstore.hpp:873-887)
+
+    // 5. allocate segments: buckets + edge offsets
+    st->nseg.assign((size_t)NP * 2, seg_t{});
+    uint64_t bucket_cursor = 0, edge_cursor = 0;
+    std::vector<uint64_t> seg_edge_start((size_t)NP * 2, 0);
+    uint64_t iseg_edge_start[2] = {0, 0};
+    auto alloc_seg = [&](seg_t &sg, uint64_t keys) {
+        if (!keys) return;
+        sg.bucket_start = bucket_cursor;
+        sg.num_buckets = (keys + 4) / 5;
+        bucket_cursor += sg.num_buckets;
+    };
+    for (uint32_t p = 0; p <= max_pid; p++) {
+        alloc_seg(st->nseg[p * 2 + DIR_OUT], out_keys[p]);
+        seg_edge_start[p * 2 + DIR_OUT] = edge_cursor; edge_cursor += out_edges[p];
+        alloc_seg(st->nseg[p * 2 + DIR_IN], in_keys[p]);
+        seg_edge_start[p * 2 + DIR_IN] = edge_cursor; edge_cursor += in_edges[p];
+    }
+    alloc_seg(st->iseg[DIR_IN], idx_in_keys);
+    iseg_edge_start[DIR_IN] = edge_cursor; edge_cursor += idx_in_edges;
+    alloc_seg(st->iseg[DIR_OUT], idx_out_keys);
+    iseg_edge_start[DIR_OUT] = edge_cursor; edge_cursor += idx_out_edges;
+
+    st->nbuckets_main = bucket_cursor;
+    st->nbuckets_ext = bucket_cursor / 4 + 1024;
+    st->vertices.assign((st->nbuckets_main + st->nbuckets_ext) * ASSOC, vertex_t{0, 0});
+    st->edges.resize(edge_cursor);
+    std::atomic<uint64_t> ext_next(st->nbuckets_main);
+
+    // 6. insert normal segments (parallel over (pid,dir) segments —
+    //    static_gstore.hpp:64-161 semantics: one key per (vid,pid,dir) run,
+    //    edge list = run values in sorted order)
+#pragma omp parallel for schedule(dynamic)
+    for (uint32_t w = 0; w < NP * 2; w++) {
+        uint32_t p = w / 2;
+        int dir = (int)(w & 1);
+        bool out = dir == DIR_OUT;
+        const seg_t &sg = st->nseg[w];
+        if (!sg.num_buckets) continue;
+        inserter ins{*st, sg, ext_next};
+        uint64_t off = seg_edge_start[w];
+        const std::vector<triple> &arr = out ? pso : pos;
+        int64_t lo = out ? pso_lo[p] : pos_lo[p];
+        int64_t hi = out ? pso_lo[p + 1] : pos_lo[p + 1];
+        for (int64_t i = lo; i < hi;) {
+            sid_t v = out ? arr[i].s : arr[i].o;
+            int64_t j = i + 1;
+            while (j < hi && (out ? arr[j].s : arr[j].o) == v) j++;
+            if (!out && is_tpid(v)) { i = j; continue; }
+            uint64_t slot = ins.insert(key_pack(v, p, (uint64_t)dir));
+            st->vertices[slot].ptr = ptr_pack((uint64_t)(j - i), off);
+            for (int64_t k = i; k < j; k++)
+                st->edges[off++] = out ? arr[k].o : arr[k].s;
+            i = j;
+        }
+    }
+
+    // 7. insert index segments (insert_idx, static_gstore.hpp:217-280;
+    //    ours iterates pids ascending — deterministic where the reference's
+    //    TBB iteration order was not; parity is set-level)
+    {
+        // IN: predicate-index (subjects) + type index
+        if (st->iseg[DIR_IN].num_buckets) {
+            inserter ins{*st, st->iseg[DIR_IN], ext_next};
+            uint64_t off = iseg_edge_start[DIR_IN];
+            for (uint32_t p = 2; p <= max_pid; p++) {
+                if (out_keys[p]) {
+                    uint64_t slot = ins.insert(key_pack(0, p, DIR_IN));
+                    st->vertices[slot].ptr = ptr_pack(out_keys[p], off);
+                    for (int64_t i = pso_lo[p]; i < pso_lo[p + 1];) {
+                        int64_t j = i + 1;
+                        while (j < pso_lo[p + 1] && pso[j].s == pso[i].s) j++;
+                        st->edges[off++] = pso[i].s;
+                        i = j;
+                    }
+                }
+                if (type_members[p]) {
+                    uint64_t slot = ins.insert(key_pack(0, p, DIR_IN));
+                    st->vertices[slot].ptr = ptr_pack(type_members[p], off);
+                    // members in subject order (type slice is (s,o)-sorted)
+                    for (int64_t i = pso_lo[TYPE_ID]; i < pso_lo[TYPE_ID + 1]; i++)
+                        if (pso[i].o == p) st->edges[off++] = pso[i].s;
+                }
+            }
+        }
+        // OUT: predicate-index (objects)
+        if (st->iseg[DIR_OUT].num_buckets) {
+            inserter ins{*st, st->iseg[DIR_OUT], ext_next};
+            uint64_t off = iseg_edge_start[DIR_OUT];
+            for (uint32_t p = 2; p <= max_pid; p++) {
+                if (!in_keys[p]) continue;
+                uint64_t slot = ins.insert(key_pack(0, p, DIR_OUT));
+                st->vertices[slot].ptr = ptr_pack(in_keys[p], off);
+                for (int64_t i = pos_lo[p]; i < pos_lo[p + 1];) {
+                    int64_t j = i + 1;
+                    while (j < pos_lo[p + 1] && pos[j].o == pos[i].o) j++;
+                    if (!is_tpid(pos[i].o)) st->edges[off++] = pos[i].o;
+                    i = j;
+                }
+            }
+        }
+    }
+    st->ext_used = ext_next.load() - st->nbuckets_main;
+    return st;
+}
+
+extern "C" void wk_store_free(wk_store_t *st) { delete st; }
+
+extern "C" const sid_t *wk_store_get_triples(const wk_store_t *st, sid_t vid,
+                                             sid_t pid, int32_t dir, uint64_t *sz) {
+    return store_get(*st, vid, pid, dir, sz);
+}
+
+extern "C" const sid_t *wk_store_get_index(const wk_store_t *st, sid_t pid,
+                                           int32_t dir, uint64_t *sz) {
+    return store_get(*st, 0, pid, dir, sz);
+}
+
+extern "C" uint64_t wk_store_num_slots(const wk_store_t *st) {
+    return st->vertices.size();
+}
+extern "C" uint64_t wk_store_num_edges(const wk_store_t *st) {
+    return st->edges.size();
+}
+extern "C" uint64_t wk_store_checksum(const wk_store_t *st) {
+    // FNV-1a over slots then edges
+    uint64_t h = 1469598103934665603ull;
+    auto mix = [&h](uint64_t x) {
+        for (int i = 0; i < 8; i++) { h ^= (x >> (8 * i)) & 0xff; h *= 1099511628211ull; }
+    };
+    for (const auto &v : st->vertices) { mix(v.key); mix(v.ptr); }
+    for (sid_t e : st->edges) mix(e);
+    return h;
+}

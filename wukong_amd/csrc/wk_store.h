@@ -1,0 +1,53 @@
+/*
+ * wk_store.h — host-side store object shared by the builder (store.cpp)
+ * and the GPU engine (gpu_engine.hip).  Layout semantics: DESIGN.md §2,
+ * restating core/store/gstore.hpp + static_gstore.hpp.
+ */
+#pragma once
+#include "wk_types.h"
+#include <vector>
+
+struct wk_store {
+    int32_t sid = 0, nsrv = 1;
+    uint32_t max_pid = 0;  // largest predicate/type id present
+
+    // cluster-hash slot array: (main + ext buckets) * 8 slots
+    std::vector<wk::vertex_t> vertices;
+    // all edge lists, contiguous per segment
+    std::vector<wk::sid_t> edges;
+    // normal segments indexed [pid*2 + dir], size (max_pid+1)*2
+    std::vector<wk::seg_t> nseg;
+    // index segments [dir]: keys [0|pid|dir] / [0|tid|IN]
+    wk::seg_t iseg[2];
+
+    uint64_t nbuckets_main = 0, nbuckets_ext = 0, ext_used = 0;
+
+    const wk::seg_t *seg_of(uint64_t vid, uint64_t pid, int dir) const {
+        if (vid == 0) return &iseg[dir];
+        if (pid > max_pid) return nullptr;
+        const wk::seg_t *s = &nseg[pid * 2 + dir];
+        return s->num_buckets ? s : nullptr;
+    }
+};
+
+// host probe — mirrors GStore::get_vertex_local (gstore.hpp:341-361)
+namespace wk {
+inline const sid_t *store_get(const wk_store &st, uint64_t vid, uint64_t pid,
+                              int dir, uint64_t *sz) {
+    *sz = 0;
+    const seg_t *seg = st.seg_of(vid, pid, dir);
+    if (!seg || seg->num_buckets == 0) return nullptr;
+    uint64_t key = key_pack(vid, pid, (uint64_t)dir);
+    uint64_t bucket = seg->bucket_start + hash_u64(key) % seg->num_buckets;
+    while (true) {
+        const vertex_t *b = &st.vertices[bucket * ASSOC];
+        for (int i = 0; i < ASSOC - 1; i++)
+            if (b[i].key == key) {
+                *sz = ptr_size(b[i].ptr);
+                return st.edges.data() + ptr_off(b[i].ptr);
+            }
+        if (b[ASSOC - 1].key == KEY_EMPTY) return nullptr;
+        bucket = key_vid(b[ASSOC - 1].key);  // chain (gstore.hpp:826)
+    }
+}
+}  // namespace wk

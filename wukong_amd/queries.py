@@ -1,0 +1,89 @@
+"""LUBM Q1-Q7 plans, restated from the reference's OSDI16 benchmark
+configuration (scripts/sparql_query/lubm/basic/lubm_q{1..7} +
+osdi16_plan/*.fmt).  Plan-line semantics: `N >` keeps pattern N as
+(s, p, OUT, o); `N <` reverses it to (o, p, IN, s); `N <<` starts from
+the predicate index (pid, PREDICATE_ID, IN, ?s)
+(core/parser.hpp plan format, documented in the .fmt headers).
+
+Pattern tuples here are (start, predicate, direction, end); vars are
+negative ids.  Schema ids match wukong_amd/csrc/lubm_gen.cpp.
+"""
+from . import Plan, DIR_IN, DIR_OUT
+
+# schema ids (lubm_gen.cpp enum; fixed enumeration, ids stable)
+PREDICATE_ID, TYPE_ID = 0, 1
+SUBORG, UGDEGREE, MEMBEROF, WORKSFOR, TEACHEROF, ADVISOR, TAKESCOURSE = range(2, 9)
+NAME, EMAIL, TELEPHONE, HEADOF, DOCDEGREE = range(9, 14)
+(UNIVERSITY, DEPARTMENT, FULLPROF, ASSOCPROF, ASSTPROF, LECTURER,
+ UGSTUDENT, GRADSTUDENT, COURSE, GRADCOURSE, RESEARCHGROUP) = range(14, 25)
+
+# deterministic generator constants (lubm_gen.cpp allocation order):
+UNIV0 = 1 << 17            # university 0 entity
+DEPT0_UNIV0 = UNIV0 + 2    # univ, univ-name-literal, then first dept
+
+X, Y, Z, Y1, Y2, Y3 = -1, -2, -3, -2, -3, -4
+
+# Q1 — osdi16_plan/lubm_q1.fmt: 3< 4> 2> 1> 5< 6>
+Q1 = Plan([
+    (GRADSTUDENT, TYPE_ID, DIR_IN, X),      # ?X type GraduateStudent (idx)
+    (X, MEMBEROF, DIR_OUT, Z),
+    (X, UGDEGREE, DIR_OUT, Y),
+    (Y, TYPE_ID, DIR_OUT, UNIVERSITY),
+    (Y, SUBORG, DIR_IN, Z),                 # ?Z subOrgOf ?Y reversed -> k2k
+    (Z, TYPE_ID, DIR_OUT, DEPARTMENT),
+], nvars=3, required_vars=[X, Y, Z])
+
+# Q2 — 1< 2>
+Q2 = Plan([
+    (COURSE, TYPE_ID, DIR_IN, X),
+    (X, NAME, DIR_OUT, Y),
+], nvars=2, required_vars=[X, Y])
+
+# Q3 — 1<< 2> 3> 4> 5> 6> 1>   (result is empty by schema: UG students
+# have no undergraduateDegreeFrom — matches the reference's #R=0)
+Q3 = Plan([
+    (UGDEGREE, PREDICATE_ID, DIR_IN, X),    # predicate-index start
+    (X, TYPE_ID, DIR_OUT, UGSTUDENT),
+    (X, MEMBEROF, DIR_OUT, Z),
+    (Z, TYPE_ID, DIR_OUT, DEPARTMENT),
+    (Z, SUBORG, DIR_OUT, Y),
+    (Y, TYPE_ID, DIR_OUT, UNIVERSITY),
+    (X, UGDEGREE, DIR_OUT, Y),              # k2k
+], nvars=3, required_vars=[X, Y, Z])
+
+# Q4 — 1< 2> 3> 4> 5>
+Q4 = Plan([
+    (DEPT0_UNIV0, WORKSFOR, DIR_IN, X),     # const start
+    (X, TYPE_ID, DIR_OUT, FULLPROF),
+    (X, NAME, DIR_OUT, Y1),
+    (X, EMAIL, DIR_OUT, Y2),
+    (X, TELEPHONE, DIR_OUT, Y3),
+], nvars=4, required_vars=[X, Y1, Y2, Y3])
+
+# Q5 — 1< 2>
+Q5 = Plan([
+    (DEPT0_UNIV0, SUBORG, DIR_IN, X),
+    (X, TYPE_ID, DIR_OUT, RESEARCHGROUP),
+], nvars=1, required_vars=[X])
+
+# Q6 — 1< 2> 3< 4>
+Q6 = Plan([
+    (UNIV0, SUBORG, DIR_IN, Y),
+    (Y, TYPE_ID, DIR_OUT, DEPARTMENT),
+    (Y, WORKSFOR, DIR_IN, X),
+    (X, TYPE_ID, DIR_OUT, FULLPROF),
+], nvars=2, required_vars=[X, Y])
+
+# Q7 — 1< 2< 3> 4> 5> 6<
+Q7 = Plan([
+    (FULLPROF, TYPE_ID, DIR_IN, Y),
+    (Y, ADVISOR, DIR_IN, X),
+    (X, TYPE_ID, DIR_OUT, UGSTUDENT),
+    (X, TAKESCOURSE, DIR_OUT, Z),
+    (Z, TYPE_ID, DIR_OUT, COURSE),
+    (Z, TEACHEROF, DIR_IN, Y),              # k2k
+], nvars=3, required_vars=[X, Y, Z])
+
+ALL = {"q1": Q1, "q2": Q2, "q3": Q3, "q4": Q4, "q5": Q5, "q6": Q6, "q7": Q7}
+HEAVY = ["q1", "q2", "q3", "q7"]
+LIGHT = ["q4", "q5", "q6"]
