@@ -17,7 +17,19 @@
 #include <atomic>
 #include <cstring>
 #include <cstdio>
+#include <chrono>
 #include <omp.h>
+
+static double now_s() {
+    return std::chrono::duration<double>(
+               std::chrono::steady_clock::now().time_since_epoch()).count();
+}
+static bool wk_verbose() {
+    static int v = -1;
+    if (v < 0) { const char *e = getenv("WK_VERBOSE"); v = e && atoi(e); }
+    return v;
+}
+#define WK_LOG(...) do { if (wk_verbose()) { fprintf(stderr, __VA_ARGS__); } } while (0)
 
 using namespace wk;
 
@@ -130,6 +142,7 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     if (!spo || ntriples < 0 || nsrv <= 0 || sid < 0 || sid >= nsrv) return nullptr;
     wk_store *st = new wk_store();
     st->sid = sid; st->nsrv = nsrv;
+    double t0 = now_s();
 
     // 1. partition into pso (s%n==sid) / pos (o%n==sid) — base_loader.hpp:344-352
     std::vector<triple> pso, pos;
@@ -147,11 +160,15 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     st->max_pid = max_pid;
     uint32_t max_blk = max_id >> NBITS_IDX;
 
+    WK_LOG("[store] partition: %.1fs (pso=%zu pos=%zu)\n", now_s() - t0, pso.size(), pos.size());
+    t0 = now_s();
     // 2. sort + dedup — base_loader.hpp:367-377
     sort_triples(pso, true, max_pid, max_blk);
     sort_triples(pos, false, max_pid, max_blk);
+    WK_LOG("[store] sort: %.1fs\n", now_s() - t0); t0 = now_s();
     dedup_triples(pso);
     dedup_triples(pos);
+    WK_LOG("[store] dedup: %.1fs\n", now_s() - t0); t0 = now_s();
 
     // 3. pid slice boundaries in each sorted array
     const uint32_t NP = max_pid + 1;
@@ -175,6 +192,7 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
         if (i != (int64_t)pos.size()) { fprintf(stderr, "pos slice error\n"); abort(); }
     }
 
+    WK_LOG("[store] slices: %.1fs\n", now_s() - t0); t0 = now_s();
     // 4. count keys/edges per segment.
     //    OUT seg of pid: (s,p) runs in pso slice.
     //    IN  seg of pid: (o,p) runs in pos slice, skipping is_tpid(o) runs
@@ -196,6 +214,7 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
         }
     }
 
+    WK_LOG("[store] seg counts: %.1fs\n", now_s() - t0); t0 = now_s();
     // index segment contents (gstore.hpp:858-888):
     //   [0|pid|IN]  = subjects of pid (from OUT keys, pid != TYPE_ID)
     //   [0|pid|OUT] = objects  of pid (from IN keys)
@@ -234,11 +253,15 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     alloc_seg(st->iseg[DIR_OUT], idx_out_keys);
     iseg_edge_start[DIR_OUT] = edge_cursor; edge_cursor += idx_out_edges;
 
+    WK_LOG("[store] idx counts: %.1fs\n", now_s() - t0); t0 = now_s();
     st->nbuckets_main = bucket_cursor;
     st->nbuckets_ext = bucket_cursor / 4 + 1024;
-    st->vertices.assign((st->nbuckets_main + st->nbuckets_ext) * ASSOC, vertex_t{0, 0});
-    st->edges.resize(edge_cursor);
+    st->vertices.alloc((st->nbuckets_main + st->nbuckets_ext) * ASSOC, /*zero=*/true);
+    st->edges.alloc(edge_cursor, /*zero=*/false);
     std::atomic<uint64_t> ext_next(st->nbuckets_main);
+    WK_LOG("[store] count+alloc: %.1fs (buckets=%lu edges=%lu)\n", now_s() - t0,
+           (unsigned long)st->nbuckets_main, (unsigned long)edge_cursor);
+    t0 = now_s();
 
     // 6. insert normal segments (parallel over (pid,dir) segments —
     //    static_gstore.hpp:64-161 semantics: one key per (vid,pid,dir) run,
@@ -268,6 +291,7 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
         }
     }
 
+    WK_LOG("[store] insert normal: %.1fs\n", now_s() - t0); t0 = now_s();
     // 7. insert index segments (insert_idx, static_gstore.hpp:217-280;
     //    ours iterates pids ascending — deterministic where the reference's
     //    TBB iteration order was not; parity is set-level)
@@ -313,6 +337,8 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
             }
         }
     }
+    WK_LOG("[store] insert index: %.1fs (ext used %lu)\n", now_s() - t0,
+           (unsigned long)(ext_next.load() - st->nbuckets_main));
     st->ext_used = ext_next.load() - st->nbuckets_main;
     return st;
 }

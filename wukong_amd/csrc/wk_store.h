@@ -6,15 +6,51 @@
 #pragma once
 #include "wk_types.h"
 #include <vector>
+#include <cstdlib>
+#include <cstring>
+#include <algorithm>
+
+// flat POD array with parallel zeroing (std::vector's value-init is
+// single-threaded and page-fault-bound: ~6 s for the 1 GB slot array)
+template <class T>
+struct pod_array {
+    T *ptr = nullptr;
+    size_t n = 0;
+    pod_array() = default;
+    pod_array(const pod_array &) = delete;
+    pod_array &operator=(const pod_array &) = delete;
+    ~pod_array() { free(ptr); }
+    void alloc(size_t count, bool zero) {
+        free(ptr);
+        ptr = (T *)malloc(count * sizeof(T) + 16);
+        n = count;
+        if (zero) {
+#pragma omp parallel for schedule(static)
+            for (long long i = 0; i < (long long)count; i += (1 << 20)) {
+                size_t len = std::min<size_t>(1 << 20, count - (size_t)i);
+                memset(ptr + i, 0, len * sizeof(T));
+            }
+        }
+    }
+    T *data() { return ptr; }
+    const T *data() const { return ptr; }
+    size_t size() const { return n; }
+    T *begin() { return ptr; }
+    T *end() { return ptr + n; }
+    const T *begin() const { return ptr; }
+    const T *end() const { return ptr + n; }
+    T &operator[](size_t i) { return ptr[i]; }
+    const T &operator[](size_t i) const { return ptr[i]; }
+};
 
 struct wk_store {
     int32_t sid = 0, nsrv = 1;
     uint32_t max_pid = 0;  // largest predicate/type id present
 
     // cluster-hash slot array: (main + ext buckets) * 8 slots
-    std::vector<wk::vertex_t> vertices;
+    pod_array<wk::vertex_t> vertices;
     // all edge lists, contiguous per segment
-    std::vector<wk::sid_t> edges;
+    pod_array<wk::sid_t> edges;
     // normal segments indexed [pid*2 + dir], size (max_pid+1)*2
     std::vector<wk::seg_t> nseg;
     // index segments [dir]: keys [0|pid|dir] / [0|tid|IN]
