@@ -1,0 +1,240 @@
+#!/usr/bin/env python3
+"""bench.py — LUBM Q1-Q7 suite throughput on N MI355X GPUs.
+
+One step = one pass of the 7-query OSDI16-plan suite over the LUBM-2560
+synthetic store resident in HBM (BASELINE.json configs[1]).  value =
+whole-job queries/sec.  N>1: launched by torch.distributed.run, one rank
+per GPU, store partitioned by vid % N, per-step RCCL all-to-allv.
+
+Env knobs: WK_UNIV (default 2560), WK_CPU_UNIV (cpu_baseline sample,
+default 256), WK_SKIP_CPU_BASELINE=1.
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, HERE)
+
+import numpy as np  # noqa: E402
+
+os.environ.setdefault("WK_KERNEL_TIMING", "1")
+
+HBM_PEAK_GBS = 8000.0  # MI355X spec (MI355X_MICROARCH.md)
+
+
+def log(*a):
+    print(*a, file=sys.stderr, flush=True)
+
+
+def pctl(xs, p):
+    return float(np.percentile(np.asarray(xs), p)) if xs else None
+
+
+def cpu_baseline(nuniv_workload, seed):
+    """Oracle engine (DESIGN.md §4 — 'port' of the reference CPU engine)
+    timed on this box's host cores, bounded sample (~10-30 s CPU)."""
+    import multiprocessing
+    from tests.oracle_util import OracleCtx
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+
+    cores = multiprocessing.cpu_count()
+    nuniv = int(os.environ.get("WK_CPU_UNIV", "256"))
+    t0 = time.time()
+    triples = wk.lubm_gen(nuniv, seed=seed)
+    ctx = OracleCtx(triples)
+    build_s = time.time() - t0
+    # one warm pass, then timed passes until ~10 s
+    for plan in Q.ALL.values():
+        ctx.run_query(plan, mt=cores)
+    passes, t0 = 0, time.time()
+    while time.time() - t0 < 10.0:
+        for plan in Q.ALL.values():
+            ctx.run_query(plan, mt=cores)
+        passes += 1
+    secs = time.time() - t0
+    qps = 7.0 * passes / secs
+    return {
+        "value": round(qps, 3),
+        "unit": "queries/s",
+        "cores": cores,
+        "kind": "port",
+        "sample": (f"LUBM-{nuniv} (workload is LUBM-{nuniv_workload}), "
+                   f"Q1-Q7 suite x{passes} in {secs:.1f}s, oracle engine "
+                   f"(restated reference CPU engine), mt={cores}; "
+                   f"oracle build {build_s:.1f}s excluded"),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=4)
+    ap.add_argument("--warmup", type=int, default=1)
+    args = ap.parse_args()
+
+    import torch
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    ngpus = max(args.gpus, world)
+    distributed = world > 1
+
+    seed = 42
+    nuniv = int(os.environ.get("WK_UNIV", "2560"))
+    dataset = f"LUBM-{nuniv}"
+
+    if distributed:
+        import torch.distributed as dist
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+        from wukong_amd.dist import DistQuery, GpuExecutor
+
+    t0 = time.time()
+    triples = wk.lubm_gen(nuniv, seed=seed, sid=rank, nsrv=world)
+    log(f"[rank {rank}] gen {dataset}: {triples.shape[0]} triples "
+        f"({time.time()-t0:.1f}s)")
+    t0 = time.time()
+    store = wk.Store(triples, sid=rank, nsrv=world)
+    del triples
+    log(f"[rank {rank}] store: {store.num_slots} slots, {store.num_edges} edges "
+        f"({time.time()-t0:.1f}s)")
+    t0 = time.time()
+    eng = wk.Engine(store, device=local_rank)
+    log(f"[rank {rank}] HBM upload {((store.num_slots*16+store.num_edges*4)/1e9):.2f} GB "
+        f"({time.time()-t0:.1f}s)")
+
+    names = list(Q.ALL)
+
+    def run_suite(collect=None):
+        for name in names:
+            plan = Q.ALL[name]
+            tq = time.time()
+            if distributed:
+                ex = GpuExecutor(eng, plan)
+                dq = DistQuery(ex, plan, rank, world, device=f"cuda:{local_rank}")
+                dq.run()
+                part = ex.finalize()
+                cnt = torch.tensor([len(part)], dtype=torch.int64,
+                                   device=f"cuda:{local_rank}")
+                dist.all_reduce(cnt)
+                nrows = int(cnt.item())
+            else:
+                tbl = eng.run_query(plan)
+                nrows = tbl.shape[0]
+            if collect is not None:
+                collect.setdefault(name, []).append((time.time() - tq) * 1e3)
+                collect.setdefault("_rows", {})[name] = nrows
+
+    def sync():
+        torch.cuda.synchronize(local_rank)
+        if distributed:
+            dist.barrier()
+            torch.cuda.synchronize(local_rank)
+
+    # warmup
+    for _ in range(args.warmup):
+        run_suite()
+    sync()
+    stats0 = eng.kernel_stats()
+
+    lat = {}
+    t_start = time.time()
+    for _ in range(args.steps):
+        run_suite(collect=lat)
+    sync()
+    elapsed = time.time() - t_start
+    stats1 = eng.kernel_stats()
+
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=f"cuda:{local_rank}")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    if rank != 0:
+        if distributed:
+            dist.destroy_process_group()
+        return
+
+    total_queries = 7 * args.steps
+    qps = total_queries / elapsed
+
+    # roofline: dominant kernel = k2u expand
+    dk = "expand"
+    d_us = stats1[dk]["usec"] - stats0[dk]["usec"]
+    d_by = stats1[dk]["bytes"] - stats0[dk]["bytes"]
+    d_n = stats1[dk]["launches"] - stats0[dk]["launches"]
+    achieved = (d_by / d_us / 1e3) if d_us > 0 else None  # GB/s
+    traffic = None
+    tj = os.path.join(HERE, "profiles", "traffic.json")
+    if os.path.exists(tj):
+        with open(tj) as f:
+            tdata = json.load(f)
+        traffic = tdata.get("expand_bytes_per_launch")
+    roofline = {
+        "bound": "hbm",
+        "kernel": "k_expand",
+        "achieved": round(achieved, 1) if achieved else None,
+        "peak": HBM_PEAK_GBS,
+        "unit": "GB/s",
+        "frac": round(achieved / HBM_PEAK_GBS, 4) if achieved else None,
+        "traffic": traffic,
+        "launches": int(d_n),
+        "avg_launch_us": round(d_us / d_n, 2) if d_n else None,
+    }
+
+    cb = None
+    if os.environ.get("WK_SKIP_CPU_BASELINE") != "1" and not distributed:
+        log("[rank 0] timing cpu_baseline (oracle engine)...")
+        cb = cpu_baseline(nuniv, seed)
+
+    latency = {}
+    for name in names:
+        xs = lat.get(name, [])
+        latency[name] = {"p50_ms": round(pctl(xs, 50), 3),
+                         "p99_ms": round(pctl(xs, 99), 3),
+                         "rows": lat.get("_rows", {}).get(name)}
+
+    out = {
+        "metric": "queries/sec, LUBM-2560 Q1-Q7 mix (OSDI16 plans)",
+        "value": round(qps, 3),
+        "unit": "queries/s",
+        "n_gpus": ngpus,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(elapsed * 1e3 / args.steps, 3),
+        "higher_is_better": True,
+        "scaling": "strong",
+        "vs_baseline": None,
+        "dtype": "u32",
+        "data": "synthetic",
+        "config": {
+            "workload": "LUBM-2560 on 1xMI355X, full graph in HBM, "
+                        "Q1-Q7 mixed light/heavy" if ngpus == 1 else
+                        f"LUBM-2560 subject-hash-partitioned across "
+                        f"{ngpus}xMI355X, RCCL all-to-all sub-query shipping",
+            "dataset": dataset,
+            "queries": "lubm q1-q7, osdi16 plans",
+            "parallelism": f"graph-partitioned x{ngpus} + per-step all-to-allv",
+        },
+        "roofline": roofline,
+        "cpu_baseline": cb,
+        "latency": latency,
+        "kernel_stats": {k: {"usec": round(stats1[k]["usec"] - stats0[k]["usec"], 1),
+                             "launches": int(stats1[k]["launches"] - stats0[k]["launches"])}
+                         for k in stats1},
+    }
+    print(json.dumps(out), flush=True)
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

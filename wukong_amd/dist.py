@@ -76,9 +76,15 @@ class DistQuery:
         dist.all_to_all_single(recv_rows, send_rows)
         in_splits = [int(s) * ncols for s in sizes]
         out_list = [int(x) * ncols for x in recv_rows.tolist()]
+        # global-empty guard (e.g. a query emptied on every rank): all
+        # ranks must agree before skipping the collective
+        tot = torch.tensor([sum(in_splits)], dtype=torch.int64, device=self.device)
+        dist.all_reduce(tot)
+        if int(tot.item()) == 0:
+            return torch.empty(1, dtype=torch.int32, device=self.device), 0
         recv = torch.empty(max(sum(out_list), 1), dtype=torch.int32,
                            device=self.device)
-        dist.all_to_all_single(recv[:sum(out_list)] if sum(out_list) else recv[:0],
+        dist.all_to_all_single(recv[:sum(out_list)],
                                buf[:sum(in_splits)], out_list, in_splits)
         return recv, sum(out_list) // ncols if ncols else 0
 
