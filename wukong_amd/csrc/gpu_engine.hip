@@ -32,6 +32,17 @@
 #include <cstring>
 #include <cstdio>
 #include <cstdlib>
+#include <chrono>
+
+static double now_us() {
+    return std::chrono::duration<double, std::micro>(
+               std::chrono::steady_clock::now().time_since_epoch()).count();
+}
+static int wk_verbose_lvl() {
+    static int v = -1;
+    if (v < 0) { const char *e = getenv("WK_VERBOSE"); v = e ? atoi(e) : 0; }
+    return v;
+}
 
 using namespace wk;
 
@@ -437,13 +448,28 @@ static int32_t scan_counts(wk_engine *e, int64_t n, uint64_t *total_out) {
     hipcub::DeviceScan::ExclusiveSum(e->cubtmp.p, tmp, d_cnt, d_pre, n + 1, e->stream);
     TIME_END(e, CAT_SCAN, (double)(n + 1) * 16);
     HIP_CHECK(hipMemcpyAsync(e->h_pin, d_pre + n, 8, hipMemcpyDeviceToHost, e->stream));
+    double ts = now_us();
     HIP_CHECK(hipStreamSynchronize(e->stream));
+    if (wk_verbose_lvl() >= 2)
+        fprintf(stderr, "[scan] n=%lld sync_us=%.0f\n", (long long)n, now_us() - ts);
     *total_out = e->h_pin[0];
     return WK_OK;
 }
 
 // Run one pattern — dispatch per sparql.hpp:1016-1058.
+extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *nrows_out);
 extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_out) {
+    if (wk_verbose_lvl() < 2) return wk_engine_execute_one_pattern_impl(e, nrows_out);
+    double t0 = now_us();
+    int64_t rin = e ? e->nrows : -1;
+    int step = e ? e->step : -1;
+    int32_t rc = wk_engine_execute_one_pattern_impl(e, nrows_out);
+    fprintf(stderr, "[pat] step=%d rows_in=%lld rows_out=%lld host_us=%.0f\n",
+            step, (long long)rin, (long long)(e ? e->nrows : -1), now_us() - t0);
+    return rc;
+}
+
+extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *nrows_out) {
     if (!e || e->step >= (int)e->pats.size()) return WK_ERR_STATE;
     const wk_store *st = e->st;
     const wk_pattern_t pat = e->pats[e->step];
@@ -732,13 +758,19 @@ static int32_t finalize_result(wk_engine *e, const wk_plan_t *plan,
 extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
                                           wk_result_t *out) {
     if (!e || !plan || !out) return WK_ERR_STATE;
+    double t0 = now_us();
     std::vector<sid_t> tbl((size_t)e->nrows * e->ncols);
     if (!tbl.empty())
         HIP_CHECK(hipMemcpyAsync(tbl.data(), e->tbl[e->cur].p, tbl.size() * 4,
                                  hipMemcpyDeviceToHost, e->stream));
     HIP_CHECK(hipStreamSynchronize(e->stream));
     resolve_timing(e);
-    return finalize_result(e, plan, tbl, out);
+    double t1 = now_us();
+    int32_t rc = finalize_result(e, plan, tbl, out);
+    if (wk_verbose_lvl() >= 2)
+        fprintf(stderr, "[fetch] rows=%lld d2h_us=%.0f final_us=%.0f\n",
+                (long long)e->nrows, t1 - t0, now_us() - t1);
+    return rc;
 }
 
 // raw current table (no final ops) — for the gloo-path exchange in tests
