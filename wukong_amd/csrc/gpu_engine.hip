@@ -164,6 +164,69 @@ __global__ void k_probe(const vertex_t *__restrict__ verts,
     }
 }
 
+// Thread-per-row probe variant: one LANE walks one row's bucket chain
+// (7 slot compares in one 128-B line, L1-resident after the first 16-B
+// fetch).  64 rows in flight per wave vs the cooperative kernel's 8 —
+// 8x the memory-level parallelism for DRAM-random probe patterns.
+__global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
+                            const sid_t *__restrict__ edges,
+                            uint64_t bucket_start, uint64_t num_buckets,
+                            const sid_t *__restrict__ tbl, int64_t nrows, int ncols,
+                            int col, uint32_t pid, int dir, int key_mode,
+                            int probe_mode, int col2, sid_t cval,
+                            uint64_t list_off, uint64_t list_sz,
+                            uint64_t *__restrict__ d_eoff,
+                            uint64_t *__restrict__ d_cnt)
+{
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
+         r += (int64_t)gridDim.x * blockDim.x) {
+        if (probe_mode == PM_LIST) {
+            sid_t v = tbl[r * ncols + col];
+            d_cnt[r] = bsearch_u32(edges + list_off, list_sz, v) ? 1 : 0;
+            continue;
+        }
+        sid_t v = tbl[r * ncols + col];
+        uint64_t key = (key_mode == PK_NORMAL) ? key_pack(v, pid, (uint64_t)dir)
+                                               : key_pack(0, v, (uint64_t)dir);
+        uint64_t bucket = bucket_start + hash_u64(key) % num_buckets;
+        uint64_t eoff = 0, esz = 0;
+        while (true) {
+            const vertex_t *b = &verts[bucket * ASSOC];
+            // one 128-B line; issue wide reads, compare in registers
+            uint64_t k0 = b[0].key, k1 = b[1].key, k2 = b[2].key, k3 = b[3].key;
+            uint64_t k4 = b[4].key, k5 = b[5].key, k6 = b[6].key, k7 = b[7].key;
+            int hit = -1;
+            if (k0 == key) hit = 0;
+            else if (k1 == key) hit = 1;
+            else if (k2 == key) hit = 2;
+            else if (k3 == key) hit = 3;
+            else if (k4 == key) hit = 4;
+            else if (k5 == key) hit = 5;
+            else if (k6 == key) hit = 6;
+            if (hit >= 0) {
+                uint64_t p = b[hit].ptr;
+                eoff = ptr_off(p); esz = ptr_size(p);
+                break;
+            }
+            if (k7 == KEY_EMPTY) { esz = 0; break; }
+            bucket = key_vid(k7);
+        }
+        switch (probe_mode) {
+        case PM_SIZE:
+            d_eoff[r] = eoff;
+            d_cnt[r] = esz;
+            break;
+        case PM_CONST:
+            d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz, cval)) ? 1 : 0;
+            break;
+        case PM_COL:
+            d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz,
+                                           tbl[r * ncols + col2])) ? 1 : 0;
+            break;
+        }
+    }
+}
+
 // Output-centric expansion: one thread per OUTPUT row; binary-search the
 // prefix array for the source row.  Consecutive threads write consecutive
 // rows -> coalesced stores (known_to_unknown semantics, sparql.hpp:325-367).
@@ -309,6 +372,8 @@ struct wk_engine {
     int step = 0;
     int status = 0;
 
+    bool probe_coop = false;  // WK_PROBE=coop selects the 8-lane kernel
+
     // timing
     bool timing = false;
     std::vector<timed_launch> pending;
@@ -372,6 +437,7 @@ extern "C" wk_engine_t *wk_engine_create(const wk_store_t *st, int32_t device) {
         return nullptr;
     }
     if (const char *t = getenv("WK_KERNEL_TIMING")) e->timing = atoi(t) != 0;
+    if (const char *pv = getenv("WK_PROBE")) e->probe_coop = !strcmp(pv, "coop");
     return e;
 }
 
@@ -536,10 +602,16 @@ extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *n
         if (e->cnt.ensure((size_t)(R + 1) * 8)) return WK_ERR_HIP;
         HIP_CHECK(hipMemsetAsync((uint64_t *)e->cnt.p + R, 0, 8, e->stream));
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_probe, dim3(grid_for(R * 8)), dim3(BLOCK), 0, e->stream,
-                           e->d_verts, e->d_edges, 0, 1, cur_tbl, R, e->ncols, col,
-                           0u, dir, PK_NORMAL, PM_LIST, 0, 0u, off, sz,
-                           (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+        if (e->probe_coop)
+            hipLaunchKernelGGL(k_probe, dim3(grid_for(R * 8)), dim3(BLOCK), 0, e->stream,
+                               e->d_verts, e->d_edges, 0, 1, cur_tbl, R, e->ncols, col,
+                               0u, dir, PK_NORMAL, PM_LIST, 0, 0u, off, sz,
+                               (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+        else
+            hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(R)), dim3(BLOCK), 0, e->stream,
+                               e->d_verts, e->d_edges, 0, 1, cur_tbl, R, e->ncols, col,
+                               0u, dir, PK_NORMAL, PM_LIST, 0, 0u, off, sz,
+                               (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
         TIME_END(e, CAT_FILTER, (double)R * 12);
         uint64_t total = 0;
         int32_t rc = scan_counts(e, R, &total);
@@ -605,10 +677,16 @@ extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *n
     sid_t cval = (ostat == 2) ? (sid_t)o : 0;
     if (R) {
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_probe, dim3(grid_for(R * 8)), dim3(BLOCK), 0, e->stream,
-                           e->d_verts, e->d_edges, seg->bucket_start, seg->num_buckets,
-                           cur_tbl, R, e->ncols, col, (uint32_t)p, dir, key_mode, pmode,
-                           col2, cval, 0, 0, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+        if (e->probe_coop)
+            hipLaunchKernelGGL(k_probe, dim3(grid_for(R * 8)), dim3(BLOCK), 0, e->stream,
+                               e->d_verts, e->d_edges, seg->bucket_start, seg->num_buckets,
+                               cur_tbl, R, e->ncols, col, (uint32_t)p, dir, key_mode, pmode,
+                               col2, cval, 0, 0, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+        else
+            hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(R)), dim3(BLOCK), 0, e->stream,
+                               e->d_verts, e->d_edges, seg->bucket_start, seg->num_buckets,
+                               cur_tbl, R, e->ncols, col, (uint32_t)p, dir, key_mode, pmode,
+                               col2, cval, 0, 0, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
         // bytes: key col + bucket + (scan: log2(deg) touches ~1 cacheline)
         TIME_END(e, CAT_PROBE, (double)R * (4 + 128 + 8 + (pmode != PM_SIZE ? 64 : 0)));
     }
