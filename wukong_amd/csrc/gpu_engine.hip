@@ -403,10 +403,16 @@ static hipEvent_t ev_get() { hipEvent_t e; (void)hipEventCreate(&e); return e; }
     if ((eng)->timing) { (void)hipEventRecord(t_e_, (eng)->stream);         \
         (eng)->pending.push_back({t_s_, t_e_, (category), (double)(nbytes)}); }
 
+static const char *CAT_NAMES[7] = {"probe", "scan", "expand", "filter",
+                                   "copy", "split", "other"};
+
 static void resolve_timing(wk_engine *e) {
     for (auto &t : e->pending) {
         float ms = 0;
         (void)hipEventElapsedTime(&ms, t.start, t.stop);
+        if (wk_verbose_lvl() >= 3)
+            fprintf(stderr, "[launch] %s %.0fus %.2fGB/s\n", CAT_NAMES[t.cat],
+                    ms * 1e3, ms > 0 ? t.bytes / (ms * 1e6) : 0.0);
         e->cat_usec[t.cat] += ms * 1000.0;
         e->cat_bytes[t.cat] += t.bytes;
         e->cat_n[t.cat]++;
@@ -510,14 +516,32 @@ static int32_t scan_counts(wk_engine *e, int64_t n, uint64_t *total_out) {
     uint64_t *d_pre = (uint64_t *)e->prefix.p;
     hipcub::DeviceScan::ExclusiveSum(nullptr, tmp, d_cnt, d_pre, n + 1, e->stream);
     if (e->cubtmp.ensure(tmp)) return WK_ERR_HIP;
+    // verbose>=2: bracket every stream op to localise stalls
+    const bool dbg = wk_verbose_lvl() >= 2;
+    hipEvent_t ev[4];
+    if (dbg) for (int i = 0; i < 4; i++) (void)hipEventCreate(&ev[i]);
+    if (dbg) (void)hipEventRecord(ev[0], e->stream);
+    HIP_CHECK(hipMemsetAsync(d_cnt + n, 0, 8, e->stream));
+    if (dbg) (void)hipEventRecord(ev[1], e->stream);
     TIME_BEGIN(e);
     hipcub::DeviceScan::ExclusiveSum(e->cubtmp.p, tmp, d_cnt, d_pre, n + 1, e->stream);
     TIME_END(e, CAT_SCAN, (double)(n + 1) * 16);
+    if (dbg) (void)hipEventRecord(ev[2], e->stream);
     HIP_CHECK(hipMemcpyAsync(e->h_pin, d_pre + n, 8, hipMemcpyDeviceToHost, e->stream));
+    if (dbg) (void)hipEventRecord(ev[3], e->stream);
     double ts = now_us();
     HIP_CHECK(hipStreamSynchronize(e->stream));
-    if (wk_verbose_lvl() >= 2)
-        fprintf(stderr, "[scan] n=%lld sync_us=%.0f\n", (long long)n, now_us() - ts);
+    if (dbg) {
+        float pre = 0, ms = 0, sc = 0, d2h = 0;
+        (void)hipEventElapsedTime(&ms, ev[0], ev[1]);
+        (void)hipEventElapsedTime(&sc, ev[1], ev[2]);
+        (void)hipEventElapsedTime(&d2h, ev[2], ev[3]);
+        fprintf(stderr,
+                "[scan] n=%lld sync_us=%.0f memset_us=%.0f scan_us=%.0f d2h_us=%.0f\n",
+                (long long)n, now_us() - ts, ms * 1e3, sc * 1e3, d2h * 1e3);
+        (void)pre;
+        for (int i = 0; i < 4; i++) (void)hipEventDestroy(ev[i]);
+    }
     *total_out = e->h_pin[0];
     return WK_OK;
 }
@@ -600,7 +624,6 @@ extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *n
         if (col < 0) return WK_ERR_PLAN;
         int64_t R = e->nrows;
         if (e->cnt.ensure((size_t)(R + 1) * 8)) return WK_ERR_HIP;
-        HIP_CHECK(hipMemsetAsync((uint64_t *)e->cnt.p + R, 0, 8, e->stream));
         TIME_BEGIN(e);
         if (e->probe_coop)
             hipLaunchKernelGGL(k_probe, dim3(grid_for(R * 8)), dim3(BLOCK), 0, e->stream,
@@ -670,7 +693,6 @@ extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *n
 
     if (e->cnt.ensure((size_t)(R + 1) * 8) || e->eoff.ensure((size_t)(R ? R : 1) * 8))
         return WK_ERR_HIP;
-    HIP_CHECK(hipMemsetAsync((uint64_t *)e->cnt.p + R, 0, 8, e->stream));
 
     int pmode = (ostat == 0) ? PM_SIZE : (ostat == 2 ? PM_CONST : PM_COL);
     int col2 = (ostat == 1) ? e->var2col(o) : 0;
