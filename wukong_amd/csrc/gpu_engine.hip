@@ -394,6 +394,20 @@ static int grid_for(int64_t work) {
 
 static hipEvent_t ev_get() { hipEvent_t e; (void)hipEventCreate(&e); return e; }
 
+// WK_SPIN_SYNC=1: poll hipStreamQuery instead of blocking sync (the
+// blocking path showed intermittent ~25 ms interrupt-latency stalls)
+static bool wk_spin_sync() {
+    static int v = -1;
+    if (v < 0) { const char *e = getenv("WK_SPIN_SYNC"); v = e ? atoi(e) : 0; }
+    return v;
+}
+static hipError_t stream_sync(hipStream_t s) {
+    if (!wk_spin_sync()) return hipStreamSynchronize(s);
+    hipError_t rc;
+    while ((rc = hipStreamQuery(s)) == hipErrorNotReady) {}
+    return rc;
+}
+
 #define TIME_BEGIN(eng)                                                     \
     hipEvent_t t_s_ = nullptr, t_e_ = nullptr;                              \
     if ((eng)->timing) { t_s_ = ev_get(); t_e_ = ev_get();                  \
@@ -530,7 +544,7 @@ static int32_t scan_counts(wk_engine *e, int64_t n, uint64_t *total_out) {
     HIP_CHECK(hipMemcpyAsync(e->h_pin, d_pre + n, 8, hipMemcpyDeviceToHost, e->stream));
     if (dbg) (void)hipEventRecord(ev[3], e->stream);
     double ts = now_us();
-    HIP_CHECK(hipStreamSynchronize(e->stream));
+    HIP_CHECK(stream_sync(e->stream));
     if (dbg) {
         float pre = 0, ms = 0, sc = 0, d2h = 0;
         (void)hipEventElapsedTime(&ms, ev[0], ev[1]);
@@ -781,7 +795,7 @@ extern "C" int32_t wk_engine_generate_sub_query(wk_engine_t *e, int32_t ndst,
     unsigned long long h_hist[64];
     HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, ndst * sizeof(unsigned long long),
                              hipMemcpyDeviceToHost, e->stream));
-    HIP_CHECK(hipStreamSynchronize(e->stream));
+    HIP_CHECK(stream_sync(e->stream));
     unsigned long long base[64], acc = 0;
     for (int i = 0; i < ndst; i++) { base[i] = acc; acc += h_hist[i]; rows_per_dst[i] = (int64_t)h_hist[i]; }
     HIP_CHECK(hipMemcpyAsync(d_hist, base, ndst * sizeof(unsigned long long),
@@ -792,7 +806,7 @@ extern "C" int32_t wk_engine_generate_sub_query(wk_engine_t *e, int32_t ndst,
                            cur_tbl, R, e->ncols, col, ndst, d_hist, d_cursor, dev_out);
         TIME_END(e, CAT_SPLIT, (double)R * e->ncols * 8);
     }
-    HIP_CHECK(hipStreamSynchronize(e->stream));
+    HIP_CHECK(stream_sync(e->stream));
     return WK_OK;
 }
 
@@ -863,7 +877,7 @@ extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
     if (!tbl.empty())
         HIP_CHECK(hipMemcpyAsync(tbl.data(), e->tbl[e->cur].p, tbl.size() * 4,
                                  hipMemcpyDeviceToHost, e->stream));
-    HIP_CHECK(hipStreamSynchronize(e->stream));
+    HIP_CHECK(stream_sync(e->stream));
     resolve_timing(e);
     double t1 = now_us();
     int32_t rc = finalize_result(e, plan, tbl, out);
@@ -881,7 +895,7 @@ extern "C" int32_t wk_engine_fetch_raw(wk_engine_t *e, wk_result_t *out) {
     if (n)
         HIP_CHECK(hipMemcpyAsync(res, e->tbl[e->cur].p, n * 4,
                                  hipMemcpyDeviceToHost, e->stream));
-    HIP_CHECK(hipStreamSynchronize(e->stream));
+    HIP_CHECK(stream_sync(e->stream));
     out->col_num = e->ncols;
     out->row_num = e->nrows;
     out->table = res;
