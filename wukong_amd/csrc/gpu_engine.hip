@@ -279,6 +279,15 @@ __global__ void k_copy_list(const sid_t *__restrict__ edges, uint64_t off,
         out[t] = edges[off + t];
 }
 
+// zero one u64 + publish the scan total straight into pinned host memory
+// (a 1-thread kernel each: keeps the hot path on the compute queue; the
+// 8-byte hipMemsetAsync/hipMemcpyAsync SDMA hops behind a large kernel
+// showed ~25 ms cross-queue stalls on this box)
+__global__ void k_zero_u64(uint64_t *p) { *p = 0; }
+__global__ void k_publish_u64(const uint64_t *src, uint64_t *dst_pinned) {
+    *dst_pinned = *src;
+}
+
 // fork-join split (generate_sub_query sparql.hpp:772-796): dst = vid % ndst
 // (hash_mod, utils/math.hpp:51-55)
 __global__ void k_dst_histogram(const sid_t *__restrict__ tbl, int64_t nrows,
@@ -535,13 +544,14 @@ static int32_t scan_counts(wk_engine *e, int64_t n, uint64_t *total_out) {
     hipEvent_t ev[4];
     if (dbg) for (int i = 0; i < 4; i++) (void)hipEventCreate(&ev[i]);
     if (dbg) (void)hipEventRecord(ev[0], e->stream);
-    HIP_CHECK(hipMemsetAsync(d_cnt + n, 0, 8, e->stream));
+    hipLaunchKernelGGL(k_zero_u64, dim3(1), dim3(1), 0, e->stream, d_cnt + n);
     if (dbg) (void)hipEventRecord(ev[1], e->stream);
     TIME_BEGIN(e);
     hipcub::DeviceScan::ExclusiveSum(e->cubtmp.p, tmp, d_cnt, d_pre, n + 1, e->stream);
     TIME_END(e, CAT_SCAN, (double)(n + 1) * 16);
     if (dbg) (void)hipEventRecord(ev[2], e->stream);
-    HIP_CHECK(hipMemcpyAsync(e->h_pin, d_pre + n, 8, hipMemcpyDeviceToHost, e->stream));
+    hipLaunchKernelGGL(k_publish_u64, dim3(1), dim3(1), 0, e->stream, d_pre + n,
+                       e->h_pin);
     if (dbg) (void)hipEventRecord(ev[3], e->stream);
     double ts = now_us();
     HIP_CHECK(stream_sync(e->stream));
