@@ -373,6 +373,8 @@ struct wk_engine {
     // scratch
     devbuf eoff, cnt, prefix, cubtmp, misc;
     uint64_t *h_pin = nullptr;  // pinned: totals
+    void *h_stage = nullptr;    // pinned staging for result downloads
+    size_t h_stage_cap = 0;
 
     // query state (host mirror of SPARQLQuery, query.hpp:560-594)
     std::vector<wk_pattern_t> pats;
@@ -479,6 +481,7 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
     if (e->d_verts) (void)hipFree(e->d_verts);
     if (e->d_edges) (void)hipFree(e->d_edges);
     if (e->h_pin) (void)hipHostFree(e->h_pin);
+    if (e->h_stage) (void)hipHostFree(e->h_stage);
     if (e->stream) (void)hipStreamDestroy(e->stream);
     delete e;
 }
@@ -820,6 +823,28 @@ extern "C" int32_t wk_engine_generate_sub_query(wk_engine_t *e, int32_t ndst,
     return WK_OK;
 }
 
+// download the current device table via a persistent PINNED staging
+// buffer: a pageable-destination hipMemcpyAsync makes the NEXT queries'
+// kernels stall ~25 ms (driver pin/unpin behind a large result copy,
+// measured in tools/qloop.py)
+static int32_t download_table(wk_engine *e, sid_t *dst, size_t n) {
+    if (!n) return WK_OK;
+    size_t bytes = n * 4;
+    if (bytes > e->h_stage_cap) {
+        if (e->h_stage) (void)hipHostFree(e->h_stage);
+        e->h_stage = nullptr;
+        e->h_stage_cap = 0;
+        size_t want = bytes + bytes / 2;
+        if (hipHostMalloc(&e->h_stage, want) != hipSuccess) return WK_ERR_HIP;
+        e->h_stage_cap = want;
+    }
+    HIP_CHECK(hipMemcpyAsync(e->h_stage, e->tbl[e->cur].p, bytes,
+                             hipMemcpyDeviceToHost, e->stream));
+    HIP_CHECK(stream_sync(e->stream));
+    memcpy(dst, e->h_stage, bytes);
+    return WK_OK;
+}
+
 // ---- host-side final ops (final_process, sparql.hpp:1424-1551) ----
 static int32_t finalize_result(wk_engine *e, const wk_plan_t *plan,
                                std::vector<sid_t> &tbl, wk_result_t *out) {
@@ -884,9 +909,8 @@ extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
     if (!e || !plan || !out) return WK_ERR_STATE;
     double t0 = now_us();
     std::vector<sid_t> tbl((size_t)e->nrows * e->ncols);
-    if (!tbl.empty())
-        HIP_CHECK(hipMemcpyAsync(tbl.data(), e->tbl[e->cur].p, tbl.size() * 4,
-                                 hipMemcpyDeviceToHost, e->stream));
+    int32_t drc = download_table(e, tbl.data(), tbl.size());
+    if (drc) return drc;
     HIP_CHECK(stream_sync(e->stream));
     resolve_timing(e);
     double t1 = now_us();
@@ -902,9 +926,8 @@ extern "C" int32_t wk_engine_fetch_raw(wk_engine_t *e, wk_result_t *out) {
     if (!e || !out) return WK_ERR_STATE;
     size_t n = (size_t)e->nrows * e->ncols;
     wk_sid_t *res = (wk_sid_t *)malloc(n ? n * 4 : 4);
-    if (n)
-        HIP_CHECK(hipMemcpyAsync(res, e->tbl[e->cur].p, n * 4,
-                                 hipMemcpyDeviceToHost, e->stream));
+    int32_t drc = download_table(e, res, n);
+    if (drc) { free(res); return drc; }
     HIP_CHECK(stream_sync(e->stream));
     out->col_num = e->ncols;
     out->row_num = e->nrows;
