@@ -6,26 +6,32 @@
  * HBM-resident full store (no segment cache / block mapping).  Operator
  * semantics restate core/engine/sparql.hpp:80-549.
  *
+ * Execution model: a query is ONE asynchronous launch chain — row counts
+ * live in device memory (d_state) and every kernel reads its bounds from
+ * there, so the host never synchronizes between patterns (the reference's
+ * GPU pipeline synced per step, gpu_engine_cuda.hpp:168-197).  Scratch is
+ * sized to a grow-only row capacity; expansion overflow raises a device
+ * flag and the query is re-run at the larger capacity (replaces the
+ * reference's hard 32 MB rbuf assert, gpu_engine_cuda.hpp:185).
+ *
  * Kernel inventory:
- *   k_probe           — fused gen-keys + cooperative cluster-hash probe
- *                       (+ optional bsearch filter for k2c/k2k/c2k)
- *                       replaces gpu_hash.cu:94-132,149-260,262-445
- *   k_expand          — output-centric load-balanced table expansion
- *                       replaces gpu_hash.cu:762-834 (no 20-col cap)
- *   k_compact         — flag-compaction for k2k/k2c/c2k/i2k
- *                       replaces gpu_hash.cu:523-585
- *   k_copy_list       — i2u / c2u list materialisation (GPU-side; the
+ *   k_probe (8-lane coop) / k_probe_tpr (thread-per-row, default)
+ *                     — fused gen-keys + cluster-hash probe + per-mode
+ *                       epilogue; replaces gpu_hash.cu:94-132,149-445
+ *   k_scan_part/mid/add — 3-phase exclusive scan with DEVICE-side length
+ *                       (replaces the thrust scans, gpu_hash.cu:587-596)
+ *   k_expand          — output-centric load-balanced expansion
+ *                       (replaces gpu_hash.cu:762-834; no 20-col cap)
+ *   k_compact         — flag compaction (replaces gpu_hash.cu:523-585)
+ *   k_copy_list       — i2u/c2u list materialisation on device (the
  *                       reference ran these on CPU, gpu_engine.hpp:63-123)
  *   k_dst_histogram / k_dst_scatter — fork-join split by vid % ndst
- *                       replaces gpu_hash.cu:600-760
- *   scans             — hipCUB (rocPRIM) exclusive sums, replacing thrust
- *                       (gpu_hash.cu:587-596,750-751)
+ *                       (replaces gpu_hash.cu:600-760)
  */
 #include "wk_store.h"
 #include "../../include/wukong_abi.h"
 
 #include <hip/hip_runtime.h>
-#include <hipcub/hipcub.hpp>
 
 #include <vector>
 #include <algorithm>
@@ -75,6 +81,18 @@ enum { PM_SIZE = 0,    // k2u: write edge count per row
        PM_COL = 2,     // k2k: flag = (row's other col in edge list)
        PM_LIST = 3 };  // c2k/i2k: flag = (row col value in a FIXED list)
 
+// device query state (see engine): [0]=nrows [1]=scan total [2]=overflow
+// flag [3]=required rows; stats[0..6] = algorithmic bytes per category
+enum { S_NROWS = 0, S_TOTAL = 1, S_ERR = 2, S_REQ = 3, S_WORDS = 4 };
+enum { CAT_PROBE = 0, CAT_SCAN, CAT_EXPAND, CAT_FILTER, CAT_COPY, CAT_SPLIT,
+       CAT_OTHER, CAT_COUNT };
+
+__device__ __forceinline__ void count_bytes(uint64_t *stats, int cat,
+                                            uint64_t bytes) {
+    if (blockIdx.x == 0 && threadIdx.x == 0 && stats)
+        atomicAdd((unsigned long long *)&stats[cat], (unsigned long long)bytes);
+}
+
 // sorted-membership test: edge lists are ascending (loader sort,
 // base_loader.hpp:367-377) so binary search replaces the reference's
 // linear scan (sparql.hpp:430-470) with identical keep-row semantics.
@@ -90,94 +108,27 @@ __device__ __forceinline__ bool bsearch_u32(const sid_t *a, uint64_t n, sid_t x)
     return false;
 }
 
-// Cooperative cluster-hash probe: 8 lanes per input row load one 128-B
-// bucket coalesced (8 x 16 B slots), ballot-match, follow the chain slot
-// (gstore.hpp:341-361 semantics).  Fused with the per-mode epilogue.
-__global__ void k_probe(const vertex_t *__restrict__ verts,
-                        const sid_t *__restrict__ edges,
-                        uint64_t bucket_start, uint64_t num_buckets,
-                        const sid_t *__restrict__ tbl, int64_t nrows, int ncols,
-                        int col,            // column holding the start var
-                        uint32_t pid, int dir, int key_mode, int probe_mode,
-                        int col2,           // PM_COL: column of the bound var
-                        sid_t cval,         // PM_CONST: the constant
-                        uint64_t list_off, uint64_t list_sz,  // PM_LIST
-                        uint64_t *__restrict__ d_eoff,
-                        uint64_t *__restrict__ d_cnt)
-{
-    const int sub = threadIdx.x & 7;                    // lane within group
-    const int64_t grp0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 3;
-    const int64_t ngrp = ((int64_t)gridDim.x * blockDim.x) >> 3;
-
-    for (int64_t r = grp0; r < nrows; r += ngrp) {
-        uint64_t key;
-        if (probe_mode == PM_LIST) {
-            // no probe at all: membership of tbl[r,col] in the fixed list
-            if (sub == 0) {
-                sid_t v = tbl[r * ncols + col];
-                d_cnt[r] = bsearch_u32(edges + list_off, list_sz, v) ? 1 : 0;
-            }
-            continue;
-        }
-        {
-            sid_t v = tbl[r * ncols + col];
-            key = (key_mode == PK_NORMAL) ? key_pack(v, pid, (uint64_t)dir)
-                                          : key_pack(0, v, (uint64_t)dir);
-        }
-        uint64_t bucket = bucket_start + hash_u64(key) % num_buckets;
-        uint64_t eoff = 0, esz = 0;
-        while (true) {
-            const vertex_t slot = verts[bucket * ASSOC + sub];
-            bool match = (sub < ASSOC - 1) && (slot.key == key);
-            uint64_t ball = __ballot(match);
-            uint32_t gmask = (uint32_t)((ball >> ((threadIdx.x & 63) & ~7)) & 0xffu);
-            if (gmask) {
-                if (match) { eoff = ptr_off(slot.ptr); esz = ptr_size(slot.ptr); }
-                // broadcast ptr to the whole group via shuffle from the hit lane
-                int src = (threadIdx.x & ~7 & 63) + (__ffs(gmask) - 1);
-                uint64_t p2 = __shfl((unsigned long long)slot.ptr, src);
-                eoff = ptr_off(p2); esz = ptr_size(p2);
-                break;
-            }
-            // chain via slot 7 (gstore.hpp:826)
-            uint64_t chain = __shfl((unsigned long long)slot.key,
-                                    (threadIdx.x & ~7 & 63) + (ASSOC - 1));
-            if (chain == KEY_EMPTY) { esz = 0; break; }
-            bucket = key_vid(chain);
-        }
-        if (sub == 0) {
-            switch (probe_mode) {
-            case PM_SIZE:
-                d_eoff[r] = eoff;
-                d_cnt[r] = esz;
-                break;
-            case PM_CONST:
-                d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz, cval)) ? 1 : 0;
-                break;
-            case PM_COL: {
-                sid_t tgt = tbl[r * ncols + col2];
-                d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz, tgt)) ? 1 : 0;
-                break;
-            }
-            }
-        }
-    }
-}
-
-// Thread-per-row probe variant: one LANE walks one row's bucket chain
-// (7 slot compares in one 128-B line, L1-resident after the first 16-B
-// fetch).  64 rows in flight per wave vs the cooperative kernel's 8 —
-// 8x the memory-level parallelism for DRAM-random probe patterns.
+// Thread-per-row probe (default): one LANE walks one row's bucket chain
+// (7 slot compares in one 128-B line).  64 rows in flight per wave —
+// maximal memory-level parallelism for DRAM-random probes
+// (gstore.hpp:341-361 semantics; dispatch per sparql.hpp:1016-1058).
 __global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
                             const sid_t *__restrict__ edges,
                             uint64_t bucket_start, uint64_t num_buckets,
-                            const sid_t *__restrict__ tbl, int64_t nrows, int ncols,
+                            const sid_t *__restrict__ tbl, int ncols,
                             int col, uint32_t pid, int dir, int key_mode,
                             int probe_mode, int col2, sid_t cval,
                             uint64_t list_off, uint64_t list_sz,
+                            const uint64_t *__restrict__ d_state,
+                            uint64_t *__restrict__ d_stats,
                             uint64_t *__restrict__ d_eoff,
                             uint64_t *__restrict__ d_cnt)
 {
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, probe_mode == PM_LIST ? CAT_FILTER : CAT_PROBE,
+                (uint64_t)nrows * (probe_mode == PM_LIST
+                                       ? 12
+                                       : (4 + 128 + 8 + (probe_mode != PM_SIZE ? 64 : 0))));
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
          r += (int64_t)gridDim.x * blockDim.x) {
         if (probe_mode == PM_LIST) {
@@ -192,7 +143,6 @@ __global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
         uint64_t eoff = 0, esz = 0;
         while (true) {
             const vertex_t *b = &verts[bucket * ASSOC];
-            // one 128-B line; issue wide reads, compare in registers
             uint64_t k0 = b[0].key, k1 = b[1].key, k2 = b[2].key, k3 = b[3].key;
             uint64_t k4 = b[4].key, k5 = b[5].key, k6 = b[6].key, k7 = b[7].key;
             int hit = -1;
@@ -204,8 +154,8 @@ __global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
             else if (k5 == key) hit = 5;
             else if (k6 == key) hit = 6;
             if (hit >= 0) {
-                uint64_t p = b[hit].ptr;
-                eoff = ptr_off(p); esz = ptr_size(p);
+                uint64_t pp = b[hit].ptr;
+                eoff = ptr_off(pp); esz = ptr_size(pp);
                 break;
             }
             if (k7 == KEY_EMPTY) { esz = 0; break; }
@@ -227,19 +177,192 @@ __global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
     }
 }
 
+// Cooperative 8-lane-per-row probe (WK_PROBE=coop, kept for A/B): lanes
+// load one 128-B bucket coalesced and ballot-match.
+__global__ void k_probe(const vertex_t *__restrict__ verts,
+                        const sid_t *__restrict__ edges,
+                        uint64_t bucket_start, uint64_t num_buckets,
+                        const sid_t *__restrict__ tbl, int ncols,
+                        int col, uint32_t pid, int dir, int key_mode,
+                        int probe_mode, int col2, sid_t cval,
+                        uint64_t list_off, uint64_t list_sz,
+                        const uint64_t *__restrict__ d_state,
+                        uint64_t *__restrict__ d_stats,
+                        uint64_t *__restrict__ d_eoff,
+                        uint64_t *__restrict__ d_cnt)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, probe_mode == PM_LIST ? CAT_FILTER : CAT_PROBE,
+                (uint64_t)nrows * (probe_mode == PM_LIST
+                                       ? 12
+                                       : (4 + 128 + 8 + (probe_mode != PM_SIZE ? 64 : 0))));
+    const int sub = threadIdx.x & 7;
+    const int64_t grp0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 3;
+    const int64_t ngrp = ((int64_t)gridDim.x * blockDim.x) >> 3;
+
+    for (int64_t r = grp0; r < nrows; r += ngrp) {
+        uint64_t key;
+        if (probe_mode == PM_LIST) {
+            if (sub == 0) {
+                sid_t v = tbl[r * ncols + col];
+                d_cnt[r] = bsearch_u32(edges + list_off, list_sz, v) ? 1 : 0;
+            }
+            continue;
+        }
+        {
+            sid_t v = tbl[r * ncols + col];
+            key = (key_mode == PK_NORMAL) ? key_pack(v, pid, (uint64_t)dir)
+                                          : key_pack(0, v, (uint64_t)dir);
+        }
+        uint64_t bucket = bucket_start + hash_u64(key) % num_buckets;
+        uint64_t eoff = 0, esz = 0;
+        while (true) {
+            const vertex_t slot = verts[bucket * ASSOC + sub];
+            bool match = (sub < ASSOC - 1) && (slot.key == key);
+            uint64_t ball = __ballot(match);
+            uint32_t gmask = (uint32_t)((ball >> ((threadIdx.x & 63) & ~7)) & 0xffu);
+            if (gmask) {
+                int src = (threadIdx.x & ~7 & 63) + (__ffs(gmask) - 1);
+                uint64_t p2 = __shfl((unsigned long long)slot.ptr, src);
+                eoff = ptr_off(p2); esz = ptr_size(p2);
+                break;
+            }
+            uint64_t chain = __shfl((unsigned long long)slot.key,
+                                    (threadIdx.x & ~7 & 63) + (ASSOC - 1));
+            if (chain == KEY_EMPTY) { esz = 0; break; }
+            bucket = key_vid(chain);
+        }
+        if (sub == 0) {
+            switch (probe_mode) {
+            case PM_SIZE: d_eoff[r] = eoff; d_cnt[r] = esz; break;
+            case PM_CONST:
+                d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz, cval)) ? 1 : 0;
+                break;
+            case PM_COL: {
+                sid_t tgt = tbl[r * ncols + col2];
+                d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz, tgt)) ? 1 : 0;
+                break;
+            }
+            }
+        }
+    }
+}
+
+// ---- 3-phase exclusive scan with device-side length ------------------
+// m = state[S_NROWS] + 1 elements of cnt (cnt[m-1] pre-zeroed by
+// k_zero_at) -> prefix; total -> state[S_TOTAL].
+static const int SCAN_T = 256;
+
+__global__ void k_zero_at(uint64_t *cnt, const uint64_t *__restrict__ d_state) {
+    cnt[d_state[S_NROWS]] = 0;
+}
+
+__global__ void k_scan_part(const uint64_t *__restrict__ cnt,
+                            const uint64_t *__restrict__ d_state,
+                            uint64_t *__restrict__ pre,
+                            uint64_t *__restrict__ bsums,
+                            uint64_t *__restrict__ d_stats)
+{
+    count_bytes(d_stats, CAT_SCAN, (d_state[S_NROWS] + 1) * 8 * 2);
+    const int64_t m = (int64_t)d_state[S_NROWS] + 1;
+    const int64_t chunk = (m + gridDim.x - 1) / gridDim.x;
+    const int64_t start = (int64_t)blockIdx.x * chunk;
+    const int64_t end = min(start + chunk, m);
+    __shared__ uint64_t sh[SCAN_T];
+    uint64_t carry = 0;
+    for (int64_t base = start; base < end; base += SCAN_T) {
+        int64_t i = base + threadIdx.x;
+        uint64_t x = (i < end) ? cnt[i] : 0;
+        sh[threadIdx.x] = x;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint64_t v = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += v;
+            __syncthreads();
+        }
+        if (i < end) pre[i] = carry + sh[threadIdx.x] - x;
+        carry += sh[SCAN_T - 1];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
+}
+
+// single block: exclusive-scan the G block sums in place, total -> state
+__global__ void k_scan_mid(uint64_t *__restrict__ bsums, int G,
+                           uint64_t *__restrict__ d_state)
+{
+    __shared__ uint64_t sh[SCAN_T];
+    uint64_t carry = 0;
+    for (int base = 0; base < G; base += SCAN_T) {
+        int i = base + threadIdx.x;
+        uint64_t x = (i < G) ? bsums[i] : 0;
+        sh[threadIdx.x] = x;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint64_t v = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += v;
+            __syncthreads();
+        }
+        if (i < G) bsums[i] = carry + sh[threadIdx.x] - x;
+        carry += sh[SCAN_T - 1];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) d_state[S_TOTAL] = carry;
+}
+
+__global__ void k_scan_add(uint64_t *__restrict__ pre,
+                           const uint64_t *__restrict__ bsums, int G,
+                           const uint64_t *__restrict__ d_state)
+{
+    const int64_t m = (int64_t)d_state[S_NROWS] + 1;
+    const int64_t chunk = (m + G - 1) / G;
+    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < m;
+         i += (int64_t)gridDim.x * blockDim.x)
+        pre[i] += bsums[i / chunk];
+}
+
+// advance: nrows = min(total, cap); flag overflow for the host re-run
+// (replaces the reference's rbuf-overflow assert, gpu_engine_cuda.hpp:185)
+__global__ void k_commit(uint64_t *__restrict__ d_state, uint64_t cap) {
+    uint64_t t = d_state[S_TOTAL];
+    if (t > cap) {
+        d_state[S_ERR] = 1;
+        d_state[S_REQ] = max(d_state[S_REQ], t);
+        t = cap;
+    }
+    d_state[S_NROWS] = t;
+}
+
+__global__ void k_set_state(uint64_t *__restrict__ d_state, uint64_t nrows) {
+    d_state[S_NROWS] = nrows;
+}
+
+__global__ void k_publish_state(const uint64_t *__restrict__ d_state,
+                                const uint64_t *__restrict__ d_stats,
+                                uint64_t *__restrict__ h_pin) {
+    for (int i = 0; i < S_WORDS; i++) h_pin[i] = d_state[i];
+    for (int i = 0; i < CAT_COUNT; i++) h_pin[8 + i] = d_stats[i];
+}
+
 // Output-centric expansion: one thread per OUTPUT row; binary-search the
 // prefix array for the source row.  Consecutive threads write consecutive
 // rows -> coalesced stores (known_to_unknown semantics, sparql.hpp:325-367).
 __global__ void k_expand(const sid_t *__restrict__ tbl, int ncols,
                          const sid_t *__restrict__ edges,
                          const uint64_t *__restrict__ d_eoff,
-                         const uint64_t *__restrict__ d_prefix, int64_t nrows_in,
-                         sid_t *__restrict__ out, int64_t total)
+                         const uint64_t *__restrict__ d_prefix,
+                         const uint64_t *__restrict__ d_state, uint64_t cap,
+                         uint64_t *__restrict__ d_stats,
+                         sid_t *__restrict__ out)
 {
+    const int64_t nrows_in = (int64_t)d_state[S_NROWS];
+    const int64_t total = (int64_t)min(d_state[S_TOTAL], cap);
     const int oc = ncols + 1;
+    count_bytes(d_stats, CAT_EXPAND, (uint64_t)total * (4 + 4 * ncols + 4 * oc));
     for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < total;
          t += (int64_t)gridDim.x * blockDim.x) {
-        // upper_bound(prefix, t) - 1
         int64_t lo = 0, hi = nrows_in;
         while (lo + 1 < hi) {
             int64_t mid = (lo + hi) >> 1;
@@ -254,12 +377,17 @@ __global__ void k_expand(const sid_t *__restrict__ tbl, int ncols,
     }
 }
 
-// Flag-compaction (k2k/k2c/c2k/i2k keep-row semantics, sparql.hpp:455-476)
+// Flag-compaction (k2k/k2c/c2k/i2k keep-row semantics, sparql.hpp:455-476).
+// Output rows <= input rows <= cap: never overflows.
 __global__ void k_compact(const sid_t *__restrict__ tbl, int ncols,
                           const uint64_t *__restrict__ d_cnt,
                           const uint64_t *__restrict__ d_prefix,
-                          int64_t nrows_in, sid_t *__restrict__ out)
+                          const uint64_t *__restrict__ d_state,
+                          uint64_t *__restrict__ d_stats,
+                          sid_t *__restrict__ out)
 {
+    const int64_t nrows_in = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows_in * (16 + 4 * ncols));
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows_in;
          r += (int64_t)gridDim.x * blockDim.x) {
         if (!d_cnt[r]) continue;
@@ -272,20 +400,13 @@ __global__ void k_compact(const sid_t *__restrict__ tbl, int ncols,
 // i2u / c2u: materialise an edge/index list as a 1-column table
 // (index_to_unknown sparql.hpp:194-231 / const_to_unknown :238-285)
 __global__ void k_copy_list(const sid_t *__restrict__ edges, uint64_t off,
-                            uint64_t n, sid_t *__restrict__ out)
+                            uint64_t n, uint64_t *__restrict__ d_stats,
+                            sid_t *__restrict__ out)
 {
+    count_bytes(d_stats, CAT_COPY, n * 8);
     for (uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; t < n;
          t += (uint64_t)gridDim.x * blockDim.x)
         out[t] = edges[off + t];
-}
-
-// zero one u64 + publish the scan total straight into pinned host memory
-// (a 1-thread kernel each: keeps the hot path on the compute queue; the
-// 8-byte hipMemsetAsync/hipMemcpyAsync SDMA hops behind a large kernel
-// showed ~25 ms cross-queue stalls on this box)
-__global__ void k_zero_u64(uint64_t *p) { *p = 0; }
-__global__ void k_publish_u64(const uint64_t *src, uint64_t *dst_pinned) {
-    *dst_pinned = *src;
 }
 
 // fork-join split (generate_sub_query sparql.hpp:772-796): dst = vid % ndst
@@ -320,6 +441,10 @@ __global__ void k_dst_scatter(const sid_t *__restrict__ tbl, int64_t nrows,
     }
 }
 
+__global__ void k_zero_words(uint64_t *p, int n) {
+    for (int i = threadIdx.x; i < n; i += blockDim.x) p[i] = 0;
+}
+
 // ---------------------------------------------------------------------
 // engine
 // ---------------------------------------------------------------------
@@ -332,14 +457,9 @@ struct devbuf {
     int ensure(size_t bytes) {
         if (bytes <= cap) return WK_OK;
         if (p) (void)hipFree(p);
-        size_t want = bytes + bytes / 2;
-        if (hipMalloc(&p, want) != hipSuccess) {
-            p = nullptr; cap = 0;
-            if (hipMalloc(&p, bytes) != hipSuccess) return WK_ERR_HIP;
-            cap = bytes;
-            return WK_OK;
-        }
-        cap = want;
+        p = nullptr; cap = 0;
+        if (hipMalloc(&p, bytes) != hipSuccess) return WK_ERR_HIP;
+        cap = bytes;
         return WK_OK;
     }
     void release() { if (p) (void)hipFree(p); p = nullptr; cap = 0; }
@@ -348,10 +468,7 @@ struct devbuf {
 struct timed_launch {
     hipEvent_t start, stop;
     int cat;
-    double bytes;
 };
-
-enum { CAT_PROBE = 0, CAT_SCAN, CAT_EXPAND, CAT_FILTER, CAT_COPY, CAT_SPLIT, CAT_OTHER };
 
 }  // namespace
 
@@ -364,36 +481,40 @@ struct wk_engine {
     vertex_t *d_verts = nullptr;
     sid_t *d_edges = nullptr;
 
-    // dual result buffer (gpu_mem.hpp:116-124)
+    // dual result buffer (gpu_mem.hpp:116-124) + scratch, all sized by
+    // cap_rows (grow-only; overflow -> re-run)
     devbuf tbl[2];
-    int cur = 0;
-    int64_t nrows = 0;
-    int ncols = 0;
+    devbuf eoff, cnt, prefix, bsums, misc;
+    int64_t cap_rows = 0;
+    int cap_cols = 0;
 
-    // scratch
-    devbuf eoff, cnt, prefix, cubtmp, misc;
-    uint64_t *h_pin = nullptr;  // pinned: totals
-    void *h_stage = nullptr;    // pinned staging for result downloads
+    uint64_t *d_state = nullptr;  // S_WORDS words
+    uint64_t *d_stats = nullptr;  // CAT_COUNT words (algorithmic bytes)
+    uint64_t *h_pin = nullptr;    // pinned: state + stats snapshot
+    void *h_stage = nullptr;      // pinned staging for result downloads
     size_t h_stage_cap = 0;
+
+    int cur = 0;
+    int64_t nrows = 0;   // valid only after a sync point
+    int ncols = 0;
+    int64_t bound = 0;   // host-side upper bound on rows (grid sizing)
 
     // query state (host mirror of SPARQLQuery, query.hpp:560-594)
     std::vector<wk_pattern_t> pats;
     std::vector<int32_t> v2c;  // idx -> col (query.hpp:352-374)
     int nvars = 0;
     int step = 0;
-    int status = 0;
 
     bool probe_coop = false;  // WK_PROBE=coop selects the 8-lane kernel
 
-    // timing
+    // timing (WK_KERNEL_TIMING=1)
     bool timing = false;
     std::vector<timed_launch> pending;
-    double cat_usec[7] = {0};
-    double cat_bytes[7] = {0};
-    int64_t cat_n[7] = {0};
+    double cat_usec[CAT_COUNT] = {0};
+    double cat_bytes[CAT_COUNT] = {0};
+    int64_t cat_n[CAT_COUNT] = {0};
 
     int var2col(ssid_t v) const { return v < 0 ? v2c[-(v + 1)] : -1; }
-    // var_stat (query.hpp:341-349): >=0 CONST; known iff column assigned
     int var_stat(ssid_t v) const { return v >= 0 ? 2 : (var2col(v) >= 0 ? 1 : 0); }
 };
 
@@ -402,11 +523,14 @@ static int grid_for(int64_t work) {
     int64_t g = (work + BLOCK - 1) / BLOCK;
     return (int)(g < 1 ? 1 : (g > 2048 ? 2048 : g));
 }
+static int scan_grid(int64_t bound) {
+    int64_t g = (bound + SCAN_T) / SCAN_T + 1;
+    return (int)(g < 1 ? 1 : (g > 512 ? 512 : g));
+}
 
 static hipEvent_t ev_get() { hipEvent_t e; (void)hipEventCreate(&e); return e; }
 
-// WK_SPIN_SYNC=1: poll hipStreamQuery instead of blocking sync (the
-// blocking path showed intermittent ~25 ms interrupt-latency stalls)
+// WK_SPIN_SYNC=1: poll instead of blocking (diagnostic)
 static bool wk_spin_sync() {
     static int v = -1;
     if (v < 0) { const char *e = getenv("WK_SPIN_SYNC"); v = e ? atoi(e) : 0; }
@@ -424,27 +548,44 @@ static hipError_t stream_sync(hipStream_t s) {
     if ((eng)->timing) { t_s_ = ev_get(); t_e_ = ev_get();                  \
         (void)hipEventRecord(t_s_, (eng)->stream); }
 
-#define TIME_END(eng, category, nbytes)                                     \
+#define TIME_END(eng, category)                                             \
     if ((eng)->timing) { (void)hipEventRecord(t_e_, (eng)->stream);         \
-        (eng)->pending.push_back({t_s_, t_e_, (category), (double)(nbytes)}); }
+        (eng)->pending.push_back({t_s_, t_e_, (category)}); }
 
-static const char *CAT_NAMES[7] = {"probe", "scan", "expand", "filter",
-                                   "copy", "split", "other"};
+static const char *CAT_NAMES[CAT_COUNT] = {"probe", "scan", "expand", "filter",
+                                           "copy", "split", "other"};
 
 static void resolve_timing(wk_engine *e) {
     for (auto &t : e->pending) {
         float ms = 0;
         (void)hipEventElapsedTime(&ms, t.start, t.stop);
         if (wk_verbose_lvl() >= 3)
-            fprintf(stderr, "[launch] %s %.0fus %.2fGB/s\n", CAT_NAMES[t.cat],
-                    ms * 1e3, ms > 0 ? t.bytes / (ms * 1e6) : 0.0);
+            fprintf(stderr, "[launch] %s %.0fus\n", CAT_NAMES[t.cat], ms * 1e3);
         e->cat_usec[t.cat] += ms * 1000.0;
-        e->cat_bytes[t.cat] += t.bytes;
         e->cat_n[t.cat]++;
         (void)hipEventDestroy(t.start);
         (void)hipEventDestroy(t.stop);
     }
     e->pending.clear();
+}
+
+// grow scratch + rbufs to hold `rows` rows of up to `cols` columns
+static int32_t grow_caps(wk_engine *e, int64_t rows, int cols) {
+    rows = std::max<int64_t>(rows, 1 << 20);
+    cols = std::max(cols, std::max(e->cap_cols, 1));
+    if (rows <= e->cap_rows && cols <= e->cap_cols) return WK_OK;
+    rows = std::max(rows, e->cap_rows);
+    // one sync before freeing buffers that in-flight kernels may use
+    HIP_CHECK(stream_sync(e->stream));
+    if (e->tbl[0].ensure((size_t)rows * cols * 4)) return WK_ERR_HIP;
+    if (e->tbl[1].ensure((size_t)rows * cols * 4)) return WK_ERR_HIP;
+    if (e->cnt.ensure((size_t)(rows + 1) * 8)) return WK_ERR_HIP;
+    if (e->eoff.ensure((size_t)(rows + 1) * 8)) return WK_ERR_HIP;
+    if (e->prefix.ensure((size_t)(rows + 1) * 8)) return WK_ERR_HIP;
+    if (e->bsums.ensure(520 * 8)) return WK_ERR_HIP;
+    e->cap_rows = rows;
+    e->cap_cols = cols;
+    return WK_OK;
 }
 
 extern "C" wk_engine_t *wk_engine_create(const wk_store_t *st, int32_t device) {
@@ -458,17 +599,21 @@ extern "C" wk_engine_t *wk_engine_create(const wk_store_t *st, int32_t device) {
     size_t eb = st->edges.size() * sizeof(sid_t);
     if (hipMalloc(&e->d_verts, vb ? vb : 16) != hipSuccess ||
         hipMalloc(&e->d_edges, eb ? eb : 16) != hipSuccess ||
+        hipMalloc(&e->d_state, (S_WORDS + CAT_COUNT + 2) * 8) != hipSuccess ||
         hipHostMalloc(&e->h_pin, 64 * sizeof(uint64_t)) != hipSuccess) {
         wk_engine_destroy(e);
         return nullptr;
     }
+    e->d_stats = e->d_state + S_WORDS;
     if (hipMemcpy(e->d_verts, st->vertices.data(), vb, hipMemcpyHostToDevice) != hipSuccess ||
-        hipMemcpy(e->d_edges, st->edges.data(), eb, hipMemcpyHostToDevice) != hipSuccess) {
+        hipMemcpy(e->d_edges, st->edges.data(), eb, hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemset(e->d_state, 0, (S_WORDS + CAT_COUNT + 2) * 8) != hipSuccess) {
         wk_engine_destroy(e);
         return nullptr;
     }
     if (const char *t = getenv("WK_KERNEL_TIMING")) e->timing = atoi(t) != 0;
     if (const char *pv = getenv("WK_PROBE")) e->probe_coop = !strcmp(pv, "coop");
+    if (grow_caps(e, 1 << 20, 4) != WK_OK) { wk_engine_destroy(e); return nullptr; }
     return e;
 }
 
@@ -477,9 +622,10 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
     resolve_timing(e);
     for (int i = 0; i < 2; i++) e->tbl[i].release();
     e->eoff.release(); e->cnt.release(); e->prefix.release();
-    e->cubtmp.release(); e->misc.release();
+    e->bsums.release(); e->misc.release();
     if (e->d_verts) (void)hipFree(e->d_verts);
     if (e->d_edges) (void)hipFree(e->d_edges);
+    if (e->d_state) (void)hipFree(e->d_state);
     if (e->h_pin) (void)hipHostFree(e->h_pin);
     if (e->h_stage) (void)hipHostFree(e->h_stage);
     if (e->stream) (void)hipStreamDestroy(e->stream);
@@ -495,7 +641,25 @@ extern "C" int32_t wk_engine_begin_query(wk_engine_t *e, const wk_plan_t *plan) 
     e->nrows = 0;
     e->ncols = 0;
     e->cur = 0;
-    e->status = 0;
+    e->bound = 0;
+    int32_t rc = grow_caps(e, e->cap_rows, plan->nvars);
+    if (rc) return rc;
+    // reset nrows + overflow flags on device (async, cheap kernel)
+    hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream, e->d_state,
+                       S_WORDS);
+    return WK_OK;
+}
+
+static int32_t load_common(wk_engine *e, int64_t nrows, int32_t ncols,
+                           const int32_t *v2c_map, int32_t pattern_step) {
+    e->cur = 0;
+    e->nrows = nrows;
+    e->bound = nrows;
+    e->ncols = ncols;
+    if (v2c_map) e->v2c.assign(v2c_map, v2c_map + e->nvars);
+    e->step = pattern_step;
+    hipLaunchKernelGGL(k_set_state, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                       (uint64_t)nrows);
     return WK_OK;
 }
 
@@ -503,192 +667,142 @@ extern "C" int32_t wk_engine_load_rbuf(wk_engine_t *e, const sid_t *table,
                                        int64_t nrows, int32_t ncols,
                                        const int32_t *v2c_map, int32_t pattern_step) {
     if (!e) return WK_ERR_STATE;
+    int32_t rc = grow_caps(e, nrows, std::max(ncols, 1));
+    if (rc) return rc;
     size_t bytes = (size_t)nrows * ncols * sizeof(sid_t);
-    if (e->tbl[0].ensure(bytes ? bytes : 4)) return WK_ERR_HIP;
     if (bytes)
-        if (hipMemcpyAsync(e->tbl[0].p, table, bytes, hipMemcpyHostToDevice,
-                           e->stream) != hipSuccess) return WK_ERR_HIP;
-    e->cur = 0;
-    e->nrows = nrows;
-    e->ncols = ncols;
-    if (v2c_map) e->v2c.assign(v2c_map, v2c_map + e->nvars);
-    e->step = pattern_step;
-    return WK_OK;
+        HIP_CHECK(hipMemcpyAsync(e->tbl[0].p, table, bytes, hipMemcpyHostToDevice,
+                                 e->stream));
+    return load_common(e, nrows, ncols, v2c_map, pattern_step);
 }
 
 extern "C" int32_t wk_engine_load_rbuf_device(wk_engine_t *e, const sid_t *dev_table,
                                               int64_t nrows, int32_t ncols,
                                               const int32_t *v2c_map, int32_t pattern_step) {
     if (!e) return WK_ERR_STATE;
+    int32_t rc = grow_caps(e, nrows, std::max(ncols, 1));
+    if (rc) return rc;
     size_t bytes = (size_t)nrows * ncols * sizeof(sid_t);
-    if (e->tbl[0].ensure(bytes ? bytes : 4)) return WK_ERR_HIP;
     if (bytes)
-        if (hipMemcpyAsync(e->tbl[0].p, dev_table, bytes, hipMemcpyDeviceToDevice,
-                           e->stream) != hipSuccess) return WK_ERR_HIP;
-    e->cur = 0;
-    e->nrows = nrows;
-    e->ncols = ncols;
-    if (v2c_map) e->v2c.assign(v2c_map, v2c_map + e->nvars);
-    e->step = pattern_step;
-    return WK_OK;
+        HIP_CHECK(hipMemcpyAsync(e->tbl[0].p, dev_table, bytes,
+                                 hipMemcpyDeviceToDevice, e->stream));
+    return load_common(e, nrows, ncols, v2c_map, pattern_step);
 }
 
-// exclusive scan of cnt[0..n] (n+1 entries; cnt[n] pre-zeroed) into prefix;
-// returns total via h_pin[0]
-static int32_t scan_counts(wk_engine *e, int64_t n, uint64_t *total_out) {
-    if (e->prefix.ensure((size_t)(n + 1) * 8)) return WK_ERR_HIP;
-    size_t tmp = 0;
-    uint64_t *d_cnt = (uint64_t *)e->cnt.p;
-    uint64_t *d_pre = (uint64_t *)e->prefix.p;
-    hipcub::DeviceScan::ExclusiveSum(nullptr, tmp, d_cnt, d_pre, n + 1, e->stream);
-    if (e->cubtmp.ensure(tmp)) return WK_ERR_HIP;
-    // verbose>=2: bracket every stream op to localise stalls
-    const bool dbg = wk_verbose_lvl() >= 2;
-    hipEvent_t ev[4];
-    if (dbg) for (int i = 0; i < 4; i++) (void)hipEventCreate(&ev[i]);
-    if (dbg) (void)hipEventRecord(ev[0], e->stream);
-    hipLaunchKernelGGL(k_zero_u64, dim3(1), dim3(1), 0, e->stream, d_cnt + n);
-    if (dbg) (void)hipEventRecord(ev[1], e->stream);
+// enqueue the device-length scan chain: cnt[0..n] -> prefix, total ->
+// state[S_TOTAL] (no host sync)
+static void enqueue_scan(wk_engine *e) {
+    const int G = scan_grid(e->bound);
+    hipLaunchKernelGGL(k_zero_at, dim3(1), dim3(1), 0, e->stream,
+                       (uint64_t *)e->cnt.p, e->d_state);
     TIME_BEGIN(e);
-    hipcub::DeviceScan::ExclusiveSum(e->cubtmp.p, tmp, d_cnt, d_pre, n + 1, e->stream);
-    TIME_END(e, CAT_SCAN, (double)(n + 1) * 16);
-    if (dbg) (void)hipEventRecord(ev[2], e->stream);
-    hipLaunchKernelGGL(k_publish_u64, dim3(1), dim3(1), 0, e->stream, d_pre + n,
-                       e->h_pin);
-    if (dbg) (void)hipEventRecord(ev[3], e->stream);
-    double ts = now_us();
+    hipLaunchKernelGGL(k_scan_part, dim3(G), dim3(SCAN_T), 0, e->stream,
+                       (uint64_t *)e->cnt.p, e->d_state, (uint64_t *)e->prefix.p,
+                       (uint64_t *)e->bsums.p, e->d_stats);
+    hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0, e->stream,
+                       (uint64_t *)e->bsums.p, G, e->d_state);
+    hipLaunchKernelGGL(k_scan_add, dim3(G), dim3(SCAN_T), 0, e->stream,
+                       (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
+                       e->d_state);
+    TIME_END(e, CAT_SCAN);
+}
+
+// publish device state to pinned memory + sync; refresh host nrows/stats
+static int32_t sync_state(wk_engine *e) {
+    hipLaunchKernelGGL(k_publish_state, dim3(1), dim3(1), 0, e->stream,
+                       e->d_state, e->d_stats, e->h_pin);
     HIP_CHECK(stream_sync(e->stream));
-    if (dbg) {
-        float pre = 0, ms = 0, sc = 0, d2h = 0;
-        (void)hipEventElapsedTime(&ms, ev[0], ev[1]);
-        (void)hipEventElapsedTime(&sc, ev[1], ev[2]);
-        (void)hipEventElapsedTime(&d2h, ev[2], ev[3]);
-        fprintf(stderr,
-                "[scan] n=%lld sync_us=%.0f memset_us=%.0f scan_us=%.0f d2h_us=%.0f\n",
-                (long long)n, now_us() - ts, ms * 1e3, sc * 1e3, d2h * 1e3);
-        (void)pre;
-        for (int i = 0; i < 4; i++) (void)hipEventDestroy(ev[i]);
-    }
-    *total_out = e->h_pin[0];
+    resolve_timing(e);
+    e->nrows = (int64_t)e->h_pin[S_NROWS];
+    e->bound = e->nrows;
+    for (int i = 0; i < CAT_COUNT; i++) e->cat_bytes[i] = (double)e->h_pin[8 + i];
+    if (e->h_pin[S_ERR]) return WK_ERR_CAP;
     return WK_OK;
 }
 
-// Run one pattern — dispatch per sparql.hpp:1016-1058.
-extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *nrows_out);
-extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_out) {
-    if (wk_verbose_lvl() < 2) return wk_engine_execute_one_pattern_impl(e, nrows_out);
-    double t0 = now_us();
-    int64_t rin = e ? e->nrows : -1;
-    int step = e ? e->step : -1;
-    int32_t rc = wk_engine_execute_one_pattern_impl(e, nrows_out);
-    fprintf(stderr, "[pat] step=%d rows_in=%lld rows_out=%lld host_us=%.0f\n",
-            step, (long long)rin, (long long)(e ? e->nrows : -1), now_us() - t0);
-    return rc;
-}
-
-extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *nrows_out) {
+// Run one pattern — dispatch per sparql.hpp:1016-1058.  Fully async: row
+// counts live on device; a non-NULL nrows_out forces a sync (step API).
+static int32_t exec_pattern(wk_engine *e) {
     if (!e || e->step >= (int)e->pats.size()) return WK_ERR_STATE;
     const wk_store *st = e->st;
     const wk_pattern_t pat = e->pats[e->step];
     const ssid_t s = pat.subject, p = pat.predicate, o = pat.object;
     const int dir = pat.direction;
     sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
-    sid_t *out_tbl;
+    sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
 
-    // ---- step 0, start from index (query.hpp:660-682) ----
-    if (e->step == 0 && s >= 0 && is_tpid(s)) {
-        // index_to_unknown (sparql.hpp:194-231); i2k not reachable at step 0
+    // ---- step 0: index start (query.hpp:660-682) / const start ----
+    const bool index_start = (e->step == 0 && s >= 0 && is_tpid(s));
+    if (index_start || (s >= 0 && o < 0 && e->var2col(o) < 0)) {
+        // index_to_unknown (sparql.hpp:194-231) or const_to_unknown
+        // (:238-285): materialise a host-probed list as a 1-col table
+        if (!index_start && e->ncols != 0) return WK_ERR_PLAN;
         uint64_t sz = 0;
-        const seg_t *seg = st->seg_of(0, (uint64_t)s, dir);
-        uint64_t off = 0;
-        if (seg) {
-            const sid_t *ptr = store_get(*st, 0, (uint64_t)s, dir, &sz);
-            off = ptr ? (uint64_t)(ptr - st->edges.data()) : 0;
-        }
-        if (e->tbl[1].ensure((size_t)(sz ? sz : 1) * 4)) return WK_ERR_HIP;
-        out_tbl = (sid_t *)e->tbl[1].p;
+        const sid_t *ptr = index_start ? store_get(*st, 0, (uint64_t)s, dir, &sz)
+                                       : store_get(*st, (uint64_t)s, (uint64_t)p, dir, &sz);
+        uint64_t off = ptr ? (uint64_t)(ptr - st->edges.data()) : 0;
+        int32_t rc = grow_caps(e, (int64_t)sz, e->nvars);
+        if (rc) return rc;
+        out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
         if (sz) {
             TIME_BEGIN(e);
-            hipLaunchKernelGGL(k_copy_list, dim3(grid_for((int64_t)sz)), dim3(BLOCK), 0,
-                               e->stream, e->d_edges, off, sz, out_tbl);
-            TIME_END(e, CAT_COPY, (double)sz * 8);
+            hipLaunchKernelGGL(k_copy_list, dim3(grid_for((int64_t)sz)), dim3(BLOCK),
+                               0, e->stream, e->d_edges, off, sz, e->d_stats, out_tbl);
+            TIME_END(e, CAT_COPY);
         }
-        e->cur = 1;
+        hipLaunchKernelGGL(k_set_state, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                           sz);
+        e->cur ^= 1;
         e->nrows = (int64_t)sz;
+        e->bound = (int64_t)sz;
         e->ncols = 1;
         e->v2c[-(o + 1)] = 0;
         e->step++;
-        if (nrows_out) *nrows_out = e->nrows;
         return WK_OK;
     }
 
-    // ---- const start ----
+    // ---- const_to_known (sparql.hpp:138-186): fixed-list membership ----
     if (s >= 0) {
+        int col = e->var2col(o);
+        if (col < 0) return WK_ERR_PLAN;
         uint64_t sz = 0;
         const sid_t *ptr = store_get(*st, (uint64_t)s, (uint64_t)p, dir, &sz);
         uint64_t off = ptr ? (uint64_t)(ptr - st->edges.data()) : 0;
-        if (o < 0 && e->var2col(o) < 0) {
-            // const_to_unknown (sparql.hpp:238-285): MUST be first pattern
-            if (e->ncols != 0) return WK_ERR_PLAN;
-            if (e->tbl[1].ensure((size_t)(sz ? sz : 1) * 4)) return WK_ERR_HIP;
-            out_tbl = (sid_t *)e->tbl[1].p;
-            if (sz) {
-                TIME_BEGIN(e);
-                hipLaunchKernelGGL(k_copy_list, dim3(grid_for((int64_t)sz)), dim3(BLOCK),
-                                   0, e->stream, e->d_edges, off, sz, out_tbl);
-                TIME_END(e, CAT_COPY, (double)sz * 8);
-            }
-            e->cur = 1;
-            e->nrows = (int64_t)sz;
-            e->ncols = 1;
-            e->v2c[-(o + 1)] = 0;
-            e->step++;
-            if (nrows_out) *nrows_out = e->nrows;
-            return WK_OK;
-        }
-        // const_to_known (sparql.hpp:138-186): membership filter on col(o)
-        int col = e->var2col(o);
-        if (col < 0) return WK_ERR_PLAN;
-        int64_t R = e->nrows;
-        if (e->cnt.ensure((size_t)(R + 1) * 8)) return WK_ERR_HIP;
         TIME_BEGIN(e);
         if (e->probe_coop)
-            hipLaunchKernelGGL(k_probe, dim3(grid_for(R * 8)), dim3(BLOCK), 0, e->stream,
-                               e->d_verts, e->d_edges, 0, 1, cur_tbl, R, e->ncols, col,
-                               0u, dir, PK_NORMAL, PM_LIST, 0, 0u, off, sz,
+            hipLaunchKernelGGL(k_probe, dim3(grid_for(e->bound * 8)), dim3(BLOCK), 0,
+                               e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
+                               e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
+                               off, sz, e->d_state, e->d_stats,
                                (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
         else
-            hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(R)), dim3(BLOCK), 0, e->stream,
-                               e->d_verts, e->d_edges, 0, 1, cur_tbl, R, e->ncols, col,
-                               0u, dir, PK_NORMAL, PM_LIST, 0, 0u, off, sz,
+            hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
+                               e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
+                               e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
+                               off, sz, e->d_state, e->d_stats,
                                (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
-        TIME_END(e, CAT_FILTER, (double)R * 12);
-        uint64_t total = 0;
-        int32_t rc = scan_counts(e, R, &total);
-        if (rc) return rc;
-        if (e->tbl[e->cur ^ 1].ensure((size_t)(total ? total : 1) * e->ncols * 4))
-            return WK_ERR_HIP;
-        out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
-        if (R) {
+        TIME_END(e, CAT_FILTER);
+        enqueue_scan(e);
+        {
             TIME_BEGIN(e);
-            hipLaunchKernelGGL(k_compact, dim3(grid_for(R)), dim3(BLOCK), 0, e->stream,
-                               cur_tbl, e->ncols, (uint64_t *)e->cnt.p,
-                               (uint64_t *)e->prefix.p, R, out_tbl);
-            TIME_END(e, CAT_FILTER, (double)(R * 16 + (double)total * e->ncols * 8));
+            hipLaunchKernelGGL(k_compact, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
+                               e->stream, cur_tbl, e->ncols, (uint64_t *)e->cnt.p,
+                               (uint64_t *)e->prefix.p, e->d_state, e->d_stats,
+                               out_tbl);
+            TIME_END(e, CAT_FILTER);
         }
+        hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                           (uint64_t)e->cap_rows);
         e->cur ^= 1;
-        e->nrows = (int64_t)total;
         e->step++;
-        if (nrows_out) *nrows_out = e->nrows;
         return WK_OK;
     }
 
     // ---- known start ----
     int col = e->var2col(s);
     if (col < 0) return WK_ERR_PLAN;  // UNKNOWN start: invalid plan (sparql.hpp:1044-1049)
-    int64_t R = e->nrows;
 
-    // segment of the fixed (pid,dir) — known_to_unknown's TYPE_ID/IN case
+    // segment of the fixed (pid,dir); known_to_unknown's TYPE_ID/IN case
     // probes the index segment with per-row type keys (sparql.hpp:340-343)
     int key_mode = PK_NORMAL;
     const seg_t *seg;
@@ -696,89 +810,111 @@ extern "C" int32_t wk_engine_execute_one_pattern_impl(wk_engine_t *e, int64_t *n
         key_mode = PK_INDEX;
         seg = &st->iseg[DIR_IN];
     } else {
-        seg = st->seg_of((uint64_t)1 << NBITS_IDX /*any normal vid*/, (uint64_t)p, dir);
+        seg = st->seg_of((uint64_t)1 << NBITS_IDX, (uint64_t)p, dir);
     }
 
     const int ostat = (o >= 0) ? 2 : (e->var2col(o) >= 0 ? 1 : 0);
-    uint64_t total = 0;
     if (!seg || seg->num_buckets == 0) {
-        // segment absent: every probe misses
+        // segment absent: every probe misses -> empty table
+        hipLaunchKernelGGL(k_set_state, dim3(1), dim3(1), 0, e->stream, e->d_state, 0);
+        e->nrows = 0;
+        e->bound = 0;
         if (ostat == 0) {
-            // k2u with no matches -> empty table, still adds the column
-            e->nrows = 0;
             e->v2c[-(o + 1)] = e->ncols;
             e->ncols += 1;
-            e->step++;
-            if (nrows_out) *nrows_out = 0;
-            return WK_OK;
         }
-        e->nrows = 0;
         e->step++;
-        if (nrows_out) *nrows_out = 0;
         return WK_OK;
     }
-
-    if (e->cnt.ensure((size_t)(R + 1) * 8) || e->eoff.ensure((size_t)(R ? R : 1) * 8))
-        return WK_ERR_HIP;
 
     int pmode = (ostat == 0) ? PM_SIZE : (ostat == 2 ? PM_CONST : PM_COL);
     int col2 = (ostat == 1) ? e->var2col(o) : 0;
     sid_t cval = (ostat == 2) ? (sid_t)o : 0;
-    if (R) {
+    {
         TIME_BEGIN(e);
         if (e->probe_coop)
-            hipLaunchKernelGGL(k_probe, dim3(grid_for(R * 8)), dim3(BLOCK), 0, e->stream,
-                               e->d_verts, e->d_edges, seg->bucket_start, seg->num_buckets,
-                               cur_tbl, R, e->ncols, col, (uint32_t)p, dir, key_mode, pmode,
-                               col2, cval, 0, 0, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+            hipLaunchKernelGGL(k_probe, dim3(grid_for(e->bound * 8)), dim3(BLOCK), 0,
+                               e->stream, e->d_verts, e->d_edges, seg->bucket_start,
+                               seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
+                               dir, key_mode, pmode, col2, cval, 0, 0, e->d_state,
+                               e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
         else
-            hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(R)), dim3(BLOCK), 0, e->stream,
-                               e->d_verts, e->d_edges, seg->bucket_start, seg->num_buckets,
-                               cur_tbl, R, e->ncols, col, (uint32_t)p, dir, key_mode, pmode,
-                               col2, cval, 0, 0, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
-        // bytes: key col + bucket + (scan: log2(deg) touches ~1 cacheline)
-        TIME_END(e, CAT_PROBE, (double)R * (4 + 128 + 8 + (pmode != PM_SIZE ? 64 : 0)));
+            hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
+                               e->stream, e->d_verts, e->d_edges, seg->bucket_start,
+                               seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
+                               dir, key_mode, pmode, col2, cval, 0, 0, e->d_state,
+                               e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+        TIME_END(e, CAT_PROBE);
     }
-    int32_t rc = scan_counts(e, R, &total);
-    if (rc) return rc;
+    enqueue_scan(e);
 
     if (pmode == PM_SIZE) {
-        // known_to_unknown: expand
+        // known_to_unknown: expand (bound multiplies by unknown fan-out;
+        // grid heuristic x16, correctness by grid-stride)
         int oc = e->ncols + 1;
-        if (e->tbl[e->cur ^ 1].ensure((size_t)(total ? total : 1) * oc * 4))
-            return WK_ERR_HIP;
-        out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
-        if (total) {
-            TIME_BEGIN(e);
-            hipLaunchKernelGGL(k_expand, dim3(grid_for((int64_t)total)), dim3(BLOCK), 0,
-                               e->stream, cur_tbl, e->ncols, e->d_edges,
-                               (uint64_t *)e->eoff.p, (uint64_t *)e->prefix.p, R,
-                               out_tbl, (int64_t)total);
-            TIME_END(e, CAT_EXPAND,
-                     (double)total * (4 + 4 * e->ncols + 4 * oc));
-        }
-        e->cur ^= 1;
-        e->nrows = (int64_t)total;
+        if (oc > e->cap_cols) return WK_ERR_STATE;  // begin_query sizes cap_cols>=nvars
+        TIME_BEGIN(e);
+        hipLaunchKernelGGL(k_expand, dim3(grid_for(e->cap_rows)), dim3(BLOCK), 0,
+                           e->stream, cur_tbl, e->ncols, e->d_edges,
+                           (uint64_t *)e->eoff.p, (uint64_t *)e->prefix.p,
+                           e->d_state, (uint64_t)e->cap_rows, e->d_stats, out_tbl);
+        TIME_END(e, CAT_EXPAND);
         e->v2c[-(o + 1)] = e->ncols;
         e->ncols = oc;
+        e->bound = e->cap_rows;  // fan-out unknown until a sync point
     } else {
-        // known_to_known / known_to_const: compact
-        if (e->tbl[e->cur ^ 1].ensure((size_t)(total ? total : 1) * e->ncols * 4))
-            return WK_ERR_HIP;
-        out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
-        if (R) {
-            TIME_BEGIN(e);
-            hipLaunchKernelGGL(k_compact, dim3(grid_for(R)), dim3(BLOCK), 0, e->stream,
-                               cur_tbl, e->ncols, (uint64_t *)e->cnt.p,
-                               (uint64_t *)e->prefix.p, R, out_tbl);
-            TIME_END(e, CAT_FILTER, (double)(R * 16 + (double)total * e->ncols * 8));
-        }
-        e->cur ^= 1;
-        e->nrows = (int64_t)total;
+        // known_to_known / known_to_const: compact (rows only shrink)
+        TIME_BEGIN(e);
+        hipLaunchKernelGGL(k_compact, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
+                           e->stream, cur_tbl, e->ncols, (uint64_t *)e->cnt.p,
+                           (uint64_t *)e->prefix.p, e->d_state, e->d_stats, out_tbl);
+        TIME_END(e, CAT_FILTER);
     }
+    hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                       (uint64_t)e->cap_rows);
+    e->cur ^= 1;
     e->step++;
-    if (nrows_out) *nrows_out = e->nrows;
     return WK_OK;
+}
+
+extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_out) {
+    double t0 = wk_verbose_lvl() >= 2 ? now_us() : 0;
+    if (!e) return WK_ERR_STATE;
+    const int64_t in_rows = e->nrows;  // valid in step mode (synced per step)
+    const int in_cur = e->cur;
+    int32_t rc = exec_pattern(e);
+    if (rc == WK_OK && nrows_out) {
+        // step-API callers get a synchronous row count; on overflow grow
+        // and re-run THIS pattern.  NOTE: grow_caps reallocates BOTH rbufs,
+        // so the input table must be restaged — we re-upload is impossible
+        // here; instead grow only the scratch by re-running via a fresh
+        // copy: since grow_caps frees tbl[], the step-API rerun only works
+        // when the input table can be regenerated (step 0) — otherwise
+        // callers should size caps up front.  In practice overflow implies
+        // a large table which implies caps already grown by run_query.
+        for (int attempt = 0; attempt < 6; attempt++) {
+            rc = sync_state(e);
+            if (rc != WK_ERR_CAP) break;
+            if (e->step - 1 != 0) return WK_ERR_CAP;  // cannot rerun mid-plan
+            int64_t need = (int64_t)e->h_pin[S_REQ];
+            e->cur = in_cur;
+            e->step--;
+            int32_t rc2 = grow_caps(e, need + need / 4, e->cap_cols);
+            if (rc2) return rc2;
+            hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                               e->d_state, S_WORDS);
+            hipLaunchKernelGGL(k_set_state, dim3(1), dim3(1), 0, e->stream,
+                               e->d_state, (uint64_t)in_rows);
+            rc = exec_pattern(e);
+            if (rc) return rc;
+        }
+        if (rc == WK_OK && nrows_out) *nrows_out = e->nrows;
+    }
+    if (wk_verbose_lvl() >= 2)
+        fprintf(stderr, "[pat] step=%d rows_out=%lld host_us=%.0f\n",
+                e ? e->step - 1 : -1, (long long)(e ? e->nrows : -1),
+                now_us() - t0);
+    return rc;
 }
 
 extern "C" int32_t wk_engine_pattern_step(const wk_engine_t *e) { return e->step; }
@@ -792,32 +928,40 @@ extern "C" int32_t wk_engine_generate_sub_query(wk_engine_t *e, int32_t ndst,
     const wk_pattern_t pat = e->pats[e->step];
     int col = e->var2col(pat.subject);
     if (col < 0) return WK_ERR_PLAN;
+    int32_t rc = sync_state(e);
+    if (rc) return rc;
     int64_t R = e->nrows;
     if (R > cap_rows) return WK_ERR_CAP;
     if (e->misc.ensure(128 * sizeof(unsigned long long))) return WK_ERR_HIP;
     unsigned long long *d_hist = (unsigned long long *)e->misc.p;
     unsigned long long *d_cursor = d_hist + 64;
-    HIP_CHECK(hipMemsetAsync(d_hist, 0, 128 * sizeof(unsigned long long), e->stream));
+    hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(128), 0, e->stream,
+                       (uint64_t *)e->misc.p, 128);
     sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
     if (R) {
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_dst_histogram, dim3(grid_for(R)), dim3(BLOCK), 0, e->stream,
-                           cur_tbl, R, e->ncols, col, ndst, d_hist);
-        TIME_END(e, CAT_SPLIT, (double)R * 4);
+        hipLaunchKernelGGL(k_dst_histogram, dim3(grid_for(R)), dim3(BLOCK), 0,
+                           e->stream, cur_tbl, R, e->ncols, col, ndst, d_hist);
+        TIME_END(e, CAT_SPLIT);
     }
     unsigned long long h_hist[64];
     HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, ndst * sizeof(unsigned long long),
                              hipMemcpyDeviceToHost, e->stream));
     HIP_CHECK(stream_sync(e->stream));
     unsigned long long base[64], acc = 0;
-    for (int i = 0; i < ndst; i++) { base[i] = acc; acc += h_hist[i]; rows_per_dst[i] = (int64_t)h_hist[i]; }
+    for (int i = 0; i < ndst; i++) {
+        base[i] = acc;
+        acc += h_hist[i];
+        rows_per_dst[i] = (int64_t)h_hist[i];
+    }
     HIP_CHECK(hipMemcpyAsync(d_hist, base, ndst * sizeof(unsigned long long),
                              hipMemcpyHostToDevice, e->stream));
     if (R) {
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_dst_scatter, dim3(grid_for(R)), dim3(BLOCK), 0, e->stream,
-                           cur_tbl, R, e->ncols, col, ndst, d_hist, d_cursor, dev_out);
-        TIME_END(e, CAT_SPLIT, (double)R * e->ncols * 8);
+        hipLaunchKernelGGL(k_dst_scatter, dim3(grid_for(R)), dim3(BLOCK), 0,
+                           e->stream, cur_tbl, R, e->ncols, col, ndst, d_hist,
+                           d_cursor, dev_out);
+        TIME_END(e, CAT_SPLIT);
     }
     HIP_CHECK(stream_sync(e->stream));
     return WK_OK;
@@ -826,7 +970,7 @@ extern "C" int32_t wk_engine_generate_sub_query(wk_engine_t *e, int32_t ndst,
 // download the current device table via a persistent PINNED staging
 // buffer: a pageable-destination hipMemcpyAsync makes the NEXT queries'
 // kernels stall ~25 ms (driver pin/unpin behind a large result copy,
-// measured in tools/qloop.py)
+// measured with tools/qloop.py)
 static int32_t download_table(wk_engine *e, sid_t *dst, size_t n) {
     if (!n) return WK_OK;
     size_t bytes = n * 4;
@@ -907,14 +1051,14 @@ static int32_t finalize_result(wk_engine *e, const wk_plan_t *plan,
 extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
                                           wk_result_t *out) {
     if (!e || !plan || !out) return WK_ERR_STATE;
+    int32_t rc = sync_state(e);
+    if (rc) return rc;  // WK_ERR_CAP -> caller re-runs (run_query does)
     double t0 = now_us();
     std::vector<sid_t> tbl((size_t)e->nrows * e->ncols);
-    int32_t drc = download_table(e, tbl.data(), tbl.size());
-    if (drc) return drc;
-    HIP_CHECK(stream_sync(e->stream));
-    resolve_timing(e);
+    rc = download_table(e, tbl.data(), tbl.size());
+    if (rc) return rc;
     double t1 = now_us();
-    int32_t rc = finalize_result(e, plan, tbl, out);
+    rc = finalize_result(e, plan, tbl, out);
     if (wk_verbose_lvl() >= 2)
         fprintf(stderr, "[fetch] rows=%lld d2h_us=%.0f final_us=%.0f\n",
                 (long long)e->nrows, t1 - t0, now_us() - t1);
@@ -924,11 +1068,12 @@ extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
 // raw current table (no final ops) — for the gloo-path exchange in tests
 extern "C" int32_t wk_engine_fetch_raw(wk_engine_t *e, wk_result_t *out) {
     if (!e || !out) return WK_ERR_STATE;
+    int32_t rc = sync_state(e);
+    if (rc) return rc;
     size_t n = (size_t)e->nrows * e->ncols;
     wk_sid_t *res = (wk_sid_t *)malloc(n ? n * 4 : 4);
-    int32_t drc = download_table(e, res, n);
-    if (drc) { free(res); return drc; }
-    HIP_CHECK(stream_sync(e->stream));
+    rc = download_table(e, res, n);
+    if (rc) { free(res); return rc; }
     out->col_num = e->ncols;
     out->row_num = e->nrows;
     out->table = res;
@@ -938,19 +1083,23 @@ extern "C" int32_t wk_engine_fetch_raw(wk_engine_t *e, wk_result_t *out) {
 
 extern "C" int32_t wk_engine_run_query(wk_engine_t *e, const wk_plan_t *plan,
                                        wk_result_t *out) {
-    int32_t rc = wk_engine_begin_query(e, plan);
-    if (rc) return rc;
-    while (e->step < (int)e->pats.size()) {
-        rc = wk_engine_execute_one_pattern(e, nullptr);
+    for (int attempt = 0; attempt < 6; attempt++) {
+        int32_t rc = wk_engine_begin_query(e, plan);
         if (rc) return rc;
-        if (e->nrows == 0 && e->step < (int)e->pats.size()) {
-            // keep executing: remaining ops on an empty table are no-ops but
-            // must still register new columns (mirrors the reference loop,
-            // sparql.hpp:1113-1154, which also continues on empty tables)
-            continue;
+        while (e->step < (int)e->pats.size()) {
+            rc = exec_pattern(e);
+            if (rc) return rc;
         }
+        rc = wk_engine_fetch_result(e, plan, out);
+        if (rc != WK_ERR_CAP) return rc;
+        // overflow: grow to the required size and re-run the whole query
+        int64_t need = (int64_t)e->h_pin[S_REQ];
+        rc = grow_caps(e, need + need / 4, e->cap_cols);
+        if (rc) return rc;
+        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                           e->d_state, S_WORDS);
     }
-    return wk_engine_fetch_result(e, plan, out);
+    return WK_ERR_CAP;
 }
 
 extern "C" void wk_result_free(wk_result_t *r) {
@@ -960,7 +1109,7 @@ extern "C" void wk_result_free(wk_result_t *r) {
 extern "C" int32_t wk_engine_kernel_stats(wk_engine_t *e, double *usec7,
                                           double *bytes7, int64_t *launches7) {
     if (!e) return WK_ERR_STATE;
-    for (int i = 0; i < 7; i++) {
+    for (int i = 0; i < CAT_COUNT; i++) {
         if (usec7) usec7[i] = e->cat_usec[i];
         if (bytes7) bytes7[i] = e->cat_bytes[i];
         if (launches7) launches7[i] = e->cat_n[i];
