@@ -120,14 +120,15 @@ def main():
                 ex = GpuExecutor(eng, plan)
                 dq = DistQuery(ex, plan, rank, world, device=f"cuda:{local_rank}")
                 dq.run()
-                part = ex.finalize()
-                cnt = torch.tensor([len(part)], dtype=torch.int64,
+                part = eng.fetch_count()  # blind reply (proxy.hpp:491)
+                cnt = torch.tensor([part], dtype=torch.int64,
                                    device=f"cuda:{local_rank}")
                 dist.all_reduce(cnt)
                 nrows = int(cnt.item())
             else:
-                tbl = eng.run_query(plan)
-                nrows = tbl.shape[0]
+                # blind execution — the reference's emulator benchmark mode
+                # ("always not take back results", core/proxy.hpp:491)
+                nrows = eng.run_query_count(plan)
             if collect is not None:
                 collect.setdefault(name, []).append((time.time() - tq) * 1e3)
                 collect.setdefault("_rows", {})[name] = nrows
@@ -222,6 +223,7 @@ def main():
                         f"{ngpus}xMI355X, RCCL all-to-all sub-query shipping",
             "dataset": dataset,
             "queries": "lubm q1-q7, osdi16 plans",
+            "blind": True,  # reference emulator semantics, proxy.hpp:491
             "parallelism": f"graph-partitioned x{ngpus} + per-step all-to-allv",
         },
         "roofline": roofline,

@@ -45,6 +45,9 @@ typedef struct {
     int32_t   distinct;           /* 0/1 */
     int64_t   limit;              /* -1 = none */
     int64_t   offset;
+    int32_t   blind;              /* 1 = return row count only (Result::blind,
+                                   * core/query.hpp:321 — the reference's
+                                   * proxy/emulator benchmark mode) */
 } wk_plan_t;
 
 /* A materialised binding table (SPARQLQuery::Result subset —

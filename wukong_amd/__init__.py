@@ -33,7 +33,8 @@ class WkPlan(ctypes.Structure):
                 ("nrequired", ctypes.c_int32),
                 ("distinct", ctypes.c_int32),
                 ("limit", ctypes.c_int64),
-                ("offset", ctypes.c_int64)]
+                ("offset", ctypes.c_int64),
+                ("blind", ctypes.c_int32)]
 
 
 class WkResult(ctypes.Structure):
@@ -110,13 +111,14 @@ class Plan:
     """A planner-ordered pattern list (vars negative: -1..-nvars)."""
 
     def __init__(self, patterns, nvars, required_vars, distinct=False,
-                 limit=-1, offset=0):
+                 limit=-1, offset=0, blind=False):
         self.patterns = list(patterns)
         self.nvars = nvars
         self.required_vars = list(required_vars)
         self.distinct = distinct
         self.limit = limit
         self.offset = offset
+        self.blind = blind
 
     def to_c(self):
         pats = (WkPattern * len(self.patterns))()
@@ -127,7 +129,7 @@ class Plan:
                       len(self.patterns), self.nvars,
                       ctypes.cast(req, ctypes.POINTER(ctypes.c_int32)),
                       len(self.required_vars), 1 if self.distinct else 0,
-                      self.limit, self.offset)
+                      self.limit, self.offset, 1 if self.blind else 0)
         plan._keepalive = (pats, req)
         return plan
 
@@ -187,6 +189,20 @@ class Engine:
             _eng_destroy(self._h)
             self._h = None
 
+    def run_query_count(self, plan):
+        """Blind execution (Result::blind): row count only, no download."""
+        import copy
+        p2 = copy.copy(plan)
+        p2.blind = True
+        cplan = p2.to_c()
+        res = WkResult()
+        rc = _eng_run(self._h, ctypes.byref(cplan), ctypes.byref(res))
+        if rc != 0:
+            raise RuntimeError(f"wk_engine_run_query(blind) rc={rc}")
+        n = res.row_num
+        _res_free(ctypes.byref(res))
+        return n
+
     def run_query(self, plan):
         cplan = plan.to_c()
         res = WkResult()
@@ -194,7 +210,7 @@ class Engine:
         if rc != 0:
             raise RuntimeError(f"wk_engine_run_query rc={rc}")
         try:
-            if res.row_num and res.col_num:
+            if res.row_num and res.col_num and res.table:
                 tbl = np.ctypeslib.as_array(
                     res.table, shape=(res.row_num, res.col_num)).copy()
             else:
@@ -247,6 +263,20 @@ class Engine:
         if rc != 0:
             raise RuntimeError(f"generate_sub_query rc={rc}")
         return list(rows)
+
+    def fetch_count(self):
+        """Blind fetch of the current query (Result::blind): row count."""
+        cplan = self._cplan
+        old = cplan.blind
+        cplan.blind = 1
+        res = WkResult()
+        rc = _eng_fetch(self._h, ctypes.byref(cplan), ctypes.byref(res))
+        cplan.blind = old
+        if rc != 0:
+            raise RuntimeError(f"fetch_count rc={rc}")
+        n = res.row_num
+        _res_free(ctypes.byref(res))
+        return n
 
     def fetch_result(self, plan=None):
         cplan = self._cplan if plan is None else plan.to_c()

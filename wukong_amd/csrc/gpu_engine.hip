@@ -119,10 +119,11 @@ __global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
                             int col, uint32_t pid, int dir, int key_mode,
                             int probe_mode, int col2, sid_t cval,
                             uint64_t list_off, uint64_t list_sz,
-                            const uint64_t *__restrict__ d_state,
+                            uint64_t *__restrict__ d_state,
                             uint64_t *__restrict__ d_stats,
                             uint64_t *__restrict__ d_eoff,
-                            uint64_t *__restrict__ d_cnt)
+                            uint64_t *__restrict__ d_cnt,
+                            sid_t *__restrict__ out_tbl)
 {
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     count_bytes(d_stats, probe_mode == PM_LIST ? CAT_FILTER : CAT_PROBE,
@@ -133,7 +134,13 @@ __global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
          r += (int64_t)gridDim.x * blockDim.x) {
         if (probe_mode == PM_LIST) {
             sid_t v = tbl[r * ncols + col];
-            d_cnt[r] = bsearch_u32(edges + list_off, list_sz, v) ? 1 : 0;
+            if (bsearch_u32(edges + list_off, list_sz, v)) {
+                unsigned long long pos = atomicAdd(
+                    (unsigned long long *)&d_state[S_TOTAL], 1ull);
+                sid_t *dst = out_tbl + (int64_t)pos * ncols;
+                const sid_t *src = tbl + r * ncols;
+                for (int c = 0; c < ncols; c++) dst[c] = src[c];
+            }
             continue;
         }
         sid_t v = tbl[r * ncols + col];
@@ -161,18 +168,21 @@ __global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
             if (k7 == KEY_EMPTY) { esz = 0; break; }
             bucket = key_vid(k7);
         }
-        switch (probe_mode) {
-        case PM_SIZE:
+        if (probe_mode == PM_SIZE) {
             d_eoff[r] = eoff;
             d_cnt[r] = esz;
-            break;
-        case PM_CONST:
-            d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz, cval)) ? 1 : 0;
-            break;
-        case PM_COL:
-            d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz,
-                                           tbl[r * ncols + col2])) ? 1 : 0;
-            break;
+        } else {
+            // fused filter + atomic compaction (keep-row semantics of
+            // sparql.hpp:455-476; row order is engine-internal, parity
+            // is set-level): replaces probe->scan->compact
+            sid_t tgt = (probe_mode == PM_CONST) ? cval : tbl[r * ncols + col2];
+            if (esz && bsearch_u32(edges + eoff, esz, tgt)) {
+                unsigned long long pos = atomicAdd(
+                    (unsigned long long *)&d_state[S_TOTAL], 1ull);
+                sid_t *dst = out_tbl + (int64_t)pos * ncols;
+                const sid_t *src = tbl + r * ncols;
+                for (int c = 0; c < ncols; c++) dst[c] = src[c];
+            }
         }
     }
 }
@@ -186,10 +196,11 @@ __global__ void k_probe(const vertex_t *__restrict__ verts,
                         int col, uint32_t pid, int dir, int key_mode,
                         int probe_mode, int col2, sid_t cval,
                         uint64_t list_off, uint64_t list_sz,
-                        const uint64_t *__restrict__ d_state,
+                        uint64_t *__restrict__ d_state,
                         uint64_t *__restrict__ d_stats,
                         uint64_t *__restrict__ d_eoff,
-                        uint64_t *__restrict__ d_cnt)
+                        uint64_t *__restrict__ d_cnt,
+                        sid_t *__restrict__ out_tbl)
 {
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     count_bytes(d_stats, probe_mode == PM_LIST ? CAT_FILTER : CAT_PROBE,
@@ -205,7 +216,13 @@ __global__ void k_probe(const vertex_t *__restrict__ verts,
         if (probe_mode == PM_LIST) {
             if (sub == 0) {
                 sid_t v = tbl[r * ncols + col];
-                d_cnt[r] = bsearch_u32(edges + list_off, list_sz, v) ? 1 : 0;
+                if (bsearch_u32(edges + list_off, list_sz, v)) {
+                    unsigned long long pos = atomicAdd(
+                        (unsigned long long *)&d_state[S_TOTAL], 1ull);
+                    sid_t *dst = out_tbl + (int64_t)pos * ncols;
+                    const sid_t *src = tbl + r * ncols;
+                    for (int c = 0; c < ncols; c++) dst[c] = src[c];
+                }
             }
             continue;
         }
@@ -233,16 +250,18 @@ __global__ void k_probe(const vertex_t *__restrict__ verts,
             bucket = key_vid(chain);
         }
         if (sub == 0) {
-            switch (probe_mode) {
-            case PM_SIZE: d_eoff[r] = eoff; d_cnt[r] = esz; break;
-            case PM_CONST:
-                d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz, cval)) ? 1 : 0;
-                break;
-            case PM_COL: {
-                sid_t tgt = tbl[r * ncols + col2];
-                d_cnt[r] = (esz && bsearch_u32(edges + eoff, esz, tgt)) ? 1 : 0;
-                break;
-            }
+            if (probe_mode == PM_SIZE) {
+                d_eoff[r] = eoff;
+                d_cnt[r] = esz;
+            } else {
+                sid_t tgt = (probe_mode == PM_CONST) ? cval : tbl[r * ncols + col2];
+                if (esz && bsearch_u32(edges + eoff, esz, tgt)) {
+                    unsigned long long pos = atomicAdd(
+                        (unsigned long long *)&d_state[S_TOTAL], 1ull);
+                    sid_t *dst = out_tbl + (int64_t)pos * ncols;
+                    const sid_t *src = tbl + r * ncols;
+                    for (int c = 0; c < ncols; c++) dst[c] = src[c];
+                }
             }
         }
     }
@@ -337,6 +356,22 @@ __global__ void k_commit(uint64_t *__restrict__ d_state, uint64_t cap) {
 
 __global__ void k_set_state(uint64_t *__restrict__ d_state, uint64_t nrows) {
     d_state[S_NROWS] = nrows;
+}
+
+// device projection to required-var columns (sparql.hpp:1510-1536)
+struct cols8 { int32_t c[8]; };
+__global__ void k_project(const sid_t *__restrict__ tbl, int ncols,
+                          const uint64_t *__restrict__ d_state, cols8 cols,
+                          int rc, sid_t *__restrict__ out)
+{
+    const int64_t n = (int64_t)d_state[S_NROWS];
+    for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < n * rc;
+         t += (int64_t)gridDim.x * blockDim.x) {
+        int64_t i = t / rc;
+        int j = (int)(t - i * rc);
+        int c = cols.c[j];
+        out[t] = (c >= 0) ? tbl[i * ncols + c] : (sid_t)0xFFFFFFFFu;
+    }
 }
 
 __global__ void k_publish_state(const uint64_t *__restrict__ d_state,
@@ -768,29 +803,22 @@ static int32_t exec_pattern(wk_engine *e) {
         uint64_t sz = 0;
         const sid_t *ptr = store_get(*st, (uint64_t)s, (uint64_t)p, dir, &sz);
         uint64_t off = ptr ? (uint64_t)(ptr - st->edges.data()) : 0;
+        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                           e->d_state + S_TOTAL, 1);
         TIME_BEGIN(e);
         if (e->probe_coop)
             hipLaunchKernelGGL(k_probe, dim3(grid_for(e->bound * 8)), dim3(BLOCK), 0,
                                e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
                                e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
                                off, sz, e->d_state, e->d_stats,
-                               (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+                               (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p, out_tbl);
         else
             hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
                                e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
                                e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
                                off, sz, e->d_state, e->d_stats,
-                               (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+                               (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p, out_tbl);
         TIME_END(e, CAT_FILTER);
-        enqueue_scan(e);
-        {
-            TIME_BEGIN(e);
-            hipLaunchKernelGGL(k_compact, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
-                               e->stream, cur_tbl, e->ncols, (uint64_t *)e->cnt.p,
-                               (uint64_t *)e->prefix.p, e->d_state, e->d_stats,
-                               out_tbl);
-            TIME_END(e, CAT_FILTER);
-        }
         hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
                            (uint64_t)e->cap_rows);
         e->cur ^= 1;
@@ -830,6 +858,9 @@ static int32_t exec_pattern(wk_engine *e) {
     int pmode = (ostat == 0) ? PM_SIZE : (ostat == 2 ? PM_CONST : PM_COL);
     int col2 = (ostat == 1) ? e->var2col(o) : 0;
     sid_t cval = (ostat == 2) ? (sid_t)o : 0;
+    if (pmode != PM_SIZE)
+        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                           e->d_state + S_TOTAL, 1);
     {
         TIME_BEGIN(e);
         if (e->probe_coop)
@@ -837,18 +868,20 @@ static int32_t exec_pattern(wk_engine *e) {
                                e->stream, e->d_verts, e->d_edges, seg->bucket_start,
                                seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
                                dir, key_mode, pmode, col2, cval, 0, 0, e->d_state,
-                               e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
+                               e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p,
+                               out_tbl);
         else
             hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
                                e->stream, e->d_verts, e->d_edges, seg->bucket_start,
                                seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
                                dir, key_mode, pmode, col2, cval, 0, 0, e->d_state,
-                               e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p);
-        TIME_END(e, CAT_PROBE);
+                               e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p,
+                               out_tbl);
+        TIME_END(e, pmode == PM_SIZE ? CAT_PROBE : CAT_FILTER);
     }
-    enqueue_scan(e);
 
     if (pmode == PM_SIZE) {
+        enqueue_scan(e);
         // known_to_unknown: expand (bound multiplies by unknown fan-out;
         // grid heuristic x16, correctness by grid-stride)
         int oc = e->ncols + 1;
@@ -862,13 +895,6 @@ static int32_t exec_pattern(wk_engine *e) {
         e->v2c[-(o + 1)] = e->ncols;
         e->ncols = oc;
         e->bound = e->cap_rows;  // fan-out unknown until a sync point
-    } else {
-        // known_to_known / known_to_const: compact (rows only shrink)
-        TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_compact, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
-                           e->stream, cur_tbl, e->ncols, (uint64_t *)e->cnt.p,
-                           (uint64_t *)e->prefix.p, e->d_state, e->d_stats, out_tbl);
-        TIME_END(e, CAT_FILTER);
     }
     hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
                        (uint64_t)e->cap_rows);
@@ -1051,8 +1077,48 @@ static int32_t finalize_result(wk_engine *e, const wk_plan_t *plan,
 extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
                                           wk_result_t *out) {
     if (!e || !plan || !out) return WK_ERR_STATE;
+    const bool simple = !plan->distinct && plan->offset <= 0 && plan->limit < 0 &&
+                        plan->nrequired <= 8;
+    if (!plan->blind && simple && plan->nrequired > 0) {
+        // project on DEVICE (sparql.hpp:1510-1536 semantics) and download
+        // the projected table directly
+        cols8 cols{};
+        for (int j = 0; j < plan->nrequired; j++)
+            cols.c[j] = e->var2col(plan->required_vars[j]);
+        sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
+        TIME_BEGIN(e);
+        hipLaunchKernelGGL(k_project, dim3(grid_for(e->cap_rows)), dim3(BLOCK), 0,
+                           e->stream, (sid_t *)e->tbl[e->cur].p, e->ncols,
+                           e->d_state, cols, plan->nrequired, out_tbl);
+        TIME_END(e, CAT_OTHER);
+        int32_t rc = sync_state(e);
+        if (rc) return rc;
+        double t0 = now_us();
+        size_t n = (size_t)e->nrows * plan->nrequired;
+        wk_sid_t *res = (wk_sid_t *)malloc(n ? n * 4 : 4);
+        e->cur ^= 1;  // projected table is current for the download
+        rc = download_table(e, res, n);
+        e->cur ^= 1;
+        if (rc) { free(res); return rc; }
+        out->col_num = plan->nrequired;
+        out->row_num = e->nrows;
+        out->table = res;
+        out->status_code = 0;
+        if (wk_verbose_lvl() >= 2)
+            fprintf(stderr, "[fetch] rows=%lld proj+d2h_us=%.0f\n",
+                    (long long)e->nrows, now_us() - t0);
+        return WK_OK;
+    }
     int32_t rc = sync_state(e);
     if (rc) return rc;  // WK_ERR_CAP -> caller re-runs (run_query does)
+    if (plan->blind) {
+        // Result::blind (query.hpp:321): row count only, no table
+        out->col_num = plan->nrequired;
+        out->row_num = e->nrows;
+        out->table = nullptr;
+        out->status_code = 0;
+        return WK_OK;
+    }
     double t0 = now_us();
     std::vector<sid_t> tbl((size_t)e->nrows * e->ncols);
     rc = download_table(e, tbl.data(), tbl.size());
@@ -1105,6 +1171,7 @@ extern "C" int32_t wk_engine_run_query(wk_engine_t *e, const wk_plan_t *plan,
 extern "C" void wk_result_free(wk_result_t *r) {
     if (r && r->table) { free(r->table); r->table = nullptr; }
 }
+
 
 extern "C" int32_t wk_engine_kernel_stats(wk_engine_t *e, double *usec7,
                                           double *bytes7, int64_t *launches7) {
