@@ -187,6 +187,92 @@ __global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
     }
 }
 
+// Filter operators (k2c/k2k/c2k/i2k): probe + membership test + BLOCK-
+// AGGREGATED atomic compaction (one global atomicAdd per 256 rows — a
+// single per-row counter serializes at ~88 adds/us, MI355X_MICROARCH.md
+// row `dequeue`).  Row order is engine-internal; parity is set-level.
+__global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
+                             const sid_t *__restrict__ edges,
+                             uint64_t bucket_start, uint64_t num_buckets,
+                             const sid_t *__restrict__ tbl, int ncols,
+                             int col, uint32_t pid, int dir, int key_mode,
+                             int probe_mode, int col2, sid_t cval,
+                             uint64_t list_off, uint64_t list_sz,
+                             uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats,
+                             sid_t *__restrict__ out_tbl)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_FILTER,
+                (uint64_t)nrows * (probe_mode == PM_LIST ? 12 : (4 + 128 + 8 + 64)));
+    __shared__ unsigned long long s_base;
+    __shared__ uint32_t wtot[4];
+    const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < nrows;
+         base += stride) {
+        const int64_t r = base + threadIdx.x;
+        bool keep = false;
+        if (r < nrows) {
+            sid_t v = tbl[r * ncols + col];
+            if (probe_mode == PM_LIST) {
+                keep = bsearch_u32(edges + list_off, list_sz, v);
+            } else {
+                uint64_t key = (key_mode == PK_NORMAL)
+                                   ? key_pack(v, pid, (uint64_t)dir)
+                                   : key_pack(0, v, (uint64_t)dir);
+                uint64_t bucket = bucket_start + hash_u64(key) % num_buckets;
+                uint64_t eoff = 0, esz = 0;
+                while (true) {
+                    const vertex_t *b = &verts[bucket * ASSOC];
+                    uint64_t k0 = b[0].key, k1 = b[1].key, k2 = b[2].key,
+                             k3 = b[3].key, k4 = b[4].key, k5 = b[5].key,
+                             k6 = b[6].key, k7 = b[7].key;
+                    int hit = -1;
+                    if (k0 == key) hit = 0;
+                    else if (k1 == key) hit = 1;
+                    else if (k2 == key) hit = 2;
+                    else if (k3 == key) hit = 3;
+                    else if (k4 == key) hit = 4;
+                    else if (k5 == key) hit = 5;
+                    else if (k6 == key) hit = 6;
+                    if (hit >= 0) {
+                        uint64_t pp = b[hit].ptr;
+                        eoff = ptr_off(pp); esz = ptr_size(pp);
+                        break;
+                    }
+                    if (k7 == KEY_EMPTY) { esz = 0; break; }
+                    bucket = key_vid(k7);
+                }
+                sid_t tgt = (probe_mode == PM_CONST) ? cval : tbl[r * ncols + col2];
+                keep = esz && bsearch_u32(edges + eoff, esz, tgt);
+            }
+        }
+        // block-wide compaction: wave ballot -> per-wave totals -> one
+        // atomic for the block -> in-block exclusive offsets
+        uint64_t wmask = __ballot(keep);
+        if (lane == 0) wtot[wid] = (uint32_t)__popcll(wmask);
+        __syncthreads();
+        if (threadIdx.x == 0) {
+            uint32_t t = wtot[0] + wtot[1] + wtot[2] + wtot[3];
+            s_base = t ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
+                                   (unsigned long long)t)
+                       : 0;
+        }
+        __syncthreads();
+        if (keep) {
+            uint32_t woff = 0;
+            for (int k = 0; k < wid; k++) woff += wtot[k];
+            uint64_t pos = (uint64_t)s_base + woff +
+                           __popcll(wmask & ((1ull << lane) - 1));
+            sid_t *dst = out_tbl + (int64_t)pos * ncols;
+            const sid_t *src = tbl + r * ncols;
+            for (int c = 0; c < ncols; c++) dst[c] = src[c];
+        }
+        __syncthreads();
+    }
+}
+
 // Cooperative 8-lane-per-row probe (WK_PROBE=coop, kept for A/B): lanes
 // load one 128-B bucket coalesced and ballot-match.
 __global__ void k_probe(const vertex_t *__restrict__ verts,
@@ -806,18 +892,10 @@ static int32_t exec_pattern(wk_engine *e) {
         hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
                            e->d_state + S_TOTAL, 1);
         TIME_BEGIN(e);
-        if (e->probe_coop)
-            hipLaunchKernelGGL(k_probe, dim3(grid_for(e->bound * 8)), dim3(BLOCK), 0,
-                               e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
-                               e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
-                               off, sz, e->d_state, e->d_stats,
-                               (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p, out_tbl);
-        else
-            hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
-                               e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
-                               e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
-                               off, sz, e->d_state, e->d_stats,
-                               (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p, out_tbl);
+        hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
+                           e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
+                           e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
+                           off, sz, e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
         hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
                            (uint64_t)e->cap_rows);
@@ -858,10 +936,17 @@ static int32_t exec_pattern(wk_engine *e) {
     int pmode = (ostat == 0) ? PM_SIZE : (ostat == 2 ? PM_CONST : PM_COL);
     int col2 = (ostat == 1) ? e->var2col(o) : 0;
     sid_t cval = (ostat == 2) ? (sid_t)o : 0;
-    if (pmode != PM_SIZE)
+    if (pmode != PM_SIZE) {
         hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
                            e->d_state + S_TOTAL, 1);
-    {
+        TIME_BEGIN(e);
+        hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
+                           e->stream, e->d_verts, e->d_edges, seg->bucket_start,
+                           seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
+                           dir, key_mode, pmode, col2, cval, 0, 0, e->d_state,
+                           e->d_stats, out_tbl);
+        TIME_END(e, CAT_FILTER);
+    } else {
         TIME_BEGIN(e);
         if (e->probe_coop)
             hipLaunchKernelGGL(k_probe, dim3(grid_for(e->bound * 8)), dim3(BLOCK), 0,
@@ -877,7 +962,7 @@ static int32_t exec_pattern(wk_engine *e) {
                                dir, key_mode, pmode, col2, cval, 0, 0, e->d_state,
                                e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p,
                                out_tbl);
-        TIME_END(e, pmode == PM_SIZE ? CAT_PROBE : CAT_FILTER);
+        TIME_END(e, CAT_PROBE);
     }
 
     if (pmode == PM_SIZE) {
