@@ -148,6 +148,11 @@ void wk_result_free(wk_result_t *);
 int32_t wk_engine_kernel_stats(wk_engine_t *, double *usec7, double *bytes7,
                                int64_t *launches7);
 
+/* Raw device allocation (tests / exchange buffers). */
+void   *wk_dev_alloc(uint64_t bytes);
+void    wk_dev_free(void *);
+int32_t wk_dev_download(const void *dev, void *host, uint64_t bytes);
+
 /* Library/build info: returns gfx arch string the .so was built for. */
 const char *wk_build_arch(void);
 int32_t     wk_device_count(void);

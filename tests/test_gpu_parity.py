@@ -53,7 +53,6 @@ def test_distinct_limit_gpu(eng4, oracle4):
 
 def test_step_api_and_split(store4, oracle4):
     """Step-level API + generate_sub_query chunks (fork-join split)."""
-    import torch
     eng = wk.Engine(store4, device=0)
     plan = Q.Q1
     eng.begin_query(plan)
@@ -62,11 +61,12 @@ def test_step_api_and_split(store4, oracle4):
     assert n1 > 0
     # split the current table by next pattern's start var (?X, col 0)
     ncols = eng.col_num
-    buf = torch.empty(n1 * ncols, dtype=torch.int32, device="cuda:0")
-    rows = eng.generate_sub_query(2, buf.data_ptr(), n1)
+    buf = wk.dev_alloc(n1 * ncols * 4)
+    rows = eng.generate_sub_query(2, buf, n1)
     assert sum(rows) == n1
     tbl = eng.fetch_raw()
-    packed = buf[: n1 * ncols].reshape(n1, ncols).cpu().numpy().view(np.uint32)
+    packed = wk.dev_download_u32(buf, n1 * ncols).reshape(n1, ncols)
+    wk.dev_free(buf)
     # chunk d rows all hash to d; union of chunks == table (multiset)
     xcol = 0
     assert np.all(packed[: rows[0], xcol] % 2 == 0)

@@ -80,6 +80,30 @@ _res_free = _sig("wk_result_free", None, [ctypes.POINTER(WkResult)])
 _kstats = _sig("wk_engine_kernel_stats", c_i32, [c_vp, ctypes.POINTER(ctypes.c_double), ctypes.POINTER(ctypes.c_double), ctypes.POINTER(c_i64)])
 _arch = _sig("wk_build_arch", ctypes.c_char_p, [])
 _devcount = _sig("wk_device_count", c_i32, [])
+_dev_alloc = _sig("wk_dev_alloc", c_vp, [c_u64])
+_dev_free = _sig("wk_dev_free", None, [c_vp])
+_dev_download = _sig("wk_dev_download", c_i32, [c_vp, c_vp, c_u64])
+
+
+def dev_alloc(nbytes):
+    p = _dev_alloc(nbytes)
+    if not p:
+        raise RuntimeError("wk_dev_alloc failed")
+    return p
+
+
+def dev_free(p):
+    _dev_free(p)
+
+
+def dev_download_u32(ptr, count):
+    out = np.empty(count, dtype=np.uint32)
+    rc = _dev_download(ptr, out.ctypes.data_as(c_vp), count * 4)
+    if rc != 0:
+        raise RuntimeError("wk_dev_download failed")
+    return out
+
+
 hash_u64 = _sig("wk_hash_u64", c_u64, [c_u64])
 key_pack = _sig("wk_key_pack", c_u64, [c_u64, c_u64, c_u64])
 ptr_pack = _sig("wk_ptr_pack", c_u64, [c_u64, c_u64, c_u64])

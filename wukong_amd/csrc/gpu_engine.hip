@@ -1278,6 +1278,18 @@ extern "C" uint64_t wk_ptr_pack(uint64_t size, uint64_t off, uint64_t type) {
     return ptr_pack(size, off, type);
 }
 
+// raw device allocation helpers (tests / exchange buffers without torch)
+extern "C" void *wk_dev_alloc(uint64_t bytes) {
+    void *p = nullptr;
+    if (hipMalloc(&p, bytes ? bytes : 4) != hipSuccess) return nullptr;
+    return p;
+}
+extern "C" void wk_dev_free(void *p) { if (p) (void)hipFree(p); }
+extern "C" int32_t wk_dev_download(const void *dev, void *host, uint64_t bytes) {
+    HIP_CHECK(hipMemcpy(host, dev, bytes, hipMemcpyDeviceToHost));
+    return 0;
+}
+
 extern "C" const char *wk_build_arch(void) {
 #if defined(__HIP_PLATFORM_AMD__)
     return "gfx950";
