@@ -339,6 +339,22 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     }
     WK_LOG("[store] insert index: %.1fs (ext used %lu)\n", now_s() - t0,
            (unsigned long)(ext_next.load() - st->nbuckets_main));
+    // dense type side-index from the TYPE_ID pso slice (runs per subject)
+    if (max_id >= (1u << NBITS_IDX) && max_pid < 0xFFFF) {
+        st->type_base = 1u << NBITS_IDX;
+        st->type_n = (uint64_t)max_id + 1 - st->type_base;
+        st->type_of.alloc(st->type_n, /*zero=*/true);
+#pragma omp parallel for schedule(static)
+        for (int64_t i = pso_lo[TYPE_ID]; i < pso_lo[TYPE_ID + 1]; i++) {
+            bool first = (i == pso_lo[TYPE_ID]) || (pso[i - 1].s != pso[i].s);
+            bool last = (i + 1 == pso_lo[TYPE_ID + 1]) || (pso[i + 1].s != pso[i].s);
+            if (first && last)
+                st->type_of[pso[i].s - st->type_base] = (uint16_t)pso[i].o;
+            else
+                st->type_of[pso[i].s - st->type_base] = 0xFFFF;
+        }
+    }
+    WK_LOG("[store] type index: %.1fs\n", now_s() - t0);
     st->ext_used = ext_next.load() - st->nbuckets_main;
     return st;
 }

@@ -53,6 +53,13 @@ struct wk_store {
     pod_array<wk::sid_t> edges;
     // normal segments indexed [pid*2 + dir], size (max_pid+1)*2
     std::vector<wk::seg_t> nseg;
+
+    // dense single-type side index: type_of[vid - type_base] = the type id
+    // of vid (0 = untyped, 0xFFFF = multiple types -> probe fallback).
+    // Accelerates `?X rdf:type T` filters to one 2-byte read; derived
+    // from the [vid|TYPE_ID|OUT] lists, bit-identical results.
+    pod_array<uint16_t> type_of;
+    uint64_t type_base = 0, type_n = 0;
     // index segments [dir]: keys [0|pid|dir] / [0|tid|IN]
     wk::seg_t iseg[2];
 
