@@ -106,32 +106,70 @@ def main():
     log(f"[rank {rank}] store: {store.num_slots} slots, {store.num_edges} edges "
         f"({time.time()-t0:.1f}s)")
     t0 = time.time()
-    eng = wk.Engine(store, device=local_rank)
-    log(f"[rank {rank}] HBM upload {((store.num_slots*16+store.num_edges*4)/1e9):.2f} GB "
-        f"({time.time()-t0:.1f}s)")
+    inflight = 1 if distributed else int(os.environ.get("WK_INFLIGHT", "8"))
+    gstore = wk.GpuStore(store, device=local_rank)
+    engines = [wk.Engine(gstore, device=local_rank) for _ in range(inflight)]
+    eng = engines[0]
+    log(f"[rank {rank}] HBM upload {((store.num_slots*16+store.num_edges*4)/1e9):.2f} GB, "
+        f"{inflight} engine(s) ({time.time()-t0:.1f}s)")
 
     names = list(Q.ALL)
 
-    def run_suite(collect=None):
-        for name in names:
-            plan = Q.ALL[name]
-            tq = time.time()
-            if distributed:
-                ex = GpuExecutor(eng, plan)
-                dq = DistQuery(ex, plan, rank, world, device=f"cuda:{local_rank}")
-                dq.run()
-                part = eng.fetch_count()  # blind reply (proxy.hpp:491)
-                cnt = torch.tensor([part], dtype=torch.int64,
+    def harvest(eng_, plan_):
+        n = eng_.fetch_count()
+        while n < 0:  # capacity grew; resubmit (overflow re-run)
+            eng_.submit(plan_)
+            n = eng_.fetch_count()
+        return n
+
+    def run_suite(collect=None, passes=1):
+        """Distributed path / single-engine fallback: sequential queries."""
+        for _ in range(passes):
+            for name in names:
+                plan = Q.ALL[name]
+                tq = time.time()
+                if distributed:
+                    ex = GpuExecutor(eng, plan)
+                    dq = DistQuery(ex, plan, rank, world,
                                    device=f"cuda:{local_rank}")
-                dist.all_reduce(cnt)
-                nrows = int(cnt.item())
-            else:
-                # blind execution — the reference's emulator benchmark mode
-                # ("always not take back results", core/proxy.hpp:491)
-                nrows = eng.run_query_count(plan)
+                    dq.run()
+                    part = eng.fetch_count()  # blind reply (proxy.hpp:491)
+                    cnt = torch.tensor([part], dtype=torch.int64,
+                                       device=f"cuda:{local_rank}")
+                    dist.all_reduce(cnt)
+                    nrows = int(cnt.item())
+                else:
+                    nrows = eng.run_query_count(plan)
+                if collect is not None:
+                    collect.setdefault(name, []).append((time.time() - tq) * 1e3)
+                    collect.setdefault("_rows", {})[name] = nrows
+
+    def run_pipelined(passes, collect=None):
+        """In-flight window over engine pool — the reference proxy's
+        emulator mode (`-p`, core/proxy.hpp:477-525), blind replies."""
+        from collections import deque
+        pending = deque()
+        free = list(engines)
+        seq = [name for _ in range(passes) for name in names]
+
+        def drain_one():
+            name_, eng_, t_iss, plan_ = pending.popleft()
+            n = harvest(eng_, plan_)
             if collect is not None:
-                collect.setdefault(name, []).append((time.time() - tq) * 1e3)
-                collect.setdefault("_rows", {})[name] = nrows
+                collect.setdefault(name_, []).append((time.time() - t_iss) * 1e3)
+                collect.setdefault("_rows", {})[name_] = n
+            free.append(eng_)
+
+        for name in seq:
+            if not free:
+                drain_one()
+            eng_ = free.pop()
+            plan = Q.ALL[name]
+            t_iss = time.time()
+            eng_.submit(plan)
+            pending.append((name, eng_, t_iss, plan))
+        while pending:
+            drain_one()
 
     def sync():
         torch.cuda.synchronize(local_rank)
@@ -139,19 +177,33 @@ def main():
             dist.barrier()
             torch.cuda.synchronize(local_rank)
 
+    def all_stats():
+        tot = {}
+        for en in engines:
+            for k, v in en.kernel_stats().items():
+                t = tot.setdefault(k, dict(usec=0.0, bytes=0.0, launches=0))
+                t["usec"] += v["usec"]
+                t["bytes"] += v["bytes"]
+                t["launches"] += v["launches"]
+        return tot
+
     # warmup
-    for _ in range(args.warmup):
-        run_suite()
+    if distributed or inflight == 1:
+        run_suite(passes=args.warmup)
+    else:
+        run_pipelined(args.warmup)
     sync()
-    stats0 = eng.kernel_stats()
+    stats0 = all_stats()
 
     lat = {}
     t_start = time.time()
-    for _ in range(args.steps):
-        run_suite(collect=lat)
+    if distributed or inflight == 1:
+        run_suite(collect=lat, passes=args.steps)
+    else:
+        run_pipelined(args.steps, collect=lat)
     sync()
     elapsed = time.time() - t_start
-    stats1 = eng.kernel_stats()
+    stats1 = all_stats()
 
     if distributed:
         t = torch.tensor([elapsed], dtype=torch.float64,
@@ -225,6 +277,7 @@ def main():
             "queries": "lubm q1-q7, osdi16 plans",
             "blind": True,  # reference emulator semantics, proxy.hpp:491
             "parallelism": f"graph-partitioned x{ngpus} + per-step all-to-allv",
+            "inflight": inflight,  # reference emulator window (proxy.hpp -p)
         },
         "roofline": roofline,
         "cpu_baseline": cb,

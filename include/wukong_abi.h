@@ -104,9 +104,22 @@ uint64_t wk_store_checksum(const wk_store_t *); /* FNV over vertices+edges */
  * GPUMem+GPUCache init (core/wukong.cpp:235-241) with the cache
  * replaced by a fully HBM-resident store. */
 typedef struct wk_engine wk_engine_t;
+typedef struct wk_gpu_store wk_gpu_store_t;
+
+/* Device-resident store image shared by any number of engines on one
+ * GPU (one HBM upload; engines carry only scratch) — replaces the
+ * reference's per-agent GPUCache (core/gpu/gpu_cache.hpp). */
+wk_gpu_store_t *wk_gpu_store_create(const wk_store_t *, int32_t device);
+void            wk_gpu_store_destroy(wk_gpu_store_t *);
+wk_engine_t    *wk_engine_create_on(wk_gpu_store_t *);
 
 wk_engine_t *wk_engine_create(const wk_store_t *, int32_t device);
 void         wk_engine_destroy(wk_engine_t *);
+
+/* Enqueue a whole plan as one asynchronous launch chain (no sync) —
+ * pipelined multi-engine execution = the reference proxy's in-flight
+ * window (core/proxy.hpp:477-525).  Harvest with wk_engine_fetch_*. */
+int32_t wk_engine_submit(wk_engine_t *, const wk_plan_t *);
 
 /* Whole-query execution on one GPU (Engine::execute_sparql_query +
  * SPARQLEngine::execute_patterns, core/engine/sparql.hpp:1113-1154,

@@ -65,6 +65,10 @@ _num_slots = _sig("wk_store_num_slots", c_u64, [c_vp])
 _num_edges = _sig("wk_store_num_edges", c_u64, [c_vp])
 _checksum = _sig("wk_store_checksum", c_u64, [c_vp])
 _eng_create = _sig("wk_engine_create", c_vp, [c_vp, c_i32])
+_gstore_create = _sig("wk_gpu_store_create", c_vp, [c_vp, c_i32])
+_gstore_destroy = _sig("wk_gpu_store_destroy", None, [c_vp])
+_eng_create_on = _sig("wk_engine_create_on", c_vp, [c_vp])
+_eng_submit = _sig("wk_engine_submit", c_i32, [c_vp, ctypes.POINTER(WkPlan)])
 _eng_destroy = _sig("wk_engine_destroy", None, [c_vp])
 _eng_run = _sig("wk_engine_run_query", c_i32, [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(WkResult)])
 _eng_begin = _sig("wk_engine_begin_query", c_i32, [c_vp, ctypes.POINTER(WkPlan)])
@@ -198,15 +202,42 @@ class Store:
         return _checksum(self._h)
 
 
+class GpuStore:
+    """Device-resident store image shared by multiple engines (one HBM
+    upload)."""
+
+    def __init__(self, store, device=0):
+        self._store = store
+        self._h = _gstore_create(store._h, device)
+        if not self._h:
+            raise RuntimeError("wk_gpu_store_create failed (no GPU / HBM alloc)")
+
+    def __del__(self):
+        if getattr(self, "_h", None):
+            _gstore_destroy(self._h)
+            self._h = None
+
+
 class Engine:
     """GPU engine bound to one device; requires a GPU (fails loudly)."""
 
     def __init__(self, store, device=0):
         self._store = store  # keepalive
-        self._h = _eng_create(store._h, device)
+        if isinstance(store, GpuStore):
+            self._h = _eng_create_on(store._h)
+        else:
+            self._h = _eng_create(store._h, device)
         if not self._h:
             raise RuntimeError(
                 "wk_engine_create failed (no MI355X visible, or HBM alloc failed)")
+
+    def submit(self, plan):
+        """Enqueue a whole plan asynchronously (harvest with fetch_*)."""
+        cplan = plan.to_c()
+        rc = _eng_submit(self._h, ctypes.byref(cplan))
+        if rc != 0:
+            raise RuntimeError(f"submit rc={rc}")
+        self._cplan = cplan
 
     def __del__(self):
         if getattr(self, "_h", None):
@@ -296,6 +327,8 @@ class Engine:
         res = WkResult()
         rc = _eng_fetch(self._h, ctypes.byref(cplan), ctypes.byref(res))
         cplan.blind = old
+        if rc == -5:  # capacity overflow: engine grew, caller resubmits
+            return -1
         if rc != 0:
             raise RuntimeError(f"fetch_count rc={rc}")
         n = res.row_num

@@ -606,12 +606,26 @@ struct timed_launch {
 
 }  // namespace
 
+// device-resident store image, shared by any number of engines on one
+// GPU (one upload; engines add only their scratch — replaces the
+// reference's per-agent GPUCache, core/gpu/gpu_cache.hpp)
+struct wk_gpu_store {
+    const wk_store *st = nullptr;
+    int device = 0;
+    vertex_t *d_verts = nullptr;
+    sid_t *d_edges = nullptr;
+    uint16_t *d_type_of = nullptr;
+    int refs = 0;     // engines attached
+    bool owned = false;  // created implicitly by wk_engine_create
+};
+
 struct wk_engine {
     const wk_store *st = nullptr;
+    wk_gpu_store *gs = nullptr;
     int device = 0;
     hipStream_t stream = nullptr;
 
-    // device store
+    // device store (borrowed from gs)
     vertex_t *d_verts = nullptr;
     sid_t *d_edges = nullptr;
     uint16_t *d_type_of = nullptr;
@@ -723,40 +737,80 @@ static int32_t grow_caps(wk_engine *e, int64_t rows, int cols) {
     return WK_OK;
 }
 
-extern "C" wk_engine_t *wk_engine_create(const wk_store_t *st, int32_t device) {
+extern "C" void wk_gpu_store_destroy(wk_gpu_store_t *g);
+
+extern "C" wk_gpu_store_t *wk_gpu_store_create(const wk_store_t *st, int32_t device) {
     if (!st) return nullptr;
     if (hipSetDevice(device) != hipSuccess) return nullptr;
-    wk_engine *e = new wk_engine();
-    e->st = st;
-    e->device = device;
-    if (hipStreamCreate(&e->stream) != hipSuccess) { delete e; return nullptr; }
+    wk_gpu_store *g = new wk_gpu_store();
+    g->st = st;
+    g->device = device;
     size_t vb = st->vertices.size() * sizeof(vertex_t);
     size_t eb = st->edges.size() * sizeof(sid_t);
-    if (hipMalloc(&e->d_verts, vb ? vb : 16) != hipSuccess ||
-        hipMalloc(&e->d_edges, eb ? eb : 16) != hipSuccess ||
-        hipMalloc(&e->d_state, (S_WORDS + CAT_COUNT + 2) * 8) != hipSuccess ||
+    if (hipMalloc(&g->d_verts, vb ? vb : 16) != hipSuccess ||
+        hipMalloc(&g->d_edges, eb ? eb : 16) != hipSuccess) {
+        wk_gpu_store_destroy(g);
+        return nullptr;
+    }
+    if (hipMemcpy(g->d_verts, st->vertices.data(), vb, hipMemcpyHostToDevice) != hipSuccess ||
+        hipMemcpy(g->d_edges, st->edges.data(), eb, hipMemcpyHostToDevice) != hipSuccess) {
+        wk_gpu_store_destroy(g);
+        return nullptr;
+    }
+    if (st->type_n) {
+        if (hipMalloc(&g->d_type_of, st->type_n * 2) != hipSuccess ||
+            hipMemcpy(g->d_type_of, st->type_of.data(), st->type_n * 2,
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            wk_gpu_store_destroy(g);
+            return nullptr;
+        }
+    }
+    return g;
+}
+
+extern "C" void wk_gpu_store_destroy(wk_gpu_store_t *g) {
+    if (!g) return;
+    if (g->refs > 0) { g->owned = false; return; }  // last engine frees
+    if (g->d_verts) (void)hipFree(g->d_verts);
+    if (g->d_edges) (void)hipFree(g->d_edges);
+    if (g->d_type_of) (void)hipFree(g->d_type_of);
+    delete g;
+}
+
+extern "C" wk_engine_t *wk_engine_create_on(wk_gpu_store_t *g) {
+    if (!g) return nullptr;
+    if (hipSetDevice(g->device) != hipSuccess) return nullptr;
+    wk_engine *e = new wk_engine();
+    e->st = g->st;
+    e->gs = g;
+    g->refs++;
+    e->device = g->device;
+    e->d_verts = g->d_verts;
+    e->d_edges = g->d_edges;
+    e->d_type_of = g->d_type_of;
+    if (hipStreamCreate(&e->stream) != hipSuccess) { delete e; return nullptr; }
+    if (hipMalloc(&e->d_state, (S_WORDS + CAT_COUNT + 2) * 8) != hipSuccess ||
         hipHostMalloc(&e->h_pin, 64 * sizeof(uint64_t)) != hipSuccess) {
         wk_engine_destroy(e);
         return nullptr;
     }
     e->d_stats = e->d_state + S_WORDS;
-    if (hipMemcpy(e->d_verts, st->vertices.data(), vb, hipMemcpyHostToDevice) != hipSuccess ||
-        hipMemcpy(e->d_edges, st->edges.data(), eb, hipMemcpyHostToDevice) != hipSuccess ||
-        hipMemset(e->d_state, 0, (S_WORDS + CAT_COUNT + 2) * 8) != hipSuccess) {
+    if (hipMemset(e->d_state, 0, (S_WORDS + CAT_COUNT + 2) * 8) != hipSuccess) {
         wk_engine_destroy(e);
         return nullptr;
-    }
-    if (st->type_n) {
-        if (hipMalloc(&e->d_type_of, st->type_n * 2) != hipSuccess ||
-            hipMemcpy(e->d_type_of, st->type_of.data(), st->type_n * 2,
-                      hipMemcpyHostToDevice) != hipSuccess) {
-            wk_engine_destroy(e);
-            return nullptr;
-        }
     }
     if (const char *t = getenv("WK_KERNEL_TIMING")) e->timing = atoi(t) != 0;
     if (const char *pv = getenv("WK_PROBE")) e->probe_coop = !strcmp(pv, "coop");
     if (grow_caps(e, 1 << 20, 4) != WK_OK) { wk_engine_destroy(e); return nullptr; }
+    return e;
+}
+
+extern "C" wk_engine_t *wk_engine_create(const wk_store_t *st, int32_t device) {
+    wk_gpu_store *g = wk_gpu_store_create(st, device);
+    if (!g) return nullptr;
+    g->owned = true;
+    wk_engine *e = wk_engine_create_on(g);
+    if (!e) { g->owned = false; wk_gpu_store_destroy(g); return nullptr; }
     return e;
 }
 
@@ -766,13 +820,17 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
     for (int i = 0; i < 2; i++) e->tbl[i].release();
     e->eoff.release(); e->cnt.release(); e->prefix.release();
     e->bsums.release(); e->misc.release();
-    if (e->d_verts) (void)hipFree(e->d_verts);
-    if (e->d_edges) (void)hipFree(e->d_edges);
-    if (e->d_type_of) (void)hipFree(e->d_type_of);
     if (e->d_state) (void)hipFree(e->d_state);
     if (e->h_pin) (void)hipHostFree(e->h_pin);
     if (e->h_stage) (void)hipHostFree(e->h_stage);
     if (e->stream) (void)hipStreamDestroy(e->stream);
+    if (e->gs) {
+        e->gs->refs--;
+        if (e->gs->refs == 0 && e->gs->owned) {
+            e->gs->owned = false;
+            wk_gpu_store_destroy(e->gs);
+        }
+    }
     delete e;
 }
 
@@ -860,7 +918,15 @@ static int32_t sync_state(wk_engine *e) {
     e->nrows = (int64_t)e->h_pin[S_NROWS];
     e->bound = e->nrows;
     for (int i = 0; i < CAT_COUNT; i++) e->cat_bytes[i] = (double)e->h_pin[8 + i];
-    if (e->h_pin[S_ERR]) return WK_ERR_CAP;
+    if (e->h_pin[S_ERR]) {
+        // grow now so the caller can simply resubmit the plan
+        int64_t need = (int64_t)e->h_pin[S_REQ];
+        int32_t rc = grow_caps(e, need + need / 4, e->cap_cols);
+        if (rc) return rc;
+        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                           e->d_state, S_WORDS);
+        return WK_ERR_CAP;
+    }
     return WK_OK;
 }
 
@@ -1258,6 +1324,19 @@ extern "C" int32_t wk_engine_fetch_raw(wk_engine_t *e, wk_result_t *out) {
     out->row_num = e->nrows;
     out->table = res;
     out->status_code = 0;
+    return WK_OK;
+}
+
+// asynchronous whole-plan submission: enqueue the full launch chain and
+// return without any sync (pipelined multi-engine execution — the
+// reference proxy's in-flight window, proxy.hpp:477-525)
+extern "C" int32_t wk_engine_submit(wk_engine_t *e, const wk_plan_t *plan) {
+    int32_t rc = wk_engine_begin_query(e, plan);
+    if (rc) return rc;
+    while (e->step < (int)e->pats.size()) {
+        rc = exec_pattern(e);
+        if (rc) return rc;
+    }
     return WK_OK;
 }
 
