@@ -83,9 +83,11 @@ enum { PM_SIZE = 0,    // k2u: write edge count per row
 
 // device query state (see engine): [0]=nrows [1]=scan total [2]=overflow
 // flag [3]=required rows; stats[0..6] = algorithmic bytes per category
-enum { S_NROWS = 0, S_TOTAL = 1, S_ERR = 2, S_REQ = 3, S_WORDS = 4 };
+enum { S_NROWS = 0, S_TOTAL = 1, S_ERR = 2, S_REQ = 3, S_OVF = 4, S_WORDS = 5 };
 enum { CAT_PROBE = 0, CAT_SCAN, CAT_EXPAND, CAT_FILTER, CAT_COPY, CAT_SPLIT,
        CAT_OTHER, CAT_COUNT };
+
+constexpr int SCAN_T = 256;  // scan tile = block size of the fused kernels
 
 __device__ __forceinline__ void count_bytes(uint64_t *stats, int cat,
                                             uint64_t bytes) {
@@ -108,83 +110,84 @@ __device__ __forceinline__ bool bsearch_u32(const sid_t *a, uint64_t n, sid_t x)
     return false;
 }
 
-// Thread-per-row probe (default): one LANE walks one row's bucket chain
-// (7 slot compares in one 128-B line).  64 rows in flight per wave —
-// maximal memory-level parallelism for DRAM-random probes
-// (gstore.hpp:341-361 semantics; dispatch per sparql.hpp:1016-1058).
-__global__ void k_probe_tpr(const vertex_t *__restrict__ verts,
-                            const sid_t *__restrict__ edges,
-                            uint64_t bucket_start, uint64_t num_buckets,
-                            const sid_t *__restrict__ tbl, int ncols,
-                            int col, uint32_t pid, int dir, int key_mode,
-                            int probe_mode, int col2, sid_t cval,
-                            uint64_t list_off, uint64_t list_sz,
-                            uint64_t *__restrict__ d_state,
-                            uint64_t *__restrict__ d_stats,
-                            uint64_t *__restrict__ d_eoff,
-                            uint64_t *__restrict__ d_cnt,
-                            sid_t *__restrict__ out_tbl)
+// one cluster-hash lookup: walk the bucket chain, 7 slot compares per
+// 128-B bucket (gstore.hpp:341-361 semantics)
+__device__ __forceinline__ void probe_one(const vertex_t *__restrict__ verts,
+                                          uint64_t bucket_start,
+                                          uint64_t num_buckets, uint64_t key,
+                                          uint64_t &eoff, uint64_t &esz) {
+    uint64_t bucket = bucket_start + hash_u64(key) % num_buckets;
+    while (true) {
+        const vertex_t *b = &verts[bucket * ASSOC];
+        uint64_t k0 = b[0].key, k1 = b[1].key, k2 = b[2].key, k3 = b[3].key;
+        uint64_t k4 = b[4].key, k5 = b[5].key, k6 = b[6].key, k7 = b[7].key;
+        int hit = -1;
+        if (k0 == key) hit = 0;
+        else if (k1 == key) hit = 1;
+        else if (k2 == key) hit = 2;
+        else if (k3 == key) hit = 3;
+        else if (k4 == key) hit = 4;
+        else if (k5 == key) hit = 5;
+        else if (k6 == key) hit = 6;
+        if (hit >= 0) {
+            uint64_t pp = b[hit].ptr;
+            eoff = ptr_off(pp);
+            esz = ptr_size(pp);
+            return;
+        }
+        if (k7 == KEY_EMPTY) { eoff = 0; esz = 0; return; }
+        bucket = key_vid(k7);
+    }
+}
+
+// Fused known_to_unknown front half: contiguous-chunk thread-per-row
+// probe + in-kernel exclusive prefix of the edge counts (local prefix +
+// per-block sums; k_scan_mid finishes across blocks).  64 rows in
+// flight per wave between scan barriers — MLP for DRAM-random probes
+// (gstore.hpp:341-361; replaces gpu_hash.cu:94-324 + the thrust scan).
+__global__ void k_probe_scan(const vertex_t *__restrict__ verts,
+                             const sid_t *__restrict__ tbl, int ncols,
+                             int col, uint32_t pid, int dir, int key_mode,
+                             uint64_t bucket_start, uint64_t num_buckets,
+                             uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats,
+                             uint64_t *__restrict__ d_eoff,
+                             uint32_t *__restrict__ d_cnt,
+                             uint64_t *__restrict__ d_pre,
+                             uint64_t *__restrict__ bsums)
 {
     const int64_t nrows = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, probe_mode == PM_LIST ? CAT_FILTER : CAT_PROBE,
-                (uint64_t)nrows * (probe_mode == PM_LIST
-                                       ? 12
-                                       : (4 + 128 + 8 + (probe_mode != PM_SIZE ? 64 : 0))));
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
-         r += (int64_t)gridDim.x * blockDim.x) {
-        if (probe_mode == PM_LIST) {
-            sid_t v = tbl[r * ncols + col];
-            if (bsearch_u32(edges + list_off, list_sz, v)) {
-                unsigned long long pos = atomicAdd(
-                    (unsigned long long *)&d_state[S_TOTAL], 1ull);
-                sid_t *dst = out_tbl + (int64_t)pos * ncols;
-                const sid_t *src = tbl + r * ncols;
-                for (int c = 0; c < ncols; c++) dst[c] = src[c];
-            }
-            continue;
-        }
-        sid_t v = tbl[r * ncols + col];
-        uint64_t key = (key_mode == PK_NORMAL) ? key_pack(v, pid, (uint64_t)dir)
-                                               : key_pack(0, v, (uint64_t)dir);
-        uint64_t bucket = bucket_start + hash_u64(key) % num_buckets;
+    count_bytes(d_stats, CAT_PROBE, (uint64_t)nrows * (4 + 128 + 8 + 12));
+    const int64_t chunk = (nrows + gridDim.x - 1) / gridDim.x;
+    const int64_t start = (int64_t)blockIdx.x * chunk;
+    const int64_t end = min(start + chunk, nrows);
+    __shared__ uint64_t sh[SCAN_T];
+    uint64_t carry = 0;
+    for (int64_t base = start; base < end; base += SCAN_T) {
+        const int64_t r = base + threadIdx.x;
         uint64_t eoff = 0, esz = 0;
-        while (true) {
-            const vertex_t *b = &verts[bucket * ASSOC];
-            uint64_t k0 = b[0].key, k1 = b[1].key, k2 = b[2].key, k3 = b[3].key;
-            uint64_t k4 = b[4].key, k5 = b[5].key, k6 = b[6].key, k7 = b[7].key;
-            int hit = -1;
-            if (k0 == key) hit = 0;
-            else if (k1 == key) hit = 1;
-            else if (k2 == key) hit = 2;
-            else if (k3 == key) hit = 3;
-            else if (k4 == key) hit = 4;
-            else if (k5 == key) hit = 5;
-            else if (k6 == key) hit = 6;
-            if (hit >= 0) {
-                uint64_t pp = b[hit].ptr;
-                eoff = ptr_off(pp); esz = ptr_size(pp);
-                break;
-            }
-            if (k7 == KEY_EMPTY) { esz = 0; break; }
-            bucket = key_vid(k7);
-        }
-        if (probe_mode == PM_SIZE) {
+        if (r < end) {
+            sid_t v = tbl[r * ncols + col];
+            uint64_t key = (key_mode == PK_NORMAL)
+                               ? key_pack(v, pid, (uint64_t)dir)
+                               : key_pack(0, v, (uint64_t)dir);
+            probe_one(verts, bucket_start, num_buckets, key, eoff, esz);
             d_eoff[r] = eoff;
-            d_cnt[r] = esz;
-        } else {
-            // fused filter + atomic compaction (keep-row semantics of
-            // sparql.hpp:455-476; row order is engine-internal, parity
-            // is set-level): replaces probe->scan->compact
-            sid_t tgt = (probe_mode == PM_CONST) ? cval : tbl[r * ncols + col2];
-            if (esz && bsearch_u32(edges + eoff, esz, tgt)) {
-                unsigned long long pos = atomicAdd(
-                    (unsigned long long *)&d_state[S_TOTAL], 1ull);
-                sid_t *dst = out_tbl + (int64_t)pos * ncols;
-                const sid_t *src = tbl + r * ncols;
-                for (int c = 0; c < ncols; c++) dst[c] = src[c];
-            }
+            d_cnt[r] = (uint32_t)esz;
         }
+        sh[threadIdx.x] = esz;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint64_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += x;
+            __syncthreads();
+        }
+        if (r < end) d_pre[r] = carry + sh[threadIdx.x] - esz;
+        carry += sh[SCAN_T - 1];
+        __syncthreads();
     }
+    if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
 }
 
 // Filter operators (k2c/k2k/c2k/i2k): probe + membership test + BLOCK-
@@ -366,47 +369,8 @@ __global__ void k_probe(const vertex_t *__restrict__ verts,
     }
 }
 
-// ---- 3-phase exclusive scan with device-side length ------------------
-// m = state[S_NROWS] + 1 elements of cnt (cnt[m-1] pre-zeroed by
-// k_zero_at) -> prefix; total -> state[S_TOTAL].
-static const int SCAN_T = 256;
-
-__global__ void k_zero_at(uint64_t *cnt, const uint64_t *__restrict__ d_state) {
-    cnt[d_state[S_NROWS]] = 0;
-}
-
-__global__ void k_scan_part(const uint64_t *__restrict__ cnt,
-                            const uint64_t *__restrict__ d_state,
-                            uint64_t *__restrict__ pre,
-                            uint64_t *__restrict__ bsums,
-                            uint64_t *__restrict__ d_stats)
-{
-    count_bytes(d_stats, CAT_SCAN, (d_state[S_NROWS] + 1) * 8 * 2);
-    const int64_t m = (int64_t)d_state[S_NROWS] + 1;
-    const int64_t chunk = (m + gridDim.x - 1) / gridDim.x;
-    const int64_t start = (int64_t)blockIdx.x * chunk;
-    const int64_t end = min(start + chunk, m);
-    __shared__ uint64_t sh[SCAN_T];
-    uint64_t carry = 0;
-    for (int64_t base = start; base < end; base += SCAN_T) {
-        int64_t i = base + threadIdx.x;
-        uint64_t x = (i < end) ? cnt[i] : 0;
-        sh[threadIdx.x] = x;
-        __syncthreads();
-        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
-            uint64_t v = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
-            __syncthreads();
-            sh[threadIdx.x] += v;
-            __syncthreads();
-        }
-        if (i < end) pre[i] = carry + sh[threadIdx.x] - x;
-        carry += sh[SCAN_T - 1];
-        __syncthreads();
-    }
-    if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
-}
-
-// single block: exclusive-scan the G block sums in place, total -> state
+// finish the scan across blocks: exclusive over the G block sums
+// (single block), total -> state[S_TOTAL]
 __global__ void k_scan_mid(uint64_t *__restrict__ bsums, int G,
                            uint64_t *__restrict__ d_state)
 {
@@ -428,17 +392,6 @@ __global__ void k_scan_mid(uint64_t *__restrict__ bsums, int G,
         __syncthreads();
     }
     if (threadIdx.x == 0) d_state[S_TOTAL] = carry;
-}
-
-__global__ void k_scan_add(uint64_t *__restrict__ pre,
-                           const uint64_t *__restrict__ bsums, int G,
-                           const uint64_t *__restrict__ d_state)
-{
-    const int64_t m = (int64_t)d_state[S_NROWS] + 1;
-    const int64_t chunk = (m + G - 1) / G;
-    for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < m;
-         i += (int64_t)gridDim.x * blockDim.x)
-        pre[i] += bsums[i / chunk];
 }
 
 // advance: nrows = min(total, cap); flag overflow for the host re-run
@@ -480,54 +433,92 @@ __global__ void k_publish_state(const uint64_t *__restrict__ d_state,
     for (int i = 0; i < CAT_COUNT; i++) h_pin[8 + i] = d_stats[i];
 }
 
-// Output-centric expansion: one thread per OUTPUT row; binary-search the
-// prefix array for the source row.  Consecutive threads write consecutive
-// rows -> coalesced stores (known_to_unknown semantics, sparql.hpp:325-367).
-__global__ void k_expand(const sid_t *__restrict__ tbl, int ncols,
-                         const sid_t *__restrict__ edges,
-                         const uint64_t *__restrict__ d_eoff,
-                         const uint64_t *__restrict__ d_prefix,
-                         const uint64_t *__restrict__ d_state, uint64_t cap,
-                         uint64_t *__restrict__ d_stats,
-                         sid_t *__restrict__ out)
+// Input-centric expansion (known_to_unknown back half,
+// sparql.hpp:325-367): thread r writes its deg outputs at
+// pre[r]+bsums[chunk(r)] — no per-output binary search (the
+// output-centric version paid 2-3 DRAM lines of prefix walk per output
+// row).  Rows with deg > 32 go to an overflow queue handled by
+// k_expand_big with one WAVE per row (lanes stride the edge list).
+__global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
+                            const sid_t *__restrict__ edges,
+                            const uint64_t *__restrict__ d_eoff,
+                            const uint32_t *__restrict__ d_cnt,
+                            const uint64_t *__restrict__ d_pre,
+                            const uint64_t *__restrict__ bsums, int G,
+                            uint64_t *__restrict__ d_state, uint64_t cap,
+                            uint64_t *__restrict__ d_stats,
+                            uint32_t *__restrict__ ovf,
+                            sid_t *__restrict__ out)
 {
-    const int64_t nrows_in = (int64_t)d_state[S_NROWS];
-    const int64_t total = (int64_t)min(d_state[S_TOTAL], cap);
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int64_t chunk = (nrows + G - 1) / G;
     const int oc = ncols + 1;
-    count_bytes(d_stats, CAT_EXPAND, (uint64_t)total * (4 + 4 * ncols + 4 * oc));
-    for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < total;
-         t += (int64_t)gridDim.x * blockDim.x) {
-        int64_t lo = 0, hi = nrows_in;
-        while (lo + 1 < hi) {
-            int64_t mid = (lo + hi) >> 1;
-            if (d_prefix[mid] <= (uint64_t)t) lo = mid; else hi = mid;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
+         r += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t deg = d_cnt[r];
+        if (!deg) continue;
+        uint64_t basep = d_pre[r] + bsums[r / chunk];
+        if (basep >= cap) continue;  // overflow: flagged by k_commit
+        if (deg > 32) {
+            unsigned long long i = atomicAdd(
+                (unsigned long long *)&d_state[S_OVF], 1ull);
+            ovf[i] = (uint32_t)r;
+            continue;
         }
-        const int64_t r = lo;
-        const uint64_t k = (uint64_t)t - d_prefix[r];
-        sid_t *dst = out + t * oc;
-        const sid_t *src = tbl + r * ncols;
-        for (int c = 0; c < ncols; c++) dst[c] = src[c];
-        dst[ncols] = edges[d_eoff[r] + k];
+        if (basep + deg > cap) deg = (uint32_t)(cap - basep);
+        sid_t row[8];
+        for (int c = 0; c < ncols; c++) row[c] = tbl[r * ncols + c];
+        const sid_t *el = edges + d_eoff[r];
+        sid_t *dst = out + (int64_t)basep * oc;
+        for (uint32_t k = 0; k < deg; k++) {
+            for (int c = 0; c < ncols; c++) dst[c] = row[c];
+            dst[ncols] = el[k];
+            dst += oc;
+        }
     }
+    // algorithmic bytes for the whole expansion (counted once; includes
+    // the big-row pass): total*(edge 4 + write 4*oc) + nrows*(row 4*ncols
+    // + cnt/pre/eoff 20)
+    if (blockIdx.x == 0 && threadIdx.x == 0)
+        atomicAdd((unsigned long long *)&d_stats[CAT_EXPAND],
+                  (unsigned long long)(min(d_state[S_TOTAL], cap) * (4 + 4 * oc) +
+                                       (uint64_t)nrows * (4 * ncols + 20)));
 }
 
-// Flag-compaction (k2k/k2c/c2k/i2k keep-row semantics, sparql.hpp:455-476).
-// Output rows <= input rows <= cap: never overflows.
-__global__ void k_compact(const sid_t *__restrict__ tbl, int ncols,
-                          const uint64_t *__restrict__ d_cnt,
-                          const uint64_t *__restrict__ d_prefix,
-                          const uint64_t *__restrict__ d_state,
-                          uint64_t *__restrict__ d_stats,
-                          sid_t *__restrict__ out)
+// big-fanout rows: one wave per queued row, lanes stride the edge list
+// (coalesced writes: adjacent lanes write adjacent output rows)
+__global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
+                             const sid_t *__restrict__ edges,
+                             const uint64_t *__restrict__ d_eoff,
+                             const uint32_t *__restrict__ d_cnt,
+                             const uint64_t *__restrict__ d_pre,
+                             const uint64_t *__restrict__ bsums, int G,
+                             const uint64_t *__restrict__ d_state, uint64_t cap,
+                             const uint32_t *__restrict__ ovf,
+                             sid_t *__restrict__ out)
 {
-    const int64_t nrows_in = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows_in * (16 + 4 * ncols));
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows_in;
-         r += (int64_t)gridDim.x * blockDim.x) {
-        if (!d_cnt[r]) continue;
-        sid_t *dst = out + (int64_t)d_prefix[r] * ncols;
-        const sid_t *src = tbl + r * ncols;
-        for (int c = 0; c < ncols; c++) dst[c] = src[c];
+    const int64_t nq = (int64_t)d_state[S_OVF];
+    if (!nq) return;
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int64_t chunk = (nrows + G - 1) / G;
+    const int oc = ncols + 1;
+    const int lane = threadIdx.x & 63;
+    const int64_t w0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    for (int64_t q = w0; q < nq; q += nw) {
+        const int64_t r = ovf[q];
+        uint64_t deg = d_cnt[r];
+        uint64_t basep = d_pre[r] + bsums[r / chunk];
+        if (basep >= cap) continue;
+        if (basep + deg > cap) deg = cap - basep;
+        sid_t row[8];
+        for (int c = 0; c < ncols; c++) row[c] = tbl[r * ncols + c];
+        const sid_t *el = edges + d_eoff[r];
+        for (uint64_t k = lane; k < deg; k += 64) {
+            sid_t *dst = out + (int64_t)(basep + k) * oc;
+            for (int c = 0; c < ncols; c++) dst[c] = row[c];
+            dst[ncols] = el[k];
+        }
     }
 }
 
@@ -633,7 +624,7 @@ struct wk_engine {
     // dual result buffer (gpu_mem.hpp:116-124) + scratch, all sized by
     // cap_rows (grow-only; overflow -> re-run)
     devbuf tbl[2];
-    devbuf eoff, cnt, prefix, bsums, misc;
+    devbuf eoff, cnt, prefix, bsums, misc, ovf;
     int64_t cap_rows = 0;
     int cap_cols = 0;
 
@@ -654,8 +645,6 @@ struct wk_engine {
     int nvars = 0;
     int step = 0;
 
-    bool probe_coop = false;  // WK_PROBE=coop selects the 8-lane kernel
-
     // timing (WK_KERNEL_TIMING=1)
     bool timing = false;
     std::vector<timed_launch> pending;
@@ -674,7 +663,7 @@ static int grid_for(int64_t work) {
 }
 static int scan_grid(int64_t bound) {
     int64_t g = (bound + SCAN_T) / SCAN_T + 1;
-    return (int)(g < 1 ? 1 : (g > 512 ? 512 : g));
+    return (int)(g < 1 ? 1 : (g > 2048 ? 2048 : g));
 }
 
 static hipEvent_t ev_get() { hipEvent_t e; (void)hipEventCreate(&e); return e; }
@@ -728,10 +717,11 @@ static int32_t grow_caps(wk_engine *e, int64_t rows, int cols) {
     HIP_CHECK(stream_sync(e->stream));
     if (e->tbl[0].ensure((size_t)rows * cols * 4)) return WK_ERR_HIP;
     if (e->tbl[1].ensure((size_t)rows * cols * 4)) return WK_ERR_HIP;
-    if (e->cnt.ensure((size_t)(rows + 1) * 8)) return WK_ERR_HIP;
+    if (e->cnt.ensure((size_t)(rows + 1) * 4)) return WK_ERR_HIP;   // u32 degs
     if (e->eoff.ensure((size_t)(rows + 1) * 8)) return WK_ERR_HIP;
     if (e->prefix.ensure((size_t)(rows + 1) * 8)) return WK_ERR_HIP;
-    if (e->bsums.ensure(520 * 8)) return WK_ERR_HIP;
+    if (e->ovf.ensure((size_t)(rows + 1) * 4)) return WK_ERR_HIP;
+    if (e->bsums.ensure(2056 * 8)) return WK_ERR_HIP;
     e->cap_rows = rows;
     e->cap_cols = cols;
     return WK_OK;
@@ -800,7 +790,6 @@ extern "C" wk_engine_t *wk_engine_create_on(wk_gpu_store_t *g) {
         return nullptr;
     }
     if (const char *t = getenv("WK_KERNEL_TIMING")) e->timing = atoi(t) != 0;
-    if (const char *pv = getenv("WK_PROBE")) e->probe_coop = !strcmp(pv, "coop");
     if (grow_caps(e, 1 << 20, 4) != WK_OK) { wk_engine_destroy(e); return nullptr; }
     return e;
 }
@@ -819,7 +808,7 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
     resolve_timing(e);
     for (int i = 0; i < 2; i++) e->tbl[i].release();
     e->eoff.release(); e->cnt.release(); e->prefix.release();
-    e->bsums.release(); e->misc.release();
+    e->bsums.release(); e->misc.release(); e->ovf.release();
     if (e->d_state) (void)hipFree(e->d_state);
     if (e->h_pin) (void)hipHostFree(e->h_pin);
     if (e->h_stage) (void)hipHostFree(e->h_stage);
@@ -889,24 +878,6 @@ extern "C" int32_t wk_engine_load_rbuf_device(wk_engine_t *e, const sid_t *dev_t
         HIP_CHECK(hipMemcpyAsync(e->tbl[0].p, dev_table, bytes,
                                  hipMemcpyDeviceToDevice, e->stream));
     return load_common(e, nrows, ncols, v2c_map, pattern_step);
-}
-
-// enqueue the device-length scan chain: cnt[0..n] -> prefix, total ->
-// state[S_TOTAL] (no host sync)
-static void enqueue_scan(wk_engine *e) {
-    const int G = scan_grid(e->bound);
-    hipLaunchKernelGGL(k_zero_at, dim3(1), dim3(1), 0, e->stream,
-                       (uint64_t *)e->cnt.p, e->d_state);
-    TIME_BEGIN(e);
-    hipLaunchKernelGGL(k_scan_part, dim3(G), dim3(SCAN_T), 0, e->stream,
-                       (uint64_t *)e->cnt.p, e->d_state, (uint64_t *)e->prefix.p,
-                       (uint64_t *)e->bsums.p, e->d_stats);
-    hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0, e->stream,
-                       (uint64_t *)e->bsums.p, G, e->d_state);
-    hipLaunchKernelGGL(k_scan_add, dim3(G), dim3(SCAN_T), 0, e->stream,
-                       (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
-                       e->d_state);
-    TIME_END(e, CAT_SCAN);
 }
 
 // publish device state to pinned memory + sync; refresh host nrows/stats
@@ -1042,36 +1013,43 @@ static int32_t exec_pattern(wk_engine *e) {
                            use_typeof, e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
     } else {
-        TIME_BEGIN(e);
-        if (e->probe_coop)
-            hipLaunchKernelGGL(k_probe, dim3(grid_for(e->bound * 8)), dim3(BLOCK), 0,
-                               e->stream, e->d_verts, e->d_edges, seg->bucket_start,
-                               seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
-                               dir, key_mode, pmode, col2, cval, 0, 0, e->d_state,
-                               e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p,
-                               out_tbl);
-        else
-            hipLaunchKernelGGL(k_probe_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
-                               e->stream, e->d_verts, e->d_edges, seg->bucket_start,
-                               seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
-                               dir, key_mode, pmode, col2, cval, 0, 0, e->d_state,
-                               e->d_stats, (uint64_t *)e->eoff.p, (uint64_t *)e->cnt.p,
-                               out_tbl);
-        TIME_END(e, CAT_PROBE);
-    }
-
-    if (pmode == PM_SIZE) {
-        enqueue_scan(e);
-        // known_to_unknown: expand (bound multiplies by unknown fan-out;
-        // grid heuristic x16, correctness by grid-stride)
+        // known_to_unknown: fused probe+scan -> cross-block scan ->
+        // input-centric expansion (+ big-row wave pass)
+        const int G = scan_grid(e->bound);
         int oc = e->ncols + 1;
-        if (oc > e->cap_cols) return WK_ERR_STATE;  // begin_query sizes cap_cols>=nvars
+        if (oc > e->cap_cols) return WK_ERR_STATE;  // begin_query sizes cap_cols
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_expand, dim3(grid_for(e->cap_rows)), dim3(BLOCK), 0,
-                           e->stream, cur_tbl, e->ncols, e->d_edges,
-                           (uint64_t *)e->eoff.p, (uint64_t *)e->prefix.p,
-                           e->d_state, (uint64_t)e->cap_rows, e->d_stats, out_tbl);
-        TIME_END(e, CAT_EXPAND);
+        hipLaunchKernelGGL(k_probe_scan, dim3(G), dim3(SCAN_T), 0, e->stream,
+                           e->d_verts, cur_tbl, e->ncols, col, (uint32_t)p, dir,
+                           key_mode, seg->bucket_start, seg->num_buckets,
+                           e->d_state, e->d_stats, (uint64_t *)e->eoff.p,
+                           (uint32_t *)e->cnt.p, (uint64_t *)e->prefix.p,
+                           (uint64_t *)e->bsums.p);
+        TIME_END(e, CAT_PROBE);
+        {
+            TIME_BEGIN(e);
+            hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0, e->stream,
+                               (uint64_t *)e->bsums.p, G, e->d_state);
+            TIME_END(e, CAT_SCAN);
+        }
+        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                           e->d_state + S_OVF, 1);
+        {
+            TIME_BEGIN(e);
+            hipLaunchKernelGGL(k_expand_in, dim3(grid_for(e->bound)), dim3(BLOCK),
+                               0, e->stream, cur_tbl, e->ncols, e->d_edges,
+                               (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p,
+                               (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
+                               e->d_state, (uint64_t)e->cap_rows, e->d_stats,
+                               (uint32_t *)e->ovf.p, out_tbl);
+            hipLaunchKernelGGL(k_expand_big, dim3(512), dim3(BLOCK), 0, e->stream,
+                               cur_tbl, e->ncols, e->d_edges,
+                               (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p,
+                               (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
+                               e->d_state, (uint64_t)e->cap_rows,
+                               (uint32_t *)e->ovf.p, out_tbl);
+            TIME_END(e, CAT_EXPAND);
+        }
         e->v2c[-(o + 1)] = e->ncols;
         e->ncols = oc;
         e->bound = e->cap_rows;  // fan-out unknown until a sync point
