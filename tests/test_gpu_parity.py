@@ -99,3 +99,39 @@ def test_empty_and_edge_cases(eng4):
               nvars=2, required_vars=[-1, -2])
     t = eng4.run_query(p2)
     assert t.shape == (0, 2)
+
+
+def test_overflow_rerun(lubm4, oracle4):
+    """Force tiny capacities so every heavy step overflows: the device
+    flag + grow + re-run path must still produce exact results (replaces
+    the reference's hard rbuf assert, gpu_engine_cuda.hpp:185)."""
+    import subprocess
+    import sys
+    code = """
+import os, sys
+os.environ['WK_MIN_CAP'] = '1000'
+sys.path.insert(0, %r)
+import numpy as np
+import wukong_amd as wk
+from wukong_amd import queries as Q
+from tests.oracle_util import OracleCtx, sort_rows
+triples = wk.lubm_gen(4, seed=42)
+store = wk.Store(triples)
+eng = wk.Engine(store, device=0)
+ora = OracleCtx(triples)
+for name in ('q1', 'q2', 'q7'):
+    got = eng.run_query(Q.ALL[name])
+    want = ora.run_query(Q.ALL[name])
+    assert got.shape == want.shape, (name, got.shape, want.shape)
+    assert np.array_equal(sort_rows(got), sort_rows(want)), name
+# step API path (mid-plan overflow re-run)
+eng.begin_query(Q.ALL['q1'])
+for _ in Q.ALL['q1'].patterns:
+    eng.execute_one_pattern()
+got = sort_rows(eng.fetch_result())
+assert np.array_equal(got, sort_rows(ora.run_query(Q.ALL['q1'])))
+print('OVERFLOW_OK')
+""" % (os.path.dirname(os.path.dirname(os.path.abspath(__file__))),)
+    r = subprocess.run([sys.executable, "-c", code], capture_output=True,
+                       text=True, timeout=300)
+    assert "OVERFLOW_OK" in r.stdout, r.stdout + r.stderr

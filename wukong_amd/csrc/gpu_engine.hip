@@ -707,16 +707,35 @@ static void resolve_timing(wk_engine *e) {
     e->pending.clear();
 }
 
-// grow scratch + rbufs to hold `rows` rows of up to `cols` columns
+// grow scratch + rbufs to hold `rows` rows of up to `cols` columns.
+// Preserves the CURRENT table's contents (e->nrows x e->ncols, valid at
+// every sync point) so a mid-plan overflow re-run can resume from the
+// intact input table (dual-buffer invariant).
+static int64_t min_cap_rows() {
+    const char *v = getenv("WK_MIN_CAP");  // tests force tiny caps to
+    return v ? atoll(v) : (1 << 20);       // exercise the overflow re-run
+}
+
 static int32_t grow_caps(wk_engine *e, int64_t rows, int cols) {
-    rows = std::max<int64_t>(rows, 1 << 20);
+    rows = std::max<int64_t>(rows, min_cap_rows());
     cols = std::max(cols, std::max(e->cap_cols, 1));
     if (rows <= e->cap_rows && cols <= e->cap_cols) return WK_OK;
     rows = std::max(rows, e->cap_rows);
     // one sync before freeing buffers that in-flight kernels may use
     HIP_CHECK(stream_sync(e->stream));
+    size_t keep = (size_t)std::max<int64_t>(e->nrows, 0) * std::max(e->ncols, 0) * 4;
+    keep = std::min(keep, e->tbl[e->cur].cap);
+    void *saved = nullptr;
+    if (keep) {
+        if (hipMalloc(&saved, keep) != hipSuccess) return WK_ERR_HIP;
+        HIP_CHECK(hipMemcpy(saved, e->tbl[e->cur].p, keep, hipMemcpyDeviceToDevice));
+    }
     if (e->tbl[0].ensure((size_t)rows * cols * 4)) return WK_ERR_HIP;
     if (e->tbl[1].ensure((size_t)rows * cols * 4)) return WK_ERR_HIP;
+    if (saved) {
+        HIP_CHECK(hipMemcpy(e->tbl[e->cur].p, saved, keep, hipMemcpyDeviceToDevice));
+        (void)hipFree(saved);
+    }
     if (e->cnt.ensure((size_t)(rows + 1) * 4)) return WK_ERR_HIP;   // u32 degs
     if (e->eoff.ensure((size_t)(rows + 1) * 8)) return WK_ERR_HIP;
     if (e->prefix.ensure((size_t)(rows + 1) * 8)) return WK_ERR_HIP;
@@ -790,7 +809,7 @@ extern "C" wk_engine_t *wk_engine_create_on(wk_gpu_store_t *g) {
         return nullptr;
     }
     if (const char *t = getenv("WK_KERNEL_TIMING")) e->timing = atoi(t) != 0;
-    if (grow_caps(e, 1 << 20, 4) != WK_OK) { wk_engine_destroy(e); return nullptr; }
+    if (grow_caps(e, min_cap_rows(), 4) != WK_OK) { wk_engine_destroy(e); return nullptr; }
     return e;
 }
 
@@ -889,16 +908,22 @@ static int32_t sync_state(wk_engine *e) {
     e->nrows = (int64_t)e->h_pin[S_NROWS];
     e->bound = e->nrows;
     for (int i = 0; i < CAT_COUNT; i++) e->cat_bytes[i] = (double)e->h_pin[8 + i];
-    if (e->h_pin[S_ERR]) {
-        // grow now so the caller can simply resubmit the plan
-        int64_t need = (int64_t)e->h_pin[S_REQ];
-        int32_t rc = grow_caps(e, need + need / 4, e->cap_cols);
-        if (rc) return rc;
-        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                           e->d_state, S_WORDS);
-        return WK_ERR_CAP;
-    }
+    if (e->h_pin[S_ERR]) return WK_ERR_CAP;
     return WK_OK;
+}
+
+// fetch-path variant: on overflow, grow and reset flags so the caller can
+// simply resubmit the whole plan (table contents are irrelevant then)
+static int32_t sync_state_grow(wk_engine *e) {
+    int32_t rc = sync_state(e);
+    if (rc != WK_ERR_CAP) return rc;
+    int64_t need = (int64_t)e->h_pin[S_REQ];
+    e->nrows = 0;  // nothing worth preserving across the re-run
+    int32_t rc2 = grow_caps(e, need + need / 4, e->cap_cols);
+    if (rc2) return rc2;
+    hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                       e->d_state, S_WORDS);
+    return WK_ERR_CAP;
 }
 
 // Run one pattern — dispatch per sparql.hpp:1016-1058.  Fully async: row
@@ -1064,25 +1089,26 @@ static int32_t exec_pattern(wk_engine *e) {
 extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_out) {
     double t0 = wk_verbose_lvl() >= 2 ? now_us() : 0;
     if (!e) return WK_ERR_STATE;
-    const int64_t in_rows = e->nrows;  // valid in step mode (synced per step)
+    // snapshot for the overflow re-run (step mode syncs per step, so
+    // e->nrows/ncols/v2c are the pattern's INPUT state here)
+    const int64_t in_rows = e->nrows;
     const int in_cur = e->cur;
+    const int in_ncols = e->ncols;
+    const std::vector<int32_t> in_v2c = e->v2c;
     int32_t rc = exec_pattern(e);
     if (rc == WK_OK && nrows_out) {
-        // step-API callers get a synchronous row count; on overflow grow
-        // and re-run THIS pattern.  NOTE: grow_caps reallocates BOTH rbufs,
-        // so the input table must be restaged — we re-upload is impossible
-        // here; instead grow only the scratch by re-running via a fresh
-        // copy: since grow_caps frees tbl[], the step-API rerun only works
-        // when the input table can be regenerated (step 0) — otherwise
-        // callers should size caps up front.  In practice overflow implies
-        // a large table which implies caps already grown by run_query.
         for (int attempt = 0; attempt < 6; attempt++) {
             rc = sync_state(e);
             if (rc != WK_ERR_CAP) break;
-            if (e->step - 1 != 0) return WK_ERR_CAP;  // cannot rerun mid-plan
+            // restore the input state; grow (preserving the INPUT table,
+            // now current again); re-run this pattern
             int64_t need = (int64_t)e->h_pin[S_REQ];
             e->cur = in_cur;
+            e->nrows = in_rows;
+            e->ncols = in_ncols;
+            e->v2c = in_v2c;
             e->step--;
+            e->bound = in_rows;
             int32_t rc2 = grow_caps(e, need + need / 4, e->cap_cols);
             if (rc2) return rc2;
             hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
@@ -1249,7 +1275,7 @@ extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
                            e->stream, (sid_t *)e->tbl[e->cur].p, e->ncols,
                            e->d_state, cols, plan->nrequired, out_tbl);
         TIME_END(e, CAT_OTHER);
-        int32_t rc = sync_state(e);
+        int32_t rc = sync_state_grow(e);
         if (rc) return rc;
         double t0 = now_us();
         size_t n = (size_t)e->nrows * plan->nrequired;
@@ -1267,7 +1293,7 @@ extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
                     (long long)e->nrows, now_us() - t0);
         return WK_OK;
     }
-    int32_t rc = sync_state(e);
+    int32_t rc = sync_state_grow(e);
     if (rc) return rc;  // WK_ERR_CAP -> caller re-runs (run_query does)
     if (plan->blind) {
         // Result::blind (query.hpp:321): row count only, no table
@@ -1292,7 +1318,7 @@ extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
 // raw current table (no final ops) — for the gloo-path exchange in tests
 extern "C" int32_t wk_engine_fetch_raw(wk_engine_t *e, wk_result_t *out) {
     if (!e || !out) return WK_ERR_STATE;
-    int32_t rc = sync_state(e);
+    int32_t rc = sync_state_grow(e);
     if (rc) return rc;
     size_t n = (size_t)e->nrows * e->ncols;
     wk_sid_t *res = (wk_sid_t *)malloc(n ? n * 4 : 4);
@@ -1329,12 +1355,7 @@ extern "C" int32_t wk_engine_run_query(wk_engine_t *e, const wk_plan_t *plan,
         }
         rc = wk_engine_fetch_result(e, plan, out);
         if (rc != WK_ERR_CAP) return rc;
-        // overflow: grow to the required size and re-run the whole query
-        int64_t need = (int64_t)e->h_pin[S_REQ];
-        rc = grow_caps(e, need + need / 4, e->cap_cols);
-        if (rc) return rc;
-        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                           e->d_state, S_WORDS);
+        // overflow: fetch grew the caps; re-run the whole query
     }
     return WK_ERR_CAP;
 }
