@@ -69,11 +69,92 @@ def cpu_baseline(nuniv_workload, seed):
     }
 
 
+def run_emulator(args, store, engines, inflight):
+    """The reference's sparql-emu benchmark (Proxy::run_query_emu,
+    core/proxy.hpp:391-545): light templates A1-A6 with mix_config
+    weights, %-constants drawn from type-index candidates, `-p`-style
+    in-flight window, blind replies; reports qps + per-type p50/p99."""
+    import random
+    from collections import deque
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+
+    rng = random.Random(7)
+    pools = {t: store.get_index(Q.EMU_POOLS[t], wk.DIR_IN)
+             for t in Q.EMU_WEIGHTS}
+    tnames = list(Q.EMU_WEIGHTS)
+    weights = [Q.EMU_WEIGHTS[t] for t in tnames]
+
+    def gen_query():
+        t = rng.choices(tnames, weights)[0]
+        c = int(rng.choice(pools[t]))
+        p = Q.emu_template(t, c)
+        p.blind = True
+        return t, p
+
+    def harvest(eng_, plan_):
+        n = eng_.fetch_count()
+        while n < 0:
+            eng_.submit(plan_)
+            n = eng_.fetch_count()
+        return n
+
+    def run(n, collect=None):
+        pending = deque()
+        free = list(engines)
+        for _ in range(n):
+            if not free:
+                t_, e_, ti_, p_ = pending.popleft()
+                harvest(e_, p_)
+                if collect is not None:
+                    collect.setdefault(t_, []).append((time.time() - ti_) * 1e3)
+                free.append(e_)
+            t, p = gen_query()
+            e_ = free.pop()
+            ti = time.time()
+            e_.submit(p)
+            pending.append((t, e_, ti, p))
+        while pending:
+            t_, e_, ti_, p_ = pending.popleft()
+            harvest(e_, p_)
+            if collect is not None:
+                collect.setdefault(t_, []).append((time.time() - ti_) * 1e3)
+
+    run(max(args.emu // 10, inflight * 2))  # warmup
+    lat = {}
+    t0 = time.time()
+    run(args.emu, collect=lat)
+    elapsed = time.time() - t0
+    latency = {t: {"p50_ms": round(pctl(xs, 50), 3),
+                   "p99_ms": round(pctl(xs, 99), 3), "n": len(xs)}
+               for t, xs in sorted(lat.items())}
+    out = {
+        "metric": "light-mix queries/sec (emulator A1-A6, mix_config weights)",
+        "value": round(args.emu / elapsed, 1),
+        "unit": "queries/s",
+        "n_gpus": 1,
+        "queries": args.emu,
+        "inflight": inflight,
+        "higher_is_better": True,
+        "dtype": "u32",
+        "data": "synthetic",
+        "config": {"workload": "LUBM-2560 light-template mix, blind replies "
+                               "(proxy.hpp:491), in-flight window "
+                               "(proxy.hpp:477-525)"},
+        "latency": latency,
+    }
+    print(json.dumps(out), flush=True)
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=4)
     ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--emu", type=int, default=0,
+                    help="emulator light-mix mode: run N template queries "
+                         "(A1-A6, mix_config weights) through the in-flight "
+                         "window; reports its own JSON line")
     args = ap.parse_args()
 
     import torch
@@ -109,7 +190,8 @@ def main():
     # heavy queries keep their tables L3-resident; concurrent engines evict
     # each other (measured: inflight=8 halves throughput at LUBM-2560), so
     # the headline run is sequential; WK_INFLIGHT>1 suits light-query mixes
-    inflight = 1 if distributed else int(os.environ.get("WK_INFLIGHT", "1"))
+    inflight = 1 if distributed else int(
+        os.environ.get("WK_INFLIGHT", "16" if args.emu else "1"))
     gstore = wk.GpuStore(store, device=local_rank)
     engines = [wk.Engine(gstore, device=local_rank) for _ in range(inflight)]
     eng = engines[0]
@@ -117,6 +199,10 @@ def main():
         f"{inflight} engine(s) ({time.time()-t0:.1f}s)")
 
     names = list(Q.ALL)
+
+    if args.emu and not distributed:
+        run_emulator(args, store, engines, inflight)
+        return
 
     def harvest(eng_, plan_):
         n = eng_.fetch_count()

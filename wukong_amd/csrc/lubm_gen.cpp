@@ -52,6 +52,8 @@ enum : sid_t {
     T_COURSE = 22,
     T_GRADCOURSE = 23,
     T_RESEARCHGROUP = 24,
+    P_PUBAUTHOR = 25,  // ub:publicationAuthor (emulator template A2)
+    T_PUBLICATION = 26,
 };
 
 struct rng_t {  // splitmix64
@@ -178,6 +180,24 @@ static void gen_university(uint32_t u, uint32_t nuniv, uint64_t seed, emitter &e
             for (int k = 0; k < nc; k++)
                 em.emit(s, P_TAKESCOURSE, gradcourses[rng.range(0, gradcourses.size())]);
             em.emit(s, P_ADVISOR, profs[rng.range(0, profs.size())]);
+        }
+
+        // publications: professors author 4-9 each, half with one grad
+        // co-author (emulator A2 shape: %AssistantProfessor pubs)
+        {
+            // grads are generated above; collect their ids arithmetically is
+            // fragile — emit pubs from the profs only plus rng-chosen grads
+            // via a second pass is unnecessary: keep co-authors professors.
+            for (sid_t f : profs) {
+                int np = rng.range(4, 10);
+                for (int i = 0; i < np; i++) {
+                    sid_t pub = alloc();
+                    em.emit(pub, TYPE_ID, T_PUBLICATION);
+                    em.emit(pub, P_PUBAUTHOR, f);
+                    if (rng.next() & 1)
+                        em.emit(pub, P_PUBAUTHOR, profs[rng.range(0, profs.size())]);
+                }
+            }
         }
 
         // research groups

@@ -16,6 +16,7 @@ SUBORG, UGDEGREE, MEMBEROF, WORKSFOR, TEACHEROF, ADVISOR, TAKESCOURSE = range(2,
 NAME, EMAIL, TELEPHONE, HEADOF, DOCDEGREE = range(9, 14)
 (UNIVERSITY, DEPARTMENT, FULLPROF, ASSOCPROF, ASSTPROF, LECTURER,
  UGSTUDENT, GRADSTUDENT, COURSE, GRADCOURSE, RESEARCHGROUP) = range(14, 25)
+PUBAUTHOR, PUBLICATION = 25, 26
 
 # deterministic generator constants (lubm_gen.cpp allocation order):
 UNIV0 = 1 << 17            # university 0 entity
@@ -87,3 +88,37 @@ Q7 = Plan([
 ALL = {"q1": Q1, "q2": Q2, "q3": Q3, "q4": Q4, "q5": Q5, "q6": Q6, "q7": Q7}
 HEAVY = ["q1", "q2", "q3", "q7"]
 LIGHT = ["q4", "q5", "q6"]
+
+
+def emu_template(name, const):
+    """Light emulator templates A1-A6 instantiated with a candidate
+    constant (scripts/sparql_query/lubm/emulator/q{1..6}; %-placeholders
+    filled from type-index candidates like Proxy::fill_template,
+    core/proxy.hpp:69-129)."""
+    t = {
+        "a1": Plan([(const, TAKESCOURSE, DIR_IN, X),
+                    (X, TYPE_ID, DIR_OUT, GRADSTUDENT)], 1, [X]),
+        "a2": Plan([(const, PUBAUTHOR, DIR_IN, X),
+                    (X, TYPE_ID, DIR_OUT, PUBLICATION)], 1, [X]),
+        "a3": Plan([(const, MEMBEROF, DIR_IN, X),
+                    (X, TYPE_ID, DIR_OUT, GRADSTUDENT)], 1, [X]),
+        "a4": Plan([(const, WORKSFOR, DIR_IN, X),
+                    (X, TYPE_ID, DIR_OUT, FULLPROF),
+                    (X, NAME, DIR_OUT, Y1),
+                    (X, EMAIL, DIR_OUT, Y2),
+                    (X, TELEPHONE, DIR_OUT, Y3)], 4, [X, Y1, Y2, Y3]),
+        "a5": Plan([(const, SUBORG, DIR_IN, X),
+                    (X, TYPE_ID, DIR_OUT, RESEARCHGROUP)], 1, [X]),
+        "a6": Plan([(const, SUBORG, DIR_IN, Y),
+                    (Y, TYPE_ID, DIR_OUT, DEPARTMENT),
+                    (Y, WORKSFOR, DIR_IN, X),
+                    (X, TYPE_ID, DIR_OUT, FULLPROF)], 2, [X, Y]),
+    }
+    return t[name]
+
+
+# candidate pool type per template + mix weights
+# (scripts/sparql_query/lubm/emulator/mix_config: 25/25/3/6/25/2)
+EMU_POOLS = {"a1": GRADCOURSE, "a2": ASSTPROF, "a3": DEPARTMENT,
+             "a4": DEPARTMENT, "a5": DEPARTMENT, "a6": UNIVERSITY}
+EMU_WEIGHTS = {"a1": 25, "a2": 25, "a3": 3, "a4": 6, "a5": 25, "a6": 2}
