@@ -99,9 +99,23 @@ def run_emulator(args, store, engines, inflight):
             n = eng_.fetch_count()
         return n
 
-    def run(n, collect=None):
+    nthreads = int(os.environ.get("WK_EMU_THREADS", "6"))
+    nthreads = max(1, min(nthreads, len(engines)))
+
+    def run_window(engs, n, seed, collect=None):
+        # one submitter's pipeline over its engine subset (ctypes releases
+        # the GIL inside submit/fetch, so threads overlap on the host too)
+        lrng = random.Random(seed)
         pending = deque()
-        free = list(engines)
+        free = list(engs)
+
+        def gen_q():
+            t = lrng.choices(tnames, weights)[0]
+            c = int(lrng.choice(pools[t]))
+            p = Q.emu_template(t, c)
+            p.blind = True
+            return t, p
+
         for _ in range(n):
             if not free:
                 t_, e_, ti_, p_ = pending.popleft()
@@ -109,7 +123,7 @@ def run_emulator(args, store, engines, inflight):
                 if collect is not None:
                     collect.setdefault(t_, []).append((time.time() - ti_) * 1e3)
                 free.append(e_)
-            t, p = gen_query()
+            t, p = gen_q()
             e_ = free.pop()
             ti = time.time()
             e_.submit(p)
@@ -119,6 +133,24 @@ def run_emulator(args, store, engines, inflight):
             harvest(e_, p_)
             if collect is not None:
                 collect.setdefault(t_, []).append((time.time() - ti_) * 1e3)
+
+    def run(n, collect=None):
+        import threading
+        chunks = [engines[i::nthreads] for i in range(nthreads)]
+        per = n // nthreads
+        cols = [dict() for _ in range(nthreads)]
+        ths = [threading.Thread(target=run_window,
+                                args=(chunks[i], per, 100 + i,
+                                      cols[i] if collect is not None else None))
+               for i in range(nthreads)]
+        for t in ths:
+            t.start()
+        for t in ths:
+            t.join()
+        if collect is not None:
+            for c in cols:
+                for k, v in c.items():
+                    collect.setdefault(k, []).extend(v)
 
     run(max(args.emu // 10, inflight * 2))  # warmup
     lat = {}
@@ -135,6 +167,7 @@ def run_emulator(args, store, engines, inflight):
         "n_gpus": 1,
         "queries": args.emu,
         "inflight": inflight,
+        "submit_threads": nthreads,
         "higher_is_better": True,
         "dtype": "u32",
         "data": "synthetic",
