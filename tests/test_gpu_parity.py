@@ -135,3 +135,22 @@ print('OVERFLOW_OK')
     r = subprocess.run([sys.executable, "-c", code], capture_output=True,
                        text=True, timeout=300)
     assert "OVERFLOW_OK" in r.stdout, r.stdout + r.stderr
+
+
+def test_emulator_templates_parity(lubm4, store4, oracle4):
+    """Light emulator templates (A1-A6) through the async submit path —
+    including the single-kernel light-query fast path — vs the oracle."""
+    import random
+    eng = wk.Engine(store4, device=0)
+    rng = random.Random(3)
+    for tname in Q.EMU_WEIGHTS:
+        pool = store4.get_index(Q.EMU_POOLS[tname], wk.DIR_IN)
+        assert pool.size > 0, tname
+        for _ in range(4):
+            const = int(rng.choice(pool))
+            plan = Q.emu_template(tname, const)
+            eng.submit(plan)
+            got = eng.fetch_result()
+            want = oracle4.run_query(plan)
+            assert got.shape == want.shape, (tname, const, got.shape, want.shape)
+            assert np.array_equal(sort_rows(got), sort_rows(want)), (tname, const)

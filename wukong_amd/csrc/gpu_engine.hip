@@ -566,6 +566,45 @@ __global__ void k_dst_scatter(const sid_t *__restrict__ tbl, int64_t nrows,
     }
 }
 
+// whole-query fast path for the dominant light-template shape
+// (const_to_unknown + rdf:type constant filter — emulator A1/A2/A3/A5,
+// proxy.hpp:391-545): ONE single-block kernel executes both patterns and
+// publishes the row count straight to pinned memory.  Cuts ~20 dispatches
+// to 1 (the launch rate, not the kernels, bounds light-query throughput).
+__global__ void k_light2(const sid_t *__restrict__ edges, uint64_t list_off,
+                         uint64_t sz, sid_t cval,
+                         const uint16_t *__restrict__ type_of,
+                         uint64_t type_base, uint64_t type_n,
+                         uint64_t *__restrict__ d_state,
+                         uint64_t *__restrict__ d_stats,
+                         sid_t *__restrict__ out,
+                         uint64_t *__restrict__ h_pin)
+{
+    __shared__ unsigned int cnt;
+    if (threadIdx.x == 0) cnt = 0;
+    __syncthreads();
+    for (uint64_t i = threadIdx.x; i < sz; i += blockDim.x) {
+        sid_t v = edges[list_off + i];
+        uint64_t idx = (uint64_t)v - type_base;
+        uint16_t t = (idx < type_n) ? type_of[idx] : 0;
+        if ((sid_t)t == cval) {
+            unsigned p = atomicAdd(&cnt, 1u);
+            out[p] = v;
+        }
+        // multi-type entities (t==0xFFFF) would need the probe path; the
+        // host only routes here when the store has no such entities
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        d_state[S_NROWS] = cnt;
+        d_state[S_ERR] = 0;
+        for (int i = 0; i < S_WORDS; i++) h_pin[i] = d_state[i];
+        for (int i = 0; i < CAT_COUNT; i++) h_pin[8 + i] = d_stats[i];
+        atomicAdd((unsigned long long *)&d_stats[CAT_FILTER],
+                  (unsigned long long)(sz * 6 + cnt * 4));
+    }
+}
+
 __global__ void k_zero_words(uint64_t *p, int n) {
     for (int i = threadIdx.x; i < n; i += blockDim.x) p[i] = 0;
 }
@@ -1334,9 +1373,44 @@ extern "C" int32_t wk_engine_fetch_raw(wk_engine_t *e, wk_result_t *out) {
 // asynchronous whole-plan submission: enqueue the full launch chain and
 // return without any sync (pipelined multi-engine execution — the
 // reference proxy's in-flight window, proxy.hpp:477-525)
+// true when the whole plan is [c2u, typeof-filter] and the store has a
+// complete single-type index (no 0xFFFF fallbacks anywhere)
+static bool light2_eligible(wk_engine *e, const wk_plan_t *plan) {
+    if (plan->npatterns != 2 || !e->d_type_of || e->st->type_multi) return false;
+    const wk_pattern_t &p0 = plan->patterns[0];
+    const wk_pattern_t &p1 = plan->patterns[1];
+    if (p0.subject < 0 || is_tpid(p0.subject) || p0.object >= 0) return false;
+    if (p1.subject != p0.object) return false;
+    if (p1.predicate != (ssid_t)TYPE_ID || p1.direction != DIR_OUT ||
+        p1.object <= 0)
+        return false;
+    return true;
+}
+
 extern "C" int32_t wk_engine_submit(wk_engine_t *e, const wk_plan_t *plan) {
     int32_t rc = wk_engine_begin_query(e, plan);
     if (rc) return rc;
+    if (light2_eligible(e, plan)) {
+        const wk_pattern_t &p0 = plan->patterns[0];
+        uint64_t sz = 0;
+        const sid_t *ptr = store_get(*e->st, (uint64_t)p0.subject,
+                                     (uint64_t)p0.predicate, p0.direction, &sz);
+        uint64_t off = ptr ? (uint64_t)(ptr - e->st->edges.data()) : 0;
+        rc = grow_caps(e, (int64_t)sz, e->nvars);
+        if (rc) return rc;
+        TIME_BEGIN(e);
+        hipLaunchKernelGGL(k_light2, dim3(1), dim3(BLOCK), 0, e->stream,
+                           e->d_edges, off, sz, (sid_t)plan->patterns[1].object,
+                           e->d_type_of, e->st->type_base, e->st->type_n,
+                           e->d_state, e->d_stats, (sid_t *)e->tbl[1].p, e->h_pin);
+        TIME_END(e, CAT_FILTER);
+        e->cur = 1;
+        e->ncols = 1;
+        e->bound = (int64_t)sz;
+        e->v2c[-(p0.object + 1)] = 0;
+        e->step = 2;
+        return WK_OK;
+    }
     while (e->step < (int)e->pats.size()) {
         rc = exec_pattern(e);
         if (rc) return rc;

@@ -348,10 +348,12 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
         for (int64_t i = pso_lo[TYPE_ID]; i < pso_lo[TYPE_ID + 1]; i++) {
             bool first = (i == pso_lo[TYPE_ID]) || (pso[i - 1].s != pso[i].s);
             bool last = (i + 1 == pso_lo[TYPE_ID + 1]) || (pso[i + 1].s != pso[i].s);
-            if (first && last)
+            if (first && last) {
                 st->type_of[pso[i].s - st->type_base] = (uint16_t)pso[i].o;
-            else
+            } else {
                 st->type_of[pso[i].s - st->type_base] = 0xFFFF;
+                st->type_multi = true;
+            }
         }
     }
     WK_LOG("[store] type index: %.1fs\n", now_s() - t0);

@@ -60,6 +60,7 @@ struct wk_store {
     // from the [vid|TYPE_ID|OUT] lists, bit-identical results.
     pod_array<uint16_t> type_of;
     uint64_t type_base = 0, type_n = 0;
+    bool type_multi = false;  // any entity with >1 type (0xFFFF fallbacks)
     // index segments [dir]: keys [0|pid|dir] / [0|tid|IN]
     wk::seg_t iseg[2];
 
