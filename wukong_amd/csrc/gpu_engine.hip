@@ -678,6 +678,8 @@ struct wk_engine {
     int ncols = 0;
     int64_t bound = 0;   // host-side upper bound on rows (grid sizing)
 
+    bool light = false;  // single-kernel fast path ran (h_pin self-published)
+
     // query state (host mirror of SPARQLQuery, query.hpp:560-594)
     std::vector<wk_pattern_t> pats;
     std::vector<int32_t> v2c;  // idx -> col (query.hpp:352-374)
@@ -891,12 +893,29 @@ extern "C" int32_t wk_engine_begin_query(wk_engine_t *e, const wk_plan_t *plan) 
     e->ncols = 0;
     e->cur = 0;
     e->bound = 0;
+    e->light = false;
     int32_t rc = grow_caps(e, e->cap_rows, plan->nvars);
     if (rc) return rc;
     // reset nrows + overflow flags on device (async, cheap kernel)
     hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream, e->d_state,
                        S_WORDS);
     return WK_OK;
+}
+
+// begin without the device-state zeroing kernel (the light fast path's
+// kernel writes NROWS/ERR itself and publishes to pinned memory)
+static int32_t begin_light(wk_engine *e, const wk_plan_t *plan) {
+    if (!e || !plan || plan->npatterns <= 0 || plan->nvars <= 0) return WK_ERR_PLAN;
+    e->pats.assign(plan->patterns, plan->patterns + plan->npatterns);
+    e->v2c.assign(plan->nvars, -1);
+    e->nvars = plan->nvars;
+    e->step = 0;
+    e->nrows = 0;
+    e->ncols = 0;
+    e->cur = 0;
+    e->bound = 0;
+    e->light = true;
+    return grow_caps(e, e->cap_rows, plan->nvars);
 }
 
 static int32_t load_common(wk_engine *e, int64_t nrows, int32_t ncols,
@@ -940,8 +959,9 @@ extern "C" int32_t wk_engine_load_rbuf_device(wk_engine_t *e, const sid_t *dev_t
 
 // publish device state to pinned memory + sync; refresh host nrows/stats
 static int32_t sync_state(wk_engine *e) {
-    hipLaunchKernelGGL(k_publish_state, dim3(1), dim3(1), 0, e->stream,
-                       e->d_state, e->d_stats, e->h_pin);
+    if (!e->light)
+        hipLaunchKernelGGL(k_publish_state, dim3(1), dim3(1), 0, e->stream,
+                           e->d_state, e->d_stats, e->h_pin);
     HIP_CHECK(stream_sync(e->stream));
     resolve_timing(e);
     e->nrows = (int64_t)e->h_pin[S_NROWS];
@@ -1388,9 +1408,11 @@ static bool light2_eligible(wk_engine *e, const wk_plan_t *plan) {
 }
 
 extern "C" int32_t wk_engine_submit(wk_engine_t *e, const wk_plan_t *plan) {
-    int32_t rc = wk_engine_begin_query(e, plan);
+    if (!e || !plan) return WK_ERR_STATE;
+    const bool light = light2_eligible(e, plan);
+    int32_t rc = light ? begin_light(e, plan) : wk_engine_begin_query(e, plan);
     if (rc) return rc;
-    if (light2_eligible(e, plan)) {
+    if (light) {
         const wk_pattern_t &p0 = plan->patterns[0];
         uint64_t sz = 0;
         const sid_t *ptr = store_get(*e->st, (uint64_t)p0.subject,
