@@ -184,6 +184,10 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=4)
     ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--watdiv", type=int, default=0,
+                    help="WatDiv mode: generate N products (~55N triples) and "
+                         "run the star/linear/snowflake templates instead of "
+                         "the LUBM suite")
     ap.add_argument("--emu", type=int, default=0,
                     help="emulator light-mix mode: run N template queries "
                          "(A1-A6, mix_config weights) through the in-flight "
@@ -203,6 +207,8 @@ def main():
     seed = 42
     nuniv = int(os.environ.get("WK_UNIV", "2560"))
     dataset = f"LUBM-{nuniv}"
+    if args.watdiv:
+        dataset = f"WatDiv-{args.watdiv}p"
 
     if distributed:
         import torch.distributed as dist
@@ -211,7 +217,10 @@ def main():
         from wukong_amd.dist import DistQuery, GpuExecutor
 
     t0 = time.time()
-    triples = wk.lubm_gen(nuniv, seed=seed, sid=rank, nsrv=world)
+    if args.watdiv:
+        triples = wk.watdiv_gen(args.watdiv, seed=seed, sid=rank, nsrv=world)
+    else:
+        triples = wk.lubm_gen(nuniv, seed=seed, sid=rank, nsrv=world)
     log(f"[rank {rank}] gen {dataset}: {triples.shape[0]} triples "
         f"({time.time()-t0:.1f}s)")
     t0 = time.time()
@@ -231,7 +240,12 @@ def main():
     log(f"[rank {rank}] HBM upload {((store.num_slots*16+store.num_edges*4)/1e9):.2f} GB, "
         f"{inflight} engine(s) ({time.time()-t0:.1f}s)")
 
-    names = list(Q.ALL)
+    if args.watdiv:
+        from wukong_amd import watdiv as W
+        Q_ALL = W.ALL
+    else:
+        Q_ALL = Q.ALL
+    names = list(Q_ALL)
 
     if args.emu and not distributed:
         run_emulator(args, store, engines, inflight)
@@ -248,7 +262,7 @@ def main():
         """Distributed path / single-engine fallback: sequential queries."""
         for _ in range(passes):
             for name in names:
-                plan = Q.ALL[name]
+                plan = Q_ALL[name]
                 tq = time.time()
                 if distributed:
                     ex = GpuExecutor(eng, plan)
@@ -286,7 +300,7 @@ def main():
             if not free:
                 drain_one()
             eng_ = free.pop()
-            plan = Q.ALL[name]
+            plan = Q_ALL[name]
             t_iss = time.time()
             eng_.submit(plan)
             pending.append((name, eng_, t_iss, plan))
@@ -347,6 +361,8 @@ def main():
     def roofline_probe():
         if distributed:
             return {}, {}
+        if args.watdiv:
+            return {}, {}
         e0 = engines[0]
         e0.begin_query(Q.ALL["q1"])
         e0.execute_one_pattern()               # i2u
@@ -392,7 +408,8 @@ def main():
     }
 
     cb = None
-    if os.environ.get("WK_SKIP_CPU_BASELINE") != "1" and not distributed:
+    if (os.environ.get("WK_SKIP_CPU_BASELINE") != "1" and not distributed
+            and not args.watdiv):
         log("[rank 0] timing cpu_baseline (oracle engine)...")
         cb = cpu_baseline(nuniv, seed)
 
@@ -404,7 +421,9 @@ def main():
                          "rows": lat.get("_rows", {}).get(name)}
 
     out = {
-        "metric": "queries/sec, LUBM-2560 Q1-Q7 mix (OSDI16 plans)",
+        "metric": ("queries/sec, WatDiv star/linear/snowflake templates"
+                   if args.watdiv else
+                   "queries/sec, LUBM-2560 Q1-Q7 mix (OSDI16 plans)"),
         "value": round(qps, 3),
         "unit": "queries/s",
         "n_gpus": ngpus,

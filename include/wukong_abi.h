@@ -70,6 +70,9 @@ typedef struct {
  * Caller frees with wk_free_triples. */
 int64_t wk_lubm_gen(int32_t nuniv, uint64_t seed, int32_t sid, int32_t nsrv,
                     wk_sid_t **out_spo /* 3*n entries */);
+/* WatDiv-shaped generator (configs[3]); ~55 triples per product. */
+int64_t wk_watdiv_gen(int64_t nproducts, uint64_t seed, int32_t sid,
+                      int32_t nsrv, wk_sid_t **out_spo);
 void    wk_free_triples(wk_sid_t *spo);
 
 /* ---------- store ---------------------------------------------- */

@@ -56,6 +56,7 @@ c_u64 = ctypes.c_uint64
 c_vp = ctypes.c_void_p
 
 _lubm_gen = _sig("wk_lubm_gen", c_i64, [c_i32, c_u64, c_i32, c_i32, ctypes.POINTER(c_u32p)])
+_watdiv_gen = _sig("wk_watdiv_gen", c_i64, [c_i64, c_u64, c_i32, c_i32, ctypes.POINTER(c_u32p)])
 _free_triples = _sig("wk_free_triples", None, [c_u32p])
 _store_build = _sig("wk_store_build", c_vp, [c_u32p, c_i64, c_i32, c_i32])
 _store_free = _sig("wk_store_free", None, [c_vp])
@@ -130,6 +131,17 @@ def lubm_gen(nuniv, seed=42, sid=0, nsrv=1):
     n = _lubm_gen(nuniv, seed, sid, nsrv, ctypes.byref(out))
     if n < 0:
         raise RuntimeError("wk_lubm_gen failed")
+    arr = np.ctypeslib.as_array(out, shape=(n, 3)).copy()
+    _free_triples(out)
+    return arr
+
+
+def watdiv_gen(nproducts, seed=42, sid=0, nsrv=1):
+    """Seeded WatDiv-shaped synthetic ID-triples (~55 triples/product)."""
+    out = c_u32p()
+    n = _watdiv_gen(nproducts, seed, sid, nsrv, ctypes.byref(out))
+    if n < 0:
+        raise RuntimeError("wk_watdiv_gen failed")
     arr = np.ctypeslib.as_array(out, shape=(n, 3)).copy()
     _free_triples(out)
     return arr

@@ -10,7 +10,7 @@ import sys
 
 HERE = os.path.dirname(os.path.abspath(__file__))
 SRC = [os.path.join(HERE, "csrc", f)
-       for f in ("lubm_gen.cpp", "store.cpp", "gpu_engine.hip")]
+       for f in ("lubm_gen.cpp", "watdiv_gen.cpp", "store.cpp", "gpu_engine.hip")]
 OUT = os.path.join(HERE, "libwukong_hip.so")
 HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
 
