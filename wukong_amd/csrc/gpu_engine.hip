@@ -439,6 +439,7 @@ __global__ void k_publish_state(const uint64_t *__restrict__ d_state,
 // output-centric version paid 2-3 DRAM lines of prefix walk per output
 // row).  Rows with deg > 32 go to an overflow queue handled by
 // k_expand_big with one WAVE per row (lanes stride the edge list).
+template <int NC>
 __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
                             const sid_t *__restrict__ edges,
                             const uint64_t *__restrict__ d_eoff,
@@ -452,7 +453,8 @@ __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
 {
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     const int64_t chunk = (nrows + G - 1) / G;
-    const int oc = ncols + 1;
+    constexpr int oc = NC + 1;
+    (void)ncols;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
          r += (int64_t)gridDim.x * blockDim.x) {
         uint32_t deg = d_cnt[r];
@@ -466,13 +468,15 @@ __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
             continue;
         }
         if (basep + deg > cap) deg = (uint32_t)(cap - basep);
-        sid_t row[8];
-        for (int c = 0; c < ncols; c++) row[c] = tbl[r * ncols + c];
+        sid_t row[NC];
+#pragma unroll
+        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
         const sid_t *el = edges + d_eoff[r];
         sid_t *dst = out + (int64_t)basep * oc;
         for (uint32_t k = 0; k < deg; k++) {
-            for (int c = 0; c < ncols; c++) dst[c] = row[c];
-            dst[ncols] = el[k];
+#pragma unroll
+            for (int c = 0; c < NC; c++) dst[c] = row[c];
+            dst[NC] = el[k];
             dst += oc;
         }
     }
@@ -487,6 +491,7 @@ __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
 
 // big-fanout rows: one wave per queued row, lanes stride the edge list
 // (coalesced writes: adjacent lanes write adjacent output rows)
+template <int NC>
 __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
                              const sid_t *__restrict__ edges,
                              const uint64_t *__restrict__ d_eoff,
@@ -501,7 +506,8 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
     if (!nq) return;
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     const int64_t chunk = (nrows + G - 1) / G;
-    const int oc = ncols + 1;
+    constexpr int oc = NC + 1;
+    (void)ncols;
     const int lane = threadIdx.x & 63;
     const int64_t w0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
     const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
@@ -511,13 +517,15 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
         uint64_t basep = d_pre[r] + bsums[r / chunk];
         if (basep >= cap) continue;
         if (basep + deg > cap) deg = cap - basep;
-        sid_t row[8];
-        for (int c = 0; c < ncols; c++) row[c] = tbl[r * ncols + c];
+        sid_t row[NC];
+#pragma unroll
+        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
         const sid_t *el = edges + d_eoff[r];
         for (uint64_t k = lane; k < deg; k += 64) {
             sid_t *dst = out + (int64_t)(basep + k) * oc;
-            for (int c = 0; c < ncols; c++) dst[c] = row[c];
-            dst[ncols] = el[k];
+#pragma unroll
+            for (int c = 0; c < NC; c++) dst[c] = row[c];
+            dst[NC] = el[k];
         }
     }
 }
@@ -885,6 +893,7 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
 
 extern "C" int32_t wk_engine_begin_query(wk_engine_t *e, const wk_plan_t *plan) {
     if (!e || !plan || plan->npatterns <= 0 || plan->nvars <= 0) return WK_ERR_PLAN;
+    if (plan->nvars > 8) return WK_ERR_PLAN;  // col count cap (templated kernels)
     e->pats.assign(plan->patterns, plan->patterns + plan->npatterns);
     e->v2c.assign(plan->nvars, -1);
     e->nvars = plan->nvars;
@@ -983,6 +992,39 @@ static int32_t sync_state_grow(wk_engine *e) {
     hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
                        e->d_state, S_WORDS);
     return WK_ERR_CAP;
+}
+
+// expansion dispatch, specialised on the input column count so the row
+// copy unrolls/vectorises (runtime-indexed local arrays spill to scratch
+// — cdna_hip_programming.md §5.4 rule 20)
+template <int NC>
+static void launch_expand_t(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
+                            int G) {
+    hipLaunchKernelGGL(k_expand_in<NC>, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
+                       e->stream, cur_tbl, e->ncols, e->d_edges,
+                       (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p,
+                       (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
+                       e->d_state, (uint64_t)e->cap_rows, e->d_stats,
+                       (uint32_t *)e->ovf.p, out_tbl);
+    hipLaunchKernelGGL(k_expand_big<NC>, dim3(512), dim3(BLOCK), 0, e->stream,
+                       cur_tbl, e->ncols, e->d_edges, (uint64_t *)e->eoff.p,
+                       (uint32_t *)e->cnt.p, (uint64_t *)e->prefix.p,
+                       (uint64_t *)e->bsums.p, G, e->d_state,
+                       (uint64_t)e->cap_rows, (uint32_t *)e->ovf.p, out_tbl);
+}
+
+static void launch_expand(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
+                          int G) {
+    switch (e->ncols) {
+    case 1: launch_expand_t<1>(e, cur_tbl, out_tbl, G); break;
+    case 2: launch_expand_t<2>(e, cur_tbl, out_tbl, G); break;
+    case 3: launch_expand_t<3>(e, cur_tbl, out_tbl, G); break;
+    case 4: launch_expand_t<4>(e, cur_tbl, out_tbl, G); break;
+    case 5: launch_expand_t<5>(e, cur_tbl, out_tbl, G); break;
+    case 6: launch_expand_t<6>(e, cur_tbl, out_tbl, G); break;
+    case 7: launch_expand_t<7>(e, cur_tbl, out_tbl, G); break;
+    default: launch_expand_t<8>(e, cur_tbl, out_tbl, G); break;
+    }
 }
 
 // Run one pattern — dispatch per sparql.hpp:1016-1058.  Fully async: row
@@ -1119,20 +1161,11 @@ static int32_t exec_pattern(wk_engine *e) {
         hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
                            e->d_state + S_OVF, 1);
         {
+        {
             TIME_BEGIN(e);
-            hipLaunchKernelGGL(k_expand_in, dim3(grid_for(e->bound)), dim3(BLOCK),
-                               0, e->stream, cur_tbl, e->ncols, e->d_edges,
-                               (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p,
-                               (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
-                               e->d_state, (uint64_t)e->cap_rows, e->d_stats,
-                               (uint32_t *)e->ovf.p, out_tbl);
-            hipLaunchKernelGGL(k_expand_big, dim3(512), dim3(BLOCK), 0, e->stream,
-                               cur_tbl, e->ncols, e->d_edges,
-                               (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p,
-                               (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
-                               e->d_state, (uint64_t)e->cap_rows,
-                               (uint32_t *)e->ovf.p, out_tbl);
+            launch_expand(e, cur_tbl, out_tbl, G);
             TIME_END(e, CAT_EXPAND);
+        }
         }
         e->v2c[-(o + 1)] = e->ncols;
         e->ncols = oc;
