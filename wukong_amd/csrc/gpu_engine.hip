@@ -140,6 +140,26 @@ __device__ __forceinline__ void probe_one(const vertex_t *__restrict__ verts,
     }
 }
 
+// continue a probe from a chained bucket (absolute bucket id)
+__device__ __forceinline__ void probe_chain(const vertex_t *__restrict__ verts,
+                                            uint64_t bucket, uint64_t key,
+                                            uint64_t &eoff, uint64_t &esz) {
+    while (true) {
+        const vertex_t *b = &verts[bucket * ASSOC];
+        int hit = -1;
+#pragma unroll
+        for (int i = 0; i < ASSOC - 1; i++)
+            if (b[i].key == key && hit < 0) hit = i;
+        if (hit >= 0) {
+            eoff = ptr_off(b[hit].ptr);
+            esz = ptr_size(b[hit].ptr);
+            return;
+        }
+        if (b[ASSOC - 1].key == KEY_EMPTY) { eoff = 0; esz = 0; return; }
+        bucket = key_vid(b[ASSOC - 1].key);
+    }
+}
+
 // Fused known_to_unknown front half: contiguous-chunk thread-per-row
 // probe + in-kernel exclusive prefix of the edge counts (local prefix +
 // per-block sums; k_scan_mid finishes across blocks).  64 rows in
@@ -163,15 +183,69 @@ __global__ void k_probe_scan(const vertex_t *__restrict__ verts,
     const int64_t end = min(start + chunk, nrows);
     __shared__ uint64_t sh[SCAN_T];
     uint64_t carry = 0;
+
+    // 2-deep pipeline: the NEXT tile row's first bucket is issued before
+    // this tile's scan phase, so its ~900-cycle HBM latency hides under
+    // the LDS scan + barriers (loads to registers stay in flight across
+    // s_barrier; hipcc waits at first use)
+    auto key_of = [&](int64_t r) {
+        sid_t v = tbl[r * ncols + col];
+        return (key_mode == PK_NORMAL) ? key_pack(v, pid, (uint64_t)dir)
+                                       : key_pack(0, v, (uint64_t)dir);
+    };
+    uint64_t nkey = 0, nbucket = 0;
+    uint64_t pk[ASSOC];
+    uint64_t pp[ASSOC - 1];
+    bool have_pref = false;
+    if (start + threadIdx.x < end) {
+        nkey = key_of(start + threadIdx.x);
+        nbucket = bucket_start + hash_u64(nkey) % num_buckets;
+        const vertex_t *b = &verts[nbucket * ASSOC];
+#pragma unroll
+        for (int i = 0; i < ASSOC; i++) pk[i] = b[i].key;
+#pragma unroll
+        for (int i = 0; i < ASSOC - 1; i++) pp[i] = b[i].ptr;
+        have_pref = true;
+    }
     for (int64_t base = start; base < end; base += SCAN_T) {
         const int64_t r = base + threadIdx.x;
         uint64_t eoff = 0, esz = 0;
-        if (r < end) {
-            sid_t v = tbl[r * ncols + col];
-            uint64_t key = (key_mode == PK_NORMAL)
-                               ? key_pack(v, pid, (uint64_t)dir)
-                               : key_pack(0, v, (uint64_t)dir);
-            probe_one(verts, bucket_start, num_buckets, key, eoff, esz);
+        uint64_t key = nkey;
+        uint64_t bucket = nbucket;
+        uint64_t ck[ASSOC], cp[ASSOC - 1];
+        bool have = have_pref;
+        if (have) {
+#pragma unroll
+            for (int i = 0; i < ASSOC; i++) ck[i] = pk[i];
+#pragma unroll
+            for (int i = 0; i < ASSOC - 1; i++) cp[i] = pp[i];
+        }
+        // issue next tile's first bucket
+        const int64_t rn = base + SCAN_T + threadIdx.x;
+        have_pref = false;
+        if (rn < end) {
+            nkey = key_of(rn);
+            nbucket = bucket_start + hash_u64(nkey) % num_buckets;
+            const vertex_t *b = &verts[nbucket * ASSOC];
+#pragma unroll
+            for (int i = 0; i < ASSOC; i++) pk[i] = b[i].key;
+#pragma unroll
+            for (int i = 0; i < ASSOC - 1; i++) pp[i] = b[i].ptr;
+            have_pref = true;
+        }
+        if (have) {
+            // resolve the prefetched bucket; rare chains fall back to the walk
+            int hit = -1;
+#pragma unroll
+            for (int i = 0; i < ASSOC - 1; i++)
+                if (ck[i] == key && hit < 0) hit = i;
+            if (hit >= 0) {
+                eoff = ptr_off(cp[hit]);
+                esz = ptr_size(cp[hit]);
+            } else if (ck[ASSOC - 1] != KEY_EMPTY) {
+                // rare (~13% of buckets): continue down the chain
+                probe_chain(verts, key_vid(ck[ASSOC - 1]), key, eoff, esz);
+            }
             d_eoff[r] = eoff;
             d_cnt[r] = (uint32_t)esz;
         }
