@@ -290,18 +290,43 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
     __shared__ uint32_t wtot[4];
     const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
     const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    // typeof pipeline: issue the next tile's 2-byte type read before this
+    // tile's compact phase so its HBM latency hides under the barriers
+    uint16_t nt = 0;
+    sid_t nv = 0;
+    bool have_pref = false;
+    if (use_typeof) {
+        const int64_t r0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+        if (r0 < nrows) {
+            nv = tbl[r0 * ncols + col];
+            uint64_t idx = (uint64_t)nv - type_base;
+            nt = (idx < type_n) ? type_of[idx] : 0;
+            have_pref = true;
+        }
+    }
     for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < nrows;
          base += stride) {
         const int64_t r = base + threadIdx.x;
         bool keep = false;
         bool need_probe = false;
+        uint16_t ct = nt;
+        bool chave = have_pref;
+        if (use_typeof) {
+            const int64_t rn = base + stride + threadIdx.x;
+            have_pref = false;
+            if (rn < nrows) {
+                nv = tbl[rn * ncols + col];
+                uint64_t idx = (uint64_t)nv - type_base;
+                nt = (idx < type_n) ? type_of[idx] : 0;
+                have_pref = true;
+            }
+        }
         if (r < nrows) {
             sid_t v = tbl[r * ncols + col];
             if (use_typeof) {
                 // dense type check: `?X rdf:type T` (the hottest LUBM
                 // filter) is one 2-byte read; 0xFFFF falls back to probe
-                uint64_t idx = (uint64_t)v - type_base;
-                uint16_t t = (idx < type_n) ? type_of[idx] : 0;
+                uint16_t t = chave ? ct : 0;
                 if (t == 0xFFFF) need_probe = true;
                 else keep = ((sid_t)t == cval);
             }
