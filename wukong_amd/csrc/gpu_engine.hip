@@ -264,10 +264,11 @@ __global__ void k_probe_scan(const vertex_t *__restrict__ verts,
     if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
 }
 
-// Filter operators (k2c/k2k/c2k/i2k): probe + membership test + BLOCK-
-// AGGREGATED atomic compaction (one global atomicAdd per 256 rows — a
-// single per-row counter serializes at ~88 adds/us, MI355X_MICROARCH.md
-// row `dequeue`).  Row order is engine-internal; parity is set-level.
+// Filter operators (k2c/k2k/c2k/i2k): probe + membership test + block-
+// aggregated compaction.  Each thread handles 4 rows per tile (4 loads in
+// flight, and 4x fewer global atomics: a single counter saturates at
+// ~88 adds/us — MI355X_MICROARCH.md row `dequeue`).  Row order is
+// engine-internal; parity is set-level (sparql.hpp:455-476 semantics).
 __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                              const sid_t *__restrict__ edges,
                              uint64_t bucket_start, uint64_t num_buckets,
@@ -284,105 +285,74 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
 {
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     count_bytes(d_stats, CAT_FILTER,
-                (uint64_t)nrows * (use_typeof ? 6
-                                  : probe_mode == PM_LIST ? 12 : (4 + 128 + 8 + 64)));
+                (uint64_t)nrows * ((use_typeof ? 6
+                                    : probe_mode == PM_LIST ? 12
+                                                            : (4 + 128 + 8 + 64)) +
+                                   8 * ncols));
+    constexpr int K = 4;
     __shared__ unsigned long long s_base;
-    __shared__ uint32_t wtot[4];
-    const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
-    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
-    // typeof pipeline: issue the next tile's 2-byte type read before this
-    // tile's compact phase so its HBM latency hides under the barriers
-    uint16_t nt = 0;
-    sid_t nv = 0;
-    bool have_pref = false;
-    if (use_typeof) {
-        const int64_t r0 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-        if (r0 < nrows) {
-            nv = tbl[r0 * ncols + col];
-            uint64_t idx = (uint64_t)nv - type_base;
-            nt = (idx < type_n) ? type_of[idx] : 0;
-            have_pref = true;
-        }
-    }
-    for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < nrows;
-         base += stride) {
-        const int64_t r = base + threadIdx.x;
-        bool keep = false;
-        bool need_probe = false;
-        uint16_t ct = nt;
-        bool chave = have_pref;
-        if (use_typeof) {
-            const int64_t rn = base + stride + threadIdx.x;
-            have_pref = false;
-            if (rn < nrows) {
-                nv = tbl[rn * ncols + col];
-                uint64_t idx = (uint64_t)nv - type_base;
-                nt = (idx < type_n) ? type_of[idx] : 0;
-                have_pref = true;
-            }
-        }
-        if (r < nrows) {
+    __shared__ uint32_t sh[SCAN_T];
+    const int64_t tile = (int64_t)blockDim.x * K;
+    const int64_t stride = (int64_t)gridDim.x * tile;
+
+    for (int64_t base = (int64_t)blockIdx.x * tile; base < nrows; base += stride) {
+        bool keep[K];
+        int64_t rr[K];
+        uint32_t cnt = 0;
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            // row layout: thread t owns rows base + t*K + k (contiguous per
+            // thread, so kept rows stay grouped and writes coalesce-ish)
+            const int64_t r = base + (int64_t)threadIdx.x * K + k;
+            rr[k] = r;
+            keep[k] = false;
+            if (r >= nrows) continue;
             sid_t v = tbl[r * ncols + col];
             if (use_typeof) {
-                // dense type check: `?X rdf:type T` (the hottest LUBM
-                // filter) is one 2-byte read; 0xFFFF falls back to probe
-                uint16_t t = chave ? ct : 0;
-                if (t == 0xFFFF) need_probe = true;
-                else keep = ((sid_t)t == cval);
+                uint64_t idx = (uint64_t)v - type_base;
+                uint16_t t = (idx < type_n) ? type_of[idx] : 0;
+                if (t != 0xFFFF) { keep[k] = ((sid_t)t == cval); continue; }
             }
-            if (probe_mode == PM_LIST) {
-                keep = bsearch_u32(edges + list_off, list_sz, v);
-            } else if (!use_typeof || need_probe) {
+            if (probe_mode == PM_LIST && !use_typeof) {
+                keep[k] = bsearch_u32(edges + list_off, list_sz, v);
+            } else {
                 uint64_t key = (key_mode == PK_NORMAL)
                                    ? key_pack(v, pid, (uint64_t)dir)
                                    : key_pack(0, v, (uint64_t)dir);
-                uint64_t bucket = bucket_start + hash_u64(key) % num_buckets;
                 uint64_t eoff = 0, esz = 0;
-                while (true) {
-                    const vertex_t *b = &verts[bucket * ASSOC];
-                    uint64_t k0 = b[0].key, k1 = b[1].key, k2 = b[2].key,
-                             k3 = b[3].key, k4 = b[4].key, k5 = b[5].key,
-                             k6 = b[6].key, k7 = b[7].key;
-                    int hit = -1;
-                    if (k0 == key) hit = 0;
-                    else if (k1 == key) hit = 1;
-                    else if (k2 == key) hit = 2;
-                    else if (k3 == key) hit = 3;
-                    else if (k4 == key) hit = 4;
-                    else if (k5 == key) hit = 5;
-                    else if (k6 == key) hit = 6;
-                    if (hit >= 0) {
-                        uint64_t pp = b[hit].ptr;
-                        eoff = ptr_off(pp); esz = ptr_size(pp);
-                        break;
-                    }
-                    if (k7 == KEY_EMPTY) { esz = 0; break; }
-                    bucket = key_vid(k7);
-                }
-                sid_t tgt = (probe_mode == PM_CONST) ? cval : tbl[r * ncols + col2];
-                keep = esz && bsearch_u32(edges + eoff, esz, tgt);
+                probe_one(verts, bucket_start, num_buckets, key, eoff, esz);
+                sid_t tgt = (probe_mode == PM_CONST) ? cval
+                                                     : tbl[r * ncols + col2];
+                keep[k] = esz && bsearch_u32(edges + eoff, esz, tgt);
             }
         }
-        // block-wide compaction: wave ballot -> per-wave totals -> one
-        // atomic for the block -> in-block exclusive offsets
-        uint64_t wmask = __ballot(keep);
-        if (lane == 0) wtot[wid] = (uint32_t)__popcll(wmask);
+#pragma unroll
+        for (int k = 0; k < K; k++) cnt += keep[k] ? 1u : 0u;
+        // block exclusive scan of per-thread counts
+        sh[threadIdx.x] = cnt;
         __syncthreads();
-        if (threadIdx.x == 0) {
-            uint32_t t = wtot[0] + wtot[1] + wtot[2] + wtot[3];
-            s_base = t ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
-                                   (unsigned long long)t)
-                       : 0;
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint32_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += x;
+            __syncthreads();
         }
+        const uint32_t my_end = sh[threadIdx.x];
+        const uint32_t block_total = sh[SCAN_T - 1];
+        if (threadIdx.x == SCAN_T - 1)
+            s_base = block_total
+                         ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
+                                     (unsigned long long)block_total)
+                         : 0;
         __syncthreads();
-        if (keep) {
-            uint32_t woff = 0;
-            for (int k = 0; k < wid; k++) woff += wtot[k];
-            uint64_t pos = (uint64_t)s_base + woff +
-                           __popcll(wmask & ((1ull << lane) - 1));
+        uint64_t pos = s_base + my_end - cnt;
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            if (!keep[k]) continue;
             sid_t *dst = out_tbl + (int64_t)pos * ncols;
-            const sid_t *src = tbl + r * ncols;
+            const sid_t *src = tbl + rr[k] * ncols;
             for (int c = 0; c < ncols; c++) dst[c] = src[c];
+            pos++;
         }
         __syncthreads();
     }
