@@ -15,9 +15,13 @@
  * reference's hard 32 MB rbuf assert, gpu_engine_cuda.hpp:185).
  *
  * Kernel inventory:
- *   k_probe (8-lane coop) / k_probe_tpr (thread-per-row, default)
- *                     — fused gen-keys + cluster-hash probe + per-mode
- *                       epilogue; replaces gpu_hash.cu:94-132,149-445
+ *   k_probe_scan      — fused gen-keys + thread-per-row cluster-hash
+ *                       probe + in-kernel prefix scan (2-deep bucket
+ *                       prefetch); replaces gpu_hash.cu:94-324
+ *   k_filter_tpr      — fused probe/typeof filter + block compaction
+ *                       (replaces gpu_hash.cu:326-445,523-585)
+ *   k_light2          — whole-query single-kernel path for 2-pattern
+ *                       light templates (emulator A1/A2/A3/A5)
  *   k_scan_part/mid/add — 3-phase exclusive scan with DEVICE-side length
  *                       (replaces the thrust scans, gpu_hash.cu:587-596)
  *   k_expand          — output-centric load-balanced expansion
@@ -355,86 +359,6 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
             pos++;
         }
         __syncthreads();
-    }
-}
-
-// Cooperative 8-lane-per-row probe (WK_PROBE=coop, kept for A/B): lanes
-// load one 128-B bucket coalesced and ballot-match.
-__global__ void k_probe(const vertex_t *__restrict__ verts,
-                        const sid_t *__restrict__ edges,
-                        uint64_t bucket_start, uint64_t num_buckets,
-                        const sid_t *__restrict__ tbl, int ncols,
-                        int col, uint32_t pid, int dir, int key_mode,
-                        int probe_mode, int col2, sid_t cval,
-                        uint64_t list_off, uint64_t list_sz,
-                        uint64_t *__restrict__ d_state,
-                        uint64_t *__restrict__ d_stats,
-                        uint64_t *__restrict__ d_eoff,
-                        uint64_t *__restrict__ d_cnt,
-                        sid_t *__restrict__ out_tbl)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, probe_mode == PM_LIST ? CAT_FILTER : CAT_PROBE,
-                (uint64_t)nrows * (probe_mode == PM_LIST
-                                       ? 12
-                                       : (4 + 128 + 8 + (probe_mode != PM_SIZE ? 64 : 0))));
-    const int sub = threadIdx.x & 7;
-    const int64_t grp0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 3;
-    const int64_t ngrp = ((int64_t)gridDim.x * blockDim.x) >> 3;
-
-    for (int64_t r = grp0; r < nrows; r += ngrp) {
-        uint64_t key;
-        if (probe_mode == PM_LIST) {
-            if (sub == 0) {
-                sid_t v = tbl[r * ncols + col];
-                if (bsearch_u32(edges + list_off, list_sz, v)) {
-                    unsigned long long pos = atomicAdd(
-                        (unsigned long long *)&d_state[S_TOTAL], 1ull);
-                    sid_t *dst = out_tbl + (int64_t)pos * ncols;
-                    const sid_t *src = tbl + r * ncols;
-                    for (int c = 0; c < ncols; c++) dst[c] = src[c];
-                }
-            }
-            continue;
-        }
-        {
-            sid_t v = tbl[r * ncols + col];
-            key = (key_mode == PK_NORMAL) ? key_pack(v, pid, (uint64_t)dir)
-                                          : key_pack(0, v, (uint64_t)dir);
-        }
-        uint64_t bucket = bucket_start + hash_u64(key) % num_buckets;
-        uint64_t eoff = 0, esz = 0;
-        while (true) {
-            const vertex_t slot = verts[bucket * ASSOC + sub];
-            bool match = (sub < ASSOC - 1) && (slot.key == key);
-            uint64_t ball = __ballot(match);
-            uint32_t gmask = (uint32_t)((ball >> ((threadIdx.x & 63) & ~7)) & 0xffu);
-            if (gmask) {
-                int src = (threadIdx.x & ~7 & 63) + (__ffs(gmask) - 1);
-                uint64_t p2 = __shfl((unsigned long long)slot.ptr, src);
-                eoff = ptr_off(p2); esz = ptr_size(p2);
-                break;
-            }
-            uint64_t chain = __shfl((unsigned long long)slot.key,
-                                    (threadIdx.x & ~7 & 63) + (ASSOC - 1));
-            if (chain == KEY_EMPTY) { esz = 0; break; }
-            bucket = key_vid(chain);
-        }
-        if (sub == 0) {
-            if (probe_mode == PM_SIZE) {
-                d_eoff[r] = eoff;
-                d_cnt[r] = esz;
-            } else {
-                sid_t tgt = (probe_mode == PM_CONST) ? cval : tbl[r * ncols + col2];
-                if (esz && bsearch_u32(edges + eoff, esz, tgt)) {
-                    unsigned long long pos = atomicAdd(
-                        (unsigned long long *)&d_state[S_TOTAL], 1ull);
-                    sid_t *dst = out_tbl + (int64_t)pos * ncols;
-                    const sid_t *src = tbl + r * ncols;
-                    for (int c = 0; c < ncols; c++) dst[c] = src[c];
-                }
-            }
-        }
     }
 }
 
