@@ -523,6 +523,163 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
     }
 }
 
+// Fused known_to_unknown + rdf:type constant filter on the NEW column
+// (plan pairs like Q1's ugDegreeFrom -> "?Y type University"): expansion
+// emits only passing rows, block-compacted — saves the follow-up
+// filter's full table read+write pass.  Multi-type entities (0xFFFF)
+// fall back to a probe of [val|TYPE_ID|OUT].
+template <int NC>
+__global__ void k_expand_filter(const sid_t *__restrict__ tbl, int ncols,
+                                const sid_t *__restrict__ edges,
+                                const uint64_t *__restrict__ d_eoff,
+                                const uint32_t *__restrict__ d_cnt,
+                                uint64_t *__restrict__ d_state, uint64_t cap,
+                                uint64_t *__restrict__ d_stats,
+                                uint32_t *__restrict__ ovf,
+                                sid_t fcval,
+                                const uint16_t *__restrict__ type_of,
+                                uint64_t type_base, uint64_t type_n,
+                                const vertex_t *__restrict__ verts,
+                                uint64_t f_bstart, uint64_t f_nbuckets,
+                                sid_t *__restrict__ out)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    constexpr int oc = NC + 1;
+    (void)ncols;
+    count_bytes(d_stats, CAT_EXPAND,
+                (uint64_t)nrows * (4 * NC + 20) + d_state[S_TOTAL] * 6);
+    __shared__ unsigned long long s_base;
+    __shared__ uint32_t sh[SCAN_T];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    auto pass = [&](sid_t val) {
+        uint64_t idx = (uint64_t)val - type_base;
+        uint16_t t = (idx < type_n) ? type_of[idx] : 0;
+        if (t != 0xFFFF) return (sid_t)t == fcval;
+        uint64_t eo = 0, es = 0;
+        probe_one(verts, f_bstart, f_nbuckets,
+                  key_pack(val, TYPE_ID, (uint64_t)DIR_OUT), eo, es);
+        return es && bsearch_u32(edges + eo, es, fcval);
+    };
+    for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < nrows;
+         base += stride) {
+        const int64_t r = base + threadIdx.x;
+        sid_t keepv[32];
+        uint32_t cnt = 0;
+        bool big = false;
+        if (r < nrows) {
+            uint32_t deg = d_cnt[r];
+            if (deg > 32) {
+                unsigned long long i = atomicAdd(
+                    (unsigned long long *)&d_state[S_OVF], 1ull);
+                ovf[i] = (uint32_t)r;
+                big = true;
+            } else if (deg) {
+                const sid_t *el = edges + d_eoff[r];
+                for (uint32_t k = 0; k < deg; k++) {
+                    sid_t v = el[k];
+                    if (pass(v)) keepv[cnt++] = v;
+                }
+            }
+        }
+        (void)big;
+        sh[threadIdx.x] = cnt;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint32_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += x;
+            __syncthreads();
+        }
+        const uint32_t my_end = sh[threadIdx.x];
+        const uint32_t btot = sh[SCAN_T - 1];
+        if (threadIdx.x == SCAN_T - 1)
+            s_base = btot ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
+                                      (unsigned long long)btot)
+                          : 0;
+        __syncthreads();
+        if (cnt) {
+            uint64_t pos = s_base + my_end - cnt;
+            sid_t row[NC];
+#pragma unroll
+            for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
+            for (uint32_t k = 0; k < cnt && pos + k < cap; k++) {
+                sid_t *dst = out + (int64_t)(pos + k) * oc;
+#pragma unroll
+                for (int c = 0; c < NC; c++) dst[c] = row[c];
+                dst[NC] = keepv[k];
+            }
+        }
+        __syncthreads();
+    }
+}
+
+// big-fanout rows of the fused path: one wave per row, ballot compaction,
+// one atomic per 64 edges
+template <int NC>
+__global__ void k_expand_filter_big(const sid_t *__restrict__ tbl, int ncols,
+                                    const sid_t *__restrict__ edges,
+                                    const uint64_t *__restrict__ d_eoff,
+                                    const uint32_t *__restrict__ d_cnt,
+                                    uint64_t *__restrict__ d_state, uint64_t cap,
+                                    const uint32_t *__restrict__ ovf,
+                                    sid_t fcval,
+                                    const uint16_t *__restrict__ type_of,
+                                    uint64_t type_base, uint64_t type_n,
+                                    const vertex_t *__restrict__ verts,
+                                    uint64_t f_bstart, uint64_t f_nbuckets,
+                                    sid_t *__restrict__ out)
+{
+    const int64_t nq = (int64_t)d_state[S_OVF];
+    if (!nq) return;
+    constexpr int oc = NC + 1;
+    (void)ncols;
+    const int lane = threadIdx.x & 63;
+    const int64_t w0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    for (int64_t q = w0; q < nq; q += nw) {
+        const int64_t r = ovf[q];
+        const uint64_t deg = d_cnt[r];
+        const sid_t *el = edges + d_eoff[r];
+        sid_t row[NC];
+#pragma unroll
+        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
+        for (uint64_t k0 = 0; k0 < deg; k0 += 64) {
+            uint64_t k = k0 + lane;
+            sid_t v = 0;
+            bool keep = false;
+            if (k < deg) {
+                v = el[k];
+                uint64_t idx = (uint64_t)v - type_base;
+                uint16_t t = (idx < type_n) ? type_of[idx] : 0;
+                if (t != 0xFFFF) {
+                    keep = ((sid_t)t == fcval);
+                } else {
+                    uint64_t eo = 0, es = 0;
+                    probe_one(verts, f_bstart, f_nbuckets,
+                              key_pack(v, TYPE_ID, (uint64_t)DIR_OUT), eo, es);
+                    keep = es && bsearch_u32(edges + eo, es, fcval);
+                }
+            }
+            uint64_t m = __ballot(keep);
+            int wcnt = __popcll(m);
+            unsigned long long wbase = 0;
+            if (lane == 0 && wcnt)
+                wbase = atomicAdd((unsigned long long *)&d_state[S_TOTAL],
+                                  (unsigned long long)wcnt);
+            wbase = __shfl((unsigned long long)wbase, 0);
+            if (keep) {
+                uint64_t pos = wbase + __popcll(m & ((1ull << lane) - 1));
+                if (pos < cap) {
+                    sid_t *dst = out + (int64_t)pos * oc;
+#pragma unroll
+                    for (int c = 0; c < NC; c++) dst[c] = row[c];
+                    dst[NC] = v;
+                }
+            }
+        }
+    }
+}
+
 // i2u / c2u: materialise an edge/index list as a 1-column table
 // (index_to_unknown sparql.hpp:194-231 / const_to_unknown :238-285)
 __global__ void k_copy_list(const sid_t *__restrict__ edges, uint64_t off,
@@ -1006,6 +1163,38 @@ static void launch_expand_t(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
                        (uint64_t)e->cap_rows, (uint32_t *)e->ovf.p, out_tbl);
 }
 
+template <int NC>
+static void launch_expand_filter_t(wk_engine *e, const sid_t *cur_tbl,
+                                   sid_t *out_tbl, sid_t fcval,
+                                   const seg_t *fseg) {
+    hipLaunchKernelGGL(k_expand_filter<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
+                       0, e->stream, cur_tbl, e->ncols, e->d_edges,
+                       (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p, e->d_state,
+                       (uint64_t)e->cap_rows, e->d_stats, (uint32_t *)e->ovf.p,
+                       fcval, e->d_type_of, e->st->type_base, e->st->type_n,
+                       e->d_verts, fseg->bucket_start, fseg->num_buckets, out_tbl);
+    hipLaunchKernelGGL(k_expand_filter_big<NC>, dim3(512), dim3(BLOCK), 0,
+                       e->stream, cur_tbl, e->ncols, e->d_edges,
+                       (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p, e->d_state,
+                       (uint64_t)e->cap_rows, (uint32_t *)e->ovf.p, fcval,
+                       e->d_type_of, e->st->type_base, e->st->type_n,
+                       e->d_verts, fseg->bucket_start, fseg->num_buckets, out_tbl);
+}
+
+static void launch_expand_filter(wk_engine *e, const sid_t *cur_tbl,
+                                 sid_t *out_tbl, sid_t fcval, const seg_t *fseg) {
+    switch (e->ncols) {
+    case 1: launch_expand_filter_t<1>(e, cur_tbl, out_tbl, fcval, fseg); break;
+    case 2: launch_expand_filter_t<2>(e, cur_tbl, out_tbl, fcval, fseg); break;
+    case 3: launch_expand_filter_t<3>(e, cur_tbl, out_tbl, fcval, fseg); break;
+    case 4: launch_expand_filter_t<4>(e, cur_tbl, out_tbl, fcval, fseg); break;
+    case 5: launch_expand_filter_t<5>(e, cur_tbl, out_tbl, fcval, fseg); break;
+    case 6: launch_expand_filter_t<6>(e, cur_tbl, out_tbl, fcval, fseg); break;
+    case 7: launch_expand_filter_t<7>(e, cur_tbl, out_tbl, fcval, fseg); break;
+    default: launch_expand_filter_t<8>(e, cur_tbl, out_tbl, fcval, fseg); break;
+    }
+}
+
 static void launch_expand(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
                           int G) {
     switch (e->ncols) {
@@ -1137,6 +1326,23 @@ static int32_t exec_pattern(wk_engine *e) {
         const int G = scan_grid(e->bound);
         int oc = e->ncols + 1;
         if (oc > e->cap_cols) return WK_ERR_STATE;  // begin_query sizes cap_cols
+
+        // look-ahead: a `?v rdf:type CONST` filter on THIS k2u's output
+        // var fuses into the expansion (saves a full table pass)
+        bool fuse = false;
+        sid_t fcval = 0;
+        const seg_t *fseg = nullptr;
+        if (e->step + 1 < (int)e->pats.size() && e->d_type_of) {
+            const wk_pattern_t &nx = e->pats[e->step + 1];
+            if (nx.subject == o && nx.predicate == (ssid_t)TYPE_ID &&
+                nx.direction == DIR_OUT && nx.object > 0) {
+                fseg = st->seg_of((uint64_t)1 << NBITS_IDX, TYPE_ID, DIR_OUT);
+                if (fseg && fseg->num_buckets) {
+                    fuse = true;
+                    fcval = (sid_t)nx.object;
+                }
+            }
+        }
         TIME_BEGIN(e);
         hipLaunchKernelGGL(k_probe_scan, dim3(G), dim3(SCAN_T), 0, e->stream,
                            e->d_verts, cur_tbl, e->ncols, col, (uint32_t)p, dir,
@@ -1145,6 +1351,25 @@ static int32_t exec_pattern(wk_engine *e) {
                            (uint32_t *)e->cnt.p, (uint64_t *)e->prefix.p,
                            (uint64_t *)e->bsums.p);
         TIME_END(e, CAT_PROBE);
+        if (fuse) {
+            hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                               e->d_state + S_TOTAL, 1);
+            hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                               e->d_state + S_OVF, 1);
+            {
+                TIME_BEGIN(e);
+                launch_expand_filter(e, cur_tbl, out_tbl, fcval, fseg);
+                TIME_END(e, CAT_EXPAND);
+            }
+            e->v2c[-(o + 1)] = e->ncols;
+            e->ncols = oc;
+            e->bound = e->cap_rows;
+            hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
+                               e->d_state, (uint64_t)e->cap_rows);
+            e->cur ^= 1;
+            e->step += 2;  // consumed the fused filter pattern too
+            return WK_OK;
+        }
         {
             TIME_BEGIN(e);
             hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0, e->stream,
@@ -1154,11 +1379,9 @@ static int32_t exec_pattern(wk_engine *e) {
         hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
                            e->d_state + S_OVF, 1);
         {
-        {
             TIME_BEGIN(e);
             launch_expand(e, cur_tbl, out_tbl, G);
             TIME_END(e, CAT_EXPAND);
-        }
         }
         e->v2c[-(o + 1)] = e->ncols;
         e->ncols = oc;
