@@ -560,28 +560,36 @@ __global__ void k_expand_filter(const sid_t *__restrict__ tbl, int ncols,
                   key_pack(val, TYPE_ID, (uint64_t)DIR_OUT), eo, es);
         return es && bsearch_u32(edges + eo, es, fcval);
     };
+    // fast path handles deg <= DF with fixed-index registers (a runtime-
+    // indexed local array goes to scratch — HIP guide rule 20); bigger
+    // rows go to the wave pass.  All of a thread's edge+type loads issue
+    // before any compare, so their latencies overlap.
+    constexpr int DF = 4;
     for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < nrows;
          base += stride) {
         const int64_t r = base + threadIdx.x;
-        sid_t keepv[32];
-        uint32_t cnt = 0;
-        bool big = false;
+        sid_t ev[DF];
+        bool pv[DF];
+        uint32_t deg = 0, cnt = 0;
         if (r < nrows) {
-            uint32_t deg = d_cnt[r];
-            if (deg > 32) {
+            deg = d_cnt[r];
+            if (deg > DF) {
                 unsigned long long i = atomicAdd(
                     (unsigned long long *)&d_state[S_OVF], 1ull);
                 ovf[i] = (uint32_t)r;
-                big = true;
+                deg = 0;
             } else if (deg) {
                 const sid_t *el = edges + d_eoff[r];
-                for (uint32_t k = 0; k < deg; k++) {
-                    sid_t v = el[k];
-                    if (pass(v)) keepv[cnt++] = v;
-                }
+#pragma unroll
+                for (int k = 0; k < DF; k++)
+                    ev[k] = (k < (int)deg) ? el[k] : 0;
+#pragma unroll
+                for (int k = 0; k < DF; k++)
+                    pv[k] = (k < (int)deg) && pass(ev[k]);
+#pragma unroll
+                for (int k = 0; k < DF; k++) cnt += pv[k] ? 1u : 0u;
             }
         }
-        (void)big;
         sh[threadIdx.x] = cnt;
         __syncthreads();
         for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
@@ -602,11 +610,14 @@ __global__ void k_expand_filter(const sid_t *__restrict__ tbl, int ncols,
             sid_t row[NC];
 #pragma unroll
             for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
-            for (uint32_t k = 0; k < cnt && pos + k < cap; k++) {
-                sid_t *dst = out + (int64_t)(pos + k) * oc;
+#pragma unroll
+            for (int k = 0; k < DF; k++) {
+                if (!pv[k] || pos >= cap) continue;
+                sid_t *dst = out + (int64_t)pos * oc;
 #pragma unroll
                 for (int c = 0; c < NC; c++) dst[c] = row[c];
-                dst[NC] = keepv[k];
+                dst[NC] = ev[k];
+                pos++;
             }
         }
         __syncthreads();
@@ -1332,7 +1343,12 @@ static int32_t exec_pattern(wk_engine *e) {
         bool fuse = false;
         sid_t fcval = 0;
         const seg_t *fseg = nullptr;
-        if (e->step + 1 < (int)e->pats.size() && e->d_type_of) {
+        // fusion needs the NEW column's type info locally: only valid on a
+        // single-partition store (type_of covers local subjects only), and
+        // it consumes two plan steps (incompatible with the per-pattern
+        // distributed driver)
+        if (e->step + 1 < (int)e->pats.size() && e->d_type_of &&
+            e->st->nsrv == 1) {
             const wk_pattern_t &nx = e->pats[e->step + 1];
             if (nx.subject == o && nx.predicate == (ssid_t)TYPE_ID &&
                 nx.direction == DIR_OUT && nx.object > 0) {
@@ -1397,6 +1413,15 @@ static int32_t exec_pattern(wk_engine *e) {
 extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_out) {
     double t0 = wk_verbose_lvl() >= 2 ? now_us() : 0;
     if (!e) return WK_ERR_STATE;
+    if (e->step >= (int)e->pats.size()) {
+        // already done (a fused step may consume two patterns): no-op
+        if (nrows_out) {
+            int32_t rc0 = sync_state(e);
+            if (rc0) return rc0;
+            *nrows_out = e->nrows;
+        }
+        return WK_OK;
+    }
     // snapshot for the overflow re-run (step mode syncs per step, so
     // e->nrows/ncols/v2c are the pattern's INPUT state here)
     const int64_t in_rows = e->nrows;
