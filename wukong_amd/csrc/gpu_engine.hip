@@ -1346,8 +1346,16 @@ static int32_t exec_pattern(wk_engine *e) {
         // fusion needs the NEW column's type info locally: only valid on a
         // single-partition store (type_of covers local subjects only), and
         // it consumes two plan steps (incompatible with the per-pattern
-        // distributed driver)
-        if (e->step + 1 < (int)e->pats.size() && e->d_type_of &&
+        // distributed driver).  DEFAULT OFF: measured net-negative until
+        // the fused kernels get the filter's 4-row batching (the per-tile
+        // counter atomics and the wave-pass per-row atomics serialize —
+        // q7's advisor expansion went 150us -> 4.2ms).  WK_FUSE=1 enables
+        // for experiments; parity is tested either way.
+        static const bool fuse_enabled = [] {
+            const char *v = getenv("WK_FUSE");
+            return v && atoi(v);
+        }();
+        if (fuse_enabled && e->step + 1 < (int)e->pats.size() && e->d_type_of &&
             e->st->nsrv == 1) {
             const wk_pattern_t &nx = e->pats[e->step + 1];
             if (nx.subject == o && nx.predicate == (ssid_t)TYPE_ID &&
