@@ -98,6 +98,9 @@ const wk_sid_t *wk_store_get_index(const wk_store_t *, wk_sid_t pid,
 uint64_t wk_store_num_slots(const wk_store_t *);
 uint64_t wk_store_num_edges(const wk_store_t *);
 uint64_t wk_store_checksum(const wk_store_t *); /* FNV over vertices+edges */
+/* Full-store integrity scan (the `gsck` command, core/store/
+ * gchecker.hpp:364-392): 0 = consistent, else #violations. */
+uint64_t wk_store_check(const wk_store_t *);
 
 /* ---------- GPU engine ------------------------------------------ */
 /* Mirrors the five-call GPU surface: load_result_buf

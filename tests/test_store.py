@@ -91,3 +91,13 @@ def test_store_partitioned_union(lubm2):
             b = owner.get_triples(int(v), pid, 1)
             assert np.array_equal(a, b)
             assert other.get_triples(int(v), pid, 1).size == 0
+
+
+def test_store_integrity_gsck(store4):
+    """The reference's gsck full-store scan (gchecker.hpp:364-392)."""
+    assert store4.check() == 0
+
+
+def test_store_integrity_watdiv():
+    st = wk.Store(wk.watdiv_gen(500, seed=1))
+    assert st.check() == 0

@@ -65,6 +65,7 @@ _get_index = _sig("wk_store_get_index", c_u32p, [c_vp, ctypes.c_uint32, c_i32, c
 _num_slots = _sig("wk_store_num_slots", c_u64, [c_vp])
 _num_edges = _sig("wk_store_num_edges", c_u64, [c_vp])
 _checksum = _sig("wk_store_checksum", c_u64, [c_vp])
+_store_check = _sig("wk_store_check", c_u64, [c_vp])
 _eng_create = _sig("wk_engine_create", c_vp, [c_vp, c_i32])
 _gstore_create = _sig("wk_gpu_store_create", c_vp, [c_vp, c_i32])
 _gstore_destroy = _sig("wk_gpu_store_destroy", None, [c_vp])
@@ -212,6 +213,10 @@ class Store:
 
     def checksum(self):
         return _checksum(self._h)
+
+    def check(self):
+        """gsck-style full integrity scan; returns #violations (0 = ok)."""
+        return _store_check(self._h)
 
 
 class GpuStore:

@@ -379,6 +379,58 @@ extern "C" uint64_t wk_store_num_slots(const wk_store_t *st) {
 extern "C" uint64_t wk_store_num_edges(const wk_store_t *st) {
     return st->edges.size();
 }
+// Full-store integrity scan — the reference's `gsck` console command
+// (GChecker::gstore_check, core/store/gchecker.hpp:364-392): every
+// slot's edge extent in bounds; every predicate-index subject/object has
+// the matching normal key; every type-index member lists that type.
+// Returns 0 if consistent, else the number of violations.
+extern "C" uint64_t wk_store_check(const wk_store_t *st) {
+    std::atomic<uint64_t> bad(0);
+    const uint64_t nslots = st->vertices.size();
+    const uint64_t nedges = st->edges.size();
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < (int64_t)nslots; i++) {
+        if (i % ASSOC == ASSOC - 1) continue;  // chain slot
+        const vertex_t &v = st->vertices[i];
+        if (v.key == KEY_EMPTY) continue;
+        uint64_t off = ptr_off(v.ptr), sz = ptr_size(v.ptr);
+        if (off + sz > nedges) { bad++; continue; }
+        // normal-key edge lists must be ascending (loader sort invariant
+        // the GPU filters rely on)
+        if (key_vid(v.key) != 0) {
+            for (uint64_t k = 1; k < sz; k++)
+                if (st->edges[off + k - 1] >= st->edges[off + k]) { bad++; break; }
+        }
+    }
+    // index <-> normal cross-check
+    for (uint32_t pid = 2; pid <= st->max_pid; pid++) {
+        for (int d = 0; d < 2; d++) {
+            uint64_t sz = 0;
+            const sid_t *lst = store_get(*st, 0, pid, d, &sz);
+            if (!lst) continue;
+#pragma omp parallel for schedule(static)
+            for (int64_t k = 0; k < (int64_t)sz; k++) {
+                sid_t v = lst[k];
+                uint64_t esz = 0;
+                if (is_tpid(v)) { bad++; continue; }
+                // [0|pid|IN] lists subjects (OUT keys); [0|pid|OUT] objects
+                // (IN keys); [0|tid|IN] lists type members ([v|TYPE|OUT])
+                const sid_t *e1 = store_get(*st, v, pid, d == DIR_IN ? DIR_OUT : DIR_IN, &esz);
+                if (e1 && esz) continue;
+                // maybe pid is a type id: member must have [v|TYPE|OUT]
+                // containing pid
+                const sid_t *t = store_get(*st, v, TYPE_ID, DIR_OUT, &esz);
+                bool ok = false;
+                if (t)
+                    for (uint64_t j = 0; j < esz; j++)
+                        if (t[j] == pid) { ok = true; break; }
+                if (!ok) bad++;
+            }
+        }
+    }
+    return bad.load();
+}
+
 extern "C" uint64_t wk_store_checksum(const wk_store_t *st) {
     // FNV-1a over slots then edges
     uint64_t h = 1469598103934665603ull;
