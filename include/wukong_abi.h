@@ -127,6 +127,23 @@ void         wk_engine_destroy(wk_engine_t *);
  * window (core/proxy.hpp:477-525).  Harvest with wk_engine_fetch_*. */
 int32_t wk_engine_submit(wk_engine_t *, const wk_plan_t *);
 
+/* Batched light-query window: one asynchronous launch executes n
+ * queries of the dominant light-template shape (const_to_unknown +
+ * rdf:type constant filter — the emulator's A1/A2/A3/A5,
+ * core/proxy.hpp:391-545), one wavefront workgroup per query.  SoA
+ * inputs: subject constants, predicates, directions, type filter
+ * constants.  Removes the kernel-dispatch-rate wall at 1024 in-flight
+ * light queries.  WK_ERR_PLAN = store has no complete single-type
+ * index; fall back to per-query wk_engine_submit. */
+int32_t wk_engine_submit_light_batch(wk_engine_t *, const int64_t *subj,
+                                     const int32_t *pred,
+                                     const int32_t *dir,
+                                     const uint32_t *cval, int32_t n);
+/* Blocks until the window completes; fills the n per-query row counts
+ * (blind replies, Result::blind — proxy.hpp:491). */
+int32_t wk_engine_light_batch_wait(wk_engine_t *, uint64_t *counts,
+                                   int32_t n);
+
 /* Whole-query execution on one GPU (Engine::execute_sparql_query +
  * SPARQLEngine::execute_patterns, core/engine/sparql.hpp:1113-1154,
  * 1564-1672, single-server path).  Returns 0 on success. */

@@ -154,3 +154,43 @@ def test_emulator_templates_parity(lubm4, store4, oracle4):
             want = oracle4.run_query(plan)
             assert got.shape == want.shape, (tname, const, got.shape, want.shape)
             assert np.array_equal(sort_rows(got), sort_rows(want)), (tname, const)
+
+
+def test_light_batch_parity(store4, oracle4):
+    """Batched light-query window (wk_engine_submit_light_batch — one
+    launch per window, one wavefront workgroup per query) vs per-query
+    oracle counts, including empty-pool constants and double-buffered
+    back-to-back windows."""
+    import random
+    eng = wk.Engine(store4, device=0)
+    rng = random.Random(11)
+    light = {}
+    for tname in Q.EMU_WEIGHTS:
+        p = Q.emu_template(tname, 1)
+        if len(p.patterns) == 2 and p.patterns[1][1] == Q.TYPE_ID:
+            light[tname] = (p.patterns[0][1], p.patterns[0][2],
+                            p.patterns[1][3])
+    assert set(light) == {"a1", "a2", "a3", "a5"}
+    for trial in range(3):
+        subj, pred, dirs, cval, plans = [], [], [], [], []
+        for _ in range(37 + trial):
+            tname = rng.choice(list(light))
+            pool = store4.get_index(Q.EMU_POOLS[tname], wk.DIR_IN)
+            const = int(rng.choice(pool))
+            pr, dr, cv = light[tname]
+            subj.append(const)
+            pred.append(pr)
+            dirs.append(dr)
+            cval.append(cv)
+            plans.append(Q.emu_template(tname, const))
+        # one nonexistent constant: must count 0, not crash
+        subj.append(3)
+        pred.append(Q.TAKESCOURSE)
+        dirs.append(wk.DIR_IN)
+        cval.append(Q.GRADSTUDENT)
+        plans.append(None)
+        eng.submit_light_batch(subj, pred, dirs, cval)
+        counts = eng.wait_light_batch()
+        for i, plan in enumerate(plans):
+            want = 0 if plan is None else len(oracle4.run_query(plan))
+            assert counts[i] == want, (trial, i, counts[i], want)

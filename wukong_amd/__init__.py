@@ -72,6 +72,12 @@ _gstore_destroy = _sig("wk_gpu_store_destroy", None, [c_vp])
 _eng_create_on = _sig("wk_engine_create_on", c_vp, [c_vp])
 _eng_submit = _sig("wk_engine_submit", c_i32, [c_vp, ctypes.POINTER(WkPlan)])
 _eng_destroy = _sig("wk_engine_destroy", None, [c_vp])
+_eng_submit_lb = _sig("wk_engine_submit_light_batch", c_i32,
+                      [c_vp, ctypes.POINTER(c_i64), ctypes.POINTER(c_i32),
+                       ctypes.POINTER(c_i32), ctypes.POINTER(ctypes.c_uint32),
+                       c_i32])
+_eng_lb_wait = _sig("wk_engine_light_batch_wait", c_i32,
+                    [c_vp, ctypes.POINTER(c_u64), c_i32])
 _eng_run = _sig("wk_engine_run_query", c_i32, [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(WkResult)])
 _eng_begin = _sig("wk_engine_begin_query", c_i32, [c_vp, ctypes.POINTER(WkPlan)])
 _eng_load = _sig("wk_engine_load_rbuf", c_i32, [c_vp, c_u32p, c_i64, c_i32, ctypes.POINTER(c_i32), c_i32])
@@ -260,6 +266,39 @@ class Engine:
         if getattr(self, "_h", None):
             _eng_destroy(self._h)
             self._h = None
+
+    def submit_light_batch(self, subj, pred, dirs, cval):
+        """One asynchronous launch for a whole window of light
+        (c2u + rdf:type filter) queries — numpy SoA in, counts out via
+        wait_light_batch.  Raises ValueError when the store cannot take
+        the fast path (caller falls back to per-query submit)."""
+        subj = np.ascontiguousarray(subj, dtype=np.int64)
+        pred = np.ascontiguousarray(pred, dtype=np.int32)
+        dirs = np.ascontiguousarray(dirs, dtype=np.int32)
+        cval = np.ascontiguousarray(cval, dtype=np.uint32)
+        n = len(subj)
+        assert len(pred) == n and len(dirs) == n and len(cval) == n
+        rc = _eng_submit_lb(
+            self._h,
+            subj.ctypes.data_as(ctypes.POINTER(c_i64)),
+            pred.ctypes.data_as(ctypes.POINTER(c_i32)),
+            dirs.ctypes.data_as(ctypes.POINTER(c_i32)),
+            cval.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)), n)
+        if rc == -3:
+            raise ValueError("store has no single-type index (fast path off)")
+        if rc != 0:
+            raise RuntimeError(f"submit_light_batch rc={rc}")
+        self._lb_n = n
+
+    def wait_light_batch(self):
+        """Block until the window completes; per-query row counts."""
+        n = self._lb_n
+        counts = np.empty(n, dtype=np.uint64)
+        rc = _eng_lb_wait(self._h, counts.ctypes.data_as(ctypes.POINTER(c_u64)), n)
+        if rc != 0:
+            raise RuntimeError(f"light_batch_wait rc={rc}")
+        self._lb_n = None
+        return counts
 
     def run_query_count(self, plan):
         """Blind execution (Result::blind): row count only, no download."""

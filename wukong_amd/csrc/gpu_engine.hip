@@ -774,6 +774,52 @@ __global__ void k_light2(const sid_t *__restrict__ edges, uint64_t list_off,
     }
 }
 
+// batched light queries: ONE launch executes a whole window of
+// light2-shaped queries (const_to_unknown + rdf:type filter), one
+// wavefront-sized workgroup per query.  This is the engine-side answer
+// to the reference proxy's in-flight window (proxy.hpp:477-525): at
+// 1024 in-flight light queries the bottleneck is kernel DISPATCH rate
+// (~10us/launch), so the scheduler packs the window into one grid.
+struct light_desc {
+    uint64_t off;      // edge-list offset of the c2u constant
+    uint64_t sz;       // edge-list length
+    uint64_t out_off;  // this query's region in the shared out buffer
+    uint32_t cval;     // rdf:type filter constant
+    uint32_t pad;
+};
+
+static const int LBLOCK = 64;  // one wavefront per query
+
+__global__ void k_light_batch(const sid_t *__restrict__ edges,
+                              const light_desc *__restrict__ descs,
+                              const uint16_t *__restrict__ type_of,
+                              uint64_t type_base, uint64_t type_n,
+                              sid_t *__restrict__ out,
+                              uint64_t *__restrict__ d_counts,
+                              uint64_t *__restrict__ d_stats)
+{
+    const light_desc d = descs[blockIdx.x];
+    __shared__ unsigned int cnt;
+    if (threadIdx.x == 0) cnt = 0;
+    __syncthreads();
+    sid_t *qout = out + d.out_off;
+    for (uint64_t i = threadIdx.x; i < d.sz; i += blockDim.x) {
+        sid_t v = edges[d.off + i];
+        uint64_t idx = (uint64_t)v - type_base;
+        uint16_t t = (idx < type_n) ? type_of[idx] : 0;
+        if ((sid_t)t == d.cval) {
+            unsigned p = atomicAdd(&cnt, 1u);
+            qout[p] = v;
+        }
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        d_counts[blockIdx.x] = cnt;
+        atomicAdd((unsigned long long *)&d_stats[CAT_FILTER],
+                  (unsigned long long)(d.sz * 6 + cnt * 4));
+    }
+}
+
 __global__ void k_zero_words(uint64_t *p, int n) {
     for (int i = threadIdx.x; i < n; i += blockDim.x) p[i] = 0;
 }
@@ -848,6 +894,13 @@ struct wk_engine {
     int64_t bound = 0;   // host-side upper bound on rows (grid sizing)
 
     bool light = false;  // single-kernel fast path ran (h_pin self-published)
+
+    // batched light-query window (wk_engine_submit_light_batch)
+    devbuf lbd;               // device light_desc array
+    devbuf lbcnt;             // device per-query counts
+    uint64_t *h_lb = nullptr; // pinned: descs staging, then counts
+    size_t h_lb_cap = 0;
+    int lb_n = -1;            // queries in flight in the batch (-1 = none)
 
     // query state (host mirror of SPARQLQuery, query.hpp:560-594)
     std::vector<wk_pattern_t> pats;
@@ -1038,6 +1091,8 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
     for (int i = 0; i < 2; i++) e->tbl[i].release();
     e->eoff.release(); e->cnt.release(); e->prefix.release();
     e->bsums.release(); e->misc.release(); e->ovf.release();
+    e->lbd.release(); e->lbcnt.release();
+    if (e->h_lb) (void)hipHostFree(e->h_lb);
     if (e->d_state) (void)hipFree(e->d_state);
     if (e->h_pin) (void)hipHostFree(e->h_pin);
     if (e->h_stage) (void)hipHostFree(e->h_stage);
@@ -1719,6 +1774,83 @@ extern "C" int32_t wk_engine_submit(wk_engine_t *e, const wk_plan_t *plan) {
         rc = exec_pattern(e);
         if (rc) return rc;
     }
+    return WK_OK;
+}
+
+// Batched light-query window: n queries of the light2 shape
+// (subject=constant, predicate, direction → ?x; rdf:type ?x == cval),
+// SoA arrays so the host hands numpy buffers straight through.  One
+// async launch for the whole window; harvest with
+// wk_engine_light_batch_wait.  Returns WK_ERR_PLAN when the store
+// cannot take the fast path (no single-type index) — callers fall back
+// to per-query wk_engine_submit.
+extern "C" int32_t wk_engine_submit_light_batch(wk_engine_t *e,
+                                                const int64_t *subj,
+                                                const int32_t *pred,
+                                                const int32_t *dirs,
+                                                const uint32_t *cval,
+                                                int32_t n) {
+    if (!e || !subj || !pred || !dirs || !cval || n <= 0 || n > (1 << 20))
+        return WK_ERR_STATE;
+    if (!e->d_type_of || e->st->type_multi) return WK_ERR_PLAN;
+    if (e->lb_n >= 0) return WK_ERR_STATE;  // previous batch not harvested
+    resolve_timing(e);
+
+    size_t need = (size_t)n * (sizeof(light_desc) + 8);
+    if (e->h_lb_cap < need) {
+        if (e->h_lb) (void)hipHostFree(e->h_lb);
+        e->h_lb = nullptr; e->h_lb_cap = 0;
+        if (hipHostMalloc(&e->h_lb, need) != hipSuccess) return WK_ERR_HIP;
+        e->h_lb_cap = need;
+    }
+    if (e->lbd.ensure((size_t)n * sizeof(light_desc))) return WK_ERR_HIP;
+    if (e->lbcnt.ensure((size_t)n * 8)) return WK_ERR_HIP;
+
+    light_desc *descs = (light_desc *)e->h_lb;
+    uint64_t total = 0;
+    for (int i = 0; i < n; i++) {
+        uint64_t sz = 0;
+        const sid_t *ptr = store_get(*e->st, (uint64_t)subj[i],
+                                     (uint64_t)pred[i], (int)dirs[i], &sz);
+        descs[i].off = ptr ? (uint64_t)(ptr - e->st->edges.data()) : 0;
+        descs[i].sz = sz;
+        descs[i].out_off = total;
+        descs[i].cval = cval[i];
+        descs[i].pad = 0;
+        total += sz;
+    }
+    int32_t rc = grow_caps(e, (int64_t)total, 1);
+    if (rc) return rc;
+    HIP_CHECK(hipMemcpyAsync(e->lbd.p, descs, (size_t)n * sizeof(light_desc),
+                             hipMemcpyHostToDevice, e->stream));
+    TIME_BEGIN(e);
+    hipLaunchKernelGGL(k_light_batch, dim3(n), dim3(LBLOCK), 0, e->stream,
+                       e->d_edges, (const light_desc *)e->lbd.p,
+                       e->d_type_of, e->st->type_base, e->st->type_n,
+                       (sid_t *)e->tbl[1].p, (uint64_t *)e->lbcnt.p,
+                       e->d_stats);
+    TIME_END(e, CAT_FILTER);
+    uint64_t *h_counts = (uint64_t *)(descs + n);
+    HIP_CHECK(hipMemcpyAsync(h_counts, e->lbcnt.p, (size_t)n * 8,
+                             hipMemcpyDeviceToHost, e->stream));
+    e->lb_n = n;
+    e->cur = 1;
+    e->ncols = 1;
+    e->light = false;
+    return WK_OK;
+}
+
+// Blocks until the window completes; writes the n per-query row counts
+// (blind replies, Result::blind — proxy.hpp:491).
+extern "C" int32_t wk_engine_light_batch_wait(wk_engine_t *e,
+                                              uint64_t *counts, int32_t n) {
+    if (!e || e->lb_n < 0 || !counts || n != e->lb_n) return WK_ERR_STATE;
+    HIP_CHECK(stream_sync(e->stream));
+    const uint64_t *h_counts =
+        (const uint64_t *)((const light_desc *)e->h_lb + e->lb_n);
+    memcpy(counts, h_counts, (size_t)n * 8);
+    e->lb_n = -1;
+    resolve_timing(e);
     return WK_OK;
 }
 
