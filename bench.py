@@ -73,11 +73,12 @@ def run_emulator_batched(args, store, engines, inflight):
     """Batched emulator: the light templates (A1/A2/A3/A5 — 78/86 of
     the mix) go through wk_engine_submit_light_batch, ONE kernel launch
     per window of queries (one wavefront workgroup each), double-
-    buffered across two engines; the heavy templates (A4/A6) run on the
-    per-query pipelined path on side threads.  Same mix, same blind
+    buffered across two engines; the heavy templates (A4/A6) batch per
+    template through the LDS plan interpreter
+    (wk_engine_submit_plan_batch — one wavefront workgroup interprets
+    the whole plan, binding table in LDS).  Same mix, same blind
     replies — the batching is engine-side scheduling, not a workload
     change."""
-    import threading
     from collections import deque
     import numpy as np
     import wukong_amd as wk
@@ -101,10 +102,17 @@ def run_emulator_batched(args, store, engines, inflight):
     light_idx = np.array([tn in light_meta for tn in tnames])
 
     B = int(os.environ.get("WK_EMU_WINDOW", str(max(inflight, 512))))
-    light_engines = engines[:2]
-    heavy_engines = engines[2:] or engines[:1]
-    nhthreads = min(int(os.environ.get("WK_EMU_HTHREADS", "3")),
-                    len(heavy_engines))
+    heavy_names = [t for t in tnames if t not in light_meta]
+    light_engines = list(engines[:2])
+    hv, k = {}, 2
+    for t in heavy_names:
+        hv[t] = list(engines[k:k + 2]) or [engines[0]]
+        k += 2
+    # heavy templates run through the LDS plan interpreter, batched per
+    # template (wk_engine_submit_plan_batch); subject is per-query
+    heavy_plans = {t: Q.emu_template(t, 1 << 17) for t in heavy_names}
+    for p in heavy_plans.values():
+        p.blind = True
 
     class Batch:
         __slots__ = ("labels", "subj", "pred", "dirs", "cval", "t_gen",
@@ -132,7 +140,7 @@ def run_emulator_batched(args, store, engines, inflight):
                 dirs[mask] = dr
                 cval[mask] = cv
             else:
-                heavy.extend((tn, int(c)) for c in consts)
+                heavy.append((tn, consts))
         b.labels = [tnames[i] for i in ti[lmask]]
         b.subj, b.pred = subj[lmask], pred[lmask]
         b.dirs, b.cval = dirs[lmask], cval[lmask]
@@ -140,88 +148,87 @@ def run_emulator_batched(args, store, engines, inflight):
         return b, heavy
 
     def run(total, collect=None):
-        hq = deque()
-        done = threading.Event()
-
-        def heavy_worker(engs):
-            free = list(engs)
-            pend = deque()
-            while True:
-                item = None
-                try:
-                    item = hq.popleft()
-                except IndexError:
-                    if not pend:
-                        if done.is_set():
-                            return
-                        time.sleep(0.0002)
-                        continue
-                if item is not None and free:
-                    tn, c, tg = item
-                    p = Q.emu_template(tn, c)
-                    p.blind = True
-                    e = free.pop()
-                    e.submit(p)
-                    pend.append((tn, e, tg, p))
-                    continue
-                if item is not None:
-                    hq.appendleft(item)  # no free engine: drain first
-                tn, e, tg, p = pend.popleft()
-                n_ = e.fetch_count()
-                while n_ < 0:
-                    e.submit(p)
-                    n_ = e.fetch_count()
-                if collect is not None:
-                    collect.setdefault(tn, []).append((time.time() - tg) * 1e3)
-                free.append(e)
-
-        chunks = [heavy_engines[i::nhthreads] for i in range(nhthreads)]
-        hts = [threading.Thread(target=heavy_worker, args=(chunks[i],))
-               for i in range(nhthreads)]
-        for t in hts:
-            t.start()
         produced = 0
         pending = deque()
-        free = deque(light_engines)
-        while produced < total or pending:
-            if produced < total and free:
-                n = min(B, total - produced)
+        free_light = deque(light_engines)
+        free_heavy = {t: deque(hv[t]) for t in heavy_names}
+        acc = {t: [] for t in heavy_names}    # accumulated consts arrays
+        accg = {t: [] for t in heavy_names}   # their gen times
+        rerun = [0]
+
+        def harvest_one():
+            kind, e, payload = pending.popleft()
+            if kind == "light":
+                b = payload
+                e.wait_light_batch()
+                now = time.time()
+                if collect is not None:
+                    ms = (now - b.t_gen) * 1e3
+                    for tn in b.labels:
+                        collect.setdefault(tn, []).append(ms)
+                free_light.append(e)
+            else:
+                tn, consts, gens = payload
+                counts = e.wait_light_batch()
+                for j in np.nonzero(counts == np.uint64(wk.LP_OVERFLOW))[0]:
+                    p = Q.emu_template(tn, int(consts[j]))
+                    p.blind = True
+                    e.run_query_count(p)
+                    rerun[0] += 1
+                now = time.time()
+                if collect is not None:
+                    for tg, m in gens:
+                        ms = (now - tg) * 1e3
+                        collect.setdefault(tn, []).extend([ms] * m)
+                free_heavy[tn].append(e)
+
+        while produced < total or pending or any(acc[t] for t in heavy_names):
+            progress = False
+            if produced < total and free_light:
                 tg = time.time()
-                b, heavy = make_batch(n)
-                for tn, c in heavy:
-                    hq.append((tn, c, tg))
+                b, heavy = make_batch(min(B, total - produced))
                 produced += b.total
+                for tn, consts in heavy:
+                    acc[tn].append(consts)
+                    accg[tn].append((tg, len(consts)))
                 if b.nlight:
-                    e = free.popleft()
+                    e = free_light.popleft()
                     e.submit_light_batch(b.subj, b.pred, b.dirs, b.cval)
                     b.t_gen = tg
-                    pending.append((e, b))
-                continue
-            e, b = pending.popleft()
-            e.wait_light_batch()
-            now = time.time()
-            if collect is not None:
-                ms = (now - b.t_gen) * 1e3
-                for tn in b.labels:
-                    collect.setdefault(tn, []).append(ms)
-            free.append(e)
-        done.set()
-        for t in hts:
-            t.join()
+                    pending.append(("light", e, b))
+                progress = True
+            for tn in heavy_names:
+                if acc[tn] and free_heavy[tn]:
+                    e = free_heavy[tn].popleft()
+                    consts = np.concatenate(acc[tn])
+                    gens = accg[tn]
+                    acc[tn], accg[tn] = [], []
+                    e.submit_plan_batch(heavy_plans[tn], consts)
+                    pending.append(("heavy", e, (tn, consts, gens)))
+                    progress = True
+            if pending and (not progress or len(pending) >= 6):
+                harvest_one()
+        return rerun[0]
 
-    # gate: store must support the fast path (single-type index)
+    # gate: store must support the fast paths (single-type index +
+    # interpretable heavy templates); otherwise per-query emulator
     try:
         probe, _ = make_batch(4)
-        light_engines[0].submit_light_batch(probe.subj, probe.pred,
-                                            probe.dirs, probe.cval)
-        light_engines[0].wait_light_batch()
+        if probe.nlight:
+            light_engines[0].submit_light_batch(probe.subj, probe.pred,
+                                                probe.dirs, probe.cval)
+            light_engines[0].wait_light_batch()
+        for tn in heavy_names:
+            hv[tn][0].submit_plan_batch(heavy_plans[tn],
+                                        pools[tn][:2].astype(np.int64))
+            hv[tn][0].wait_light_batch()
     except ValueError:
         return False
 
     run(max(args.emu // 10, 4 * B))  # warmup
     lat = {}
     t0 = time.time()
-    run(args.emu, collect=lat)
+    nrerun = run(args.emu, collect=lat)
     elapsed = time.time() - t0
     latency = {t: {"p50_ms": round(pctl(xs, 50), 3),
                    "p99_ms": round(pctl(xs, 99), 3), "n": len(xs)}
@@ -234,7 +241,9 @@ def run_emulator_batched(args, store, engines, inflight):
         "queries": args.emu,
         "inflight": 2 * B,
         "window": B,
-        "mode": "batched (one launch per window, wavefront/query)",
+        "mode": "batched (one launch per window, wavefront/query; "
+                "heavy templates via LDS plan interpreter)",
+        "plan_batch_overflow_reruns": nrerun,
         "higher_is_better": True,
         "dtype": "u32",
         "data": "synthetic",

@@ -139,6 +139,16 @@ int32_t wk_engine_submit_light_batch(wk_engine_t *, const int64_t *subj,
                                      const int32_t *pred,
                                      const int32_t *dir,
                                      const uint32_t *cval, int32_t n);
+/* Batched whole-plan light queries (the LDS plan interpreter): n
+ * same-template queries, ONE launch; each wavefront workgroup
+ * interprets the compiled plan (const-start, then typeof/expand/
+ * filter patterns) with its binding table staged in LDS, covering the
+ * emulator's multi-pattern templates A4/A6.  consts[i] replaces
+ * patterns[0].subject.  Blind replies only; a query whose table
+ * outgrows LDS reports count UINT64_MAX and must be re-run on the
+ * per-pattern path.  WK_ERR_PLAN = template shape not interpretable. */
+int32_t wk_engine_submit_plan_batch(wk_engine_t *, const wk_plan_t *tmpl,
+                                    const int64_t *consts, int32_t n);
 /* Blocks until the window completes; fills the n per-query row counts
  * (blind replies, Result::blind — proxy.hpp:491). */
 int32_t wk_engine_light_batch_wait(wk_engine_t *, uint64_t *counts,

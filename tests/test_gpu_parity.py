@@ -194,3 +194,31 @@ def test_light_batch_parity(store4, oracle4):
         for i, plan in enumerate(plans):
             want = 0 if plan is None else len(oracle4.run_query(plan))
             assert counts[i] == want, (trial, i, counts[i], want)
+
+
+def test_plan_batch_parity(store4, oracle4):
+    """LDS plan interpreter (wk_engine_submit_plan_batch): whole
+    multi-pattern templates batched one workgroup per query — counts vs
+    the oracle, including a nonexistent constant."""
+    import random
+    eng = wk.Engine(store4, device=0)
+    rng = random.Random(5)
+    for tname in ["a4", "a6", "a1"]:
+        pool = store4.get_index(Q.EMU_POOLS[tname], wk.DIR_IN)
+        consts = [int(rng.choice(pool)) for _ in range(23)]
+        consts.append(1 << 17)  # vid likely without this predicate
+        tmpl = Q.emu_template(tname, 1 << 17)
+        eng.submit_plan_batch(tmpl, consts)
+        counts = eng.wait_light_batch()
+        assert not np.any(counts == np.uint64(wk.LP_OVERFLOW)), tname
+        for i, c in enumerate(consts):
+            want = len(oracle4.run_query(Q.emu_template(tname, c)))
+            assert counts[i] == want, (tname, c, counts[i], want)
+
+
+def test_plan_batch_rejects_uninterpretable(store4):
+    """Index-start plans (Q1) are not LDS-interpretable — must raise
+    (fallback contract), not mis-execute."""
+    eng = wk.Engine(store4, device=0)
+    with pytest.raises(ValueError):
+        eng.submit_plan_batch(Q.Q1, [1 << 17])

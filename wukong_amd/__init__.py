@@ -78,6 +78,12 @@ _eng_submit_lb = _sig("wk_engine_submit_light_batch", c_i32,
                        c_i32])
 _eng_lb_wait = _sig("wk_engine_light_batch_wait", c_i32,
                     [c_vp, ctypes.POINTER(c_u64), c_i32])
+_eng_submit_pb = _sig("wk_engine_submit_plan_batch", c_i32,
+                      [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(c_i64),
+                       c_i32])
+
+#: plan-batch count sentinel: table outgrew LDS, re-run per-pattern
+LP_OVERFLOW = 0xFFFFFFFFFFFFFFFF
 _eng_run = _sig("wk_engine_run_query", c_i32, [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(WkResult)])
 _eng_begin = _sig("wk_engine_begin_query", c_i32, [c_vp, ctypes.POINTER(WkPlan)])
 _eng_load = _sig("wk_engine_load_rbuf", c_i32, [c_vp, c_u32p, c_i64, c_i32, ctypes.POINTER(c_i32), c_i32])
@@ -289,6 +295,24 @@ class Engine:
         if rc != 0:
             raise RuntimeError(f"submit_light_batch rc={rc}")
         self._lb_n = n
+
+    def submit_plan_batch(self, plan, consts):
+        """One launch for a window of same-template whole plans (LDS
+        plan interpreter — A4/A6 shapes).  consts replaces the
+        template's first-pattern subject per query.  Blind counts via
+        wait_light_batch; LP_OVERFLOW entries must be re-run per-query.
+        Raises ValueError when the template is not interpretable."""
+        consts = np.ascontiguousarray(consts, dtype=np.int64)
+        cplan = plan.to_c()
+        rc = _eng_submit_pb(self._h, ctypes.byref(cplan),
+                            consts.ctypes.data_as(ctypes.POINTER(c_i64)),
+                            len(consts))
+        if rc == -3:
+            raise ValueError("plan not interpretable by the LDS plan batch")
+        if rc != 0:
+            raise RuntimeError(f"submit_plan_batch rc={rc}")
+        self._lb_n = len(consts)
+        self._cplan_batch = cplan  # keepalive for pattern array
 
     def wait_light_batch(self):
         """Block until the window completes; per-query row counts."""

@@ -820,6 +820,156 @@ __global__ void k_light_batch(const sid_t *__restrict__ edges,
     }
 }
 
+// ---------------------------------------------------------------------
+// batched whole-plan light queries: the LDS plan interpreter.
+//
+// The emulator's heavy templates (A4: 5 patterns, A6: 4 patterns,
+// proxy.hpp:391-545) have tiny intermediate tables (tens-hundreds of
+// rows) — per-pattern global kernels waste ~20 dispatches/query.  Here
+// ONE launch runs a window of same-template queries; each wavefront-
+// sized workgroup interprets the whole compiled plan with its binding
+// table staged in LDS (ping-pong buffers), probing the cluster hash in
+// global memory (64 independent probes in flight per wave).  A query
+// whose table outgrows LDS publishes WK_LP_OVERFLOW and the host
+// re-runs it on the general per-pattern path — results identical.
+// ---------------------------------------------------------------------
+enum { LOP_C2U = 0, LOP_TYPEOF, LOP_K2U, LOP_K2C, LOP_K2K };
+struct lop_t {
+    int32_t op, col, col2, dir;
+    uint32_t cval, pid;
+    uint64_t bucket_start, num_buckets;  // segment of (pid,dir)
+};
+struct lplan_t {
+    lop_t ops[8];
+    int32_t nops;
+};
+static const int LP_CAP = 6144;            // u32 slots per LDS buffer
+static const uint64_t WK_LP_OVERFLOW = ~0ull;
+
+__global__ void
+__launch_bounds__(64)
+k_plan_batch(const vertex_t *__restrict__ verts,
+             const sid_t *__restrict__ edges, const lplan_t lp,
+             const int64_t *__restrict__ consts,
+             const uint16_t *__restrict__ type_of, uint64_t type_base,
+             uint64_t type_n, uint64_t *__restrict__ d_counts,
+             uint64_t *__restrict__ d_stats)
+{
+    __shared__ sid_t bufA[LP_CAP], bufB[LP_CAP];
+    __shared__ unsigned s_n, s_cur, s_err;
+    __shared__ uint64_t s_eoff, s_esz, s_bytes;
+    sid_t *cur = bufA, *nxt = bufB;
+    int nc = 0;
+    uint64_t my_bytes = 0;
+    if (threadIdx.x == 0) { s_n = 0; s_err = 0; s_bytes = 0; }
+    __syncthreads();
+
+    for (int o = 0; o < lp.nops && !s_err; o++) {
+        const lop_t op = lp.ops[o];
+        if (op.op == LOP_C2U) {
+            if (threadIdx.x == 0) {
+                uint64_t key = key_pack((uint64_t)consts[blockIdx.x],
+                                        (uint64_t)op.pid, (uint64_t)op.dir);
+                s_eoff = 0; s_esz = 0;
+                if (op.num_buckets)
+                    probe_one(verts, op.bucket_start, op.num_buckets, key,
+                              s_eoff, s_esz);
+                if (s_esz > (uint64_t)LP_CAP) s_err = 1;
+                s_n = (unsigned)s_esz;
+                my_bytes += 148;
+            }
+            __syncthreads();
+            if (s_err) break;
+            for (unsigned i = threadIdx.x; i < s_n; i += blockDim.x)
+                cur[i] = edges[s_eoff + i];
+            if (threadIdx.x == 0) my_bytes += (uint64_t)s_n * 8;
+            nc = 1;
+            __syncthreads();
+        } else if (op.op == LOP_TYPEOF) {
+            if (threadIdx.x == 0) s_cur = 0;
+            __syncthreads();
+            for (unsigned r = threadIdx.x; r < s_n; r += blockDim.x) {
+                sid_t v = cur[r * nc + op.col];
+                uint64_t idx = (uint64_t)v - type_base;
+                uint16_t t = (idx < type_n) ? type_of[idx] : 0;
+                my_bytes += 6;
+                if ((sid_t)t == (sid_t)op.cval) {
+                    unsigned p = atomicAdd(&s_cur, 1u);
+                    for (int c = 0; c < nc; c++)
+                        nxt[p * nc + c] = cur[r * nc + c];
+                    my_bytes += (uint64_t)nc * 4;
+                }
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) s_n = s_cur;
+            sid_t *t_ = cur; cur = nxt; nxt = t_;
+            __syncthreads();
+        } else if (op.op == LOP_K2C || op.op == LOP_K2K) {
+            if (threadIdx.x == 0) s_cur = 0;
+            __syncthreads();
+            for (unsigned r = threadIdx.x; r < s_n; r += blockDim.x) {
+                sid_t v = cur[r * nc + op.col];
+                uint64_t eoff = 0, esz = 0;
+                uint64_t key = key_pack((uint64_t)v, (uint64_t)op.pid,
+                                        (uint64_t)op.dir);
+                if (op.num_buckets)
+                    probe_one(verts, op.bucket_start, op.num_buckets, key,
+                              eoff, esz);
+                sid_t tgt = (op.op == LOP_K2C) ? (sid_t)op.cval
+                                               : cur[r * nc + op.col2];
+                my_bytes += 148 + 32;  // probe + ~log2 bsearch lines
+                if (esz && bsearch_u32(edges + eoff, esz, tgt)) {
+                    unsigned p = atomicAdd(&s_cur, 1u);
+                    for (int c = 0; c < nc; c++)
+                        nxt[p * nc + c] = cur[r * nc + c];
+                    my_bytes += (uint64_t)nc * 4;
+                }
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) s_n = s_cur;
+            sid_t *t_ = cur; cur = nxt; nxt = t_;
+            __syncthreads();
+        } else {  // LOP_K2U
+            if (threadIdx.x == 0) s_cur = 0;
+            __syncthreads();
+            for (unsigned r = threadIdx.x; r < s_n; r += blockDim.x) {
+                sid_t v = cur[r * nc + op.col];
+                uint64_t eoff = 0, esz = 0;
+                uint64_t key = key_pack((uint64_t)v, (uint64_t)op.pid,
+                                        (uint64_t)op.dir);
+                if (op.num_buckets)
+                    probe_one(verts, op.bucket_start, op.num_buckets, key,
+                              eoff, esz);
+                my_bytes += 148;
+                unsigned base = atomicAdd(&s_cur, (unsigned)esz);
+                if (((uint64_t)base + esz) * (nc + 1) > (uint64_t)LP_CAP) {
+                    s_err = 1;
+                    continue;
+                }
+                for (uint64_t e = 0; e < esz; e++) {
+                    sid_t *dst = nxt + (base + e) * (nc + 1);
+                    for (int c = 0; c < nc; c++) dst[c] = cur[r * nc + c];
+                    dst[nc] = edges[eoff + e];
+                }
+                my_bytes += esz * (uint64_t)(nc + 1) * 4 + esz * 4;
+            }
+            __syncthreads();
+            if (threadIdx.x == 0) s_n = s_cur;
+            sid_t *t_ = cur; cur = nxt; nxt = t_;
+            nc++;
+            __syncthreads();
+        }
+    }
+    __syncthreads();
+    atomicAdd((unsigned long long *)&s_bytes, (unsigned long long)my_bytes);
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        d_counts[blockIdx.x] = s_err ? WK_LP_OVERFLOW : (uint64_t)s_n;
+        atomicAdd((unsigned long long *)&d_stats[CAT_FILTER],
+                  (unsigned long long)s_bytes);
+    }
+}
+
 __global__ void k_zero_words(uint64_t *p, int n) {
     for (int i = threadIdx.x; i < n; i += blockDim.x) p[i] = 0;
 }
@@ -898,8 +1048,9 @@ struct wk_engine {
     // batched light-query window (wk_engine_submit_light_batch)
     devbuf lbd;               // device light_desc array
     devbuf lbcnt;             // device per-query counts
-    uint64_t *h_lb = nullptr; // pinned: descs staging, then counts
+    uint64_t *h_lb = nullptr; // pinned: descs/consts staging, then counts
     size_t h_lb_cap = 0;
+    size_t lb_coff = 0;       // byte offset of the counts region in h_lb
     int lb_n = -1;            // queries in flight in the batch (-1 = none)
 
     // query state (host mirror of SPARQLQuery, query.hpp:560-594)
@@ -1833,10 +1984,121 @@ extern "C" int32_t wk_engine_submit_light_batch(wk_engine_t *e,
     uint64_t *h_counts = (uint64_t *)(descs + n);
     HIP_CHECK(hipMemcpyAsync(h_counts, e->lbcnt.p, (size_t)n * 8,
                              hipMemcpyDeviceToHost, e->stream));
+    e->lb_coff = (size_t)n * sizeof(light_desc);
     e->lb_n = n;
     e->cur = 1;
     e->ncols = 1;
     e->light = false;
+    return WK_OK;
+}
+
+// Compile a light whole-plan template for the LDS interpreter; the
+// per-query constant replaces patterns[0].subject.  WK_ERR_PLAN = shape
+// not interpretable (caller uses the per-pattern path instead).
+static int32_t compile_light_plan(const wk_engine *e, const wk_plan_t *plan,
+                                  lplan_t *lp) {
+    if (!plan || plan->npatterns < 1 || plan->npatterns > 8) return WK_ERR_PLAN;
+    if (plan->nvars < 1 || plan->nvars > 8) return WK_ERR_PLAN;
+    if (plan->distinct || plan->limit >= 0) return WK_ERR_PLAN;
+    const wk_store *st = e->st;
+    int32_t v2c[8];
+    for (int i = 0; i < 8; i++) v2c[i] = -1;
+
+    const wk_pattern_t &p0 = plan->patterns[0];
+    if (p0.subject < 0 || is_tpid(p0.subject) || p0.object >= 0)
+        return WK_ERR_PLAN;  // must be const_to_unknown
+    const wk::seg_t *seg = st->seg_of(1ull << NBITS_IDX,
+                                      (uint64_t)p0.predicate, p0.direction);
+    lp->ops[0] = {LOP_C2U, 0, 0, p0.direction, 0, (uint32_t)p0.predicate,
+                  seg ? seg->bucket_start : 0, seg ? seg->num_buckets : 0};
+    v2c[-(p0.object + 1)] = 0;
+    int nc = 1, no = 1;
+
+    for (int i = 1; i < plan->npatterns; i++) {
+        const wk_pattern_t &p = plan->patterns[i];
+        if (p.subject >= 0) return WK_ERR_PLAN;  // const-start mid-plan
+        int sidx = -(p.subject + 1);
+        if (sidx >= plan->nvars || v2c[sidx] < 0) return WK_ERR_PLAN;
+        int scol = v2c[sidx];
+        lop_t op = {};
+        op.col = scol;
+        op.dir = p.direction;
+        op.pid = (uint32_t)p.predicate;
+        if (p.object >= 0) {  // constant object
+            if ((sid_t)p.predicate == TYPE_ID && p.direction == DIR_OUT &&
+                st->type_n && !st->type_multi) {
+                op.op = LOP_TYPEOF;
+                op.cval = (uint32_t)p.object;
+            } else {
+                const wk::seg_t *s2 = st->seg_of(
+                    1ull << NBITS_IDX, (uint64_t)p.predicate, p.direction);
+                op.op = LOP_K2C;
+                op.cval = (uint32_t)p.object;
+                op.bucket_start = s2 ? s2->bucket_start : 0;
+                op.num_buckets = s2 ? s2->num_buckets : 0;
+            }
+        } else {
+            int oidx = -(p.object + 1);
+            if (oidx >= plan->nvars) return WK_ERR_PLAN;
+            const wk::seg_t *s2 = st->seg_of(
+                1ull << NBITS_IDX, (uint64_t)p.predicate, p.direction);
+            op.bucket_start = s2 ? s2->bucket_start : 0;
+            op.num_buckets = s2 ? s2->num_buckets : 0;
+            if (v2c[oidx] >= 0) {
+                op.op = LOP_K2K;
+                op.col2 = v2c[oidx];
+            } else {
+                op.op = LOP_K2U;
+                if (nc >= 8) return WK_ERR_PLAN;
+                v2c[oidx] = nc++;
+            }
+        }
+        lp->ops[no++] = op;
+    }
+    lp->nops = no;
+    return WK_OK;
+}
+
+// Batched whole-plan light queries: n same-template queries, ONE launch,
+// one wavefront workgroup interpreting the compiled plan per query with
+// the binding table in LDS.  Blind replies only (counts); a query whose
+// table outgrows LDS reports WK_LP_OVERFLOW_COUNT and must be re-run on
+// the per-pattern path.  Harvest with wk_engine_light_batch_wait.
+extern "C" int32_t wk_engine_submit_plan_batch(wk_engine_t *e,
+                                               const wk_plan_t *tmpl,
+                                               const int64_t *consts,
+                                               int32_t n) {
+    if (!e || !tmpl || !consts || n <= 0 || n > (1 << 20)) return WK_ERR_STATE;
+    if (e->lb_n >= 0) return WK_ERR_STATE;
+    lplan_t lp = {};
+    int32_t rc = compile_light_plan(e, tmpl, &lp);
+    if (rc) return rc;
+    resolve_timing(e);
+
+    size_t need = (size_t)n * 16;  // consts then counts
+    if (e->h_lb_cap < need) {
+        if (e->h_lb) (void)hipHostFree(e->h_lb);
+        e->h_lb = nullptr; e->h_lb_cap = 0;
+        if (hipHostMalloc(&e->h_lb, need) != hipSuccess) return WK_ERR_HIP;
+        e->h_lb_cap = need;
+    }
+    if (e->lbd.ensure((size_t)n * 8)) return WK_ERR_HIP;
+    if (e->lbcnt.ensure((size_t)n * 8)) return WK_ERR_HIP;
+    memcpy(e->h_lb, consts, (size_t)n * 8);
+    HIP_CHECK(hipMemcpyAsync(e->lbd.p, e->h_lb, (size_t)n * 8,
+                             hipMemcpyHostToDevice, e->stream));
+    TIME_BEGIN(e);
+    hipLaunchKernelGGL(k_plan_batch, dim3(n), dim3(64), 0, e->stream,
+                       e->d_verts, e->d_edges, lp,
+                       (const int64_t *)e->lbd.p, e->d_type_of,
+                       e->st->type_base, e->st->type_n,
+                       (uint64_t *)e->lbcnt.p, e->d_stats);
+    TIME_END(e, CAT_FILTER);
+    uint64_t *h_counts = (uint64_t *)((char *)e->h_lb + (size_t)n * 8);
+    HIP_CHECK(hipMemcpyAsync(h_counts, e->lbcnt.p, (size_t)n * 8,
+                             hipMemcpyDeviceToHost, e->stream));
+    e->lb_coff = (size_t)n * 8;
+    e->lb_n = n;
     return WK_OK;
 }
 
@@ -1847,7 +2109,7 @@ extern "C" int32_t wk_engine_light_batch_wait(wk_engine_t *e,
     if (!e || e->lb_n < 0 || !counts || n != e->lb_n) return WK_ERR_STATE;
     HIP_CHECK(stream_sync(e->stream));
     const uint64_t *h_counts =
-        (const uint64_t *)((const light_desc *)e->h_lb + e->lb_n);
+        (const uint64_t *)((const char *)e->h_lb + e->lb_coff);
     memcpy(counts, h_counts, (size_t)n * 8);
     e->lb_n = -1;
     resolve_timing(e);
