@@ -63,6 +63,89 @@ int64_t ok_brute_query(void *cv, const ok_pattern_t *pats, int32_t npat,
             v2c[-(o + 1)] = 0;
             continue;
         }
+        // p < 0: VERSATILE (predicate variable) — match every predicate,
+        // the predicate id becomes a column.  Independent restatement:
+        // no predicate lists, just raw triple matching.
+        if (p < 0) {
+            int ostat = (o >= 0) ? 2 : (col_of(o) >= 0 ? 1 : 0);
+            std::vector<sid_t> next;
+            auto each = [&](auto &&f) {
+                for (const auto &t : T) {
+                    if (d == 1) f(t.s, t.o, t.p);
+                    else f(t.o, t.s, t.p);
+                }
+            };
+            if (a >= 0) {  // const start (first pattern)
+                if (ostat == 0) {
+                    each([&](sid_t va, sid_t vb, sid_t pp) {
+                        if (va == (sid_t)a) {
+                            next.push_back(pp);
+                            next.push_back(vb);
+                        }
+                    });
+                    table.swap(next);
+                    v2c[-(p + 1)] = C;
+                    v2c[-(o + 1)] = C + 1;
+                    C += 2;
+                } else {  // const_unknown_const: distinct linking preds
+                    std::unordered_set<sid_t> preds;
+                    each([&](sid_t va, sid_t vb, sid_t pp) {
+                        if (va == (sid_t)a && vb == (sid_t)o) preds.insert(pp);
+                    });
+                    std::vector<sid_t> ps(preds.begin(), preds.end());
+                    std::sort(ps.begin(), ps.end());
+                    table.assign(ps.begin(), ps.end());
+                    v2c[-(p + 1)] = C;
+                    C += 1;
+                }
+                continue;
+            }
+            int ca = col_of(a);
+            int64_t R = C ? (int64_t)table.size() / C : 0;
+            if (ostat == 0) {  // known_unknown_unknown
+                std::unordered_multimap<sid_t, std::pair<sid_t, sid_t>> mm;
+                each([&](sid_t va, sid_t vb, sid_t pp) {
+                    mm.emplace(va, std::make_pair(pp, vb));
+                });
+                for (int64_t i = 0; i < R; i++) {
+                    auto range = mm.equal_range(table[i * C + ca]);
+                    for (auto it = range.first; it != range.second; ++it) {
+                        next.insert(next.end(), table.begin() + i * C,
+                                    table.begin() + (i + 1) * C);
+                        next.push_back(it->second.first);
+                        next.push_back(it->second.second);
+                    }
+                }
+                table.swap(next);
+                v2c[-(p + 1)] = C;
+                v2c[-(o + 1)] = C + 1;
+                C += 2;
+            } else {  // known_unknown_const: one row per distinct (va, p)
+                std::unordered_set<uint64_t> vap;
+                each([&](sid_t va, sid_t vb, sid_t pp) {
+                    if (vb == (sid_t)o)
+                        vap.insert(((uint64_t)va << 32) | pp);
+                });
+                for (int64_t i = 0; i < R; i++) {
+                    sid_t va = table[i * C + ca];
+                    // iterate candidate preds in ascending order: collect
+                    std::vector<sid_t> ps;
+                    for (uint64_t key : vap)
+                        if ((sid_t)(key >> 32) == va)
+                            ps.push_back((sid_t)(key & 0xFFFFFFFFu));
+                    std::sort(ps.begin(), ps.end());
+                    for (sid_t pp : ps) {
+                        next.insert(next.end(), table.begin() + i * C,
+                                    table.begin() + (i + 1) * C);
+                        next.push_back(pp);
+                    }
+                }
+                table.swap(next);
+                v2c[-(p + 1)] = C;
+                C += 1;
+            }
+            continue;
+        }
         for (const auto &t : T) {
             if (t.p != (sid_t)p) continue;
             if (d == 1) pairs.emplace_back(t.s, t.o);

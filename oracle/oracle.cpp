@@ -126,6 +126,37 @@ static ctx *build(const sid_t *spo, int64_t n, int sid, int nsrv) {
     for (auto &kvp : pidx_in) put(0, kvp.first, DIR_IN, kvp.second);
     for (auto &kvp : pidx_out) put(0, kvp.first, DIR_OUT, kvp.second);
     for (auto &kvp : tidx) put(0, kvp.first, DIR_IN, kvp.second);
+    // VERSATILE per-vertex predicate lists [vid|PREDICATE_ID|dir] —
+    // insert_vp, static_gstore.hpp:282-374: OUT from pso runs INCLUDING
+    // type triples; IN from pos runs skipping tpid objects.  Runs are
+    // iterated in (p, vid) order, so each list is ascending-pid.
+    {
+        std::unordered_map<sid_t, std::vector<sid_t>> vp_out, vp_in;
+        for (size_t i = 0; i < pso.size();) {
+            size_t j = i + 1;
+            while (j < pso.size() && pso[j].p == pso[i].p &&
+                   pso[j].s == pso[i].s)
+                j++;
+            vp_out[pso[i].s].push_back(pso[i].p);
+            i = j;
+        }
+        for (size_t i = 0; i < pos.size();) {
+            size_t j = i + 1;
+            while (j < pos.size() && pos[j].p == pos[i].p &&
+                   pos[j].o == pos[i].o)
+                j++;
+            if (!is_tpid(pos[i].o)) vp_in[pos[i].o].push_back(pos[i].p);
+            i = j;
+        }
+        for (auto &kvp : vp_out) {
+            std::sort(kvp.second.begin(), kvp.second.end());
+            put(kvp.first, PREDICATE_ID, DIR_OUT, kvp.second);
+        }
+        for (auto &kvp : vp_in) {
+            std::sort(kvp.second.begin(), kvp.second.end());
+            put(kvp.first, PREDICATE_ID, DIR_IN, kvp.second);
+        }
+    }
     return c;
 }
 
@@ -164,6 +195,51 @@ static void exec_pattern(const ctx &c, query &q) {
         q.table.swap(out);
         q.col_num = 1;
         q.v2c[-(o + 1)] = 0;
+        q.step++;
+        return;
+    }
+
+    // VERSATILE ops (predicate variable) — sparql.hpp:556-744: read the
+    // vertex's predicate list under [vid|PREDICATE_ID|dir], then probe
+    // each (vid, p, dir); *_unknown_unknown appends (p, y) per edge,
+    // *_unknown_const appends one row per matching p.
+    if (p < 0) {
+        std::vector<sid_t> out;
+        auto run_vid = [&](sid_t vid, const sid_t *row, int rowlen) {
+            uint64_t npids = 0;
+            const sid_t *pids = c.get(vid, PREDICATE_ID, d, &npids);
+            for (uint64_t pi = 0; pi < npids; pi++) {
+                uint64_t esz = 0;
+                const sid_t *vids2 = c.get(vid, pids[pi], d, &esz);
+                if (o >= 0) {
+                    for (uint64_t k = 0; k < esz; k++)
+                        if (vids2[k] == (sid_t)o) {
+                            out.insert(out.end(), row, row + rowlen);
+                            out.push_back(pids[pi]);
+                            break;
+                        }
+                } else {
+                    for (uint64_t k = 0; k < esz; k++) {
+                        out.insert(out.end(), row, row + rowlen);
+                        out.push_back(pids[pi]);
+                        out.push_back(vids2[k]);
+                    }
+                }
+            }
+        };
+        if (s >= 0) {  // const_unknown_* (first pattern, sparql.hpp:719)
+            run_vid((sid_t)s, nullptr, 0);
+        } else {
+            int col = q.var2col(s);
+            int64_t R = q.nrows();
+            for (int64_t i = 0; i < R; i++)
+                run_vid(q.table[i * q.col_num + col],
+                        q.table.data() + i * q.col_num, q.col_num);
+        }
+        q.table.swap(out);
+        q.v2c[-(p + 1)] = q.col_num;
+        if (o < 0) q.v2c[-(o + 1)] = q.col_num + 1;
+        q.col_num += (o >= 0) ? 1 : 2;
         q.step++;
         return;
     }

@@ -970,6 +970,81 @@ k_plan_batch(const vertex_t *__restrict__ verts,
     }
 }
 
+// ---------------------------------------------------------------------
+// VERSATILE predicate-variable ops (sparql.hpp:556-744): one kernel
+// covers known/const_unknown_unknown (append p [, y] columns) and
+// known/const_unknown_const (keep one row per matching p).  Block per
+// input row; the row vid's predicate list comes from the dense vp CSR;
+// each predicate probes its own (pid,dir) segment via the device
+// segment table.  Appends use the global S_TOTAL cursor (k_commit
+// finalises; overflow -> S_ERR/S_REQ re-run like every other op).
+// ---------------------------------------------------------------------
+enum { VU_END_NEW = 0, VU_END_CONST = 1 };
+__global__ void k_vu(const vertex_t *__restrict__ verts,
+                     const sid_t *__restrict__ edges,
+                     const sid_t *__restrict__ in_tbl, int ncols, int col,
+                     sid_t const_start,
+                     const uint32_t *__restrict__ vp_off,
+                     const sid_t *__restrict__ vp_edges,
+                     uint64_t vp_base, uint64_t vp_n,
+                     const seg_t *__restrict__ segtab, uint32_t max_pid,
+                     int dir, int end_mode, sid_t cval,
+                     sid_t *__restrict__ out, int out_cols, uint64_t cap_rows,
+                     uint64_t *__restrict__ d_state,
+                     uint64_t *__restrict__ d_stats)
+{
+    const uint64_t nrows = (col >= 0) ? d_state[S_NROWS] : 1;
+    uint64_t bytes = 0;
+    for (uint64_t r = blockIdx.x; r < nrows; r += gridDim.x) {
+        sid_t vid = (col >= 0) ? in_tbl[r * ncols + col] : const_start;
+        uint64_t idx = (uint64_t)vid - vp_base;
+        if (idx >= vp_n) continue;
+        uint64_t plo = vp_off[idx], phi = vp_off[idx + 1];
+        bytes += 8;
+        for (uint64_t pi = plo + threadIdx.x; pi < phi; pi += blockDim.x) {
+            sid_t p = vp_edges[pi];
+            uint64_t eoff = 0, esz = 0;
+            if (p <= max_pid) {
+                const seg_t sg = segtab[p * 2 + dir];
+                if (sg.num_buckets)
+                    probe_one(verts, sg.bucket_start, sg.num_buckets,
+                              key_pack((uint64_t)vid, (uint64_t)p,
+                                       (uint64_t)dir),
+                              eoff, esz);
+            }
+            bytes += 4 + 148;
+            if (end_mode == VU_END_CONST) {
+                if (esz && bsearch_u32(edges + eoff, esz, cval)) {
+                    uint64_t pos = atomicAdd(
+                        (unsigned long long *)&d_state[S_TOTAL], 1ull);
+                    if (pos < cap_rows) {
+                        sid_t *dst = out + pos * out_cols;
+                        for (int c = 0; c < ncols; c++)
+                            dst[c] = in_tbl[r * ncols + c];
+                        dst[ncols] = p;
+                        bytes += (uint64_t)out_cols * 4;
+                    }
+                }
+            } else {
+                uint64_t pos = atomicAdd(
+                    (unsigned long long *)&d_state[S_TOTAL],
+                    (unsigned long long)esz);
+                for (uint64_t k = 0; k < esz && pos + k < cap_rows; k++) {
+                    sid_t *dst = out + (pos + k) * out_cols;
+                    for (int c = 0; c < ncols; c++)
+                        dst[c] = in_tbl[r * ncols + c];
+                    dst[ncols] = p;
+                    dst[ncols + 1] = edges[eoff + k];
+                }
+                bytes += esz * ((uint64_t)out_cols * 4 + 4);
+            }
+        }
+    }
+    if (bytes)
+        atomicAdd((unsigned long long *)&d_stats[CAT_EXPAND],
+                  (unsigned long long)bytes);
+}
+
 __global__ void k_zero_words(uint64_t *p, int n) {
     for (int i = threadIdx.x; i < n; i += blockDim.x) p[i] = 0;
 }
@@ -1010,6 +1085,11 @@ struct wk_gpu_store {
     vertex_t *d_verts = nullptr;
     sid_t *d_edges = nullptr;
     uint16_t *d_type_of = nullptr;
+    // VERSATILE dense CSR (per-vertex predicate lists) + the per-
+    // (pid,dir) segment table the predicate-variable ops probe through
+    uint32_t *d_vp_off[2] = {nullptr, nullptr};
+    sid_t *d_vp_edges[2] = {nullptr, nullptr};
+    seg_t *d_segtab = nullptr;
     int refs = 0;     // engines attached
     bool owned = false;  // created implicitly by wk_engine_create
 };
@@ -1188,6 +1268,28 @@ extern "C" wk_gpu_store_t *wk_gpu_store_create(const wk_store_t *st, int32_t dev
             return nullptr;
         }
     }
+    if (st->vp_n) {
+        for (int d = 0; d < 2; d++) {
+            size_t ob = (st->vp_n + 1) * 4;
+            size_t eb = st->vp_edges[d].size() * 4;
+            if (hipMalloc(&g->d_vp_off[d], ob) != hipSuccess ||
+                hipMalloc(&g->d_vp_edges[d], eb ? eb : 16) != hipSuccess ||
+                hipMemcpy(g->d_vp_off[d], st->vp_off[d].data(), ob,
+                          hipMemcpyHostToDevice) != hipSuccess ||
+                (eb && hipMemcpy(g->d_vp_edges[d], st->vp_edges[d].data(), eb,
+                                 hipMemcpyHostToDevice) != hipSuccess)) {
+                wk_gpu_store_destroy(g);
+                return nullptr;
+            }
+        }
+        size_t sb = st->nseg.size() * sizeof(seg_t);
+        if (hipMalloc(&g->d_segtab, sb) != hipSuccess ||
+            hipMemcpy(g->d_segtab, st->nseg.data(), sb,
+                      hipMemcpyHostToDevice) != hipSuccess) {
+            wk_gpu_store_destroy(g);
+            return nullptr;
+        }
+    }
     return g;
 }
 
@@ -1197,6 +1299,11 @@ extern "C" void wk_gpu_store_destroy(wk_gpu_store_t *g) {
     if (g->d_verts) (void)hipFree(g->d_verts);
     if (g->d_edges) (void)hipFree(g->d_edges);
     if (g->d_type_of) (void)hipFree(g->d_type_of);
+    for (int d = 0; d < 2; d++) {
+        if (g->d_vp_off[d]) (void)hipFree(g->d_vp_off[d]);
+        if (g->d_vp_edges[d]) (void)hipFree(g->d_vp_edges[d]);
+    }
+    if (g->d_segtab) (void)hipFree(g->d_segtab);
     delete g;
 }
 
@@ -1436,6 +1543,51 @@ static int32_t exec_pattern(wk_engine *e) {
     const int dir = pat.direction;
     sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
     sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
+
+    // ---- VERSATILE: predicate variable (sparql.hpp:556-744) ----
+    if (p < 0) {
+        if (!st->vp_n || !e->gs || !e->gs->d_segtab) return WK_ERR_PLAN;
+        int pvar = -(p + 1);
+        if (pvar >= e->nvars || e->v2c[pvar] >= 0) return WK_ERR_PLAN;
+        const bool cstart = s >= 0;
+        int col = -1;
+        if (cstart) {
+            // const_unknown_* must be the first pattern (sparql.hpp:719)
+            if (e->ncols != 0 || is_tpid(s)) return WK_ERR_PLAN;
+        } else {
+            col = e->var2col(s);
+            if (col < 0) return WK_ERR_PLAN;
+        }
+        const int ostat = (o >= 0) ? 2 : (e->var2col(o) >= 0 ? 1 : 0);
+        if (ostat == 1) return WK_ERR_PLAN;  // known_unknown_known: absent
+                                             // in the reference too
+        const int end_mode = (ostat == 2) ? VU_END_CONST : VU_END_NEW;
+        const int oc = e->ncols + (end_mode == VU_END_CONST ? 1 : 2);
+        if (oc > e->cap_cols) return WK_ERR_STATE;
+        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                           e->d_state + S_TOTAL, 1);
+        int G = (int)std::min<int64_t>(std::max<int64_t>(e->bound, 1), 4096);
+        if (cstart) G = 1;
+        TIME_BEGIN(e);
+        hipLaunchKernelGGL(k_vu, dim3(G), dim3(64), 0, e->stream,
+                           e->d_verts, e->d_edges, cur_tbl, e->ncols, col,
+                           cstart ? (sid_t)s : 0,
+                           e->gs->d_vp_off[dir], e->gs->d_vp_edges[dir],
+                           st->vp_base, st->vp_n, e->gs->d_segtab,
+                           st->max_pid, dir, end_mode,
+                           ostat == 2 ? (sid_t)o : 0, out_tbl, oc,
+                           (uint64_t)e->cap_rows, e->d_state, e->d_stats);
+        TIME_END(e, CAT_EXPAND);
+        hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
+                           e->d_state, (uint64_t)e->cap_rows);
+        e->v2c[pvar] = e->ncols;
+        if (end_mode == VU_END_NEW) e->v2c[-(o + 1)] = e->ncols + 1;
+        e->ncols = oc;
+        e->cur ^= 1;
+        e->bound = e->cap_rows;
+        e->step++;
+        return WK_OK;
+    }
 
     // ---- step 0: index start (query.hpp:660-682) / const start ----
     const bool index_start = (e->step == 0 && s >= 0 && is_tpid(s));
@@ -2005,8 +2157,9 @@ static int32_t compile_light_plan(const wk_engine *e, const wk_plan_t *plan,
     for (int i = 0; i < 8; i++) v2c[i] = -1;
 
     const wk_pattern_t &p0 = plan->patterns[0];
-    if (p0.subject < 0 || is_tpid(p0.subject) || p0.object >= 0)
-        return WK_ERR_PLAN;  // must be const_to_unknown
+    if (p0.subject < 0 || is_tpid(p0.subject) || p0.object >= 0 ||
+        p0.predicate < 0)
+        return WK_ERR_PLAN;  // must be const_to_unknown, fixed predicate
     const wk::seg_t *seg = st->seg_of(1ull << NBITS_IDX,
                                       (uint64_t)p0.predicate, p0.direction);
     lp->ops[0] = {LOP_C2U, 0, 0, p0.direction, 0, (uint32_t)p0.predicate,
@@ -2017,6 +2170,7 @@ static int32_t compile_light_plan(const wk_engine *e, const wk_plan_t *plan,
     for (int i = 1; i < plan->npatterns; i++) {
         const wk_pattern_t &p = plan->patterns[i];
         if (p.subject >= 0) return WK_ERR_PLAN;  // const-start mid-plan
+        if (p.predicate < 0) return WK_ERR_PLAN;  // predicate variable
         int sidx = -(p.subject + 1);
         if (sidx >= plan->nvars || v2c[sidx] < 0) return WK_ERR_PLAN;
         int scol = v2c[sidx];

@@ -357,6 +357,72 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
         }
     }
     WK_LOG("[store] type index: %.1fs\n", now_s() - t0);
+    t0 = now_s();
+    // VERSATILE per-vertex predicate lists ([vid|PREDICATE_ID|dir],
+    // static_gstore.hpp:282-374: OUT from pso runs INCLUDING type
+    // triples; IN from pos runs skipping tpid objects) — dense CSR over
+    // the vid range instead of hash keys (vids are dense; probe = 2
+    // loads).  Lists are ascending-pid (built by ascending-p slices),
+    // matching the sorted-edge-list invariant.  WK_VERSATILE=0 disables.
+    {
+        const char *vv = getenv("WK_VERSATILE");
+        bool want = !(vv && !atoi(vv));
+        if (want && max_id >= (1u << NBITS_IDX)) {
+            st->vp_base = 1u << NBITS_IDX;
+            st->vp_n = (uint64_t)max_id + 1 - st->vp_base;
+            for (int dir = 0; dir < 2; dir++) {
+                const bool out = dir == DIR_OUT;
+                const std::vector<triple> &arr = out ? pso : pos;
+                const std::vector<int64_t> &lo = out ? pso_lo : pos_lo;
+                st->vp_off[dir].alloc(st->vp_n + 1, /*zero=*/true);
+                uint32_t *offp = st->vp_off[dir].data();
+                // count distinct (vid, p) pairs per vid (shifted by 1
+                // for the in-place exclusive scan)
+#pragma omp parallel for schedule(dynamic)
+                for (uint32_t p = 0; p <= max_pid; p++) {
+                    for (int64_t i = lo[p]; i < lo[p + 1];) {
+                        sid_t v = out ? arr[i].s : arr[i].o;
+                        int64_t j = i + 1;
+                        while (j < lo[p + 1] && (out ? arr[j].s : arr[j].o) == v)
+                            j++;
+                        if ((uint64_t)v >= st->vp_base && !(!out && is_tpid(v)))
+                            __atomic_fetch_add(&offp[v - st->vp_base + 1], 1u,
+                                               __ATOMIC_RELAXED);
+                        i = j;
+                    }
+                }
+                uint64_t total = 0;
+                for (uint64_t i = 1; i <= st->vp_n; i++) {
+                    total += offp[i];
+                    if (total > 0xFFFFFFFFull) { total = 0; break; }
+                    offp[i] = (uint32_t)total;
+                }
+                if (!total && st->vp_n) {  // u32 overflow: disable VERSATILE
+                    st->vp_n = 0;
+                    st->vp_off[0].alloc(0, false);
+                    st->vp_off[1].alloc(0, false);
+                    break;
+                }
+                st->vp_edges[dir].alloc(total, /*zero=*/false);
+                std::vector<uint32_t> cursor(offp, offp + st->vp_n);
+                // fill: sequential ascending p, parallel over runs (each
+                // vid appears once per slice -> no cursor races)
+                for (uint32_t p = 0; p <= max_pid; p++) {
+#pragma omp parallel for schedule(static)
+                    for (int64_t i = lo[p]; i < lo[p + 1]; i++) {
+                        sid_t v = out ? arr[i].s : arr[i].o;
+                        bool first = (i == lo[p]) ||
+                                     (out ? arr[i - 1].s : arr[i - 1].o) != v;
+                        if (!first) continue;
+                        if ((uint64_t)v < st->vp_base || (!out && is_tpid(v)))
+                            continue;
+                        st->vp_edges[dir][cursor[v - st->vp_base]++] = p;
+                    }
+                }
+            }
+        }
+    }
+    WK_LOG("[store] versatile vp: %.1fs\n", now_s() - t0);
     st->ext_used = ext_next.load() - st->nbuckets_main;
     return st;
 }

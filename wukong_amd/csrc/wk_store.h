@@ -64,6 +64,15 @@ struct wk_store {
     // index segments [dir]: keys [0|pid|dir] / [0|tid|IN]
     wk::seg_t iseg[2];
 
+    // VERSATILE per-vertex predicate lists ([vid|PREDICATE_ID|dir],
+    // static_gstore.hpp:282-374 semantics: OUT includes rdf:type, IN
+    // skips type triples) stored as a dense CSR over the vid range —
+    // vids are dense, so 2 loads replace the cluster-hash walk
+    // (same deliberate densification as type_of).
+    pod_array<uint32_t> vp_off[2];   // size vp_n+1 (CSR offsets)
+    pod_array<wk::sid_t> vp_edges[2];
+    uint64_t vp_base = 0, vp_n = 0;
+
     uint64_t nbuckets_main = 0, nbuckets_ext = 0, ext_used = 0;
 
     const wk::seg_t *seg_of(uint64_t vid, uint64_t pid, int dir) const {
@@ -79,6 +88,15 @@ namespace wk {
 inline const sid_t *store_get(const wk_store &st, uint64_t vid, uint64_t pid,
                               int dir, uint64_t *sz) {
     *sz = 0;
+    // [vid|PREDICATE_ID|dir] = the vertex's predicate list (VERSATILE,
+    // gstore get_triples surface) — served from the dense CSR
+    if (pid == PREDICATE_ID && vid != 0) {
+        uint64_t idx = vid - st.vp_base;
+        if (idx >= st.vp_n) return nullptr;
+        uint64_t lo = st.vp_off[dir][idx], hi = st.vp_off[dir][idx + 1];
+        *sz = hi - lo;
+        return *sz ? st.vp_edges[dir].data() + lo : nullptr;
+    }
     const seg_t *seg = st.seg_of(vid, pid, dir);
     if (!seg || seg->num_buckets == 0) return nullptr;
     uint64_t key = key_pack(vid, pid, (uint64_t)dir);

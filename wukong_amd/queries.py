@@ -122,3 +122,49 @@ def emu_template(name, const):
 EMU_POOLS = {"a1": GRADCOURSE, "a2": ASSTPROF, "a3": DEPARTMENT,
              "a4": DEPARTMENT, "a5": DEPARTMENT, "a6": UNIVERSITY}
 EMU_WEIGHTS = {"a1": 25, "a2": 25, "a3": 3, "a4": 6, "a5": 25, "a6": 2}
+
+
+def versatile_plans(store):
+    """LUBM Q8-Q12 — the VERSATILE (predicate-variable) queries
+    (scripts/sparql_query/lubm/basic/lubm_q{8..12}; operators
+    sparql.hpp:556-744).  The reference names its constants by URI
+    (Department3.University0/FullProfessor1, ...); our generator ids are
+    allocation-order, so the constants are derived positionally from the
+    store: dept k of university 0, the first FullProfessor of dept 3, an
+    UndergraduateStudent member of dept 7.  Same shapes/selectivity."""
+    import numpy as np
+
+    def tris(v, p, d):
+        return np.sort(np.asarray(store.get_triples(int(v), p, d)))
+
+    def first_of_type(cands, t):
+        for c in cands:
+            if t in tris(c, TYPE_ID, DIR_OUT):
+                return int(c)
+        raise RuntimeError("no candidate of requested type")
+
+    depts = tris(UNIV0, SUBORG, DIR_IN)
+    dept3 = int(depts[3 % len(depts)])
+    dept7 = int(depts[7 % len(depts)])
+    prof = first_of_type(tris(dept3, WORKSFOR, DIR_IN), FULLPROF)
+    members = tris(dept7, MEMBEROF, DIR_IN)
+    ug = [int(c) for c in members if UGSTUDENT in tris(c, TYPE_ID, DIR_OUT)]
+    student = ug[201 % len(ug)]
+
+    # Q8: ?D subOrgOf Univ0 . ?X ?P ?D  (known_unknown_unknown, d=IN)
+    q8 = Plan([(UNIV0, SUBORG, DIR_IN, -1),
+               (-1, -2, DIR_IN, -3)], 3, [-3, -2, -1])
+    # Q9: ?X ?P Dept0.Univ0  (const_unknown_unknown, d=IN)
+    q9 = Plan([(DEPT0_UNIV0, -1, DIR_IN, -2)], 2, [-2, -1])
+    # Q10: ?X type UGStudent . ?X ?Y Prof  (planner starts from the
+    # constant: const_unknown_unknown + typeof filter)
+    q10 = Plan([(prof, -1, DIR_IN, -2),
+                (-2, TYPE_ID, DIR_OUT, UGSTUDENT)], 2, [-2, -1])
+    # Q11: Student ?X Dept7  (const_unknown_const)
+    q11 = Plan([(student, -1, DIR_OUT, dept7)], 1, [-1])
+    # Q12: ?Y subOrgOf Univ0 . ?X worksFor ?Y . ?Z advisor ?X
+    # (fixed predicates; ?Z is generated but not selected)
+    q12 = Plan([(UNIV0, SUBORG, DIR_IN, -1),
+                (-1, WORKSFOR, DIR_IN, -2),
+                (-2, ADVISOR, DIR_IN, -3)], 3, [-2, -1])
+    return {"q8": q8, "q9": q9, "q10": q10, "q11": q11, "q12": q12}
