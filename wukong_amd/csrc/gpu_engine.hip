@@ -283,6 +283,8 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                              const uint16_t *__restrict__ type_of,
                              uint64_t type_base, uint64_t type_n,
                              int use_typeof,
+                             const sid_t *__restrict__ fn, uint64_t fn_base,
+                             uint64_t fn_n,
                              uint64_t *__restrict__ d_state,
                              uint64_t *__restrict__ d_stats,
                              sid_t *__restrict__ out_tbl)
@@ -290,6 +292,7 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     count_bytes(d_stats, CAT_FILTER,
                 (uint64_t)nrows * ((use_typeof ? 6
+                                    : fn ? 8
                                     : probe_mode == PM_LIST ? 12
                                                             : (4 + 128 + 8 + 64)) +
                                    8 * ncols));
@@ -319,6 +322,14 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
             }
             if (probe_mode == PM_LIST && !use_typeof) {
                 keep[k] = bsearch_u32(edges + list_off, list_sz, v);
+            } else if (fn) {
+                // functional predicate: the row's single object replaces
+                // the probe + edge-list search (4 bytes vs 148)
+                uint64_t idx = (uint64_t)v - fn_base;
+                sid_t tv = (idx < fn_n) ? fn[idx] : 0;
+                sid_t tgt = (probe_mode == PM_CONST) ? cval
+                                                     : tbl[r * ncols + col2];
+                keep[k] = tv && tv == tgt;
             } else {
                 uint64_t key = (key_mode == PK_NORMAL)
                                    ? key_pack(v, pid, (uint64_t)dir)
@@ -520,6 +531,106 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
             for (int c = 0; c < NC; c++) dst[c] = row[c];
             dst[NC] = el[k];
         }
+    }
+}
+
+// known_to_unknown over a FUNCTIONAL predicate (every key deg==1, dense
+// vid->object map): the probe+scan+expand pipeline collapses into one
+// 4-byte gather + block-compacted append — 148 bytes of cluster-hash
+// traffic per row become 4.  An optional fused `?v rdf:type CONST`
+// filter on the NEW column (plan pairs like Q1's ugDegreeFrom ->
+// University) drops rows in flight with the same block-aggregated
+// compaction (one atomic per 1024 rows — no contention, unlike the
+// wave-pass fusion this replaces).  Multi-type (0xFFFF) falls back to a
+// probe of [val|TYPE_ID|OUT].  Output rows <= input rows, so capacity
+// can never overflow.
+template <int NC>
+__global__ void k_expand_fn(const sid_t *__restrict__ tbl,
+                            const sid_t *__restrict__ fn, uint64_t fn_base,
+                            uint64_t fn_n, int col, int use_typeof,
+                            sid_t fcval,
+                            const uint16_t *__restrict__ type_of,
+                            uint64_t type_base, uint64_t type_n,
+                            const vertex_t *__restrict__ verts,
+                            const sid_t *__restrict__ edges,
+                            uint64_t f_bstart, uint64_t f_nbuckets,
+                            uint64_t *__restrict__ d_state,
+                            uint64_t *__restrict__ d_stats,
+                            sid_t *__restrict__ out)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    constexpr int oc = NC + 1;
+    count_bytes(d_stats, CAT_EXPAND,
+                (uint64_t)nrows * (8 + (use_typeof ? 2 : 0)));
+    constexpr int K = 4;
+    __shared__ unsigned long long s_base;
+    __shared__ uint32_t sh[SCAN_T];
+    const int64_t tile = (int64_t)blockDim.x * K;
+    const int64_t stride = (int64_t)gridDim.x * tile;
+
+    for (int64_t base = (int64_t)blockIdx.x * tile; base < nrows; base += stride) {
+        bool keep[K];
+        sid_t val[K];
+        int64_t rr[K];
+        uint32_t cnt = 0;
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            const int64_t r = base + (int64_t)threadIdx.x * K + k;
+            rr[k] = r;
+            keep[k] = false;
+            val[k] = 0;
+            if (r >= nrows) continue;
+            sid_t v = tbl[r * NC + col];
+            uint64_t idx = (uint64_t)v - fn_base;
+            sid_t tv = (idx < fn_n) ? fn[idx] : 0;
+            val[k] = tv;
+            if (!tv) continue;
+            if (!use_typeof) { keep[k] = true; continue; }
+            uint64_t tix = (uint64_t)tv - type_base;
+            uint16_t t = (tix < type_n) ? type_of[tix] : 0;
+            if (t != 0xFFFF) {
+                keep[k] = ((sid_t)t == fcval);
+            } else {
+                uint64_t eo = 0, es = 0;
+                probe_one(verts, f_bstart, f_nbuckets,
+                          key_pack(tv, TYPE_ID, (uint64_t)DIR_OUT), eo, es);
+                keep[k] = es && bsearch_u32(edges + eo, es, fcval);
+            }
+        }
+#pragma unroll
+        for (int k = 0; k < K; k++) cnt += keep[k] ? 1u : 0u;
+        sh[threadIdx.x] = cnt;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint32_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += x;
+            __syncthreads();
+        }
+        const uint32_t my_end = sh[threadIdx.x];
+        const uint32_t block_total = sh[SCAN_T - 1];
+        if (threadIdx.x == SCAN_T - 1) {
+            s_base = block_total
+                         ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
+                                     (unsigned long long)block_total)
+                         : 0;
+            if (block_total)
+                atomicAdd((unsigned long long *)&d_stats[CAT_EXPAND],
+                          (unsigned long long)block_total * (NC * 4 + oc * 4));
+        }
+        __syncthreads();
+        uint64_t pos = s_base + my_end - cnt;
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            if (!keep[k]) continue;
+            sid_t *dst = out + (int64_t)pos * oc;
+            const sid_t *src = tbl + rr[k] * NC;
+#pragma unroll
+            for (int c = 0; c < NC; c++) dst[c] = src[c];
+            dst[NC] = val[k];
+            pos++;
+        }
+        __syncthreads();
     }
 }
 
@@ -1090,6 +1201,9 @@ struct wk_gpu_store {
     uint32_t *d_vp_off[2] = {nullptr, nullptr};
     sid_t *d_vp_edges[2] = {nullptr, nullptr};
     seg_t *d_segtab = nullptr;
+    // functional-predicate dense maps, indexed [pid*2+dir] (host vector
+    // of device pointers; null = absent)
+    std::vector<sid_t *> d_fn;
     int refs = 0;     // engines attached
     bool owned = false;  // created implicitly by wk_engine_create
 };
@@ -1290,6 +1404,19 @@ extern "C" wk_gpu_store_t *wk_gpu_store_create(const wk_store_t *st, int32_t dev
             return nullptr;
         }
     }
+    if (st->fn_n) {
+        g->d_fn.assign(st->fn.size(), nullptr);
+        for (size_t w = 0; w < st->fn.size(); w++) {
+            if (st->fn[w].empty()) continue;
+            size_t fb = st->fn_n * 4;
+            if (hipMalloc(&g->d_fn[w], fb) != hipSuccess ||
+                hipMemcpy(g->d_fn[w], st->fn[w].data(), fb,
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                wk_gpu_store_destroy(g);
+                return nullptr;
+            }
+        }
+    }
     return g;
 }
 
@@ -1304,6 +1431,8 @@ extern "C" void wk_gpu_store_destroy(wk_gpu_store_t *g) {
         if (g->d_vp_edges[d]) (void)hipFree(g->d_vp_edges[d]);
     }
     if (g->d_segtab) (void)hipFree(g->d_segtab);
+    for (sid_t *p : g->d_fn)
+        if (p) (void)hipFree(p);
     delete g;
 }
 
@@ -1519,6 +1648,34 @@ static void launch_expand_filter(wk_engine *e, const sid_t *cur_tbl,
     }
 }
 
+template <int NC>
+static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
+                               sid_t *out_tbl, const sid_t *d_fn, int col,
+                               bool fuse, sid_t fcval, const seg_t *fseg) {
+    hipLaunchKernelGGL(k_expand_fn<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
+                       0, e->stream, cur_tbl, d_fn, e->st->fn_base,
+                       e->st->fn_n, col, fuse ? 1 : 0, fcval, e->d_type_of,
+                       e->st->type_base, e->st->type_n, e->d_verts,
+                       e->d_edges, fseg ? fseg->bucket_start : 0,
+                       fseg ? fseg->num_buckets : 0, e->d_state, e->d_stats,
+                       out_tbl);
+}
+
+static void launch_expand_fn(wk_engine *e, const sid_t *cur_tbl,
+                             sid_t *out_tbl, const sid_t *d_fn, int col,
+                             bool fuse, sid_t fcval, const seg_t *fseg) {
+    switch (e->ncols) {
+    case 1: launch_expand_fn_t<1>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    case 2: launch_expand_fn_t<2>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    case 3: launch_expand_fn_t<3>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    case 4: launch_expand_fn_t<4>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    case 5: launch_expand_fn_t<5>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    case 6: launch_expand_fn_t<6>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    case 7: launch_expand_fn_t<7>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    default: launch_expand_fn_t<8>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    }
+}
+
 static void launch_expand(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
                           int G) {
     switch (e->ncols) {
@@ -1633,6 +1790,7 @@ static int32_t exec_pattern(wk_engine *e) {
                            e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
                            e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
                            off, sz, e->d_type_of, 0, 0, 0,
+                           (const sid_t *)nullptr, 0, 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
         hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
@@ -1682,12 +1840,18 @@ static int32_t exec_pattern(wk_engine *e) {
                           dir == DIR_OUT && key_mode == PK_NORMAL &&
                           e->d_type_of != nullptr)
                              ? 1 : 0;
+        // functional predicate: row's single object replaces the probe
+        const sid_t *d_fn = (!use_typeof && key_mode == PK_NORMAL &&
+                             e->gs && !e->gs->d_fn.empty())
+                                ? e->gs->d_fn[(size_t)p * 2 + dir]
+                                : nullptr;
         hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
                            e->stream, e->d_verts, e->d_edges, seg->bucket_start,
                            seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
                            dir, key_mode, pmode, col2, cval, 0, 0,
                            e->d_type_of, e->st->type_base, e->st->type_n,
-                           use_typeof, e->d_state, e->d_stats, out_tbl);
+                           use_typeof, d_fn, st->fn_base, st->fn_n,
+                           e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
     } else {
         // known_to_unknown: fused probe+scan -> cross-block scan ->
@@ -1695,6 +1859,46 @@ static int32_t exec_pattern(wk_engine *e) {
         const int G = scan_grid(e->bound);
         int oc = e->ncols + 1;
         if (oc > e->cap_cols) return WK_ERR_STATE;  // begin_query sizes cap_cols
+
+        // FUNCTIONAL predicate (deg==1 dense map): probe+scan+expand
+        // collapse to one gather + block-compacted append, with an
+        // optional fused typeof filter on the new column.  Output rows
+        // <= input rows, so no capacity risk; e->bound is unchanged.
+        const sid_t *d_fn = (key_mode == PK_NORMAL && e->gs &&
+                             !e->gs->d_fn.empty())
+                                ? e->gs->d_fn[(size_t)p * 2 + dir]
+                                : nullptr;
+        if (d_fn) {
+            bool fuse2 = false;
+            sid_t fcval2 = 0;
+            const seg_t *fseg2 = nullptr;
+            if (e->step + 1 < (int)e->pats.size() && e->d_type_of &&
+                e->st->nsrv == 1) {
+                const wk_pattern_t &nx = e->pats[e->step + 1];
+                if (nx.subject == o && nx.predicate == (ssid_t)TYPE_ID &&
+                    nx.direction == DIR_OUT && nx.object > 0) {
+                    fseg2 = st->seg_of((uint64_t)1 << NBITS_IDX, TYPE_ID,
+                                       DIR_OUT);
+                    if (fseg2 && fseg2->num_buckets) {
+                        fuse2 = true;
+                        fcval2 = (sid_t)nx.object;
+                    }
+                }
+            }
+            hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
+                               e->d_state + S_TOTAL, 1);
+            TIME_BEGIN(e);
+            launch_expand_fn(e, cur_tbl, out_tbl, d_fn, col, fuse2, fcval2,
+                             fseg2);
+            TIME_END(e, CAT_EXPAND);
+            hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
+                               e->d_state, (uint64_t)e->cap_rows);
+            e->v2c[-(o + 1)] = e->ncols;
+            e->ncols = oc;
+            e->cur ^= 1;
+            e->step += fuse2 ? 2 : 1;
+            return WK_OK;
+        }
 
         // look-ahead: a `?v rdf:type CONST` filter on THIS k2u's output
         // var fuses into the expansion (saves a full table pass)

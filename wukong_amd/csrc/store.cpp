@@ -423,6 +423,54 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
         }
     }
     WK_LOG("[store] versatile vp: %.1fs\n", now_s() - t0);
+    t0 = now_s();
+    // functional-predicate dense maps (keys == edges -> every key deg 1).
+    // Budgeted: largest segments first until WK_FN_BUDGET_GB (default 24)
+    // of host+HBM side-index space is spent.  WK_FN=0 disables.
+    {
+        const char *fv = getenv("WK_FN");
+        bool want = !(fv && !atoi(fv));
+        if (want && max_id >= (1u << NBITS_IDX)) {
+            st->fn_base = 1u << NBITS_IDX;
+            st->fn_n = (uint64_t)max_id + 1 - st->fn_base;
+            st->fn.assign((size_t)NP * 2, {});
+            const char *bv = getenv("WK_FN_BUDGET_GB");
+            uint64_t budget = (bv ? strtoull(bv, nullptr, 10) : 24) << 30;
+            std::vector<std::pair<uint64_t, uint32_t>> cand;  // (-edges, w)
+            for (uint32_t p = 1; p <= max_pid; p++) {
+                if (out_keys[p] && out_keys[p] == out_edges[p])
+                    cand.push_back({out_edges[p], p * 2 + DIR_OUT});
+                if (in_keys[p] && in_keys[p] == in_edges[p])
+                    cand.push_back({in_edges[p], p * 2 + DIR_IN});
+            }
+            std::sort(cand.begin(), cand.end(),
+                      [](auto &a, auto &b) { return a.first > b.first; });
+            std::vector<uint32_t> picked;
+            uint64_t spent = 0;
+            for (auto &c : cand) {
+                if (spent + st->fn_n * 4 > budget) break;
+                spent += st->fn_n * 4;
+                picked.push_back(c.second);
+            }
+#pragma omp parallel for schedule(dynamic)
+            for (size_t ci = 0; ci < picked.size(); ci++) {
+                uint32_t w = picked[ci];
+                uint32_t p = w / 2;
+                bool out = (w & 1) == DIR_OUT;
+                auto &m = st->fn[w];
+                m.assign(st->fn_n, 0);
+                const std::vector<triple> &arr = out ? pso : pos;
+                const std::vector<int64_t> &lo = out ? pso_lo : pos_lo;
+                for (int64_t i = lo[p]; i < lo[p + 1]; i++) {
+                    sid_t v = out ? arr[i].s : arr[i].o;
+                    if ((uint64_t)v < st->fn_base) continue;
+                    m[v - st->fn_base] = out ? arr[i].o : arr[i].s;
+                }
+            }
+            WK_LOG("[store] fn maps: %zu segments, %.1f GB (%.1fs)\n",
+                   picked.size(), spent / 1e9, now_s() - t0);
+        }
+    }
     st->ext_used = ext_next.load() - st->nbuckets_main;
     return st;
 }

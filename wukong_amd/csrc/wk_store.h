@@ -73,6 +73,16 @@ struct wk_store {
     pod_array<wk::sid_t> vp_edges[2];
     uint64_t vp_base = 0, vp_n = 0;
 
+    // functional-predicate dense maps: for a (pid,dir) segment where
+    // EVERY key has degree exactly 1 (LUBM: memberOf, ugDegreeFrom,
+    // worksFor, advisor, name, ... — detected as keys==edges), a dense
+    // vid→object array replaces the 148-byte cluster-hash probe with one
+    // 4-byte read (0 = vid has no such edge; 0 is never a valid id).
+    // Results identical; an HBM-capacity-funded densification like
+    // type_of.  Indexed [pid*2+dir]; empty = absent.
+    std::vector<std::vector<wk::sid_t>> fn;
+    uint64_t fn_base = 0, fn_n = 0;
+
     uint64_t nbuckets_main = 0, nbuckets_ext = 0, ext_used = 0;
 
     const wk::seg_t *seg_of(uint64_t vid, uint64_t pid, int dir) const {
