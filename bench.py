@@ -550,26 +550,42 @@ def main():
     # algorithmic bytes (device-counted) / HIP-event launch time
     def roofline_probe():
         if distributed:
-            return {}, {}
+            return {}, {}, {}
         if args.watdiv:
-            return {}, {}
-        e0 = engines[0]
-        e0.begin_query(Q.ALL["q1"])
-        e0.execute_one_pattern()               # i2u (6.4M grad students)
-        e0.execute_one_pattern()               # k2u memberOf
-        s0 = e0.kernel_stats()
-        e0.execute_one_pattern()               # k2u ugDegreeFrom (the largest
-        s1 = e0.kernel_stats()                 # expansion launch: 6.4M x 3 cols)
-        e0.fetch_count()
-        def delta(cat):
-            du = s1[cat]["usec"] - s0[cat]["usec"]
-            db = s1[cat]["bytes"] - s0[cat]["bytes"]
-            dn = s1[cat]["launches"] - s0[cat]["launches"]
-            return {"usec": round(du, 1), "bytes": db, "launches": dn,
-                    "gbs": round(db / du / 1e3, 1) if du > 0 else None}
-        return delta("expand"), delta("probe")
+            return {}, {}, {}
 
-    rl_expand, rl_probe = roofline_probe()
+        def run_steps(e0):
+            e0.begin_query(Q.ALL["q1"])
+            e0.execute_one_pattern()           # i2u (6.4M grad students)
+            e0.execute_one_pattern()           # k2u memberOf
+            s0 = e0.kernel_stats()
+            e0.execute_one_pattern()           # k2u ugDegreeFrom (the largest
+            s1 = e0.kernel_stats()             # expansion launch: 6.4M x 3 cols)
+            e0.fetch_count()
+            def delta(cat):
+                du = s1[cat]["usec"] - s0[cat]["usec"]
+                db = s1[cat]["bytes"] - s0[cat]["bytes"]
+                dn = s1[cat]["launches"] - s0[cat]["launches"]
+                return {"usec": round(du, 1), "bytes": db, "launches": dn,
+                        "gbs": round(db / du / 1e3, 1) if du > 0 else None}
+            return delta
+
+        e0 = engines[0]
+        # classic probe+scan+expand pipeline (the production path for
+        # every non-functional predicate: takesCourse, advisor, pub...)
+        os.environ["WK_FN_DISPATCH"] = "0"
+        try:
+            d = run_steps(e0)
+            rl_expand, rl_probe = d("expand"), d("probe")
+        finally:
+            os.environ.pop("WK_FN_DISPATCH", None)
+        # functional-predicate dense-map path (the production path for
+        # this specific step): one 4-byte gather replaces the probe
+        d2 = run_steps(e0)
+        rl_fn = d2("expand")
+        return rl_expand, rl_probe, rl_fn
+
+    rl_expand, rl_probe, rl_fn = roofline_probe()
     dk = "expand"
     d_us = stats1[dk]["usec"] - stats0[dk]["usec"]
     d_by = stats1[dk]["bytes"] - stats0[dk]["bytes"]
@@ -594,9 +610,15 @@ def main():
         "launches": int(d_n),
         "avg_launch_us": round(d_us / d_n, 2) if d_n else None,
         "measured_on": "Q1 known_to_unknown(ugDegreeFrom), 6.4M rows -> 3 "
-                       "cols, LUBM-2560 (largest expansion launch)"
+                       "cols, LUBM-2560 (largest expansion launch; classic "
+                       "probe+scan+expand pipeline, WK_FN_DISPATCH=0 — the "
+                       "production path for non-functional predicates)"
                        if rl_expand else "category aggregate",
         "probe_kernel": rl_probe or None,
+        # same step through the functional-predicate dense map (the
+        # default production path for this step): algorithmic bytes are
+        # 148->8 per row, so GB/s is lower while wall time is ~2x faster
+        "fn_map_kernel": rl_fn or None,
     }
 
     cb = None

@@ -1276,6 +1276,14 @@ static int scan_grid(int64_t bound) {
 
 static hipEvent_t ev_get() { hipEvent_t e; (void)hipEventCreate(&e); return e; }
 
+// WK_FN_DISPATCH=0: route around the functional-predicate dense maps at
+// dispatch time (diagnostic / roofline probe of the classic pipeline).
+// Read per call — bench.py flips it mid-process.
+static bool wk_fn_dispatch() {
+    const char *v = getenv("WK_FN_DISPATCH");
+    return !(v && !atoi(v));
+}
+
 // WK_SPIN_SYNC=1: poll instead of blocking (diagnostic)
 static bool wk_spin_sync() {
     static int v = -1;
@@ -1842,7 +1850,8 @@ static int32_t exec_pattern(wk_engine *e) {
                              ? 1 : 0;
         // functional predicate: row's single object replaces the probe
         const sid_t *d_fn = (!use_typeof && key_mode == PK_NORMAL &&
-                             e->gs && !e->gs->d_fn.empty())
+                             e->gs && !e->gs->d_fn.empty() &&
+                             wk_fn_dispatch())
                                 ? e->gs->d_fn[(size_t)p * 2 + dir]
                                 : nullptr;
         hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
@@ -1865,7 +1874,7 @@ static int32_t exec_pattern(wk_engine *e) {
         // optional fused typeof filter on the new column.  Output rows
         // <= input rows, so no capacity risk; e->bound is unchanged.
         const sid_t *d_fn = (key_mode == PK_NORMAL && e->gs &&
-                             !e->gs->d_fn.empty())
+                             !e->gs->d_fn.empty() && wk_fn_dispatch())
                                 ? e->gs->d_fn[(size_t)p * 2 + dir]
                                 : nullptr;
         if (d_fn) {
