@@ -83,7 +83,8 @@ enum { PK_NORMAL = 0, PK_INDEX = 1 };   // key = [vid|pid|dir] vs [0|val|dir]
 enum { PM_SIZE = 0,    // k2u: write edge count per row
        PM_CONST = 1,   // k2c: flag = (const in edge list)
        PM_COL = 2,     // k2k: flag = (row's other col in edge list)
-       PM_LIST = 3 };  // c2k/i2k: flag = (row col value in a FIXED list)
+       PM_LIST = 3,    // c2k/i2k: flag = (row col value in a FIXED list)
+       PM_EQ = 4 };    // k2c via reversed functional map: flag = (col == cval)
 
 // device query state (see engine): [0]=nrows [1]=scan total [2]=overflow
 // flag [3]=required rows; stats[0..6] = algorithmic bytes per category
@@ -284,7 +285,7 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                              uint64_t type_base, uint64_t type_n,
                              int use_typeof,
                              const sid_t *__restrict__ fn, uint64_t fn_base,
-                             uint64_t fn_n,
+                             uint64_t fn_n, int fn_swap,
                              uint64_t *__restrict__ d_state,
                              uint64_t *__restrict__ d_stats,
                              sid_t *__restrict__ out_tbl)
@@ -320,16 +321,24 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                 uint16_t t = (idx < type_n) ? type_of[idx] : 0;
                 if (t != 0xFFFF) { keep[k] = ((sid_t)t == cval); continue; }
             }
-            if (probe_mode == PM_LIST && !use_typeof) {
+            if (probe_mode == PM_EQ) {
+                // reversed functional map resolved host-side: the edge
+                // exists iff the row value IS the precomputed vertex
+                keep[k] = (v == cval);
+            } else if (probe_mode == PM_LIST && !use_typeof) {
                 keep[k] = bsearch_u32(edges + list_off, list_sz, v);
             } else if (fn) {
                 // functional predicate: the row's single object replaces
-                // the probe + edge-list search (4 bytes vs 148)
-                uint64_t idx = (uint64_t)v - fn_base;
+                // the probe + edge-list search (4 bytes vs 148).
+                // fn_swap: the REVERSED direction is functional — check
+                // fn[other col] == this col instead.
+                sid_t a = v, b;
+                if (probe_mode == PM_CONST) b = cval;
+                else b = tbl[r * ncols + col2];
+                if (fn_swap) { sid_t t_ = a; a = b; b = t_; }
+                uint64_t idx = (uint64_t)a - fn_base;
                 sid_t tv = (idx < fn_n) ? fn[idx] : 0;
-                sid_t tgt = (probe_mode == PM_CONST) ? cval
-                                                     : tbl[r * ncols + col2];
-                keep[k] = tv && tv == tgt;
+                keep[k] = tv && tv == b;
             } else {
                 uint64_t key = (key_mode == PK_NORMAL)
                                    ? key_pack(v, pid, (uint64_t)dir)
@@ -408,10 +417,16 @@ __global__ void k_commit(uint64_t *__restrict__ d_state, uint64_t cap) {
         t = cap;
     }
     d_state[S_NROWS] = t;
+    // reset the accumulators for the next step (saves a 3us k_zero_words
+    // launch per step — the state kernels were 12% of suite GPU time)
+    d_state[S_TOTAL] = 0;
+    d_state[S_OVF] = 0;
 }
 
 __global__ void k_set_state(uint64_t *__restrict__ d_state, uint64_t nrows) {
     d_state[S_NROWS] = nrows;
+    d_state[S_TOTAL] = 0;
+    d_state[S_OVF] = 0;
 }
 
 // device projection to required-var columns (sparql.hpp:1510-1536)
@@ -1729,8 +1744,6 @@ static int32_t exec_pattern(wk_engine *e) {
         const int end_mode = (ostat == 2) ? VU_END_CONST : VU_END_NEW;
         const int oc = e->ncols + (end_mode == VU_END_CONST ? 1 : 2);
         if (oc > e->cap_cols) return WK_ERR_STATE;
-        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                           e->d_state + S_TOTAL, 1);
         int G = (int)std::min<int64_t>(std::max<int64_t>(e->bound, 1), 4096);
         if (cstart) G = 1;
         TIME_BEGIN(e);
@@ -1791,14 +1804,12 @@ static int32_t exec_pattern(wk_engine *e) {
         uint64_t sz = 0;
         const sid_t *ptr = store_get(*st, (uint64_t)s, (uint64_t)p, dir, &sz);
         uint64_t off = ptr ? (uint64_t)(ptr - st->edges.data()) : 0;
-        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                           e->d_state + S_TOTAL, 1);
         TIME_BEGIN(e);
         hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
                            e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
                            e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
                            off, sz, e->d_type_of, 0, 0, 0,
-                           (const sid_t *)nullptr, 0, 0,
+                           (const sid_t *)nullptr, 0, 0, 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
         hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
@@ -1841,25 +1852,42 @@ static int32_t exec_pattern(wk_engine *e) {
     int col2 = (ostat == 1) ? e->var2col(o) : 0;
     sid_t cval = (ostat == 2) ? (sid_t)o : 0;
     if (pmode != PM_SIZE) {
-        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                           e->d_state + S_TOTAL, 1);
         TIME_BEGIN(e);
         int use_typeof = (pmode == PM_CONST && (sid_t)p == TYPE_ID &&
                           dir == DIR_OUT && key_mode == PK_NORMAL &&
                           e->d_type_of != nullptr)
                              ? 1 : 0;
-        // functional predicate: row's single object replaces the probe
-        const sid_t *d_fn = (!use_typeof && key_mode == PK_NORMAL &&
-                             e->gs && !e->gs->d_fn.empty() &&
-                             wk_fn_dispatch())
-                                ? e->gs->d_fn[(size_t)p * 2 + dir]
-                                : nullptr;
+        // functional predicate: row's single object replaces the probe.
+        // If only the REVERSED direction is functional, k2k checks
+        // fn[other col] == col (fn_swap); k2c resolves the single edge
+        // endpoint HOST-side and becomes a pure equality compare (PM_EQ).
+        const sid_t *d_fn = nullptr;
+        int fn_swap = 0;
+        if (!use_typeof && key_mode == PK_NORMAL && e->gs &&
+            !e->gs->d_fn.empty() && wk_fn_dispatch()) {
+            d_fn = e->gs->d_fn[(size_t)p * 2 + dir];
+            if (!d_fn && (pmode == PM_COL || pmode == PM_CONST)) {
+                const sid_t *rev = e->gs->d_fn[(size_t)p * 2 + (dir ^ 1)];
+                if (rev && pmode == PM_COL) {
+                    d_fn = rev;
+                    fn_swap = 1;
+                } else if (rev && pmode == PM_CONST) {
+                    // host lookup: the const's single neighbour
+                    uint64_t idx = (uint64_t)cval - st->fn_base;
+                    sid_t tv = (idx < st->fn_n)
+                                   ? st->fn[(size_t)p * 2 + (dir ^ 1)][idx]
+                                   : 0;
+                    pmode = PM_EQ;
+                    cval = tv;  // 0 never matches a vid -> empty result
+                }
+            }
+        }
         hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
                            e->stream, e->d_verts, e->d_edges, seg->bucket_start,
                            seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
                            dir, key_mode, pmode, col2, cval, 0, 0,
                            e->d_type_of, e->st->type_base, e->st->type_n,
-                           use_typeof, d_fn, st->fn_base, st->fn_n,
+                           use_typeof, d_fn, st->fn_base, st->fn_n, fn_swap,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
     } else {
@@ -1894,8 +1922,6 @@ static int32_t exec_pattern(wk_engine *e) {
                     }
                 }
             }
-            hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                               e->d_state + S_TOTAL, 1);
             TIME_BEGIN(e);
             launch_expand_fn(e, cur_tbl, out_tbl, d_fn, col, fuse2, fcval2,
                              fseg2);
@@ -1947,10 +1973,6 @@ static int32_t exec_pattern(wk_engine *e) {
                            (uint64_t *)e->bsums.p);
         TIME_END(e, CAT_PROBE);
         if (fuse) {
-            hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                               e->d_state + S_TOTAL, 1);
-            hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                               e->d_state + S_OVF, 1);
             {
                 TIME_BEGIN(e);
                 launch_expand_filter(e, cur_tbl, out_tbl, fcval, fseg);
@@ -1971,8 +1993,6 @@ static int32_t exec_pattern(wk_engine *e) {
                                (uint64_t *)e->bsums.p, G, e->d_state);
             TIME_END(e, CAT_SCAN);
         }
-        hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(64), 0, e->stream,
-                           e->d_state + S_OVF, 1);
         {
             TIME_BEGIN(e);
             launch_expand(e, cur_tbl, out_tbl, G);
