@@ -1866,7 +1866,11 @@ static int32_t exec_pattern(wk_engine *e) {
         if (!use_typeof && key_mode == PK_NORMAL && e->gs &&
             !e->gs->d_fn.empty() && wk_fn_dispatch()) {
             d_fn = e->gs->d_fn[(size_t)p * 2 + dir];
-            if (!d_fn && (pmode == PM_COL || pmode == PM_CONST)) {
+            // the REVERSED map keys the non-routed endpoint, which is
+            // only complete on a single-partition store (the forward map
+            // is partitioned on the same axis as the probe routing)
+            if (!d_fn && st->nsrv == 1 &&
+                (pmode == PM_COL || pmode == PM_CONST)) {
                 const sid_t *rev = e->gs->d_fn[(size_t)p * 2 + (dir ^ 1)];
                 if (rev && pmode == PM_COL) {
                     d_fn = rev;
