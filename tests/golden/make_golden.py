@@ -21,7 +21,9 @@ def main():
     ora = OracleCtx(triples)
     out = {"dataset": "lubm4", "seed": 42, "ntriples": int(triples.shape[0]),
            "queries": {}}
-    for name, plan in Q.ALL.items():
+    plans = dict(Q.ALL)
+    plans.update(Q.versatile_plans(ora))  # Q8-Q12 (VERSATILE)
+    for name, plan in plans.items():
         t = ora.run_query(plan)
         out["queries"][name] = {"rows": int(t.shape[0]), "cols": int(t.shape[1]),
                                 "sha": fnv1a_fast(t)}

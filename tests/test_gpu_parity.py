@@ -30,14 +30,16 @@ def test_query_parity_lubm4(name, eng4, oracle4):
     assert np.array_equal(sort_rows(got), sort_rows(want))
 
 
-def test_golden_fixture_gpu(eng4):
+def test_golden_fixture_gpu(eng4, store4):
     if not os.path.exists(GOLD):
         pytest.skip("golden fixture not generated")
     from tests.test_queries_cpu import fnv1a_fast
     with open(GOLD) as f:
         gold = json.load(f)
+    plans = dict(Q.ALL)
+    plans.update(Q.versatile_plans(store4))
     for name, rec in gold["queries"].items():
-        t = eng4.run_query(Q.ALL[name])
+        t = eng4.run_query(plans[name])
         assert t.shape[0] == rec["rows"], name
         assert fnv1a_fast(t) == rec["sha"], name
 
