@@ -48,6 +48,21 @@ typedef struct {
     int32_t   blind;              /* 1 = return row count only (Result::blind,
                                    * core/query.hpp:321 — the reference's
                                    * proxy/emulator benchmark mode) */
+    /* OPTIONAL pattern group (BGP-only, matching the reference's own
+     * limit — query.hpp:722-733; unmatched rows keep BLANK_ID in the
+     * optional-bound columns, sparql.hpp:100-170,316-375).  NULL/0 when
+     * absent. */
+    const wk_pattern_t *opt_patterns;
+    int32_t   nopt;
+    /* UNION branches (query.hpp:708-718; sparql.hpp:1593-1614): the
+     * branch pattern groups concatenated into one array, with
+     * union_sizes[i] patterns per branch.  Each branch continues from
+     * the main-group result table; the final result is the row concat
+     * of the branch results (rmap.hpp:57-87).  All branches must bind
+     * the same variable->column layout.  NULL/0 when absent. */
+    const wk_pattern_t *union_pats;
+    const int32_t *union_sizes;
+    int32_t   nunion;
 } wk_plan_t;
 
 /* A materialised binding table (SPARQLQuery::Result subset —

@@ -172,9 +172,23 @@ struct query {
     int step = 0;
     int mt_tid = 0, mt_factor = 1;  // sparql.hpp:210-221
 
+    // OPTIONAL group state (optional_matched_rows + optional_new_vars,
+    // query.hpp:722-813): rows are kept; unmatched rows carry BLANK_ID
+    // in the columns first bound inside the group.
+    int opt_mode = 0;
+    uint32_t opt_mask = 0;           // columns to blank on mismatch
+    std::vector<uint8_t> matched;
+
     int var2col(ssid_t v) const { return v < 0 ? v2c[-(v + 1)] : -1; }
     int64_t nrows() const { return col_num ? (int64_t)table.size() / col_num : 0; }
+
+    void blank_row(int64_t i) {      // correct_optional_result, query.hpp:805
+        for (int c = 0; c < col_num; c++)
+            if ((opt_mask >> c) & 1)
+                table[i * col_num + c] = 0xFFFFFFFFu;  // BLANK_ID
+    }
 };
+static const sid_t OBLANK = 0xFFFFFFFFu;
 
 // one pattern — dispatch per sparql.hpp:1016-1058
 static void exec_pattern(const ctx &c, query &q) {
@@ -253,16 +267,26 @@ static void exec_pattern(const ctx &c, query &q) {
             q.v2c[-(o + 1)] = q.col_num;
             q.col_num += 1;
         } else {
-            // const_to_known — sparql.hpp:138-186
+            // const_to_known — sparql.hpp:138-186 (+ OPTIONAL branch
+            // :160-170: keep rows, blank + clear flag on mismatch)
             std::unordered_set<sid_t> set(vids, vids + sz);
             int col = q.var2col(o);
-            std::vector<sid_t> out;
             int64_t R = q.nrows();
-            for (int64_t i = 0; i < R; i++)
-                if (set.count(q.table[i * q.col_num + col]))
-                    out.insert(out.end(), q.table.begin() + i * q.col_num,
-                               q.table.begin() + (i + 1) * q.col_num);
-            q.table.swap(out);
+            if (q.opt_mode) {
+                for (int64_t i = 0; i < R; i++) {
+                    if (!set.count(q.table[i * q.col_num + col])) {
+                        if (q.matched[i]) q.blank_row(i);
+                        q.matched[i] = 0;
+                    }
+                }
+            } else {
+                std::vector<sid_t> out;
+                for (int64_t i = 0; i < R; i++)
+                    if (set.count(q.table[i * q.col_num + col]))
+                        out.insert(out.end(), q.table.begin() + i * q.col_num,
+                                   q.table.begin() + (i + 1) * q.col_num);
+                q.table.swap(out);
+            }
         }
         q.step++;
         return;
@@ -278,7 +302,69 @@ static void exec_pattern(const ctx &c, query &q) {
     sz = 0;
     std::vector<sid_t> out;
 
-    if (ostat == 0) {
+    if (ostat == 0 && q.opt_mode) {
+        // known_to_unknown under OPTIONAL — sparql.hpp:316-375: skip
+        // unmatched/BLANK rows (keep + BLANK, flag unchanged); deg-0
+        // matched rows keep + BLANK with flag still true
+        std::vector<uint8_t> nm;
+        for (int64_t i = 0; i < R; i++) {
+            sid_t cur = q.table[i * q.col_num + col];
+            if (!q.matched[i] || cur == OBLANK) {
+                out.insert(out.end(), q.table.begin() + i * q.col_num,
+                           q.table.begin() + (i + 1) * q.col_num);
+                out.push_back(OBLANK);
+                nm.push_back(q.matched[i]);
+                continue;
+            }
+            if (cur != cached) {
+                cached = cur;
+                if ((sid_t)p == TYPE_ID && d == DIR_IN)
+                    vids = c.get(0, cur, d, &sz);
+                else
+                    vids = c.get(cur, (uint64_t)p, d, &sz);
+            }
+            if (sz > 0) {
+                for (uint64_t k = 0; k < sz; k++) {
+                    out.insert(out.end(), q.table.begin() + i * q.col_num,
+                               q.table.begin() + (i + 1) * q.col_num);
+                    out.push_back(vids[k]);
+                    nm.push_back(1);
+                }
+            } else {
+                out.insert(out.end(), q.table.begin() + i * q.col_num,
+                           q.table.begin() + (i + 1) * q.col_num);
+                out.push_back(OBLANK);
+                nm.push_back(1);
+            }
+        }
+        q.table.swap(out);
+        q.matched.swap(nm);
+        q.v2c[-(o + 1)] = q.col_num;
+        q.opt_mask |= 1u << q.col_num;
+        q.col_num += 1;
+    } else if ((ostat == 1 || ostat == 2) && q.opt_mode) {
+        // known_to_known / known_to_const under OPTIONAL: keep rows,
+        // blank + clear flag on mismatch (sparql.hpp:416-549 OPTIONAL)
+        for (int64_t i = 0; i < R; i++) {
+            sid_t cur = q.table[i * q.col_num + col];
+            bool ok = false;
+            if (cur != OBLANK) {
+                if (cur != cached) {
+                    cached = cur;
+                    vids = c.get(cur, (uint64_t)p, d, &sz);
+                }
+                sid_t tgt = (ostat == 2) ? (sid_t)o
+                                         : q.table[i * q.col_num + q.var2col(o)];
+                if (tgt != OBLANK)
+                    for (uint64_t k = 0; k < sz; k++)
+                        if (vids[k] == tgt) { ok = true; break; }
+            }
+            if (!ok) {
+                if (q.matched[i]) q.blank_row(i);
+                q.matched[i] = 0;
+            }
+        }
+    } else if (ostat == 0) {
         // known_to_unknown — sparql.hpp:295-407
         for (int64_t i = 0; i < R; i++) {
             sid_t cur = q.table[i * q.col_num + col];
@@ -451,6 +537,66 @@ int64_t ok_run_query(void *cv, const ok_pattern_t *pats, int32_t npat,
     *out_cols = q0.col_num;
     *out = (uint32_t *)malloc(std::max<size_t>(q0.table.size() * 4, 4));
     memcpy(*out, q0.table.data(), q0.table.size() * 4);
+    return R;
+}
+
+// Extended run: main BGP + UNION branches + OPTIONAL group
+// (execute_sparql_query order, sparql.hpp:1564-1662; union merge =
+// rmap.hpp:57-87 row concat; optional = matched-flag + BLANK fill).
+int64_t ok_run_query_ex(void *cv, const ok_pattern_t *pats, int32_t npat,
+                        int32_t nvars, const ok_pattern_t *opt, int32_t nopt,
+                        const ok_pattern_t *upats, const int32_t *usizes,
+                        int32_t nunion, const int32_t *req, int32_t nreq,
+                        int32_t distinct, int64_t limit, int64_t offset,
+                        uint32_t **out, int32_t *out_cols) {
+    ok::ctx *c = (ok::ctx *)cv;
+    ok::query q;
+    for (int i = 0; i < npat; i++)
+        q.pats.push_back({pats[i].subject, pats[i].predicate, pats[i].object,
+                          pats[i].direction});
+    q.nvars = nvars;
+    q.v2c.assign(nvars, -1);
+    while (q.step < (int)q.pats.size()) ok::exec_pattern(*c, q);
+
+    if (nunion > 0) {
+        std::vector<ok::sid_t> merged;
+        std::vector<int> bv2c;
+        int bc = -1;
+        const ok_pattern_t *bp = upats;
+        for (int b = 0; b < nunion; b++) {
+            ok::query qb = q;  // branch inherits the parent result
+            qb.pats.clear();
+            for (int i = 0; i < usizes[b]; i++)
+                qb.pats.push_back({bp[i].subject, bp[i].predicate,
+                                   bp[i].object, bp[i].direction});
+            bp += usizes[b];
+            qb.step = 0;
+            while (qb.step < (int)qb.pats.size()) ok::exec_pattern(*c, qb);
+            if (bc < 0) { bc = qb.col_num; bv2c = qb.v2c; }
+            merged.insert(merged.end(), qb.table.begin(), qb.table.end());
+        }
+        q.table.swap(merged);
+        q.col_num = bc;
+        q.v2c = bv2c;
+    }
+    if (nopt > 0) {
+        q.opt_mode = 1;
+        q.opt_mask = 0;
+        q.matched.assign(q.nrows(), 1);
+        q.pats.clear();
+        for (int i = 0; i < nopt; i++)
+            q.pats.push_back({opt[i].subject, opt[i].predicate, opt[i].object,
+                              opt[i].direction});
+        q.step = 0;
+        while (q.step < (int)q.pats.size()) ok::exec_pattern(*c, q);
+        q.opt_mode = 0;
+    }
+
+    ok::final_process(q, req, nreq, distinct, limit, offset);
+    int64_t R = q.nrows();
+    *out_cols = q.col_num;
+    *out = (uint32_t *)malloc(std::max<size_t>(q.table.size() * 4, 4));
+    memcpy(*out, q.table.data(), q.table.size() * 4);
     return R;
 }
 

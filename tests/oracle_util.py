@@ -110,12 +110,44 @@ class OracleCtx:
         req = (ctypes.c_int32 * len(plan.required_vars))(*plan.required_vars)
         out = ctypes.POINTER(ctypes.c_uint32)()
         cols = ctypes.c_int32()
-        n = lib().ok_run_query(self._h, ctypes.cast(pats, ctypes.c_void_p),
-                               len(plan.patterns), plan.nvars, req,
-                               len(plan.required_vars),
-                               1 if plan.distinct else 0, plan.limit,
-                               plan.offset, mt, ctypes.byref(out),
-                               ctypes.byref(cols))
+        optional = getattr(plan, "optional", [])
+        unions = getattr(plan, "unions", [])
+        if optional or unions:
+            class _P(ctypes.Structure):
+                _fields_ = [("s", ctypes.c_int32), ("p", ctypes.c_int32),
+                            ("o", ctypes.c_int32), ("d", ctypes.c_int32)]
+            def mk(pl):
+                a = (_P * max(len(pl), 1))()
+                for i, (s, p, d, o) in enumerate(pl):
+                    a[i] = _P(s, p, o, d)
+                return a
+            opt = mk(optional)
+            flat = [p for u in unions for p in u]
+            up = mk(flat)
+            us = (ctypes.c_int32 * max(len(unions), 1))(*[len(u) for u in unions])
+            i32, i64, vp = ctypes.c_int32, ctypes.c_int64, ctypes.c_void_p
+            u32p = ctypes.POINTER(ctypes.c_uint32)
+            f = lib().ok_run_query_ex
+            f.restype = i64
+            f.argtypes = [vp, vp, i32, i32, vp, i32, vp,
+                          ctypes.POINTER(i32), i32, ctypes.POINTER(i32), i32,
+                          i32, i64, i64, ctypes.POINTER(u32p),
+                          ctypes.POINTER(i32)]
+            n = f(self._h, ctypes.cast(pats, vp),
+                  len(plan.patterns), plan.nvars,
+                  ctypes.cast(opt, vp), len(optional),
+                  ctypes.cast(up, vp), us, len(unions),
+                  req, len(plan.required_vars),
+                  1 if plan.distinct else 0,
+                  plan.limit, plan.offset,
+                  ctypes.byref(out), ctypes.byref(cols))
+        else:
+            n = lib().ok_run_query(self._h, ctypes.cast(pats, ctypes.c_void_p),
+                                   len(plan.patterns), plan.nvars, req,
+                                   len(plan.required_vars),
+                                   1 if plan.distinct else 0, plan.limit,
+                                   plan.offset, mt, ctypes.byref(out),
+                                   ctypes.byref(cols))
         if n and cols.value:
             tbl = np.ctypeslib.as_array(out, shape=(n, cols.value)).copy()
         else:

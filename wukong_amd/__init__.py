@@ -34,7 +34,13 @@ class WkPlan(ctypes.Structure):
                 ("distinct", ctypes.c_int32),
                 ("limit", ctypes.c_int64),
                 ("offset", ctypes.c_int64),
-                ("blind", ctypes.c_int32)]
+                ("blind", ctypes.c_int32),
+                # OPTIONAL group + UNION branches (wukong_abi.h)
+                ("opt_patterns", ctypes.POINTER(WkPattern)),
+                ("nopt", ctypes.c_int32),
+                ("union_pats", ctypes.POINTER(WkPattern)),
+                ("union_sizes", ctypes.POINTER(ctypes.c_int32)),
+                ("nunion", ctypes.c_int32)]
 
 
 class WkResult(ctypes.Structure):
@@ -164,7 +170,7 @@ class Plan:
     """A planner-ordered pattern list (vars negative: -1..-nvars)."""
 
     def __init__(self, patterns, nvars, required_vars, distinct=False,
-                 limit=-1, offset=0, blind=False):
+                 limit=-1, offset=0, blind=False, optional=None, unions=None):
         self.patterns = list(patterns)
         self.nvars = nvars
         self.required_vars = list(required_vars)
@@ -172,18 +178,37 @@ class Plan:
         self.limit = limit
         self.offset = offset
         self.blind = blind
+        self.optional = list(optional or [])   # OPTIONAL pattern group
+        self.unions = [list(u) for u in (unions or [])]  # UNION branches
 
     def to_c(self):
-        pats = (WkPattern * len(self.patterns))()
-        for i, (s, p, d, o) in enumerate(self.patterns):
-            pats[i] = WkPattern(s, p, o, d)  # note arg order (s,p,d,o) -> struct
+        def mk(pat_list):
+            arr = (WkPattern * max(len(pat_list), 1))()
+            for i, (s, p, d, o) in enumerate(pat_list):
+                arr[i] = WkPattern(s, p, o, d)  # arg order (s,p,d,o) -> struct
+            return arr
+
+        pats = mk(self.patterns)
         req = (ctypes.c_int32 * len(self.required_vars))(*self.required_vars)
         plan = WkPlan(ctypes.cast(pats, ctypes.POINTER(WkPattern)),
                       len(self.patterns), self.nvars,
                       ctypes.cast(req, ctypes.POINTER(ctypes.c_int32)),
                       len(self.required_vars), 1 if self.distinct else 0,
                       self.limit, self.offset, 1 if self.blind else 0)
-        plan._keepalive = (pats, req)
+        plan._keepalive = [pats, req]
+        if self.optional:
+            opt = mk(self.optional)
+            plan.opt_patterns = ctypes.cast(opt, ctypes.POINTER(WkPattern))
+            plan.nopt = len(self.optional)
+            plan._keepalive.append(opt)
+        if self.unions:
+            flat = [p for u in self.unions for p in u]
+            up = mk(flat)
+            us = (ctypes.c_int32 * len(self.unions))(*[len(u) for u in self.unions])
+            plan.union_pats = ctypes.cast(up, ctypes.POINTER(WkPattern))
+            plan.union_sizes = ctypes.cast(us, ctypes.POINTER(ctypes.c_int32))
+            plan.nunion = len(self.unions)
+            plan._keepalive.extend((up, us))
         return plan
 
 

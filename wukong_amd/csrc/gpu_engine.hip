@@ -1171,6 +1171,104 @@ __global__ void k_vu(const vertex_t *__restrict__ verts,
                   (unsigned long long)bytes);
 }
 
+// ---------------------------------------------------------------------
+// OPTIONAL-mode operators (sparql.hpp:100-170, 316-375; BGP-only like
+// the reference, query.hpp:722-733): rows are never dropped — a row
+// whose pattern fails keeps BLANK_ID in every column first bound inside
+// the OPTIONAL group (blank_mask) and clears its matched flag.  This is
+// a correctness/coverage path (no LUBM/WatDiv benchmark uses OPTIONAL),
+// so the expansion uses a plain global append cursor.
+// ---------------------------------------------------------------------
+__global__ void k_filter_opt(const vertex_t *__restrict__ verts,
+                             const sid_t *__restrict__ edges,
+                             uint64_t bucket_start, uint64_t num_buckets,
+                             sid_t *__restrict__ tbl, int ncols, int col,
+                             uint32_t pid, int dir, int probe_mode, int col2,
+                             sid_t cval, uint64_t list_off, uint64_t list_sz,
+                             uint32_t blank_mask, uint8_t *__restrict__ flags,
+                             uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows * (4 + 128 + 8 + 64));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        sid_t v = tbl[r * ncols + col];
+        bool ok;
+        if (probe_mode == PM_LIST) {
+            ok = bsearch_u32(edges + list_off, list_sz, v);
+        } else {
+            uint64_t eoff = 0, esz = 0;
+            if (v != BLANK_ID && num_buckets)
+                probe_one(verts, bucket_start, num_buckets,
+                          key_pack((uint64_t)v, (uint64_t)pid, (uint64_t)dir),
+                          eoff, esz);
+            sid_t tgt = (probe_mode == PM_CONST) ? cval
+                                                 : tbl[r * ncols + col2];
+            ok = esz && tgt != BLANK_ID && bsearch_u32(edges + eoff, esz, tgt);
+        }
+        if (!ok) {
+            if (flags[r]) {
+                for (int c = 0; c < ncols; c++)
+                    if ((blank_mask >> c) & 1)
+                        tbl[r * ncols + c] = BLANK_ID;
+            }
+            flags[r] = 0;
+        }
+    }
+}
+
+__global__ void k_expand_opt(const vertex_t *__restrict__ verts,
+                             const sid_t *__restrict__ edges,
+                             const sid_t *__restrict__ tbl, int ncols,
+                             int col, uint32_t pid, int dir, int key_mode,
+                             uint64_t bucket_start, uint64_t num_buckets,
+                             const uint8_t *__restrict__ flags_in,
+                             uint8_t *__restrict__ flags_out,
+                             sid_t *__restrict__ out, uint64_t cap,
+                             uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int oc = ncols + 1;
+    count_bytes(d_stats, CAT_EXPAND, (uint64_t)nrows * (4 + 128 + 8));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        sid_t v = tbl[r * ncols + col];
+        uint8_t m = flags_in[r];
+        uint64_t eoff = 0, esz = 0;
+        if (m && v != BLANK_ID && num_buckets) {
+            uint64_t key = (key_mode == PK_NORMAL)
+                               ? key_pack((uint64_t)v, (uint64_t)pid,
+                                          (uint64_t)dir)
+                               : key_pack(0, (uint64_t)v, (uint64_t)dir);
+            probe_one(verts, bucket_start, num_buckets, key, eoff, esz);
+        }
+        if (!m || v == BLANK_ID || esz == 0) {
+            // unmatched / blank / no edges: keep the row, BLANK new col;
+            // the matched flag survives unchanged (sparql.hpp:328-334,
+            // 352-356)
+            uint64_t pos =
+                atomicAdd((unsigned long long *)&d_state[S_TOTAL], 1ull);
+            if (pos < cap) {
+                sid_t *dst = out + pos * oc;
+                for (int c = 0; c < ncols; c++) dst[c] = tbl[r * ncols + c];
+                dst[ncols] = BLANK_ID;
+                flags_out[pos] = m;
+            }
+        } else {
+            uint64_t pos = atomicAdd((unsigned long long *)&d_state[S_TOTAL],
+                                     (unsigned long long)esz);
+            for (uint64_t k = 0; k < esz && pos + k < cap; k++) {
+                sid_t *dst = out + (pos + k) * oc;
+                for (int c = 0; c < ncols; c++) dst[c] = tbl[r * ncols + c];
+                dst[ncols] = edges[eoff + k];
+                flags_out[pos + k] = 1;
+            }
+        }
+    }
+}
+
 __global__ void k_zero_words(uint64_t *p, int n) {
     for (int i = threadIdx.x; i < n; i += blockDim.x) p[i] = 0;
 }
@@ -1253,6 +1351,12 @@ struct wk_engine {
     int64_t bound = 0;   // host-side upper bound on rows (grid sizing)
 
     bool light = false;  // single-kernel fast path ran (h_pin self-published)
+
+    // OPTIONAL group execution state (wk_engine_run_query only)
+    devbuf oflag[2];          // per-row matched flags (u8), ping-pong
+    int opt_mode = 0;         // inside an OPTIONAL pattern group
+    int ocur = 0;             // which oflag is current
+    uint32_t opt_mask = 0;    // columns first bound inside the group
 
     // batched light-query window (wk_engine_submit_light_batch)
     devbuf lbd;               // device light_desc array
@@ -1501,6 +1605,7 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
     for (int i = 0; i < 2; i++) e->tbl[i].release();
     e->eoff.release(); e->cnt.release(); e->prefix.release();
     e->bsums.release(); e->misc.release(); e->ovf.release();
+    e->oflag[0].release(); e->oflag[1].release();
     e->lbd.release(); e->lbcnt.release();
     if (e->h_lb) (void)hipHostFree(e->h_lb);
     if (e->d_state) (void)hipFree(e->d_state);
@@ -1518,9 +1623,13 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
 }
 
 extern "C" int32_t wk_engine_begin_query(wk_engine_t *e, const wk_plan_t *plan) {
-    if (!e || !plan || plan->npatterns <= 0 || plan->nvars <= 0) return WK_ERR_PLAN;
+    if (!e || !plan || plan->nvars <= 0) return WK_ERR_PLAN;
+    if (plan->npatterns <= 0 && plan->nunion <= 0) return WK_ERR_PLAN;
     if (plan->nvars > 8) return WK_ERR_PLAN;  // col count cap (templated kernels)
-    e->pats.assign(plan->patterns, plan->patterns + plan->npatterns);
+    if (plan->npatterns > 0)
+        e->pats.assign(plan->patterns, plan->patterns + plan->npatterns);
+    else
+        e->pats.clear();
     e->v2c.assign(plan->nvars, -1);
     e->nvars = plan->nvars;
     e->step = 0;
@@ -1723,6 +1832,78 @@ static int32_t exec_pattern(wk_engine *e) {
     const int dir = pat.direction;
     sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
     sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
+
+    // ---- OPTIONAL-mode dispatch (sparql.hpp:100-170,316-375) ----
+    if (e->opt_mode) {
+        if (p < 0) return WK_ERR_PLAN;  // BGP-only, like the reference
+        const bool cstart = s >= 0;
+        if (cstart && is_tpid(s)) return WK_ERR_PLAN;  // no index starts
+        int ocol = cstart ? e->var2col(o) : e->var2col(s);
+        if (ocol < 0 && cstart) return WK_ERR_PLAN;  // c2u inside OPTIONAL
+        if (!cstart && e->var2col(s) < 0) return WK_ERR_PLAN;
+        const int ostat = (o >= 0) ? 2 : (e->var2col(o) >= 0 ? 1 : 0);
+        const seg_t *oseg = st->seg_of((uint64_t)1 << NBITS_IDX,
+                                       (uint64_t)p, dir);
+        int okey_mode = PK_NORMAL;
+        if (!cstart && ostat == 0 && (sid_t)p == TYPE_ID && dir == DIR_IN) {
+            okey_mode = PK_INDEX;
+            oseg = &st->iseg[DIR_IN];
+        }
+        uint64_t bs = oseg ? oseg->bucket_start : 0;
+        uint64_t nb = oseg ? oseg->num_buckets : 0;
+        if (cstart || ostat != 0) {
+            // filter-style: in place, flags transition, no compaction
+            uint64_t loff = 0, lsz = 0;
+            int pm;
+            int fcol, fcol2 = 0;
+            sid_t fcval = 0;
+            if (cstart) {  // const_to_known (sparql.hpp:138-186, OPTIONAL)
+                const sid_t *ptr =
+                    store_get(*st, (uint64_t)s, (uint64_t)p, dir, &lsz);
+                loff = ptr ? (uint64_t)(ptr - st->edges.data()) : 0;
+                pm = PM_LIST;
+                fcol = ocol;
+            } else {
+                pm = (ostat == 2) ? PM_CONST : PM_COL;
+                fcol = e->var2col(s);
+                fcol2 = (ostat == 1) ? e->var2col(o) : 0;
+                fcval = (ostat == 2) ? (sid_t)o : 0;
+            }
+            TIME_BEGIN(e);
+            hipLaunchKernelGGL(k_filter_opt, dim3(grid_for(e->bound)),
+                               dim3(BLOCK), 0, e->stream, e->d_verts,
+                               e->d_edges, bs, nb, cur_tbl, e->ncols, fcol,
+                               (uint32_t)p, dir, pm, fcol2, fcval, loff, lsz,
+                               e->opt_mask, (uint8_t *)e->oflag[e->ocur].p,
+                               e->d_state, e->d_stats);
+            TIME_END(e, CAT_FILTER);
+            // nrows unchanged; no commit (S_NROWS stays), no table flip
+            e->step++;
+            return WK_OK;
+        }
+        // known_to_unknown inside OPTIONAL: append column, keep rows
+        const int oc = e->ncols + 1;
+        if (oc > e->cap_cols) return WK_ERR_STATE;
+        TIME_BEGIN(e);
+        hipLaunchKernelGGL(k_expand_opt, dim3(grid_for(e->bound)),
+                           dim3(BLOCK), 0, e->stream, e->d_verts, e->d_edges,
+                           cur_tbl, e->ncols, e->var2col(s), (uint32_t)p, dir,
+                           okey_mode, bs, nb,
+                           (const uint8_t *)e->oflag[e->ocur].p,
+                           (uint8_t *)e->oflag[e->ocur ^ 1].p, out_tbl,
+                           (uint64_t)e->cap_rows, e->d_state, e->d_stats);
+        TIME_END(e, CAT_EXPAND);
+        hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
+                           e->d_state, (uint64_t)e->cap_rows);
+        e->v2c[-(o + 1)] = e->ncols;
+        e->opt_mask |= 1u << e->ncols;
+        e->ncols = oc;
+        e->cur ^= 1;
+        e->ocur ^= 1;
+        e->bound = e->cap_rows;
+        e->step++;
+        return WK_OK;
+    }
 
     // ---- VERSATILE: predicate variable (sparql.hpp:556-744) ----
     if (p < 0) {
@@ -2286,6 +2467,8 @@ static bool light2_eligible(wk_engine *e, const wk_plan_t *plan) {
 
 extern "C" int32_t wk_engine_submit(wk_engine_t *e, const wk_plan_t *plan) {
     if (!e || !plan) return WK_ERR_STATE;
+    if (plan->nopt > 0 || plan->nunion > 0)
+        return WK_ERR_PLAN;  // UNION/OPTIONAL need the sync run_query path
     const bool light = light2_eligible(e, plan);
     int32_t rc = light ? begin_light(e, plan) : wk_engine_begin_query(e, plan);
     if (rc) return rc;
@@ -2507,6 +2690,103 @@ extern "C" int32_t wk_engine_light_batch_wait(wk_engine_t *e,
     return WK_OK;
 }
 
+// UNION branches (sparql.hpp:1593-1614 + rmap.hpp:57-87): each branch
+// continues from the main group's result table; the final table is the
+// row-concat of the branch results.  All branches must end with the
+// same variable->column layout.
+static int32_t run_union(wk_engine *e, const wk_plan_t *plan) {
+    int32_t rc = sync_state(e);
+    if (rc) return rc;
+    const int64_t pr = e->nrows;
+    const int pc = e->ncols;
+    const std::vector<int32_t> pv2c = e->v2c;
+    const int pcur = e->cur;
+    devbuf snap;
+    size_t bytes = (size_t)pr * std::max(pc, 1) * 4;
+    if (bytes && pc) {
+        if (snap.ensure(bytes)) return WK_ERR_HIP;
+        HIP_CHECK(hipMemcpyAsync(snap.p, e->tbl[pcur].p, bytes,
+                                 hipMemcpyDeviceToDevice, e->stream));
+    }
+    std::vector<sid_t> merged;
+    std::vector<int32_t> bv2c;
+    int bc = -1;
+    const wk_pattern_t *bp = plan->union_pats;
+    for (int b = 0; b < plan->nunion; b++) {
+        if (b) {  // restore the parent state for the next branch
+            e->v2c = pv2c;
+            e->ncols = pc;
+            e->cur = pcur;
+            if (bytes && pc)
+                HIP_CHECK(hipMemcpyAsync(e->tbl[pcur].p, snap.p, bytes,
+                                         hipMemcpyDeviceToDevice, e->stream));
+            hipLaunchKernelGGL(k_set_state, dim3(1), dim3(1), 0, e->stream,
+                               e->d_state, (uint64_t)pr);
+            e->nrows = pr;
+            e->bound = pr;
+        }
+        e->pats.assign(bp, bp + plan->union_sizes[b]);
+        e->step = 0;
+        bp += plan->union_sizes[b];
+        while (e->step < (int)e->pats.size()) {
+            rc = exec_pattern(e);
+            if (rc) { snap.release(); return rc; }
+        }
+        rc = sync_state_grow(e);
+        if (rc) { snap.release(); return rc; }
+        if (bc < 0) {
+            bc = e->ncols;
+            bv2c = e->v2c;
+        } else if (bc != e->ncols || bv2c != e->v2c) {
+            snap.release();
+            return WK_ERR_PLAN;
+        }
+        size_t off = merged.size();
+        merged.resize(off + (size_t)e->nrows * bc);
+        if (e->nrows)
+            HIP_CHECK(hipMemcpy(merged.data() + off, e->tbl[e->cur].p,
+                                (size_t)e->nrows * bc * 4,
+                                hipMemcpyDeviceToHost));
+    }
+    snap.release();
+    int64_t mrows = bc > 0 ? (int64_t)(merged.size() / bc) : 0;
+    rc = wk_engine_load_rbuf(e, merged.empty() ? nullptr : merged.data(),
+                             mrows, bc, bv2c.data(), 0);
+    if (rc) return rc;
+    HIP_CHECK(stream_sync(e->stream));  // merged[] is about to go away
+    return WK_OK;
+}
+
+// OPTIONAL pattern group (sparql.hpp:1616-1649, BGP-only): rows are
+// kept; unmatched rows carry BLANK_ID in the columns first bound inside
+// the group.
+static int32_t run_optional(wk_engine *e, const wk_plan_t *plan) {
+    int32_t rc = sync_state(e);
+    if (rc) return rc;
+    if (e->oflag[0].ensure((size_t)e->cap_rows) ||
+        e->oflag[1].ensure((size_t)e->cap_rows))
+        return WK_ERR_HIP;
+    HIP_CHECK(hipMemsetAsync(e->oflag[0].p, 1, (size_t)e->cap_rows,
+                             e->stream));
+    HIP_CHECK(hipMemsetAsync(e->oflag[1].p, 1, (size_t)e->cap_rows,
+                             e->stream));
+    e->opt_mode = 1;
+    e->ocur = 0;
+    e->opt_mask = 0;
+    e->pats.assign(plan->opt_patterns, plan->opt_patterns + plan->nopt);
+    e->step = 0;
+    while (e->step < (int)e->pats.size()) {
+        rc = exec_pattern(e);
+        if (!rc) rc = sync_state_grow(e);
+        if (rc) {
+            e->opt_mode = 0;
+            return rc;
+        }
+    }
+    e->opt_mode = 0;
+    return WK_OK;
+}
+
 extern "C" int32_t wk_engine_run_query(wk_engine_t *e, const wk_plan_t *plan,
                                        wk_result_t *out) {
     for (int attempt = 0; attempt < 6; attempt++) {
@@ -2514,6 +2794,18 @@ extern "C" int32_t wk_engine_run_query(wk_engine_t *e, const wk_plan_t *plan,
         if (rc) return rc;
         while (e->step < (int)e->pats.size()) {
             rc = exec_pattern(e);
+            if (rc) return rc;
+        }
+        // UNION then OPTIONAL, the reference's order
+        // (execute_sparql_query, sparql.hpp:1564-1662)
+        if (plan->nunion > 0 && plan->union_pats && plan->union_sizes) {
+            rc = run_union(e, plan);
+            if (rc == WK_ERR_CAP) continue;
+            if (rc) return rc;
+        }
+        if (plan->nopt > 0 && plan->opt_patterns) {
+            rc = run_optional(e, plan);
+            if (rc == WK_ERR_CAP) continue;
             if (rc) return rc;
         }
         rc = wk_engine_fetch_result(e, plan, out);
