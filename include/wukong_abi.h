@@ -112,6 +112,11 @@ const wk_sid_t *wk_store_get_index(const wk_store_t *, wk_sid_t pid,
  * idea, core/store/gchecker.hpp:364-392). */
 uint64_t wk_store_num_slots(const wk_store_t *);
 uint64_t wk_store_num_edges(const wk_store_t *);
+/* Per-(pid,dir) segment statistics — the planner's cost-model inputs
+ * (the reference's type-centric stats, core/optimizer/stats.hpp):
+ * distinct keys and total edges of the segment. */
+int32_t wk_store_seg_stats(const wk_store_t *, uint32_t pid, int32_t dir,
+                           uint64_t *keys, uint64_t *edges);
 uint64_t wk_store_checksum(const wk_store_t *); /* FNV over vertices+edges */
 /* Full-store integrity scan (the `gsck` command, core/store/
  * gchecker.hpp:364-392): 0 = consistent, else #violations. */

@@ -71,6 +71,9 @@ _get_index = _sig("wk_store_get_index", c_u32p, [c_vp, ctypes.c_uint32, c_i32, c
 _num_slots = _sig("wk_store_num_slots", c_u64, [c_vp])
 _num_edges = _sig("wk_store_num_edges", c_u64, [c_vp])
 _checksum = _sig("wk_store_checksum", c_u64, [c_vp])
+_seg_stats = _sig("wk_store_seg_stats", c_i32,
+                  [c_vp, ctypes.c_uint32, c_i32, ctypes.POINTER(c_u64),
+                   ctypes.POINTER(c_u64)])
 _store_check = _sig("wk_store_check", c_u64, [c_vp])
 _eng_create = _sig("wk_engine_create", c_vp, [c_vp, c_i32])
 _gstore_create = _sig("wk_gpu_store_create", c_vp, [c_vp, c_i32])
@@ -250,6 +253,13 @@ class Store:
 
     def checksum(self):
         return _checksum(self._h)
+
+    def seg_stats(self, pid, direction):
+        """(distinct keys, total edges) of the (pid,dir) segment — the
+        planner's cost-model inputs (reference stats.hpp)."""
+        k, e = c_u64(), c_u64()
+        _seg_stats(self._h, pid, direction, ctypes.byref(k), ctypes.byref(e))
+        return int(k.value), int(e.value)
 
     def check(self):
         """gsck-style full integrity scan; returns #violations (0 = ok)."""

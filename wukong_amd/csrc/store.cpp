@@ -214,6 +214,14 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
         }
     }
 
+    st->seg_keys.assign((size_t)NP * 2, 0);
+    st->seg_edges.assign((size_t)NP * 2, 0);
+    for (uint32_t p = 0; p <= max_pid; p++) {
+        st->seg_keys[p * 2 + DIR_OUT] = out_keys[p];
+        st->seg_edges[p * 2 + DIR_OUT] = out_edges[p];
+        st->seg_keys[p * 2 + DIR_IN] = in_keys[p];
+        st->seg_edges[p * 2 + DIR_IN] = in_edges[p];
+    }
     WK_LOG("[store] seg counts: %.1fs\n", now_s() - t0); t0 = now_s();
     // index segment contents (gstore.hpp:858-888):
     //   [0|pid|IN]  = subjects of pid (from OUT keys, pid != TYPE_ID)
@@ -485,6 +493,23 @@ extern "C" const sid_t *wk_store_get_triples(const wk_store_t *st, sid_t vid,
 extern "C" const sid_t *wk_store_get_index(const wk_store_t *st, sid_t pid,
                                            int32_t dir, uint64_t *sz) {
     return store_get(*st, 0, pid, dir, sz);
+}
+
+// per-(pid,dir) key/edge counts — the planner's cost-model inputs
+// (the reference's type-centric statistics, core/optimizer/stats.hpp)
+extern "C" int32_t wk_store_seg_stats(const wk_store_t *st, uint32_t pid,
+                                      int32_t dir, uint64_t *keys,
+                                      uint64_t *edges) {
+    if (!st || dir < 0 || dir > 1) return -1;
+    size_t w = (size_t)pid * 2 + dir;
+    if (w >= st->seg_keys.size()) {
+        if (keys) *keys = 0;
+        if (edges) *edges = 0;
+        return 0;
+    }
+    if (keys) *keys = st->seg_keys[w];
+    if (edges) *edges = st->seg_edges[w];
+    return 0;
 }
 
 extern "C" uint64_t wk_store_num_slots(const wk_store_t *st) {

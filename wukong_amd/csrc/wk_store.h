@@ -83,6 +83,10 @@ struct wk_store {
     std::vector<std::vector<wk::sid_t>> fn;
     uint64_t fn_base = 0, fn_n = 0;
 
+    // per-(pid,dir) key/edge counts — the planner's cost-model inputs
+    // (the reference's type-centric stats, core/optimizer/stats.hpp)
+    std::vector<uint64_t> seg_keys, seg_edges;  // [pid*2+dir]
+
     uint64_t nbuckets_main = 0, nbuckets_ext = 0, ext_used = 0;
 
     const wk::seg_t *seg_of(uint64_t vid, uint64_t pid, int dir) const {

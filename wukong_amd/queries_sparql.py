@@ -64,3 +64,31 @@ PLAN = {
     "q6": ["1 <", "2 >", "3 <", "4 >"],
     "q7": ["1 <", "2 <", "3 >", "4 >", "5 >", "6 <"],
 }
+
+
+# Q8-Q12 — the VERSATILE (predicate-variable) queries
+# (scripts/sparql_query/lubm/basic/lubm_q{8..12}).  No OSDI16 plans
+# exist for these; they go through the greedy planner
+# (wukong_amd.planner).  Entity URIs resolve positionally through
+# sparql.lubm_entity_vocab(store).
+TEXT_VERSATILE = {
+    "q8": _P + """SELECT ?X ?P ?D WHERE {
+        ?D ub:subOrganizationOf <http://www.University0.edu> .
+        ?X ?P ?D .
+    }""",
+    "q9": _P + """SELECT ?X ?P WHERE {
+        ?X ?P <http://www.Department0.University0.edu> .
+    }""",
+    "q10": _P + """SELECT ?X ?Y WHERE {
+        ?X rdf:type ub:UndergraduateStudent .
+        ?X ?Y <http://www.Department3.University0.edu/FullProfessor1> .
+    }""",
+    "q11": _P + """SELECT ?X WHERE {
+        <http://www.Department7.University0.edu/UndergraduateStudent201> ?X <http://www.Department7.University0.edu> .
+    }""",
+    "q12": _P + """SELECT ?X ?Y WHERE {
+        ?Y ub:subOrganizationOf <http://www.University0.edu> .
+        ?X ub:worksFor ?Y .
+        ?Z ub:advisor ?X .
+    }""",
+}

@@ -99,6 +99,11 @@ def parse(text, vocab, plan_lines=None, distinct=None, limit=None,
             return var_id(tok)
         if tok in vocab:
             return vocab[tok]
+        resolver = vocab.get("__resolver__")
+        if resolver is not None:
+            v = resolver(tok)
+            if v is not None:
+                return v
         if ":" in tok and tok.split(":")[0] + ":" in prefixes:
             raise ParseError(f"unknown vocabulary term {tok!r}")
         if tok.startswith("<") and tok.endswith(">"):
@@ -164,3 +169,61 @@ def parse(text, vocab, plan_lines=None, distinct=None, limit=None,
     req = [vars_[v] for v in sel]
     return Plan(patterns, nvars=len(vars_), required_vars=req,
                 distinct=distinct, limit=limit, offset=offset)
+
+
+def lubm_entity_vocab(store):
+    """lubm_vocab + a positional entity-URI resolver: the reference maps
+    URIs through its StringServer; our synthetic ids are allocation-
+    order, so DepartmentK.UniversityN / FullProfessorJ / ... resolve
+    positionally through the store (same shape and selectivity)."""
+    from . import queries as Q
+    import numpy as np
+
+    def tris(v, p, d):
+        return np.sort(np.asarray(store.get_triples(int(v), p, d)))
+
+    cache = {}
+
+    def depts(n):
+        if ("d", n) not in cache:
+            cache[("d", n)] = tris((n + 1) << 17, Q.SUBORG, DIR_IN)
+        return cache[("d", n)]
+
+    def members_of_type(dept, pred, t):
+        key = ("m", dept, pred, t)
+        if key not in cache:
+            cache[key] = [int(c) for c in tris(dept, pred, DIR_IN)
+                          if t in tris(c, TYPE_ID, DIR_OUT)]
+        return cache[key]
+
+    import re as _re
+
+    def resolver(tok):
+        m = _re.fullmatch(r"<http://www\.University(\d+)\.edu>", tok)
+        if m:
+            return (int(m.group(1)) + 1) << 17
+        m = _re.fullmatch(
+            r"<http://www\.Department(\d+)\.University(\d+)\.edu>", tok)
+        if m:
+            ds = depts(int(m.group(2)))
+            return int(ds[int(m.group(1)) % len(ds)])
+        m = _re.fullmatch(
+            r"<http://www\.Department(\d+)\.University(\d+)\.edu/"
+            r"(FullProfessor|UndergraduateStudent|GraduateStudent)(\d+)>",
+            tok)
+        if m:
+            ds = depts(int(m.group(2)))
+            dept = int(ds[int(m.group(1)) % len(ds)])
+            kind, j = m.group(3), int(m.group(4))
+            if kind == "FullProfessor":
+                pool = members_of_type(dept, Q.WORKSFOR, Q.FULLPROF)
+            elif kind == "UndergraduateStudent":
+                pool = members_of_type(dept, Q.MEMBEROF, Q.UGSTUDENT)
+            else:
+                pool = members_of_type(dept, Q.MEMBEROF, Q.GRADSTUDENT)
+            return pool[j % len(pool)] if pool else None
+        return None
+
+    v = lubm_vocab()
+    v["__resolver__"] = resolver
+    return v
