@@ -577,6 +577,8 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
     constexpr int oc = NC + 1;
     count_bytes(d_stats, CAT_EXPAND,
                 (uint64_t)nrows * (8 + (use_typeof ? 2 : 0)));
+    // K=4 measured best: K=8 (deeper gather MLP) costs more VGPRs than
+    // it hides latency — q1's fn expands regressed 184->216us / 230->341us
     constexpr int K = 4;
     __shared__ unsigned long long s_base;
     __shared__ uint32_t sh[SCAN_T];
