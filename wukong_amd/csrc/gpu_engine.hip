@@ -578,39 +578,64 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
     count_bytes(d_stats, CAT_EXPAND,
                 (uint64_t)nrows * (8 + (use_typeof ? 2 : 0)));
     // K=4 measured best: K=8 (deeper gather MLP) costs more VGPRs than
-    // it hides latency — q1's fn expands regressed 184->216us / 230->341us
+    // it hides latency AND halves the tile count (load imbalance) —
+    // q1's fn expands regressed 184->216us / 230->341us.  Instead, a
+    // 2-deep software pipeline (the k_probe_scan idiom) issues the NEXT
+    // tile's dense-map gathers before this tile's scan barriers, so
+    // their ~1us HBM latency hides under the scan + writes.
     constexpr int K = 4;
     __shared__ unsigned long long s_base;
     __shared__ uint32_t sh[SCAN_T];
     const int64_t tile = (int64_t)blockDim.x * K;
     const int64_t stride = (int64_t)gridDim.x * tile;
 
-    for (int64_t base = (int64_t)blockIdx.x * tile; base < nrows; base += stride) {
+    sid_t pv[K], ptv[K];
+    uint16_t pt16[K];
+    auto load_tile = [&](int64_t b) {
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            const int64_t r = b + (int64_t)threadIdx.x * K + k;
+            sid_t v = (r < nrows) ? tbl[r * NC + col] : 0;
+            pv[k] = v;
+            uint64_t idx = (uint64_t)v - fn_base;
+            sid_t tv = (idx < fn_n) ? fn[idx] : 0;
+            ptv[k] = tv;
+            if (use_typeof) {
+                uint64_t tix = (uint64_t)tv - type_base;
+                pt16[k] = (tv && tix < type_n) ? type_of[tix] : 0;
+            }
+        }
+    };
+    const int64_t base0 = (int64_t)blockIdx.x * tile;
+    if (base0 < nrows) load_tile(base0);
+
+    for (int64_t base = base0; base < nrows; base += stride) {
         bool keep[K];
         sid_t val[K];
         int64_t rr[K];
         uint32_t cnt = 0;
+        sid_t cv[K];
+        uint16_t ct16[K];
 #pragma unroll
         for (int k = 0; k < K; k++) {
-            const int64_t r = base + (int64_t)threadIdx.x * K + k;
-            rr[k] = r;
+            cv[k] = pv[k];
+            val[k] = ptv[k];
+            ct16[k] = pt16[k];
+            rr[k] = base + (int64_t)threadIdx.x * K + k;
+        }
+        if (base + stride < nrows) load_tile(base + stride);  // prefetch
+#pragma unroll
+        for (int k = 0; k < K; k++) {
             keep[k] = false;
-            val[k] = 0;
-            if (r >= nrows) continue;
-            sid_t v = tbl[r * NC + col];
-            uint64_t idx = (uint64_t)v - fn_base;
-            sid_t tv = (idx < fn_n) ? fn[idx] : 0;
-            val[k] = tv;
-            if (!tv) continue;
+            if (rr[k] >= nrows || !val[k]) continue;
             if (!use_typeof) { keep[k] = true; continue; }
-            uint64_t tix = (uint64_t)tv - type_base;
-            uint16_t t = (tix < type_n) ? type_of[tix] : 0;
-            if (t != 0xFFFF) {
-                keep[k] = ((sid_t)t == fcval);
+            if (ct16[k] != 0xFFFF) {
+                keep[k] = ((sid_t)ct16[k] == fcval);
             } else {
                 uint64_t eo = 0, es = 0;
                 probe_one(verts, f_bstart, f_nbuckets,
-                          key_pack(tv, TYPE_ID, (uint64_t)DIR_OUT), eo, es);
+                          key_pack(val[k], TYPE_ID, (uint64_t)DIR_OUT), eo,
+                          es);
                 keep[k] = es && bsearch_u32(edges + eo, es, fcval);
             }
         }
