@@ -448,6 +448,20 @@ def main():
             n = eng_.fetch_count()
         return n
 
+    graph_ids = {}  # plan name -> captured hipGraph id (single-GPU path)
+
+    def build_graphs():
+        """Capture each suite plan as a hipGraph after warmup (the
+        launch chain replays in ONE hipGraphLaunch).  WK_GRAPH=0
+        disables; any per-plan failure falls back to submit."""
+        if distributed or os.environ.get("WK_GRAPH", "1") == "0":
+            return
+        for name in names:
+            try:
+                graph_ids[name] = eng.graph_build(Q_ALL[name])
+            except Exception as ex:  # fall back silently to submit path
+                log(f"[graph] {name}: fallback to submit ({ex})")
+
     def run_suite(collect=None, passes=1):
         """Distributed path / single-engine fallback: sequential queries."""
         for _ in range(passes):
@@ -464,6 +478,12 @@ def main():
                                        device=f"cuda:{local_rank}")
                     dist.all_reduce(cnt)
                     nrows = int(cnt.item())
+                elif name in graph_ids:
+                    try:
+                        nrows = eng.graph_run(graph_ids[name])
+                    except OverflowError:
+                        del graph_ids[name]
+                        nrows = eng.run_query_count(plan)
                 else:
                     nrows = eng.run_query_count(plan)
                 if collect is not None:
@@ -516,6 +536,8 @@ def main():
     # warmup
     if distributed or inflight == 1:
         run_suite(passes=args.warmup)
+        build_graphs()
+        run_suite(passes=1)  # one graph-mode warm pass
     else:
         run_pipelined(args.warmup)
     sync()

@@ -174,6 +174,16 @@ int32_t wk_engine_submit_plan_batch(wk_engine_t *, const wk_plan_t *tmpl,
 int32_t wk_engine_light_batch_wait(wk_engine_t *, uint64_t *counts,
                                    int32_t n);
 
+/* hipGraph replay of a whole fixed plan: build captures the submit
+ * launch chain once (after a warm pass settles capacities); run
+ * replays it in ONE hipGraphLaunch and returns the blind row count.
+ * WK_ERR_CAP from run = capacity overflow (data changed since build):
+ * fall back to wk_engine_submit. */
+int32_t wk_engine_graph_build(wk_engine_t *, const wk_plan_t *,
+                              int32_t *graph_id);
+int32_t wk_engine_graph_run(wk_engine_t *, int32_t graph_id,
+                            int64_t *nrows);
+
 /* Whole-query execution on one GPU (Engine::execute_sparql_query +
  * SPARQLEngine::execute_patterns, core/engine/sparql.hpp:1113-1154,
  * 1564-1672, single-server path).  Returns 0 on success. */

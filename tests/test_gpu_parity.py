@@ -224,3 +224,15 @@ def test_plan_batch_rejects_uninterpretable(store4):
     eng = wk.Engine(store4, device=0)
     with pytest.raises(ValueError):
         eng.submit_plan_batch(Q.Q1, [1 << 17])
+
+
+def test_graph_replay_counts(store4, oracle4):
+    """hipGraph capture/replay (wk_engine_graph_build/run): one
+    hipGraphLaunch per whole query, counts equal the oracle across
+    repeated replays (device state resets inside the graph)."""
+    eng = wk.Engine(store4, device=0)
+    for name, plan in Q.ALL.items():
+        gid = eng.graph_build(plan)
+        want = len(oracle4.run_query(plan))
+        for _ in range(3):
+            assert eng.graph_run(gid) == want, name

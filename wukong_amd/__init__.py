@@ -90,6 +90,10 @@ _eng_lb_wait = _sig("wk_engine_light_batch_wait", c_i32,
 _eng_submit_pb = _sig("wk_engine_submit_plan_batch", c_i32,
                       [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(c_i64),
                        c_i32])
+_eng_graph_build = _sig("wk_engine_graph_build", c_i32,
+                        [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(c_i32)])
+_eng_graph_run = _sig("wk_engine_graph_run", c_i32,
+                      [c_vp, c_i32, ctypes.POINTER(c_i64)])
 
 #: plan-batch count sentinel: table outgrew LDS, re-run per-pattern
 LP_OVERFLOW = 0xFFFFFFFFFFFFFFFF
@@ -307,6 +311,28 @@ class Engine:
         if getattr(self, "_h", None):
             _eng_destroy(self._h)
             self._h = None
+
+    def graph_build(self, plan):
+        """Capture the plan's whole launch chain as a hipGraph (one
+        warm pass first); returns a graph id for graph_run."""
+        cplan = plan.to_c()
+        gid = c_i32()
+        rc = _eng_graph_build(self._h, ctypes.byref(cplan), ctypes.byref(gid))
+        if rc != 0:
+            raise RuntimeError(f"graph_build rc={rc}")
+        return int(gid.value)
+
+    def graph_run(self, gid):
+        """Replay a captured plan (ONE hipGraphLaunch); blind row count.
+        Raises OverflowError on capacity overflow — fall back to
+        submit()."""
+        n = c_i64()
+        rc = _eng_graph_run(self._h, gid, ctypes.byref(n))
+        if rc == -5:
+            raise OverflowError("graph replay overflow: use submit()")
+        if rc != 0:
+            raise RuntimeError(f"graph_run rc={rc}")
+        return int(n.value)
 
     def submit_light_batch(self, subj, pred, dirs, cval):
         """One asynchronous launch for a whole window of light

@@ -1399,6 +1399,17 @@ struct wk_engine {
     int nvars = 0;
     int step = 0;
 
+    // hipGraph replay of whole fixed plans (wk_engine_graph_build/run)
+    struct wk_graph {
+        hipGraphExec_t exec = nullptr;
+        std::vector<int32_t> v2c;
+        int ncols = 0, cur = 0, nvars = 0;
+        int64_t bound = 0;
+        bool light = false;
+    };
+    std::vector<wk_graph> graphs;
+    int capturing = 0;
+
     // timing (WK_KERNEL_TIMING=1)
     bool timing = false;
     std::vector<timed_launch> pending;
@@ -1445,11 +1456,13 @@ static hipError_t stream_sync(hipStream_t s) {
 
 #define TIME_BEGIN(eng)                                                     \
     hipEvent_t t_s_ = nullptr, t_e_ = nullptr;                              \
-    if ((eng)->timing) { t_s_ = ev_get(); t_e_ = ev_get();                  \
+    if ((eng)->timing && !(eng)->capturing) {                               \
+        t_s_ = ev_get(); t_e_ = ev_get();                                   \
         (void)hipEventRecord(t_s_, (eng)->stream); }
 
 #define TIME_END(eng, category)                                             \
-    if ((eng)->timing) { (void)hipEventRecord(t_e_, (eng)->stream);         \
+    if ((eng)->timing && !(eng)->capturing) {                               \
+        (void)hipEventRecord(t_e_, (eng)->stream);                          \
         (eng)->pending.push_back({t_s_, t_e_, (category)}); }
 
 static const char *CAT_NAMES[CAT_COUNT] = {"probe", "scan", "expand", "filter",
@@ -1482,6 +1495,7 @@ static int32_t grow_caps(wk_engine *e, int64_t rows, int cols) {
     rows = std::max<int64_t>(rows, min_cap_rows());
     cols = std::max(cols, std::max(e->cap_cols, 1));
     if (rows <= e->cap_rows && cols <= e->cap_cols) return WK_OK;
+    if (e->capturing) return WK_ERR_STATE;  // no alloc/sync inside capture
     rows = std::max(rows, e->cap_rows);
     // one sync before freeing buffers that in-flight kernels may use
     HIP_CHECK(stream_sync(e->stream));
@@ -1633,6 +1647,8 @@ extern "C" void wk_engine_destroy(wk_engine_t *e) {
     e->eoff.release(); e->cnt.release(); e->prefix.release();
     e->bsums.release(); e->misc.release(); e->ovf.release();
     e->oflag[0].release(); e->oflag[1].release();
+    for (auto &g : e->graphs)
+        if (g.exec) (void)hipGraphExecDestroy(g.exec);
     e->lbd.release(); e->lbcnt.release();
     if (e->h_lb) (void)hipHostFree(e->h_lb);
     if (e->d_state) (void)hipFree(e->d_state);
@@ -2524,6 +2540,86 @@ extern "C" int32_t wk_engine_submit(wk_engine_t *e, const wk_plan_t *plan) {
         rc = exec_pattern(e);
         if (rc) return rc;
     }
+    return WK_OK;
+}
+
+// ---------------------------------------------------------------------
+// hipGraph replay of a whole fixed plan: the benchmark suite re-runs
+// the same 7 queries every step, so the ~20-launch chain (with its
+// per-launch dispatch gaps) collapses into ONE hipGraphLaunch.  Build
+// captures the submit chain (after a warm pass settles capacities —
+// growth is illegal mid-capture); run replays it and returns the blind
+// row count.  S_ERR on replay = capacity overflow: caller falls back
+// to the plain submit path.
+// ---------------------------------------------------------------------
+extern "C" int32_t wk_engine_graph_build(wk_engine_t *e,
+                                         const wk_plan_t *plan,
+                                         int32_t *gid) {
+    if (!e || !plan || !gid) return WK_ERR_STATE;
+    if (plan->nopt > 0 || plan->nunion > 0) return WK_ERR_PLAN;
+    // warm pass: settle scratch capacities so capture never allocates
+    for (int attempt = 0; attempt < 4; attempt++) {
+        int32_t rc = wk_engine_submit(e, plan);
+        if (rc) return rc;
+        rc = sync_state_grow(e);
+        if (rc == WK_OK) break;
+        if (rc != WK_ERR_CAP) return rc;
+    }
+    e->capturing = 1;
+    hipGraph_t g = nullptr;
+    if (hipStreamBeginCapture(e->stream, hipStreamCaptureModeThreadLocal) !=
+        hipSuccess) {
+        e->capturing = 0;
+        return WK_ERR_HIP;
+    }
+    int32_t rc = wk_engine_submit(e, plan);
+    if (rc == WK_OK && !e->light)
+        hipLaunchKernelGGL(k_publish_state, dim3(1), dim3(1), 0, e->stream,
+                           e->d_state, e->d_stats, e->h_pin);
+    hipError_t ce = hipStreamEndCapture(e->stream, &g);
+    e->capturing = 0;
+    if (rc != WK_OK || ce != hipSuccess || !g) {
+        if (g) (void)hipGraphDestroy(g);
+        // the capture aborted mid-chain: resynchronise engine state
+        (void)stream_sync(e->stream);
+        return rc != WK_OK ? rc : WK_ERR_HIP;
+    }
+    hipGraphExec_t ex = nullptr;
+    if (hipGraphInstantiate(&ex, g, nullptr, nullptr, 0) != hipSuccess) {
+        (void)hipGraphDestroy(g);
+        return WK_ERR_HIP;
+    }
+    (void)hipGraphDestroy(g);
+    wk_engine::wk_graph wg;
+    wg.exec = ex;
+    wg.v2c = e->v2c;
+    wg.ncols = e->ncols;
+    wg.cur = e->cur;
+    wg.nvars = e->nvars;
+    wg.bound = e->bound;
+    wg.light = e->light;
+    e->graphs.push_back(std::move(wg));
+    *gid = (int32_t)e->graphs.size() - 1;
+    return WK_OK;
+}
+
+extern "C" int32_t wk_engine_graph_run(wk_engine_t *e, int32_t gid,
+                                       int64_t *nrows) {
+    if (!e || gid < 0 || gid >= (int32_t)e->graphs.size())
+        return WK_ERR_STATE;
+    wk_engine::wk_graph &g = e->graphs[gid];
+    resolve_timing(e);
+    if (hipGraphLaunch(g.exec, e->stream) != hipSuccess) return WK_ERR_HIP;
+    HIP_CHECK(stream_sync(e->stream));
+    e->v2c = g.v2c;
+    e->ncols = g.ncols;
+    e->cur = g.cur;
+    e->nvars = g.nvars;
+    e->bound = g.bound;
+    e->light = g.light;
+    if (e->h_pin[S_ERR]) return WK_ERR_CAP;
+    e->nrows = (int64_t)e->h_pin[S_NROWS];
+    if (nrows) *nrows = e->nrows;
     return WK_OK;
 }
 
