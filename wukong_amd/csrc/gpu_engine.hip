@@ -284,6 +284,7 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                              const uint16_t *__restrict__ type_of,
                              uint64_t type_base, uint64_t type_n,
                              int use_typeof,
+                             const uint64_t *__restrict__ tbm,
                              const sid_t *__restrict__ fn, uint64_t fn_base,
                              uint64_t fn_n, int fn_swap,
                              uint64_t *__restrict__ d_state,
@@ -318,6 +319,13 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
             sid_t v = tbl[r * ncols + col];
             if (use_typeof) {
                 uint64_t idx = (uint64_t)v - type_base;
+                if (tbm) {
+                    // per-type bitmap: 1 bit/vid, whole map LLC-resident
+                    // — exact (multi-type included), no probe fallback
+                    keep[k] = idx < type_n &&
+                              ((tbm[idx >> 6] >> (idx & 63)) & 1);
+                    continue;
+                }
                 uint16_t t = (idx < type_n) ? type_of[idx] : 0;
                 if (t != 0xFFFF) { keep[k] = ((sid_t)t == cval); continue; }
             }
@@ -565,6 +573,7 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
                             uint64_t fn_n, int col, int use_typeof,
                             sid_t fcval,
                             const uint16_t *__restrict__ type_of,
+                            const uint64_t *__restrict__ tbm,
                             uint64_t type_base, uint64_t type_n,
                             const vertex_t *__restrict__ verts,
                             const sid_t *__restrict__ edges,
@@ -602,7 +611,12 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
             ptv[k] = tv;
             if (use_typeof) {
                 uint64_t tix = (uint64_t)tv - type_base;
-                pt16[k] = (tv && tix < type_n) ? type_of[tix] : 0;
+                if (tbm)  // bitmap: decision is exact (multi-type incl.)
+                    pt16[k] = (tv && tix < type_n &&
+                               ((tbm[tix >> 6] >> (tix & 63)) & 1))
+                                  ? 1 : 0;
+                else
+                    pt16[k] = (tv && tix < type_n) ? type_of[tix] : 0;
             }
         }
     };
@@ -629,6 +643,7 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
             keep[k] = false;
             if (rr[k] >= nrows || !val[k]) continue;
             if (!use_typeof) { keep[k] = true; continue; }
+            if (tbm) { keep[k] = ct16[k] != 0; continue; }
             if (ct16[k] != 0xFFFF) {
                 keep[k] = ((sid_t)ct16[k] == fcval);
             } else {
@@ -1344,6 +1359,8 @@ struct wk_gpu_store {
     // functional-predicate dense maps, indexed [pid*2+dir] (host vector
     // of device pointers; null = absent)
     std::vector<sid_t *> d_fn;
+    // per-type membership bitmaps (LLC-resident typeof filters)
+    std::vector<uint64_t *> d_tbm;
     int refs = 0;     // engines attached
     bool owned = false;  // created implicitly by wk_engine_create
 };
@@ -1585,6 +1602,19 @@ extern "C" wk_gpu_store_t *wk_gpu_store_create(const wk_store_t *st, int32_t dev
             }
         }
     }
+    if (!st->tbm.empty()) {
+        g->d_tbm.assign(st->tbm.size(), nullptr);
+        for (size_t t = 0; t < st->tbm.size(); t++) {
+            if (st->tbm[t].empty()) continue;
+            size_t tb = st->tbm[t].size() * 8;
+            if (hipMalloc(&g->d_tbm[t], tb) != hipSuccess ||
+                hipMemcpy(g->d_tbm[t], st->tbm[t].data(), tb,
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                wk_gpu_store_destroy(g);
+                return nullptr;
+            }
+        }
+    }
     return g;
 }
 
@@ -1600,6 +1630,8 @@ extern "C" void wk_gpu_store_destroy(wk_gpu_store_t *g) {
     }
     if (g->d_segtab) (void)hipFree(g->d_segtab);
     for (sid_t *p : g->d_fn)
+        if (p) (void)hipFree(p);
+    for (uint64_t *p : g->d_tbm)
         if (p) (void)hipFree(p);
     delete g;
 }
@@ -1827,10 +1859,14 @@ template <int NC>
 static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
                                sid_t *out_tbl, const sid_t *d_fn, int col,
                                bool fuse, sid_t fcval, const seg_t *fseg) {
+    const uint64_t *tbm =
+        (fuse && e->gs && (size_t)fcval < e->gs->d_tbm.size())
+            ? e->gs->d_tbm[fcval]
+            : nullptr;
     hipLaunchKernelGGL(k_expand_fn<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
                        0, e->stream, cur_tbl, d_fn, e->st->fn_base,
                        e->st->fn_n, col, fuse ? 1 : 0, fcval, e->d_type_of,
-                       e->st->type_base, e->st->type_n, e->d_verts,
+                       tbm, e->st->type_base, e->st->type_n, e->d_verts,
                        e->d_edges, fseg ? fseg->bucket_start : 0,
                        fseg ? fseg->num_buckets : 0, e->d_state, e->d_stats,
                        out_tbl);
@@ -2033,6 +2069,7 @@ static int32_t exec_pattern(wk_engine *e) {
                            e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
                            e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
                            off, sz, e->d_type_of, 0, 0, 0,
+                           (const uint64_t *)nullptr,
                            (const sid_t *)nullptr, 0, 0, 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
@@ -2081,6 +2118,10 @@ static int32_t exec_pattern(wk_engine *e) {
                           dir == DIR_OUT && key_mode == PK_NORMAL &&
                           e->d_type_of != nullptr)
                              ? 1 : 0;
+        const uint64_t *d_tbm =
+            (use_typeof && e->gs && (size_t)cval < e->gs->d_tbm.size())
+                ? e->gs->d_tbm[cval]
+                : nullptr;
         // functional predicate: row's single object replaces the probe.
         // If only the REVERSED direction is functional, k2k checks
         // fn[other col] == col (fn_swap); k2c resolves the single edge
@@ -2115,8 +2156,8 @@ static int32_t exec_pattern(wk_engine *e) {
                            seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
                            dir, key_mode, pmode, col2, cval, 0, 0,
                            e->d_type_of, e->st->type_base, e->st->type_n,
-                           use_typeof, d_fn, st->fn_base, st->fn_n, fn_swap,
-                           e->d_state, e->d_stats, out_tbl);
+                           use_typeof, d_tbm, d_fn, st->fn_base, st->fn_n,
+                           fn_swap, e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
     } else {
         // known_to_unknown: fused probe+scan -> cross-block scan ->

@@ -479,6 +479,50 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
                    picked.size(), spent / 1e9, now_s() - t0);
         }
     }
+    t0 = now_s();
+    // per-type membership bitmaps from the TYPE_ID pso slice (exact:
+    // includes multi-type vids).  Budgeted; WK_TBM=0 disables.
+    {
+        const char *tv = getenv("WK_TBM");
+        bool want = !(tv && !atoi(tv));
+        if (want && st->type_n) {
+            const char *bv = getenv("WK_TBM_BUDGET_GB");
+            uint64_t budget = (bv ? strtoull(bv, nullptr, 10) : 4) << 30;
+            uint64_t words = (st->type_n + 63) / 64;
+            uint64_t per = words * 8;
+            st->tbm.assign((size_t)NP, {});
+            // rank types by member count, biggest first
+            std::vector<std::pair<uint64_t, uint32_t>> order;
+            for (uint32_t t2 = 2; t2 < NP; t2++)
+                if (type_members[t2]) order.push_back({type_members[t2], t2});
+            std::sort(order.begin(), order.end(),
+                      [](auto &a, auto &b) { return a.first > b.first; });
+            uint64_t spent = 0;
+            std::vector<uint32_t> picked;
+            for (auto &pr : order) {
+                if (spent + per > budget) break;
+                spent += per;
+                picked.push_back(pr.second);
+            }
+            for (uint32_t t2 : picked) st->tbm[t2].assign(words, 0);
+            // one parallel sweep of the type slice (subject-disjoint
+            // runs; same vid can repeat per type, but only within one
+            // run -> races only on the same word for NEARBY subjects:
+            // use atomic or bit OR; subjects within a 64-run may share a
+            // word across threads, so use atomic OR
+#pragma omp parallel for schedule(static)
+            for (int64_t i = pso_lo[TYPE_ID]; i < pso_lo[TYPE_ID + 1]; i++) {
+                uint32_t t2 = pso[i].o;
+                if (t2 >= NP || st->tbm[t2].empty()) continue;
+                uint64_t idx = (uint64_t)pso[i].s - st->type_base;
+                if (idx >= st->type_n) continue;
+                __atomic_fetch_or(&st->tbm[t2][idx >> 6],
+                                  1ull << (idx & 63), __ATOMIC_RELAXED);
+            }
+            WK_LOG("[store] type bitmaps: %zu types, %.2f GB (%.1fs)\n",
+                   picked.size(), spent / 1e9, now_s() - t0);
+        }
+    }
     st->ext_used = ext_next.load() - st->nbuckets_main;
     return st;
 }

@@ -87,6 +87,14 @@ struct wk_store {
     // (the reference's type-centric stats, core/optimizer/stats.hpp)
     std::vector<uint64_t> seg_keys, seg_edges;  // [pid*2+dir]
 
+    // per-type membership bitmaps: tbm[tid] bit (vid - vp_base) set iff
+    // (vid, TYPE_ID, tid) exists.  1 bit/vid (~22 MB per type at
+    // LUBM-2560) keeps the whole bitmap LLC-resident, so `?x rdf:type T`
+    // filters read cache lines instead of HBM-random 2-byte gathers —
+    // and the semantics are EXACT (multi-type vids included; no 0xFFFF
+    // fallback).  Indexed by type id; empty = absent.
+    std::vector<std::vector<uint64_t>> tbm;
+
     uint64_t nbuckets_main = 0, nbuckets_ext = 0, ext_used = 0;
 
     const wk::seg_t *seg_of(uint64_t vid, uint64_t pid, int dir) const {
