@@ -443,7 +443,12 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
             st->fn_n = (uint64_t)max_id + 1 - st->fn_base;
             st->fn.assign((size_t)NP * 2, {});
             const char *bv = getenv("WK_FN_BUDGET_GB");
-            uint64_t budget = (bv ? strtoull(bv, nullptr, 10) : 24) << 30;
+            // default budget shrinks with the partition count: each
+            // rank's maps still span the FULL vid range (mostly zeros
+            // off-partition), so 8 ranks would cost 8x the host+HBM
+            uint64_t gb = bv ? strtoull(bv, nullptr, 10)
+                             : (uint64_t)(24 / (nsrv > 0 ? nsrv : 1));
+            uint64_t budget = std::max<uint64_t>(gb, 2) << 30;
             std::vector<std::pair<uint64_t, uint32_t>> cand;  // (-edges, w)
             for (uint32_t p = 1; p <= max_pid; p++) {
                 if (out_keys[p] && out_keys[p] == out_edges[p])
