@@ -1311,6 +1311,67 @@ __global__ void k_expand_opt(const vertex_t *__restrict__ verts,
     }
 }
 
+// OPTIMISTIC functional k2u: used ONLY inside captured graphs for
+// steps whose warm pass dropped no rows (store is immutable, so the
+// observation holds for replays; a miss at runtime still flips S_ERR
+// via k_commit_map and the caller falls back to the safe path).  The
+// 1:1 write needs no scan/compaction and coalesces perfectly.
+template <int NC>
+__global__ void k_expand_fn_map(const sid_t *__restrict__ tbl,
+                                const sid_t *__restrict__ fn,
+                                uint64_t fn_base, uint64_t fn_n, int col,
+                                int use_typeof,
+                                const uint64_t *__restrict__ tbm,
+                                const uint16_t *__restrict__ type_of,
+                                uint64_t type_base, uint64_t type_n,
+                                sid_t fcval,
+                                uint64_t *__restrict__ d_state,
+                                uint64_t *__restrict__ d_stats,
+                                sid_t *__restrict__ out)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    constexpr int oc = NC + 1;
+    count_bytes(d_stats, CAT_EXPAND,
+                (uint64_t)nrows * (8 + (use_typeof ? 1 : 0) + oc * 4));
+    uint32_t miss = 0;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        sid_t v = tbl[r * NC + col];
+        uint64_t idx = (uint64_t)v - fn_base;
+        sid_t tv = (idx < fn_n) ? fn[idx] : 0;
+        bool ok = tv != 0;
+        if (ok && use_typeof) {
+            uint64_t tix = (uint64_t)tv - type_base;
+            if (tbm) {
+                ok = tix < type_n && ((tbm[tix >> 6] >> (tix & 63)) & 1);
+            } else {
+                uint16_t t = (tix < type_n) ? type_of[tix] : 0;
+                ok = ((sid_t)t == fcval);  // 0xFFFF never matches ->
+                                           // counts as miss -> fallback
+            }
+        }
+        miss += ok ? 0u : 1u;
+        sid_t *dst = out + r * oc;
+#pragma unroll
+        for (int c = 0; c < NC; c++) dst[c] = tbl[r * NC + c];
+        dst[NC] = tv;
+    }
+    if (miss)
+        atomicAdd((unsigned long long *)&d_state[S_OVF],
+                  (unsigned long long)miss);
+}
+
+// commit for the optimistic map: row count unchanged; any miss flags
+// S_ERR so the replay result is discarded and the safe path re-runs
+__global__ void k_commit_map(uint64_t *__restrict__ d_state) {
+    if (d_state[S_OVF]) {
+        d_state[S_ERR] = 1;
+        d_state[S_REQ] = max(d_state[S_REQ], d_state[S_NROWS]);
+    }
+    d_state[S_TOTAL] = 0;
+    d_state[S_OVF] = 0;
+}
+
 __global__ void k_zero_words(uint64_t *p, int n) {
     for (int i = threadIdx.x; i < n; i += blockDim.x) p[i] = 0;
 }
@@ -1426,6 +1487,9 @@ struct wk_engine {
     };
     std::vector<wk_graph> graphs;
     int capturing = 0;
+    // per-pattern-step no-drop hints from the graph build's warm pass
+    // (consulted ONLY while capturing)
+    std::vector<uint8_t> capture_hint;
 
     // timing (WK_KERNEL_TIMING=1)
     bool timing = false;
@@ -1872,6 +1936,36 @@ static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
                        out_tbl);
 }
 
+template <int NC>
+static void launch_expand_fn_map_t(wk_engine *e, const sid_t *cur_tbl,
+                                   sid_t *out_tbl, const sid_t *d_fn,
+                                   int col, bool fuse, sid_t fcval) {
+    const uint64_t *tbm =
+        (fuse && e->gs && (size_t)fcval < e->gs->d_tbm.size())
+            ? e->gs->d_tbm[fcval]
+            : nullptr;
+    hipLaunchKernelGGL(k_expand_fn_map<NC>, dim3(grid_for(e->bound)),
+                       dim3(BLOCK), 0, e->stream, cur_tbl, d_fn,
+                       e->st->fn_base, e->st->fn_n, col, fuse ? 1 : 0, tbm,
+                       e->d_type_of, e->st->type_base, e->st->type_n, fcval,
+                       e->d_state, e->d_stats, out_tbl);
+}
+
+static void launch_expand_fn_map(wk_engine *e, const sid_t *cur_tbl,
+                                 sid_t *out_tbl, const sid_t *d_fn, int col,
+                                 bool fuse, sid_t fcval) {
+    switch (e->ncols) {
+    case 1: launch_expand_fn_map_t<1>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    case 2: launch_expand_fn_map_t<2>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    case 3: launch_expand_fn_map_t<3>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    case 4: launch_expand_fn_map_t<4>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    case 5: launch_expand_fn_map_t<5>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    case 6: launch_expand_fn_map_t<6>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    case 7: launch_expand_fn_map_t<7>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    default: launch_expand_fn_map_t<8>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    }
+}
+
 static void launch_expand_fn(wk_engine *e, const sid_t *cur_tbl,
                              sid_t *out_tbl, const sid_t *d_fn, int col,
                              bool fuse, sid_t fcval, const seg_t *fseg) {
@@ -2113,6 +2207,14 @@ static int32_t exec_pattern(wk_engine *e) {
     int col2 = (ostat == 1) ? e->var2col(o) : 0;
     sid_t cval = (ostat == 2) ? (sid_t)o : 0;
     if (pmode != PM_SIZE) {
+        // identity filter: the graph build's warm pass saw this step
+        // keep every row — the store is immutable, so inside a captured
+        // graph the whole pass is skipped (q7's COURSE filter: 86us -> 0)
+        if (e->capturing && e->step < (int)e->capture_hint.size() &&
+            e->capture_hint[e->step]) {
+            e->step++;
+            return WK_OK;
+        }
         TIME_BEGIN(e);
         int use_typeof = (pmode == PM_CONST && (sid_t)p == TYPE_ID &&
                           dir == DIR_OUT && key_mode == PK_NORMAL &&
@@ -2191,12 +2293,25 @@ static int32_t exec_pattern(wk_engine *e) {
                     }
                 }
             }
+            // optimistic 1:1 map: only while capturing a graph, and only
+            // when the build's warm pass saw this step drop no rows
+            const bool opt = e->capturing &&
+                             e->step < (int)e->capture_hint.size() &&
+                             e->capture_hint[e->step];
             TIME_BEGIN(e);
-            launch_expand_fn(e, cur_tbl, out_tbl, d_fn, col, fuse2, fcval2,
-                             fseg2);
+            if (opt)
+                launch_expand_fn_map(e, cur_tbl, out_tbl, d_fn, col, fuse2,
+                                     fcval2);
+            else
+                launch_expand_fn(e, cur_tbl, out_tbl, d_fn, col, fuse2,
+                                 fcval2, fseg2);
             TIME_END(e, CAT_EXPAND);
-            hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
-                               e->d_state, (uint64_t)e->cap_rows);
+            if (opt)
+                hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0,
+                                   e->stream, e->d_state);
+            else
+                hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
+                                   e->d_state, (uint64_t)e->cap_rows);
             e->v2c[-(o + 1)] = e->ncols;
             e->ncols = oc;
             e->cur ^= 1;
@@ -2606,11 +2721,30 @@ extern "C" int32_t wk_engine_graph_build(wk_engine_t *e,
         if (rc == WK_OK) break;
         if (rc != WK_ERR_CAP) return rc;
     }
+    // step-wise pass: record which pattern steps dropped no rows (the
+    // store is immutable, so these hints hold for every replay; the
+    // optimistic kernels still guard at runtime via S_ERR)
+    e->capture_hint.assign(std::max(plan->npatterns, 0), 0);
+    {
+        int32_t rc = wk_engine_begin_query(e, plan);
+        if (rc) return rc;
+        int64_t prev = -1;
+        while (e->step < (int)e->pats.size()) {
+            const int at = e->step;
+            int64_t n = 0;
+            rc = wk_engine_execute_one_pattern(e, &n);
+            if (rc) return rc;
+            if (n == prev && at < (int)e->capture_hint.size())
+                e->capture_hint[at] = 1;
+            prev = n;
+        }
+    }
     e->capturing = 1;
     hipGraph_t g = nullptr;
     if (hipStreamBeginCapture(e->stream, hipStreamCaptureModeThreadLocal) !=
         hipSuccess) {
         e->capturing = 0;
+        e->capture_hint.clear();
         return WK_ERR_HIP;
     }
     int32_t rc = wk_engine_submit(e, plan);
@@ -2619,6 +2753,7 @@ extern "C" int32_t wk_engine_graph_build(wk_engine_t *e,
                            e->d_state, e->d_stats, e->h_pin);
     hipError_t ce = hipStreamEndCapture(e->stream, &g);
     e->capturing = 0;
+    e->capture_hint.clear();
     if (rc != WK_OK || ce != hipSuccess || !g) {
         if (g) (void)hipGraphDestroy(g);
         // the capture aborted mid-chain: resynchronise engine state
