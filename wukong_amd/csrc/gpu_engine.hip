@@ -286,7 +286,7 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                              int use_typeof,
                              const uint64_t *__restrict__ tbm,
                              const sid_t *__restrict__ fn, uint64_t fn_base,
-                             uint64_t fn_n, int fn_swap,
+                             uint64_t fn_n, int fn_swap, int verify_only,
                              uint64_t *__restrict__ d_state,
                              uint64_t *__restrict__ d_stats,
                              sid_t *__restrict__ out_tbl)
@@ -360,6 +360,20 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
         }
 #pragma unroll
         for (int k = 0; k < K; k++) cnt += keep[k] ? 1u : 0u;
+        if (verify_only) {
+            // identity-verified filter (captured graphs): the warm pass
+            // saw zero drops; every replay still checks every row and
+            // flags S_ERR on any miss (no writes, no table flip)
+            uint32_t miss = 0;
+#pragma unroll
+            for (int k = 0; k < K; k++)
+                miss += (rr[k] < nrows && !keep[k]) ? 1u : 0u;
+            if (miss)
+                atomicAdd((unsigned long long *)&d_state[S_OVF],
+                          (unsigned long long)miss);
+            __syncthreads();
+            continue;
+        }
         // block exclusive scan of per-thread counts
         sh[threadIdx.x] = cnt;
         __syncthreads();
@@ -2164,7 +2178,7 @@ static int32_t exec_pattern(wk_engine *e) {
                            e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
                            off, sz, e->d_type_of, 0, 0, 0,
                            (const uint64_t *)nullptr,
-                           (const sid_t *)nullptr, 0, 0, 0,
+                           (const sid_t *)nullptr, 0, 0, 0, 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
         hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
@@ -2207,14 +2221,14 @@ static int32_t exec_pattern(wk_engine *e) {
     int col2 = (ostat == 1) ? e->var2col(o) : 0;
     sid_t cval = (ostat == 2) ? (sid_t)o : 0;
     if (pmode != PM_SIZE) {
-        // identity filter: the graph build's warm pass saw this step
-        // keep every row — the store is immutable, so inside a captured
-        // graph the whole pass is skipped (q7's COURSE filter: 86us -> 0)
-        if (e->capturing && e->step < (int)e->capture_hint.size() &&
-            e->capture_hint[e->step]) {
-            e->step++;
-            return WK_OK;
-        }
+        // identity-verified filter: the graph build's warm pass saw this
+        // step keep every row, so the captured graph runs a read-only
+        // verification (typeof/membership checks on every row, misses ->
+        // S_ERR -> replay discarded + safe fallback) with no compaction
+        // write and no table flip (q7's COURSE filter: 86us -> ~40us)
+        const bool verify_only =
+            e->capturing && e->step < (int)e->capture_hint.size() &&
+            e->capture_hint[e->step];
         TIME_BEGIN(e);
         int use_typeof = (pmode == PM_CONST && (sid_t)p == TYPE_ID &&
                           dir == DIR_OUT && key_mode == PK_NORMAL &&
@@ -2259,8 +2273,15 @@ static int32_t exec_pattern(wk_engine *e) {
                            dir, key_mode, pmode, col2, cval, 0, 0,
                            e->d_type_of, e->st->type_base, e->st->type_n,
                            use_typeof, d_tbm, d_fn, st->fn_base, st->fn_n,
-                           fn_swap, e->d_state, e->d_stats, out_tbl);
+                           fn_swap, verify_only ? 1 : 0,
+                           e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
+        if (verify_only) {
+            hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0, e->stream,
+                               e->d_state);
+            e->step++;
+            return WK_OK;
+        }
     } else {
         // known_to_unknown: fused probe+scan -> cross-block scan ->
         // input-centric expansion (+ big-row wave pass)
