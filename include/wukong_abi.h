@@ -118,6 +118,12 @@ uint64_t wk_store_num_edges(const wk_store_t *);
 int32_t wk_store_seg_stats(const wk_store_t *, uint32_t pid, int32_t dir,
                            uint64_t *keys, uint64_t *edges);
 uint64_t wk_store_checksum(const wk_store_t *); /* FNV over vertices+edges */
+/* Memory-usage report (GStore::print_mem_usage, gstore.hpp:1062-1103):
+ * cluster-hash slots, edge arrays, and the side indexes (type_of, vp
+ * CSR, functional maps, type bitmaps). */
+int32_t wk_store_mem_usage(const wk_store_t *, uint64_t *slots_bytes,
+                           uint64_t *edges_bytes,
+                           uint64_t *side_index_bytes);
 /* Full-store integrity scan (the `gsck` command, core/store/
  * gchecker.hpp:364-392): 0 = consistent, else #violations. */
 uint64_t wk_store_check(const wk_store_t *);

@@ -101,3 +101,12 @@ def test_store_integrity_gsck(store4):
 def test_store_integrity_watdiv():
     st = wk.Store(wk.watdiv_gen(500, seed=1))
     assert st.check() == 0
+
+
+def test_mem_usage(lubm2):
+    import wukong_amd as wk
+    st = wk.Store(lubm2)
+    slots, edges, side = st.mem_usage()
+    assert slots == st.num_slots * 16
+    assert edges == st.num_edges * 4
+    assert side > 0  # type_of + vp CSR + fn maps + type bitmaps

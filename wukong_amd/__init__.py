@@ -74,6 +74,9 @@ _checksum = _sig("wk_store_checksum", c_u64, [c_vp])
 _seg_stats = _sig("wk_store_seg_stats", c_i32,
                   [c_vp, ctypes.c_uint32, c_i32, ctypes.POINTER(c_u64),
                    ctypes.POINTER(c_u64)])
+_mem_usage = _sig("wk_store_mem_usage", c_i32,
+                  [c_vp, ctypes.POINTER(c_u64), ctypes.POINTER(c_u64),
+                   ctypes.POINTER(c_u64)])
 _store_check = _sig("wk_store_check", c_u64, [c_vp])
 _eng_create = _sig("wk_engine_create", c_vp, [c_vp, c_i32])
 _gstore_create = _sig("wk_gpu_store_create", c_vp, [c_vp, c_i32])
@@ -257,6 +260,13 @@ class Store:
 
     def checksum(self):
         return _checksum(self._h)
+
+    def mem_usage(self):
+        """(slots_bytes, edges_bytes, side_index_bytes) — the
+        reference's print_mem_usage report (gstore.hpp:1062-1103)."""
+        a, b, c = c_u64(), c_u64(), c_u64()
+        _mem_usage(self._h, ctypes.byref(a), ctypes.byref(b), ctypes.byref(c))
+        return int(a.value), int(b.value), int(c.value)
 
     def seg_stats(self, pid, direction):
         """(distinct keys, total edges) of the (pid,dir) segment — the

@@ -561,6 +561,24 @@ extern "C" int32_t wk_store_seg_stats(const wk_store_t *st, uint32_t pid,
     return 0;
 }
 
+// memory-usage report (the reference's GStore::print_mem_usage,
+// core/store/gstore.hpp:1062-1103): bytes per region, via out params.
+extern "C" int32_t wk_store_mem_usage(const wk_store_t *st,
+                                      uint64_t *slots_bytes,
+                                      uint64_t *edges_bytes,
+                                      uint64_t *side_index_bytes) {
+    if (!st) return -1;
+    if (slots_bytes) *slots_bytes = st->vertices.size() * sizeof(vertex_t);
+    if (edges_bytes) *edges_bytes = st->edges.size() * sizeof(sid_t);
+    uint64_t side = st->type_n * 2;  // type_of
+    for (int d = 0; d < 2; d++)
+        side += st->vp_off[d].size() * 4 + st->vp_edges[d].size() * 4;
+    for (auto &m : st->fn) side += m.size() * 4;
+    for (auto &b : st->tbm) side += b.size() * 8;
+    if (side_index_bytes) *side_index_bytes = side;
+    return 0;
+}
+
 extern "C" uint64_t wk_store_num_slots(const wk_store_t *st) {
     return st->vertices.size();
 }
