@@ -544,8 +544,27 @@ def main():
     stats0 = all_stats()
 
     lat = {}
+    # full graph coverage: per-query latencies from a dedicated
+    # sequential pass, then the TIMED loop enqueues all replays of a
+    # pass back-to-back with ONE sync (the reference proxy's in-flight
+    # window applied to replays; graphs serialize safely on one stream)
+    graphs_all = (not distributed and inflight == 1
+                  and all(n in graph_ids for n in names))
+    if graphs_all:
+        run_suite(collect=lat, passes=3)
     t_start = time.time()
-    if distributed or inflight == 1:
+    if graphs_all:
+        try:
+            for _ in range(args.steps):
+                for name in names:
+                    eng.graph_launch(graph_ids[name])
+                eng.sync()
+        except OverflowError:
+            log("[graph] pipelined replay overflow: classic timed loop")
+            graphs_all = False
+            t_start = time.time()
+            run_suite(collect=lat, passes=args.steps)
+    elif distributed or inflight == 1:
         run_suite(collect=lat, passes=args.steps)
     else:
         run_pipelined(args.steps, collect=lat)

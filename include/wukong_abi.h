@@ -189,6 +189,11 @@ int32_t wk_engine_graph_build(wk_engine_t *, const wk_plan_t *,
                               int32_t *graph_id);
 int32_t wk_engine_graph_run(wk_engine_t *, int32_t graph_id,
                             int64_t *nrows);
+/* Asynchronous replay (no sync): back-to-back graphs on one stream
+ * serialize safely, so a whole suite pass costs ONE wk_engine_sync —
+ * the reference proxy's in-flight window applied to replays. */
+int32_t wk_engine_graph_launch(wk_engine_t *, int32_t graph_id);
+int32_t wk_engine_sync(wk_engine_t *);
 
 /* Whole-query execution on one GPU (Engine::execute_sparql_query +
  * SPARQLEngine::execute_patterns, core/engine/sparql.hpp:1113-1154,

@@ -97,6 +97,8 @@ _eng_graph_build = _sig("wk_engine_graph_build", c_i32,
                         [c_vp, ctypes.POINTER(WkPlan), ctypes.POINTER(c_i32)])
 _eng_graph_run = _sig("wk_engine_graph_run", c_i32,
                       [c_vp, c_i32, ctypes.POINTER(c_i64)])
+_eng_graph_launch = _sig("wk_engine_graph_launch", c_i32, [c_vp, c_i32])
+_eng_sync = _sig("wk_engine_sync", c_i32, [c_vp])
 
 #: plan-batch count sentinel: table outgrew LDS, re-run per-pattern
 LP_OVERFLOW = 0xFFFFFFFFFFFFFFFF
@@ -343,6 +345,21 @@ class Engine:
         if rc != 0:
             raise RuntimeError(f"graph_run rc={rc}")
         return int(n.value)
+
+    def graph_launch(self, gid):
+        """Async replay (no sync); pair with sync()."""
+        rc = _eng_graph_launch(self._h, gid)
+        if rc != 0:
+            raise RuntimeError(f"graph_launch rc={rc}")
+
+    def sync(self):
+        """One stream sync; raises OverflowError if any replayed graph
+        flagged S_ERR (fall back to per-query submit)."""
+        rc = _eng_sync(self._h)
+        if rc == -5:
+            raise OverflowError("replayed graph overflow")
+        if rc != 0:
+            raise RuntimeError(f"sync rc={rc}")
 
     def submit_light_batch(self, subj, pred, dirs, cval):
         """One asynchronous launch for a whole window of light

@@ -472,6 +472,10 @@ __global__ void k_publish_state(const uint64_t *__restrict__ d_state,
                                 uint64_t *__restrict__ h_pin) {
     for (int i = 0; i < S_WORDS; i++) h_pin[i] = d_state[i];
     for (int i = 0; i < CAT_COUNT; i++) h_pin[8 + i] = d_stats[i];
+    // sticky error flag (h_pin slot 7): survives across back-to-back
+    // graph replays whose own begin kernels clear d_state — the
+    // one-sync-per-pass path checks and clears it host-side
+    if (d_state[S_ERR]) h_pin[7] = 1;
 }
 
 // Input-centric expansion (known_to_unknown back half,
@@ -2798,6 +2802,29 @@ extern "C" int32_t wk_engine_graph_build(wk_engine_t *e,
     e->graphs.push_back(std::move(wg));
     *gid = (int32_t)e->graphs.size() - 1;
     return WK_OK;
+}
+
+// Asynchronous replay: enqueue the graph with NO sync — back-to-back
+// graphs on one stream serialize safely (each begins with its own
+// state-reset kernel), so a whole suite pass costs ONE host sync
+// (wk_engine_sync).  Counts are not read back (throughput loops).
+extern "C" int32_t wk_engine_graph_launch(wk_engine_t *e, int32_t gid) {
+    if (!e || gid < 0 || gid >= (int32_t)e->graphs.size())
+        return WK_ERR_STATE;
+    if (hipGraphLaunch(e->graphs[gid].exec, e->stream) != hipSuccess)
+        return WK_ERR_HIP;
+    return WK_OK;
+}
+
+extern "C" int32_t wk_engine_sync(wk_engine_t *e) {
+    if (!e) return WK_ERR_STATE;
+    HIP_CHECK(stream_sync(e->stream));
+    resolve_timing(e);
+    if (e->h_pin[7]) {  // sticky S_ERR from ANY replay in the window
+        e->h_pin[7] = 0;
+        return WK_ERR_CAP;
+    }
+    return e->h_pin[S_ERR] ? WK_ERR_CAP : WK_OK;
 }
 
 extern "C" int32_t wk_engine_graph_run(wk_engine_t *e, int32_t gid,
