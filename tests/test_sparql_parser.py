@@ -66,3 +66,41 @@ def test_unplanned_textual_order():
     p = sparql.parse(QS.TEXT["q2"], vocab)  # no plan: textual order, OUT
     assert p.patterns[0][2] == sparql.DIR_OUT
     assert len(p.patterns) == 2
+
+
+def test_parse_optional_union(lubm2):
+    """OPTIONAL { } and { } UNION { } parse into Plan.optional/unions
+    and execute (oracle) identically to the hand-built groups."""
+    import wukong_amd as wk
+    from wukong_amd import Plan
+    from tests.oracle_util import OracleCtx
+    ora = OracleCtx(lubm2)
+    vocab = sparql.lubm_vocab()
+    text = vocab and (
+        "PREFIX rdf: <http://www.w3.org/1999/02/22-rdf-syntax-ns#>\n"
+        "PREFIX ub: <http://swat.cse.lehigh.edu/onto/univ-bench.owl#>\n"
+        "SELECT ?x ?y WHERE {\n"
+        "  ?x rdf:type ub:UndergraduateStudent .\n"
+        "  OPTIONAL { ?x ub:advisor ?y . }\n"
+        "}")
+    from wukong_amd import planner
+    store = wk.Store(lubm2)
+    plan = planner.plan_text(store, text, vocab)
+    assert len(plan.optional) == 1 and not plan.unions
+    got = ora.run_query(plan)
+    want = ora.run_query(Plan(plan.patterns, plan.nvars, plan.required_vars,
+                              optional=plan.optional))
+    assert np.array_equal(sort_rows(got), sort_rows(want))
+    assert (got == 0xFFFFFFFF).any()  # BLANKs present
+
+    text2 = (
+        "PREFIX rdf: <http://www.w3.org/1999/02/22-rdf-syntax-ns#>\n"
+        "PREFIX ub: <http://swat.cse.lehigh.edu/onto/univ-bench.owl#>\n"
+        "SELECT ?x ?y WHERE {\n"
+        "  ?x rdf:type ub:GraduateStudent .\n"
+        "  { ?x ub:memberOf ?y . } UNION { ?x ub:undergraduateDegreeFrom ?y . }\n"
+        "}")
+    plan2 = planner.plan_text(store, text2, vocab)
+    assert len(plan2.unions) == 2 and not plan2.optional
+    got2 = ora.run_query(plan2)
+    assert len(got2) > 0

@@ -124,4 +124,5 @@ def plan_text(store, text, vocab, **kw):
     from . import sparql
     p = sparql.parse(text, vocab)
     return plan_patterns(store, p.patterns, p.nvars, p.required_vars,
-                         distinct=p.distinct, limit=p.limit, offset=p.offset)
+                         distinct=p.distinct, limit=p.limit, offset=p.offset,
+                         optional=p.optional, unions=p.unions)

@@ -80,6 +80,22 @@ def parse(text, vocab, plan_lines=None, distinct=None, limit=None,
         distinct = bool(m.group(1))
     sel = m.group(2).split()
     body = m.group(3)
+
+    # OPTIONAL { ... } group and { A } UNION { B } branches (SPARQL
+    # subset matching the engine's support: one OPTIONAL BGP, one
+    # UNION chain — query.hpp:708-733)
+    opt_body = None
+    union_bodies = []
+    om2 = re.search(r"OPTIONAL\s*\{([^{}]*)\}", body, re.I | re.S)
+    if om2:
+        opt_body = om2.group(1)
+        body = body[:om2.start()] + body[om2.end():]
+    um = re.search(
+        r"\{([^{}]*)\}(\s*UNION\s*\{([^{}]*)\})+", body, re.I | re.S)
+    if um:
+        whole = um.group(0)
+        union_bodies = re.findall(r"\{([^{}]*)\}", whole)
+        body = body[:um.start()] + body[um.end():]
     lm = re.search(r"LIMIT\s+(\d+)", text, re.I)
     om = re.search(r"OFFSET\s+(\d+)", text, re.I)
     if limit is None:
@@ -110,31 +126,35 @@ def parse(text, vocab, plan_lines=None, distinct=None, limit=None,
             raise ParseError(f"unknown IRI {tok!r}")
         raise ParseError(f"cannot resolve term {tok!r}")
 
-    raw = []  # (s, p, o) textual patterns, in order
-    cur = []
-    for tok in body.split():
-        # '.' separates patterns (IRIs contain dots, so split on the
-        # standalone token, not on the character)
-        if tok == ".":
-            if cur:
-                if len(cur) != 3:
-                    raise ParseError(f"expected triple pattern, got {cur!r}")
-                raw.append(tuple(cur))
-                cur = []
-            continue
-        if tok.endswith(".") and not tok.endswith(">."):
-            cur.append(tok[:-1])
-            tok = "."
-            if len(cur) != 3:
-                raise ParseError(f"expected triple pattern, got {cur!r}")
-            raw.append(tuple(cur))
-            cur = []
-            continue
-        cur.append(tok)
-    if cur:
-        if len(cur) != 3:
-            raise ParseError(f"expected triple pattern, got {cur!r}")
-        raw.append(tuple(cur))
+    def split_triples(text_body):
+        out = []
+        cur2 = []
+        for tok in text_body.split():
+            # '.' separates patterns (IRIs contain dots, so split on the
+            # standalone token, not on the character)
+            if tok == ".":
+                if cur2:
+                    if len(cur2) != 3:
+                        raise ParseError(
+                            f"expected triple pattern, got {cur2!r}")
+                    out.append(tuple(cur2))
+                    cur2 = []
+                continue
+            if tok.endswith(".") and not tok.endswith(">."):
+                cur2.append(tok[:-1])
+                if len(cur2) != 3:
+                    raise ParseError(f"expected triple pattern, got {cur2!r}")
+                out.append(tuple(cur2))
+                cur2 = []
+                continue
+            cur2.append(tok)
+        if cur2:
+            if len(cur2) != 3:
+                raise ParseError(f"expected triple pattern, got {cur2!r}")
+            out.append(tuple(cur2))
+        return out
+
+    raw = split_triples(body)
 
     def resolve(i, direction):
         s, p, o = raw[i]
@@ -166,9 +186,17 @@ def parse(text, vocab, plan_lines=None, distinct=None, limit=None,
     # the .fmt 'N <' on "?X rdf:type T" yields (T, TYPE_ID, IN, ?X) which
     # start_from_index() recognises (query.hpp:660-682) — nothing special
     # to do here, the ids compose.
+    def to_pats(text_body):
+        return [(term(a), term(b), DIR_OUT, term(c))
+                for (a, b, c) in split_triples(text_body)]
+
+    optional = to_pats(opt_body) if opt_body else []
+    unions = [to_pats(b) for b in union_bodies] if union_bodies else []
+
     req = [vars_[v] for v in sel]
     return Plan(patterns, nvars=len(vars_), required_vars=req,
-                distinct=distinct, limit=limit, offset=offset)
+                distinct=distinct, limit=limit, offset=offset,
+                optional=optional, unions=unions)
 
 
 def lubm_entity_vocab(store):
