@@ -96,12 +96,6 @@ def test_optional_two_patterns_oracle_vs_numpy(lubm4, oracle4, opt2_plan):
     assert np.array_equal(sort_rows(got), sort_rows(want))
 
 
-def test_union_requires_same_layout(oracle4):
-    # mismatched branch layouts must error, not mis-merge (engine-side
-    # WK_ERR_PLAN; oracle concat would differ -> only engine test on GPU)
-    pass
-
-
 @pytest.mark.gpu
 def test_gpu_union_parity(store4, oracle4, union_plan):
     eng = wk.Engine(store4, device=0)

@@ -1403,6 +1403,11 @@ namespace {
 struct devbuf {
     void *p = nullptr;
     size_t cap = 0;
+    devbuf() = default;
+    devbuf(const devbuf &) = delete;
+    devbuf &operator=(const devbuf &) = delete;
+    ~devbuf() { release(); }  // locals (e.g. run_union's snapshot) free
+                              // on every early-return path
     int ensure(size_t bytes) {
         if (bytes <= cap) return WK_OK;
         if (p) (void)hipFree(p);
