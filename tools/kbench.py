@@ -2,6 +2,10 @@
 Usage: python tools/kbench.py [--univ=2560] q1 q7 q2
 Prints, per pattern step: rows in/out and per-category usec/bytes deltas
 (device-counted algorithmic bytes, HIP-event usec) -> achieved GB/s.
+
+NB: a fused step consumes TWO plan patterns, so subsequent printed
+labels shift by one relative to what actually executed (the engine's
+step counter advances by 2; this tool prints plan.patterns[i]).
 """
 import os
 import sys

@@ -2743,6 +2743,10 @@ extern "C" int32_t wk_engine_graph_build(wk_engine_t *e,
                                          int32_t *gid) {
     if (!e || !plan || !gid) return WK_ERR_STATE;
     if (plan->nopt > 0 || plan->nunion > 0) return WK_ERR_PLAN;
+    // replay returns the BLIND row count (pre-DISTINCT/LIMIT): reject
+    // plans whose count the caller would misread
+    if (plan->distinct || plan->limit >= 0 || plan->offset > 0)
+        return WK_ERR_PLAN;
     // warm pass: settle scratch capacities so capture never allocates
     for (int attempt = 0; attempt < 4; attempt++) {
         int32_t rc = wk_engine_submit(e, plan);
