@@ -126,3 +126,29 @@ def test_gpu_union_optional_combined(store4, oracle4):
     want = oracle4.run_query(plan)
     assert got.shape == want.shape, (got.shape, want.shape)
     assert np.array_equal(sort_rows(got), sort_rows(want))
+
+
+def test_union_then_optional_oracle(lubm4, oracle4):
+    """The reference's execution order (union THEN optional,
+    sparql.hpp:1564-1662) on the CPU oracle."""
+    from wukong_amd import Plan
+    plan = Plan([(Q.GRADSTUDENT, Q.TYPE_ID, wk.DIR_IN, X)], 3, [X, Y, Z],
+                unions=[[(X, Q.MEMBEROF, wk.DIR_OUT, Y)],
+                        [(X, Q.UGDEGREE, wk.DIR_OUT, Y)]],
+                optional=[(X, Q.ADVISOR, wk.DIR_OUT, Z)])
+    got = oracle4.run_query(plan)
+    # independent restatement: union concat, then left-join advisor
+    a = oracle4.run_query(Plan([(Q.GRADSTUDENT, Q.TYPE_ID, wk.DIR_IN, X),
+                                (X, Q.MEMBEROF, wk.DIR_OUT, Y)], 2, [X, Y]))
+    b = oracle4.run_query(Plan([(Q.GRADSTUDENT, Q.TYPE_ID, wk.DIR_IN, X),
+                                (X, Q.UGDEGREE, wk.DIR_OUT, Y)], 2, [X, Y]))
+    adv = tri_map(lubm4, Q.ADVISOR)
+    rows = []
+    for x, y in np.vstack([a, b]):
+        if int(x) in adv:
+            rows += [[x, y, v] for v in adv[int(x)]]
+        else:
+            rows.append([x, y, BLANK])
+    want = np.array(rows, dtype=np.uint32)
+    assert got.shape == want.shape, (got.shape, want.shape)
+    assert np.array_equal(sort_rows(got), sort_rows(want))
