@@ -31,6 +31,20 @@
  *                       reference ran these on CPU, gpu_engine.hpp:63-123)
  *   k_dst_histogram / k_dst_scatter — fork-join split by vid % ndst
  *                       (replaces gpu_hash.cu:600-760)
+ *   k_light_batch     — batched light-query window (one wavefront
+ *                       workgroup per query; proxy.hpp:477-525 window)
+ *   k_plan_batch      — LDS plan interpreter: whole multi-pattern
+ *                       const-start templates, binding table in LDS
+ *   k_vu              — VERSATILE predicate-variable ops over the dense
+ *                       vp CSR (sparql.hpp:556-744)
+ *   k_expand_fn[_map] — functional-predicate dense-map k2u (deg==1
+ *                       segments); _map = optimistic 1:1 inside graphs
+ *   k_expand_opt / k_filter_opt — OPTIONAL-group ops (BLANK fill,
+ *                       matched flags; sparql.hpp:100-170,316-375)
+ *   per-type bitmaps  — rdf:type filters read 1 bit/vid (LLC-resident)
+ *   hipGraph replay   — wk_engine_graph_build/run/launch: a fixed plan
+ *                       replays as ONE launch, with warm-pass-derived
+ *                       no-drop specialization verified on every replay
  */
 #include "wk_store.h"
 #include "../../include/wukong_abi.h"
