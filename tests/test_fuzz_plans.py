@@ -112,3 +112,44 @@ def test_fuzz_gpu_vs_oracle(lubm2, fuzz_store):
             (trial, plan.patterns)
         ran += 1
     assert ran > 30
+
+
+def test_fuzz_watdiv_oracle_vs_brute():
+    """Same fuzz net over the WatDiv schema (different shape: hubs,
+    deep chains, genre fan-outs)."""
+    from tests.oracle_util import OracleCtx
+    from wukong_amd import watdiv as W
+    tri = wk.watdiv_gen(2000, seed=7)
+    store = wk.Store(tri)
+    oc = OracleCtx(tri)
+    preds = [W.HASGENRE, W.OFFER_PRODUCT, W.RETAILER, W.REVIEW_PRODUCT,
+             W.REVIEWER, W.PURCHASED, W.FRIEND]
+    types = [W.T_PRODUCT, W.T_OFFER, W.T_REVIEW, W.T_USER, W.T_GENRE,
+             W.T_RETAILER]
+    rng = random.Random(5)
+    ran = 0
+    for trial in range(40):
+        nv = rng.randint(2, 4)
+        vars_ = [-(i + 1) for i in range(nv)]
+        t = rng.choice(types)
+        pats = [(t, Q.TYPE_ID, wk.DIR_IN, vars_[0])]
+        bound = [vars_[0]]
+        free = vars_[1:]
+        for _ in range(rng.randint(1, 3)):
+            s = rng.choice(bound)
+            if rng.random() < 0.5 and free:
+                o = free.pop(0)
+                pats.append((s, rng.choice(preds), rng.choice([0, 1]), o))
+                bound.append(o)
+            else:
+                pats.append((s, Q.TYPE_ID, wk.DIR_OUT, rng.choice(types)))
+        plan = Plan(pats, nvars=nv, required_vars=bound)
+        try:
+            a = oc.run_query(plan)
+        except RuntimeError:
+            continue
+        b = oc.brute_query(plan)
+        assert a.shape == b.shape, (trial, pats, a.shape, b.shape)
+        assert np.array_equal(sort_rows(a), sort_rows(b)), (trial, pats)
+        ran += 1
+    assert ran > 20
