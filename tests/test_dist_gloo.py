@@ -85,3 +85,27 @@ def test_dist_four_ranks_equal_single():
         want = sort_rows(full.run_query(Q.ALL[name]))
         assert got[name].shape == want.shape, (name, got[name].shape, want.shape)
         assert np.array_equal(got[name], want), name
+
+
+@pytest.mark.timeout(900)
+def test_dist_eight_ranks_equal_single():
+    """world_size=8 — the driver's full-node scaling shape."""
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    port = 29873
+    procs = [ctx.Process(target=_worker, args=(r, 8, port, results))
+             for r in range(8)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=840)
+    for p in procs:
+        p.join(timeout=60)
+
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from tests.oracle_util import OracleCtx, sort_rows
+    full = OracleCtx(wk.lubm_gen(2, seed=42))
+    for name in QUERIES:
+        want = sort_rows(full.run_query(Q.ALL[name]))
+        assert got[name].shape == want.shape, (name, got[name].shape, want.shape)
+        assert np.array_equal(got[name], want), name
