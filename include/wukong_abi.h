@@ -215,6 +215,14 @@ int32_t wk_engine_load_rbuf_device(wk_engine_t *, const wk_sid_t *dev_table,
 int32_t wk_engine_execute_one_pattern(wk_engine_t *, int64_t *nrows_out);
 int32_t wk_engine_pattern_step(const wk_engine_t *);
 int32_t wk_engine_col_num(const wk_engine_t *);
+/* Execute the CURRENT pattern (a mid-plan const-/index-start membership
+ * filter, sparql.hpp:80-186) against a caller-supplied SORTED edge list
+ * instead of the local store.  Distributed driver use: the constant's
+ * edge list lives only on rank `const % nsrv`, so the driver broadcasts
+ * the owner's list first (the reference reads it in place over
+ * one-sided RDMA, gstore.hpp:260-338).  Advances the step. */
+int32_t wk_engine_execute_filter_list(wk_engine_t *, const wk_sid_t *sorted_list,
+                                      uint64_t n, int64_t *nrows_out);
 /* Split the current table by hash of the next pattern's start var
  * (sparql.hpp:746-799; gpu_hash.cu:600-760): fills per-destination row
  * counts and packs rows into dev_out (device buffer, row-major,

@@ -199,6 +199,22 @@ class OracleExecutor:
                             t.shape[0] if t.size else 0,
                             t.shape[1] if t.ndim == 2 else 0, v, step)
 
+    def get_triples(self, vid, pid, d):
+        return self.ctx.get_triples(vid, pid, d)
+
+    def get_index(self, pid, d):
+        return self.ctx.get_index(pid, d)
+
+    def filter_with_list(self, sorted_list, step, v2c, col):
+        """Membership filter of row[col] against a merged (broadcast)
+        edge list, then skip past the pattern — the dist driver's
+        mid-plan const-start path (reference: in-place one-sided read,
+        gstore.hpp:260-338)."""
+        t = self.table()
+        kept = t[np.isin(t[:, col], sorted_list)] if t.size else t
+        self.load(kept, v2c, step + 1)
+        return len(kept)
+
     def step(self):
         return lib().ok_query_step(self._h)
 

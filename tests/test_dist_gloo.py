@@ -40,6 +40,79 @@ def _worker(rank, world, port, results):
         dist.destroy_process_group()
 
 
+def _worker_modifiers(rank, world, port, results):
+    """Plans with DISTINCT/LIMIT/OFFSET and a mid-plan const-start
+    membership filter: final ops must run ONCE after the rank merge
+    (sparql.hpp:1424), and the constant's edge list must be broadcast
+    from its owner rank (gstore.hpp:260-338 one-sided-read analog)."""
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch.distributed as dist
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from wukong_amd.dist import DistQuery
+    from tests.oracle_util import OracleCtx, OracleExecutor, sort_rows
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        triples = wk.lubm_gen(2, seed=42, sid=rank, nsrv=world)
+        ctx = OracleCtx(triples, sid=rank, nsrv=world)
+        out = {}
+        for name, plan in _modifier_plans(Q, wk).items():
+            ex = OracleExecutor(ctx, plan)
+            dq = DistQuery(ex, plan, rank, world)
+            dq.run()
+            out[name] = sort_rows(dq.gather_result())
+        if rank == 0:
+            results.put(out)
+    finally:
+        dist.destroy_process_group()
+
+
+def _modifier_plans(Q, wk):
+    """Shapes the advisor flagged: DISTINCT (cross-rank duplicates),
+    DISTINCT+OFFSET+LIMIT (world*limit rows), and a const-start
+    membership filter after step 0 (owner-only edge list)."""
+    X, Z = -1, -2
+    distinct = wk.Plan([(Q.GRADSTUDENT, Q.TYPE_ID, Q.DIR_IN, X),
+                        (X, Q.MEMBEROF, Q.DIR_OUT, Z)],
+                       nvars=2, required_vars=[Z], distinct=True)
+    dlo = wk.Plan([(Q.GRADSTUDENT, Q.TYPE_ID, Q.DIR_IN, X),
+                   (X, Q.MEMBEROF, Q.DIR_OUT, Z)],
+                  nvars=2, required_vars=[Z], distinct=True,
+                  limit=5, offset=3)
+    cstart = wk.Plan([(Q.FULLPROF, Q.TYPE_ID, Q.DIR_IN, X),
+                      (Q.DEPT0_UNIV0, Q.WORKSFOR, Q.DIR_IN, X)],
+                     nvars=1, required_vars=[X])
+    return {"distinct": distinct, "distinct_lim_off": dlo,
+            "const_mid": cstart}
+
+
+@pytest.mark.timeout(600)
+def test_dist_modifiers_and_const_start():
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    port = 29874
+    procs = [ctx.Process(target=_worker_modifiers, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=540)
+    for p in procs:
+        p.join(timeout=60)
+
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from tests.oracle_util import OracleCtx, sort_rows
+    full = OracleCtx(wk.lubm_gen(2, seed=42))
+    for name, plan in _modifier_plans(Q, wk).items():
+        want = sort_rows(full.run_query(plan))
+        assert got[name].shape == want.shape, (name, got[name].shape, want.shape)
+        assert np.array_equal(got[name], want), name
+
+
 @pytest.mark.timeout(600)
 def test_dist_two_ranks_equal_single():
     ctx = mp.get_context("spawn")

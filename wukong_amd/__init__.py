@@ -107,6 +107,8 @@ _eng_begin = _sig("wk_engine_begin_query", c_i32, [c_vp, ctypes.POINTER(WkPlan)]
 _eng_load = _sig("wk_engine_load_rbuf", c_i32, [c_vp, c_u32p, c_i64, c_i32, ctypes.POINTER(c_i32), c_i32])
 _eng_load_dev = _sig("wk_engine_load_rbuf_device", c_i32, [c_vp, c_vp, c_i64, c_i32, ctypes.POINTER(c_i32), c_i32])
 _eng_step = _sig("wk_engine_execute_one_pattern", c_i32, [c_vp, ctypes.POINTER(c_i64)])
+_eng_filter_list = _sig("wk_engine_execute_filter_list", c_i32,
+                        [c_vp, c_u32p, c_u64, ctypes.POINTER(c_i64)])
 _eng_pattern_step = _sig("wk_engine_pattern_step", c_i32, [c_vp])
 _eng_col_num = _sig("wk_engine_col_num", c_i32, [c_vp])
 _eng_subq = _sig("wk_engine_generate_sub_query", c_i32, [c_vp, c_i32, c_vp, c_i64, ctypes.POINTER(c_i64)])
@@ -470,6 +472,19 @@ class Engine:
         rc = _eng_step(self._h, ctypes.byref(n))
         if rc != 0:
             raise RuntimeError(f"execute_one_pattern rc={rc}")
+        return n.value
+
+    def execute_filter_list(self, sorted_list):
+        """Run the current (mid-plan const-/index-start) pattern as a
+        membership filter against a caller-supplied sorted list — the
+        distributed driver broadcasts the owner rank's edge list first
+        (reference: one-sided remote read, gstore.hpp:260-338)."""
+        lst = np.ascontiguousarray(sorted_list, dtype=np.uint32)
+        n = c_i64()
+        rc = _eng_filter_list(self._h, lst.ctypes.data_as(c_u32p),
+                              len(lst), ctypes.byref(n))
+        if rc != 0:
+            raise RuntimeError(f"execute_filter_list rc={rc}")
         return n.value
 
     @property

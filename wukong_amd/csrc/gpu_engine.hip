@@ -2454,6 +2454,7 @@ extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_
     const int64_t in_rows = e->nrows;
     const int in_cur = e->cur;
     const int in_ncols = e->ncols;
+    const int in_step = e->step;  // a fused step consumes TWO patterns
     const std::vector<int32_t> in_v2c = e->v2c;
     int32_t rc = exec_pattern(e);
     if (rc == WK_OK && nrows_out) {
@@ -2467,7 +2468,7 @@ extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_
             e->nrows = in_rows;
             e->ncols = in_ncols;
             e->v2c = in_v2c;
-            e->step--;
+            e->step = in_step;
             e->bound = in_rows;
             int32_t rc2 = grow_caps(e, need + need / 4, e->cap_cols);
             if (rc2) return rc2;
@@ -2489,6 +2490,48 @@ extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_
 
 extern "C" int32_t wk_engine_pattern_step(const wk_engine_t *e) { return e->step; }
 extern "C" int32_t wk_engine_col_num(const wk_engine_t *e) { return e->ncols; }
+
+// Execute the CURRENT pattern (const- or index-start membership filter,
+// sparql.hpp:80-186) against a CALLER-SUPPLIED sorted edge list instead
+// of the local store.  The distributed driver broadcasts the owner
+// rank's list first (the reference reads it in place over one-sided
+// RDMA, gstore.hpp:260-338): a mid-plan constant's edge list lives only
+// on rank `const % nsrv`, so filtering against the local store would
+// drop every row on the other ranks.
+extern "C" int32_t wk_engine_execute_filter_list(wk_engine_t *e,
+                                                 const sid_t *host_list,
+                                                 uint64_t n,
+                                                 int64_t *nrows_out) {
+    if (!e || e->step >= (int)e->pats.size()) return WK_ERR_STATE;
+    const wk_pattern_t pat = e->pats[e->step];
+    if (pat.subject < 0) return WK_ERR_PLAN;
+    int col = e->var2col(pat.object);
+    if (col < 0) return WK_ERR_PLAN;
+    int32_t rc0 = sync_state(e);  // misc scratch may be in use upstream
+    if (rc0) return rc0;
+    if (e->misc.ensure(std::max<uint64_t>(n * sizeof(sid_t), 4)))
+        return WK_ERR_HIP;
+    if (n && hipMemcpyAsync(e->misc.p, host_list, n * sizeof(sid_t),
+                            hipMemcpyHostToDevice, e->stream) != hipSuccess)
+        return WK_ERR_HIP;
+    sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
+    sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
+    TIME_BEGIN(e);
+    hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
+                       e->stream, e->d_verts, (const sid_t *)e->misc.p, 0, 1,
+                       cur_tbl, e->ncols, col, 0u, pat.direction, PK_NORMAL,
+                       PM_LIST, 0, 0u, 0, n, e->d_type_of, 0, 0, 0,
+                       (const uint64_t *)nullptr, (const sid_t *)nullptr,
+                       0, 0, 0, 0, e->d_state, e->d_stats, out_tbl);
+    TIME_END(e, CAT_FILTER);
+    hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                       (uint64_t)e->cap_rows);
+    e->cur ^= 1;
+    e->step++;
+    int32_t rc = sync_state(e);
+    if (rc == WK_OK && nrows_out) *nrows_out = e->nrows;
+    return rc;
+}
 
 extern "C" int32_t wk_engine_generate_sub_query(wk_engine_t *e, int32_t ndst,
                                                 sid_t *dev_out, int64_t cap_rows,

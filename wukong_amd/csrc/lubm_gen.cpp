@@ -54,6 +54,8 @@ enum : sid_t {
     T_RESEARCHGROUP = 24,
     P_PUBAUTHOR = 25,  // ub:publicationAuthor (emulator template A2)
     T_PUBLICATION = 26,
+    P_MASTERSDEGREE = 27,  // ub:mastersDegreeFrom
+    P_RESEARCHINT = 28,    // ub:researchInterest
 };
 
 struct rng_t {  // splitmix64
@@ -101,7 +103,8 @@ static void gen_university(uint32_t u, uint32_t nuniv, uint64_t seed, emitter &e
         em.emit(dept, P_SUBORG, univ);
         em.emit(dept, P_NAME, alloc());
 
-        // faculty
+        // faculty (real-LUBM degree shape: ug + masters + doctoral for
+        // professors, ug + masters for lecturers; researchInterest all)
         std::vector<sid_t> profs, faculty;
         int nfull = rng.range(7, 11), nassoc = rng.range(10, 15),
             nasst = rng.range(8, 12), nlect = rng.range(5, 8);
@@ -113,6 +116,8 @@ static void gen_university(uint32_t u, uint32_t nuniv, uint64_t seed, emitter &e
             em.emit(p, P_EMAIL, alloc());
             em.emit(p, P_TELEPHONE, alloc());
             em.emit(p, P_UGDEGREE, rand_univ_id());
+            em.emit(p, P_MASTERSDEGREE, rand_univ_id());
+            em.emit(p, P_RESEARCHINT, alloc());
             if (is_prof) {
                 em.emit(p, P_DOCDEGREE, rand_univ_id());
                 profs.push_back(p);
@@ -126,13 +131,16 @@ static void gen_university(uint32_t u, uint32_t nuniv, uint64_t seed, emitter &e
         for (int i = 0; i < nlect; i++) person(T_LECTURER, false);
         em.emit(profs[0], P_HEADOF, dept);
 
-        // courses: each faculty member teaches 1-2; ~1/3 graduate courses
+        // courses: each faculty member teaches 1-2 undergraduate AND 1-2
+        // graduate courses (undergraduate Course count per dept ~54
+        // matches the published LUBM-2560 Q2 #R = 2,765,067 — one Course
+        // row per course, S1C24-LUBM2560-20181203.md:124)
         std::vector<sid_t> courses, gradcourses;
         for (sid_t f : faculty) {
-            int nteach = rng.range(1, 3);
-            for (int i = 0; i < nteach; i++) {
+            int nug_c = rng.range(1, 3), ngrad_c = rng.range(1, 3);
+            for (int i = 0; i < nug_c + ngrad_c; i++) {
                 sid_t c = alloc();
-                bool grad = (rng.next() % 3u) == 0;
+                bool grad = i >= nug_c;
                 em.emit(c, TYPE_ID, grad ? T_GRADCOURSE : T_COURSE);
                 em.emit(c, P_NAME, alloc());
                 em.emit(f, P_TEACHEROF, c);
@@ -150,8 +158,11 @@ static void gen_university(uint32_t u, uint32_t nuniv, uint64_t seed, emitter &e
             em.emit(faculty[0], P_TEACHEROF, c); gradcourses.push_back(c);
         }
 
-        // undergraduate students (NO ugDegreeFrom: Q3 must yield 0 rows)
-        int nug = (int)faculty.size() * rng.range(8, 15);
+        // undergraduate students (NO ugDegreeFrom: Q3 must yield 0 rows);
+        // counts + 2/9 advisor fraction calibrated so Q7 (UG taking a
+        // course taught by their advisor) lands near the published
+        // 112,559 at LUBM-2560
+        int nug = (int)faculty.size() * rng.range(9, 16);
         for (int i = 0; i < nug; i++) {
             sid_t s = alloc();
             em.emit(s, TYPE_ID, T_UGSTUDENT);
@@ -162,7 +173,7 @@ static void gen_university(uint32_t u, uint32_t nuniv, uint64_t seed, emitter &e
             int nc = rng.range(2, 5);
             for (int k = 0; k < nc; k++)
                 em.emit(s, P_TAKESCOURSE, courses[rng.range(0, courses.size())]);
-            if (rng.next() % 5u == 0)  // 1/5 of UG students have an advisor
+            if (rng.next() % 9u < 2)  // 2/9 of UG students have an advisor
                 em.emit(s, P_ADVISOR, profs[rng.range(0, profs.size())]);
         }
 
@@ -182,22 +193,27 @@ static void gen_university(uint32_t u, uint32_t nuniv, uint64_t seed, emitter &e
             em.emit(s, P_ADVISOR, profs[rng.range(0, profs.size())]);
         }
 
-        // publications: professors author 4-9 each, half with one grad
-        // co-author (emulator A2 shape: %AssistantProfessor pubs)
+        // publications (real-LUBM per-rank ranges: FullProf 15-20,
+        // AssocProf 10-17, AsstProf 5-10, Lecturer 0-5), each with a
+        // name literal; half with one professor co-author (emulator A2
+        // shape: %AssistantProfessor pubs)
         {
-            // grads are generated above; collect their ids arithmetically is
-            // fragile — emit pubs from the profs only plus rng-chosen grads
-            // via a second pass is unnecessary: keep co-authors professors.
-            for (sid_t f : profs) {
-                int np = rng.range(4, 10);
+            int fi = 0;
+            auto pubs_for = [&](sid_t f, int lo, int hi) {
+                int np = rng.range(lo, hi);
                 for (int i = 0; i < np; i++) {
                     sid_t pub = alloc();
                     em.emit(pub, TYPE_ID, T_PUBLICATION);
+                    em.emit(pub, P_NAME, alloc());
                     em.emit(pub, P_PUBAUTHOR, f);
                     if (rng.next() & 1)
                         em.emit(pub, P_PUBAUTHOR, profs[rng.range(0, profs.size())]);
                 }
-            }
+            };
+            for (int i = 0; i < nfull; i++) pubs_for(faculty[fi++], 15, 21);
+            for (int i = 0; i < nassoc; i++) pubs_for(faculty[fi++], 10, 18);
+            for (int i = 0; i < nasst; i++) pubs_for(faculty[fi++], 5, 11);
+            for (int i = 0; i < nlect; i++) pubs_for(faculty[fi++], 0, 6);
         }
 
         // research groups
@@ -225,7 +241,7 @@ extern "C" int64_t wk_lubm_gen(int32_t nuniv, uint64_t seed, int32_t sid,
     std::vector<std::vector<sid_t>> parts(nuniv);
 #pragma omp parallel for schedule(dynamic, 4) num_threads(nthr)
     for (int u = 0; u < nuniv; u++) {
-        parts[u].reserve(140000 * 3 / nsrv + 1024);
+        parts[u].reserve(137000 * 3 / nsrv + 1024);
         lubm::emitter em{parts[u], sid, nsrv};
         lubm::gen_university((uint32_t)u, (uint32_t)nuniv, seed, em);
     }

@@ -264,8 +264,14 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     WK_LOG("[store] idx counts: %.1fs\n", now_s() - t0); t0 = now_s();
     st->nbuckets_main = bucket_cursor;
     st->nbuckets_ext = bucket_cursor / 4 + 1024;
-    st->vertices.alloc((st->nbuckets_main + st->nbuckets_ext) * ASSOC, /*zero=*/true);
-    st->edges.alloc(edge_cursor, /*zero=*/false);
+    if (!st->vertices.alloc((st->nbuckets_main + st->nbuckets_ext) * ASSOC,
+                            /*zero=*/true) ||
+        !st->edges.alloc(edge_cursor, /*zero=*/false)) {
+        WK_LOG("[store] allocation failed (%lu buckets, %lu edges)\n",
+               (unsigned long)st->nbuckets_main, (unsigned long)edge_cursor);
+        delete st;
+        return nullptr;
+    }
     std::atomic<uint64_t> ext_next(st->nbuckets_main);
     WK_LOG("[store] count+alloc: %.1fs (buckets=%lu edges=%lu)\n", now_s() - t0,
            (unsigned long)st->nbuckets_main, (unsigned long)edge_cursor);
@@ -351,16 +357,19 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     if (max_id >= (1u << NBITS_IDX) && max_pid < 0xFFFF) {
         st->type_base = 1u << NBITS_IDX;
         st->type_n = (uint64_t)max_id + 1 - st->type_base;
-        st->type_of.alloc(st->type_n, /*zero=*/true);
+        if (!st->type_of.alloc(st->type_n, /*zero=*/true)) {
+            st->type_n = 0;  // side index optional: probe fallback
+        } else {
 #pragma omp parallel for schedule(static)
-        for (int64_t i = pso_lo[TYPE_ID]; i < pso_lo[TYPE_ID + 1]; i++) {
-            bool first = (i == pso_lo[TYPE_ID]) || (pso[i - 1].s != pso[i].s);
-            bool last = (i + 1 == pso_lo[TYPE_ID + 1]) || (pso[i + 1].s != pso[i].s);
-            if (first && last) {
-                st->type_of[pso[i].s - st->type_base] = (uint16_t)pso[i].o;
-            } else {
-                st->type_of[pso[i].s - st->type_base] = 0xFFFF;
-                st->type_multi = true;
+            for (int64_t i = pso_lo[TYPE_ID]; i < pso_lo[TYPE_ID + 1]; i++) {
+                bool first = (i == pso_lo[TYPE_ID]) || (pso[i - 1].s != pso[i].s);
+                bool last = (i + 1 == pso_lo[TYPE_ID + 1]) || (pso[i + 1].s != pso[i].s);
+                if (first && last) {
+                    st->type_of[pso[i].s - st->type_base] = (uint16_t)pso[i].o;
+                } else {
+                    st->type_of[pso[i].s - st->type_base] = 0xFFFF;
+                    st->type_multi = true;
+                }
             }
         }
     }
@@ -382,7 +391,12 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
                 const bool out = dir == DIR_OUT;
                 const std::vector<triple> &arr = out ? pso : pos;
                 const std::vector<int64_t> &lo = out ? pso_lo : pos_lo;
-                st->vp_off[dir].alloc(st->vp_n + 1, /*zero=*/true);
+                if (!st->vp_off[dir].alloc(st->vp_n + 1, /*zero=*/true)) {
+                    st->vp_n = 0;  // VERSATILE optional: disable on OOM
+                    st->vp_off[0].alloc(0, false);
+                    st->vp_off[1].alloc(0, false);
+                    break;
+                }
                 uint32_t *offp = st->vp_off[dir].data();
                 // count distinct (vid, p) pairs per vid (shifted by 1
                 // for the in-place exclusive scan)
@@ -411,7 +425,12 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
                     st->vp_off[1].alloc(0, false);
                     break;
                 }
-                st->vp_edges[dir].alloc(total, /*zero=*/false);
+                if (!st->vp_edges[dir].alloc(total, /*zero=*/false)) {
+                    st->vp_n = 0;
+                    st->vp_off[0].alloc(0, false);
+                    st->vp_off[1].alloc(0, false);
+                    break;
+                }
                 std::vector<uint32_t> cursor(offp, offp + st->vp_n);
                 // fill: sequential ascending p, parallel over runs (each
                 // vid appears once per slice -> no cursor races)

@@ -20,9 +20,13 @@ struct pod_array {
     pod_array(const pod_array &) = delete;
     pod_array &operator=(const pod_array &) = delete;
     ~pod_array() { free(ptr); }
-    void alloc(size_t count, bool zero) {
+    // returns false on allocation failure (n stays 0; wk_store_build
+    // propagates it as a NULL store instead of faulting in the memset)
+    bool alloc(size_t count, bool zero) {
         free(ptr);
         ptr = (T *)malloc(count * sizeof(T) + 16);
+        n = 0;
+        if (!ptr) return false;
         n = count;
         if (zero) {
 #pragma omp parallel for schedule(static)
@@ -31,6 +35,7 @@ struct pod_array {
                 memset(ptr + i, 0, len * sizeof(T));
             }
         }
+        return true;
     }
     T *data() { return ptr; }
     const T *data() const { return ptr; }

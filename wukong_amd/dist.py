@@ -93,6 +93,24 @@ class DistQuery:
         local_var = None
         nccl = dist.is_initialized() and dist.get_backend() == "nccl"
         for i, (s, p, d, o) in enumerate(plan.patterns):
+            if i > 0 and s >= 0:
+                # mid-plan const-/index-start membership filter: the
+                # start's edge list lives only on its owner rank
+                # (`s % world`), so filtering against the local store
+                # would drop every row elsewhere.  Broadcast the owner's
+                # list and filter once against the merged set — the
+                # reference reads it in place over one-sided RDMA
+                # (gstore.hpp:260-338).
+                local = (ex.get_index(s, d) if _is_tpid(s)
+                         else ex.get_triples(s, p, d))
+                gathered = [None] * self.world
+                dist.all_gather_object(
+                    gathered, np.asarray(local, dtype=np.uint32))
+                merged = np.unique(np.concatenate(gathered))  # sorted
+                v2c_prev, _ = self.states[i - 1]
+                ex.filter_with_list(merged, i, v2c_prev,
+                                    v2c_prev[-(o + 1)])
+                continue
             if i > 0 and s < 0 and local_var != s:
                 # fork-join exchange (need_fork_join, sparql.hpp:802-814;
                 # round 1: no remote-read fallback, always exchange)
@@ -115,12 +133,43 @@ class DistQuery:
                 local_var = o if (s >= 0 and _is_tpid(s)) else None
 
     def gather_result(self):
-        """Final table: per-rank finalize (projection), then gather rows to
-        every rank (rmap merge semantics, rmap.hpp:57-87)."""
-        part = self.ex.finalize()
+        """Final table: gather rank results (rmap merge semantics,
+        rmap.hpp:57-87), with the reference's final ops run ONCE after
+        the merge (final_process, sparql.hpp:1424-1551) — per-rank
+        DISTINCT would keep cross-rank duplicates; per-rank LIMIT/OFFSET
+        would return world*limit / drop world*offset rows."""
+        p = self.plan
+        if not (p.distinct or p.limit >= 0 or p.offset > 0):
+            part = self.ex.finalize()  # projection only — order-free
+            gathered = [None] * self.world
+            dist.all_gather_object(gathered, part)
+            return np.concatenate([g for g in gathered if g.size] or [part],
+                                  axis=0)
+        raw = self.ex.table()
         gathered = [None] * self.world
-        dist.all_gather_object(gathered, part)
-        return np.concatenate([g for g in gathered if g.size] or [part], axis=0)
+        dist.all_gather_object(gathered, raw)
+        full = np.concatenate([g for g in gathered if g.size] or [raw], axis=0)
+        v2c, _ = self.states[-1]
+        return final_process(full, v2c, p)
+
+
+def final_process(table, v2c, plan):
+    """The reference's final ops (final_process, sparql.hpp:1424-1551):
+    DISTINCT = full-row sort + adjacent equal-on-required-vars removal,
+    then OFFSET, LIMIT, projection to the required-var columns."""
+    cols = [v2c[-(v + 1)] for v in plan.required_vars]
+    t = np.asarray(table)
+    if plan.distinct and len(t):
+        t = t[np.lexsort(t.T[::-1])]
+        reqt = t[:, cols]
+        keep = np.ones(len(t), dtype=bool)
+        keep[1:] = (reqt[1:] != reqt[:-1]).any(axis=1)
+        t = t[keep]
+    if plan.offset > 0:
+        t = t[plan.offset:]
+    if plan.limit >= 0:
+        t = t[:plan.limit]
+    return np.ascontiguousarray(t[:, cols])
 
 
 def engine_rows(engine):
@@ -138,6 +187,20 @@ class GpuExecutor:
         self.engine.begin_query(plan)
         self.engine.last_rows = 0
         self._host_table = None
+        # host store (edge-list reads for the const-start broadcast)
+        s = wk_engine._store
+        self.store = s if hasattr(s, "get_triples") else s._store
+
+    def get_triples(self, vid, pid, d):
+        return self.store.get_triples(vid, pid, d)
+
+    def get_index(self, pid, d):
+        return self.store.get_index(pid, d)
+
+    def filter_with_list(self, sorted_list, step, v2c, col):
+        n = self.engine.execute_filter_list(sorted_list)
+        self.engine.last_rows = n
+        return n
 
     def load(self, table, v2c, step):
         self.engine.load_rbuf(table, v2c, step)
