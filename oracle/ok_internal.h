@@ -29,12 +29,22 @@ struct triple { sid_t s, p, o; };
 struct ctx {
     std::vector<triple> triples;  // dedup'd pso-side input (brute.cpp)
     std::vector<sid_t> edges;     // all edge lists
-    std::unordered_map<uint64_t, std::pair<uint64_t, uint64_t>> kv;
+    // kv sharded by a key mix so the build can insert shards in
+    // parallel (bench.py's cpu_baseline builds this at LUBM-2560);
+    // lookup semantics unchanged
+    static constexpr int NSHARD = 64;
+    std::unordered_map<uint64_t, std::pair<uint64_t, uint64_t>> kv[NSHARD];
     int sid = 0, nsrv = 1;
 
+    static inline int shard_of(uint64_t key) {
+        return (int)((key * 0x9e3779b97f4a7c15ull) >> 58);
+    }
+
     const sid_t *get(uint64_t vid, uint64_t pid, int dir, uint64_t *sz) const {
-        auto it = kv.find(key_pack(vid, pid, (uint64_t)dir));
-        if (it == kv.end()) { *sz = 0; return nullptr; }
+        uint64_t k = key_pack(vid, pid, (uint64_t)dir);
+        auto &m = kv[shard_of(k)];
+        auto it = m.find(k);
+        if (it == m.end()) { *sz = 0; return nullptr; }
         *sz = it->second.second;
         return edges.data() + it->second.first;
     }

@@ -61,23 +61,74 @@ static inline uint64_t hash_u64(uint64_t key) {
     return key;
 }
 
+// parallel merge sort (chunk std::sort + pairwise inplace_merge): same
+// comparator, same total order as a single std::sort
+template <class T, class Cmp>
+static void par_sort(std::vector<T> &v, Cmp cmp) {
+    const size_t n = v.size();
+    int chunks = omp_get_max_threads();
+    if (chunks <= 1 || n < (1u << 16)) {
+        std::sort(v.begin(), v.end(), cmp);
+        return;
+    }
+    std::vector<size_t> bnd(chunks + 1);
+    for (int i = 0; i <= chunks; i++) bnd[i] = n * (size_t)i / chunks;
+#pragma omp parallel for schedule(dynamic, 1)
+    for (int i = 0; i < chunks; i++)
+        std::sort(v.begin() + bnd[i], v.begin() + bnd[i + 1], cmp);
+    for (int step = 1; step < chunks; step *= 2) {
+#pragma omp parallel for schedule(dynamic, 1)
+        for (int i = 0; i < chunks; i += 2 * step) {
+            int mid = i + step, end = std::min(i + 2 * step, chunks);
+            if (mid < end)
+                std::inplace_merge(v.begin() + bnd[i], v.begin() + bnd[mid],
+                                   v.begin() + bnd[end], cmp);
+        }
+    }
+}
+
 // ---- build (loader + static store semantics) -------------------------
+// The build phase is OpenMP-parallel (partition, sort, fills) purely for
+// wall-clock — bench.py's cpu_baseline leg builds this store at full
+// LUBM-2560 scale.  Query semantics and all stored orders are identical
+// to the sequential form.
 static ctx *build(const sid_t *spo, int64_t n, int sid, int nsrv) {
     ctx *c = new ctx();
     c->sid = sid; c->nsrv = nsrv;
 
     std::vector<triple> pso, pos;
-    for (int64_t i = 0; i < n; i++) {
-        triple t{spo[3 * i], spo[3 * i + 1], spo[3 * i + 2]};
-        // partition — base_loader.hpp:344-352
-        if ((int)(t.s % (sid_t)nsrv) == sid) pso.push_back(t);
-        if ((int)(t.o % (sid_t)nsrv) == sid) pos.push_back(t);
+    {
+        int nt = omp_get_max_threads();
+        std::vector<std::vector<triple>> lso(nt), los(nt);
+#pragma omp parallel num_threads(nt)
+        {
+            int t = omp_get_thread_num();
+            lso[t].reserve((size_t)(n / nt) + 64);
+            los[t].reserve((size_t)(n / nt) + 64);
+#pragma omp for schedule(static)
+            for (int64_t i = 0; i < n; i++) {
+                triple tr{spo[3 * i], spo[3 * i + 1], spo[3 * i + 2]};
+                // partition — base_loader.hpp:344-352
+                if ((int)(tr.s % (sid_t)nsrv) == sid) lso[t].push_back(tr);
+                if ((int)(tr.o % (sid_t)nsrv) == sid) los[t].push_back(tr);
+            }
+        }
+        size_t tso = 0, tos = 0;
+        for (int t = 0; t < nt; t++) { tso += lso[t].size(); tos += los[t].size(); }
+        pso.resize(tso); pos.resize(tos);
+        size_t oso = 0, oos = 0;
+        for (int t = 0; t < nt; t++) {
+            memcpy(pso.data() + oso, lso[t].data(), lso[t].size() * sizeof(triple));
+            memcpy(pos.data() + oos, los[t].data(), los[t].size() * sizeof(triple));
+            oso += lso[t].size(); oos += los[t].size();
+        }
     }
-    // sort + dedup — base_loader.hpp:367-377
-    std::sort(pso.begin(), pso.end(), [](const triple &a, const triple &b) {
+    // sort + dedup — base_loader.hpp:367-377 (parallel merge sort: same
+    // comparator, same total order as std::sort)
+    par_sort(pso, [](const triple &a, const triple &b) {
         return a.p != b.p ? a.p < b.p : (a.s != b.s ? a.s < b.s : a.o < b.o);
     });
-    std::sort(pos.begin(), pos.end(), [](const triple &a, const triple &b) {
+    par_sort(pos, [](const triple &a, const triple &b) {
         return a.p != b.p ? a.p < b.p : (a.o != b.o ? a.o < b.o : a.s < b.s);
     });
     auto eq = [](const triple &a, const triple &b) {
@@ -85,78 +136,140 @@ static ctx *build(const sid_t *spo, int64_t n, int sid, int nsrv) {
     };
     pso.erase(std::unique(pso.begin(), pso.end(), eq), pso.end());
     pos.erase(std::unique(pos.begin(), pos.end(), eq), pos.end());
-    c->triples = pso;  // brute.cpp uses the pso-side dedup'd set
 
+    // ---- slab geometry -------------------------------------------------
+    // OUT edge lists are exactly pso's object column in order (one edge
+    // per triple), so the OUT slab is [0, |pso|) and a run starting at i
+    // has offset i.  The IN slab skips tpid-object runs; tpid objects
+    // sort FIRST within each predicate slice (ids < 2^17), so each
+    // slice's kept region is contiguous and offsets stay derivable.
+    sid_t max_pid = 1;
+#pragma omp parallel for reduction(max : max_pid) schedule(static)
+    for (int64_t i = 0; i < (int64_t)pos.size(); i++)
+        max_pid = std::max(max_pid, pos[i].p);
+    const size_t NPD = (size_t)max_pid + 1;
+    std::vector<size_t> slice_lo(NPD + 1, 0), skip(NPD, 0), seg(NPD + 1, 0);
+    {
+        size_t i = 0;
+        for (size_t p = 0; p < NPD; p++) {
+            slice_lo[p] = i;
+            while (i < pos.size() && pos[i].p == (sid_t)p) i++;
+        }
+        slice_lo[NPD] = pos.size();
+        for (size_t p = 0; p < NPD; p++) {
+            size_t lo = slice_lo[p], hi = slice_lo[p + 1];
+            size_t s2 = lo;
+            while (s2 < hi && is_tpid(pos[s2].o)) s2++;
+            skip[p] = s2 - lo;
+            seg[p + 1] = seg[p] + (hi - s2);
+        }
+    }
+    const size_t out_n = pso.size(), in_n = seg[NPD];
+    c->edges.resize(out_n + in_n);
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < (int64_t)out_n; i++) c->edges[i] = pso[i].o;
+#pragma omp parallel for schedule(dynamic)
+    for (size_t p = 0; p < NPD; p++)
+        for (size_t i = slice_lo[p] + skip[p]; i < slice_lo[p + 1]; i++)
+            c->edges[out_n + seg[p] + (i - slice_lo[p] - skip[p])] = pos[i].s;
+
+    // ---- kv inserts, parallel over shards ------------------------------
+    // Each shard task scans the run starts and keeps only its keys; a
+    // run's offset is derivable from its start index (above), so shards
+    // are fully independent.  Same (key -> off,len) map as the
+    // sequential form.
+    auto insert_runs = [&](bool out) {
+        const std::vector<triple> &arr = out ? pso : pos;
+        const size_t base = out ? 0 : out_n;
+#pragma omp parallel for schedule(dynamic, 1)
+        for (int sh = 0; sh < ctx::NSHARD; sh++) {
+            auto &m = c->kv[sh];
+            for (size_t i = 0; i < arr.size();) {
+                sid_t v = out ? arr[i].s : arr[i].o;
+                size_t j = i + 1;
+                while (j < arr.size() && arr[j].p == arr[i].p &&
+                       (out ? arr[j].s : arr[j].o) == v)
+                    j++;
+                if (!out && is_tpid(v)) { i = j; continue; }
+                uint64_t key = key_pack(v, arr[i].p, (uint64_t)(out ? DIR_OUT : DIR_IN));
+                if (ctx::shard_of(key) == sh) {
+                    size_t p = arr[i].p;
+                    size_t off = out ? i
+                                     : base + seg[p] + (i - slice_lo[p] - skip[p]);
+                    m[key] = {off, j - i};
+                }
+                i = j;
+            }
+        }
+    };
+    insert_runs(true);   // OUT keys [s|p|OUT] — static_gstore.hpp:95-117
+    insert_runs(false);  // IN keys [o|p|IN], tpid objects skipped — :125-152
+
+    // ---- index keys + VERSATILE ----------------------------------------
+    // pidx/tidx (gstore.hpp:858-888) are tiny (one list per predicate /
+    // type): sequential run scan, no edge copies
     auto put = [&](uint64_t vid, uint64_t pid, int dir, const std::vector<sid_t> &vals) {
-        c->kv[key_pack(vid, pid, (uint64_t)dir)] = {c->edges.size(), vals.size()};
+        uint64_t key = key_pack(vid, pid, (uint64_t)dir);
+        c->kv[ctx::shard_of(key)][key] = {c->edges.size(), vals.size()};
         c->edges.insert(c->edges.end(), vals.begin(), vals.end());
     };
-
-    // OUT keys [s|p|OUT] from pso runs — static_gstore.hpp:95-117; also
-    // collect pidx_in (subjects) and tidx (type members) — gstore.hpp:858-888
     std::unordered_map<sid_t, std::vector<sid_t>> pidx_in, pidx_out, tidx;
+    std::vector<std::pair<sid_t, sid_t>> vpo, vpi;  // (vid, pid) per run
+    vpo.reserve(out_n / 2); vpi.reserve(in_n / 2);
     for (size_t i = 0; i < pso.size();) {
         size_t j = i + 1;
         while (j < pso.size() && pso[j].p == pso[i].p && pso[j].s == pso[i].s) j++;
-        std::vector<sid_t> vals;
-        for (size_t k = i; k < j; k++) vals.push_back(pso[k].o);
-        put(pso[i].s, pso[i].p, DIR_OUT, vals);
+        vpo.push_back({pso[i].s, pso[i].p});
         if (pso[i].p == TYPE_ID)
-            for (sid_t t : vals) tidx[t].push_back(pso[i].s);
+            for (size_t k = i; k < j; k++) tidx[pso[k].o].push_back(pso[i].s);
         else if (pso[i].p != PREDICATE_ID)
             pidx_in[pso[i].p].push_back(pso[i].s);
         i = j;
     }
-    // IN keys [o|p|IN] from pos runs, skipping tpid objects —
-    // static_gstore.hpp:125-152; collect pidx_out (objects)
     for (size_t i = 0; i < pos.size();) {
         size_t j = i + 1;
         while (j < pos.size() && pos[j].p == pos[i].p && pos[j].o == pos[i].o) j++;
         if (!is_tpid(pos[i].o)) {
-            std::vector<sid_t> vals;
-            for (size_t k = i; k < j; k++) vals.push_back(pos[k].s);
-            put(pos[i].o, pos[i].p, DIR_IN, vals);
+            vpi.push_back({pos[i].o, pos[i].p});
             if (pos[i].p != PREDICATE_ID && pos[i].p != TYPE_ID)
                 pidx_out[pos[i].p].push_back(pos[i].o);
         }
         i = j;
     }
-    // index keys — insert_idx, static_gstore.hpp:217-280:
-    //   [0|pid|IN] = subjects, [0|pid|OUT] = objects, [0|tid|IN] = members
+    // [0|pid|IN] = subjects, [0|pid|OUT] = objects, [0|tid|IN] = members
     for (auto &kvp : pidx_in) put(0, kvp.first, DIR_IN, kvp.second);
     for (auto &kvp : pidx_out) put(0, kvp.first, DIR_OUT, kvp.second);
     for (auto &kvp : tidx) put(0, kvp.first, DIR_IN, kvp.second);
     // VERSATILE per-vertex predicate lists [vid|PREDICATE_ID|dir] —
     // insert_vp, static_gstore.hpp:282-374: OUT from pso runs INCLUDING
-    // type triples; IN from pos runs skipping tpid objects.  Runs are
-    // iterated in (p, vid) order, so each list is ascending-pid.
-    {
-        std::unordered_map<sid_t, std::vector<sid_t>> vp_out, vp_in;
-        for (size_t i = 0; i < pso.size();) {
-            size_t j = i + 1;
-            while (j < pso.size() && pso[j].p == pso[i].p &&
-                   pso[j].s == pso[i].s)
-                j++;
-            vp_out[pso[i].s].push_back(pso[i].p);
-            i = j;
-        }
-        for (size_t i = 0; i < pos.size();) {
-            size_t j = i + 1;
-            while (j < pos.size() && pos[j].p == pos[i].p &&
-                   pos[j].o == pos[i].o)
-                j++;
-            if (!is_tpid(pos[i].o)) vp_in[pos[i].o].push_back(pos[i].p);
-            i = j;
-        }
-        for (auto &kvp : vp_out) {
-            std::sort(kvp.second.begin(), kvp.second.end());
-            put(kvp.first, PREDICATE_ID, DIR_OUT, kvp.second);
-        }
-        for (auto &kvp : vp_in) {
-            std::sort(kvp.second.begin(), kvp.second.end());
-            put(kvp.first, PREDICATE_ID, DIR_IN, kvp.second);
+    // type triples; IN from pos runs skipping tpid objects.  One
+    // (vid, pid) pair per run; (vid, pid)-sorting yields each vid's
+    // predicate list ascending — the same lists the per-vid-map + sort
+    // form produced.  One edge per pair, so a vid run starting at k has
+    // offset vp_base + k (shard-parallel inserts as above).
+    for (int dir = 0; dir < 2; dir++) {
+        auto &vp = dir == DIR_OUT ? vpo : vpi;
+        par_sort(vp, [](const std::pair<sid_t, sid_t> &a,
+                        const std::pair<sid_t, sid_t> &b) { return a < b; });
+        const size_t vp_base = c->edges.size();
+        c->edges.resize(vp_base + vp.size());
+#pragma omp parallel for schedule(static)
+        for (int64_t k = 0; k < (int64_t)vp.size(); k++)
+            c->edges[vp_base + k] = vp[k].second;
+#pragma omp parallel for schedule(dynamic, 1)
+        for (int sh = 0; sh < ctx::NSHARD; sh++) {
+            auto &m = c->kv[sh];
+            for (size_t i = 0; i < vp.size();) {
+                size_t j = i + 1;
+                while (j < vp.size() && vp[j].first == vp[i].first) j++;
+                uint64_t key =
+                    key_pack(vp[i].first, PREDICATE_ID, (uint64_t)dir);
+                if (ctx::shard_of(key) == sh) m[key] = {vp_base + i, j - i};
+                i = j;
+            }
         }
     }
+    c->triples = std::move(pso);  // brute.cpp uses the pso-side dedup'd set
     return c;
 }
 
