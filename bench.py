@@ -33,25 +33,30 @@ def pctl(xs, p):
     return float(np.percentile(np.asarray(xs), p)) if xs else None
 
 
-def cpu_baseline(nuniv_workload, seed):
+def cpu_baseline(nuniv_workload, seed, triples=None):
     """Oracle engine (DESIGN.md §4 — 'port' of the reference CPU engine)
-    timed on this box's host cores, bounded sample (~10-30 s CPU)."""
+    timed on this box's host cores, SAME inputs as the GPU workload
+    (LUBM-2560 by default), bounded passes (~10-20 s; build excluded).
+    WK_CPU_UNIV overrides the sample scale (debug)."""
     import multiprocessing
     from tests.oracle_util import OracleCtx
     import wukong_amd as wk
     from wukong_amd import queries as Q
 
     cores = multiprocessing.cpu_count()
-    nuniv = int(os.environ.get("WK_CPU_UNIV", "256"))
+    nuniv = int(os.environ.get("WK_CPU_UNIV", str(nuniv_workload)))
     t0 = time.time()
-    triples = wk.lubm_gen(nuniv, seed=seed)
+    if triples is None or nuniv != nuniv_workload:
+        triples = wk.lubm_gen(nuniv, seed=seed)
     ctx = OracleCtx(triples)
     build_s = time.time() - t0
-    # one warm pass, then timed passes until ~10 s
+    # one warm pass, then timed passes until ~10 s (min 1)
+    t0 = time.time()
     for plan in Q.ALL.values():
         ctx.run_query(plan, mt=cores)
+    warm_s = time.time() - t0
     passes, t0 = 0, time.time()
-    while time.time() - t0 < 10.0:
+    while passes == 0 or time.time() - t0 < 10.0:
         for plan in Q.ALL.values():
             ctx.run_query(plan, mt=cores)
         passes += 1
@@ -62,10 +67,13 @@ def cpu_baseline(nuniv_workload, seed):
         "unit": "queries/s",
         "cores": cores,
         "kind": "port",
-        "sample": (f"LUBM-{nuniv} (workload is LUBM-{nuniv_workload}), "
-                   f"Q1-Q7 suite x{passes} in {secs:.1f}s, oracle engine "
-                   f"(restated reference CPU engine), mt={cores}; "
-                   f"oracle build {build_s:.1f}s excluded"),
+        "sample": (f"LUBM-{nuniv}"
+                   + ("" if nuniv == nuniv_workload else
+                      f" (workload is LUBM-{nuniv_workload})")
+                   + f", same seeded inputs, Q1-Q7 suite x{passes} in "
+                     f"{secs:.1f}s, oracle engine (restated reference CPU "
+                     f"engine), mt={cores}; oracle build {build_s:.1f}s + "
+                     f"warm pass {warm_s:.1f}s excluded"),
     }
 
 
@@ -233,7 +241,7 @@ def run_emulator_batched(args, store, engines, inflight):
     latency = {t: {"p50_ms": round(pctl(xs, 50), 3),
                    "p99_ms": round(pctl(xs, 99), 3), "n": len(xs)}
                for t, xs in sorted(lat.items())}
-    out = {
+    return {
         "metric": "light-mix queries/sec (emulator A1-A6, mix_config weights)",
         "value": round(args.emu / elapsed, 1),
         "unit": "queries/s",
@@ -252,8 +260,6 @@ def run_emulator_batched(args, store, engines, inflight):
                                "(proxy.hpp:477-525)"},
         "latency": latency,
     }
-    print(json.dumps(out), flush=True)
-    return True
 
 
 def run_emulator(args, store, engines, inflight):
@@ -369,7 +375,8 @@ def run_emulator(args, store, engines, inflight):
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=4)
+    # default steps size the timed region to >= 1 s (~0.9 ms/pass)
+    ap.add_argument("--steps", type=int, default=1500)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--watdiv", type=int, default=0,
                     help="WatDiv mode: generate N products (~55N triples) and "
@@ -412,7 +419,14 @@ def main():
         f"({time.time()-t0:.1f}s)")
     t0 = time.time()
     store = wk.Store(triples, sid=rank, nsrv=world)
-    del triples
+    # the cpu_baseline leg reuses the SAME generated inputs (north_star:
+    # reference CPU engine timed on the same inputs)
+    keep_triples = (not distributed and not args.watdiv
+                    and os.environ.get("WK_SKIP_CPU_BASELINE") != "1"
+                    and int(os.environ.get("WK_CPU_UNIV", str(nuniv))) == nuniv)
+    if not keep_triples:
+        del triples
+        triples = None
     log(f"[rank {rank}] store: {store.num_slots} slots, {store.num_edges} edges "
         f"({time.time()-t0:.1f}s)")
     t0 = time.time()
@@ -436,7 +450,9 @@ def main():
 
     if args.emu and not distributed:
         if os.environ.get("WK_EMU_MODE", "batched") == "batched":
-            if run_emulator_batched(args, store, engines, inflight):
+            rec = run_emulator_batched(args, store, engines, inflight)
+            if rec:
+                print(json.dumps(rec), flush=True)
                 return
         run_emulator(args, store, engines, inflight)
         return
@@ -541,6 +557,41 @@ def main():
     else:
         run_pipelined(args.warmup)
     sync()
+
+    # at-scale self-parity gates (the gsck idea, gchecker.hpp:364-392):
+    # store integrity scan + per-query counts must agree between the
+    # functional-map and classic dispatch paths AND between graph replay
+    # and submit — asserted on the full workload store every bench run
+    gates = None
+    if not distributed and os.environ.get("WK_SKIP_GATES") != "1":
+        t0 = time.time()
+        viol = store.check()
+        if viol:
+            raise RuntimeError(f"parity gate: store check violations={viol}")
+        gate_counts = {}
+        for name in names:
+            plan = Q_ALL[name]
+            n_fn = eng.run_query_count(plan)
+            os.environ["WK_FN_DISPATCH"] = "0"
+            try:
+                n_classic = eng.run_query_count(plan)
+            finally:
+                os.environ.pop("WK_FN_DISPATCH", None)
+            if n_fn != n_classic:
+                raise RuntimeError(
+                    f"parity gate: {name} fn-dispatch {n_fn} != classic "
+                    f"{n_classic}")
+            if name in graph_ids:
+                n_graph = eng.graph_run(graph_ids[name])
+                if n_graph != n_fn:
+                    raise RuntimeError(
+                        f"parity gate: {name} graph {n_graph} != submit {n_fn}")
+            gate_counts[name] = n_fn
+        gates = {"store_check": "ok", "counts": gate_counts,
+                 "paths": "fn==classic" + ("==graph" if graph_ids else ""),
+                 "secs": round(time.time() - t0, 1)}
+        log(f"[gates] ok: {gates}")
+        sync()
     stats0 = all_stats()
 
     lat = {}
@@ -571,6 +622,22 @@ def main():
     sync()
     elapsed = time.time() - t_start
     stats1 = all_stats()
+
+    # run-to-run spread: repeat the identical timed region (diagnostic
+    # only — `value` comes from the official K-step region above)
+    spread = None
+    if graphs_all and elapsed < 5.0:
+        reps = [elapsed]
+        for _ in range(2):
+            t0 = time.time()
+            for _ in range(args.steps):
+                for name in names:
+                    eng.graph_launch(graph_ids[name])
+                eng.sync()
+            sync()
+            reps.append(time.time() - t0)
+        spread = {"region_s": [round(r, 4) for r in reps],
+                  "pct": round(100 * (max(reps) - min(reps)) / min(reps), 2)}
 
     if distributed:
         t = torch.tensor([elapsed], dtype=torch.float64,
@@ -662,11 +729,29 @@ def main():
         "fn_map_kernel": rl_fn or None,
     }
 
+    # embedded emulator leg (driver-reproducible evidence for the
+    # light-mix claim): short batched run on the same resident store,
+    # reported inside THIS json line.  WK_EMU_EMBED=0 disables.
+    emu_rec = None
+    n_emu = int(os.environ.get("WK_EMU_EMBED", "200000"))
+    if not distributed and not args.watdiv and n_emu > 0:
+        import types
+        log("[rank 0] embedded emulator leg...")
+        emu_engines = [wk.Engine(gstore, device=local_rank) for _ in range(6)]
+        try:
+            emu_rec = run_emulator_batched(
+                types.SimpleNamespace(emu=n_emu), store, emu_engines, 512)
+            if emu_rec is False:
+                emu_rec = None
+        finally:
+            del emu_engines
+
     cb = None
     if (os.environ.get("WK_SKIP_CPU_BASELINE") != "1" and not distributed
             and not args.watdiv):
-        log("[rank 0] timing cpu_baseline (oracle engine)...")
-        cb = cpu_baseline(nuniv, seed)
+        log("[rank 0] timing cpu_baseline (oracle engine, same inputs)...")
+        cb = cpu_baseline(nuniv, seed, triples=triples)
+        del triples
 
     latency = {}
     for name in names:
@@ -701,8 +786,12 @@ def main():
             "parallelism": f"graph-partitioned x{ngpus} + per-step all-to-allv",
             "inflight": inflight,  # reference emulator window (proxy.hpp -p)
         },
+        "timed_region_s": round(elapsed, 4),
+        "spread": spread,
+        "parity_gates": gates,
         "roofline": roofline,
         "cpu_baseline": cb,
+        "emulator": emu_rec,
         "latency": latency,
         "kernel_stats": {k: {"usec": round(stats1[k]["usec"] - stats0[k]["usec"], 1),
                              "launches": int(stats1[k]["launches"] - stats0[k]["launches"])}
