@@ -299,7 +299,9 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                              uint64_t type_base, uint64_t type_n,
                              int use_typeof,
                              const uint64_t *__restrict__ tbm,
-                             const sid_t *__restrict__ fn, uint64_t fn_base,
+                             const fnpage_t *__restrict__ fn_pg,
+                             const sid_t *__restrict__ fn_vals,
+                             uint64_t fn_base,
                              uint64_t fn_n, int fn_swap, int verify_only,
                              uint64_t *__restrict__ d_state,
                              uint64_t *__restrict__ d_stats,
@@ -308,7 +310,7 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     count_bytes(d_stats, CAT_FILTER,
                 (uint64_t)nrows * ((use_typeof ? 6
-                                    : fn ? 8
+                                    : fn_pg ? 24
                                     : probe_mode == PM_LIST ? 12
                                                             : (4 + 128 + 8 + 64)) +
                                    8 * ncols));
@@ -349,17 +351,17 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                 keep[k] = (v == cval);
             } else if (probe_mode == PM_LIST && !use_typeof) {
                 keep[k] = bsearch_u32(edges + list_off, list_sz, v);
-            } else if (fn) {
+            } else if (fn_pg) {
                 // functional predicate: the row's single object replaces
-                // the probe + edge-list search (4 bytes vs 148).
+                // the probe + edge-list search (rank-compressed map:
+                // 16-B page + 4-B value vs 148 bytes of hash traffic).
                 // fn_swap: the REVERSED direction is functional — check
                 // fn[other col] == this col instead.
                 sid_t a = v, b;
                 if (probe_mode == PM_CONST) b = cval;
                 else b = tbl[r * ncols + col2];
                 if (fn_swap) { sid_t t_ = a; a = b; b = t_; }
-                uint64_t idx = (uint64_t)a - fn_base;
-                sid_t tv = (idx < fn_n) ? fn[idx] : 0;
+                sid_t tv = fn_lookup(fn_pg, fn_vals, fn_base, fn_n, a);
                 keep[k] = tv && tv == b;
             } else {
                 uint64_t key = (key_mode == PK_NORMAL)
@@ -601,7 +603,9 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
 // can never overflow.
 template <int NC>
 __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
-                            const sid_t *__restrict__ fn, uint64_t fn_base,
+                            const fnpage_t *__restrict__ fn_pg,
+                            const sid_t *__restrict__ fn_vals,
+                            uint64_t fn_base,
                             uint64_t fn_n, int col, int use_typeof,
                             sid_t fcval,
                             const uint16_t *__restrict__ type_of,
@@ -617,7 +621,7 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     constexpr int oc = NC + 1;
     count_bytes(d_stats, CAT_EXPAND,
-                (uint64_t)nrows * (8 + (use_typeof ? 2 : 0)));
+                (uint64_t)nrows * (24 + (use_typeof ? 2 : 0)));
     // K=4 measured best: K=8 (deeper gather MLP) costs more VGPRs than
     // it hides latency AND halves the tile count (load imbalance) —
     // q1's fn expands regressed 184->216us / 230->341us.  Instead, a
@@ -633,13 +637,28 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
     sid_t pv[K], ptv[K];
     uint16_t pt16[K];
     auto load_tile = [&](int64_t b) {
+        // two gather waves: all K page loads issue together, then the
+        // K dependent value loads — each wave's HBM latency overlaps
+        // across the K independent requests
+        fnpage_t pg[K];
 #pragma unroll
         for (int k = 0; k < K; k++) {
             const int64_t r = b + (int64_t)threadIdx.x * K + k;
             sid_t v = (r < nrows) ? tbl[r * NC + col] : 0;
             pv[k] = v;
             uint64_t idx = (uint64_t)v - fn_base;
-            sid_t tv = (idx < fn_n) ? fn[idx] : 0;
+            pg[k] = (idx < fn_n) ? fn_pg[idx >> 6] : fnpage_t{0, 0, 0};
+        }
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            uint64_t idx = (uint64_t)pv[k] - fn_base;
+            sid_t tv = 0;
+            if (idx < fn_n && ((pg[k].bits >> (idx & 63)) & 1)) {
+                uint32_t rk = pg[k].rank +
+                              (uint32_t)__popcll(pg[k].bits &
+                                                 ((1ull << (idx & 63)) - 1));
+                tv = fn_vals[rk];
+            }
             ptv[k] = tv;
             if (use_typeof) {
                 uint64_t tix = (uint64_t)tv - type_base;
@@ -1350,7 +1369,8 @@ __global__ void k_expand_opt(const vertex_t *__restrict__ verts,
 // 1:1 write needs no scan/compaction and coalesces perfectly.
 template <int NC>
 __global__ void k_expand_fn_map(const sid_t *__restrict__ tbl,
-                                const sid_t *__restrict__ fn,
+                                const fnpage_t *__restrict__ fn_pg,
+                                const sid_t *__restrict__ fn_vals,
                                 uint64_t fn_base, uint64_t fn_n, int col,
                                 int use_typeof,
                                 const uint64_t *__restrict__ tbm,
@@ -1364,13 +1384,12 @@ __global__ void k_expand_fn_map(const sid_t *__restrict__ tbl,
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     constexpr int oc = NC + 1;
     count_bytes(d_stats, CAT_EXPAND,
-                (uint64_t)nrows * (8 + (use_typeof ? 1 : 0) + oc * 4));
+                (uint64_t)nrows * (24 + (use_typeof ? 1 : 0) + oc * 4));
     uint32_t miss = 0;
     for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
          r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
         sid_t v = tbl[r * NC + col];
-        uint64_t idx = (uint64_t)v - fn_base;
-        sid_t tv = (idx < fn_n) ? fn[idx] : 0;
+        sid_t tv = fn_lookup(fn_pg, fn_vals, fn_base, fn_n, v);
         bool ok = tv != 0;
         if (ok && use_typeof) {
             uint64_t tix = (uint64_t)tv - type_base;
@@ -1454,9 +1473,11 @@ struct wk_gpu_store {
     uint32_t *d_vp_off[2] = {nullptr, nullptr};
     sid_t *d_vp_edges[2] = {nullptr, nullptr};
     seg_t *d_segtab = nullptr;
-    // functional-predicate dense maps, indexed [pid*2+dir] (host vector
-    // of device pointers; null = absent)
-    std::vector<sid_t *> d_fn;
+    // functional-predicate rank-compressed maps, indexed [pid*2+dir]
+    // (host vectors of device pointers; null = absent): presence-bitmap
+    // pages + packed values (wk_types.h fnpage_t)
+    std::vector<fnpage_t *> d_fn_pages;
+    std::vector<sid_t *> d_fn_vals;
     // per-type membership bitmaps (LLC-resident typeof filters)
     std::vector<uint64_t *> d_tbm;
     int refs = 0;     // engines attached
@@ -1691,12 +1712,18 @@ extern "C" wk_gpu_store_t *wk_gpu_store_create(const wk_store_t *st, int32_t dev
         }
     }
     if (st->fn_n) {
-        g->d_fn.assign(st->fn.size(), nullptr);
+        g->d_fn_pages.assign(st->fn.size(), nullptr);
+        g->d_fn_vals.assign(st->fn.size(), nullptr);
         for (size_t w = 0; w < st->fn.size(); w++) {
-            if (st->fn[w].empty()) continue;
-            size_t fb = st->fn_n * 4;
-            if (hipMalloc(&g->d_fn[w], fb) != hipSuccess ||
-                hipMemcpy(g->d_fn[w], st->fn[w].data(), fb,
+            if (!st->fn[w].present()) continue;
+            size_t pb = st->fn[w].pages.size() * sizeof(fnpage_t);
+            size_t vb = std::max<size_t>(st->fn[w].vals.size() * 4, 4);
+            if (hipMalloc(&g->d_fn_pages[w], pb) != hipSuccess ||
+                hipMemcpy(g->d_fn_pages[w], st->fn[w].pages.data(), pb,
+                          hipMemcpyHostToDevice) != hipSuccess ||
+                hipMalloc(&g->d_fn_vals[w], vb) != hipSuccess ||
+                hipMemcpy(g->d_fn_vals[w], st->fn[w].vals.data(),
+                          st->fn[w].vals.size() * 4,
                           hipMemcpyHostToDevice) != hipSuccess) {
                 wk_gpu_store_destroy(g);
                 return nullptr;
@@ -1730,7 +1757,9 @@ extern "C" void wk_gpu_store_destroy(wk_gpu_store_t *g) {
         if (g->d_vp_edges[d]) (void)hipFree(g->d_vp_edges[d]);
     }
     if (g->d_segtab) (void)hipFree(g->d_segtab);
-    for (sid_t *p : g->d_fn)
+    for (fnpage_t *p : g->d_fn_pages)
+        if (p) (void)hipFree(p);
+    for (sid_t *p : g->d_fn_vals)
         if (p) (void)hipFree(p);
     for (uint64_t *p : g->d_tbm)
         if (p) (void)hipFree(p);
@@ -1958,14 +1987,15 @@ static void launch_expand_filter(wk_engine *e, const sid_t *cur_tbl,
 
 template <int NC>
 static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
-                               sid_t *out_tbl, const sid_t *d_fn, int col,
+                               sid_t *out_tbl, const fnpage_t *d_pg,
+                               const sid_t *d_vals, int col,
                                bool fuse, sid_t fcval, const seg_t *fseg) {
     const uint64_t *tbm =
         (fuse && e->gs && (size_t)fcval < e->gs->d_tbm.size())
             ? e->gs->d_tbm[fcval]
             : nullptr;
     hipLaunchKernelGGL(k_expand_fn<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
-                       0, e->stream, cur_tbl, d_fn, e->st->fn_base,
+                       0, e->stream, cur_tbl, d_pg, d_vals, e->st->fn_base,
                        e->st->fn_n, col, fuse ? 1 : 0, fcval, e->d_type_of,
                        tbm, e->st->type_base, e->st->type_n, e->d_verts,
                        e->d_edges, fseg ? fseg->bucket_start : 0,
@@ -1975,46 +2005,49 @@ static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
 
 template <int NC>
 static void launch_expand_fn_map_t(wk_engine *e, const sid_t *cur_tbl,
-                                   sid_t *out_tbl, const sid_t *d_fn,
+                                   sid_t *out_tbl, const fnpage_t *d_pg,
+                                   const sid_t *d_vals,
                                    int col, bool fuse, sid_t fcval) {
     const uint64_t *tbm =
         (fuse && e->gs && (size_t)fcval < e->gs->d_tbm.size())
             ? e->gs->d_tbm[fcval]
             : nullptr;
     hipLaunchKernelGGL(k_expand_fn_map<NC>, dim3(grid_for(e->bound)),
-                       dim3(BLOCK), 0, e->stream, cur_tbl, d_fn,
+                       dim3(BLOCK), 0, e->stream, cur_tbl, d_pg, d_vals,
                        e->st->fn_base, e->st->fn_n, col, fuse ? 1 : 0, tbm,
                        e->d_type_of, e->st->type_base, e->st->type_n, fcval,
                        e->d_state, e->d_stats, out_tbl);
 }
 
 static void launch_expand_fn_map(wk_engine *e, const sid_t *cur_tbl,
-                                 sid_t *out_tbl, const sid_t *d_fn, int col,
+                                 sid_t *out_tbl, const fnpage_t *d_pg,
+                                 const sid_t *d_vals, int col,
                                  bool fuse, sid_t fcval) {
     switch (e->ncols) {
-    case 1: launch_expand_fn_map_t<1>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
-    case 2: launch_expand_fn_map_t<2>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
-    case 3: launch_expand_fn_map_t<3>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
-    case 4: launch_expand_fn_map_t<4>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
-    case 5: launch_expand_fn_map_t<5>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
-    case 6: launch_expand_fn_map_t<6>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
-    case 7: launch_expand_fn_map_t<7>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
-    default: launch_expand_fn_map_t<8>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval); break;
+    case 1: launch_expand_fn_map_t<1>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval); break;
+    case 2: launch_expand_fn_map_t<2>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval); break;
+    case 3: launch_expand_fn_map_t<3>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval); break;
+    case 4: launch_expand_fn_map_t<4>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval); break;
+    case 5: launch_expand_fn_map_t<5>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval); break;
+    case 6: launch_expand_fn_map_t<6>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval); break;
+    case 7: launch_expand_fn_map_t<7>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval); break;
+    default: launch_expand_fn_map_t<8>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval); break;
     }
 }
 
 static void launch_expand_fn(wk_engine *e, const sid_t *cur_tbl,
-                             sid_t *out_tbl, const sid_t *d_fn, int col,
+                             sid_t *out_tbl, const fnpage_t *d_pg,
+                             const sid_t *d_vals, int col,
                              bool fuse, sid_t fcval, const seg_t *fseg) {
     switch (e->ncols) {
-    case 1: launch_expand_fn_t<1>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
-    case 2: launch_expand_fn_t<2>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
-    case 3: launch_expand_fn_t<3>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
-    case 4: launch_expand_fn_t<4>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
-    case 5: launch_expand_fn_t<5>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
-    case 6: launch_expand_fn_t<6>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
-    case 7: launch_expand_fn_t<7>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
-    default: launch_expand_fn_t<8>(e, cur_tbl, out_tbl, d_fn, col, fuse, fcval, fseg); break;
+    case 1: launch_expand_fn_t<1>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval, fseg); break;
+    case 2: launch_expand_fn_t<2>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval, fseg); break;
+    case 3: launch_expand_fn_t<3>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval, fseg); break;
+    case 4: launch_expand_fn_t<4>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval, fseg); break;
+    case 5: launch_expand_fn_t<5>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval, fseg); break;
+    case 6: launch_expand_fn_t<6>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval, fseg); break;
+    case 7: launch_expand_fn_t<7>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval, fseg); break;
+    default: launch_expand_fn_t<8>(e, cur_tbl, out_tbl, d_pg, d_vals, col, fuse, fcval, fseg); break;
     }
 }
 
@@ -2201,6 +2234,7 @@ static int32_t exec_pattern(wk_engine *e) {
                            e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
                            off, sz, e->d_type_of, 0, 0, 0,
                            (const uint64_t *)nullptr,
+                           (const fnpage_t *)nullptr,
                            (const sid_t *)nullptr, 0, 0, 0, 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
@@ -2265,26 +2299,26 @@ static int32_t exec_pattern(wk_engine *e) {
         // If only the REVERSED direction is functional, k2k checks
         // fn[other col] == col (fn_swap); k2c resolves the single edge
         // endpoint HOST-side and becomes a pure equality compare (PM_EQ).
-        const sid_t *d_fn = nullptr;
+        const fnpage_t *d_pg = nullptr;
+        const sid_t *d_vals = nullptr;
         int fn_swap = 0;
         if (!use_typeof && key_mode == PK_NORMAL && e->gs &&
-            !e->gs->d_fn.empty() && wk_fn_dispatch()) {
-            d_fn = e->gs->d_fn[(size_t)p * 2 + dir];
+            !e->gs->d_fn_pages.empty() && wk_fn_dispatch()) {
+            d_pg = e->gs->d_fn_pages[(size_t)p * 2 + dir];
+            d_vals = e->gs->d_fn_vals[(size_t)p * 2 + dir];
             // the REVERSED map keys the non-routed endpoint, which is
             // only complete on a single-partition store (the forward map
             // is partitioned on the same axis as the probe routing)
-            if (!d_fn && st->nsrv == 1 &&
+            if (!d_pg && st->nsrv == 1 &&
                 (pmode == PM_COL || pmode == PM_CONST)) {
-                const sid_t *rev = e->gs->d_fn[(size_t)p * 2 + (dir ^ 1)];
-                if (rev && pmode == PM_COL) {
-                    d_fn = rev;
+                const fnpage_t *rpg = e->gs->d_fn_pages[(size_t)p * 2 + (dir ^ 1)];
+                if (rpg && pmode == PM_COL) {
+                    d_pg = rpg;
+                    d_vals = e->gs->d_fn_vals[(size_t)p * 2 + (dir ^ 1)];
                     fn_swap = 1;
-                } else if (rev && pmode == PM_CONST) {
+                } else if (rpg && pmode == PM_CONST) {
                     // host lookup: the const's single neighbour
-                    uint64_t idx = (uint64_t)cval - st->fn_base;
-                    sid_t tv = (idx < st->fn_n)
-                                   ? st->fn[(size_t)p * 2 + (dir ^ 1)][idx]
-                                   : 0;
+                    sid_t tv = st->fn_get((size_t)p * 2 + (dir ^ 1), cval);
                     pmode = PM_EQ;
                     cval = tv;  // 0 never matches a vid -> empty result
                 }
@@ -2295,7 +2329,7 @@ static int32_t exec_pattern(wk_engine *e) {
                            seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
                            dir, key_mode, pmode, col2, cval, 0, 0,
                            e->d_type_of, e->st->type_base, e->st->type_n,
-                           use_typeof, d_tbm, d_fn, st->fn_base, st->fn_n,
+                           use_typeof, d_tbm, d_pg, d_vals, st->fn_base, st->fn_n,
                            fn_swap, verify_only ? 1 : 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
@@ -2312,15 +2346,19 @@ static int32_t exec_pattern(wk_engine *e) {
         int oc = e->ncols + 1;
         if (oc > e->cap_cols) return WK_ERR_STATE;  // begin_query sizes cap_cols
 
-        // FUNCTIONAL predicate (deg==1 dense map): probe+scan+expand
-        // collapse to one gather + block-compacted append, with an
-        // optional fused typeof filter on the new column.  Output rows
-        // <= input rows, so no capacity risk; e->bound is unchanged.
-        const sid_t *d_fn = (key_mode == PK_NORMAL && e->gs &&
-                             !e->gs->d_fn.empty() && wk_fn_dispatch())
-                                ? e->gs->d_fn[(size_t)p * 2 + dir]
-                                : nullptr;
-        if (d_fn) {
+        // FUNCTIONAL predicate (deg==1 rank-compressed map):
+        // probe+scan+expand collapse to a page+value gather pair +
+        // block-compacted append, with an optional fused typeof filter
+        // on the new column.  Output rows <= input rows, so no capacity
+        // risk; e->bound is unchanged.
+        const bool have_fn = key_mode == PK_NORMAL && e->gs &&
+                             !e->gs->d_fn_pages.empty() && wk_fn_dispatch() &&
+                             e->gs->d_fn_pages[(size_t)p * 2 + dir];
+        const fnpage_t *d_pg =
+            have_fn ? e->gs->d_fn_pages[(size_t)p * 2 + dir] : nullptr;
+        const sid_t *d_vals =
+            have_fn ? e->gs->d_fn_vals[(size_t)p * 2 + dir] : nullptr;
+        if (d_pg) {
             bool fuse2 = false;
             sid_t fcval2 = 0;
             const seg_t *fseg2 = nullptr;
@@ -2344,11 +2382,11 @@ static int32_t exec_pattern(wk_engine *e) {
                              e->capture_hint[e->step];
             TIME_BEGIN(e);
             if (opt)
-                launch_expand_fn_map(e, cur_tbl, out_tbl, d_fn, col, fuse2,
-                                     fcval2);
+                launch_expand_fn_map(e, cur_tbl, out_tbl, d_pg, d_vals, col,
+                                     fuse2, fcval2);
             else
-                launch_expand_fn(e, cur_tbl, out_tbl, d_fn, col, fuse2,
-                                 fcval2, fseg2);
+                launch_expand_fn(e, cur_tbl, out_tbl, d_pg, d_vals, col,
+                                 fuse2, fcval2, fseg2);
             TIME_END(e, CAT_EXPAND);
             if (opt)
                 hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0,
@@ -2521,7 +2559,8 @@ extern "C" int32_t wk_engine_execute_filter_list(wk_engine_t *e,
                        e->stream, e->d_verts, (const sid_t *)e->misc.p, 0, 1,
                        cur_tbl, e->ncols, col, 0u, pat.direction, PK_NORMAL,
                        PM_LIST, 0, 0u, 0, n, e->d_type_of, 0, 0, 0,
-                       (const uint64_t *)nullptr, (const sid_t *)nullptr,
+                       (const uint64_t *)nullptr, (const fnpage_t *)nullptr,
+                       (const sid_t *)nullptr,
                        0, 0, 0, 0, e->d_state, e->d_stats, out_tbl);
     TIME_END(e, CAT_FILTER);
     hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,

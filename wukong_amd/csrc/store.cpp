@@ -479,9 +479,11 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
                       [](auto &a, auto &b) { return a.first > b.first; });
             std::vector<uint32_t> picked;
             uint64_t spent = 0;
+            const uint64_t npages = (st->fn_n + 63) / 64;
             for (auto &c : cand) {
-                if (spent + st->fn_n * 4 > budget) break;
-                spent += st->fn_n * 4;
+                uint64_t bytes = npages * sizeof(fnpage_t) + c.first * 4;
+                if (spent + bytes > budget) break;
+                spent += bytes;
                 picked.push_back(c.second);
             }
 #pragma omp parallel for schedule(dynamic)
@@ -490,13 +492,23 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
                 uint32_t p = w / 2;
                 bool out = (w & 1) == DIR_OUT;
                 auto &m = st->fn[w];
-                m.assign(st->fn_n, 0);
+                m.pages.assign(npages, fnpage_t{0, 0, 0});
                 const std::vector<triple> &arr = out ? pso : pos;
                 const std::vector<int64_t> &lo = out ? pso_lo : pos_lo;
+                // keys ascend within the slice (deg==1: one triple per
+                // key), so vals fill in rank order directly
+                m.vals.reserve((size_t)(lo[p + 1] - lo[p]));
                 for (int64_t i = lo[p]; i < lo[p + 1]; i++) {
                     sid_t v = out ? arr[i].s : arr[i].o;
                     if ((uint64_t)v < st->fn_base) continue;
-                    m[v - st->fn_base] = out ? arr[i].o : arr[i].s;
+                    uint64_t idx = v - st->fn_base;
+                    m.pages[idx >> 6].bits |= 1ull << (idx & 63);
+                    m.vals.push_back(out ? arr[i].o : arr[i].s);
+                }
+                uint32_t rank = 0;
+                for (uint64_t g = 0; g < npages; g++) {
+                    m.pages[g].rank = rank;
+                    rank += (uint32_t)__builtin_popcountll(m.pages[g].bits);
                 }
             }
             WK_LOG("[store] fn maps: %zu segments, %.1f GB (%.1fs)\n",
@@ -592,7 +604,8 @@ extern "C" int32_t wk_store_mem_usage(const wk_store_t *st,
     uint64_t side = st->type_n * 2;  // type_of
     for (int d = 0; d < 2; d++)
         side += st->vp_off[d].size() * 4 + st->vp_edges[d].size() * 4;
-    for (auto &m : st->fn) side += m.size() * 4;
+    for (auto &m : st->fn)
+        side += m.pages.size() * sizeof(fnpage_t) + m.vals.size() * 4;
     for (auto &b : st->tbm) side += b.size() * 8;
     if (side_index_bytes) *side_index_bytes = side;
     return 0;

@@ -78,15 +78,28 @@ struct wk_store {
     pod_array<wk::sid_t> vp_edges[2];
     uint64_t vp_base = 0, vp_n = 0;
 
-    // functional-predicate dense maps: for a (pid,dir) segment where
-    // EVERY key has degree exactly 1 (LUBM: memberOf, ugDegreeFrom,
-    // worksFor, advisor, name, ... — detected as keys==edges), a dense
-    // vid→object array replaces the 148-byte cluster-hash probe with one
-    // 4-byte read (0 = vid has no such edge; 0 is never a valid id).
+    // functional-predicate maps: for a (pid,dir) segment where EVERY
+    // key has degree exactly 1 (LUBM: memberOf, ugDegreeFrom, worksFor,
+    // advisor, name, ... — detected as keys==edges), a rank-compressed
+    // map (presence-bitmap pages + packed values, wk_types.h fnpage_t)
+    // replaces the 148-byte cluster-hash probe with a 16-B page load +
+    // 4-B value load (0 = vid has no such edge; 0 is never a valid id).
     // Results identical; an HBM-capacity-funded densification like
     // type_of.  Indexed [pid*2+dir]; empty = absent.
-    std::vector<std::vector<wk::sid_t>> fn;
+    struct fnmap {
+        std::vector<wk::fnpage_t> pages;  // (fn_n+63)/64
+        std::vector<wk::sid_t> vals;      // nkeys, rank-indexed
+        bool present() const { return !pages.empty(); }
+    };
+    std::vector<fnmap> fn;
     uint64_t fn_base = 0, fn_n = 0;
+
+    // host-side lookup (mirrors the device fn_lookup)
+    wk::sid_t fn_get(size_t w, wk::sid_t v) const {
+        if (w >= fn.size() || !fn[w].present()) return 0;
+        return wk::fn_lookup(fn[w].pages.data(), fn[w].vals.data(),
+                             fn_base, fn_n, v);
+    }
 
     // per-(pid,dir) key/edge counts — the planner's cost-model inputs
     // (the reference's type-centric stats, core/optimizer/stats.hpp)
