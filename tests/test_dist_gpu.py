@@ -30,8 +30,13 @@ def _worker(rank, world, port, results):
         store = wk.Store(triples, sid=rank, nsrv=world)
         eng = wk.Engine(store, device=0)
         out = {}
-        for name in QUERIES:
-            plan = Q.ALL[name]
+        plans = dict((n, Q.ALL[n]) for n in QUERIES)
+        # + the advisor-flagged shapes: DISTINCT / LIMIT+OFFSET final
+        # ops once after the merge, and a mid-plan const-start filter
+        # through the device wk_engine_execute_filter_list path
+        from tests.test_dist_gloo import _modifier_plans
+        plans.update(_modifier_plans(Q, wk))
+        for name, plan in plans.items():
             ex = GpuExecutor(eng, plan)
             dq = DistQuery(ex, plan, rank, world)
             dq.run()
@@ -57,9 +62,12 @@ def test_gpu_dist_two_ranks_equal_oracle():
 
     import wukong_amd as wk
     from wukong_amd import queries as Q
+    from tests.test_dist_gloo import _modifier_plans
     from tests.oracle_util import OracleCtx, sort_rows
     full = OracleCtx(wk.lubm_gen(2, seed=42))
-    for name in QUERIES:
-        want = sort_rows(full.run_query(Q.ALL[name]))
+    plans = dict((n, Q.ALL[n]) for n in QUERIES)
+    plans.update(_modifier_plans(Q, wk))
+    for name, plan in plans.items():
+        want = sort_rows(full.run_query(plan))
         assert got[name].shape == want.shape, (name, got[name].shape, want.shape)
         assert np.array_equal(got[name], want), name
