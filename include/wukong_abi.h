@@ -223,6 +223,39 @@ int32_t wk_engine_col_num(const wk_engine_t *);
  * one-sided RDMA, gstore.hpp:260-338).  Advances the step. */
 int32_t wk_engine_execute_filter_list(wk_engine_t *, const wk_sid_t *sorted_list,
                                       uint64_t n, int64_t *nrows_out);
+
+/* ---- xGMI peer mappings (small-table remote reads) ------------------
+ * The reference reads sub-threshold tables' remote edge lists in place
+ * over one-sided RDMA instead of fork-joining (need_fork_join,
+ * core/engine/sparql.hpp:802-814; GStore::get_edges remote leg,
+ * core/store/gstore.hpp:260-338).  The MI355X equivalent: each rank
+ * exports its HBM store image as HIP-IPC handles; every other rank maps
+ * them once over xGMI and k_peer_step probes the owner's store
+ * directly.  The driver exchanges blobs + segment tables host-side
+ * (torch.distributed), then calls import. */
+#define WK_IPC_HANDLE_BYTES 64
+typedef struct {
+    int32_t device, sid, nsrv;
+    int32_t has_type_of;
+    uint64_t type_base, type_n;
+    uint8_t verts_h[WK_IPC_HANDLE_BYTES];
+    uint8_t edges_h[WK_IPC_HANDLE_BYTES];
+    uint8_t type_of_h[WK_IPC_HANDLE_BYTES];
+    int64_t nseg; /* seg-table entries ({bucket_start, num_buckets}) */
+} wk_peer_blob_t;
+
+int32_t wk_gpu_store_export(wk_gpu_store_t *, wk_peer_blob_t *out);
+/* Flattened segment table: nseg x {bucket_start, num_buckets} u64 pairs.
+ * Returns entry count; out=NULL queries the size. */
+int64_t wk_store_seg_table(const wk_store_t *, uint64_t *out,
+                           int64_t cap_entries);
+int32_t wk_gpu_store_import_peers(wk_gpu_store_t *, const wk_peer_blob_t *blobs,
+                                  const uint64_t *segtabs, int64_t nseg,
+                                  int32_t nsrv);
+/* Run the CURRENT pattern via the peer-probe path (tables below the
+ * fork-join threshold).  Rejects const-/index-start and per-row
+ * type-index shapes (WK_ERR_PLAN): callers fall back to the exchange. */
+int32_t wk_engine_execute_one_pattern_remote(wk_engine_t *, int64_t *nrows_out);
 /* Split the current table by hash of the next pattern's start var
  * (sparql.hpp:746-799; gpu_hash.cu:600-760): fills per-destination row
  * counts and packs rows into dev_out (device buffer, row-major,

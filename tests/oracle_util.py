@@ -175,11 +175,15 @@ class OracleCtx:
 class OracleExecutor:
     """Step-level executor over the oracle (the dist-driver interface:
     begin/load/step/table/col_num/step_no/finalize) — CPU stand-in for
-    wukong_amd.Engine in the gloo multi-process tests."""
+    wukong_amd.Engine in the gloo multi-process tests.  `peers` (a
+    rank-ordered list of every partition's OracleCtx) enables the
+    remote-read path: sub-threshold tables probe the owner partition in
+    place, mirroring the GPU k_peer_step semantics."""
 
-    def __init__(self, ctx, plan):
+    def __init__(self, ctx, plan, peers=None):
         self.ctx = ctx
         self.plan = plan
+        self.peers = peers
         self.npat = len(plan.patterns)
         pats = _pats(plan)
         self._keep = pats
@@ -213,6 +217,46 @@ class OracleExecutor:
         t = self.table()
         kept = t[np.isin(t[:, col], sorted_list)] if t.size else t
         self.load(kept, v2c, step + 1)
+        return len(kept)
+
+    def rows(self):
+        data = ctypes.POINTER(ctypes.c_uint32)()
+        return int(lib().ok_query_table(self._h, ctypes.byref(data)))
+
+    def supports_remote(self):
+        return bool(self.peers)
+
+    def step_remote(self, i, pat, v2c_prev, v2c_next):
+        """Per-row probe of the OWNER partition's store (mirrors the GPU
+        k_peer_step): expansion appends the owner's edge list; filters
+        keep rows by owner-list membership.  Tables are sub-threshold,
+        so the Python loop is fine."""
+        s, p, d, o = pat
+        if s >= 0 or (p == 1 and d == 0):  # const start / per-row type index
+            raise ValueError("pattern shape needs the exchange path")
+        t = self.table()
+        col = v2c_prev[-(s + 1)]
+        world = len(self.peers)
+        expand = o < 0 and v2c_prev[-(o + 1)] < 0
+        if expand:
+            out = []
+            for row in t:
+                v = int(row[col])
+                edges = self.peers[v % world].get_triples(v, p, d)
+                for e_ in edges:
+                    out.append(np.append(row, e_))
+            new = (np.array(out, dtype=np.uint32)
+                   if out else np.empty((0, t.shape[1] + 1), dtype=np.uint32))
+            self.load(new, v2c_next, i + 1)
+            return len(new)
+        keep = np.zeros(len(t), dtype=bool)
+        for r, row in enumerate(t):
+            v = int(row[col])
+            edges = self.peers[v % world].get_triples(v, p, d)
+            tgt = o if o >= 0 else int(row[v2c_prev[-(o + 1)]])
+            keep[r] = bool(np.isin(tgt, edges))
+        kept = t[keep] if t.size else t
+        self.load(kept, v2c_prev, i + 1)
         return len(kept)
 
     def step(self):

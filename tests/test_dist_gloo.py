@@ -90,6 +90,69 @@ def _modifier_plans(Q, wk):
             "const_mid": cstart}
 
 
+def _worker_remote(rank, world, port, results):
+    """Small-table remote-read path (need_fork_join threshold,
+    sparql.hpp:802-814): with a huge threshold every sub-query avoids
+    the exchange and probes the owner partition in place (oracle peers
+    = the xGMI stand-in); results must equal both the exchange path and
+    the single-partition oracle."""
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch.distributed as dist
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from wukong_amd.dist import DistQuery
+    from tests.oracle_util import OracleCtx, OracleExecutor, sort_rows
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        ctxs = [OracleCtx(wk.lubm_gen(2, seed=42, sid=r, nsrv=world),
+                          sid=r, nsrv=world) for r in range(world)]
+        ctx = ctxs[rank]
+        out = {}
+        for thr_name, thr in (("remote", 10**9), ("mixed", 300)):
+            for name in QUERIES:
+                plan = Q.ALL[name]
+                ex = OracleExecutor(ctx, plan, peers=ctxs)
+                dq = DistQuery(ex, plan, rank, world, threshold=thr)
+                dq.run()
+                out[f"{thr_name}:{name}"] = sort_rows(dq.gather_result())
+        if rank == 0:
+            results.put(out)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_dist_remote_reads_equal_single():
+    """Both sides of the rdma_threshold gate: threshold=inf forces the
+    in-place remote-read path on every eligible step; threshold=300
+    (the reference default) mixes remote and exchange."""
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    port = 29875
+    procs = [ctx.Process(target=_worker_remote, args=(r, 2, port, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=540)
+    for p in procs:
+        p.join(timeout=60)
+
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from tests.oracle_util import OracleCtx, sort_rows
+    full = OracleCtx(wk.lubm_gen(2, seed=42))
+    for name in QUERIES:
+        want = sort_rows(full.run_query(Q.ALL[name]))
+        for mode in ("remote", "mixed"):
+            g = got[f"{mode}:{name}"]
+            assert g.shape == want.shape, (mode, name, g.shape, want.shape)
+            assert np.array_equal(g, want), (mode, name)
+
+
 @pytest.mark.timeout(600)
 def test_dist_modifiers_and_const_start():
     ctx = mp.get_context("spawn")

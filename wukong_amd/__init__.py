@@ -109,6 +109,27 @@ _eng_load_dev = _sig("wk_engine_load_rbuf_device", c_i32, [c_vp, c_vp, c_i64, c_
 _eng_step = _sig("wk_engine_execute_one_pattern", c_i32, [c_vp, ctypes.POINTER(c_i64)])
 _eng_filter_list = _sig("wk_engine_execute_filter_list", c_i32,
                         [c_vp, c_u32p, c_u64, ctypes.POINTER(c_i64)])
+_eng_step_remote = _sig("wk_engine_execute_one_pattern_remote", c_i32,
+                        [c_vp, ctypes.POINTER(c_i64)])
+
+
+class WkPeerBlob(ctypes.Structure):
+    _fields_ = [("device", c_i32), ("sid", c_i32), ("nsrv", c_i32),
+                ("has_type_of", c_i32),
+                ("type_base", c_u64), ("type_n", c_u64),
+                ("verts_h", ctypes.c_uint8 * 64),
+                ("edges_h", ctypes.c_uint8 * 64),
+                ("type_of_h", ctypes.c_uint8 * 64),
+                ("nseg", c_i64)]
+
+
+_gstore_export = _sig("wk_gpu_store_export", c_i32,
+                      [c_vp, ctypes.POINTER(WkPeerBlob)])
+_seg_table = _sig("wk_store_seg_table", c_i64,
+                  [c_vp, ctypes.POINTER(c_u64), c_i64])
+_gstore_import = _sig("wk_gpu_store_import_peers", c_i32,
+                      [c_vp, ctypes.POINTER(WkPeerBlob),
+                       ctypes.POINTER(c_u64), c_i64, c_i32])
 _eng_pattern_step = _sig("wk_engine_pattern_step", c_i32, [c_vp])
 _eng_col_num = _sig("wk_engine_col_num", c_i32, [c_vp])
 _eng_subq = _sig("wk_engine_generate_sub_query", c_i32, [c_vp, c_i32, c_vp, c_i64, ctypes.POINTER(c_i64)])
@@ -283,6 +304,16 @@ class Store:
         """gsck-style full integrity scan; returns #violations (0 = ok)."""
         return _store_check(self._h)
 
+    def seg_table(self):
+        """Flattened (pid,dir) segment table as an (n,2) uint64 array of
+        {bucket_start, num_buckets} — exchanged between ranks for the
+        xGMI peer-probe path."""
+        n = _seg_table(self._h, None, 0)
+        out = np.zeros((max(int(n), 1), 2), dtype=np.uint64)
+        if n > 0:
+            _seg_table(self._h, out.ctypes.data_as(ctypes.POINTER(c_u64)), n)
+        return out[:max(int(n), 0)]
+
 
 class GpuStore:
     """Device-resident store image shared by multiple engines (one HBM
@@ -298,6 +329,32 @@ class GpuStore:
         if getattr(self, "_h", None):
             _gstore_destroy(self._h)
             self._h = None
+
+    def export_blob(self):
+        """HIP-IPC handles + metadata of this rank's HBM store image
+        (bytes) for the xGMI peer-probe path."""
+        blob = WkPeerBlob()
+        rc = _gstore_export(self._h, ctypes.byref(blob))
+        if rc != 0:
+            raise RuntimeError(f"wk_gpu_store_export rc={rc}")
+        return bytes(blob)
+
+    def import_peers(self, blob_bytes_list, seg_tables):
+        """Open every rank's store over xGMI (HIP IPC).  blob_bytes_list
+        and seg_tables are rank-ordered (this rank's own entries
+        included)."""
+        n = len(blob_bytes_list)
+        blobs = (WkPeerBlob * n)()
+        for i, bb in enumerate(blob_bytes_list):
+            ctypes.memmove(ctypes.byref(blobs[i]), bb, ctypes.sizeof(WkPeerBlob))
+        nseg = len(seg_tables[0])
+        flat = np.ascontiguousarray(
+            np.stack([np.asarray(t, dtype=np.uint64) for t in seg_tables]))
+        rc = _gstore_import(self._h, blobs,
+                            flat.ctypes.data_as(ctypes.POINTER(c_u64)),
+                            nseg, n)
+        if rc != 0:
+            raise RuntimeError(f"wk_gpu_store_import_peers rc={rc}")
 
 
 class Engine:
@@ -472,6 +529,19 @@ class Engine:
         rc = _eng_step(self._h, ctypes.byref(n))
         if rc != 0:
             raise RuntimeError(f"execute_one_pattern rc={rc}")
+        return n.value
+
+    def execute_one_pattern_remote(self):
+        """Run the current pattern via the xGMI peer-probe path (small
+        tables: in-place remote reads instead of an exchange —
+        sparql.hpp:802-814).  Raises ValueError for shapes that must
+        exchange (caller falls back)."""
+        n = c_i64()
+        rc = _eng_step_remote(self._h, ctypes.byref(n))
+        if rc == -3:  # WK_ERR_PLAN: const-start / per-row index shape
+            raise ValueError("pattern shape needs the exchange path")
+        if rc != 0:
+            raise RuntimeError(f"execute_one_pattern_remote rc={rc}")
         return n.value
 
     def execute_filter_list(self, sorted_list):

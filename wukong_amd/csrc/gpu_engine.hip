@@ -1412,6 +1412,101 @@ __global__ void k_expand_fn_map(const sid_t *__restrict__ tbl,
                   (unsigned long long)miss);
 }
 
+// ---------------------------------------------------------------------
+// xGMI peer-probe small-table step — the reference's one-sided in-place
+// remote read (gstore.hpp:260-338), gated by the fork-join threshold
+// (need_fork_join, sparql.hpp:802-814, Global::rdma_threshold=300): a
+// binding table below the threshold probes the OWNER rank's HBM store
+// directly through HIP-IPC peer mappings over xGMI instead of shipping
+// rows with an all-to-allv.  ONE workgroup — tables are tiny, so the
+// block-local scan keeps compaction exact with zero global atomics.
+struct wk_peer_desc {
+    const vertex_t *verts;
+    const sid_t *edges;
+    const uint16_t *type_of;
+    uint64_t type_base, type_n;
+};
+struct segtab8 { uint64_t bs[8]; uint64_t nb[8]; };  // per-rank (pid,dir) segment
+
+__global__ void k_peer_step(const wk_peer_desc *__restrict__ peers, int nsrv,
+                            segtab8 seg,
+                            const sid_t *__restrict__ tbl, int ncols, int col,
+                            uint32_t pid, int dir, int pmode, int col2,
+                            sid_t cval, uint64_t cap,
+                            uint64_t *__restrict__ d_state,
+                            uint64_t *__restrict__ d_stats,
+                            sid_t *__restrict__ out)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_PROBE, (uint64_t)nrows * (4 + 128 + 8));
+    __shared__ uint64_t sh[SCAN_T];
+    const int oc = pmode == PM_SIZE ? ncols + 1 : ncols;
+    uint64_t carry = 0;
+    for (int64_t base = 0; base < nrows; base += SCAN_T) {
+        const int64_t r = base + threadIdx.x;
+        uint64_t deg = 0;  // rows to emit (expand: fanout; filter: 0/1)
+        const sid_t *el = nullptr;
+        if (r < nrows) {
+            sid_t v = tbl[r * ncols + col];
+            int owner = (int)(v % (sid_t)nsrv);
+            const wk_peer_desc P = peers[owner];
+            bool resolved = false;
+            // `?v rdf:type CONST` via the owner's dense side index;
+            // multi-type (0xFFFF) falls through to the owner probe of
+            // the same [v|TYPE|OUT] segment
+            if (pmode == PM_CONST && pid == TYPE_ID && dir == DIR_OUT &&
+                P.type_of) {
+                uint64_t tix = (uint64_t)v - P.type_base;
+                uint16_t t = (tix < P.type_n) ? P.type_of[tix] : 0;
+                if (t != 0xFFFF) {
+                    deg = ((sid_t)t == cval) ? 1 : 0;
+                    resolved = true;
+                }
+            }
+            if (!resolved && seg.nb[owner]) {
+                uint64_t eoff = 0, esz = 0;
+                probe_one(P.verts, seg.bs[owner], seg.nb[owner],
+                          key_pack(v, pid, (uint64_t)dir), eoff, esz);
+                if (pmode == PM_SIZE) {
+                    deg = esz;
+                    el = P.edges + eoff;
+                } else if (esz) {
+                    sid_t tgt = pmode == PM_CONST ? cval
+                                                  : tbl[r * ncols + col2];
+                    deg = bsearch_u32(P.edges + eoff, esz, tgt) ? 1 : 0;
+                }
+            }
+        }
+        sh[threadIdx.x] = deg;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint64_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += x;
+            __syncthreads();
+        }
+        uint64_t pos = carry + sh[threadIdx.x] - deg;
+        carry += sh[SCAN_T - 1];
+        __syncthreads();
+        if (r < nrows && deg) {
+            if (pmode == PM_SIZE) {
+                for (uint64_t k = 0; k < deg; k++) {
+                    uint64_t dst = pos + k;
+                    if (dst >= cap) break;
+                    sid_t *d = out + dst * oc;
+                    for (int c = 0; c < ncols; c++) d[c] = tbl[r * ncols + c];
+                    d[ncols] = el[k];
+                }
+            } else if (pos < cap) {
+                sid_t *d = out + pos * oc;
+                for (int c = 0; c < ncols; c++) d[c] = tbl[r * ncols + c];
+            }
+        }
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) d_state[S_TOTAL] = carry;
+}
+
 // commit for the optimistic map: row count unchanged; any miss flags
 // S_ERR so the replay result is discarded and the safe path re-runs
 __global__ void k_commit_map(uint64_t *__restrict__ d_state) {
@@ -1480,6 +1575,12 @@ struct wk_gpu_store {
     std::vector<sid_t *> d_fn_vals;
     // per-type membership bitmaps (LLC-resident typeof filters)
     std::vector<uint64_t *> d_tbm;
+    // xGMI peer mappings of the other ranks' stores (HIP IPC) — the
+    // small-table remote-read path (k_peer_step); empty = not imported
+    std::vector<wk_peer_desc> peers;            // host copy, [nsrv]
+    std::vector<std::vector<seg_t>> peer_nseg;  // [nsrv][pid*2+dir]
+    wk_peer_desc *d_peers = nullptr;            // device array [nsrv]
+    std::vector<void *> ipc_opened;             // to close on destroy
     int refs = 0;     // engines attached
     bool owned = false;  // created implicitly by wk_engine_create
 };
@@ -1514,6 +1615,7 @@ struct wk_engine {
     int64_t bound = 0;   // host-side upper bound on rows (grid sizing)
 
     bool light = false;  // single-kernel fast path ran (h_pin self-published)
+    int remote_step_idx = -1;  // pattern to run via the xGMI peer path
 
     // OPTIONAL group execution state (wk_engine_run_query only)
     devbuf oflag[2];          // per-row matched flags (u8), ping-pong
@@ -1763,7 +1865,105 @@ extern "C" void wk_gpu_store_destroy(wk_gpu_store_t *g) {
         if (p) (void)hipFree(p);
     for (uint64_t *p : g->d_tbm)
         if (p) (void)hipFree(p);
+    for (void *p : g->ipc_opened)
+        if (p) (void)hipIpcCloseMemHandle(p);
+    if (g->d_peers) (void)hipFree(g->d_peers);
     delete g;
+}
+
+// ---- xGMI peer-mapping export/import (small-table remote reads) ------
+extern "C" int32_t wk_gpu_store_export(wk_gpu_store_t *g,
+                                       wk_peer_blob_t *out) {
+    if (!g || !out) return WK_ERR_STATE;
+    memset(out, 0, sizeof(*out));
+    out->device = g->device;
+    out->sid = g->st->sid;
+    out->nsrv = g->st->nsrv;
+    out->type_base = g->st->type_base;
+    out->type_n = g->st->type_n;
+    out->nseg = (int64_t)g->st->nseg.size();
+    static_assert(sizeof(hipIpcMemHandle_t) <= WK_IPC_HANDLE_BYTES,
+                  "ipc handle size");
+    if (hipIpcGetMemHandle((hipIpcMemHandle_t *)out->verts_h, g->d_verts) !=
+            hipSuccess ||
+        hipIpcGetMemHandle((hipIpcMemHandle_t *)out->edges_h, g->d_edges) !=
+            hipSuccess)
+        return WK_ERR_HIP;
+    if (g->d_type_of) {
+        if (hipIpcGetMemHandle((hipIpcMemHandle_t *)out->type_of_h,
+                               g->d_type_of) != hipSuccess)
+            return WK_ERR_HIP;
+        out->has_type_of = 1;
+    }
+    return WK_OK;
+}
+
+extern "C" int64_t wk_store_seg_table(const wk_store_t *st, uint64_t *out,
+                                      int64_t cap_entries) {
+    if (!st) return -1;
+    int64_t n = (int64_t)st->nseg.size();
+    if (!out) return n;
+    if (cap_entries < n) return -1;
+    for (int64_t w = 0; w < n; w++) {
+        out[2 * w] = st->nseg[w].bucket_start;
+        out[2 * w + 1] = st->nseg[w].num_buckets;
+    }
+    return n;
+}
+
+extern "C" int32_t wk_gpu_store_import_peers(wk_gpu_store_t *g,
+                                             const wk_peer_blob_t *blobs,
+                                             const uint64_t *segtabs,
+                                             int64_t nseg, int32_t nsrv) {
+    if (!g || !blobs || !segtabs || nsrv <= 0 || nsrv > 8) return WK_ERR_STATE;
+    if (hipSetDevice(g->device) != hipSuccess) return WK_ERR_HIP;
+    g->peers.assign(nsrv, wk_peer_desc{});
+    g->peer_nseg.assign(nsrv, {});
+    const int me = g->st->sid;
+    for (int r = 0; r < nsrv; r++) {
+        const wk_peer_blob_t &b = blobs[r];
+        if (b.nseg != nseg) return WK_ERR_STATE;
+        wk_peer_desc &P = g->peers[r];
+        P.type_base = b.type_base;
+        P.type_n = b.type_n;
+        if (r == me) {
+            P.verts = g->d_verts;
+            P.edges = g->d_edges;
+            P.type_of = g->d_type_of;
+        } else {
+            void *pv = nullptr, *pe = nullptr, *pt = nullptr;
+            if (hipIpcOpenMemHandle(&pv, *(const hipIpcMemHandle_t *)b.verts_h,
+                                    hipIpcMemLazyEnablePeerAccess) != hipSuccess)
+                return WK_ERR_HIP;
+            g->ipc_opened.push_back(pv);
+            if (hipIpcOpenMemHandle(&pe, *(const hipIpcMemHandle_t *)b.edges_h,
+                                    hipIpcMemLazyEnablePeerAccess) != hipSuccess)
+                return WK_ERR_HIP;
+            g->ipc_opened.push_back(pe);
+            if (b.has_type_of) {
+                if (hipIpcOpenMemHandle(&pt,
+                                        *(const hipIpcMemHandle_t *)b.type_of_h,
+                                        hipIpcMemLazyEnablePeerAccess) !=
+                    hipSuccess)
+                    return WK_ERR_HIP;
+                g->ipc_opened.push_back(pt);
+            }
+            P.verts = (const vertex_t *)pv;
+            P.edges = (const sid_t *)pe;
+            P.type_of = (const uint16_t *)pt;
+        }
+        g->peer_nseg[r].resize((size_t)nseg);
+        const uint64_t *tab = segtabs + (size_t)r * nseg * 2;
+        for (int64_t w = 0; w < nseg; w++) {
+            g->peer_nseg[r][w].bucket_start = tab[2 * w];
+            g->peer_nseg[r][w].num_buckets = tab[2 * w + 1];
+        }
+    }
+    if (hipMalloc(&g->d_peers, nsrv * sizeof(wk_peer_desc)) != hipSuccess ||
+        hipMemcpy(g->d_peers, g->peers.data(), nsrv * sizeof(wk_peer_desc),
+                  hipMemcpyHostToDevice) != hipSuccess)
+        return WK_ERR_HIP;
+    return WK_OK;
 }
 
 extern "C" wk_engine_t *wk_engine_create_on(wk_gpu_store_t *g) {
@@ -2065,10 +2265,63 @@ static void launch_expand(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
     }
 }
 
+// Remote (xGMI peer-probe) variant of one pattern: small tables probe
+// the owner rank's store in place instead of exchanging rows (the
+// reference's sub-threshold one-sided read, sparql.hpp:802-814 +
+// gstore.hpp:260-338).  Routed via e->remote_step_idx so the overflow
+// re-run stays on the remote path.
+static int32_t exec_pattern_remote(wk_engine *e) {
+    wk_gpu_store *g = e->gs;
+    if (!g || !g->d_peers || g->peer_nseg.empty()) return WK_ERR_STATE;
+    const wk_pattern_t pat = e->pats[e->step];
+    const ssid_t s = pat.subject, p = pat.predicate, o = pat.object;
+    const int dir = pat.direction;
+    if (s >= 0) return WK_ERR_PLAN;  // const/index starts stay local
+    // [0|tid|IN] per-row index expansion fans out over ALL partitions —
+    // not a per-owner read; callers must exchange for this shape
+    if ((sid_t)p == TYPE_ID && dir == DIR_IN) return WK_ERR_PLAN;
+    int col = e->var2col(s);
+    if (col < 0) return WK_ERR_PLAN;
+    const int nsrv = (int)g->peers.size();
+    if (nsrv < 1 || nsrv > 8) return WK_ERR_STATE;
+    segtab8 seg{};
+    const size_t w = (size_t)p * 2 + dir;
+    for (int r = 0; r < nsrv; r++) {
+        if (w < g->peer_nseg[r].size()) {
+            seg.bs[r] = g->peer_nseg[r][w].bucket_start;
+            seg.nb[r] = g->peer_nseg[r][w].num_buckets;
+        }
+    }
+    const int ostat = (o >= 0) ? 2 : (e->var2col(o) >= 0 ? 1 : 0);
+    const int pmode = ostat == 0 ? PM_SIZE : (ostat == 2 ? PM_CONST : PM_COL);
+    if (pmode == PM_SIZE && e->ncols + 1 > e->cap_cols) return WK_ERR_STATE;
+    sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
+    sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
+    TIME_BEGIN(e);
+    hipLaunchKernelGGL(k_peer_step, dim3(1), dim3(SCAN_T), 0, e->stream,
+                       g->d_peers, nsrv, seg, cur_tbl, e->ncols, col,
+                       (uint32_t)p, dir, pmode,
+                       ostat == 1 ? e->var2col(o) : 0,
+                       ostat == 2 ? (sid_t)o : 0, (uint64_t)e->cap_rows,
+                       e->d_state, e->d_stats, out_tbl);
+    TIME_END(e, CAT_PROBE);
+    hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                       (uint64_t)e->cap_rows);
+    if (pmode == PM_SIZE) {
+        e->v2c[-(o + 1)] = e->ncols;
+        e->ncols += 1;
+        e->bound = e->cap_rows;
+    }
+    e->cur ^= 1;
+    e->step++;
+    return WK_OK;
+}
+
 // Run one pattern — dispatch per sparql.hpp:1016-1058.  Fully async: row
 // counts live on device; a non-NULL nrows_out forces a sync (step API).
 static int32_t exec_pattern(wk_engine *e) {
     if (!e || e->step >= (int)e->pats.size()) return WK_ERR_STATE;
+    if (e->step == e->remote_step_idx) return exec_pattern_remote(e);
     const wk_store *st = e->st;
     const wk_pattern_t pat = e->pats[e->step];
     const ssid_t s = pat.subject, p = pat.predicate, o = pat.object;
@@ -2528,6 +2781,19 @@ extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_
 
 extern "C" int32_t wk_engine_pattern_step(const wk_engine_t *e) { return e->step; }
 extern "C" int32_t wk_engine_col_num(const wk_engine_t *e) { return e->ncols; }
+
+// Run the CURRENT pattern via the xGMI peer-probe path (small tables:
+// in-place remote reads instead of an exchange — sparql.hpp:802-814).
+// Requires wk_gpu_store_import_peers.  The overflow re-run inherits the
+// remote routing through e->remote_step_idx.
+extern "C" int32_t wk_engine_execute_one_pattern_remote(wk_engine_t *e,
+                                                        int64_t *nrows_out) {
+    if (!e) return WK_ERR_STATE;
+    e->remote_step_idx = e->step;
+    int32_t rc = wk_engine_execute_one_pattern(e, nrows_out);
+    e->remote_step_idx = -1;
+    return rc;
+}
 
 // Execute the CURRENT pattern (const- or index-start membership filter,
 // sparql.hpp:80-186) against a CALLER-SUPPLIED sorted edge list instead

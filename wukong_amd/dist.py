@@ -11,13 +11,53 @@ its local store: probes of non-local keys miss by construction, so no
 owner special-casing is needed (index/const starts are naturally local,
 dispatch semantics of sparql.hpp:1064-1111).
 """
+import os
+import sys
+
 import numpy as np
 import torch
 import torch.distributed as dist
 
+# the reference's fork-join gate (Global::rdma_threshold=300,
+# need_fork_join sparql.hpp:802-814): tables below this read the owner
+# rank's store in place instead of exchanging
+RDMA_THRESHOLD = int(os.environ.get("WK_RDMA_THRESHOLD", "300"))
+
 
 def _is_tpid(x):
     return 1 < x < (1 << 17)
+
+
+def init_peers(gstore, store):
+    """Exchange HIP-IPC store handles + segment tables between ranks and
+    open xGMI peer mappings — the one-sided remote-read analog
+    (gstore.hpp:260-338).  All ranks must agree on the outcome (the
+    remote/exchange decision gates a collective), so the success bit is
+    min-reduced; on any failure every rank keeps the exchange-only path."""
+    if not dist.is_initialized() or dist.get_world_size() < 2:
+        return False
+    world = dist.get_world_size()
+    ok = 1
+    try:
+        blob = gstore.export_blob()
+        segs = store.seg_table()
+        gathered = [None] * world
+        dist.all_gather_object(gathered, (blob, segs))
+        nseg = len(gathered[0][1])
+        if any(len(g[1]) != nseg for g in gathered):
+            ok = 0
+        else:
+            gstore.import_peers([g[0] for g in gathered],
+                                [g[1] for g in gathered])
+    except Exception as ex:
+        print(f"[dist] peer import unavailable: {ex}", file=sys.stderr)
+        ok = 0
+    dev = "cuda" if dist.get_backend() == "nccl" else "cpu"
+    t = torch.tensor([ok], dtype=torch.int64, device=dev)
+    dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    ready = bool(int(t.item()))
+    gstore._peers_ready = ready
+    return ready
 
 
 def plan_v2c_states(plan):
@@ -46,13 +86,40 @@ class DistQuery:
     """Runs one plan across all ranks.  executor: step-level interface
     (wukong_amd.Engine on GPU; tests use the oracle executor)."""
 
-    def __init__(self, executor, plan, rank, world, device=None):
+    def __init__(self, executor, plan, rank, world, device=None,
+                 threshold=RDMA_THRESHOLD):
         self.ex = executor
         self.plan = plan
         self.rank = rank
         self.world = world
         self.device = device
+        self.threshold = threshold
         self.states = plan_v2c_states(plan)
+
+    def _global_rows(self):
+        """Total rows across ranks (the remote/exchange decision must be
+        uniform — the exchange is a collective)."""
+        dev = ("cuda" if dist.is_initialized()
+               and dist.get_backend() == "nccl" else "cpu")
+        t = torch.tensor([int(self.ex.rows())], dtype=torch.int64, device=dev)
+        dist.all_reduce(t)
+        return int(t.item())
+
+    def _try_remote(self, i, pat):
+        """Sub-threshold tables probe the owner rank's store in place
+        (xGMI peer mappings; oracle peers in CPU tests) instead of
+        exchanging — need_fork_join, sparql.hpp:802-814."""
+        if self.threshold <= 0 or not self.ex.supports_remote():
+            return False
+        if self._global_rows() >= self.threshold:
+            return False
+        v2c_prev, _ = self.states[i - 1]
+        v2c_next, _ = self.states[i]
+        try:
+            self.ex.step_remote(i, pat, v2c_prev, v2c_next)
+            return True
+        except ValueError:
+            return False  # shape needs the exchange (same on every rank)
 
     # ---- exchanges ----
     def _exchange_cpu(self, table, col):
@@ -112,8 +179,9 @@ class DistQuery:
                                     v2c_prev[-(o + 1)])
                 continue
             if i > 0 and s < 0 and local_var != s:
-                # fork-join exchange (need_fork_join, sparql.hpp:802-814;
-                # round 1: no remote-read fallback, always exchange)
+                if self._try_remote(i, (s, p, d, o)):
+                    continue  # rows stayed put; owner stores were probed
+                # fork-join exchange (need_fork_join, sparql.hpp:802-814)
                 v2c, ncols = self.states[i - 1]
                 col = v2c[-(s + 1)]
                 if nccl:
@@ -199,6 +267,17 @@ class GpuExecutor:
 
     def filter_with_list(self, sorted_list, step, v2c, col):
         n = self.engine.execute_filter_list(sorted_list)
+        self.engine.last_rows = n
+        return n
+
+    def rows(self):
+        return self.engine.last_rows
+
+    def supports_remote(self):
+        return bool(getattr(self.engine._store, "_peers_ready", False))
+
+    def step_remote(self, i, pat, v2c_prev, v2c_next):
+        n = self.engine.execute_one_pattern_remote()
         self.engine.last_rows = n
         return n
 
