@@ -211,8 +211,13 @@ int32_t wk_engine_load_rbuf(wk_engine_t *, const wk_sid_t *table,
 int32_t wk_engine_load_rbuf_device(wk_engine_t *, const wk_sid_t *dev_table,
                                    int64_t nrows, int32_t ncols,
                                    const int32_t *v2c_map, int32_t pattern_step);
-/* Run pattern [pattern_step]; advances the step.  Out: new row count. */
+/* Run pattern [pattern_step]; advances the step.  Out: new row count.
+ * nrows_out = NULL launches asynchronously (no host sync; the overflow
+ * re-run then resolves at the next synchronizing call — only safe for
+ * steps that cannot overflow, i.e. filters). */
 int32_t wk_engine_execute_one_pattern(wk_engine_t *, int64_t *nrows_out);
+/* Sync and return the current row count (pairs with async steps). */
+int32_t wk_engine_row_count(wk_engine_t *, int64_t *nrows_out);
 int32_t wk_engine_pattern_step(const wk_engine_t *);
 int32_t wk_engine_col_num(const wk_engine_t *);
 /* Execute the CURRENT pattern (a mid-plan const-/index-start membership

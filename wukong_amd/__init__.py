@@ -111,6 +111,8 @@ _eng_filter_list = _sig("wk_engine_execute_filter_list", c_i32,
                         [c_vp, c_u32p, c_u64, ctypes.POINTER(c_i64)])
 _eng_step_remote = _sig("wk_engine_execute_one_pattern_remote", c_i32,
                         [c_vp, ctypes.POINTER(c_i64)])
+_eng_row_count = _sig("wk_engine_row_count", c_i32,
+                      [c_vp, ctypes.POINTER(c_i64)])
 
 
 class WkPeerBlob(ctypes.Structure):
@@ -529,6 +531,20 @@ class Engine:
         rc = _eng_step(self._h, ctypes.byref(n))
         if rc != 0:
             raise RuntimeError(f"execute_one_pattern rc={rc}")
+        return n.value
+
+    def execute_one_pattern_async(self):
+        """No host sync — only for steps that cannot overflow (filters:
+        outputs <= inputs).  Resolve counts with row_count()."""
+        rc = _eng_step(self._h, None)
+        if rc != 0:
+            raise RuntimeError(f"execute_one_pattern(async) rc={rc}")
+
+    def row_count(self):
+        n = c_i64()
+        rc = _eng_row_count(self._h, ctypes.byref(n))
+        if rc != 0:
+            raise RuntimeError(f"row_count rc={rc}")
         return n.value
 
     def execute_one_pattern_remote(self):

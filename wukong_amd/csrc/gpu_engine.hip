@@ -2782,6 +2782,18 @@ extern "C" int32_t wk_engine_execute_one_pattern(wk_engine_t *e, int64_t *nrows_
 extern "C" int32_t wk_engine_pattern_step(const wk_engine_t *e) { return e->step; }
 extern "C" int32_t wk_engine_col_num(const wk_engine_t *e) { return e->ncols; }
 
+// Sync the stream and return the current row count — pairs with
+// async pattern execution (nrows_out = NULL): the distributed driver
+// launches filter steps (outputs <= inputs, no overflow possible)
+// without a host round-trip and resolves counts lazily at the next
+// exchange point.
+extern "C" int32_t wk_engine_row_count(wk_engine_t *e, int64_t *nrows_out) {
+    if (!e || !nrows_out) return WK_ERR_STATE;
+    int32_t rc = sync_state(e);
+    if (rc == WK_OK) *nrows_out = e->nrows;
+    return rc;
+}
+
 // Run the CURRENT pattern via the xGMI peer-probe path (small tables:
 // in-place remote reads instead of an exchange — sparql.hpp:802-814).
 // Requires wk_gpu_store_import_peers.  The overflow re-run inherits the
