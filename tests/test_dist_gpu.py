@@ -48,6 +48,67 @@ def _worker(rank, world, port, results):
         dist.destroy_process_group()
 
 
+def _worker_peer(rank, world, port, results):
+    """xGMI peer-probe path on hardware: both ranks on cuda:0 (the
+    1-GPU box), stores opened across processes via HIP IPC (dmabuf
+    mode, HSA_ENABLE_IPC_MODE_LEGACY=0); gloo carries only the control
+    plane.  threshold=inf forces every eligible step through
+    k_peer_step — the reference's in-place one-sided read."""
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch.distributed as dist
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from wukong_amd.dist import DistQuery, GpuExecutor, init_peers
+    from tests.oracle_util import sort_rows
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        triples = wk.lubm_gen(2, seed=42, sid=rank, nsrv=world)
+        store = wk.Store(triples, sid=rank, nsrv=world)
+        gstore = wk.GpuStore(store, device=0)
+        eng = wk.Engine(gstore, device=0)
+        ok = init_peers(gstore, store)
+        out = {"_peers": ok}
+        if ok:
+            for name in QUERIES:
+                plan = Q.ALL[name]
+                ex = GpuExecutor(eng, plan)
+                dq = DistQuery(ex, plan, rank, world, threshold=10**9)
+                dq.run()
+                out[name] = sort_rows(dq.gather_result())
+        if rank == 0:
+            results.put(out)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(900)
+def test_gpu_peer_probe_two_ranks_equal_oracle():
+    """k_peer_step (IPC-mapped stores) vs the 1-partition oracle."""
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_worker_peer, args=(r, 2, 29913, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=840)
+    for p in procs:
+        p.join(timeout=60)
+    assert got["_peers"], "HIP IPC peer import failed on this box"
+
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from tests.oracle_util import OracleCtx, sort_rows
+    full = OracleCtx(wk.lubm_gen(2, seed=42))
+    for name in QUERIES:
+        want = sort_rows(full.run_query(Q.ALL[name]))
+        assert got[name].shape == want.shape, (name, got[name].shape, want.shape)
+        assert np.array_equal(got[name], want), name
+
+
 @pytest.mark.timeout(900)
 def test_gpu_dist_two_ranks_equal_oracle():
     ctx = mp.get_context("spawn")
