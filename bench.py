@@ -723,9 +723,10 @@ def main():
                        "production path for non-functional predicates)"
                        if rl_expand else "category aggregate",
         "probe_kernel": rl_probe or None,
-        # same step through the functional-predicate dense map (the
-        # default production path for this step): algorithmic bytes are
-        # 148->8 per row, so GB/s is lower while wall time is ~2x faster
+        # same step through the rank-compressed functional map (submit
+        # path: k_fn_gather + k_fn_compact; the graph replay runs the
+        # 1:1 k_expand_fn_map at ~5-6 TB/s): algorithmic bytes are
+        # 148->28 per row, so GB/s reads lower while wall time is lower
         "fn_map_kernel": rl_fn or None,
     }
 

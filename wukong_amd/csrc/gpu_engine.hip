@@ -102,11 +102,47 @@ enum { PM_SIZE = 0,    // k2u: write edge count per row
 
 // device query state (see engine): [0]=nrows [1]=scan total [2]=overflow
 // flag [3]=required rows; stats[0..6] = algorithmic bytes per category
-enum { S_NROWS = 0, S_TOTAL = 1, S_ERR = 2, S_REQ = 3, S_OVF = 4, S_WORDS = 5 };
+enum { S_NROWS = 0, S_TOTAL = 1, S_ERR = 2, S_REQ = 3, S_OVF = 4,
+       S_DONE = 5, S_WORDS = 6 };
 enum { CAT_PROBE = 0, CAT_SCAN, CAT_EXPAND, CAT_FILTER, CAT_COPY, CAT_SPLIT,
        CAT_OTHER, CAT_COUNT };
 
 constexpr int SCAN_T = 256;  // scan tile = block size of the fused kernels
+
+// Commit fused into the LAST finishing block of a producing kernel
+// (mode 1 = the k_commit semantics, 2 = k_commit_map): the separate
+// 1-thread state kernels measured 4-5 us of span+boundary each, which
+// summed to ~25-30 us per replayed query.  S_DONE self-resets, so the
+// next fused launch starts clean.  Other blocks' S_TOTAL/S_OVF arrive
+// via device-scope atomics; the commit's plain stores become visible
+// to the next kernel at the dispatch boundary.
+__device__ __forceinline__ void commit_tail(uint64_t *d_state, uint64_t cap,
+                                            int mode) {
+    if (!mode) return;
+    __syncthreads();
+    if (threadIdx.x != 0) return;
+    unsigned long long done =
+        atomicAdd((unsigned long long *)&d_state[S_DONE], 1ull);
+    if (done + 1 != (unsigned long long)gridDim.x) return;
+    d_state[S_DONE] = 0;
+    uint64_t t = atomicAdd((unsigned long long *)&d_state[S_TOTAL], 0ull);
+    uint64_t ovf = atomicAdd((unsigned long long *)&d_state[S_OVF], 0ull);
+    if (mode == 2) {  // optimistic 1:1 map: rows unchanged, miss -> S_ERR
+        if (ovf) {
+            d_state[S_ERR] = 1;
+            d_state[S_REQ] = max(d_state[S_REQ], d_state[S_NROWS]);
+        }
+    } else {
+        if (t > cap) {
+            d_state[S_ERR] = 1;
+            d_state[S_REQ] = max(d_state[S_REQ], t);
+            t = cap;
+        }
+        d_state[S_NROWS] = t;
+    }
+    d_state[S_TOTAL] = 0;
+    d_state[S_OVF] = 0;
+}
 
 __device__ __forceinline__ void count_bytes(uint64_t *stats, int cat,
                                             uint64_t bytes) {
@@ -303,6 +339,7 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                              const sid_t *__restrict__ fn_vals,
                              uint64_t fn_base,
                              uint64_t fn_n, int fn_swap, int verify_only,
+                             int commit_mode, uint64_t commit_cap,
                              uint64_t *__restrict__ d_state,
                              uint64_t *__restrict__ d_stats,
                              sid_t *__restrict__ out_tbl)
@@ -418,6 +455,7 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
         }
         __syncthreads();
     }
+    commit_tail(d_state, commit_cap, commit_mode);
 }
 
 // finish the scan across blocks: exclusive over the G block sums
@@ -559,12 +597,12 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
                              const uint32_t *__restrict__ d_cnt,
                              const uint64_t *__restrict__ d_pre,
                              const uint64_t *__restrict__ bsums, int G,
-                             const uint64_t *__restrict__ d_state, uint64_t cap,
+                             uint64_t *__restrict__ d_state, uint64_t cap,
+                             int commit_mode,
                              const uint32_t *__restrict__ ovf,
                              sid_t *__restrict__ out)
 {
     const int64_t nq = (int64_t)d_state[S_OVF];
-    if (!nq) return;
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     const int64_t chunk = (nrows + G - 1) / G;
     constexpr int oc = NC + 1;
@@ -589,124 +627,93 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
             dst[NC] = el[k];
         }
     }
+    commit_tail(d_state, cap, commit_mode);
 }
 
-// known_to_unknown over a FUNCTIONAL predicate (every key deg==1, dense
-// vid->object map): the probe+scan+expand pipeline collapses into one
-// 4-byte gather + block-compacted append — 148 bytes of cluster-hash
-// traffic per row become 4.  An optional fused `?v rdf:type CONST`
-// filter on the NEW column (plan pairs like Q1's ugDegreeFrom ->
-// University) drops rows in flight with the same block-aggregated
-// compaction (one atomic per 1024 rows — no contention, unlike the
-// wave-pass fusion this replaces).  Multi-type (0xFFFF) falls back to a
-// probe of [val|TYPE_ID|OUT].  Output rows <= input rows, so capacity
-// can never overflow.
-template <int NC>
-__global__ void k_expand_fn(const sid_t *__restrict__ tbl,
+// known_to_unknown over a FUNCTIONAL predicate (every key deg==1,
+// rank-compressed map): the probe+scan+expand pipeline collapses into
+// a page+value gather pair + compacted append, with an optional fused
+// `?v rdf:type CONST` filter on the NEW column (plan pairs like Q1's
+// ugDegreeFrom -> University).  Two phases: a 1:1 grid-stride GATHER
+// (the measured-fast k_expand_fn_map shape — no barriers between the
+// dependent page/value loads) writes each row's resolved object (0 =
+// miss/filtered) to a scratch stream; the COMPACT phase then reads it
+// SEQUENTIALLY with the block-aggregated scan.  The round-1 single-pass
+// form interleaved the gathers with 16 scan barriers per tile and ran
+// 4-5x slower than its own gather cost (169 vs 36 us on Q1's 6.4M-row
+// step).  Multi-type (0xFFFF) falls back to a probe of [val|TYPE|OUT].
+// Output rows <= input rows, so capacity can never overflow.
+__global__ void k_fn_gather(const sid_t *__restrict__ tbl, int ncols,
                             const fnpage_t *__restrict__ fn_pg,
                             const sid_t *__restrict__ fn_vals,
-                            uint64_t fn_base,
-                            uint64_t fn_n, int col, int use_typeof,
-                            sid_t fcval,
+                            uint64_t fn_base, uint64_t fn_n, int col,
+                            int use_typeof, sid_t fcval,
                             const uint16_t *__restrict__ type_of,
                             const uint64_t *__restrict__ tbm,
                             uint64_t type_base, uint64_t type_n,
                             const vertex_t *__restrict__ verts,
                             const sid_t *__restrict__ edges,
                             uint64_t f_bstart, uint64_t f_nbuckets,
-                            uint64_t *__restrict__ d_state,
+                            const uint64_t *__restrict__ d_state,
                             uint64_t *__restrict__ d_stats,
-                            sid_t *__restrict__ out)
+                            sid_t *__restrict__ d_val)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_EXPAND,
+                (uint64_t)nrows * (24 + 4 + (use_typeof ? 2 : 0)));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        sid_t v = tbl[r * ncols + col];
+        sid_t tv = fn_lookup(fn_pg, fn_vals, fn_base, fn_n, v);
+        if (tv && use_typeof) {
+            uint64_t tix = (uint64_t)tv - type_base;
+            if (tbm) {  // bitmap: decision is exact (multi-type incl.)
+                if (!(tix < type_n && ((tbm[tix >> 6] >> (tix & 63)) & 1)))
+                    tv = 0;
+            } else {
+                uint16_t t = (tix < type_n) ? type_of[tix] : 0;
+                if (t != 0xFFFF) {
+                    if ((sid_t)t != fcval) tv = 0;
+                } else {
+                    uint64_t eo = 0, es = 0;
+                    probe_one(verts, f_bstart, f_nbuckets,
+                              key_pack(tv, TYPE_ID, (uint64_t)DIR_OUT), eo,
+                              es);
+                    if (!(es && bsearch_u32(edges + eo, es, fcval))) tv = 0;
+                }
+            }
+        }
+        d_val[r] = tv;
+    }
+}
+
+template <int NC>
+__global__ void k_fn_compact(const sid_t *__restrict__ tbl,
+                             const sid_t *__restrict__ d_val,
+                             uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats,
+                             uint64_t cap, int commit_mode,
+                             sid_t *__restrict__ out)
 {
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     constexpr int oc = NC + 1;
-    count_bytes(d_stats, CAT_EXPAND,
-                (uint64_t)nrows * (24 + (use_typeof ? 2 : 0)));
-    // K=4 measured best: K=8 (deeper gather MLP) costs more VGPRs than
-    // it hides latency AND halves the tile count (load imbalance) —
-    // q1's fn expands regressed 184->216us / 230->341us.  Instead, a
-    // 2-deep software pipeline (the k_probe_scan idiom) issues the NEXT
-    // tile's dense-map gathers before this tile's scan barriers, so
-    // their ~1us HBM latency hides under the scan + writes.
+    count_bytes(d_stats, CAT_EXPAND, (uint64_t)nrows * 4);
     constexpr int K = 4;
     __shared__ unsigned long long s_base;
     __shared__ uint32_t sh[SCAN_T];
     const int64_t tile = (int64_t)blockDim.x * K;
     const int64_t stride = (int64_t)gridDim.x * tile;
 
-    sid_t pv[K], ptv[K];
-    uint16_t pt16[K];
-    auto load_tile = [&](int64_t b) {
-        // two gather waves: all K page loads issue together, then the
-        // K dependent value loads — each wave's HBM latency overlaps
-        // across the K independent requests
-        fnpage_t pg[K];
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            const int64_t r = b + (int64_t)threadIdx.x * K + k;
-            sid_t v = (r < nrows) ? tbl[r * NC + col] : 0;
-            pv[k] = v;
-            uint64_t idx = (uint64_t)v - fn_base;
-            pg[k] = (idx < fn_n) ? fn_pg[idx >> 6] : fnpage_t{0, 0, 0};
-        }
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            uint64_t idx = (uint64_t)pv[k] - fn_base;
-            sid_t tv = 0;
-            if (idx < fn_n && ((pg[k].bits >> (idx & 63)) & 1)) {
-                uint32_t rk = pg[k].rank +
-                              (uint32_t)__popcll(pg[k].bits &
-                                                 ((1ull << (idx & 63)) - 1));
-                tv = fn_vals[rk];
-            }
-            ptv[k] = tv;
-            if (use_typeof) {
-                uint64_t tix = (uint64_t)tv - type_base;
-                if (tbm)  // bitmap: decision is exact (multi-type incl.)
-                    pt16[k] = (tv && tix < type_n &&
-                               ((tbm[tix >> 6] >> (tix & 63)) & 1))
-                                  ? 1 : 0;
-                else
-                    pt16[k] = (tv && tix < type_n) ? type_of[tix] : 0;
-            }
-        }
-    };
-    const int64_t base0 = (int64_t)blockIdx.x * tile;
-    if (base0 < nrows) load_tile(base0);
-
-    for (int64_t base = base0; base < nrows; base += stride) {
-        bool keep[K];
+    for (int64_t base = (int64_t)blockIdx.x * tile; base < nrows;
+         base += stride) {
         sid_t val[K];
-        int64_t rr[K];
         uint32_t cnt = 0;
-        sid_t cv[K];
-        uint16_t ct16[K];
 #pragma unroll
         for (int k = 0; k < K; k++) {
-            cv[k] = pv[k];
-            val[k] = ptv[k];
-            ct16[k] = pt16[k];
-            rr[k] = base + (int64_t)threadIdx.x * K + k;
+            const int64_t r = base + (int64_t)threadIdx.x * K + k;
+            val[k] = (r < nrows) ? d_val[r] : 0;
+            cnt += val[k] ? 1u : 0u;
         }
-        if (base + stride < nrows) load_tile(base + stride);  // prefetch
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            keep[k] = false;
-            if (rr[k] >= nrows || !val[k]) continue;
-            if (!use_typeof) { keep[k] = true; continue; }
-            if (tbm) { keep[k] = ct16[k] != 0; continue; }
-            if (ct16[k] != 0xFFFF) {
-                keep[k] = ((sid_t)ct16[k] == fcval);
-            } else {
-                uint64_t eo = 0, es = 0;
-                probe_one(verts, f_bstart, f_nbuckets,
-                          key_pack(val[k], TYPE_ID, (uint64_t)DIR_OUT), eo,
-                          es);
-                keep[k] = es && bsearch_u32(edges + eo, es, fcval);
-            }
-        }
-#pragma unroll
-        for (int k = 0; k < K; k++) cnt += keep[k] ? 1u : 0u;
         sh[threadIdx.x] = cnt;
         __syncthreads();
         for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
@@ -730,9 +737,10 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
         uint64_t pos = s_base + my_end - cnt;
 #pragma unroll
         for (int k = 0; k < K; k++) {
-            if (!keep[k]) continue;
+            if (!val[k]) continue;
+            const int64_t r = base + (int64_t)threadIdx.x * K + k;
             sid_t *dst = out + (int64_t)pos * oc;
-            const sid_t *src = tbl + rr[k] * NC;
+            const sid_t *src = tbl + r * NC;
 #pragma unroll
             for (int c = 0; c < NC; c++) dst[c] = src[c];
             dst[NC] = val[k];
@@ -740,6 +748,7 @@ __global__ void k_expand_fn(const sid_t *__restrict__ tbl,
         }
         __syncthreads();
     }
+    commit_tail(d_state, cap, commit_mode);
 }
 
 // Fused known_to_unknown + rdf:type constant filter on the NEW column
@@ -1376,7 +1385,7 @@ __global__ void k_expand_fn_map(const sid_t *__restrict__ tbl,
                                 const uint64_t *__restrict__ tbm,
                                 const uint16_t *__restrict__ type_of,
                                 uint64_t type_base, uint64_t type_n,
-                                sid_t fcval,
+                                sid_t fcval, int commit_mode,
                                 uint64_t *__restrict__ d_state,
                                 uint64_t *__restrict__ d_stats,
                                 sid_t *__restrict__ out)
@@ -1410,6 +1419,7 @@ __global__ void k_expand_fn_map(const sid_t *__restrict__ tbl,
     if (miss)
         atomicAdd((unsigned long long *)&d_state[S_OVF],
                   (unsigned long long)miss);
+    commit_tail(d_state, 0, commit_mode);
 }
 
 // ---------------------------------------------------------------------
@@ -1504,7 +1514,17 @@ __global__ void k_peer_step(const wk_peer_desc *__restrict__ peers, int nsrv,
         }
         __syncthreads();
     }
-    if (threadIdx.x == 0) d_state[S_TOTAL] = carry;
+    if (threadIdx.x == 0) {  // single block: commit inline (k_commit)
+        uint64_t t = carry;
+        if (t > cap) {
+            d_state[S_ERR] = 1;
+            d_state[S_REQ] = max(d_state[S_REQ], t);
+            t = cap;
+        }
+        d_state[S_NROWS] = t;
+        d_state[S_TOTAL] = 0;
+        d_state[S_OVF] = 0;
+    }
 }
 
 // commit for the optimistic map: row count unchanged; any miss flags
@@ -2150,7 +2170,8 @@ static void launch_expand_t(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
                        cur_tbl, e->ncols, e->d_edges, (uint64_t *)e->eoff.p,
                        (uint32_t *)e->cnt.p, (uint64_t *)e->prefix.p,
                        (uint64_t *)e->bsums.p, G, e->d_state,
-                       (uint64_t)e->cap_rows, (uint32_t *)e->ovf.p, out_tbl);
+                       (uint64_t)e->cap_rows, /*commit*/ 1,
+                       (uint32_t *)e->ovf.p, out_tbl);
 }
 
 template <int NC>
@@ -2194,13 +2215,19 @@ static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
         (fuse && e->gs && (size_t)fcval < e->gs->d_tbm.size())
             ? e->gs->d_tbm[fcval]
             : nullptr;
-    hipLaunchKernelGGL(k_expand_fn<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
-                       0, e->stream, cur_tbl, d_pg, d_vals, e->st->fn_base,
-                       e->st->fn_n, col, fuse ? 1 : 0, fcval, e->d_type_of,
-                       tbm, e->st->type_base, e->st->type_n, e->d_verts,
-                       e->d_edges, fseg ? fseg->bucket_start : 0,
+    // phase 1: 1:1 gather of resolved objects into the cnt scratch
+    // (sid_t-sized); phase 2: sequential-read compaction + fused commit
+    hipLaunchKernelGGL(k_fn_gather, dim3(grid_for(e->bound)), dim3(BLOCK),
+                       0, e->stream, cur_tbl, e->ncols, d_pg, d_vals,
+                       e->st->fn_base, e->st->fn_n, col, fuse ? 1 : 0, fcval,
+                       e->d_type_of, tbm, e->st->type_base, e->st->type_n,
+                       e->d_verts, e->d_edges, fseg ? fseg->bucket_start : 0,
                        fseg ? fseg->num_buckets : 0, e->d_state, e->d_stats,
-                       out_tbl);
+                       (sid_t *)e->cnt.p);
+    hipLaunchKernelGGL(k_fn_compact<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
+                       0, e->stream, cur_tbl, (const sid_t *)e->cnt.p,
+                       e->d_state, e->d_stats, (uint64_t)e->cap_rows,
+                       /*commit*/ 1, out_tbl);
 }
 
 template <int NC>
@@ -2216,7 +2243,7 @@ static void launch_expand_fn_map_t(wk_engine *e, const sid_t *cur_tbl,
                        dim3(BLOCK), 0, e->stream, cur_tbl, d_pg, d_vals,
                        e->st->fn_base, e->st->fn_n, col, fuse ? 1 : 0, tbm,
                        e->d_type_of, e->st->type_base, e->st->type_n, fcval,
-                       e->d_state, e->d_stats, out_tbl);
+                       /*commit*/ 2, e->d_state, e->d_stats, out_tbl);
 }
 
 static void launch_expand_fn_map(wk_engine *e, const sid_t *cur_tbl,
@@ -2305,8 +2332,7 @@ static int32_t exec_pattern_remote(wk_engine *e) {
                        ostat == 2 ? (sid_t)o : 0, (uint64_t)e->cap_rows,
                        e->d_state, e->d_stats, out_tbl);
     TIME_END(e, CAT_PROBE);
-    hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
-                       (uint64_t)e->cap_rows);
+    // commit inlined in k_peer_step (single block)
     if (pmode == PM_SIZE) {
         e->v2c[-(o + 1)] = e->ncols;
         e->ncols += 1;
@@ -2489,10 +2515,9 @@ static int32_t exec_pattern(wk_engine *e) {
                            (const uint64_t *)nullptr,
                            (const fnpage_t *)nullptr,
                            (const sid_t *)nullptr, 0, 0, 0, 0,
+                           /*commit*/ 1, (uint64_t)e->cap_rows,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
-        hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
-                           (uint64_t)e->cap_rows);
         e->cur ^= 1;
         e->step++;
         return WK_OK;
@@ -2584,14 +2609,16 @@ static int32_t exec_pattern(wk_engine *e) {
                            e->d_type_of, e->st->type_base, e->st->type_n,
                            use_typeof, d_tbm, d_pg, d_vals, st->fn_base, st->fn_n,
                            fn_swap, verify_only ? 1 : 0,
+                           verify_only ? 2 : 1, (uint64_t)e->cap_rows,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
         if (verify_only) {
-            hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0, e->stream,
-                               e->d_state);
             e->step++;
             return WK_OK;
         }
+        e->cur ^= 1;
+        e->step++;
+        return WK_OK;
     } else {
         // known_to_unknown: fused probe+scan -> cross-block scan ->
         // input-centric expansion (+ big-row wave pass)
@@ -2641,12 +2668,6 @@ static int32_t exec_pattern(wk_engine *e) {
                 launch_expand_fn(e, cur_tbl, out_tbl, d_pg, d_vals, col,
                                  fuse2, fcval2, fseg2);
             TIME_END(e, CAT_EXPAND);
-            if (opt)
-                hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0,
-                                   e->stream, e->d_state);
-            else
-                hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
-                                   e->d_state, (uint64_t)e->cap_rows);
             e->v2c[-(o + 1)] = e->ncols;
             e->ncols = oc;
             e->cur ^= 1;
@@ -2721,8 +2742,7 @@ static int32_t exec_pattern(wk_engine *e) {
         e->ncols = oc;
         e->bound = e->cap_rows;  // fan-out unknown until a sync point
     }
-    hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
-                       (uint64_t)e->cap_rows);
+    // commit is fused into the last block of k_expand_big (mode 1)
     e->cur ^= 1;
     e->step++;
     return WK_OK;
@@ -2839,10 +2859,9 @@ extern "C" int32_t wk_engine_execute_filter_list(wk_engine_t *e,
                        PM_LIST, 0, 0u, 0, n, e->d_type_of, 0, 0, 0,
                        (const uint64_t *)nullptr, (const fnpage_t *)nullptr,
                        (const sid_t *)nullptr,
-                       0, 0, 0, 0, e->d_state, e->d_stats, out_tbl);
+                       0, 0, 0, 0, /*commit*/ 1, (uint64_t)e->cap_rows,
+                       e->d_state, e->d_stats, out_tbl);
     TIME_END(e, CAT_FILTER);
-    hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
-                       (uint64_t)e->cap_rows);
     e->cur ^= 1;
     e->step++;
     int32_t rc = sync_state(e);
