@@ -54,9 +54,29 @@ def main():
               {c: (round(v['usec'] / reps, 1), v['launches'])
                for c, v in out.items()}, flush=True)
 
-    q1_steps("fn-submit")               # k_expand_fn path
+    q1_steps("fn-submit")               # k_fn_gather+compact path
     os.environ["WK_FN_DISPATCH"] = "0"
     q1_steps("classic")                 # probe+scan+expand
+    os.environ.pop("WK_FN_DISPATCH")
+
+    # whole-q1 EAGER but ASYNC (submit + one sync at fetch): isolates
+    # the per-step host-sync gaps from the kernels themselves
+    def q1_async(label):
+        s0 = eng.kernel_stats()
+        t0 = time.time()
+        for _ in range(reps):
+            eng.submit(Q.ALL["q1"])
+            eng.fetch_count()
+        wall = (time.time() - t0) * 1e6 / reps
+        s1 = eng.kernel_stats()
+        print(label, f"wall/rep={wall:.0f}us",
+              {c: round((s1[c]['usec'] - s0[c]['usec']) / reps, 1)
+               for c in ("probe", "scan", "expand", "filter", "copy")},
+              flush=True)
+
+    q1_async("fn-async")
+    os.environ["WK_FN_DISPATCH"] = "0"
+    q1_async("classic-async")
     os.environ.pop("WK_FN_DISPATCH")
 
     # graph replay of whole q1 (k_expand_fn_map 1:1 specialization)

@@ -110,12 +110,12 @@ enum { CAT_PROBE = 0, CAT_SCAN, CAT_EXPAND, CAT_FILTER, CAT_COPY, CAT_SPLIT,
 constexpr int SCAN_T = 256;  // scan tile = block size of the fused kernels
 
 // Commit fused into the LAST finishing block of a producing kernel
-// (mode 1 = the k_commit semantics, 2 = k_commit_map): the separate
-// 1-thread state kernels measured 4-5 us of span+boundary each, which
-// summed to ~25-30 us per replayed query.  S_DONE self-resets, so the
-// next fused launch starts clean.  Other blocks' S_TOTAL/S_OVF arrive
-// via device-scope atomics; the commit's plain stores become visible
-// to the next kernel at the dispatch boundary.
+// (mode 1 = the k_commit semantics, 2 = k_commit_map).  MEASURED
+// NET-NEGATIVE for wide grids and disabled there (mode 0): 2048 blocks
+// bumping one done-word serialize at ~88 atomics/us (microarch row
+// `dequeue`) = ~23 us per kernel, more than the 1-thread commit
+// kernels cost (graph-q1 207 -> 257 us with fusion on).  Kept for
+// single-/small-grid kernels (k_peer_step commits inline).
 __device__ __forceinline__ void commit_tail(uint64_t *d_state, uint64_t cap,
                                             int mode) {
     if (!mode) return;
@@ -2170,7 +2170,7 @@ static void launch_expand_t(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
                        cur_tbl, e->ncols, e->d_edges, (uint64_t *)e->eoff.p,
                        (uint32_t *)e->cnt.p, (uint64_t *)e->prefix.p,
                        (uint64_t *)e->bsums.p, G, e->d_state,
-                       (uint64_t)e->cap_rows, /*commit*/ 1,
+                       (uint64_t)e->cap_rows, /*commit*/ 0,
                        (uint32_t *)e->ovf.p, out_tbl);
 }
 
@@ -2227,7 +2227,7 @@ static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
     hipLaunchKernelGGL(k_fn_compact<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
                        0, e->stream, cur_tbl, (const sid_t *)e->cnt.p,
                        e->d_state, e->d_stats, (uint64_t)e->cap_rows,
-                       /*commit*/ 1, out_tbl);
+                       /*commit*/ 0, out_tbl);
 }
 
 template <int NC>
@@ -2243,7 +2243,7 @@ static void launch_expand_fn_map_t(wk_engine *e, const sid_t *cur_tbl,
                        dim3(BLOCK), 0, e->stream, cur_tbl, d_pg, d_vals,
                        e->st->fn_base, e->st->fn_n, col, fuse ? 1 : 0, tbm,
                        e->d_type_of, e->st->type_base, e->st->type_n, fcval,
-                       /*commit*/ 2, e->d_state, e->d_stats, out_tbl);
+                       /*commit*/ 0, e->d_state, e->d_stats, out_tbl);
 }
 
 static void launch_expand_fn_map(wk_engine *e, const sid_t *cur_tbl,
@@ -2515,9 +2515,11 @@ static int32_t exec_pattern(wk_engine *e) {
                            (const uint64_t *)nullptr,
                            (const fnpage_t *)nullptr,
                            (const sid_t *)nullptr, 0, 0, 0, 0,
-                           /*commit*/ 1, (uint64_t)e->cap_rows,
+                           /*commit*/ 0, 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
+        hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                           (uint64_t)e->cap_rows);
         e->cur ^= 1;
         e->step++;
         return WK_OK;
@@ -2609,16 +2611,15 @@ static int32_t exec_pattern(wk_engine *e) {
                            e->d_type_of, e->st->type_base, e->st->type_n,
                            use_typeof, d_tbm, d_pg, d_vals, st->fn_base, st->fn_n,
                            fn_swap, verify_only ? 1 : 0,
-                           verify_only ? 2 : 1, (uint64_t)e->cap_rows,
+                           /*commit*/ 0, 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
         if (verify_only) {
+            hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0, e->stream,
+                               e->d_state);
             e->step++;
             return WK_OK;
         }
-        e->cur ^= 1;
-        e->step++;
-        return WK_OK;
     } else {
         // known_to_unknown: fused probe+scan -> cross-block scan ->
         // input-centric expansion (+ big-row wave pass)
@@ -2668,6 +2669,12 @@ static int32_t exec_pattern(wk_engine *e) {
                 launch_expand_fn(e, cur_tbl, out_tbl, d_pg, d_vals, col,
                                  fuse2, fcval2, fseg2);
             TIME_END(e, CAT_EXPAND);
+            if (opt)
+                hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0,
+                                   e->stream, e->d_state);
+            else
+                hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
+                                   e->d_state, (uint64_t)e->cap_rows);
             e->v2c[-(o + 1)] = e->ncols;
             e->ncols = oc;
             e->cur ^= 1;
@@ -2742,7 +2749,8 @@ static int32_t exec_pattern(wk_engine *e) {
         e->ncols = oc;
         e->bound = e->cap_rows;  // fan-out unknown until a sync point
     }
-    // commit is fused into the last block of k_expand_big (mode 1)
+    hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                       (uint64_t)e->cap_rows);
     e->cur ^= 1;
     e->step++;
     return WK_OK;
@@ -2859,9 +2867,11 @@ extern "C" int32_t wk_engine_execute_filter_list(wk_engine_t *e,
                        PM_LIST, 0, 0u, 0, n, e->d_type_of, 0, 0, 0,
                        (const uint64_t *)nullptr, (const fnpage_t *)nullptr,
                        (const sid_t *)nullptr,
-                       0, 0, 0, 0, /*commit*/ 1, (uint64_t)e->cap_rows,
+                       0, 0, 0, 0, /*commit*/ 0, 0,
                        e->d_state, e->d_stats, out_tbl);
     TIME_END(e, CAT_FILTER);
+    hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
+                       (uint64_t)e->cap_rows);
     e->cur ^= 1;
     e->step++;
     int32_t rc = sync_state(e);
