@@ -354,6 +354,9 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
     constexpr int K = 4;
     __shared__ unsigned long long s_base;
     __shared__ uint32_t sh[SCAN_T];
+    // staged kept-row indices: the write loop puts adjacent output rows
+    // on adjacent lanes (coalesced stores; see k_fn_compact note)
+    __shared__ uint16_t sidx[SCAN_T * K];
     const int64_t tile = (int64_t)blockDim.x * K;
     const int64_t stride = (int64_t)gridDim.x * tile;
 
@@ -443,15 +446,16 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
                          ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
                                      (unsigned long long)block_total)
                          : 0;
-        __syncthreads();
-        uint64_t pos = s_base + my_end - cnt;
+        uint32_t p = my_end - cnt;
 #pragma unroll
-        for (int k = 0; k < K; k++) {
-            if (!keep[k]) continue;
-            sid_t *dst = out_tbl + (int64_t)pos * ncols;
-            const sid_t *src = tbl + rr[k] * ncols;
-            for (int c = 0; c < ncols; c++) dst[c] = src[c];
-            pos++;
+        for (int k = 0; k < K; k++)
+            if (keep[k]) sidx[p++] = (uint16_t)(threadIdx.x * K + k);
+        __syncthreads();
+        for (uint32_t j = threadIdx.x; j < block_total; j += SCAN_T) {
+            const int64_t src = base + sidx[j];
+            sid_t *dst = out_tbl + (int64_t)(s_base + j) * ncols;
+            const sid_t *srow = tbl + src * ncols;
+            for (int c = 0; c < ncols; c++) dst[c] = srow[c];
         }
         __syncthreads();
     }
@@ -701,6 +705,12 @@ __global__ void k_fn_compact(const sid_t *__restrict__ tbl,
     constexpr int K = 4;
     __shared__ unsigned long long s_base;
     __shared__ uint32_t sh[SCAN_T];
+    // kept rows' tile-local indices, staged so the write loop can put
+    // ADJACENT output rows on ADJACENT lanes — the round-2 first cut
+    // had each thread write its own K rows at consecutive positions,
+    // and the 48-B-per-lane store stride cost 3x (107 vs ~40 us on
+    // 6.4M rows)
+    __shared__ uint16_t sidx[SCAN_T * K];
     const int64_t tile = (int64_t)blockDim.x * K;
     const int64_t stride = (int64_t)gridDim.x * tile;
 
@@ -733,18 +743,20 @@ __global__ void k_fn_compact(const sid_t *__restrict__ tbl,
                 atomicAdd((unsigned long long *)&d_stats[CAT_EXPAND],
                           (unsigned long long)block_total * (NC * 4 + oc * 4));
         }
+        uint32_t p = my_end - cnt;
+#pragma unroll
+        for (int k = 0; k < K; k++)
+            if (val[k]) sidx[p++] = (uint16_t)(threadIdx.x * K + k);
         __syncthreads();
-        uint64_t pos = s_base + my_end - cnt;
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            if (!val[k]) continue;
-            const int64_t r = base + (int64_t)threadIdx.x * K + k;
+        for (uint32_t j = threadIdx.x; j < block_total; j += SCAN_T) {
+            const int64_t src = base + sidx[j];
+            const uint64_t pos = s_base + j;
+            if (pos >= cap) continue;
             sid_t *dst = out + (int64_t)pos * oc;
-            const sid_t *src = tbl + r * NC;
+            const sid_t *srow = tbl + src * NC;
 #pragma unroll
-            for (int c = 0; c < NC; c++) dst[c] = src[c];
-            dst[NC] = val[k];
-            pos++;
+            for (int c = 0; c < NC; c++) dst[c] = srow[c];
+            dst[NC] = d_val[src];
         }
         __syncthreads();
     }
@@ -1334,40 +1346,68 @@ __global__ void k_expand_opt(const vertex_t *__restrict__ verts,
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     const int oc = ncols + 1;
     count_bytes(d_stats, CAT_EXPAND, (uint64_t)nrows * (4 + 128 + 8));
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
-        sid_t v = tbl[r * ncols + col];
-        uint8_t m = flags_in[r];
-        uint64_t eoff = 0, esz = 0;
-        if (m && v != BLANK_ID && num_buckets) {
-            uint64_t key = (key_mode == PK_NORMAL)
-                               ? key_pack((uint64_t)v, (uint64_t)pid,
-                                          (uint64_t)dir)
-                               : key_pack(0, (uint64_t)v, (uint64_t)dir);
-            probe_one(verts, bucket_start, num_buckets, key, eoff, esz);
-        }
-        if (!m || v == BLANK_ID || esz == 0) {
-            // unmatched / blank / no edges: keep the row, BLANK new col;
-            // the matched flag survives unchanged (sparql.hpp:328-334,
-            // 352-356)
-            uint64_t pos =
-                atomicAdd((unsigned long long *)&d_state[S_TOTAL], 1ull);
-            if (pos < cap) {
-                sid_t *dst = out + pos * oc;
-                for (int c = 0; c < ncols; c++) dst[c] = tbl[r * ncols + c];
-                dst[ncols] = BLANK_ID;
-                flags_out[pos] = m;
+    // block-aggregated append (one global atomic per 256-row tile)
+    // replaces the round-1 per-row cursors — a single cursor word
+    // saturates at ~88 atomics/us (microarch row `dequeue`)
+    __shared__ unsigned long long s_base;
+    __shared__ uint64_t sh[SCAN_T];
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < nrows;
+         base += stride) {
+        const int64_t r = base + threadIdx.x;
+        uint64_t eoff = 0, esz = 0, emit = 0;
+        uint8_t m = 0;
+        sid_t v = 0;
+        if (r < nrows) {
+            v = tbl[r * ncols + col];
+            m = flags_in[r];
+            if (m && v != BLANK_ID && num_buckets) {
+                uint64_t key = (key_mode == PK_NORMAL)
+                                   ? key_pack((uint64_t)v, (uint64_t)pid,
+                                              (uint64_t)dir)
+                                   : key_pack(0, (uint64_t)v, (uint64_t)dir);
+                probe_one(verts, bucket_start, num_buckets, key, eoff, esz);
             }
-        } else {
-            uint64_t pos = atomicAdd((unsigned long long *)&d_state[S_TOTAL],
-                                     (unsigned long long)esz);
-            for (uint64_t k = 0; k < esz && pos + k < cap; k++) {
-                sid_t *dst = out + (pos + k) * oc;
-                for (int c = 0; c < ncols; c++) dst[c] = tbl[r * ncols + c];
-                dst[ncols] = edges[eoff + k];
-                flags_out[pos + k] = 1;
+            // unmatched / blank / no edges: keep the row, BLANK new
+            // col; the matched flag survives unchanged
+            // (sparql.hpp:328-334, 352-356)
+            emit = (!m || v == BLANK_ID || esz == 0) ? 1 : esz;
+        }
+        sh[threadIdx.x] = emit;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint64_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += x;
+            __syncthreads();
+        }
+        const uint64_t my_end = sh[threadIdx.x];
+        const uint64_t block_total = sh[SCAN_T - 1];
+        if (threadIdx.x == SCAN_T - 1)
+            s_base = block_total
+                         ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
+                                     (unsigned long long)block_total)
+                         : 0;
+        __syncthreads();
+        if (r < nrows) {
+            uint64_t pos = s_base + my_end - emit;
+            if (!m || v == BLANK_ID || esz == 0) {
+                if (pos < cap) {
+                    sid_t *dst = out + pos * oc;
+                    for (int c = 0; c < ncols; c++) dst[c] = tbl[r * ncols + c];
+                    dst[ncols] = BLANK_ID;
+                    flags_out[pos] = m;
+                }
+            } else {
+                for (uint64_t k = 0; k < esz && pos + k < cap; k++) {
+                    sid_t *dst = out + (pos + k) * oc;
+                    for (int c = 0; c < ncols; c++) dst[c] = tbl[r * ncols + c];
+                    dst[ncols] = edges[eoff + k];
+                    flags_out[pos + k] = 1;
+                }
             }
         }
+        __syncthreads();
     }
 }
 
