@@ -319,6 +319,76 @@ __global__ void k_probe_scan(const vertex_t *__restrict__ verts,
     if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
 }
 
+// CSR-indexed known_to_unknown front half (non-functional segments):
+// the 128-B cluster-hash bucket walk becomes a 16-B page + 8-B entry
+// lookup ({edge_off:40|len:24}, rank-compressed like the fn maps).
+// Two phases, following the two-phase fn lesson: a barrier-free 1:1
+// gather writes (eoff, cnt); k_scan_local then runs the chunked
+// prefix exactly as k_probe_scan produced it, so the expansion
+// kernels consume the identical layout.
+__global__ void k_csr_gather(const sid_t *__restrict__ tbl, int ncols,
+                             int col,
+                             const fnpage_t *__restrict__ pg,
+                             const uint64_t *__restrict__ entries,
+                             uint64_t base, uint64_t n,
+                             const uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats,
+                             uint64_t *__restrict__ d_eoff,
+                             uint32_t *__restrict__ d_cnt)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_PROBE, (uint64_t)nrows * (4 + 16 + 8 + 12));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        sid_t v = tbl[r * ncols + col];
+        uint64_t e = 0;
+        uint64_t idx = (uint64_t)v - base;
+        if (idx < n) {
+            const fnpage_t p = pg[idx >> 6];
+            if ((p.bits >> (idx & 63)) & 1) {
+                uint32_t rk = p.rank +
+                              (uint32_t)__popcll(p.bits &
+                                                 ((1ull << (idx & 63)) - 1));
+                e = entries[rk];
+            }
+        }
+        d_eoff[r] = e >> 24;
+        d_cnt[r] = (uint32_t)(e & 0xFFFFFF);
+    }
+}
+
+// chunked exclusive prefix of d_cnt -> d_pre + per-chunk sums (same
+// chunk math as k_probe_scan so the expansion kernels are unchanged)
+__global__ void k_scan_local(const uint32_t *__restrict__ d_cnt,
+                             const uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_pre,
+                             uint64_t *__restrict__ bsums)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int G = gridDim.x;
+    const int64_t chunk = (nrows + G - 1) / G;
+    const int64_t start = (int64_t)blockIdx.x * chunk;
+    const int64_t end = min(start + chunk, (int64_t)nrows);
+    __shared__ uint64_t sh[SCAN_T];
+    uint64_t carry = 0;
+    for (int64_t base = start; base < end; base += SCAN_T) {
+        const int64_t r = base + threadIdx.x;
+        uint64_t esz = (r < end) ? d_cnt[r] : 0;
+        sh[threadIdx.x] = esz;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint64_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += x;
+            __syncthreads();
+        }
+        if (r < end) d_pre[r] = carry + sh[threadIdx.x] - esz;
+        carry += sh[SCAN_T - 1];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
+}
+
 // Filter operators (k2c/k2k/c2k/i2k): probe + membership test + block-
 // aggregated compaction.  Each thread handles 4 rows per tile (4 loads in
 // flight, and 4x fewer global atomics: a single counter saturates at
@@ -1633,6 +1703,10 @@ struct wk_gpu_store {
     // pages + packed values (wk_types.h fnpage_t)
     std::vector<fnpage_t *> d_fn_pages;
     std::vector<sid_t *> d_fn_vals;
+    // rank-compressed CSR side index for non-functional segments:
+    // pages + u64 {edge_off:40|len:24} entries (24-B probes)
+    std::vector<fnpage_t *> d_csr_pages;
+    std::vector<uint64_t *> d_csr_entries;
     // per-type membership bitmaps (LLC-resident typeof filters)
     std::vector<uint64_t *> d_tbm;
     // xGMI peer mappings of the other ranks' stores (HIP IPC) — the
@@ -1892,6 +1966,25 @@ extern "C" wk_gpu_store_t *wk_gpu_store_create(const wk_store_t *st, int32_t dev
             }
         }
     }
+    if (!st->csr.empty()) {
+        g->d_csr_pages.assign(st->csr.size(), nullptr);
+        g->d_csr_entries.assign(st->csr.size(), nullptr);
+        for (size_t w = 0; w < st->csr.size(); w++) {
+            if (!st->csr[w].present()) continue;
+            size_t pb = st->csr[w].pages.size() * sizeof(fnpage_t);
+            size_t eb = std::max<size_t>(st->csr[w].entries.size() * 8, 8);
+            if (hipMalloc(&g->d_csr_pages[w], pb) != hipSuccess ||
+                hipMemcpy(g->d_csr_pages[w], st->csr[w].pages.data(), pb,
+                          hipMemcpyHostToDevice) != hipSuccess ||
+                hipMalloc(&g->d_csr_entries[w], eb) != hipSuccess ||
+                hipMemcpy(g->d_csr_entries[w], st->csr[w].entries.data(),
+                          st->csr[w].entries.size() * 8,
+                          hipMemcpyHostToDevice) != hipSuccess) {
+                wk_gpu_store_destroy(g);
+                return nullptr;
+            }
+        }
+    }
     if (!st->tbm.empty()) {
         g->d_tbm.assign(st->tbm.size(), nullptr);
         for (size_t t = 0; t < st->tbm.size(); t++) {
@@ -1922,6 +2015,10 @@ extern "C" void wk_gpu_store_destroy(wk_gpu_store_t *g) {
     for (fnpage_t *p : g->d_fn_pages)
         if (p) (void)hipFree(p);
     for (sid_t *p : g->d_fn_vals)
+        if (p) (void)hipFree(p);
+    for (fnpage_t *p : g->d_csr_pages)
+        if (p) (void)hipFree(p);
+    for (uint64_t *p : g->d_csr_entries)
         if (p) (void)hipFree(p);
     for (uint64_t *p : g->d_tbm)
         if (p) (void)hipFree(p);
@@ -2751,13 +2848,35 @@ static int32_t exec_pattern(wk_engine *e) {
                 }
             }
         }
+        // CSR side index (rank-compressed, non-functional segments):
+        // 24-B probes via a barrier-free gather + chunked scan; the
+        // cluster-hash probe remains the fallback and the
+        // WK_FN_DISPATCH=0 diagnostic path
+        const fnpage_t *c_pg =
+            (key_mode == PK_NORMAL && e->gs && wk_fn_dispatch() &&
+             !e->gs->d_csr_pages.empty())
+                ? e->gs->d_csr_pages[(size_t)p * 2 + dir]
+                : nullptr;
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_probe_scan, dim3(G), dim3(SCAN_T), 0, e->stream,
-                           e->d_verts, cur_tbl, e->ncols, col, (uint32_t)p, dir,
-                           key_mode, seg->bucket_start, seg->num_buckets,
-                           e->d_state, e->d_stats, (uint64_t *)e->eoff.p,
-                           (uint32_t *)e->cnt.p, (uint64_t *)e->prefix.p,
-                           (uint64_t *)e->bsums.p);
+        if (c_pg) {
+            hipLaunchKernelGGL(k_csr_gather, dim3(grid_for(e->bound)),
+                               dim3(BLOCK), 0, e->stream, cur_tbl, e->ncols,
+                               col, c_pg,
+                               e->gs->d_csr_entries[(size_t)p * 2 + dir],
+                               st->fn_base, st->fn_n, e->d_state, e->d_stats,
+                               (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p);
+            hipLaunchKernelGGL(k_scan_local, dim3(G), dim3(SCAN_T), 0,
+                               e->stream, (const uint32_t *)e->cnt.p,
+                               e->d_state, (uint64_t *)e->prefix.p,
+                               (uint64_t *)e->bsums.p);
+        } else {
+            hipLaunchKernelGGL(k_probe_scan, dim3(G), dim3(SCAN_T), 0, e->stream,
+                               e->d_verts, cur_tbl, e->ncols, col, (uint32_t)p, dir,
+                               key_mode, seg->bucket_start, seg->num_buckets,
+                               e->d_state, e->d_stats, (uint64_t *)e->eoff.p,
+                               (uint32_t *)e->cnt.p, (uint64_t *)e->prefix.p,
+                               (uint64_t *)e->bsums.p);
+        }
         TIME_END(e, CAT_PROBE);
         if (fuse) {
             {

@@ -516,6 +516,95 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
         }
     }
     t0 = now_s();
+    // rank-compressed CSR side index for NON-functional segments (the
+    // fn-map structure generalized to any degree): pages + u64
+    // {edge_off:40 | len:24} entries.  Probes drop from a 128-B bucket
+    // walk to 24 B.  Budgeted largest-keys first after fn maps;
+    // WK_CSR=0 disables.
+    {
+        const char *cv = getenv("WK_CSR");
+        bool want = !(cv && !atoi(cv));
+        if (want && max_id >= (1u << NBITS_IDX)) {
+            if (!st->fn_n) {  // WK_FN=0: span bookkeeping still needed
+                st->fn_base = 1u << NBITS_IDX;
+                st->fn_n = (uint64_t)max_id + 1 - st->fn_base;
+            }
+            st->csr.assign((size_t)NP * 2, {});
+            const char *bv = getenv("WK_CSR_BUDGET_GB");
+            uint64_t gb = bv ? strtoull(bv, nullptr, 10)
+                             : (uint64_t)(16 / (nsrv > 0 ? nsrv : 1));
+            uint64_t budget = std::max<uint64_t>(gb, 2) << 30;
+            const uint64_t npages = (st->fn_n + 63) / 64;
+            std::vector<std::pair<uint64_t, uint32_t>> cand;  // (keys, w)
+            auto has_fn = [&](uint32_t w) {
+                return !st->fn.empty() && st->fn[w].present();
+            };
+            for (uint32_t p = 1; p <= max_pid; p++) {
+                if (out_keys[p] && !has_fn(p * 2 + DIR_OUT))
+                    cand.push_back({out_keys[p], p * 2 + DIR_OUT});
+                if (in_keys[p] && !has_fn(p * 2 + DIR_IN))
+                    cand.push_back({in_keys[p], p * 2 + DIR_IN});
+            }
+            std::sort(cand.begin(), cand.end(),
+                      [](auto &a, auto &b) { return a.first > b.first; });
+            std::vector<uint32_t> picked;
+            uint64_t spent = 0;
+            for (auto &c : cand) {
+                uint64_t bytes = npages * sizeof(fnpage_t) + c.first * 8;
+                if (spent + bytes > budget) break;
+                spent += bytes;
+                picked.push_back(c.second);
+            }
+#pragma omp parallel for schedule(dynamic)
+            for (size_t ci = 0; ci < picked.size(); ci++) {
+                uint32_t w = picked[ci];
+                uint32_t p = w / 2;
+                bool out = (w & 1) == DIR_OUT;
+                auto &m = st->csr[w];
+                m.pages.assign(npages, fnpage_t{0, 0, 0});
+                const std::vector<triple> &arr = out ? pso : pos;
+                const std::vector<int64_t> &lo = out ? pso_lo : pos_lo;
+                m.entries.reserve((size_t)(out ? out_keys[p] : in_keys[p]));
+                // edge offsets mirror the insert pass exactly: the
+                // segment cursor advances one edge per kept triple
+                uint64_t cur = seg_edge_start[w];
+                bool ok = true;
+                for (int64_t i = lo[p]; i < lo[p + 1] && ok;) {
+                    sid_t v = out ? arr[i].s : arr[i].o;
+                    int64_t j = i + 1;
+                    while (j < lo[p + 1] && (out ? arr[j].s : arr[j].o) == v)
+                        j++;
+                    if (!out && is_tpid(v)) { i = j; continue; }
+                    uint64_t len = (uint64_t)(j - i);
+                    if ((uint64_t)v < st->fn_base || len >= (1ull << 24) ||
+                        cur >= (1ull << 40)) {
+                        ok = false;  // out of entry range: drop this map
+                        break;
+                    }
+                    uint64_t idx = v - st->fn_base;
+                    m.pages[idx >> 6].bits |= 1ull << (idx & 63);
+                    m.entries.push_back((cur << 24) | len);
+                    cur += len;
+                    i = j;
+                }
+                if (!ok) {
+                    m.pages.clear();
+                    m.entries.clear();
+                    continue;
+                }
+                uint32_t rank = 0;
+                for (uint64_t g = 0; g < npages; g++) {
+                    m.pages[g].rank = rank;
+                    rank += (uint32_t)__builtin_popcountll(m.pages[g].bits);
+                }
+            }
+            size_t nbuilt = 0;
+            for (auto &m : st->csr) nbuilt += m.present();
+            WK_LOG("[store] csr side index: %zu segments, %.1f GB (%.1fs)\n",
+                   nbuilt, spent / 1e9, now_s() - t0);
+        }
+    }
+    t0 = now_s();
     // per-type membership bitmaps from the TYPE_ID pso slice (exact:
     // includes multi-type vids).  Budgeted; WK_TBM=0 disables.
     {
@@ -606,6 +695,8 @@ extern "C" int32_t wk_store_mem_usage(const wk_store_t *st,
         side += st->vp_off[d].size() * 4 + st->vp_edges[d].size() * 4;
     for (auto &m : st->fn)
         side += m.pages.size() * sizeof(fnpage_t) + m.vals.size() * 4;
+    for (auto &m : st->csr)
+        side += m.pages.size() * sizeof(fnpage_t) + m.entries.size() * 8;
     for (auto &b : st->tbm) side += b.size() * 8;
     if (side_index_bytes) *side_index_bytes = side;
     return 0;

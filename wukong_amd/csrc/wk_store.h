@@ -94,6 +94,20 @@ struct wk_store {
     std::vector<fnmap> fn;
     uint64_t fn_base = 0, fn_n = 0;
 
+    // rank-compressed CSR side index for NON-functional segments (the
+    // fn-map idea generalized): pages as above, entries u64
+    // {edge_off:40 | len:24} rank-indexed.  A k2u probe becomes a 16-B
+    // page load + 8-B entry load instead of a 128-B bucket walk (the
+    // cluster-hash stays the keyed layout of record; this is another
+    // HBM-capacity-funded densification like type_of/fn).  Indexed
+    // [pid*2+dir]; empty = absent.
+    struct csrmap {
+        std::vector<wk::fnpage_t> pages;
+        std::vector<uint64_t> entries;
+        bool present() const { return !pages.empty(); }
+    };
+    std::vector<csrmap> csr;
+
     // host-side lookup (mirrors the device fn_lookup)
     wk::sid_t fn_get(size_t w, wk::sid_t v) const {
         if (w >= fn.size() || !fn[w].present()) return 0;
