@@ -436,9 +436,10 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
         uint32_t cnt = 0;
 #pragma unroll
         for (int k = 0; k < K; k++) {
-            // row layout: thread t owns rows base + t*K + k (contiguous per
-            // thread, so kept rows stay grouped and writes coalesce-ish)
-            const int64_t r = base + (int64_t)threadIdx.x * K + k;
+            // thread t owns rows {t, t+256, ...} within the tile: reads
+            // are lane-adjacent; the staged write loop handles output
+            // contiguity (see k_fn_compact note)
+            const int64_t r = base + k * SCAN_T + threadIdx.x;
             rr[k] = r;
             keep[k] = false;
             if (r >= nrows) continue;
@@ -519,7 +520,7 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
         uint32_t p = my_end - cnt;
 #pragma unroll
         for (int k = 0; k < K; k++)
-            if (keep[k]) sidx[p++] = (uint16_t)(threadIdx.x * K + k);
+            if (keep[k]) sidx[p++] = (uint16_t)(k * SCAN_T + threadIdx.x);
         __syncthreads();
         for (uint32_t j = threadIdx.x; j < block_total; j += SCAN_T) {
             const int64_t src = base + sidx[j];
@@ -788,9 +789,12 @@ __global__ void k_fn_compact(const sid_t *__restrict__ tbl,
          base += stride) {
         sid_t val[K];
         uint32_t cnt = 0;
+        // thread t owns rows {t, t+256, ...} within the tile: every
+        // read instruction is lane-adjacent (the thread-contiguous
+        // K-row layout gave a 16-B lane stride = 4x read over-fetch)
 #pragma unroll
         for (int k = 0; k < K; k++) {
-            const int64_t r = base + (int64_t)threadIdx.x * K + k;
+            const int64_t r = base + k * SCAN_T + threadIdx.x;
             val[k] = (r < nrows) ? d_val[r] : 0;
             cnt += val[k] ? 1u : 0u;
         }
@@ -816,7 +820,7 @@ __global__ void k_fn_compact(const sid_t *__restrict__ tbl,
         uint32_t p = my_end - cnt;
 #pragma unroll
         for (int k = 0; k < K; k++)
-            if (val[k]) sidx[p++] = (uint16_t)(threadIdx.x * K + k);
+            if (val[k]) sidx[p++] = (uint16_t)(k * SCAN_T + threadIdx.x);
         __syncthreads();
         for (uint32_t j = threadIdx.x; j < block_total; j += SCAN_T) {
             const int64_t src = base + sidx[j];
