@@ -658,9 +658,9 @@ def main():
     # algorithmic bytes (device-counted) / HIP-event launch time
     def roofline_probe():
         if distributed:
-            return {}, {}, {}
+            return {}, {}, {}, {}
         if args.watdiv:
-            return {}, {}, {}
+            return {}, {}, {}, {}
 
         def run_steps(e0):
             e0.begin_query(Q.ALL["q1"])
@@ -679,21 +679,35 @@ def main():
             return delta
 
         e0 = engines[0]
-        # classic probe+scan+expand pipeline (the production path for
-        # every non-functional predicate: takesCourse, advisor, pub...)
+        # classic probe+scan+expand pipeline (the cluster-hash layout
+        # of record; diagnostic arm, WK_FN_DISPATCH=0)
         os.environ["WK_FN_DISPATCH"] = "0"
         try:
             d = run_steps(e0)
             rl_expand, rl_probe = d("expand"), d("probe")
         finally:
             os.environ.pop("WK_FN_DISPATCH", None)
-        # functional-predicate dense-map path (the production path for
-        # this specific step): one 4-byte gather replaces the probe
+        # rank-compressed functional-map path (the production path for
+        # this step): 16-B page + 4-B value gather replaces the probe
         d2 = run_steps(e0)
         rl_fn = d2("expand")
-        return rl_expand, rl_probe, rl_fn
+        # rank-compressed CSR side-index probe (production path for
+        # NON-functional predicates): q7's advisor expansion,
+        # 1.5M rows -> 24-B probes via k_csr_gather + k_scan_local
+        e0.begin_query(Q.ALL["q7"])
+        e0.execute_one_pattern()           # i2u full professors
+        s0 = e0.kernel_stats()
+        e0.execute_one_pattern()           # k2u advisor (CSR probe)
+        s1 = e0.kernel_stats()
+        e0.fetch_count()
+        du = s1["probe"]["usec"] - s0["probe"]["usec"]
+        db = s1["probe"]["bytes"] - s0["probe"]["bytes"]
+        rl_csr = {"usec": round(du, 1), "bytes": db,
+                  "launches": s1["probe"]["launches"] - s0["probe"]["launches"],
+                  "gbs": round(db / du / 1e3, 1) if du > 0 else None}
+        return rl_expand, rl_probe, rl_fn, rl_csr
 
-    rl_expand, rl_probe, rl_fn = roofline_probe()
+    rl_expand, rl_probe, rl_fn, rl_csr = roofline_probe()
     dk = "expand"
     d_us = stats1[dk]["usec"] - stats0[dk]["usec"]
     d_by = stats1[dk]["bytes"] - stats0[dk]["bytes"]
@@ -728,6 +742,10 @@ def main():
         # 1:1 k_expand_fn_map at ~5-6 TB/s): algorithmic bytes are
         # 148->28 per row, so GB/s reads lower while wall time is lower
         "fn_map_kernel": rl_fn or None,
+        # non-functional-segment probes through the rank-compressed CSR
+        # side index (q7 advisor step; the production path for
+        # takesCourse/advisor/pubAuthor expansions)
+        "csr_probe_kernel": rl_csr or None,
     }
 
     # embedded emulator leg (driver-reproducible evidence for the
