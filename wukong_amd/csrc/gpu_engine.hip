@@ -389,46 +389,100 @@ __global__ void k_scan_local(const uint32_t *__restrict__ d_cnt,
     if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
 }
 
-// Filter operators (k2c/k2k/c2k/i2k): probe + membership test + block-
-// aggregated compaction.  Each thread handles 4 rows per tile (4 loads in
-// flight, and 4x fewer global atomics: a single counter saturates at
-// ~88 adds/us — MI355X_MICROARCH.md row `dequeue`).  Row order is
-// engine-internal; parity is set-level (sparql.hpp:455-476 semantics).
-__global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
-                             const sid_t *__restrict__ edges,
-                             uint64_t bucket_start, uint64_t num_buckets,
-                             const sid_t *__restrict__ tbl, int ncols,
-                             int col, uint32_t pid, int dir, int key_mode,
-                             int probe_mode, int col2, sid_t cval,
-                             uint64_t list_off, uint64_t list_sz,
-                             const uint16_t *__restrict__ type_of,
-                             uint64_t type_base, uint64_t type_n,
-                             int use_typeof,
-                             const uint64_t *__restrict__ tbm,
-                             const fnpage_t *__restrict__ fn_pg,
-                             const sid_t *__restrict__ fn_vals,
-                             uint64_t fn_base,
-                             uint64_t fn_n, int fn_swap, int verify_only,
+// Filter predicate (k2c/k2k/c2k/i2k keep-row decision — the shared
+// core of the one-pass and scan-pipeline filter forms below).
+struct fparams {
+    const vertex_t *verts;
+    const sid_t *edges;
+    uint64_t bucket_start, num_buckets;
+    const sid_t *tbl;
+    int ncols, col;
+    uint32_t pid;
+    int dir, key_mode, probe_mode, col2;
+    sid_t cval;
+    uint64_t list_off, list_sz;
+    const uint16_t *type_of;
+    uint64_t type_base, type_n;
+    int use_typeof;
+    const uint64_t *tbm;
+    const fnpage_t *fn_pg;
+    const sid_t *fn_vals;
+    uint64_t fn_base, fn_n;
+    int fn_swap;
+};
+
+__device__ __forceinline__ bool filter_keep(const fparams &P, int64_t r) {
+    sid_t v = P.tbl[r * P.ncols + P.col];
+    if (P.use_typeof) {
+        uint64_t idx = (uint64_t)v - P.type_base;
+        if (P.tbm)
+            // per-type bitmap: 1 bit/vid, whole map LLC-resident —
+            // exact (multi-type included), no probe fallback
+            return idx < P.type_n && ((P.tbm[idx >> 6] >> (idx & 63)) & 1);
+        uint16_t t = (idx < P.type_n) ? P.type_of[idx] : 0;
+        if (t != 0xFFFF) return (sid_t)t == P.cval;
+    }
+    if (P.probe_mode == PM_EQ) {
+        // reversed functional map resolved host-side: the edge exists
+        // iff the row value IS the precomputed vertex
+        return v == P.cval;
+    }
+    if (P.probe_mode == PM_LIST && !P.use_typeof)
+        return bsearch_u32(P.edges + P.list_off, P.list_sz, v);
+    if (P.fn_pg) {
+        // functional predicate: the row's single object replaces the
+        // probe + edge-list search (rank-compressed map: 16-B page +
+        // 4-B value vs 148 bytes of hash traffic).  fn_swap: the
+        // REVERSED direction is functional — check fn[other col] == col.
+        sid_t a = v, b;
+        if (P.probe_mode == PM_CONST) b = P.cval;
+        else b = P.tbl[r * P.ncols + P.col2];
+        if (P.fn_swap) { sid_t t_ = a; a = b; b = t_; }
+        sid_t tv = fn_lookup(P.fn_pg, P.fn_vals, P.fn_base, P.fn_n, a);
+        return tv && tv == b;
+    }
+    uint64_t key = (P.key_mode == PK_NORMAL)
+                       ? key_pack(v, P.pid, (uint64_t)P.dir)
+                       : key_pack(0, v, (uint64_t)P.dir);
+    uint64_t eoff = 0, esz = 0;
+    probe_one(P.verts, P.bucket_start, P.num_buckets, key, eoff, esz);
+    sid_t tgt = (P.probe_mode == PM_CONST) ? P.cval
+                                           : P.tbl[r * P.ncols + P.col2];
+    return esz && bsearch_u32(P.edges + eoff, esz, tgt);
+}
+
+__device__ __forceinline__ uint64_t filter_bytes_per_row(const fparams &P) {
+    return (P.use_typeof ? 6
+            : P.fn_pg ? 24
+            : P.probe_mode == PM_LIST ? 12
+                                      : (4 + 128 + 8 + 64)) +
+           8 * P.ncols;
+}
+
+// One-pass filter: keep + wavefront-ballot compaction.  Each tile
+// costs ONE global cursor atomic per block — a single word saturates
+// at ~88 atomics/us (microarch row `dequeue`), so ~2048 in-flight
+// blocks serialize ~23 us per tile ROUND; fine up to ~1-2M rows.
+// Larger tables route through the atomic-free scan pipeline below
+// (k_filter_flags + k_scan_local + k_scan_mid + k_filter_scatter).
+// Row order is engine-internal; parity is set-level
+// (sparql.hpp:455-476 semantics).
+__global__ void k_filter_tpr(fparams P, int verify_only,
                              int commit_mode, uint64_t commit_cap,
                              uint64_t *__restrict__ d_state,
                              uint64_t *__restrict__ d_stats,
                              sid_t *__restrict__ out_tbl)
 {
     const int64_t nrows = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, CAT_FILTER,
-                (uint64_t)nrows * ((use_typeof ? 6
-                                    : fn_pg ? 24
-                                    : probe_mode == PM_LIST ? 12
-                                                            : (4 + 128 + 8 + 64)) +
-                                   8 * ncols));
+    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows * filter_bytes_per_row(P));
     constexpr int K = 4;
-    // wavefront-ballot compaction (see k_fn_compact): 3 barriers per
-    // 1024-row tile, wave-ranked output positions, no LDS scan array
     __shared__ unsigned long long s_base;
     __shared__ unsigned int s_cnt;
     __shared__ unsigned int s_wbase[SCAN_T / 64];
     const int lane = threadIdx.x & 63;
     const int wid = threadIdx.x >> 6;
+    const int ncols = P.ncols;
+    const sid_t *__restrict__ tbl = P.tbl;
     const int64_t tile = (int64_t)blockDim.x * K;
     const int64_t stride = (int64_t)gridDim.x * tile;
 
@@ -437,54 +491,9 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
         int64_t rr[K];
 #pragma unroll
         for (int k = 0; k < K; k++) {
-            // thread t owns rows {t, t+256, ...} within the tile: reads
-            // are lane-adjacent; the staged write loop handles output
-            // contiguity (see k_fn_compact note)
             const int64_t r = base + k * SCAN_T + threadIdx.x;
             rr[k] = r;
-            keep[k] = false;
-            if (r >= nrows) continue;
-            sid_t v = tbl[r * ncols + col];
-            if (use_typeof) {
-                uint64_t idx = (uint64_t)v - type_base;
-                if (tbm) {
-                    // per-type bitmap: 1 bit/vid, whole map LLC-resident
-                    // — exact (multi-type included), no probe fallback
-                    keep[k] = idx < type_n &&
-                              ((tbm[idx >> 6] >> (idx & 63)) & 1);
-                    continue;
-                }
-                uint16_t t = (idx < type_n) ? type_of[idx] : 0;
-                if (t != 0xFFFF) { keep[k] = ((sid_t)t == cval); continue; }
-            }
-            if (probe_mode == PM_EQ) {
-                // reversed functional map resolved host-side: the edge
-                // exists iff the row value IS the precomputed vertex
-                keep[k] = (v == cval);
-            } else if (probe_mode == PM_LIST && !use_typeof) {
-                keep[k] = bsearch_u32(edges + list_off, list_sz, v);
-            } else if (fn_pg) {
-                // functional predicate: the row's single object replaces
-                // the probe + edge-list search (rank-compressed map:
-                // 16-B page + 4-B value vs 148 bytes of hash traffic).
-                // fn_swap: the REVERSED direction is functional — check
-                // fn[other col] == this col instead.
-                sid_t a = v, b;
-                if (probe_mode == PM_CONST) b = cval;
-                else b = tbl[r * ncols + col2];
-                if (fn_swap) { sid_t t_ = a; a = b; b = t_; }
-                sid_t tv = fn_lookup(fn_pg, fn_vals, fn_base, fn_n, a);
-                keep[k] = tv && tv == b;
-            } else {
-                uint64_t key = (key_mode == PK_NORMAL)
-                                   ? key_pack(v, pid, (uint64_t)dir)
-                                   : key_pack(0, v, (uint64_t)dir);
-                uint64_t eoff = 0, esz = 0;
-                probe_one(verts, bucket_start, num_buckets, key, eoff, esz);
-                sid_t tgt = (probe_mode == PM_CONST) ? cval
-                                                     : tbl[r * ncols + col2];
-                keep[k] = esz && bsearch_u32(edges + eoff, esz, tgt);
-            }
+            keep[k] = r < nrows && filter_keep(P, r);
         }
         if (verify_only) {
             // identity-verified filter (captured graphs): the warm pass
@@ -530,6 +539,68 @@ __global__ void k_filter_tpr(const vertex_t *__restrict__ verts,
         }
     }
     commit_tail(d_state, commit_cap, commit_mode);
+}
+
+// Scan-pipeline filter, phase 1: barrier- and atomic-free keep flags
+// into d_cnt (k_scan_local / k_scan_mid then produce deterministic
+// positions; k_filter_scatter writes).
+__global__ void k_filter_flags(fparams P,
+                               const uint64_t *__restrict__ d_state,
+                               uint64_t *__restrict__ d_stats,
+                               uint32_t *__restrict__ d_cnt)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows * filter_bytes_per_row(P));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x)
+        d_cnt[r] = filter_keep(P, r) ? 1u : 0u;
+}
+
+__global__ void k_filter_scatter(const sid_t *__restrict__ tbl, int ncols,
+                                 const uint32_t *__restrict__ d_cnt,
+                                 const uint64_t *__restrict__ d_pre,
+                                 const uint64_t *__restrict__ bsums, int G,
+                                 const uint64_t *__restrict__ d_state,
+                                 sid_t *__restrict__ out_tbl)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int64_t chunk = (nrows + G - 1) / G;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        if (!d_cnt[r]) continue;
+        uint64_t pos = d_pre[r] + bsums[r / chunk];
+        sid_t *dst = out_tbl + (int64_t)pos * ncols;
+        const sid_t *srow = tbl + r * ncols;
+        for (int c = 0; c < ncols; c++) dst[c] = srow[c];
+    }
+}
+
+// fn-expansion scatter (phase 4 of gather -> scan_local -> scan_mid ->
+// scatter): deterministic positions, no atomics, barrier-free.
+template <int NC>
+__global__ void k_fn_scatter(const sid_t *__restrict__ tbl,
+                             const sid_t *__restrict__ d_val,
+                             const uint32_t *__restrict__ d_cnt,
+                             const uint64_t *__restrict__ d_pre,
+                             const uint64_t *__restrict__ bsums, int G,
+                             const uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats,
+                             sid_t *__restrict__ out)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    constexpr int oc = NC + 1;
+    count_bytes(d_stats, CAT_EXPAND, (uint64_t)nrows * (4 + 4 * NC + 4 * oc));
+    const int64_t chunk = (nrows + G - 1) / G;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        if (!d_cnt[r]) continue;
+        uint64_t pos = d_pre[r] + bsums[r / chunk];
+        sid_t *dst = out + (int64_t)pos * oc;
+        const sid_t *srow = tbl + r * NC;
+#pragma unroll
+        for (int c = 0; c < NC; c++) dst[c] = srow[c];
+        dst[NC] = d_val[r];
+    }
 }
 
 // finish the scan across blocks: exclusive over the G block sums
@@ -730,7 +801,8 @@ __global__ void k_fn_gather(const sid_t *__restrict__ tbl, int ncols,
                             uint64_t f_bstart, uint64_t f_nbuckets,
                             const uint64_t *__restrict__ d_state,
                             uint64_t *__restrict__ d_stats,
-                            sid_t *__restrict__ d_val)
+                            sid_t *__restrict__ d_val,
+                            uint32_t *__restrict__ d_cnt)
 {
     const int64_t nrows = (int64_t)d_state[S_NROWS];
     count_bytes(d_stats, CAT_EXPAND,
@@ -758,6 +830,522 @@ __global__ void k_fn_gather(const sid_t *__restrict__ tbl, int ncols,
             }
         }
         d_val[r] = tv;
+        d_cnt[r] = tv ? 1u : 0u;
+    }
+}
+
+// CSR-indexed known_to_unknown front half (non-functional segments):
+// the 128-B cluster-hash bucket walk becomes a 16-B page + 8-B entry
+// lookup ({edge_off:40|len:24}, rank-compressed like the fn maps).
+// Two phases, following the two-phase fn lesson: a barrier-free 1:1
+// gather writes (eoff, cnt); k_scan_local then runs the chunked
+// prefix exactly as k_probe_scan produced it, so the expansion
+// kernels consume the identical layout.
+__global__ void k_csr_gather(const sid_t *__restrict__ tbl, int ncols,
+                             int col,
+                             const fnpage_t *__restrict__ pg,
+                             const uint64_t *__restrict__ entries,
+                             uint64_t base, uint64_t n,
+                             const uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats,
+                             uint64_t *__restrict__ d_eoff,
+                             uint32_t *__restrict__ d_cnt)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_PROBE, (uint64_t)nrows * (4 + 16 + 8 + 12));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        sid_t v = tbl[r * ncols + col];
+        uint64_t e = 0;
+        uint64_t idx = (uint64_t)v - base;
+        if (idx < n) {
+            const fnpage_t p = pg[idx >> 6];
+            if ((p.bits >> (idx & 63)) & 1) {
+                uint32_t rk = p.rank +
+                              (uint32_t)__popcll(p.bits &
+                                                 ((1ull << (idx & 63)) - 1));
+                e = entries[rk];
+            }
+        }
+        d_eoff[r] = e >> 24;
+        d_cnt[r] = (uint32_t)(e & 0xFFFFFF);
+    }
+}
+
+// chunked exclusive prefix of d_cnt -> d_pre + per-chunk sums (same
+// chunk math as k_probe_scan so the expansion kernels are unchanged)
+__global__ void k_scan_local(const uint32_t *__restrict__ d_cnt,
+                             const uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_pre,
+                             uint64_t *__restrict__ bsums)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int G = gridDim.x;
+    const int64_t chunk = (nrows + G - 1) / G;
+    const int64_t start = (int64_t)blockIdx.x * chunk;
+    const int64_t end = min(start + chunk, (int64_t)nrows);
+    __shared__ uint64_t sh[SCAN_T];
+    uint64_t carry = 0;
+    for (int64_t base = start; base < end; base += SCAN_T) {
+        const int64_t r = base + threadIdx.x;
+        uint64_t esz = (r < end) ? d_cnt[r] : 0;
+        sh[threadIdx.x] = esz;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint64_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += x;
+            __syncthreads();
+        }
+        if (r < end) d_pre[r] = carry + sh[threadIdx.x] - esz;
+        carry += sh[SCAN_T - 1];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
+}
+
+// Filter predicate (k2c/k2k/c2k/i2k keep-row decision — the shared
+// core of the one-pass and scan-pipeline filter forms below).
+struct fparams {
+    const vertex_t *verts;
+    const sid_t *edges;
+    uint64_t bucket_start, num_buckets;
+    const sid_t *tbl;
+    int ncols, col;
+    uint32_t pid;
+    int dir, key_mode, probe_mode, col2;
+    sid_t cval;
+    uint64_t list_off, list_sz;
+    const uint16_t *type_of;
+    uint64_t type_base, type_n;
+    int use_typeof;
+    const uint64_t *tbm;
+    const fnpage_t *fn_pg;
+    const sid_t *fn_vals;
+    uint64_t fn_base, fn_n;
+    int fn_swap;
+};
+
+__device__ __forceinline__ bool filter_keep(const fparams &P, int64_t r) {
+    sid_t v = P.tbl[r * P.ncols + P.col];
+    if (P.use_typeof) {
+        uint64_t idx = (uint64_t)v - P.type_base;
+        if (P.tbm)
+            // per-type bitmap: 1 bit/vid, whole map LLC-resident —
+            // exact (multi-type included), no probe fallback
+            return idx < P.type_n && ((P.tbm[idx >> 6] >> (idx & 63)) & 1);
+        uint16_t t = (idx < P.type_n) ? P.type_of[idx] : 0;
+        if (t != 0xFFFF) return (sid_t)t == P.cval;
+    }
+    if (P.probe_mode == PM_EQ) {
+        // reversed functional map resolved host-side: the edge exists
+        // iff the row value IS the precomputed vertex
+        return v == P.cval;
+    }
+    if (P.probe_mode == PM_LIST && !P.use_typeof)
+        return bsearch_u32(P.edges + P.list_off, P.list_sz, v);
+    if (P.fn_pg) {
+        // functional predicate: the row's single object replaces the
+        // probe + edge-list search (rank-compressed map: 16-B page +
+        // 4-B value vs 148 bytes of hash traffic).  fn_swap: the
+        // REVERSED direction is functional — check fn[other col] == col.
+        sid_t a = v, b;
+        if (P.probe_mode == PM_CONST) b = P.cval;
+        else b = P.tbl[r * P.ncols + P.col2];
+        if (P.fn_swap) { sid_t t_ = a; a = b; b = t_; }
+        sid_t tv = fn_lookup(P.fn_pg, P.fn_vals, P.fn_base, P.fn_n, a);
+        return tv && tv == b;
+    }
+    uint64_t key = (P.key_mode == PK_NORMAL)
+                       ? key_pack(v, P.pid, (uint64_t)P.dir)
+                       : key_pack(0, v, (uint64_t)P.dir);
+    uint64_t eoff = 0, esz = 0;
+    probe_one(P.verts, P.bucket_start, P.num_buckets, key, eoff, esz);
+    sid_t tgt = (P.probe_mode == PM_CONST) ? P.cval
+                                           : P.tbl[r * P.ncols + P.col2];
+    return esz && bsearch_u32(P.edges + eoff, esz, tgt);
+}
+
+__device__ __forceinline__ uint64_t filter_bytes_per_row(const fparams &P) {
+    return (P.use_typeof ? 6
+            : P.fn_pg ? 24
+            : P.probe_mode == PM_LIST ? 12
+                                      : (4 + 128 + 8 + 64)) +
+           8 * P.ncols;
+}
+
+// One-pass filter: keep + wavefront-ballot compaction.  Each tile
+// costs ONE global cursor atomic per block — a single word saturates
+// at ~88 atomics/us (microarch row `dequeue`), so ~2048 in-flight
+// blocks serialize ~23 us per tile ROUND; fine up to ~1-2M rows.
+// Larger tables route through the atomic-free scan pipeline below
+// (k_filter_flags + k_scan_local + k_scan_mid + k_filter_scatter).
+// Row order is engine-internal; parity is set-level
+// (sparql.hpp:455-476 semantics).
+__global__ void k_filter_tpr(fparams P, int verify_only,
+                             int commit_mode, uint64_t commit_cap,
+                             uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats,
+                             sid_t *__restrict__ out_tbl)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows * filter_bytes_per_row(P));
+    constexpr int K = 4;
+    __shared__ unsigned long long s_base;
+    __shared__ unsigned int s_cnt;
+    __shared__ unsigned int s_wbase[SCAN_T / 64];
+    const int lane = threadIdx.x & 63;
+    const int wid = threadIdx.x >> 6;
+    const int ncols = P.ncols;
+    const sid_t *__restrict__ tbl = P.tbl;
+    const int64_t tile = (int64_t)blockDim.x * K;
+    const int64_t stride = (int64_t)gridDim.x * tile;
+
+    for (int64_t base = (int64_t)blockIdx.x * tile; base < nrows; base += stride) {
+        bool keep[K];
+        int64_t rr[K];
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            const int64_t r = base + k * SCAN_T + threadIdx.x;
+            rr[k] = r;
+            keep[k] = r < nrows && filter_keep(P, r);
+        }
+        if (verify_only) {
+            // identity-verified filter (captured graphs): the warm pass
+            // saw zero drops; every replay still checks every row and
+            // flags S_ERR on any miss (no writes, no table flip)
+            uint32_t miss = 0;
+#pragma unroll
+            for (int k = 0; k < K; k++)
+                miss += (rr[k] < nrows && !keep[k]) ? 1u : 0u;
+            if (miss)
+                atomicAdd((unsigned long long *)&d_state[S_OVF],
+                          (unsigned long long)miss);
+            continue;
+        }
+        if (threadIdx.x == 0) s_cnt = 0;
+        __syncthreads();
+        uint64_t mask[K];
+        uint32_t wtot = 0;
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            mask[k] = __ballot(keep[k]);
+            wtot += (uint32_t)__popcll(mask[k]);
+        }
+        if (lane == 0) s_wbase[wid] = atomicAdd(&s_cnt, wtot);
+        __syncthreads();
+        if (threadIdx.x == 0)
+            s_base = s_cnt ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
+                                       (unsigned long long)s_cnt)
+                           : 0;
+        __syncthreads();
+        uint32_t wpos = s_wbase[wid];
+#pragma unroll
+        for (int k = 0; k < K; k++) {
+            if (keep[k]) {
+                uint64_t pos = s_base + wpos +
+                               (uint32_t)__popcll(mask[k] &
+                                                  ((1ull << lane) - 1));
+                sid_t *dst = out_tbl + (int64_t)pos * ncols;
+                const sid_t *srow = tbl + rr[k] * ncols;
+                for (int c = 0; c < ncols; c++) dst[c] = srow[c];
+            }
+            wpos += (uint32_t)__popcll(mask[k]);
+        }
+    }
+    commit_tail(d_state, commit_cap, commit_mode);
+}
+
+// Scan-pipeline filter, phase 1: barrier- and atomic-free keep flags
+// into d_cnt (k_scan_local / k_scan_mid then produce deterministic
+// positions; k_filter_scatter writes).
+__global__ void k_filter_flags(fparams P,
+                               const uint64_t *__restrict__ d_state,
+                               uint64_t *__restrict__ d_stats,
+                               uint32_t *__restrict__ d_cnt)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows * filter_bytes_per_row(P));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x)
+        d_cnt[r] = filter_keep(P, r) ? 1u : 0u;
+}
+
+__global__ void k_filter_scatter(const sid_t *__restrict__ tbl, int ncols,
+                                 const uint32_t *__restrict__ d_cnt,
+                                 const uint64_t *__restrict__ d_pre,
+                                 const uint64_t *__restrict__ bsums, int G,
+                                 const uint64_t *__restrict__ d_state,
+                                 sid_t *__restrict__ out_tbl)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int64_t chunk = (nrows + G - 1) / G;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        if (!d_cnt[r]) continue;
+        uint64_t pos = d_pre[r] + bsums[r / chunk];
+        sid_t *dst = out_tbl + (int64_t)pos * ncols;
+        const sid_t *srow = tbl + r * ncols;
+        for (int c = 0; c < ncols; c++) dst[c] = srow[c];
+    }
+}
+
+// fn-expansion scatter (phase 4 of gather -> scan_local -> scan_mid ->
+// scatter): deterministic positions, no atomics, barrier-free.
+template <int NC>
+__global__ void k_fn_scatter(const sid_t *__restrict__ tbl,
+                             const sid_t *__restrict__ d_val,
+                             const uint32_t *__restrict__ d_cnt,
+                             const uint64_t *__restrict__ d_pre,
+                             const uint64_t *__restrict__ bsums, int G,
+                             const uint64_t *__restrict__ d_state,
+                             uint64_t *__restrict__ d_stats,
+                             sid_t *__restrict__ out)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    constexpr int oc = NC + 1;
+    count_bytes(d_stats, CAT_EXPAND, (uint64_t)nrows * (4 + 4 * NC + 4 * oc));
+    const int64_t chunk = (nrows + G - 1) / G;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        if (!d_cnt[r]) continue;
+        uint64_t pos = d_pre[r] + bsums[r / chunk];
+        sid_t *dst = out + (int64_t)pos * oc;
+        const sid_t *srow = tbl + r * NC;
+#pragma unroll
+        for (int c = 0; c < NC; c++) dst[c] = srow[c];
+        dst[NC] = d_val[r];
+    }
+}
+
+// finish the scan across blocks: exclusive over the G block sums
+// (single block), total -> state[S_TOTAL]
+__global__ void k_scan_mid(uint64_t *__restrict__ bsums, int G,
+                           uint64_t *__restrict__ d_state)
+{
+    __shared__ uint64_t sh[SCAN_T];
+    uint64_t carry = 0;
+    for (int base = 0; base < G; base += SCAN_T) {
+        int i = base + threadIdx.x;
+        uint64_t x = (i < G) ? bsums[i] : 0;
+        sh[threadIdx.x] = x;
+        __syncthreads();
+        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
+            uint64_t v = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
+            __syncthreads();
+            sh[threadIdx.x] += v;
+            __syncthreads();
+        }
+        if (i < G) bsums[i] = carry + sh[threadIdx.x] - x;
+        carry += sh[SCAN_T - 1];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) d_state[S_TOTAL] = carry;
+}
+
+// advance: nrows = min(total, cap); flag overflow for the host re-run
+// (replaces the reference's rbuf-overflow assert, gpu_engine_cuda.hpp:185)
+__global__ void k_commit(uint64_t *__restrict__ d_state, uint64_t cap) {
+    uint64_t t = d_state[S_TOTAL];
+    if (t > cap) {
+        d_state[S_ERR] = 1;
+        d_state[S_REQ] = max(d_state[S_REQ], t);
+        t = cap;
+    }
+    d_state[S_NROWS] = t;
+    // reset the accumulators for the next step (saves a 3us k_zero_words
+    // launch per step — the state kernels were 12% of suite GPU time)
+    d_state[S_TOTAL] = 0;
+    d_state[S_OVF] = 0;
+}
+
+__global__ void k_set_state(uint64_t *__restrict__ d_state, uint64_t nrows) {
+    d_state[S_NROWS] = nrows;
+    d_state[S_TOTAL] = 0;
+    d_state[S_OVF] = 0;
+}
+
+// device projection to required-var columns (sparql.hpp:1510-1536)
+struct cols8 { int32_t c[8]; };
+__global__ void k_project(const sid_t *__restrict__ tbl, int ncols,
+                          const uint64_t *__restrict__ d_state, cols8 cols,
+                          int rc, sid_t *__restrict__ out)
+{
+    const int64_t n = (int64_t)d_state[S_NROWS];
+    for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < n * rc;
+         t += (int64_t)gridDim.x * blockDim.x) {
+        int64_t i = t / rc;
+        int j = (int)(t - i * rc);
+        int c = cols.c[j];
+        out[t] = (c >= 0) ? tbl[i * ncols + c] : (sid_t)0xFFFFFFFFu;
+    }
+}
+
+__global__ void k_publish_state(const uint64_t *__restrict__ d_state,
+                                const uint64_t *__restrict__ d_stats,
+                                uint64_t *__restrict__ h_pin) {
+    for (int i = 0; i < S_WORDS; i++) h_pin[i] = d_state[i];
+    for (int i = 0; i < CAT_COUNT; i++) h_pin[8 + i] = d_stats[i];
+    // sticky error flag (h_pin slot 7): survives across back-to-back
+    // graph replays whose own begin kernels clear d_state — the
+    // one-sync-per-pass path checks and clears it host-side
+    if (d_state[S_ERR]) h_pin[7] = 1;
+}
+
+// Input-centric expansion (known_to_unknown back half,
+// sparql.hpp:325-367): thread r writes its deg outputs at
+// pre[r]+bsums[chunk(r)] — no per-output binary search (the
+// output-centric version paid 2-3 DRAM lines of prefix walk per output
+// row).  Rows with deg > 32 go to an overflow queue handled by
+// k_expand_big with one WAVE per row (lanes stride the edge list).
+template <int NC>
+__global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
+                            const sid_t *__restrict__ edges,
+                            const uint64_t *__restrict__ d_eoff,
+                            const uint32_t *__restrict__ d_cnt,
+                            const uint64_t *__restrict__ d_pre,
+                            const uint64_t *__restrict__ bsums, int G,
+                            uint64_t *__restrict__ d_state, uint64_t cap,
+                            uint64_t *__restrict__ d_stats,
+                            uint32_t *__restrict__ ovf,
+                            sid_t *__restrict__ out)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int64_t chunk = (nrows + G - 1) / G;
+    constexpr int oc = NC + 1;
+    (void)ncols;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
+         r += (int64_t)gridDim.x * blockDim.x) {
+        uint32_t deg = d_cnt[r];
+        if (!deg) continue;
+        uint64_t basep = d_pre[r] + bsums[r / chunk];
+        if (basep >= cap) continue;  // overflow: flagged by k_commit
+        if (deg > 32) {
+            unsigned long long i = atomicAdd(
+                (unsigned long long *)&d_state[S_OVF], 1ull);
+            ovf[i] = (uint32_t)r;
+            continue;
+        }
+        if (basep + deg > cap) deg = (uint32_t)(cap - basep);
+        sid_t row[NC];
+#pragma unroll
+        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
+        const sid_t *el = edges + d_eoff[r];
+        sid_t *dst = out + (int64_t)basep * oc;
+        for (uint32_t k = 0; k < deg; k++) {
+#pragma unroll
+            for (int c = 0; c < NC; c++) dst[c] = row[c];
+            dst[NC] = el[k];
+            dst += oc;
+        }
+    }
+    // algorithmic bytes for the whole expansion (counted once; includes
+    // the big-row pass): total*(edge 4 + write 4*oc) + nrows*(row 4*ncols
+    // + cnt/pre/eoff 20)
+    if (blockIdx.x == 0 && threadIdx.x == 0)
+        atomicAdd((unsigned long long *)&d_stats[CAT_EXPAND],
+                  (unsigned long long)(min(d_state[S_TOTAL], cap) * (4 + 4 * oc) +
+                                       (uint64_t)nrows * (4 * ncols + 20)));
+}
+
+// big-fanout rows: one wave per queued row, lanes stride the edge list
+// (coalesced writes: adjacent lanes write adjacent output rows)
+template <int NC>
+__global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
+                             const sid_t *__restrict__ edges,
+                             const uint64_t *__restrict__ d_eoff,
+                             const uint32_t *__restrict__ d_cnt,
+                             const uint64_t *__restrict__ d_pre,
+                             const uint64_t *__restrict__ bsums, int G,
+                             uint64_t *__restrict__ d_state, uint64_t cap,
+                             int commit_mode,
+                             const uint32_t *__restrict__ ovf,
+                             sid_t *__restrict__ out)
+{
+    const int64_t nq = (int64_t)d_state[S_OVF];
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int64_t chunk = (nrows + G - 1) / G;
+    constexpr int oc = NC + 1;
+    (void)ncols;
+    const int lane = threadIdx.x & 63;
+    const int64_t w0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    for (int64_t q = w0; q < nq; q += nw) {
+        const int64_t r = ovf[q];
+        uint64_t deg = d_cnt[r];
+        uint64_t basep = d_pre[r] + bsums[r / chunk];
+        if (basep >= cap) continue;
+        if (basep + deg > cap) deg = cap - basep;
+        sid_t row[NC];
+#pragma unroll
+        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
+        const sid_t *el = edges + d_eoff[r];
+        for (uint64_t k = lane; k < deg; k += 64) {
+            sid_t *dst = out + (int64_t)(basep + k) * oc;
+#pragma unroll
+            for (int c = 0; c < NC; c++) dst[c] = row[c];
+            dst[NC] = el[k];
+        }
+    }
+    commit_tail(d_state, cap, commit_mode);
+}
+
+// known_to_unknown over a FUNCTIONAL predicate (every key deg==1,
+// rank-compressed map): the probe+scan+expand pipeline collapses into
+// a page+value gather pair + compacted append, with an optional fused
+// `?v rdf:type CONST` filter on the NEW column (plan pairs like Q1's
+// ugDegreeFrom -> University).  Two phases: a 1:1 grid-stride GATHER
+// (the measured-fast k_expand_fn_map shape — no barriers between the
+// dependent page/value loads) writes each row's resolved object (0 =
+// miss/filtered) to a scratch stream; the COMPACT phase then reads it
+// SEQUENTIALLY with the block-aggregated scan.  The round-1 single-pass
+// form interleaved the gathers with 16 scan barriers per tile and ran
+// 4-5x slower than its own gather cost (169 vs 36 us on Q1's 6.4M-row
+// step).  Multi-type (0xFFFF) falls back to a probe of [val|TYPE|OUT].
+// Output rows <= input rows, so capacity can never overflow.
+__global__ void k_fn_gather(const sid_t *__restrict__ tbl, int ncols,
+                            const fnpage_t *__restrict__ fn_pg,
+                            const sid_t *__restrict__ fn_vals,
+                            uint64_t fn_base, uint64_t fn_n, int col,
+                            int use_typeof, sid_t fcval,
+                            const uint16_t *__restrict__ type_of,
+                            const uint64_t *__restrict__ tbm,
+                            uint64_t type_base, uint64_t type_n,
+                            const vertex_t *__restrict__ verts,
+                            const sid_t *__restrict__ edges,
+                            uint64_t f_bstart, uint64_t f_nbuckets,
+                            const uint64_t *__restrict__ d_state,
+                            uint64_t *__restrict__ d_stats,
+                            sid_t *__restrict__ d_val,
+                            uint32_t *__restrict__ d_cnt)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_EXPAND,
+                (uint64_t)nrows * (24 + 4 + (use_typeof ? 2 : 0)));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        sid_t v = tbl[r * ncols + col];
+        sid_t tv = fn_lookup(fn_pg, fn_vals, fn_base, fn_n, v);
+        if (tv && use_typeof) {
+            uint64_t tix = (uint64_t)tv - type_base;
+            if (tbm) {  // bitmap: decision is exact (multi-type incl.)
+                if (!(tix < type_n && ((tbm[tix >> 6] >> (tix & 63)) & 1)))
+                    tv = 0;
+            } else {
+                uint16_t t = (tix < type_n) ? type_of[tix] : 0;
+                if (t != 0xFFFF) {
+                    if ((sid_t)t != fcval) tv = 0;
+                } else {
+                    uint64_t eo = 0, es = 0;
+                    probe_one(verts, f_bstart, f_nbuckets,
+                              key_pack(tv, TYPE_ID, (uint64_t)DIR_OUT), eo,
+                              es);
+                    if (!(es && bsearch_u32(edges + eo, es, fcval))) tv = 0;
+                }
+            }
+        }
+        d_val[r] = tv;
+        d_cnt[r] = tv ? 1u : 0u;
     }
 }
 
@@ -2348,19 +2936,29 @@ static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
         (fuse && e->gs && (size_t)fcval < e->gs->d_tbm.size())
             ? e->gs->d_tbm[fcval]
             : nullptr;
-    // phase 1: 1:1 gather of resolved objects into the cnt scratch
-    // (sid_t-sized); phase 2: sequential-read compaction + fused commit
+    // atomic-free pipeline: 1:1 gather of resolved objects (+0/1
+    // counts), deterministic positions via the chunked scan, then a
+    // barrier-free scatter.  A per-tile cursor atomic costs ~23 us per
+    // 2048-block round (single-word ~88 atomics/us) and held every
+    // in-kernel-aggregation variant of this step at a flat ~107 us.
+    const int G = scan_grid(e->bound);
     hipLaunchKernelGGL(k_fn_gather, dim3(grid_for(e->bound)), dim3(BLOCK),
                        0, e->stream, cur_tbl, e->ncols, d_pg, d_vals,
                        e->st->fn_base, e->st->fn_n, col, fuse ? 1 : 0, fcval,
                        e->d_type_of, tbm, e->st->type_base, e->st->type_n,
                        e->d_verts, e->d_edges, fseg ? fseg->bucket_start : 0,
                        fseg ? fseg->num_buckets : 0, e->d_state, e->d_stats,
-                       (sid_t *)e->cnt.p);
-    hipLaunchKernelGGL(k_fn_compact<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
-                       0, e->stream, cur_tbl, (const sid_t *)e->cnt.p,
-                       e->d_state, e->d_stats, (uint64_t)e->cap_rows,
-                       /*commit*/ 0, out_tbl);
+                       (sid_t *)e->ovf.p, (uint32_t *)e->cnt.p);
+    hipLaunchKernelGGL(k_scan_local, dim3(G), dim3(SCAN_T), 0, e->stream,
+                       (const uint32_t *)e->cnt.p, e->d_state,
+                       (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p);
+    hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0, e->stream,
+                       (uint64_t *)e->bsums.p, G, e->d_state);
+    hipLaunchKernelGGL(k_fn_scatter<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
+                       0, e->stream, cur_tbl, (const sid_t *)e->ovf.p,
+                       (const uint32_t *)e->cnt.p, (const uint64_t *)e->prefix.p,
+                       (const uint64_t *)e->bsums.p, G, e->d_state, e->d_stats,
+                       out_tbl);
 }
 
 template <int NC>
@@ -2641,14 +3239,11 @@ static int32_t exec_pattern(wk_engine *e) {
         const sid_t *ptr = store_get(*st, (uint64_t)s, (uint64_t)p, dir, &sz);
         uint64_t off = ptr ? (uint64_t)(ptr - st->edges.data()) : 0;
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
-                           e->stream, e->d_verts, e->d_edges, 0, 1, cur_tbl,
-                           e->ncols, col, 0u, dir, PK_NORMAL, PM_LIST, 0, 0u,
-                           off, sz, e->d_type_of, 0, 0, 0,
-                           (const uint64_t *)nullptr,
-                           (const fnpage_t *)nullptr,
-                           (const sid_t *)nullptr, 0, 0, 0, 0,
-                           /*commit*/ 0, 0,
+        fparams P{e->d_verts, e->d_edges, 0, 1, cur_tbl, e->ncols, col, 0u,
+                  dir, PK_NORMAL, PM_LIST, 0, 0u, off, sz, e->d_type_of,
+                  0, 0, 0, nullptr, nullptr, nullptr, 0, 0, 0};
+        hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK),
+                           0, e->stream, P, 0, /*commit*/ 0, 0,
                            e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
         hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
@@ -2737,15 +3332,37 @@ static int32_t exec_pattern(wk_engine *e) {
                 }
             }
         }
-        hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
-                           e->stream, e->d_verts, e->d_edges, seg->bucket_start,
-                           seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
-                           dir, key_mode, pmode, col2, cval, 0, 0,
-                           e->d_type_of, e->st->type_base, e->st->type_n,
-                           use_typeof, d_tbm, d_pg, d_vals, st->fn_base, st->fn_n,
-                           fn_swap, verify_only ? 1 : 0,
-                           /*commit*/ 0, 0,
-                           e->d_state, e->d_stats, out_tbl);
+        fparams P{e->d_verts, e->d_edges, seg->bucket_start,
+                  seg->num_buckets, cur_tbl, e->ncols, col, (uint32_t)p,
+                  dir, key_mode, pmode, col2, cval, 0, 0, e->d_type_of,
+                  e->st->type_base, e->st->type_n, use_typeof, d_tbm,
+                  d_pg, d_vals, st->fn_base, st->fn_n, fn_swap};
+        if (!verify_only && e->bound > (int64_t)(1 << 20)) {
+            // big tables: atomic-free scan pipeline (per-tile cursor
+            // atomics serialize at ~88/us — see k_filter_tpr header)
+            const int G = scan_grid(e->bound);
+            hipLaunchKernelGGL(k_filter_flags, dim3(grid_for(e->bound)),
+                               dim3(BLOCK), 0, e->stream, P, e->d_state,
+                               e->d_stats, (uint32_t *)e->cnt.p);
+            hipLaunchKernelGGL(k_scan_local, dim3(G), dim3(SCAN_T), 0,
+                               e->stream, (const uint32_t *)e->cnt.p,
+                               e->d_state, (uint64_t *)e->prefix.p,
+                               (uint64_t *)e->bsums.p);
+            hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0,
+                               e->stream, (uint64_t *)e->bsums.p, G,
+                               e->d_state);
+            hipLaunchKernelGGL(k_filter_scatter, dim3(grid_for(e->bound)),
+                               dim3(BLOCK), 0, e->stream, cur_tbl, e->ncols,
+                               (const uint32_t *)e->cnt.p,
+                               (const uint64_t *)e->prefix.p,
+                               (const uint64_t *)e->bsums.p, G, e->d_state,
+                               out_tbl);
+        } else {
+            hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)),
+                               dim3(BLOCK), 0, e->stream, P,
+                               verify_only ? 1 : 0, /*commit*/ 0, 0,
+                               e->d_state, e->d_stats, out_tbl);
+        }
         TIME_END(e, CAT_FILTER);
         if (verify_only) {
             hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0, e->stream,
@@ -3016,13 +3633,11 @@ extern "C" int32_t wk_engine_execute_filter_list(wk_engine_t *e,
     sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
     sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
     TIME_BEGIN(e);
+    fparams P{e->d_verts, (const sid_t *)e->misc.p, 0, 1, cur_tbl, e->ncols,
+              col, 0u, pat.direction, PK_NORMAL, PM_LIST, 0, 0u, 0, n,
+              e->d_type_of, 0, 0, 0, nullptr, nullptr, nullptr, 0, 0, 0};
     hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
-                       e->stream, e->d_verts, (const sid_t *)e->misc.p, 0, 1,
-                       cur_tbl, e->ncols, col, 0u, pat.direction, PK_NORMAL,
-                       PM_LIST, 0, 0u, 0, n, e->d_type_of, 0, 0, 0,
-                       (const uint64_t *)nullptr, (const fnpage_t *)nullptr,
-                       (const sid_t *)nullptr,
-                       0, 0, 0, 0, /*commit*/ 0, 0,
+                       e->stream, P, 0, /*commit*/ 0, 0,
                        e->d_state, e->d_stats, out_tbl);
     TIME_END(e, CAT_FILTER);
     hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
