@@ -834,589 +834,6 @@ __global__ void k_fn_gather(const sid_t *__restrict__ tbl, int ncols,
     }
 }
 
-// CSR-indexed known_to_unknown front half (non-functional segments):
-// the 128-B cluster-hash bucket walk becomes a 16-B page + 8-B entry
-// lookup ({edge_off:40|len:24}, rank-compressed like the fn maps).
-// Two phases, following the two-phase fn lesson: a barrier-free 1:1
-// gather writes (eoff, cnt); k_scan_local then runs the chunked
-// prefix exactly as k_probe_scan produced it, so the expansion
-// kernels consume the identical layout.
-__global__ void k_csr_gather(const sid_t *__restrict__ tbl, int ncols,
-                             int col,
-                             const fnpage_t *__restrict__ pg,
-                             const uint64_t *__restrict__ entries,
-                             uint64_t base, uint64_t n,
-                             const uint64_t *__restrict__ d_state,
-                             uint64_t *__restrict__ d_stats,
-                             uint64_t *__restrict__ d_eoff,
-                             uint32_t *__restrict__ d_cnt)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, CAT_PROBE, (uint64_t)nrows * (4 + 16 + 8 + 12));
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
-        sid_t v = tbl[r * ncols + col];
-        uint64_t e = 0;
-        uint64_t idx = (uint64_t)v - base;
-        if (idx < n) {
-            const fnpage_t p = pg[idx >> 6];
-            if ((p.bits >> (idx & 63)) & 1) {
-                uint32_t rk = p.rank +
-                              (uint32_t)__popcll(p.bits &
-                                                 ((1ull << (idx & 63)) - 1));
-                e = entries[rk];
-            }
-        }
-        d_eoff[r] = e >> 24;
-        d_cnt[r] = (uint32_t)(e & 0xFFFFFF);
-    }
-}
-
-// chunked exclusive prefix of d_cnt -> d_pre + per-chunk sums (same
-// chunk math as k_probe_scan so the expansion kernels are unchanged)
-__global__ void k_scan_local(const uint32_t *__restrict__ d_cnt,
-                             const uint64_t *__restrict__ d_state,
-                             uint64_t *__restrict__ d_pre,
-                             uint64_t *__restrict__ bsums)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    const int G = gridDim.x;
-    const int64_t chunk = (nrows + G - 1) / G;
-    const int64_t start = (int64_t)blockIdx.x * chunk;
-    const int64_t end = min(start + chunk, (int64_t)nrows);
-    __shared__ uint64_t sh[SCAN_T];
-    uint64_t carry = 0;
-    for (int64_t base = start; base < end; base += SCAN_T) {
-        const int64_t r = base + threadIdx.x;
-        uint64_t esz = (r < end) ? d_cnt[r] : 0;
-        sh[threadIdx.x] = esz;
-        __syncthreads();
-        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
-            uint64_t x = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
-            __syncthreads();
-            sh[threadIdx.x] += x;
-            __syncthreads();
-        }
-        if (r < end) d_pre[r] = carry + sh[threadIdx.x] - esz;
-        carry += sh[SCAN_T - 1];
-        __syncthreads();
-    }
-    if (threadIdx.x == 0) bsums[blockIdx.x] = (start < end) ? carry : 0;
-}
-
-// Filter predicate (k2c/k2k/c2k/i2k keep-row decision — the shared
-// core of the one-pass and scan-pipeline filter forms below).
-struct fparams {
-    const vertex_t *verts;
-    const sid_t *edges;
-    uint64_t bucket_start, num_buckets;
-    const sid_t *tbl;
-    int ncols, col;
-    uint32_t pid;
-    int dir, key_mode, probe_mode, col2;
-    sid_t cval;
-    uint64_t list_off, list_sz;
-    const uint16_t *type_of;
-    uint64_t type_base, type_n;
-    int use_typeof;
-    const uint64_t *tbm;
-    const fnpage_t *fn_pg;
-    const sid_t *fn_vals;
-    uint64_t fn_base, fn_n;
-    int fn_swap;
-};
-
-__device__ __forceinline__ bool filter_keep(const fparams &P, int64_t r) {
-    sid_t v = P.tbl[r * P.ncols + P.col];
-    if (P.use_typeof) {
-        uint64_t idx = (uint64_t)v - P.type_base;
-        if (P.tbm)
-            // per-type bitmap: 1 bit/vid, whole map LLC-resident —
-            // exact (multi-type included), no probe fallback
-            return idx < P.type_n && ((P.tbm[idx >> 6] >> (idx & 63)) & 1);
-        uint16_t t = (idx < P.type_n) ? P.type_of[idx] : 0;
-        if (t != 0xFFFF) return (sid_t)t == P.cval;
-    }
-    if (P.probe_mode == PM_EQ) {
-        // reversed functional map resolved host-side: the edge exists
-        // iff the row value IS the precomputed vertex
-        return v == P.cval;
-    }
-    if (P.probe_mode == PM_LIST && !P.use_typeof)
-        return bsearch_u32(P.edges + P.list_off, P.list_sz, v);
-    if (P.fn_pg) {
-        // functional predicate: the row's single object replaces the
-        // probe + edge-list search (rank-compressed map: 16-B page +
-        // 4-B value vs 148 bytes of hash traffic).  fn_swap: the
-        // REVERSED direction is functional — check fn[other col] == col.
-        sid_t a = v, b;
-        if (P.probe_mode == PM_CONST) b = P.cval;
-        else b = P.tbl[r * P.ncols + P.col2];
-        if (P.fn_swap) { sid_t t_ = a; a = b; b = t_; }
-        sid_t tv = fn_lookup(P.fn_pg, P.fn_vals, P.fn_base, P.fn_n, a);
-        return tv && tv == b;
-    }
-    uint64_t key = (P.key_mode == PK_NORMAL)
-                       ? key_pack(v, P.pid, (uint64_t)P.dir)
-                       : key_pack(0, v, (uint64_t)P.dir);
-    uint64_t eoff = 0, esz = 0;
-    probe_one(P.verts, P.bucket_start, P.num_buckets, key, eoff, esz);
-    sid_t tgt = (P.probe_mode == PM_CONST) ? P.cval
-                                           : P.tbl[r * P.ncols + P.col2];
-    return esz && bsearch_u32(P.edges + eoff, esz, tgt);
-}
-
-__device__ __forceinline__ uint64_t filter_bytes_per_row(const fparams &P) {
-    return (P.use_typeof ? 6
-            : P.fn_pg ? 24
-            : P.probe_mode == PM_LIST ? 12
-                                      : (4 + 128 + 8 + 64)) +
-           8 * P.ncols;
-}
-
-// One-pass filter: keep + wavefront-ballot compaction.  Each tile
-// costs ONE global cursor atomic per block — a single word saturates
-// at ~88 atomics/us (microarch row `dequeue`), so ~2048 in-flight
-// blocks serialize ~23 us per tile ROUND; fine up to ~1-2M rows.
-// Larger tables route through the atomic-free scan pipeline below
-// (k_filter_flags + k_scan_local + k_scan_mid + k_filter_scatter).
-// Row order is engine-internal; parity is set-level
-// (sparql.hpp:455-476 semantics).
-__global__ void k_filter_tpr(fparams P, int verify_only,
-                             int commit_mode, uint64_t commit_cap,
-                             uint64_t *__restrict__ d_state,
-                             uint64_t *__restrict__ d_stats,
-                             sid_t *__restrict__ out_tbl)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows * filter_bytes_per_row(P));
-    constexpr int K = 4;
-    __shared__ unsigned long long s_base;
-    __shared__ unsigned int s_cnt;
-    __shared__ unsigned int s_wbase[SCAN_T / 64];
-    const int lane = threadIdx.x & 63;
-    const int wid = threadIdx.x >> 6;
-    const int ncols = P.ncols;
-    const sid_t *__restrict__ tbl = P.tbl;
-    const int64_t tile = (int64_t)blockDim.x * K;
-    const int64_t stride = (int64_t)gridDim.x * tile;
-
-    for (int64_t base = (int64_t)blockIdx.x * tile; base < nrows; base += stride) {
-        bool keep[K];
-        int64_t rr[K];
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            const int64_t r = base + k * SCAN_T + threadIdx.x;
-            rr[k] = r;
-            keep[k] = r < nrows && filter_keep(P, r);
-        }
-        if (verify_only) {
-            // identity-verified filter (captured graphs): the warm pass
-            // saw zero drops; every replay still checks every row and
-            // flags S_ERR on any miss (no writes, no table flip)
-            uint32_t miss = 0;
-#pragma unroll
-            for (int k = 0; k < K; k++)
-                miss += (rr[k] < nrows && !keep[k]) ? 1u : 0u;
-            if (miss)
-                atomicAdd((unsigned long long *)&d_state[S_OVF],
-                          (unsigned long long)miss);
-            continue;
-        }
-        if (threadIdx.x == 0) s_cnt = 0;
-        __syncthreads();
-        uint64_t mask[K];
-        uint32_t wtot = 0;
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            mask[k] = __ballot(keep[k]);
-            wtot += (uint32_t)__popcll(mask[k]);
-        }
-        if (lane == 0) s_wbase[wid] = atomicAdd(&s_cnt, wtot);
-        __syncthreads();
-        if (threadIdx.x == 0)
-            s_base = s_cnt ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
-                                       (unsigned long long)s_cnt)
-                           : 0;
-        __syncthreads();
-        uint32_t wpos = s_wbase[wid];
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            if (keep[k]) {
-                uint64_t pos = s_base + wpos +
-                               (uint32_t)__popcll(mask[k] &
-                                                  ((1ull << lane) - 1));
-                sid_t *dst = out_tbl + (int64_t)pos * ncols;
-                const sid_t *srow = tbl + rr[k] * ncols;
-                for (int c = 0; c < ncols; c++) dst[c] = srow[c];
-            }
-            wpos += (uint32_t)__popcll(mask[k]);
-        }
-    }
-    commit_tail(d_state, commit_cap, commit_mode);
-}
-
-// Scan-pipeline filter, phase 1: barrier- and atomic-free keep flags
-// into d_cnt (k_scan_local / k_scan_mid then produce deterministic
-// positions; k_filter_scatter writes).
-__global__ void k_filter_flags(fparams P,
-                               const uint64_t *__restrict__ d_state,
-                               uint64_t *__restrict__ d_stats,
-                               uint32_t *__restrict__ d_cnt)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows * filter_bytes_per_row(P));
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < nrows; r += (int64_t)gridDim.x * blockDim.x)
-        d_cnt[r] = filter_keep(P, r) ? 1u : 0u;
-}
-
-__global__ void k_filter_scatter(const sid_t *__restrict__ tbl, int ncols,
-                                 const uint32_t *__restrict__ d_cnt,
-                                 const uint64_t *__restrict__ d_pre,
-                                 const uint64_t *__restrict__ bsums, int G,
-                                 const uint64_t *__restrict__ d_state,
-                                 sid_t *__restrict__ out_tbl)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    const int64_t chunk = (nrows + G - 1) / G;
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
-        if (!d_cnt[r]) continue;
-        uint64_t pos = d_pre[r] + bsums[r / chunk];
-        sid_t *dst = out_tbl + (int64_t)pos * ncols;
-        const sid_t *srow = tbl + r * ncols;
-        for (int c = 0; c < ncols; c++) dst[c] = srow[c];
-    }
-}
-
-// fn-expansion scatter (phase 4 of gather -> scan_local -> scan_mid ->
-// scatter): deterministic positions, no atomics, barrier-free.
-template <int NC>
-__global__ void k_fn_scatter(const sid_t *__restrict__ tbl,
-                             const sid_t *__restrict__ d_val,
-                             const uint32_t *__restrict__ d_cnt,
-                             const uint64_t *__restrict__ d_pre,
-                             const uint64_t *__restrict__ bsums, int G,
-                             const uint64_t *__restrict__ d_state,
-                             uint64_t *__restrict__ d_stats,
-                             sid_t *__restrict__ out)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    constexpr int oc = NC + 1;
-    count_bytes(d_stats, CAT_EXPAND, (uint64_t)nrows * (4 + 4 * NC + 4 * oc));
-    const int64_t chunk = (nrows + G - 1) / G;
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
-        if (!d_cnt[r]) continue;
-        uint64_t pos = d_pre[r] + bsums[r / chunk];
-        sid_t *dst = out + (int64_t)pos * oc;
-        const sid_t *srow = tbl + r * NC;
-#pragma unroll
-        for (int c = 0; c < NC; c++) dst[c] = srow[c];
-        dst[NC] = d_val[r];
-    }
-}
-
-// finish the scan across blocks: exclusive over the G block sums
-// (single block), total -> state[S_TOTAL]
-__global__ void k_scan_mid(uint64_t *__restrict__ bsums, int G,
-                           uint64_t *__restrict__ d_state)
-{
-    __shared__ uint64_t sh[SCAN_T];
-    uint64_t carry = 0;
-    for (int base = 0; base < G; base += SCAN_T) {
-        int i = base + threadIdx.x;
-        uint64_t x = (i < G) ? bsums[i] : 0;
-        sh[threadIdx.x] = x;
-        __syncthreads();
-        for (int ofs = 1; ofs < SCAN_T; ofs <<= 1) {
-            uint64_t v = (threadIdx.x >= (unsigned)ofs) ? sh[threadIdx.x - ofs] : 0;
-            __syncthreads();
-            sh[threadIdx.x] += v;
-            __syncthreads();
-        }
-        if (i < G) bsums[i] = carry + sh[threadIdx.x] - x;
-        carry += sh[SCAN_T - 1];
-        __syncthreads();
-    }
-    if (threadIdx.x == 0) d_state[S_TOTAL] = carry;
-}
-
-// advance: nrows = min(total, cap); flag overflow for the host re-run
-// (replaces the reference's rbuf-overflow assert, gpu_engine_cuda.hpp:185)
-__global__ void k_commit(uint64_t *__restrict__ d_state, uint64_t cap) {
-    uint64_t t = d_state[S_TOTAL];
-    if (t > cap) {
-        d_state[S_ERR] = 1;
-        d_state[S_REQ] = max(d_state[S_REQ], t);
-        t = cap;
-    }
-    d_state[S_NROWS] = t;
-    // reset the accumulators for the next step (saves a 3us k_zero_words
-    // launch per step — the state kernels were 12% of suite GPU time)
-    d_state[S_TOTAL] = 0;
-    d_state[S_OVF] = 0;
-}
-
-__global__ void k_set_state(uint64_t *__restrict__ d_state, uint64_t nrows) {
-    d_state[S_NROWS] = nrows;
-    d_state[S_TOTAL] = 0;
-    d_state[S_OVF] = 0;
-}
-
-// device projection to required-var columns (sparql.hpp:1510-1536)
-struct cols8 { int32_t c[8]; };
-__global__ void k_project(const sid_t *__restrict__ tbl, int ncols,
-                          const uint64_t *__restrict__ d_state, cols8 cols,
-                          int rc, sid_t *__restrict__ out)
-{
-    const int64_t n = (int64_t)d_state[S_NROWS];
-    for (int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; t < n * rc;
-         t += (int64_t)gridDim.x * blockDim.x) {
-        int64_t i = t / rc;
-        int j = (int)(t - i * rc);
-        int c = cols.c[j];
-        out[t] = (c >= 0) ? tbl[i * ncols + c] : (sid_t)0xFFFFFFFFu;
-    }
-}
-
-__global__ void k_publish_state(const uint64_t *__restrict__ d_state,
-                                const uint64_t *__restrict__ d_stats,
-                                uint64_t *__restrict__ h_pin) {
-    for (int i = 0; i < S_WORDS; i++) h_pin[i] = d_state[i];
-    for (int i = 0; i < CAT_COUNT; i++) h_pin[8 + i] = d_stats[i];
-    // sticky error flag (h_pin slot 7): survives across back-to-back
-    // graph replays whose own begin kernels clear d_state — the
-    // one-sync-per-pass path checks and clears it host-side
-    if (d_state[S_ERR]) h_pin[7] = 1;
-}
-
-// Input-centric expansion (known_to_unknown back half,
-// sparql.hpp:325-367): thread r writes its deg outputs at
-// pre[r]+bsums[chunk(r)] — no per-output binary search (the
-// output-centric version paid 2-3 DRAM lines of prefix walk per output
-// row).  Rows with deg > 32 go to an overflow queue handled by
-// k_expand_big with one WAVE per row (lanes stride the edge list).
-template <int NC>
-__global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
-                            const sid_t *__restrict__ edges,
-                            const uint64_t *__restrict__ d_eoff,
-                            const uint32_t *__restrict__ d_cnt,
-                            const uint64_t *__restrict__ d_pre,
-                            const uint64_t *__restrict__ bsums, int G,
-                            uint64_t *__restrict__ d_state, uint64_t cap,
-                            uint64_t *__restrict__ d_stats,
-                            uint32_t *__restrict__ ovf,
-                            sid_t *__restrict__ out)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    const int64_t chunk = (nrows + G - 1) / G;
-    constexpr int oc = NC + 1;
-    (void)ncols;
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
-         r += (int64_t)gridDim.x * blockDim.x) {
-        uint32_t deg = d_cnt[r];
-        if (!deg) continue;
-        uint64_t basep = d_pre[r] + bsums[r / chunk];
-        if (basep >= cap) continue;  // overflow: flagged by k_commit
-        if (deg > 32) {
-            unsigned long long i = atomicAdd(
-                (unsigned long long *)&d_state[S_OVF], 1ull);
-            ovf[i] = (uint32_t)r;
-            continue;
-        }
-        if (basep + deg > cap) deg = (uint32_t)(cap - basep);
-        sid_t row[NC];
-#pragma unroll
-        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
-        const sid_t *el = edges + d_eoff[r];
-        sid_t *dst = out + (int64_t)basep * oc;
-        for (uint32_t k = 0; k < deg; k++) {
-#pragma unroll
-            for (int c = 0; c < NC; c++) dst[c] = row[c];
-            dst[NC] = el[k];
-            dst += oc;
-        }
-    }
-    // algorithmic bytes for the whole expansion (counted once; includes
-    // the big-row pass): total*(edge 4 + write 4*oc) + nrows*(row 4*ncols
-    // + cnt/pre/eoff 20)
-    if (blockIdx.x == 0 && threadIdx.x == 0)
-        atomicAdd((unsigned long long *)&d_stats[CAT_EXPAND],
-                  (unsigned long long)(min(d_state[S_TOTAL], cap) * (4 + 4 * oc) +
-                                       (uint64_t)nrows * (4 * ncols + 20)));
-}
-
-// big-fanout rows: one wave per queued row, lanes stride the edge list
-// (coalesced writes: adjacent lanes write adjacent output rows)
-template <int NC>
-__global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
-                             const sid_t *__restrict__ edges,
-                             const uint64_t *__restrict__ d_eoff,
-                             const uint32_t *__restrict__ d_cnt,
-                             const uint64_t *__restrict__ d_pre,
-                             const uint64_t *__restrict__ bsums, int G,
-                             uint64_t *__restrict__ d_state, uint64_t cap,
-                             int commit_mode,
-                             const uint32_t *__restrict__ ovf,
-                             sid_t *__restrict__ out)
-{
-    const int64_t nq = (int64_t)d_state[S_OVF];
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    const int64_t chunk = (nrows + G - 1) / G;
-    constexpr int oc = NC + 1;
-    (void)ncols;
-    const int lane = threadIdx.x & 63;
-    const int64_t w0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
-    for (int64_t q = w0; q < nq; q += nw) {
-        const int64_t r = ovf[q];
-        uint64_t deg = d_cnt[r];
-        uint64_t basep = d_pre[r] + bsums[r / chunk];
-        if (basep >= cap) continue;
-        if (basep + deg > cap) deg = cap - basep;
-        sid_t row[NC];
-#pragma unroll
-        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
-        const sid_t *el = edges + d_eoff[r];
-        for (uint64_t k = lane; k < deg; k += 64) {
-            sid_t *dst = out + (int64_t)(basep + k) * oc;
-#pragma unroll
-            for (int c = 0; c < NC; c++) dst[c] = row[c];
-            dst[NC] = el[k];
-        }
-    }
-    commit_tail(d_state, cap, commit_mode);
-}
-
-// known_to_unknown over a FUNCTIONAL predicate (every key deg==1,
-// rank-compressed map): the probe+scan+expand pipeline collapses into
-// a page+value gather pair + compacted append, with an optional fused
-// `?v rdf:type CONST` filter on the NEW column (plan pairs like Q1's
-// ugDegreeFrom -> University).  Two phases: a 1:1 grid-stride GATHER
-// (the measured-fast k_expand_fn_map shape — no barriers between the
-// dependent page/value loads) writes each row's resolved object (0 =
-// miss/filtered) to a scratch stream; the COMPACT phase then reads it
-// SEQUENTIALLY with the block-aggregated scan.  The round-1 single-pass
-// form interleaved the gathers with 16 scan barriers per tile and ran
-// 4-5x slower than its own gather cost (169 vs 36 us on Q1's 6.4M-row
-// step).  Multi-type (0xFFFF) falls back to a probe of [val|TYPE|OUT].
-// Output rows <= input rows, so capacity can never overflow.
-__global__ void k_fn_gather(const sid_t *__restrict__ tbl, int ncols,
-                            const fnpage_t *__restrict__ fn_pg,
-                            const sid_t *__restrict__ fn_vals,
-                            uint64_t fn_base, uint64_t fn_n, int col,
-                            int use_typeof, sid_t fcval,
-                            const uint16_t *__restrict__ type_of,
-                            const uint64_t *__restrict__ tbm,
-                            uint64_t type_base, uint64_t type_n,
-                            const vertex_t *__restrict__ verts,
-                            const sid_t *__restrict__ edges,
-                            uint64_t f_bstart, uint64_t f_nbuckets,
-                            const uint64_t *__restrict__ d_state,
-                            uint64_t *__restrict__ d_stats,
-                            sid_t *__restrict__ d_val,
-                            uint32_t *__restrict__ d_cnt)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, CAT_EXPAND,
-                (uint64_t)nrows * (24 + 4 + (use_typeof ? 2 : 0)));
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
-        sid_t v = tbl[r * ncols + col];
-        sid_t tv = fn_lookup(fn_pg, fn_vals, fn_base, fn_n, v);
-        if (tv && use_typeof) {
-            uint64_t tix = (uint64_t)tv - type_base;
-            if (tbm) {  // bitmap: decision is exact (multi-type incl.)
-                if (!(tix < type_n && ((tbm[tix >> 6] >> (tix & 63)) & 1)))
-                    tv = 0;
-            } else {
-                uint16_t t = (tix < type_n) ? type_of[tix] : 0;
-                if (t != 0xFFFF) {
-                    if ((sid_t)t != fcval) tv = 0;
-                } else {
-                    uint64_t eo = 0, es = 0;
-                    probe_one(verts, f_bstart, f_nbuckets,
-                              key_pack(tv, TYPE_ID, (uint64_t)DIR_OUT), eo,
-                              es);
-                    if (!(es && bsearch_u32(edges + eo, es, fcval))) tv = 0;
-                }
-            }
-        }
-        d_val[r] = tv;
-        d_cnt[r] = tv ? 1u : 0u;
-    }
-}
-
-template <int NC>
-__global__ void k_fn_compact(const sid_t *__restrict__ tbl,
-                             const sid_t *__restrict__ d_val,
-                             uint64_t *__restrict__ d_state,
-                             uint64_t *__restrict__ d_stats,
-                             uint64_t cap, int commit_mode,
-                             sid_t *__restrict__ out)
-{
-    // wavefront-ballot compaction (the north_star idiom): each wave
-    // ranks its kept rows with __ballot+popc — no LDS scan array, and
-    // 3 block barriers per 1024-row tile instead of the Hillis-Steele
-    // form's 16 (which measured 108 us on 6.4M rows against ~40
-    // expected).  Reads are lane-adjacent; wave outputs are contiguous.
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    constexpr int oc = NC + 1;
-    count_bytes(d_stats, CAT_EXPAND, (uint64_t)nrows * (4 + 4 * NC + 4 * oc));
-    constexpr int K = 4;
-    __shared__ unsigned long long s_base;
-    __shared__ unsigned int s_cnt;
-    __shared__ unsigned int s_wbase[SCAN_T / 64];
-    const int lane = threadIdx.x & 63;
-    const int wid = threadIdx.x >> 6;
-    const int64_t tile = (int64_t)blockDim.x * K;
-    const int64_t stride = (int64_t)gridDim.x * tile;
-
-    for (int64_t base = (int64_t)blockIdx.x * tile; base < nrows;
-         base += stride) {
-        if (threadIdx.x == 0) s_cnt = 0;
-        __syncthreads();
-        sid_t val[K];
-        uint64_t mask[K];
-        uint32_t wtot = 0;
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            const int64_t r = base + k * SCAN_T + threadIdx.x;
-            val[k] = (r < nrows) ? d_val[r] : 0;
-            mask[k] = __ballot(val[k] != 0);
-            wtot += (uint32_t)__popcll(mask[k]);
-        }
-        if (lane == 0) s_wbase[wid] = atomicAdd(&s_cnt, wtot);
-        __syncthreads();
-        if (threadIdx.x == 0)
-            s_base = s_cnt ? atomicAdd((unsigned long long *)&d_state[S_TOTAL],
-                                       (unsigned long long)s_cnt)
-                           : 0;
-        __syncthreads();
-        uint32_t wpos = s_wbase[wid];
-#pragma unroll
-        for (int k = 0; k < K; k++) {
-            if (val[k]) {
-                uint64_t pos = s_base + wpos +
-                               (uint32_t)__popcll(mask[k] &
-                                                  ((1ull << lane) - 1));
-                if (pos < cap) {
-                    const int64_t r = base + k * SCAN_T + threadIdx.x;
-                    sid_t *dst = out + (int64_t)pos * oc;
-                    const sid_t *srow = tbl + r * NC;
-#pragma unroll
-                    for (int c = 0; c < NC; c++) dst[c] = srow[c];
-                    dst[NC] = val[k];
-                }
-            }
-            wpos += (uint32_t)__popcll(mask[k]);
-        }
-    }
-    commit_tail(d_state, cap, commit_mode);
-}
-
 // Fused known_to_unknown + rdf:type constant filter on the NEW column
 // (plan pairs like Q1's ugDegreeFrom -> "?Y type University"): expansion
 // emits only passing rows, block-compacted — saves the follow-up
