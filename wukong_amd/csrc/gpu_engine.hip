@@ -29,7 +29,7 @@
  *   k_compact         — flag compaction (replaces gpu_hash.cu:523-585)
  *   k_copy_list       — i2u/c2u list materialisation on device (the
  *                       reference ran these on CPU, gpu_engine.hpp:63-123)
- *   k_dst_histogram / k_dst_scatter — fork-join split by vid % ndst
+ *   k_dst_count/scan2/scatter2 — fork-join radix split by vid % ndst
  *                       (replaces gpu_hash.cu:600-760)
  *   k_light_batch     — batched light-query window (one wavefront
  *                       workgroup per query; proxy.hpp:477-525 window)
@@ -1015,32 +1015,77 @@ __global__ void k_copy_list(const sid_t *__restrict__ edges, uint64_t off,
 }
 
 // fork-join split (generate_sub_query sparql.hpp:772-796): dst = vid % ndst
-// (hash_mod, utils/math.hpp:51-55)
-__global__ void k_dst_histogram(const sid_t *__restrict__ tbl, int64_t nrows,
-                                int ncols, int col, int ndst,
-                                unsigned long long *__restrict__ hist)
+// Fork-join split as a radix partition (replaces the per-row
+// global-cursor scatter: 6.4M rows over <=8 cursor words serialize at
+// ~88 atomics/us per word ~= 9 ms; reference analog
+// gpu_hash.cu:600-760).  Phase 1: per-block LDS histograms over
+// contiguous chunks; phase 2: ONE block turns them into per-
+// (block,dst) bases + per-dst totals; phase 3: scatter with LDS
+// cursors only — no global atomics anywhere.
+__global__ void k_dst_count(const sid_t *__restrict__ tbl, int64_t nrows,
+                            int ncols, int col, int ndst,
+                            unsigned long long *__restrict__ bh)
 {
-    __shared__ unsigned long long lh[64];
+    __shared__ unsigned int lh[64];
+    const int G = gridDim.x;
+    const int64_t chunk = (nrows + G - 1) / G;
+    const int64_t start = (int64_t)blockIdx.x * chunk;
+    const int64_t end = min(start + chunk, nrows);
     if (threadIdx.x < 64) lh[threadIdx.x] = 0;
     __syncthreads();
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
-         r += (int64_t)gridDim.x * blockDim.x)
-        atomicAdd(&lh[tbl[r * ncols + col] % (sid_t)ndst], 1ull);
+    for (int64_t r = start + threadIdx.x; r < end; r += blockDim.x)
+        atomicAdd(&lh[tbl[r * ncols + col] % (sid_t)ndst], 1u);
     __syncthreads();
-    if (threadIdx.x < (unsigned)ndst) atomicAdd(&hist[threadIdx.x], lh[threadIdx.x]);
+    if (threadIdx.x < (unsigned)ndst)
+        bh[(size_t)blockIdx.x * ndst + threadIdx.x] = lh[threadIdx.x];
 }
 
-__global__ void k_dst_scatter(const sid_t *__restrict__ tbl, int64_t nrows,
-                              int ncols, int col, int ndst,
-                              const unsigned long long *__restrict__ base,
-                              unsigned long long *__restrict__ cursor,
-                              sid_t *__restrict__ out)
+__global__ void k_dst_scan2(unsigned long long *__restrict__ bh, int G,
+                            int ndst, unsigned long long *__restrict__ hist)
 {
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
-         r += (int64_t)gridDim.x * blockDim.x) {
+    __shared__ unsigned long long tot[64];
+    const int d = threadIdx.x;
+    if (d < ndst) {
+        unsigned long long run = 0;
+        for (int b = 0; b < G; b++) {
+            unsigned long long t = bh[(size_t)b * ndst + d];
+            bh[(size_t)b * ndst + d] = run;
+            run += t;
+        }
+        tot[d] = run;
+        hist[d] = run;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        unsigned long long acc = 0;
+        for (int i = 0; i < ndst; i++) {
+            unsigned long long t = tot[i];
+            tot[i] = acc;
+            acc += t;
+        }
+    }
+    __syncthreads();
+    if (d < ndst)
+        for (int b = 0; b < G; b++) bh[(size_t)b * ndst + d] += tot[d];
+}
+
+__global__ void k_dst_scatter2(const sid_t *__restrict__ tbl, int64_t nrows,
+                               int ncols, int col, int ndst,
+                               const unsigned long long *__restrict__ bh,
+                               sid_t *__restrict__ out)
+{
+    __shared__ unsigned int lc[64];
+    const int G = gridDim.x;
+    const int64_t chunk = (nrows + G - 1) / G;
+    const int64_t start = (int64_t)blockIdx.x * chunk;
+    const int64_t end = min(start + chunk, nrows);
+    if (threadIdx.x < 64) lc[threadIdx.x] = 0;
+    __syncthreads();
+    for (int64_t r = start + threadIdx.x; r < end; r += blockDim.x) {
         int d = (int)(tbl[r * ncols + col] % (sid_t)ndst);
-        unsigned long long pos = atomicAdd(&cursor[d], 1ull);
-        sid_t *dst = out + (base[d] + pos) * ncols;
+        unsigned long long pos = bh[(size_t)blockIdx.x * ndst + d] +
+                                 atomicAdd(&lc[d], 1u);
+        sid_t *dst = out + pos * ncols;
         const sid_t *src = tbl + r * ncols;
         for (int c = 0; c < ncols; c++) dst[c] = src[c];
     }
@@ -3080,33 +3125,32 @@ extern "C" int32_t wk_engine_generate_sub_query(wk_engine_t *e, int32_t ndst,
     if (R > cap_rows) return WK_ERR_CAP;
     if (e->misc.ensure(128 * sizeof(unsigned long long))) return WK_ERR_HIP;
     unsigned long long *d_hist = (unsigned long long *)e->misc.p;
-    unsigned long long *d_cursor = d_hist + 64;
     hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(128), 0, e->stream,
                        (uint64_t *)e->misc.p, 128);
     sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
+    // per-(block,dst) bases live in the prefix scratch: G2*ndst <=
+    // (R/BLOCK+1)*64 entries always fits its (cap_rows+1) u64s
+    const int G2 = (int)std::max<int64_t>(
+        1, std::min<int64_t>(1024, (R + BLOCK - 1) / BLOCK));
+    unsigned long long *bh = (unsigned long long *)e->prefix.p;
     if (R) {
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_dst_histogram, dim3(grid_for(R)), dim3(BLOCK), 0,
-                           e->stream, cur_tbl, R, e->ncols, col, ndst, d_hist);
+        hipLaunchKernelGGL(k_dst_count, dim3(G2), dim3(BLOCK), 0, e->stream,
+                           cur_tbl, R, e->ncols, col, ndst, bh);
+        hipLaunchKernelGGL(k_dst_scan2, dim3(1), dim3(64), 0, e->stream, bh,
+                           G2, ndst, d_hist);
         TIME_END(e, CAT_SPLIT);
     }
     unsigned long long h_hist[64];
     HIP_CHECK(hipMemcpyAsync(h_hist, d_hist, ndst * sizeof(unsigned long long),
                              hipMemcpyDeviceToHost, e->stream));
     HIP_CHECK(stream_sync(e->stream));
-    unsigned long long base[64], acc = 0;
-    for (int i = 0; i < ndst; i++) {
-        base[i] = acc;
-        acc += h_hist[i];
-        rows_per_dst[i] = (int64_t)h_hist[i];
-    }
-    HIP_CHECK(hipMemcpyAsync(d_hist, base, ndst * sizeof(unsigned long long),
-                             hipMemcpyHostToDevice, e->stream));
+    for (int i = 0; i < ndst; i++) rows_per_dst[i] = (int64_t)h_hist[i];
     if (R) {
         TIME_BEGIN(e);
-        hipLaunchKernelGGL(k_dst_scatter, dim3(grid_for(R)), dim3(BLOCK), 0,
-                           e->stream, cur_tbl, R, e->ncols, col, ndst, d_hist,
-                           d_cursor, dev_out);
+        hipLaunchKernelGGL(k_dst_scatter2, dim3(G2), dim3(BLOCK), 0,
+                           e->stream, cur_tbl, R, e->ncols, col, ndst, bh,
+                           dev_out);
         TIME_END(e, CAT_SPLIT);
     }
     HIP_CHECK(stream_sync(e->stream));
