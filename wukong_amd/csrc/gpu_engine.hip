@@ -2799,32 +2799,14 @@ static int32_t exec_pattern(wk_engine *e) {
                   dir, key_mode, pmode, col2, cval, 0, 0, e->d_type_of,
                   e->st->type_base, e->st->type_n, use_typeof, d_tbm,
                   d_pg, d_vals, st->fn_base, st->fn_n, fn_swap};
-        if (!verify_only && e->bound > (int64_t)(1 << 20)) {
-            // big tables: atomic-free scan pipeline (per-tile cursor
-            // atomics serialize at ~88/us — see k_filter_tpr header)
-            const int G = scan_grid(e->bound);
-            hipLaunchKernelGGL(k_filter_flags, dim3(grid_for(e->bound)),
-                               dim3(BLOCK), 0, e->stream, P, e->d_state,
-                               e->d_stats, (uint32_t *)e->cnt.p);
-            hipLaunchKernelGGL(k_scan_local, dim3(G), dim3(SCAN_T), 0,
-                               e->stream, (const uint32_t *)e->cnt.p,
-                               e->d_state, (uint64_t *)e->prefix.p,
-                               (uint64_t *)e->bsums.p);
-            hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0,
-                               e->stream, (uint64_t *)e->bsums.p, G,
-                               e->d_state);
-            hipLaunchKernelGGL(k_filter_scatter, dim3(grid_for(e->bound)),
-                               dim3(BLOCK), 0, e->stream, cur_tbl, e->ncols,
-                               (const uint32_t *)e->cnt.p,
-                               (const uint64_t *)e->prefix.p,
-                               (const uint64_t *)e->bsums.p, G, e->d_state,
-                               out_tbl);
-        } else {
-            hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)),
-                               dim3(BLOCK), 0, e->stream, P,
-                               verify_only ? 1 : 0, /*commit*/ 0, 0,
-                               e->d_state, e->d_stats, out_tbl);
-        }
+        // one-pass ballot filter: measured FASTER than the
+        // flags+scan+scatter pipeline at suite keep-rates (graph-q1
+        // 205 -> 264 us with the pipeline: the extra passes cost more
+        // than the per-tile cursor atomics they remove)
+        hipLaunchKernelGGL(k_filter_tpr, dim3(grid_for(e->bound)),
+                           dim3(BLOCK), 0, e->stream, P,
+                           verify_only ? 1 : 0, /*commit*/ 0, 0,
+                           e->d_state, e->d_stats, out_tbl);
         TIME_END(e, CAT_FILTER);
         if (verify_only) {
             hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0, e->stream,
