@@ -103,7 +103,7 @@ enum { PM_SIZE = 0,    // k2u: write edge count per row
 // device query state (see engine): [0]=nrows [1]=scan total [2]=overflow
 // flag [3]=required rows; stats[0..6] = algorithmic bytes per category
 enum { S_NROWS = 0, S_TOTAL = 1, S_ERR = 2, S_REQ = 3, S_OVF = 4,
-       S_DONE = 5, S_WORDS = 6 };
+       S_DONE = 5, S_INROWS = 6, S_WORDS = 7 };
 enum { CAT_PROBE = 0, CAT_SCAN, CAT_EXPAND, CAT_FILTER, CAT_COPY, CAT_SPLIT,
        CAT_OTHER, CAT_COUNT };
 
@@ -463,10 +463,10 @@ __device__ __forceinline__ uint64_t filter_bytes_per_row(const fparams &P) {
 // costs ONE global cursor atomic per block — a single word saturates
 // at ~88 atomics/us (microarch row `dequeue`), so ~2048 in-flight
 // blocks serialize ~23 us per tile ROUND; fine up to ~1-2M rows.
-// Larger tables route through the atomic-free scan pipeline below
-// (k_filter_flags + k_scan_local + k_scan_mid + k_filter_scatter).
-// Row order is engine-internal; parity is set-level
-// (sparql.hpp:455-476 semantics).
+// (A flags+scan+scatter pipeline was tried for larger tables and
+// measured SLOWER at suite keep-rates — extra passes cost more than
+// the atomics they save.)  Row order is engine-internal; parity is
+// set-level (sparql.hpp:455-476 semantics).
 __global__ void k_filter_tpr(fparams P, int verify_only,
                              int commit_mode, uint64_t commit_cap,
                              uint64_t *__restrict__ d_state,
@@ -541,40 +541,6 @@ __global__ void k_filter_tpr(fparams P, int verify_only,
     commit_tail(d_state, commit_cap, commit_mode);
 }
 
-// Scan-pipeline filter, phase 1: barrier- and atomic-free keep flags
-// into d_cnt (k_scan_local / k_scan_mid then produce deterministic
-// positions; k_filter_scatter writes).
-__global__ void k_filter_flags(fparams P,
-                               const uint64_t *__restrict__ d_state,
-                               uint64_t *__restrict__ d_stats,
-                               uint32_t *__restrict__ d_cnt)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    count_bytes(d_stats, CAT_FILTER, (uint64_t)nrows * filter_bytes_per_row(P));
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < nrows; r += (int64_t)gridDim.x * blockDim.x)
-        d_cnt[r] = filter_keep(P, r) ? 1u : 0u;
-}
-
-__global__ void k_filter_scatter(const sid_t *__restrict__ tbl, int ncols,
-                                 const uint32_t *__restrict__ d_cnt,
-                                 const uint64_t *__restrict__ d_pre,
-                                 const uint64_t *__restrict__ bsums, int G,
-                                 const uint64_t *__restrict__ d_state,
-                                 sid_t *__restrict__ out_tbl)
-{
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
-    const int64_t chunk = (nrows + G - 1) / G;
-    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
-        if (!d_cnt[r]) continue;
-        uint64_t pos = d_pre[r] + bsums[r / chunk];
-        sid_t *dst = out_tbl + (int64_t)pos * ncols;
-        const sid_t *srow = tbl + r * ncols;
-        for (int c = 0; c < ncols; c++) dst[c] = srow[c];
-    }
-}
-
 // fn-expansion scatter (phase 4 of gather -> scan_local -> scan_mid ->
 // scatter): deterministic positions, no atomics, barrier-free.
 template <int NC>
@@ -587,7 +553,7 @@ __global__ void k_fn_scatter(const sid_t *__restrict__ tbl,
                              uint64_t *__restrict__ d_stats,
                              sid_t *__restrict__ out)
 {
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int64_t nrows = (int64_t)d_state[S_INROWS];  // scan_mid committed
     constexpr int oc = NC + 1;
     count_bytes(d_stats, CAT_EXPAND, (uint64_t)nrows * (4 + 4 * NC + 4 * oc));
     const int64_t chunk = (nrows + G - 1) / G;
@@ -604,9 +570,13 @@ __global__ void k_fn_scatter(const sid_t *__restrict__ tbl,
 }
 
 // finish the scan across blocks: exclusive over the G block sums
-// (single block), total -> state[S_TOTAL]
+// (single block).  Fuses the step commit (k_commit semantics): the
+// pipeline total is known here, the following scatter kernels read
+// the INPUT row count from the S_INROWS snapshot, and S_OVF (the
+// big-row queue this chain is about to fill) starts clean — saving
+// the separate 1-thread commit launch per k2u/fn step.
 __global__ void k_scan_mid(uint64_t *__restrict__ bsums, int G,
-                           uint64_t *__restrict__ d_state)
+                           uint64_t cap, uint64_t *__restrict__ d_state)
 {
     __shared__ uint64_t sh[SCAN_T];
     uint64_t carry = 0;
@@ -625,7 +595,18 @@ __global__ void k_scan_mid(uint64_t *__restrict__ bsums, int G,
         carry += sh[SCAN_T - 1];
         __syncthreads();
     }
-    if (threadIdx.x == 0) d_state[S_TOTAL] = carry;
+    if (threadIdx.x == 0) {
+        d_state[S_INROWS] = d_state[S_NROWS];
+        uint64_t t = carry;
+        if (t > cap) {
+            d_state[S_ERR] = 1;
+            d_state[S_REQ] = max(d_state[S_REQ], t);
+            t = cap;
+        }
+        d_state[S_NROWS] = t;
+        d_state[S_TOTAL] = 0;
+        d_state[S_OVF] = 0;
+    }
 }
 
 // advance: nrows = min(total, cap); flag overflow for the host re-run
@@ -695,7 +676,9 @@ __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
                             uint32_t *__restrict__ ovf,
                             sid_t *__restrict__ out)
 {
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    // k_scan_mid already committed: S_INROWS = the input row count,
+    // S_NROWS = the (capped) output total
+    const int64_t nrows = (int64_t)d_state[S_INROWS];
     const int64_t chunk = (nrows + G - 1) / G;
     constexpr int oc = NC + 1;
     (void)ncols;
@@ -729,7 +712,7 @@ __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
     // + cnt/pre/eoff 20)
     if (blockIdx.x == 0 && threadIdx.x == 0)
         atomicAdd((unsigned long long *)&d_stats[CAT_EXPAND],
-                  (unsigned long long)(min(d_state[S_TOTAL], cap) * (4 + 4 * oc) +
+                  (unsigned long long)(d_state[S_NROWS] * (4 + 4 * oc) +
                                        (uint64_t)nrows * (4 * ncols + 20)));
 }
 
@@ -748,7 +731,7 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
                              sid_t *__restrict__ out)
 {
     const int64_t nq = (int64_t)d_state[S_OVF];
-    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    const int64_t nrows = (int64_t)d_state[S_INROWS];
     const int64_t chunk = (nrows + G - 1) / G;
     constexpr int oc = NC + 1;
     (void)ncols;
@@ -2415,7 +2398,8 @@ static void launch_expand_fn_t(wk_engine *e, const sid_t *cur_tbl,
                        (const uint32_t *)e->cnt.p, e->d_state,
                        (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p);
     hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0, e->stream,
-                       (uint64_t *)e->bsums.p, G, e->d_state);
+                       (uint64_t *)e->bsums.p, G, (uint64_t)e->cap_rows,
+                       e->d_state);
     hipLaunchKernelGGL(k_fn_scatter<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
                        0, e->stream, cur_tbl, (const sid_t *)e->ovf.p,
                        (const uint32_t *)e->cnt.p, (const uint64_t *)e->prefix.p,
@@ -2863,12 +2847,10 @@ static int32_t exec_pattern(wk_engine *e) {
                 launch_expand_fn(e, cur_tbl, out_tbl, d_pg, d_vals, col,
                                  fuse2, fcval2, fseg2);
             TIME_END(e, CAT_EXPAND);
-            if (opt)
+            if (opt)  // 1:1 map path: commit_map still needed
                 hipLaunchKernelGGL(k_commit_map, dim3(1), dim3(1), 0,
                                    e->stream, e->d_state);
-            else
-                hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
-                                   e->d_state, (uint64_t)e->cap_rows);
+            // scan-pipeline path: k_scan_mid committed already
             e->v2c[-(o + 1)] = e->ncols;
             e->ncols = oc;
             e->cur ^= 1;
@@ -2953,7 +2935,8 @@ static int32_t exec_pattern(wk_engine *e) {
         {
             TIME_BEGIN(e);
             hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0, e->stream,
-                               (uint64_t *)e->bsums.p, G, e->d_state);
+                               (uint64_t *)e->bsums.p, G,
+                               (uint64_t)e->cap_rows, e->d_state);
             TIME_END(e, CAT_SCAN);
         }
         {
@@ -2964,6 +2947,9 @@ static int32_t exec_pattern(wk_engine *e) {
         e->v2c[-(o + 1)] = e->ncols;
         e->ncols = oc;
         e->bound = e->cap_rows;  // fan-out unknown until a sync point
+        e->cur ^= 1;             // k_scan_mid committed for this chain
+        e->step++;
+        return WK_OK;
     }
     hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
                        (uint64_t)e->cap_rows);
