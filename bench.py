@@ -440,6 +440,14 @@ def main():
     eng = engines[0]
     log(f"[rank {rank}] HBM upload {((store.num_slots*16+store.num_edges*4)/1e9):.2f} GB, "
         f"{inflight} engine(s) ({time.time()-t0:.1f}s)")
+    if distributed:
+        # xGMI peer mappings: sub-threshold steps read the owner's store
+        # in place instead of exchanging (dist.py init_peers; falls back
+        # to exchange-only on any rank's failure, uniformly)
+        from wukong_amd.dist import init_peers
+        peers_ok = init_peers(gstore, store)
+        log(f"[rank {rank}] xGMI peer mappings: "
+            f"{'ok' if peers_ok else 'unavailable (exchange-only)'}")
 
     if args.watdiv:
         from wukong_amd import watdiv as W
