@@ -395,6 +395,9 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if os.environ.get("WK_DIST_BACKEND") == "gloo":
+        import torch as _t
+        local_rank = local_rank % max(_t.cuda.device_count(), 1)
     ngpus = max(args.gpus, world)
     distributed = world > 1
 
@@ -406,8 +409,11 @@ def main():
 
     if distributed:
         import torch.distributed as dist
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        # WK_DIST_BACKEND=gloo: single-GPU rehearsal of the full N>1
+        # code path (host exchanges; RCCL refuses 2 ranks on 1 device)
+        backend = os.environ.get("WK_DIST_BACKEND", "nccl")
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+        dist.init_process_group(backend)
         from wukong_amd.dist import DistQuery, GpuExecutor
 
     t0 = time.time()

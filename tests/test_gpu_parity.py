@@ -30,6 +30,31 @@ def test_query_parity_lubm4(name, eng4, oracle4):
     assert np.array_equal(sort_rows(got), sort_rows(want))
 
 
+def test_side_index_fallback_parity(lubm4, oracle4):
+    """Pure cluster-hash path: a store built WITHOUT the side indexes
+    (fn maps, CSR, type bitmaps — WK_FN=0 WK_CSR=0 WK_TBM=0) must give
+    identical bindings.  Covers the probe/expand/filter base kernels
+    that the side-indexed default store routes around."""
+    old = {}
+    for k in ("WK_FN", "WK_CSR", "WK_TBM"):
+        old[k] = os.environ.get(k)
+        os.environ[k] = "0"
+    try:
+        store = wk.Store(lubm4)
+        eng = wk.Engine(store, device=0)
+        for name, plan in Q.ALL.items():
+            got = sort_rows(eng.run_query(plan))
+            want = sort_rows(oracle4.run_query(plan))
+            assert got.shape == want.shape, (name, got.shape, want.shape)
+            assert np.array_equal(got, want), name
+    finally:
+        for k, v in old.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
+
+
 def test_golden_fixture_gpu(eng4, store4):
     if not os.path.exists(GOLD):
         pytest.skip("golden fixture not generated")
