@@ -232,7 +232,9 @@ class OracleExecutor:
         keep rows by owner-list membership.  Tables are sub-threshold,
         so the Python loop is fine."""
         s, p, d, o = pat
-        if s >= 0 or (p == 1 and d == 0):  # const start / per-row type index
+        # const start / per-row type index / predicate-variable shapes
+        # must exchange (mirrors exec_pattern_remote's rejections)
+        if s >= 0 or (p == 1 and d == 0) or p < 1:
             raise ValueError("pattern shape needs the exchange path")
         t = self.table()
         col = v2c_prev[-(s + 1)]

@@ -2484,6 +2484,9 @@ static int32_t exec_pattern_remote(wk_engine *e) {
     // [0|tid|IN] per-row index expansion fans out over ALL partitions —
     // not a per-owner read; callers must exchange for this shape
     if ((sid_t)p == TYPE_ID && dir == DIR_IN) return WK_ERR_PLAN;
+    // predicate variables / VERSATILE [v|PREDICATE_ID|dir] lists live in
+    // the dense vp CSR, not the peer-mapped cluster hash
+    if (p < 1) return WK_ERR_PLAN;
     int col = e->var2col(s);
     if (col < 0) return WK_ERR_PLAN;
     const int nsrv = (int)g->peers.size();
