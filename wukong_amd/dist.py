@@ -241,11 +241,6 @@ def final_process(table, v2c, plan):
     return np.ascontiguousarray(t[:, cols])
 
 
-def engine_rows(engine):
-    # row count after the last step (engine tracks it internally)
-    return engine.last_rows
-
-
 class GpuExecutor:
     """wukong_amd.Engine adapter for DistQuery (nccl path keeps tables on
     device; gloo path round-trips through host for the CPU tests)."""
@@ -268,8 +263,9 @@ class GpuExecutor:
         return self.store.get_index(pid, d)
 
     def filter_with_list(self, sorted_list, step, v2c, col):
-        n = self.engine.execute_filter_list(sorted_list)
+        n = self.engine.execute_filter_list(sorted_list)  # syncs inside
         self.engine.last_rows = n
+        self._stale = False
         return n
 
     def rows(self):
