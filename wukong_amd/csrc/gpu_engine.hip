@@ -1779,6 +1779,11 @@ struct wk_engine {
 
     bool light = false;  // single-kernel fast path ran (h_pin self-published)
     int remote_step_idx = -1;  // pattern to run via the xGMI peer path
+    // zero-copy i2u/c2u: the 1-col start table IS the (immutable) edge
+    // list in the store — later steps read it in place instead of
+    // replaying a 25-MB k_copy_list per query.  Cleared on every table
+    // flip/load; materialised before any in-place writer (OPT groups).
+    const sid_t *tbl_view = nullptr;
 
     // OPTIONAL group execution state (wk_engine_run_query only)
     devbuf oflag[2];          // per-row matched flags (u8), ping-pong
@@ -1826,9 +1831,26 @@ struct wk_engine {
 };
 
 static const int BLOCK = 256;
+static inline const sid_t *cur_table(wk_engine *e);
 static int grid_for(int64_t work) {
     int64_t g = (work + BLOCK - 1) / BLOCK;
     return (int)(g < 1 ? 1 : (g > 2048 ? 2048 : g));
+}
+static inline const sid_t *cur_table(wk_engine *e) {
+    return e->tbl_view ? e->tbl_view : (const sid_t *)e->tbl[e->cur].p;
+}
+// copy a zero-copy i2u view into the owned table buffer (needed before
+// any in-place writer, e.g. the OPTIONAL-group BLANK fill)
+static int32_t materialize_view(wk_engine *e) {
+    if (!e->tbl_view) return WK_OK;
+    size_t bytes = (size_t)std::max<int64_t>(e->nrows, 0) *
+                   std::max(e->ncols, 1) * 4;
+    if (bytes &&
+        hipMemcpyAsync(e->tbl[e->cur].p, e->tbl_view, bytes,
+                       hipMemcpyDeviceToDevice, e->stream) != hipSuccess)
+        return WK_ERR_HIP;
+    e->tbl_view = nullptr;
+    return WK_OK;
 }
 static int scan_grid(int64_t bound) {
     int64_t g = (bound + SCAN_T) / SCAN_T + 1;
@@ -2227,6 +2249,7 @@ extern "C" int32_t wk_engine_begin_query(wk_engine_t *e, const wk_plan_t *plan) 
     e->nrows = 0;
     e->ncols = 0;
     e->cur = 0;
+    e->tbl_view = nullptr;
     e->bound = 0;
     e->light = false;
     int32_t rc = grow_caps(e, e->cap_rows, plan->nvars);
@@ -2248,6 +2271,7 @@ static int32_t begin_light(wk_engine *e, const wk_plan_t *plan) {
     e->nrows = 0;
     e->ncols = 0;
     e->cur = 0;
+    e->tbl_view = nullptr;
     e->bound = 0;
     e->light = true;
     return grow_caps(e, e->cap_rows, plan->nvars);
@@ -2256,6 +2280,7 @@ static int32_t begin_light(wk_engine *e, const wk_plan_t *plan) {
 static int32_t load_common(wk_engine *e, int64_t nrows, int32_t ncols,
                            const int32_t *v2c_map, int32_t pattern_step) {
     e->cur = 0;
+    e->tbl_view = nullptr;
     e->nrows = nrows;
     e->bound = nrows;
     e->ncols = ncols;
@@ -2502,7 +2527,7 @@ static int32_t exec_pattern_remote(wk_engine *e) {
     const int ostat = (o >= 0) ? 2 : (e->var2col(o) >= 0 ? 1 : 0);
     const int pmode = ostat == 0 ? PM_SIZE : (ostat == 2 ? PM_CONST : PM_COL);
     if (pmode == PM_SIZE && e->ncols + 1 > e->cap_cols) return WK_ERR_STATE;
-    sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
+    const sid_t *cur_tbl = cur_table(e);
     sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
     TIME_BEGIN(e);
     hipLaunchKernelGGL(k_peer_step, dim3(1), dim3(SCAN_T), 0, e->stream,
@@ -2519,6 +2544,7 @@ static int32_t exec_pattern_remote(wk_engine *e) {
         e->bound = e->cap_rows;
     }
     e->cur ^= 1;
+    e->tbl_view = nullptr;
     e->step++;
     return WK_OK;
 }
@@ -2532,7 +2558,13 @@ static int32_t exec_pattern(wk_engine *e) {
     const wk_pattern_t pat = e->pats[e->step];
     const ssid_t s = pat.subject, p = pat.predicate, o = pat.object;
     const int dir = pat.direction;
-    sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
+    // OPT-group kernels write the current table IN PLACE (BLANK fill):
+    // a zero-copy i2u view must be materialised first
+    if (e->opt_mode && e->tbl_view) {
+        int32_t rcm = materialize_view(e);
+        if (rcm) return rcm;
+    }
+    const sid_t *cur_tbl = cur_table(e);
     sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
 
     // ---- OPTIONAL-mode dispatch (sparql.hpp:100-170,316-375) ----
@@ -2572,9 +2604,11 @@ static int32_t exec_pattern(wk_engine *e) {
                 fcval = (ostat == 2) ? (sid_t)o : 0;
             }
             TIME_BEGIN(e);
+            // view was materialised above: the owned buffer IS current
             hipLaunchKernelGGL(k_filter_opt, dim3(grid_for(e->bound)),
                                dim3(BLOCK), 0, e->stream, e->d_verts,
-                               e->d_edges, bs, nb, cur_tbl, e->ncols, fcol,
+                               e->d_edges, bs, nb,
+                               (sid_t *)e->tbl[e->cur].p, e->ncols, fcol,
                                (uint32_t)p, dir, pm, fcol2, fcval, loff, lsz,
                                e->opt_mask, (uint8_t *)e->oflag[e->ocur].p,
                                e->d_state, e->d_stats);
@@ -2601,6 +2635,7 @@ static int32_t exec_pattern(wk_engine *e) {
         e->opt_mask |= 1u << e->ncols;
         e->ncols = oc;
         e->cur ^= 1;
+        e->tbl_view = nullptr;
         e->ocur ^= 1;
         e->bound = e->cap_rows;
         e->step++;
@@ -2645,6 +2680,7 @@ static int32_t exec_pattern(wk_engine *e) {
         if (end_mode == VU_END_NEW) e->v2c[-(o + 1)] = e->ncols + 1;
         e->ncols = oc;
         e->cur ^= 1;
+        e->tbl_view = nullptr;
         e->bound = e->cap_rows;
         e->step++;
         return WK_OK;
@@ -2662,16 +2698,11 @@ static int32_t exec_pattern(wk_engine *e) {
         uint64_t off = ptr ? (uint64_t)(ptr - st->edges.data()) : 0;
         int32_t rc = grow_caps(e, (int64_t)sz, e->nvars);
         if (rc) return rc;
-        out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
-        if (sz) {
-            TIME_BEGIN(e);
-            hipLaunchKernelGGL(k_copy_list, dim3(grid_for((int64_t)sz)), dim3(BLOCK),
-                               0, e->stream, e->d_edges, off, sz, e->d_stats, out_tbl);
-            TIME_END(e, CAT_COPY);
-        }
         hipLaunchKernelGGL(k_set_state, dim3(1), dim3(1), 0, e->stream, e->d_state,
                            sz);
+        // zero-copy: the 1-col table IS the stored edge list (immutable)
         e->cur ^= 1;
+        e->tbl_view = sz ? e->d_edges + off : nullptr;
         e->nrows = (int64_t)sz;
         e->bound = (int64_t)sz;
         e->ncols = 1;
@@ -2698,6 +2729,7 @@ static int32_t exec_pattern(wk_engine *e) {
         hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
                            (uint64_t)e->cap_rows);
         e->cur ^= 1;
+        e->tbl_view = nullptr;
         e->step++;
         return WK_OK;
     }
@@ -2857,6 +2889,7 @@ static int32_t exec_pattern(wk_engine *e) {
             e->v2c[-(o + 1)] = e->ncols;
             e->ncols = oc;
             e->cur ^= 1;
+            e->tbl_view = nullptr;
             e->step += fuse2 ? 2 : 1;
             return WK_OK;
         }
@@ -2932,6 +2965,7 @@ static int32_t exec_pattern(wk_engine *e) {
             hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream,
                                e->d_state, (uint64_t)e->cap_rows);
             e->cur ^= 1;
+            e->tbl_view = nullptr;
             e->step += 2;  // consumed the fused filter pattern too
             return WK_OK;
         }
@@ -2950,13 +2984,15 @@ static int32_t exec_pattern(wk_engine *e) {
         e->v2c[-(o + 1)] = e->ncols;
         e->ncols = oc;
         e->bound = e->cap_rows;  // fan-out unknown until a sync point
-        e->cur ^= 1;             // k_scan_mid committed for this chain
+        e->cur ^= 1;
+        e->tbl_view = nullptr;             // k_scan_mid committed for this chain
         e->step++;
         return WK_OK;
     }
     hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
                        (uint64_t)e->cap_rows);
     e->cur ^= 1;
+    e->tbl_view = nullptr;
     e->step++;
     return WK_OK;
 }
@@ -3063,7 +3099,7 @@ extern "C" int32_t wk_engine_execute_filter_list(wk_engine_t *e,
     if (n && hipMemcpyAsync(e->misc.p, host_list, n * sizeof(sid_t),
                             hipMemcpyHostToDevice, e->stream) != hipSuccess)
         return WK_ERR_HIP;
-    sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
+    const sid_t *cur_tbl = cur_table(e);
     sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
     TIME_BEGIN(e);
     fparams P{e->d_verts, (const sid_t *)e->misc.p, 0, 1, cur_tbl, e->ncols,
@@ -3076,6 +3112,7 @@ extern "C" int32_t wk_engine_execute_filter_list(wk_engine_t *e,
     hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
                        (uint64_t)e->cap_rows);
     e->cur ^= 1;
+    e->tbl_view = nullptr;
     e->step++;
     int32_t rc = sync_state(e);
     if (rc == WK_OK && nrows_out) *nrows_out = e->nrows;
@@ -3098,7 +3135,7 @@ extern "C" int32_t wk_engine_generate_sub_query(wk_engine_t *e, int32_t ndst,
     unsigned long long *d_hist = (unsigned long long *)e->misc.p;
     hipLaunchKernelGGL(k_zero_words, dim3(1), dim3(128), 0, e->stream,
                        (uint64_t *)e->misc.p, 128);
-    sid_t *cur_tbl = (sid_t *)e->tbl[e->cur].p;
+    const sid_t *cur_tbl = cur_table(e);
     // per-(block,dst) bases live in the prefix scratch: G2*ndst <=
     // (R/BLOCK+1)*64 entries always fits its (cap_rows+1) u64s
     const int G2 = (int)std::max<int64_t>(
@@ -3143,7 +3180,7 @@ static int32_t download_table(wk_engine *e, sid_t *dst, size_t n) {
         if (hipHostMalloc(&e->h_stage, want) != hipSuccess) return WK_ERR_HIP;
         e->h_stage_cap = want;
     }
-    HIP_CHECK(hipMemcpyAsync(e->h_stage, e->tbl[e->cur].p, bytes,
+    HIP_CHECK(hipMemcpyAsync(e->h_stage, cur_table(e), bytes,
                              hipMemcpyDeviceToHost, e->stream));
     HIP_CHECK(stream_sync(e->stream));
     memcpy(dst, e->h_stage, bytes);
@@ -3223,7 +3260,7 @@ extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
         sid_t *out_tbl = (sid_t *)e->tbl[e->cur ^ 1].p;
         TIME_BEGIN(e);
         hipLaunchKernelGGL(k_project, dim3(grid_for(e->cap_rows)), dim3(BLOCK), 0,
-                           e->stream, (sid_t *)e->tbl[e->cur].p, e->ncols,
+                           e->stream, cur_table(e), e->ncols,
                            e->d_state, cols, plan->nrequired, out_tbl);
         TIME_END(e, CAT_OTHER);
         int32_t rc = sync_state_grow(e);
@@ -3231,9 +3268,11 @@ extern "C" int32_t wk_engine_fetch_result(wk_engine_t *e, const wk_plan_t *plan,
         double t0 = now_us();
         size_t n = (size_t)e->nrows * plan->nrequired;
         wk_sid_t *res = (wk_sid_t *)malloc(n ? n * 4 : 4);
-        e->cur ^= 1;  // projected table is current for the download
+        e->cur ^= 1;
+        e->tbl_view = nullptr;  // projected table is current for the download
         rc = download_table(e, res, n);
         e->cur ^= 1;
+        e->tbl_view = nullptr;
         if (rc) { free(res); return rc; }
         out->col_num = plan->nrequired;
         out->row_num = e->nrows;
@@ -3666,7 +3705,9 @@ static int32_t run_union(wk_engine *e, const wk_plan_t *plan) {
     size_t bytes = (size_t)pr * std::max(pc, 1) * 4;
     if (bytes && pc) {
         if (snap.ensure(bytes)) return WK_ERR_HIP;
-        HIP_CHECK(hipMemcpyAsync(snap.p, e->tbl[pcur].p, bytes,
+        // cur_table: the parent may be a zero-copy i2u view; the
+        // restore below rematerialises it into the owned buffer
+        HIP_CHECK(hipMemcpyAsync(snap.p, cur_table(e), bytes,
                                  hipMemcpyDeviceToDevice, e->stream));
     }
     std::vector<sid_t> merged;
@@ -3678,6 +3719,7 @@ static int32_t run_union(wk_engine *e, const wk_plan_t *plan) {
             e->v2c = pv2c;
             e->ncols = pc;
             e->cur = pcur;
+            e->tbl_view = nullptr;  // snap restore materialises the view
             if (bytes && pc)
                 HIP_CHECK(hipMemcpyAsync(e->tbl[pcur].p, snap.p, bytes,
                                          hipMemcpyDeviceToDevice, e->stream));
@@ -3705,7 +3747,7 @@ static int32_t run_union(wk_engine *e, const wk_plan_t *plan) {
         size_t off = merged.size();
         merged.resize(off + (size_t)e->nrows * bc);
         if (e->nrows)
-            HIP_CHECK(hipMemcpy(merged.data() + off, e->tbl[e->cur].p,
+            HIP_CHECK(hipMemcpy(merged.data() + off, cur_table(e),
                                 (size_t)e->nrows * bc * 4,
                                 hipMemcpyDeviceToHost));
     }
