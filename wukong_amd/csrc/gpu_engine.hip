@@ -266,7 +266,6 @@ __global__ void k_probe_scan(const vertex_t *__restrict__ verts,
         const int64_t r = base + threadIdx.x;
         uint64_t eoff = 0, esz = 0;
         uint64_t key = nkey;
-        uint64_t bucket = nbucket;
         uint64_t ck[ASSOC], cp[ASSOC - 1];
         bool have = have_pref;
         if (have) {
