@@ -192,6 +192,11 @@ int32_t wk_engine_graph_run(wk_engine_t *, int32_t graph_id,
 /* Asynchronous replay (no sync): back-to-back graphs on one stream
  * serialize safely, so a whole suite pass costs ONE wk_engine_sync —
  * the reference proxy's in-flight window applied to replays. */
+/* Whole-suite capture: N plans into ONE instantiated graph (pays the
+ * graph-replay floor once per pass).  Same gid space; replay with
+ * wk_engine_graph_launch + wk_engine_sync. */
+int32_t wk_engine_graph_build_suite(wk_engine_t *, const wk_plan_t *plans,
+                                    int32_t nplans, int32_t *gid);
 int32_t wk_engine_graph_launch(wk_engine_t *, int32_t graph_id);
 int32_t wk_engine_sync(wk_engine_t *);
 

@@ -98,6 +98,9 @@ _eng_graph_build = _sig("wk_engine_graph_build", c_i32,
 _eng_graph_run = _sig("wk_engine_graph_run", c_i32,
                       [c_vp, c_i32, ctypes.POINTER(c_i64)])
 _eng_graph_launch = _sig("wk_engine_graph_launch", c_i32, [c_vp, c_i32])
+_eng_graph_suite = _sig("wk_engine_graph_build_suite", c_i32,
+                        [c_vp, ctypes.POINTER(WkPlan), c_i32,
+                         ctypes.POINTER(c_i32)])
 _eng_sync = _sig("wk_engine_sync", c_i32, [c_vp])
 
 #: plan-batch count sentinel: table outgrew LDS, re-run per-pattern
@@ -406,6 +409,23 @@ class Engine:
         if rc != 0:
             raise RuntimeError(f"graph_run rc={rc}")
         return int(n.value)
+
+    def graph_build_suite(self, plans):
+        """Capture ALL plans back-to-back into ONE hipGraph (pays the
+        replay floor once per pass); returns a graph id for
+        graph_launch."""
+        cplans = (WkPlan * len(plans))()
+        keep = []
+        for i, p in enumerate(plans):
+            cp = p.to_c()
+            cplans[i] = cp
+            keep.append(cp)
+        gid = c_i32()
+        rc = _eng_graph_suite(self._h, cplans, len(plans), ctypes.byref(gid))
+        if rc != 0:
+            raise RuntimeError(f"graph_build_suite rc={rc}")
+        self._suite_keep = keep
+        return int(gid.value)
 
     def graph_launch(self, gid):
         """Async replay (no sync); pair with sync()."""

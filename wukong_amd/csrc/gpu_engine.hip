@@ -3456,6 +3456,88 @@ extern "C" int32_t wk_engine_graph_build(wk_engine_t *e,
     return WK_OK;
 }
 
+// Whole-SUITE capture: N plans recorded back-to-back into ONE
+// instantiated graph, so a full benchmark pass pays the ~10-16 us
+// graph-replay floor once instead of once per query (microarch row
+// `graph-replay-floor`).  Each query keeps its own in-capture publish,
+// so the sticky S_ERR overflow guard still sees every query.  Replay
+// with wk_engine_graph_launch + wk_engine_sync; counts come from the
+// per-query graphs / latency passes.
+extern "C" int32_t wk_engine_graph_build_suite(wk_engine_t *e,
+                                               const wk_plan_t *plans,
+                                               int32_t nplans,
+                                               int32_t *gid) {
+    if (!e || !plans || nplans <= 0 || !gid) return WK_ERR_STATE;
+    std::vector<std::vector<uint8_t>> hints((size_t)nplans);
+    for (int i = 0; i < nplans; i++) {
+        const wk_plan_t *plan = &plans[i];
+        if (plan->nopt > 0 || plan->nunion > 0 || plan->distinct ||
+            plan->limit >= 0 || plan->offset > 0)
+            return WK_ERR_PLAN;
+        // warm passes settle scratch capacities so capture never allocates
+        for (int attempt = 0; attempt < 4; attempt++) {
+            int32_t rc = wk_engine_submit(e, plan);
+            if (rc) return rc;
+            rc = sync_state_grow(e);
+            if (rc == WK_OK) break;
+            if (rc != WK_ERR_CAP) return rc;
+        }
+        hints[i].assign((size_t)std::max(plan->npatterns, 0), 0);
+        int32_t rc = wk_engine_begin_query(e, plan);
+        if (rc) return rc;
+        int64_t prev = -1;
+        while (e->step < (int)e->pats.size()) {
+            const int at = e->step;
+            int64_t n = 0;
+            rc = wk_engine_execute_one_pattern(e, &n);
+            if (rc) return rc;
+            if (n == prev && at < (int)hints[i].size()) hints[i][at] = 1;
+            prev = n;
+        }
+    }
+    e->capturing = 1;
+    hipGraph_t g = nullptr;
+    if (hipStreamBeginCapture(e->stream, hipStreamCaptureModeThreadLocal) !=
+        hipSuccess) {
+        e->capturing = 0;
+        e->capture_hint.clear();
+        return WK_ERR_HIP;
+    }
+    int32_t rc = WK_OK;
+    for (int i = 0; i < nplans && rc == WK_OK; i++) {
+        e->capture_hint = hints[i];
+        rc = wk_engine_submit(e, &plans[i]);
+        if (rc == WK_OK && !e->light)
+            hipLaunchKernelGGL(k_publish_state, dim3(1), dim3(1), 0,
+                               e->stream, e->d_state, e->d_stats, e->h_pin);
+    }
+    hipError_t ce = hipStreamEndCapture(e->stream, &g);
+    e->capturing = 0;
+    e->capture_hint.clear();
+    if (rc != WK_OK || ce != hipSuccess || !g) {
+        if (g) (void)hipGraphDestroy(g);
+        (void)stream_sync(e->stream);
+        return rc != WK_OK ? rc : WK_ERR_HIP;
+    }
+    hipGraphExec_t ex = nullptr;
+    if (hipGraphInstantiate(&ex, g, nullptr, nullptr, 0) != hipSuccess) {
+        (void)hipGraphDestroy(g);
+        return WK_ERR_HIP;
+    }
+    (void)hipGraphDestroy(g);
+    wk_engine::wk_graph wg;
+    wg.exec = ex;
+    wg.v2c = e->v2c;
+    wg.ncols = e->ncols;
+    wg.cur = e->cur;
+    wg.nvars = e->nvars;
+    wg.bound = e->bound;
+    wg.light = e->light;
+    e->graphs.push_back(std::move(wg));
+    *gid = (int32_t)e->graphs.size() - 1;
+    return WK_OK;
+}
+
 // Asynchronous replay: enqueue the graph with NO sync — back-to-back
 // graphs on one stream serialize safely (each begins with its own
 // state-reset kernel), so a whole suite pass costs ONE host sync
