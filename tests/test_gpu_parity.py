@@ -261,3 +261,19 @@ def test_graph_replay_counts(store4, oracle4):
         want = len(oracle4.run_query(plan))
         for _ in range(3):
             assert eng.graph_run(gid) == want, name
+
+
+def test_graph_suite_replay(store4, oracle4):
+    """Whole-suite capture (wk_engine_graph_build_suite): all plans in
+    ONE graph; replays stay parity-green — the LAST query's count is
+    published (graph_run semantics), and the sticky overflow guard
+    covers every query inside the capture (eng.sync raises on S_ERR)."""
+    eng = wk.Engine(store4, device=0)
+    plans = [Q.ALL[n] for n in ("q1", "q2", "q5", "q7")]
+    gid = eng.graph_build_suite(plans)
+    want_last = len(oracle4.run_query(plans[-1]))
+    for _ in range(3):
+        assert eng.graph_run(gid) == want_last
+    for _ in range(5):  # async replays, one sync (the timed-loop shape)
+        eng.graph_launch(gid)
+    eng.sync()
