@@ -663,6 +663,11 @@ __global__ void k_publish_state(const uint64_t *__restrict__ d_state,
 // output-centric version paid 2-3 DRAM lines of prefix walk per output
 // row).  Rows with deg > 32 go to an overflow queue handled by
 // k_expand_big with one WAVE per row (lanes stride the edge list).
+// verify-fused expansion (captured graphs only): when the NEXT pattern
+// is a no-drop `?v rdf:type CONST` filter on the NEW column (warm-pass
+// hint), each written value is checked inline (bitmap or dense type_of;
+// misses count into S_DONE -> k_verify_commit -> S_ERR -> safe re-run)
+// and the separate full-table verify pass disappears.
 template <int NC>
 __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
                             const sid_t *__restrict__ edges,
@@ -673,6 +678,10 @@ __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
                             uint64_t *__restrict__ d_state, uint64_t cap,
                             uint64_t *__restrict__ d_stats,
                             uint32_t *__restrict__ ovf,
+                            const uint64_t *__restrict__ vtbm,
+                            const uint16_t *__restrict__ v_type_of,
+                            uint64_t v_type_base, uint64_t v_type_n,
+                            sid_t v_cval, int verify,
                             sid_t *__restrict__ out)
 {
     // k_scan_mid already committed: S_INROWS = the input row count,
@@ -699,12 +708,28 @@ __global__ void k_expand_in(const sid_t *__restrict__ tbl, int ncols,
         for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
         const sid_t *el = edges + d_eoff[r];
         sid_t *dst = out + (int64_t)basep * oc;
+        uint32_t miss = 0;
         for (uint32_t k = 0; k < deg; k++) {
 #pragma unroll
             for (int c = 0; c < NC; c++) dst[c] = row[c];
-            dst[NC] = el[k];
+            sid_t v = el[k];
+            dst[NC] = v;
             dst += oc;
+            if (verify) {
+                uint64_t tix = (uint64_t)v - v_type_base;
+                bool ok;
+                if (vtbm) {
+                    ok = tix < v_type_n && ((vtbm[tix >> 6] >> (tix & 63)) & 1);
+                } else {
+                    uint16_t t = (tix < v_type_n) ? v_type_of[tix] : 0;
+                    ok = t != 0xFFFF && (sid_t)t == v_cval;
+                }
+                miss += ok ? 0u : 1u;
+            }
         }
+        if (miss)
+            atomicAdd((unsigned long long *)&d_state[S_DONE],
+                      (unsigned long long)miss);
     }
     // algorithmic bytes for the whole expansion (counted once; includes
     // the big-row pass): total*(edge 4 + write 4*oc) + nrows*(row 4*ncols
@@ -727,6 +752,10 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
                              uint64_t *__restrict__ d_state, uint64_t cap,
                              int commit_mode,
                              const uint32_t *__restrict__ ovf,
+                             const uint64_t *__restrict__ vtbm,
+                             const uint16_t *__restrict__ v_type_of,
+                             uint64_t v_type_base, uint64_t v_type_n,
+                             sid_t v_cval, int verify,
                              sid_t *__restrict__ out)
 {
     const int64_t nq = (int64_t)d_state[S_OVF];
@@ -747,12 +776,28 @@ __global__ void k_expand_big(const sid_t *__restrict__ tbl, int ncols,
 #pragma unroll
         for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
         const sid_t *el = edges + d_eoff[r];
+        uint32_t miss = 0;
         for (uint64_t k = lane; k < deg; k += 64) {
             sid_t *dst = out + (int64_t)(basep + k) * oc;
 #pragma unroll
             for (int c = 0; c < NC; c++) dst[c] = row[c];
-            dst[NC] = el[k];
+            sid_t v = el[k];
+            dst[NC] = v;
+            if (verify) {
+                uint64_t tix = (uint64_t)v - v_type_base;
+                bool ok;
+                if (vtbm) {
+                    ok = tix < v_type_n && ((vtbm[tix >> 6] >> (tix & 63)) & 1);
+                } else {
+                    uint16_t t = (tix < v_type_n) ? v_type_of[tix] : 0;
+                    ok = t != 0xFFFF && (sid_t)t == v_cval;
+                }
+                miss += ok ? 0u : 1u;
+            }
         }
+        if (miss)
+            atomicAdd((unsigned long long *)&d_state[S_DONE],
+                      (unsigned long long)miss);
     }
     commit_tail(d_state, cap, commit_mode);
 }
@@ -1665,6 +1710,16 @@ __global__ void k_peer_step(const wk_peer_desc *__restrict__ peers, int nsrv,
     }
 }
 
+// verify-fused expansion epilogue: any inline-typeof miss (S_DONE)
+// flips S_ERR so the replay is discarded and the safe path re-runs
+__global__ void k_verify_commit(uint64_t *__restrict__ d_state) {
+    if (d_state[S_DONE]) {
+        d_state[S_ERR] = 1;
+        d_state[S_REQ] = max(d_state[S_REQ], d_state[S_NROWS]);
+        d_state[S_DONE] = 0;
+    }
+}
+
 // commit for the optimistic map: row count unchanged; any miss flags
 // S_ERR so the replay result is discarded and the safe path re-runs
 __global__ void k_commit_map(uint64_t *__restrict__ d_state) {
@@ -2347,21 +2402,34 @@ static int32_t sync_state_grow(wk_engine *e) {
 // expansion dispatch, specialised on the input column count so the row
 // copy unrolls/vectorises (runtime-indexed local arrays spill to scratch
 // — cdna_hip_programming.md §5.4 rule 20)
+struct expand_verify {  // fused no-drop typeof check (captured graphs)
+    const uint64_t *tbm = nullptr;
+    const uint16_t *type_of = nullptr;
+    uint64_t base = 0, n = 0;
+    sid_t cval = 0;
+    int on = 0;
+};
+
 template <int NC>
 static void launch_expand_t(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
-                            int G) {
+                            int G, const expand_verify &vf) {
     hipLaunchKernelGGL(k_expand_in<NC>, dim3(grid_for(e->bound)), dim3(BLOCK), 0,
                        e->stream, cur_tbl, e->ncols, e->d_edges,
                        (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p,
                        (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
                        e->d_state, (uint64_t)e->cap_rows, e->d_stats,
-                       (uint32_t *)e->ovf.p, out_tbl);
+                       (uint32_t *)e->ovf.p, vf.tbm, vf.type_of, vf.base,
+                       vf.n, vf.cval, vf.on, out_tbl);
     hipLaunchKernelGGL(k_expand_big<NC>, dim3(512), dim3(BLOCK), 0, e->stream,
                        cur_tbl, e->ncols, e->d_edges, (uint64_t *)e->eoff.p,
                        (uint32_t *)e->cnt.p, (uint64_t *)e->prefix.p,
                        (uint64_t *)e->bsums.p, G, e->d_state,
                        (uint64_t)e->cap_rows, /*commit*/ 0,
-                       (uint32_t *)e->ovf.p, out_tbl);
+                       (uint32_t *)e->ovf.p, vf.tbm, vf.type_of, vf.base,
+                       vf.n, vf.cval, vf.on, out_tbl);
+    if (vf.on)
+        hipLaunchKernelGGL(k_verify_commit, dim3(1), dim3(1), 0, e->stream,
+                           e->d_state);
 }
 
 template <int NC>
@@ -2480,16 +2548,16 @@ static void launch_expand_fn(wk_engine *e, const sid_t *cur_tbl,
 }
 
 static void launch_expand(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
-                          int G) {
+                          int G, const expand_verify &vf) {
     switch (e->ncols) {
-    case 1: launch_expand_t<1>(e, cur_tbl, out_tbl, G); break;
-    case 2: launch_expand_t<2>(e, cur_tbl, out_tbl, G); break;
-    case 3: launch_expand_t<3>(e, cur_tbl, out_tbl, G); break;
-    case 4: launch_expand_t<4>(e, cur_tbl, out_tbl, G); break;
-    case 5: launch_expand_t<5>(e, cur_tbl, out_tbl, G); break;
-    case 6: launch_expand_t<6>(e, cur_tbl, out_tbl, G); break;
-    case 7: launch_expand_t<7>(e, cur_tbl, out_tbl, G); break;
-    default: launch_expand_t<8>(e, cur_tbl, out_tbl, G); break;
+    case 1: launch_expand_t<1>(e, cur_tbl, out_tbl, G, vf); break;
+    case 2: launch_expand_t<2>(e, cur_tbl, out_tbl, G, vf); break;
+    case 3: launch_expand_t<3>(e, cur_tbl, out_tbl, G, vf); break;
+    case 4: launch_expand_t<4>(e, cur_tbl, out_tbl, G, vf); break;
+    case 5: launch_expand_t<5>(e, cur_tbl, out_tbl, G, vf); break;
+    case 6: launch_expand_t<6>(e, cur_tbl, out_tbl, G, vf); break;
+    case 7: launch_expand_t<7>(e, cur_tbl, out_tbl, G, vf); break;
+    default: launch_expand_t<8>(e, cur_tbl, out_tbl, G, vf); break;
     }
 }
 
@@ -2975,9 +3043,30 @@ static int32_t exec_pattern(wk_engine *e) {
                                (uint64_t)e->cap_rows, e->d_state);
             TIME_END(e, CAT_SCAN);
         }
+        // verify-fused no-drop typeof on the NEW column (captured
+        // graphs only, warm-pass hint — same guard scheme as fn_map):
+        // the separate full-table verify pass collapses into the
+        // expansion's write loop + a 1-thread epilogue
+        expand_verify vf;
+        if (e->capturing && e->step + 1 < (int)e->pats.size() &&
+            e->step + 1 < (int)e->capture_hint.size() &&
+            e->capture_hint[e->step + 1] && e->st->nsrv == 1) {
+            const wk_pattern_t &nx = e->pats[e->step + 1];
+            if (nx.subject == o && nx.predicate == (ssid_t)TYPE_ID &&
+                nx.direction == DIR_OUT && nx.object > 0) {
+                vf.cval = (sid_t)nx.object;
+                vf.base = e->st->type_base;
+                vf.n = e->st->type_n;
+                vf.tbm = (e->gs && (size_t)vf.cval < e->gs->d_tbm.size())
+                             ? e->gs->d_tbm[vf.cval]
+                             : nullptr;
+                vf.type_of = e->d_type_of;
+                if (vf.tbm || vf.type_of) vf.on = 1;
+            }
+        }
         {
             TIME_BEGIN(e);
-            launch_expand(e, cur_tbl, out_tbl, G);
+            launch_expand(e, cur_tbl, out_tbl, G, vf);
             TIME_END(e, CAT_EXPAND);
         }
         e->v2c[-(o + 1)] = e->ncols;
@@ -2985,7 +3074,7 @@ static int32_t exec_pattern(wk_engine *e) {
         e->bound = e->cap_rows;  // fan-out unknown until a sync point
         e->cur ^= 1;
         e->tbl_view = nullptr;             // k_scan_mid committed for this chain
-        e->step++;
+        e->step += vf.on ? 2 : 1;  // fused verify consumed the filter
         return WK_OK;
     }
     hipLaunchKernelGGL(k_commit, dim3(1), dim3(1), 0, e->stream, e->d_state,
