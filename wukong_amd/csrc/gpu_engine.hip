@@ -356,6 +356,166 @@ __global__ void k_csr_gather(const sid_t *__restrict__ tbl, int ncols,
     }
 }
 
+// Counting variant for the EXACT fused `k2u + ?v rdf:type CONST`
+// compaction (nothing optimistic: output == expansion+filter output):
+// per row, CSR entry lookup, then ONE walk of the edge list counting
+// values whose type-bitmap bit is set.  d_cnt gets the FILTERED count
+// (the scan then yields exact output positions); d_eoff packs
+// {edge_off:40 | full_len:24} so the writer can re-walk the list.
+__global__ void k_csr_gather_tf(const sid_t *__restrict__ tbl, int ncols,
+                                int col,
+                                const fnpage_t *__restrict__ pg,
+                                const uint64_t *__restrict__ entries,
+                                uint64_t base, uint64_t n,
+                                const sid_t *__restrict__ edges,
+                                const uint64_t *__restrict__ tbm,
+                                uint64_t t_base, uint64_t t_n,
+                                const uint64_t *__restrict__ d_state,
+                                uint64_t *__restrict__ d_stats,
+                                uint64_t *__restrict__ d_eoff,
+                                uint32_t *__restrict__ d_cnt)
+{
+    const int64_t nrows = (int64_t)d_state[S_NROWS];
+    count_bytes(d_stats, CAT_PROBE, (uint64_t)nrows * (4 + 16 + 8 + 12));
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+         r < nrows; r += (int64_t)gridDim.x * blockDim.x) {
+        sid_t v = tbl[r * ncols + col];
+        uint64_t e = 0;
+        uint64_t idx = (uint64_t)v - base;
+        if (idx < n) {
+            const fnpage_t p = pg[idx >> 6];
+            if ((p.bits >> (idx & 63)) & 1) {
+                uint32_t rk = p.rank +
+                              (uint32_t)__popcll(p.bits &
+                                                 ((1ull << (idx & 63)) - 1));
+                e = entries[rk];
+            }
+        }
+        const uint64_t off = e >> 24;
+        const uint32_t len = (uint32_t)(e & 0xFFFFFF);
+        uint32_t keep = 0;
+        for (uint32_t k = 0; k < len; k++) {
+            uint64_t tix = (uint64_t)edges[off + k] - t_base;
+            keep += (tix < t_n && ((tbm[tix >> 6] >> (tix & 63)) & 1)) ? 1u
+                                                                       : 0u;
+        }
+        d_eoff[r] = e;  // {off:40|len:24} for the writer's re-walk
+        d_cnt[r] = keep;
+    }
+}
+
+// writer for the fused compaction: re-walk the edge list, emit only
+// passing values at the exact scanned positions (wave-ballot ranks for
+// the big-row pass below)
+template <int NC>
+__global__ void k_expand_tf(const sid_t *__restrict__ tbl, int ncols,
+                            const sid_t *__restrict__ edges,
+                            const uint64_t *__restrict__ d_eoff,
+                            const uint32_t *__restrict__ d_cnt,
+                            const uint64_t *__restrict__ d_pre,
+                            const uint64_t *__restrict__ bsums, int G,
+                            const uint64_t *__restrict__ tbm,
+                            uint64_t t_base, uint64_t t_n,
+                            uint64_t *__restrict__ d_state, uint64_t cap,
+                            uint64_t *__restrict__ d_stats,
+                            uint32_t *__restrict__ ovf,
+                            sid_t *__restrict__ out)
+{
+    const int64_t nrows = (int64_t)d_state[S_INROWS];
+    const int64_t chunk = (nrows + G - 1) / G;
+    constexpr int oc = NC + 1;
+    (void)ncols;
+    for (int64_t r = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; r < nrows;
+         r += (int64_t)gridDim.x * blockDim.x) {
+        const uint32_t keep = d_cnt[r];
+        if (!keep) continue;
+        uint64_t basep = d_pre[r] + bsums[r / chunk];
+        if (basep >= cap) continue;
+        const uint64_t e = d_eoff[r];
+        const uint64_t off = e >> 24;
+        const uint32_t len = (uint32_t)(e & 0xFFFFFF);
+        if (len > 32) {
+            unsigned long long i = atomicAdd(
+                (unsigned long long *)&d_state[S_OVF], 1ull);
+            ovf[i] = (uint32_t)r;
+            continue;
+        }
+        sid_t row[NC];
+#pragma unroll
+        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
+        sid_t *dst = out + (int64_t)basep * oc;
+        uint32_t w = 0;
+        for (uint32_t k = 0; k < len && w < keep; k++) {
+            sid_t v = edges[off + k];
+            uint64_t tix = (uint64_t)v - t_base;
+            if (!(tix < t_n && ((tbm[tix >> 6] >> (tix & 63)) & 1))) continue;
+            if (basep + w >= cap) break;
+#pragma unroll
+            for (int c = 0; c < NC; c++) dst[c] = row[c];
+            dst[NC] = v;
+            dst += oc;
+            w++;
+        }
+    }
+}
+
+template <int NC>
+__global__ void k_expand_tf_big(const sid_t *__restrict__ tbl, int ncols,
+                                const sid_t *__restrict__ edges,
+                                const uint64_t *__restrict__ d_eoff,
+                                const uint32_t *__restrict__ d_cnt,
+                                const uint64_t *__restrict__ d_pre,
+                                const uint64_t *__restrict__ bsums, int G,
+                                const uint64_t *__restrict__ tbm,
+                                uint64_t t_base, uint64_t t_n,
+                                uint64_t *__restrict__ d_state, uint64_t cap,
+                                const uint32_t *__restrict__ ovf,
+                                sid_t *__restrict__ out)
+{
+    const int64_t nq = (int64_t)d_state[S_OVF];
+    const int64_t nrows = (int64_t)d_state[S_INROWS];
+    const int64_t chunk = (nrows + G - 1) / G;
+    constexpr int oc = NC + 1;
+    (void)ncols;
+    const int lane = threadIdx.x & 63;
+    const int64_t w0 = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+    const int64_t nw = ((int64_t)gridDim.x * blockDim.x) >> 6;
+    for (int64_t q = w0; q < nq; q += nw) {
+        const int64_t r = ovf[q];
+        const uint64_t e = d_eoff[r];
+        const uint64_t off = e >> 24;
+        const uint32_t len = (uint32_t)(e & 0xFFFFFF);
+        uint64_t basep = d_pre[r] + bsums[r / chunk];
+        if (basep >= cap) continue;
+        sid_t row[NC];
+#pragma unroll
+        for (int c = 0; c < NC; c++) row[c] = tbl[r * NC + c];
+        uint64_t wbase = 0;  // passing-rank base across 64-edge chunks
+        for (uint32_t k0 = 0; k0 < len; k0 += 64) {
+            const uint32_t k = k0 + lane;
+            bool pass = false;
+            sid_t v = 0;
+            if (k < len) {
+                v = edges[off + k];
+                uint64_t tix = (uint64_t)v - t_base;
+                pass = tix < t_n && ((tbm[tix >> 6] >> (tix & 63)) & 1);
+            }
+            uint64_t mask = __ballot(pass);
+            if (pass) {
+                uint64_t pos = basep + wbase +
+                               (uint64_t)__popcll(mask & ((1ull << lane) - 1));
+                if (pos < cap) {
+                    sid_t *dst = out + (int64_t)pos * oc;
+#pragma unroll
+                    for (int c = 0; c < NC; c++) dst[c] = row[c];
+                    dst[NC] = v;
+                }
+            }
+            wbase += (uint64_t)__popcll(mask);
+        }
+    }
+}
+
 // chunked exclusive prefix of d_cnt -> d_pre + per-chunk sums (same
 // chunk math as k_probe_scan so the expansion kernels are unchanged)
 __global__ void k_scan_local(const uint32_t *__restrict__ d_cnt,
@@ -2547,6 +2707,38 @@ static void launch_expand_fn(wk_engine *e, const sid_t *cur_tbl,
     }
 }
 
+template <int NC>
+static void launch_expand_tf_t(wk_engine *e, const sid_t *cur_tbl,
+                               sid_t *out_tbl, int G, const uint64_t *tbm) {
+    hipLaunchKernelGGL(k_expand_tf<NC>, dim3(grid_for(e->bound)), dim3(BLOCK),
+                       0, e->stream, cur_tbl, e->ncols, e->d_edges,
+                       (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p,
+                       (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
+                       tbm, e->st->type_base, e->st->type_n, e->d_state,
+                       (uint64_t)e->cap_rows, e->d_stats,
+                       (uint32_t *)e->ovf.p, out_tbl);
+    hipLaunchKernelGGL(k_expand_tf_big<NC>, dim3(512), dim3(BLOCK), 0,
+                       e->stream, cur_tbl, e->ncols, e->d_edges,
+                       (uint64_t *)e->eoff.p, (uint32_t *)e->cnt.p,
+                       (uint64_t *)e->prefix.p, (uint64_t *)e->bsums.p, G,
+                       tbm, e->st->type_base, e->st->type_n, e->d_state,
+                       (uint64_t)e->cap_rows, (uint32_t *)e->ovf.p, out_tbl);
+}
+
+static void launch_expand_tf(wk_engine *e, const sid_t *cur_tbl,
+                             sid_t *out_tbl, int G, const uint64_t *tbm) {
+    switch (e->ncols) {
+    case 1: launch_expand_tf_t<1>(e, cur_tbl, out_tbl, G, tbm); break;
+    case 2: launch_expand_tf_t<2>(e, cur_tbl, out_tbl, G, tbm); break;
+    case 3: launch_expand_tf_t<3>(e, cur_tbl, out_tbl, G, tbm); break;
+    case 4: launch_expand_tf_t<4>(e, cur_tbl, out_tbl, G, tbm); break;
+    case 5: launch_expand_tf_t<5>(e, cur_tbl, out_tbl, G, tbm); break;
+    case 6: launch_expand_tf_t<6>(e, cur_tbl, out_tbl, G, tbm); break;
+    case 7: launch_expand_tf_t<7>(e, cur_tbl, out_tbl, G, tbm); break;
+    default: launch_expand_tf_t<8>(e, cur_tbl, out_tbl, G, tbm); break;
+    }
+}
+
 static void launch_expand(wk_engine *e, const sid_t *cur_tbl, sid_t *out_tbl,
                           int G, const expand_verify &vf) {
     switch (e->ncols) {
@@ -2999,6 +3191,60 @@ static int32_t exec_pattern(wk_engine *e) {
              !e->gs->d_csr_pages.empty())
                 ? e->gs->d_csr_pages[(size_t)p * 2 + dir]
                 : nullptr;
+        // EXACT fused `k2u + ?v rdf:type CONST` compaction (nothing
+        // optimistic: output == expansion-then-filter output): the CSR
+        // walk counts type-bitmap-passing values, the scan yields exact
+        // positions, the writer re-walks and emits only those.  Skipped
+        // when the capture hint says the filter drops nothing (the
+        // verify-fused expansion above is cheaper there).
+        if (c_pg && e->st->nsrv == 1 && e->gs &&
+            e->step + 1 < (int)e->pats.size()) {
+            const wk_pattern_t &nx = e->pats[e->step + 1];
+            const bool nodrop = e->capturing &&
+                                e->step + 1 < (int)e->capture_hint.size() &&
+                                e->capture_hint[e->step + 1];
+            if (!nodrop && nx.subject == o &&
+                nx.predicate == (ssid_t)TYPE_ID &&
+                nx.direction == DIR_OUT && nx.object > 0 &&
+                (size_t)nx.object < e->gs->d_tbm.size() &&
+                e->gs->d_tbm[nx.object]) {
+                const uint64_t *tf_tbm = e->gs->d_tbm[nx.object];
+                TIME_BEGIN(e);
+                hipLaunchKernelGGL(k_csr_gather_tf, dim3(grid_for(e->bound)),
+                                   dim3(BLOCK), 0, e->stream, cur_tbl,
+                                   e->ncols, col, c_pg,
+                                   e->gs->d_csr_entries[(size_t)p * 2 + dir],
+                                   st->fn_base, st->fn_n, e->d_edges, tf_tbm,
+                                   e->st->type_base, e->st->type_n,
+                                   e->d_state, e->d_stats,
+                                   (uint64_t *)e->eoff.p,
+                                   (uint32_t *)e->cnt.p);
+                hipLaunchKernelGGL(k_scan_local, dim3(G), dim3(SCAN_T), 0,
+                                   e->stream, (const uint32_t *)e->cnt.p,
+                                   e->d_state, (uint64_t *)e->prefix.p,
+                                   (uint64_t *)e->bsums.p);
+                TIME_END(e, CAT_PROBE);
+                {
+                    TIME_BEGIN(e);
+                    hipLaunchKernelGGL(k_scan_mid, dim3(1), dim3(SCAN_T), 0,
+                                       e->stream, (uint64_t *)e->bsums.p, G,
+                                       (uint64_t)e->cap_rows, e->d_state);
+                    TIME_END(e, CAT_SCAN);
+                }
+                {
+                    TIME_BEGIN(e);
+                    launch_expand_tf(e, cur_tbl, out_tbl, G, tf_tbm);
+                    TIME_END(e, CAT_EXPAND);
+                }
+                e->v2c[-(o + 1)] = e->ncols;
+                e->ncols = oc;
+                e->bound = e->cap_rows;
+                e->cur ^= 1;
+                e->tbl_view = nullptr;
+                e->step += 2;  // consumed the typeof filter pattern
+                return WK_OK;
+            }
+        }
         TIME_BEGIN(e);
         if (c_pg) {
             hipLaunchKernelGGL(k_csr_gather, dim3(grid_for(e->bound)),
