@@ -22,13 +22,19 @@
  *                       (replaces gpu_hash.cu:326-445,523-585)
  *   k_light2          — whole-query single-kernel path for 2-pattern
  *                       light templates (emulator A1/A2/A3/A5)
- *   k_scan_part/mid/add — 3-phase exclusive scan with DEVICE-side length
- *                       (replaces the thrust scans, gpu_hash.cu:587-596)
- *   k_expand          — output-centric load-balanced expansion
- *                       (replaces gpu_hash.cu:762-834; no 20-col cap)
- *   k_compact         — flag compaction (replaces gpu_hash.cu:523-585)
- *   k_copy_list       — i2u/c2u list materialisation on device (the
- *                       reference ran these on CPU, gpu_engine.hpp:63-123)
+ *   k_scan_local/mid  — chunked exclusive scan with DEVICE-side length
+ *                       (replaces the thrust scans, gpu_hash.cu:587-596);
+ *                       k_scan_mid also commits the step (S_INROWS
+ *                       snapshot for the scatter side)
+ *   k_expand_in/big   — input-centric expansion + wave-per-big-row pass
+ *                       (replaces gpu_hash.cu:762-834; no 20-col cap),
+ *                       optional inline no-drop typeof verify in graphs
+ *   k_expand_tf[_big] — EXACT fused k2u + typeof compaction over the
+ *                       CSR index (count pass -> scan -> filtered write)
+ *   k_csr_gather      — 24-B rank-compressed CSR probes for
+ *                       non-functional segments (vs 128-B bucket walks)
+ *   zero-copy i2u/c2u — the 1-col start table IS the stored edge list
+ *                       (read-only view; k_copy_list retired)
  *   k_dst_count/scan2/scatter2 — fork-join radix split by vid % ndst
  *                       (replaces gpu_hash.cu:600-760)
  *   k_light_batch     — batched light-query window (one wavefront
@@ -37,8 +43,9 @@
  *                       const-start templates, binding table in LDS
  *   k_vu              — VERSATILE predicate-variable ops over the dense
  *                       vp CSR (sparql.hpp:556-744)
- *   k_expand_fn[_map] — functional-predicate dense-map k2u (deg==1
- *                       segments); _map = optimistic 1:1 inside graphs
+ *   k_fn_gather/k_fn_scatter — functional-predicate rank-compressed
+ *                       map k2u (deg==1 segments; page+value gathers);
+ *                       k_expand_fn_map = optimistic 1:1 inside graphs
  *   k_expand_opt / k_filter_opt — OPTIONAL-group ops (BLANK fill,
  *                       matched flags; sparql.hpp:100-170,316-375)
  *   per-type bitmaps  — rdf:type filters read 1 bit/vid (LLC-resident)
@@ -1190,16 +1197,6 @@ __global__ void k_expand_filter_big(const sid_t *__restrict__ tbl, int ncols,
 }
 
 // i2u / c2u: materialise an edge/index list as a 1-column table
-// (index_to_unknown sparql.hpp:194-231 / const_to_unknown :238-285)
-__global__ void k_copy_list(const sid_t *__restrict__ edges, uint64_t off,
-                            uint64_t n, uint64_t *__restrict__ d_stats,
-                            sid_t *__restrict__ out)
-{
-    count_bytes(d_stats, CAT_COPY, n * 8);
-    for (uint64_t t = (uint64_t)blockIdx.x * blockDim.x + threadIdx.x; t < n;
-         t += (uint64_t)gridDim.x * blockDim.x)
-        out[t] = edges[off + t];
-}
 
 // fork-join split (generate_sub_query sparql.hpp:772-796): dst = vid % ndst
 // Fork-join split as a radix partition (replaces the per-row
