@@ -1989,6 +1989,8 @@ struct wk_engine {
     int64_t bound = 0;   // host-side upper bound on rows (grid sizing)
 
     bool light = false;  // single-kernel fast path ran (h_pin self-published)
+    bool hinting = false;  // graph-build hint pass: per-PATTERN row counts
+                           // needed, so multi-step fusions are disabled
     int remote_step_idx = -1;  // pattern to run via the xGMI peer path
     // zero-copy i2u/c2u: the 1-col start table IS the (immutable) edge
     // list in the store — later steps read it in place instead of
@@ -3194,7 +3196,7 @@ static int32_t exec_pattern(wk_engine *e) {
         // positions, the writer re-walks and emits only those.  Skipped
         // when the capture hint says the filter drops nothing (the
         // verify-fused expansion above is cheaper there).
-        if (c_pg && e->st->nsrv == 1 && e->gs &&
+        if (c_pg && e->st->nsrv == 1 && e->gs && !e->hinting &&
             e->step + 1 < (int)e->pats.size()) {
             const wk_pattern_t &nx = e->pats[e->step + 1];
             const bool nodrop = e->capturing &&
@@ -3737,16 +3739,18 @@ extern "C" int32_t wk_engine_graph_build(wk_engine_t *e,
     {
         int32_t rc = wk_engine_begin_query(e, plan);
         if (rc) return rc;
+        e->hinting = true;  // per-pattern counts: no multi-step fusions
         int64_t prev = -1;
         while (e->step < (int)e->pats.size()) {
             const int at = e->step;
             int64_t n = 0;
             rc = wk_engine_execute_one_pattern(e, &n);
-            if (rc) return rc;
+            if (rc) { e->hinting = false; return rc; }
             if (n == prev && at < (int)e->capture_hint.size())
                 e->capture_hint[at] = 1;
             prev = n;
         }
+        e->hinting = false;
     }
     e->capturing = 1;
     hipGraph_t g = nullptr;
@@ -3817,15 +3821,17 @@ extern "C" int32_t wk_engine_graph_build_suite(wk_engine_t *e,
         hints[i].assign((size_t)std::max(plan->npatterns, 0), 0);
         int32_t rc = wk_engine_begin_query(e, plan);
         if (rc) return rc;
+        e->hinting = true;  // per-pattern counts: no multi-step fusions
         int64_t prev = -1;
         while (e->step < (int)e->pats.size()) {
             const int at = e->step;
             int64_t n = 0;
             rc = wk_engine_execute_one_pattern(e, &n);
-            if (rc) return rc;
+            if (rc) { e->hinting = false; return rc; }
             if (n == prev && at < (int)hints[i].size()) hints[i][at] = 1;
             prev = n;
         }
+        e->hinting = false;
     }
     e->capturing = 1;
     hipGraph_t g = nullptr;
