@@ -7,7 +7,9 @@ whole-job queries/sec.  N>1: launched by torch.distributed.run, one rank
 per GPU, store partitioned by vid % N, per-step RCCL all-to-allv.
 
 Env knobs: WK_UNIV (default 2560), WK_CPU_UNIV (cpu_baseline sample,
-default 256), WK_SKIP_CPU_BASELINE=1.
+default = the workload scale: SAME inputs), WK_SKIP_CPU_BASELINE=1,
+WK_SKIP_GATES=1, WK_EMU_EMBED (embedded emulator queries, 0 = off),
+WK_DIST_BACKEND=gloo (single-GPU rehearsal of the N>1 path).
 """
 import argparse
 import json
