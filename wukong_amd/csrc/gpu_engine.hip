@@ -3196,7 +3196,15 @@ static int32_t exec_pattern(wk_engine *e) {
         // positions, the writer re-walks and emits only those.  Skipped
         // when the capture hint says the filter drops nothing (the
         // verify-fused expansion above is cheaper there).
-        if (c_pg && e->st->nsrv == 1 && e->gs && !e->hinting &&
+        // skip hub-heavy segments: the count pass walks a row's whole
+        // edge list in one thread, so a 10^5-edge hub key would
+        // serialize (fine for LUBM/WatDiv leaf predicates, avg deg <=
+        // ~8; hub segments keep the classic split pipeline)
+        const size_t w_seg = (size_t)p * 2 + dir;
+        const bool tf_deg_ok =
+            w_seg < e->st->seg_keys.size() && e->st->seg_keys[w_seg] &&
+            e->st->seg_edges[w_seg] / e->st->seg_keys[w_seg] <= 32;
+        if (c_pg && tf_deg_ok && e->st->nsrv == 1 && e->gs && !e->hinting &&
             e->step + 1 < (int)e->pats.size()) {
             const wk_pattern_t &nx = e->pats[e->step + 1];
             const bool nodrop = e->capturing &&
