@@ -3,13 +3,20 @@ partitioned by vid % world exactly as the reference partitions by server
 (core/loader/base_loader.hpp:284, gstore.hpp:1050).
 
 Per pattern step whose start variable is not the one the rows are local
-by, the binding table is split by `vid % world` (generate_sub_query,
-core/engine/sparql.hpp:746-799) and exchanged with ONE all-to-allv —
-RCCL over xGMI on the nccl backend (torch.distributed plumbing), an
-object gather on gloo (CPU tests).  Every rank executes every step on
-its local store: probes of non-local keys miss by construction, so no
-owner special-casing is needed (index/const starts are naturally local,
-dispatch semantics of sparql.hpp:1064-1111).
+by, the driver decides on the GLOBAL row total (need_fork_join,
+sparql.hpp:802-814): at or above the threshold the binding table is
+split by `vid % world` (generate_sub_query, sparql.hpp:746-799) and
+exchanged with ONE all-to-allv — RCCL over xGMI on the nccl backend,
+an object gather on gloo (CPU tests); below it the rows stay put and
+k_peer_step probes the OWNER rank's store in place through HIP-IPC
+xGMI mappings (the one-sided-RDMA analog, gstore.hpp:260-338).
+Mid-plan const-/index-start filters broadcast the owner's edge list;
+filter steps between exchanges launch asynchronously; the final
+DISTINCT/OFFSET/LIMIT run once after the rank merge.  Every rank
+executes every step on its local store: probes of non-local keys miss
+by construction, so no owner special-casing is needed (index/const
+starts are naturally local, dispatch semantics of
+sparql.hpp:1064-1111).
 """
 import os
 import sys
