@@ -44,6 +44,20 @@ static void sort_triples(std::vector<triple> &t, bool by_s /* pso vs pos */,
                          uint32_t max_pid, uint32_t max_blk) {
     const uint64_t nb = (uint64_t)(max_pid + 1) * (max_blk + 2);
     const int64_t n = (int64_t)t.size();
+    // the bucket pass needs nthreads*nb counters — only worth it (and
+    // only affordable) when the (pid, vid-block) grid is small next to
+    // the input; tiny or id-sparse inputs take the direct sort.
+    if (nb > (uint64_t)n / 4 + 4096) {
+        auto full_pso = [](const triple &a, const triple &b) {
+            return a.p != b.p ? a.p < b.p : a.s != b.s ? a.s < b.s : a.o < b.o;
+        };
+        auto full_pos = [](const triple &a, const triple &b) {
+            return a.p != b.p ? a.p < b.p : a.o != b.o ? a.o < b.o : a.s < b.s;
+        };
+        if (by_s) std::sort(t.begin(), t.end(), full_pso);
+        else      std::sort(t.begin(), t.end(), full_pos);
+        return;
+    }
     auto bucket_of = [&](const triple &x) -> uint64_t {
         uint32_t major = by_s ? x.s : x.o;
         return (uint64_t)x.p * (max_blk + 2) + (major >> NBITS_IDX);
@@ -137,8 +151,8 @@ struct inserter {
 
 }  // namespace
 
-extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
-                                      int32_t sid, int32_t nsrv) {
+static wk_store_t *store_build_impl(const sid_t *spo, int64_t ntriples,
+                                    int32_t sid, int32_t nsrv) {
     if (!spo || ntriples < 0 || nsrv <= 0 || sid < 0 || sid >= nsrv) return nullptr;
     wk_store *st = new wk_store();
     st->sid = sid; st->nsrv = nsrv;
@@ -159,6 +173,17 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     if (max_pid >= (1u << NBITS_IDX)) { delete st; return nullptr; }
     st->max_pid = max_pid;
     uint32_t max_blk = max_id >> NBITS_IDX;
+    // The dense side indexes (type_of, VERSATILE vp-CSR) are sized by
+    // the vid SPAN, not the triple count.  The generators allocate vids
+    // contiguously from 1<<17 (lubm_gen.cpp:86), so span >> input only
+    // happens on hand-built input with ids parked near 2^32 — there the
+    // dense arrays would cost GBs for a handful of triples.  Skip the
+    // optional indexes in that case; the hash-probe fallback stays
+    // correct (fn/CSR/type-bitmap builds are already byte-budgeted).
+    const uint64_t vid_span = max_id >= (1u << NBITS_IDX)
+                                  ? (uint64_t)max_id + 1 - (1u << NBITS_IDX)
+                                  : 0;
+    const bool dense_vids = vid_span <= 16ull * (uint64_t)ntriples + (1u << 22);
 
     WK_LOG("[store] partition: %.1fs (pso=%zu pos=%zu)\n", now_s() - t0, pso.size(), pos.size());
     t0 = now_s();
@@ -354,7 +379,7 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     WK_LOG("[store] insert index: %.1fs (ext used %lu)\n", now_s() - t0,
            (unsigned long)(ext_next.load() - st->nbuckets_main));
     // dense type side-index from the TYPE_ID pso slice (runs per subject)
-    if (max_id >= (1u << NBITS_IDX) && max_pid < 0xFFFF) {
+    if (dense_vids && max_id >= (1u << NBITS_IDX) && max_pid < 0xFFFF) {
         st->type_base = 1u << NBITS_IDX;
         st->type_n = (uint64_t)max_id + 1 - st->type_base;
         if (!st->type_of.alloc(st->type_n, /*zero=*/true)) {
@@ -384,7 +409,7 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     {
         const char *vv = getenv("WK_VERSATILE");
         bool want = !(vv && !atoi(vv));
-        if (want && max_id >= (1u << NBITS_IDX)) {
+        if (want && dense_vids && max_id >= (1u << NBITS_IDX)) {
             st->vp_base = 1u << NBITS_IDX;
             st->vp_n = (uint64_t)max_id + 1 - st->vp_base;
             for (int dir = 0; dir < 2; dir++) {
@@ -457,7 +482,7 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     {
         const char *fv = getenv("WK_FN");
         bool want = !(fv && !atoi(fv));
-        if (want && max_id >= (1u << NBITS_IDX)) {
+        if (want && dense_vids && max_id >= (1u << NBITS_IDX)) {
             st->fn_base = 1u << NBITS_IDX;
             st->fn_n = (uint64_t)max_id + 1 - st->fn_base;
             st->fn.assign((size_t)NP * 2, {});
@@ -524,7 +549,7 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     {
         const char *cv = getenv("WK_CSR");
         bool want = !(cv && !atoi(cv));
-        if (want && max_id >= (1u << NBITS_IDX)) {
+        if (want && dense_vids && max_id >= (1u << NBITS_IDX)) {
             if (!st->fn_n) {  // WK_FN=0: span bookkeeping still needed
                 st->fn_base = 1u << NBITS_IDX;
                 st->fn_n = (uint64_t)max_id + 1 - st->fn_base;
@@ -652,6 +677,21 @@ extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
     return st;
 }
 
+// C-ABI entry: allocation failure surfaces as nullptr, never as an
+// exception crossing the boundary (the vector/sort scaffolding inside
+// the build can throw bad_alloc on hosts smaller than the dataset; the
+// pod_array side indexes already degrade in place).  The partially
+// built store is abandoned on this path — the process is at the OOM
+// edge and the caller's contract is "nullptr = build failed".
+extern "C" wk_store_t *wk_store_build(const sid_t *spo, int64_t ntriples,
+                                      int32_t sid, int32_t nsrv) {
+    try {
+        return store_build_impl(spo, ntriples, sid, nsrv);
+    } catch (const std::bad_alloc &) {
+        return nullptr;
+    }
+}
+
 extern "C" void wk_store_free(wk_store_t *st) { delete st; }
 
 extern "C" const sid_t *wk_store_get_triples(const wk_store_t *st, sid_t vid,
@@ -761,12 +801,28 @@ extern "C" uint64_t wk_store_check(const wk_store_t *st) {
 }
 
 extern "C" uint64_t wk_store_checksum(const wk_store_t *st) {
-    // FNV-1a over slots then edges
-    uint64_t h = 1469598103934665603ull;
-    auto mix = [&h](uint64_t x) {
-        for (int i = 0; i < 8; i++) { h ^= (x >> (8 * i)) & 0xff; h *= 1099511628211ull; }
-    };
-    for (const auto &v : st->vertices) { mix(v.key); mix(v.ptr); }
-    for (sid_t e : st->edges) mix(e);
-    return h;
+    // Content fingerprint: per occupied slot, FNV-1a over (key, edge
+    // list), combined with a commutative wrapping sum.  Bucket/chain
+    // placement is left OUT on purpose — parallel ext-bucket allocation
+    // makes the physical layout run-to-run nondeterministic, but two
+    // stores with the same logical triples must fingerprint equal.
+    std::atomic<uint64_t> total(0);
+    const int64_t nslots = (int64_t)st->vertices.size();
+#pragma omp parallel for schedule(static)
+    for (int64_t i = 0; i < nslots; i++) {
+        if (i % ASSOC == ASSOC - 1) continue;  // chain slot
+        const vertex_t &v = st->vertices[i];
+        if (v.key == KEY_EMPTY) continue;
+        uint64_t h = 1469598103934665603ull;
+        auto mix = [&h](uint64_t x) {
+            for (int b = 0; b < 8; b++) { h ^= (x >> (8 * b)) & 0xff; h *= 1099511628211ull; }
+        };
+        mix(v.key);
+        uint64_t off = ptr_off(v.ptr), sz = ptr_size(v.ptr);
+        mix(sz);
+        for (uint64_t k = 0; k < sz && off + k < st->edges.size(); k++)
+            mix(st->edges[off + k]);
+        total.fetch_add(h, std::memory_order_relaxed);
+    }
+    return total.load();
 }
