@@ -111,3 +111,33 @@ def test_singleton_and_absent():
     assert list(st.get_index(2, OUT)) == [o]
     assert st.get_index(5, IN).size == 0
     assert st.get_triples(s, 2, IN).size == 0  # wrong direction is empty
+
+
+def test_random_triples_vs_brute():
+    """Randomized roundtrip: every (s,p,OUT)/(o,p,IN) list and both
+    index sides must equal a brute-force dict over the same triples."""
+    rng = np.random.default_rng(123)
+    n = 4000
+    t = np.stack([
+        rng.integers(VBASE, VBASE + 300, size=n, dtype=np.uint32),
+        rng.integers(2, 25, size=n, dtype=np.uint32),
+        rng.integers(VBASE, VBASE + 300, size=n, dtype=np.uint32),
+    ], axis=1)
+    t = np.unique(t, axis=0)
+    st = wk.Store(t)
+    assert st.check() == 0
+    out_d, in_d, subj, obj = {}, {}, {}, {}
+    for s, p, o in t:
+        out_d.setdefault((int(s), int(p)), set()).add(int(o))
+        in_d.setdefault((int(o), int(p)), set()).add(int(s))
+        subj.setdefault(int(p), set()).add(int(s))
+        obj.setdefault(int(p), set()).add(int(o))
+    for (v, p), want in out_d.items():
+        assert list(_edges(st, v, p, OUT)) == sorted(want)
+    for (v, p), want in in_d.items():
+        assert list(_edges(st, v, p, IN)) == sorted(want)
+    for p in subj:
+        assert set(st.get_index(p, IN)) == subj[p]
+        assert set(st.get_index(p, OUT)) == obj[p]
+        assert st.seg_stats(p, OUT) == (len(subj[p]),
+                                        sum(len(out_d[k]) for k in out_d if k[1] == p))
