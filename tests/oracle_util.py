@@ -203,6 +203,20 @@ class OracleExecutor:
                             t.shape[0] if t.size else 0,
                             t.shape[1] if t.ndim == 2 else 0, v, step)
 
+    def rebind(self, plan_b, table, v2c, step):
+        """UNION-branch continuation (mirrors GpuExecutor.rebind): fresh
+        query handle on the extended plan, inherited table at step."""
+        self.plan = plan_b
+        self.npat = len(plan_b.patterns)
+        pats = _pats(plan_b)
+        self._keep = pats
+        if self._h:
+            lib().ok_query_free(self._h)
+        self._h = lib().ok_query_begin(self.ctx._h,
+                                       ctypes.cast(pats, ctypes.c_void_p),
+                                       self.npat, plan_b.nvars)
+        self.load(table, v2c, step)
+
     def get_triples(self, vid, pid, d):
         return self.ctx.get_triples(vid, pid, d)
 

@@ -36,6 +36,10 @@ def _worker(rank, world, port, results):
         # through the device wk_engine_execute_filter_list path
         from tests.test_dist_gloo import _modifier_plans
         plans.update(_modifier_plans(Q, wk))
+        # + UNION branch orchestration through GpuExecutor.rebind
+        # (begin_query + load_rbuf continuation on device)
+        from tests.test_dist_union import _union_plans
+        plans.update({f"u_{n}": p for n, p in _union_plans(Q, wk).items()})
         for name, plan in plans.items():
             ex = GpuExecutor(eng, plan)
             dq = DistQuery(ex, plan, rank, world)
@@ -124,10 +128,12 @@ def test_gpu_dist_two_ranks_equal_oracle():
     import wukong_amd as wk
     from wukong_amd import queries as Q
     from tests.test_dist_gloo import _modifier_plans
+    from tests.test_dist_union import _union_plans
     from tests.oracle_util import OracleCtx, sort_rows
     full = OracleCtx(wk.lubm_gen(2, seed=42))
     plans = dict((n, Q.ALL[n]) for n in QUERIES)
     plans.update(_modifier_plans(Q, wk))
+    plans.update({f"u_{n}": p for n, p in _union_plans(Q, wk).items()})
     for name, plan in plans.items():
         want = sort_rows(full.run_query(plan))
         assert got[name].shape == want.shape, (name, got[name].shape, want.shape)

@@ -112,7 +112,7 @@ class DistQuery:
         dist.all_reduce(t)
         return int(t.item())
 
-    def _try_remote(self, i, pat):
+    def _try_remote(self, i, pat, states):
         """Sub-threshold tables probe the owner rank's store in place
         (xGMI peer mappings; oracle peers in CPU tests) instead of
         exchanging — need_fork_join, sparql.hpp:802-814."""
@@ -120,8 +120,8 @@ class DistQuery:
             return False
         if self._global_rows() >= self.threshold:
             return False
-        v2c_prev, _ = self.states[i - 1]
-        v2c_next, _ = self.states[i]
+        v2c_prev, _ = states[i - 1]
+        v2c_next, _ = states[i]
         try:
             self.ex.step_remote(i, pat, v2c_prev, v2c_next)
             return True
@@ -164,10 +164,61 @@ class DistQuery:
         return recv, sum(out_list) // ncols if ncols else 0
 
     def run(self):
-        ex, plan = self.ex, self.plan
-        local_var = None
+        plan = self.plan
+        if getattr(plan, "optional", []):
+            # OPTIONAL needs per-row matched-flag state across steps
+            # (opt_mode, sparql.hpp:1603-1662) which the step-level
+            # interface does not carry — run such plans single-GPU
+            # (wk_engine_run_query executes them on device).
+            raise NotImplementedError(
+                "OPTIONAL groups are not distributed; use the "
+                "single-GPU run_query path")
+        local_var = self._run_steps(plan, self.states, 0, None)
+        unions = getattr(plan, "unions", [])
+        if unions:
+            self._run_unions(unions, local_var)
+
+    def _run_unions(self, unions, local_var):
+        """UNION branches, distributed: each branch inherits the merged
+        main-BGP table (snapshot/restore) and runs through the SAME
+        step machinery — exchanges, remote reads and broadcast filters
+        included — then the branch outputs concatenate
+        (execute_sparql_query, sparql.hpp:1564-1601; merge semantics
+        rmap.hpp:57-87, first branch's column layout as in the oracle).
+        Final ops run once after the rank merge, in gather_result."""
+        from . import Plan
+        plan = self.plan
+        nmain = len(plan.patterns)
+        base_v2c, _ = self.states[-1]
+        self.ex.rows()  # resolve any async filter step before snapshot
+        snapshot = self.ex.table()  # host copy of this rank's shard
+        parts, final_v2c = [], None
+        for branch in unions:
+            nv = plan.nvars
+            for (s, p, d, o) in branch:
+                for v in (s, o):
+                    if v < 0:
+                        nv = max(nv, -v)
+            plan_b = Plan(list(plan.patterns) + list(branch), nv,
+                          plan.required_vars)
+            states_b = plan_v2c_states(plan_b)
+            # pad the inherited v2c to the branch's nvars (var -k lives
+            # at index k-1, so new branch vars extend at the tail)
+            v2c_b = list(base_v2c) + [-1] * (nv - plan.nvars)
+            self.ex.rebind(plan_b, snapshot, v2c_b, nmain)
+            self._run_steps(plan_b, states_b, nmain, local_var)
+            self.ex.rows()  # resolve a trailing async filter step
+            parts.append(self.ex.table())
+            if final_v2c is None:
+                final_v2c = states_b[-1][0]
+        self._union_parts = parts
+        self._union_v2c = final_v2c
+
+    def _run_steps(self, plan, states, start, local_var):
+        ex = self.ex
         nccl = dist.is_initialized() and dist.get_backend() == "nccl"
-        for i, (s, p, d, o) in enumerate(plan.patterns):
+        for i in range(start, len(plan.patterns)):
+            s, p, d, o = plan.patterns[i]
             if i > 0 and s >= 0:
                 # mid-plan const-/index-start membership filter: the
                 # start's edge list lives only on its owner rank
@@ -182,15 +233,15 @@ class DistQuery:
                 dist.all_gather_object(
                     gathered, np.asarray(local, dtype=np.uint32))
                 merged = np.unique(np.concatenate(gathered))  # sorted
-                v2c_prev, _ = self.states[i - 1]
+                v2c_prev, _ = states[i - 1]
                 ex.filter_with_list(merged, i, v2c_prev,
                                     v2c_prev[-(o + 1)])
                 continue
             if i > 0 and s < 0 and local_var != s:
-                if self._try_remote(i, (s, p, d, o)):
+                if self._try_remote(i, (s, p, d, o), states):
                     continue  # rows stayed put; owner stores were probed
                 # fork-join exchange (need_fork_join, sparql.hpp:802-814)
-                v2c, ncols = self.states[i - 1]
+                v2c, ncols = states[i - 1]
                 col = v2c[-(s + 1)]
                 if nccl:
                     recv, nrows = self._exchange_nccl(ex.engine, ncols, col)
@@ -207,6 +258,7 @@ class DistQuery:
                 # i2u sets local_var=end (sparql.hpp:230); c2u leaves it
                 # unset (rows are edge values, arbitrary ranks)
                 local_var = o if (s >= 0 and _is_tpid(s)) else None
+        return local_var
 
     def gather_result(self):
         """Final table: gather rank results (rmap merge semantics,
@@ -215,6 +267,19 @@ class DistQuery:
         DISTINCT would keep cross-rank duplicates; per-rank LIMIT/OFFSET
         would return world*limit / drop world*offset rows."""
         p = self.plan
+        if getattr(self, "_union_parts", None) is not None:
+            # union plans: branch outputs already concatenate host-side
+            # (first branch's column layout, as in the oracle); every
+            # final op — including plain projection — runs on the merge
+            raw = (np.concatenate([t for t in self._union_parts if t.size],
+                                  axis=0)
+                   if any(t.size for t in self._union_parts)
+                   else self._union_parts[0])
+            gathered = [None] * self.world
+            dist.all_gather_object(gathered, raw)
+            full = np.concatenate([g for g in gathered if g.size] or [raw],
+                                  axis=0)
+            return final_process(full, self._union_v2c, p)
         if not (p.distinct or p.limit >= 0 or p.offset > 0):
             part = self.ex.finalize()  # projection only — order-free
             gathered = [None] * self.world
@@ -294,6 +359,17 @@ class GpuExecutor:
         self.engine.load_rbuf(table, v2c, step)
         self.engine.last_rows = len(table)
         self._stale = False
+
+    def rebind(self, plan_b, table, v2c, step):
+        """Restart the engine on a continuation plan with an inherited
+        table — the UNION-branch snapshot/restore (run_union semantics,
+        sparql.hpp:1564-1601, via the sub-query continuation interface
+        begin_query + load_rbuf at step>0)."""
+        self.plan = plan_b
+        if hasattr(self, "_states"):
+            del self._states
+        self.engine.begin_query(plan_b)
+        self.load(table, v2c, step)
 
     def step(self):
         # filter steps (known/const end: outputs <= inputs, no overflow
