@@ -40,6 +40,12 @@ def _worker(rank, world, port, results):
         # (begin_query + load_rbuf continuation on device)
         from tests.test_dist_union import _union_plans
         plans.update({f"u_{n}": p for n, p in _union_plans(Q, wk).items()})
+        # + VERSATILE (k_vu over the partitioned vp CSR; owner-local
+        # const starts, exchange-then-local known starts)
+        from tests.oracle_util import OracleCtx
+        vfull = OracleCtx(wk.lubm_gen(2, seed=42))
+        plans.update({f"v_{n}": p
+                      for n, p in Q.versatile_plans(vfull).items()})
         for name, plan in plans.items():
             ex = GpuExecutor(eng, plan)
             dq = DistQuery(ex, plan, rank, world)
@@ -134,6 +140,7 @@ def test_gpu_dist_two_ranks_equal_oracle():
     plans = dict((n, Q.ALL[n]) for n in QUERIES)
     plans.update(_modifier_plans(Q, wk))
     plans.update({f"u_{n}": p for n, p in _union_plans(Q, wk).items()})
+    plans.update({f"v_{n}": p for n, p in Q.versatile_plans(full).items()})
     for name, plan in plans.items():
         want = sort_rows(full.run_query(plan))
         assert got[name].shape == want.shape, (name, got[name].shape, want.shape)

@@ -76,7 +76,17 @@ def plan_v2c_states(plan):
     for i, (s, p, d, o) in enumerate(plan.patterns):
         def known(v):
             return v < 0 and v2c[-(v + 1)] >= 0
-        if i == 0 and s >= 0 and _is_tpid(s):
+        if p < 0:
+            # VERSATILE: the predicate var binds a new column, an
+            # unknown object var the one after (sparql.hpp:556-744
+            # *_unknown_unknown / *_unknown_const column order)
+            v2c[-(p + 1)] = col
+            if o < 0:
+                v2c[-(o + 1)] = col + 1
+                col += 2
+            else:
+                col += 1
+        elif i == 0 and s >= 0 and _is_tpid(s):
             v2c[-(o + 1)] = 0
             col = 1
         elif s >= 0 and o < 0 and not known(o):
@@ -219,6 +229,13 @@ class DistQuery:
         nccl = dist.is_initialized() and dist.get_backend() == "nccl"
         for i in range(start, len(plan.patterns)):
             s, p, d, o = plan.patterns[i]
+            if i > 0 and s >= 0 and p < 0:
+                # mid-plan const predicate-variable: the engine (and the
+                # reference's dispatcher) only accept const_unknown_* as
+                # the FIRST pattern (sparql.hpp:719) — it regenerates
+                # the table from the constant, discarding prior bindings
+                raise ValueError(
+                    "const predicate-variable pattern must be first")
             if i > 0 and s >= 0:
                 # mid-plan const-/index-start membership filter: the
                 # start's edge list lives only on its owner rank
@@ -384,7 +401,9 @@ class GpuExecutor:
                 self._states = plan_v2c_states(self.plan)
             s, p, d, o = pats[i]
             known = o >= 0 or self._states[i - 1][0][-(o + 1)] >= 0
-            if s < 0 and known:
+            # p<0 (VERSATILE) can GROW the table even with a known end
+            # (one row per matching predicate) — never overflow-safe
+            if s < 0 and p >= 1 and known:
                 self.engine.execute_one_pattern_async()
                 self._stale = True
                 return None
