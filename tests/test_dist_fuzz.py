@@ -11,16 +11,21 @@ import pytest
 import torch.multiprocessing as mp
 
 import wukong_amd as wk
-from wukong_amd import queries as Q
-from tests.test_fuzz_plans import random_plan
+from wukong_amd import Plan, queries as Q
+from tests.test_fuzz_plans import random_plan, PREDS, TYPES
 from tests.oracle_util import OracleCtx, sort_rows
 
 N_PLANS = 25
+N_VU = 8      # plans containing a predicate variable (distributed vu)
+N_UNION = 8   # plans with random UNION branches
 
 
 def _gen_plans():
-    """Deterministic plan list (no predicate variables: the per-pattern
-    distributed driver covers the BGP surface)."""
+    """Deterministic plan list: plain BGPs, BGPs with a predicate
+    variable (the distributed vu path), and BGPs with two random
+    layout-compatible UNION branches (expansion branches share the
+    start/new var so every branch appends the same column; filter
+    branches append none)."""
     store = wk.Store(wk.lubm_gen(2, seed=42))
     rng = random.Random(4321)
     plans = []
@@ -28,6 +33,34 @@ def _gen_plans():
         p = random_plan(rng, store)
         if all(pp[1] >= 1 for pp in p.patterns):
             plans.append(p)
+    rng2 = random.Random(8765)
+    while len(plans) < N_PLANS + N_VU:
+        p = random_plan(rng2, store)
+        # vu steps are always known-start in random_plan (mid-plan
+        # const vu is refused by driver and engine alike)
+        if any(pp[1] < 0 for pp in p.patterns):
+            plans.append(p)
+    rng3 = random.Random(1357)
+    n_uni = 0
+    while n_uni < N_UNION:
+        base = random_plan(rng3, store)
+        if not all(pp[1] >= 1 for pp in base.patterns):
+            continue
+        bound = list(base.required_vars)
+        nv = base.nvars
+        if rng3.random() < 0.5:
+            s = rng3.choice(bound)
+            ovar = -(nv + 1)
+            branches = [[(s, rng3.choice(PREDS), rng3.choice([0, 1]), ovar)]
+                        for _ in range(2)]
+            plan = Plan(list(base.patterns), nv + 1, bound + [ovar],
+                        unions=branches)
+        else:
+            branches = [[(rng3.choice(bound), Q.TYPE_ID, wk.DIR_OUT,
+                          rng3.choice(TYPES))] for _ in range(2)]
+            plan = Plan(list(base.patterns), nv, bound, unions=branches)
+        plans.append(plan)
+        n_uni += 1
     return plans
 
 

@@ -72,7 +72,7 @@ def _worker(rank, world, port, results):
 def test_dist_union_two_ranks_equal_single():
     ctx = mp.get_context("spawn")
     results = ctx.Queue()
-    port = 29876
+    port = 29878
     procs = [ctx.Process(target=_worker, args=(r, 2, port, results))
              for r in range(2)]
     for p in procs:
