@@ -36,7 +36,13 @@ def _union_plans(Q, wk):
                     unions=[[(X, Q.MEMBEROF, wk.DIR_OUT, Y),
                              (Y, Q.TYPE_ID, wk.DIR_OUT, Q.GRADSTUDENT)],
                             [(X, Q.UGDEGREE, wk.DIR_OUT, Y)]])
-    return {"basic": basic, "exch": exch, "dlo": dlo, "empty": empty}
+    # predicate-variable branches (vu inside a union: both bind the
+    # same pred/obj vars, so the branch layouts agree)
+    vu = wk.Plan([(Q.GRADSTUDENT, Q.TYPE_ID, wk.DIR_IN, X)], 3, [X, Y, Z],
+                 unions=[[(X, Y, wk.DIR_OUT, Z)],
+                         [(X, Y, wk.DIR_IN, Z)]])
+    return {"basic": basic, "exch": exch, "dlo": dlo, "empty": empty,
+            "vu": vu}
 
 
 def _worker(rank, world, port, results):

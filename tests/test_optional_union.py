@@ -152,3 +152,34 @@ def test_union_then_optional_oracle(lubm4, oracle4):
     want = np.array(rows, dtype=np.uint32)
     assert got.shape == want.shape, (got.shape, want.shape)
     assert np.array_equal(sort_rows(got), sort_rows(want))
+
+
+def test_union_vu_oracle_vs_numpy(lubm4, oracle4):
+    """UNION branches carrying a predicate VARIABLE (vu inside the
+    union machinery, sparql.hpp:556-744 under :1564-1601): equals the
+    concat of the two plain vu continuations."""
+    plan = Plan([(Q.GRADSTUDENT, Q.TYPE_ID, wk.DIR_IN, X)], 3, [X, Y, Z],
+                unions=[[(X, Y, wk.DIR_OUT, Z)],
+                        [(X, Y, wk.DIR_IN, Z)]])
+    got = oracle4.run_query(plan)
+    a = oracle4.run_query(Plan([(Q.GRADSTUDENT, Q.TYPE_ID, wk.DIR_IN, X),
+                                (X, Y, wk.DIR_OUT, Z)], 3, [X, Y, Z]))
+    b = oracle4.run_query(Plan([(Q.GRADSTUDENT, Q.TYPE_ID, wk.DIR_IN, X),
+                                (X, Y, wk.DIR_IN, Z)], 3, [X, Y, Z]))
+    want = np.vstack([a, b])
+    assert len(want) > 0
+    assert got.shape == want.shape
+    assert np.array_equal(sort_rows(got), sort_rows(want))
+
+
+@pytest.mark.gpu
+def test_gpu_union_vu_parity(store4, oracle4):
+    """k_vu launched from inside run_union on device == oracle."""
+    plan = Plan([(Q.GRADSTUDENT, Q.TYPE_ID, wk.DIR_IN, X)], 3, [X, Y, Z],
+                unions=[[(X, Y, wk.DIR_OUT, Z)],
+                        [(X, Y, wk.DIR_IN, Z)]])
+    eng = wk.Engine(store4, device=0)
+    got = eng.run_query(plan)
+    want = oracle4.run_query(plan)
+    assert got.shape == want.shape, (got.shape, want.shape)
+    assert np.array_equal(sort_rows(got), sort_rows(want))
