@@ -110,3 +110,101 @@ def test_dist_fuzz_two_ranks():
             assert g.shape == want.shape, (mode, j, g.shape, want.shape,
                                            plan.patterns)
             assert np.array_equal(g, want), (mode, j, plan.patterns)
+
+
+def _gen_watdiv_plans():
+    """WatDiv-schema random WALKS (hubs, chains, genre fan-outs) —
+    schema-aware so most plans are non-empty: each hop follows an edge
+    the generator actually emits (watdiv_gen.cpp), tracking the bound
+    var's type."""
+    from wukong_amd import watdiv as W
+    # type -> [(pred, dir, result_type)] per the generator's schema
+    EDGES = {
+        W.T_PRODUCT: [(W.HASGENRE, wk.DIR_OUT, W.T_GENRE),
+                      (W.OFFER_PRODUCT, wk.DIR_IN, W.T_OFFER),
+                      (W.REVIEW_PRODUCT, wk.DIR_IN, W.T_REVIEW),
+                      (W.PURCHASED, wk.DIR_IN, W.T_USER)],
+        W.T_OFFER: [(W.OFFER_PRODUCT, wk.DIR_OUT, W.T_PRODUCT),
+                    (W.RETAILER, wk.DIR_OUT, W.T_RETAILER)],
+        W.T_REVIEW: [(W.REVIEW_PRODUCT, wk.DIR_OUT, W.T_PRODUCT),
+                     (W.REVIEWER, wk.DIR_OUT, W.T_USER)],
+        W.T_USER: [(W.PURCHASED, wk.DIR_OUT, W.T_PRODUCT),
+                   (W.FRIEND, wk.DIR_OUT, W.T_USER),
+                   (W.FRIEND, wk.DIR_IN, W.T_USER),
+                   (W.REVIEWER, wk.DIR_IN, W.T_REVIEW)],
+        W.T_GENRE: [(W.HASGENRE, wk.DIR_IN, W.T_PRODUCT)],
+        W.T_RETAILER: [(W.RETAILER, wk.DIR_IN, W.T_OFFER)],
+    }
+    rng = random.Random(97)
+    plans = []
+    for _ in range(15):
+        nv = rng.randint(2, 4)
+        vars_ = [-(i + 1) for i in range(nv)]
+        t = rng.choice(list(EDGES))
+        pats = [(t, Q.TYPE_ID, wk.DIR_IN, vars_[0])]
+        bound = {vars_[0]: t}
+        free = vars_[1:]
+        for _ in range(rng.randint(1, 3)):
+            s = rng.choice(list(bound))
+            pred, d, rt = rng.choice(EDGES[bound[s]])
+            if rng.random() < 0.75 and free:
+                o = free.pop(0)
+                pats.append((s, pred, d, o))
+                bound[o] = rt
+            else:  # matching typeof filter (kept) or a wrong one (empties)
+                wrong = rng.random() < 0.15
+                ft = rng.choice(list(EDGES)) if wrong else bound[s]
+                pats.append((s, Q.TYPE_ID, wk.DIR_OUT, ft))
+        plans.append(Plan(pats, nvars=nv, required_vars=list(bound)))
+    return plans
+
+
+def _worker_watdiv(rank, world, port, results):
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch.distributed as dist
+    from wukong_amd.dist import DistQuery
+    from tests.oracle_util import OracleExecutor
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        ctxs = [OracleCtx(wk.watdiv_gen(2000, seed=7, sid=r, nsrv=world),
+                          sid=r, nsrv=world) for r in range(world)]
+        out = {}
+        for j, plan in enumerate(_gen_watdiv_plans()):
+            for mode, thr in (("x", 0), ("r", 10**9)):
+                ex = OracleExecutor(ctxs[rank], plan, peers=ctxs)
+                dq = DistQuery(ex, plan, rank, world, threshold=thr)
+                dq.run()
+                out[f"{mode}:{j}"] = sort_rows(dq.gather_result())
+        if rank == 0:
+            results.put(out)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(900)
+def test_dist_fuzz_watdiv_two_ranks():
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_worker_watdiv, args=(r, 2, 29879, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=840)
+    for p in procs:
+        p.join(timeout=60)
+
+    full = OracleCtx(wk.watdiv_gen(2000, seed=7))
+    nonempty = 0
+    for j, plan in enumerate(_gen_watdiv_plans()):
+        want = sort_rows(full.run_query(plan))
+        nonempty += bool(len(want))
+        for mode in ("x", "r"):
+            g = got[f"{mode}:{j}"]
+            assert g.shape == want.shape, (mode, j, g.shape, want.shape,
+                                           plan.patterns)
+            assert np.array_equal(g, want), (mode, j, plan.patterns)
+    assert nonempty >= 8
