@@ -114,3 +114,62 @@ def test_dist_optional_refuses():
     dq = DistQuery(None, plan, 0, 1)
     with pytest.raises(NotImplementedError):
         dq.run()
+
+
+def _worker4(rank, world, port, results):
+    """World-4: union + versatile plans through the exchange path (the
+    4/8-GPU bench shape) — owner math must hold beyond 2 ranks."""
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch.distributed as dist
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from wukong_amd.dist import DistQuery
+    from tests.oracle_util import OracleCtx, OracleExecutor, sort_rows
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        full = OracleCtx(wk.lubm_gen(2, seed=42))
+        plans = dict(_union_plans(Q, wk))
+        plans.update({f"v_{n}": p
+                      for n, p in Q.versatile_plans(full).items()})
+        ctx = OracleCtx(wk.lubm_gen(2, seed=42, sid=rank, nsrv=world),
+                        sid=rank, nsrv=world)
+        out = {}
+        for name, plan in plans.items():
+            ex = OracleExecutor(ctx, plan)
+            dq = DistQuery(ex, plan, rank, world)
+            dq.run()
+            out[name] = sort_rows(dq.gather_result())
+        if rank == 0:
+            results.put(out)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_dist_union_versatile_four_ranks_equal_single():
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    port = 29880
+    procs = [ctx.Process(target=_worker4, args=(r, 4, port, results))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=540)
+    for p in procs:
+        p.join(timeout=60)
+
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from tests.oracle_util import OracleCtx, sort_rows
+    full = OracleCtx(wk.lubm_gen(2, seed=42))
+    plans = dict(_union_plans(Q, wk))
+    plans.update({f"v_{n}": p for n, p in Q.versatile_plans(full).items()})
+    for name, plan in plans.items():
+        want = sort_rows(full.run_query(plan))
+        assert got[name].shape == want.shape, (name, got[name].shape,
+                                               want.shape)
+        assert np.array_equal(got[name], want), name
