@@ -46,6 +46,10 @@ def _worker(rank, world, port, results):
         vfull = OracleCtx(wk.lubm_gen(2, seed=42))
         plans.update({f"v_{n}": p
                       for n, p in Q.versatile_plans(vfull).items()})
+        # + OPTIONAL groups (host-side matched-flag restatement over
+        # the device executor's table/get_triples)
+        from tests.test_dist_optional import _opt_plans
+        plans.update({f"o_{n}": p for n, p in _opt_plans(Q, wk).items()})
         for name, plan in plans.items():
             ex = GpuExecutor(eng, plan)
             dq = DistQuery(ex, plan, rank, world)
@@ -141,6 +145,8 @@ def test_gpu_dist_two_ranks_equal_oracle():
     plans.update(_modifier_plans(Q, wk))
     plans.update({f"u_{n}": p for n, p in _union_plans(Q, wk).items()})
     plans.update({f"v_{n}": p for n, p in Q.versatile_plans(full).items()})
+    from tests.test_dist_optional import _opt_plans
+    plans.update({f"o_{n}": p for n, p in _opt_plans(Q, wk).items()})
     for name, plan in plans.items():
         want = sort_rows(full.run_query(plan))
         assert got[name].shape == want.shape, (name, got[name].shape, want.shape)

@@ -100,22 +100,6 @@ def test_dist_union_two_ranks_equal_single():
             assert np.array_equal(g, want), (mode, name)
 
 
-def test_dist_optional_refuses():
-    """OPTIONAL groups need per-row matched-flag state the step-level
-    interface does not carry — the driver must refuse, not silently
-    drop the group (the guard fires before any collective, so no
-    process group is needed)."""
-    import wukong_amd as wk
-    from wukong_amd import queries as Q
-    from wukong_amd.dist import DistQuery
-    X, Y = -1, -2
-    plan = wk.Plan([(Q.UGSTUDENT, Q.TYPE_ID, wk.DIR_IN, X)], 2, [X, Y],
-                   optional=[(X, Q.ADVISOR, wk.DIR_OUT, Y)])
-    dq = DistQuery(None, plan, 0, 1)
-    with pytest.raises(NotImplementedError):
-        dq.run()
-
-
 def _worker4(rank, world, port, results):
     """World-4: union + versatile plans through the exchange path (the
     4/8-GPU bench shape) — owner math must hold beyond 2 ranks."""

@@ -175,18 +175,13 @@ class DistQuery:
 
     def run(self):
         plan = self.plan
-        if getattr(plan, "optional", []):
-            # OPTIONAL needs per-row matched-flag state across steps
-            # (opt_mode, sparql.hpp:1603-1662) which the step-level
-            # interface does not carry — run such plans single-GPU
-            # (wk_engine_run_query executes them on device).
-            raise NotImplementedError(
-                "OPTIONAL groups are not distributed; use the "
-                "single-GPU run_query path")
         local_var = self._run_steps(plan, self.states, 0, None)
         unions = getattr(plan, "unions", [])
         if unions:
             self._run_unions(unions, local_var)
+        optional = getattr(plan, "optional", [])
+        if optional:
+            self._run_optional(optional)
 
     def _run_unions(self, unions, local_var):
         """UNION branches, distributed: each branch inherits the merged
@@ -223,6 +218,121 @@ class DistQuery:
                 final_v2c = states_b[-1][0]
         self._union_parts = parts
         self._union_v2c = final_v2c
+
+    BLANK = 0xFFFFFFFF
+
+    def _exchange_rows_cpu(self, table, matched, col):
+        """Host-side exchange of (rows, matched flags) by owner of
+        row[col]; rows whose key is BLANK stay put (they never probe)."""
+        w = self.world
+        key_ok = table[:, col] != np.uint32(self.BLANK)
+        stay_t, stay_m = table[~key_ok], matched[~key_ok]
+        chunks = []
+        for dst in range(w):
+            sel = key_ok & (table[:, col] % w == dst)
+            chunks.append((table[sel], matched[sel]))
+        gathered = [None] * w
+        dist.all_gather_object(gathered, chunks)
+        ts = [stay_t] + [g[self.rank][0] for g in gathered]
+        ms = [stay_m] + [g[self.rank][1] for g in gathered]
+        return (np.concatenate(ts, axis=0), np.concatenate(ms))
+
+    def _run_optional(self, opt_pats):
+        """OPTIONAL group, distributed: a HOST-side restatement of the
+        reference's matched-flag mechanics (opt_mode row semantics,
+        sparql.hpp:100-170,316-375,416-549 under :1603-1662) — the
+        post-pattern passes run on the pruned table, so
+        per-row work is host-side; owner-correct probes come from
+        exchanging rows to the key's owner rank before each step and
+        broadcasting const-start edge lists.  Columns born inside the
+        group (opt_mask) blank out when a later step unmatches the row;
+        a matched degree-0 row keeps its flag (the reference's
+        left-join quirk)."""
+        ex, plan = self.ex, self.plan
+        if getattr(self, "_union_parts", None) is not None:
+            parts = [t for t in self._union_parts if t.size]
+            T = (np.concatenate(parts, axis=0) if parts
+                 else self._union_parts[0])
+            v2c = list(self._union_v2c)
+        else:
+            ex.rows()
+            T = np.asarray(ex.table())
+            v2c = list(self.states[-1][0])
+        ncols = T.shape[1] if T.ndim == 2 else 0
+        T = T.reshape(-1, max(ncols, 1)).astype(np.uint32, copy=True)
+        matched = np.ones(len(T), dtype=bool)
+        opt_cols = []
+        B = np.uint32(self.BLANK)
+
+        def blank_rows(mask):
+            for c in opt_cols:
+                T[mask, c] = B
+
+        for (s, p, d, o) in opt_pats:
+            if p < 1 or (p == 1 and d == 0):
+                # predicate variables and type-member expansion (the
+                # [0|tid|IN] index is spread over every rank) don't fit
+                # the row-owner exchange; no reference plan uses them
+                # inside OPTIONAL
+                raise ValueError(
+                    "unsupported pattern shape inside a distributed "
+                    "OPTIONAL group")
+            if s >= 0:
+                # const_to_known under OPTIONAL: broadcast the owner's
+                # edge list, then blank+unmatch rows outside it
+                local = (ex.get_index(s, d) if _is_tpid(s)
+                         else ex.get_triples(s, p, d))
+                gathered = [None] * self.world
+                dist.all_gather_object(
+                    gathered, np.asarray(local, dtype=np.uint32))
+                merged = np.unique(np.concatenate(gathered))
+                col = v2c[-(o + 1)]
+                ok = (T[:, col] != B) & np.isin(T[:, col], merged)
+                drop = ~ok
+                blank_rows(drop & matched)
+                matched &= ok
+                continue
+            col = v2c[-(s + 1)]
+            T, matched = self._exchange_rows_cpu(T, matched, col)
+            ostat = 2 if o >= 0 else (1 if v2c[-(o + 1)] >= 0 else 0)
+            if ostat == 0:
+                rows, flags = [], []
+                for i in range(len(T)):
+                    cur = T[i, col]
+                    if not matched[i] or cur == B:
+                        rows.append(np.append(T[i], B))
+                        flags.append(matched[i])
+                        continue
+                    edges = ex.get_triples(int(cur), p, d)
+                    if len(edges) == 0:
+                        rows.append(np.append(T[i], B))
+                        flags.append(True)  # deg-0 keeps the flag
+                    else:
+                        for e_ in edges:
+                            rows.append(np.append(T[i], e_))
+                            flags.append(True)
+                T = (np.array(rows, dtype=np.uint32) if rows
+                     else np.empty((0, T.shape[1] + 1), dtype=np.uint32))
+                matched = np.array(flags, dtype=bool)
+                v2c[-(o + 1)] = ncols
+                opt_cols.append(ncols)
+                ncols += 1
+            else:
+                for i in range(len(T)):
+                    cur = T[i, col]
+                    ok = False
+                    if cur != B:
+                        edges = ex.get_triples(int(cur), p, d)
+                        tgt = (np.uint32(o) if ostat == 2
+                               else T[i, v2c[-(o + 1)]])
+                        ok = tgt != B and bool(np.isin(tgt, edges))
+                    if not ok:
+                        if matched[i]:
+                            for c in opt_cols:
+                                T[i, c] = B
+                        matched[i] = False
+        self._opt_table = T
+        self._opt_v2c = v2c
 
     def _run_steps(self, plan, states, start, local_var):
         ex = self.ex
@@ -284,6 +394,14 @@ class DistQuery:
         DISTINCT would keep cross-rank duplicates; per-rank LIMIT/OFFSET
         would return world*limit / drop world*offset rows."""
         p = self.plan
+        if getattr(self, "_opt_table", None) is not None:
+            # optional ran host-side (after any union merge): the final
+            # table and its v2c live on the driver
+            gathered = [None] * self.world
+            dist.all_gather_object(gathered, self._opt_table)
+            full = np.concatenate(
+                [g for g in gathered if g.size] or [self._opt_table], axis=0)
+            return final_process(full, self._opt_v2c, p)
         if getattr(self, "_union_parts", None) is not None:
             # union plans: branch outputs already concatenate host-side
             # (first branch's column layout, as in the oracle); every
