@@ -18,6 +18,7 @@ from tests.oracle_util import OracleCtx, sort_rows
 N_PLANS = 25
 N_VU = 8      # plans containing a predicate variable (distributed vu)
 N_UNION = 8   # plans with random UNION branches
+N_OPT = 8     # plans with a random OPTIONAL group
 
 
 def _gen_plans():
@@ -61,6 +62,28 @@ def _gen_plans():
             plan = Plan(list(base.patterns), nv, bound, unions=branches)
         plans.append(plan)
         n_uni += 1
+    rng4 = random.Random(2468)
+    n_opt = 0
+    while n_opt < N_OPT:
+        base = random_plan(rng4, store)
+        if not all(pp[1] >= 1 for pp in base.patterns):
+            continue
+        bound = list(base.required_vars)
+        nv = base.nvars
+        s = rng4.choice(bound)
+        ovar = -(nv + 1)
+        group = [(s, rng4.choice(PREDS), rng4.choice([0, 1]), ovar)]
+        r = rng4.random()
+        if r < 0.4:    # typeof filter on the optional-born column
+            group.append((ovar, Q.TYPE_ID, wk.DIR_OUT, rng4.choice(TYPES)))
+        elif r < 0.6:  # k2k back onto a base column
+            other = rng4.choice(bound)
+            group.append((ovar, rng4.choice(PREDS), rng4.choice([0, 1]),
+                          other))
+        plan = Plan(list(base.patterns), nv + 1, bound + [ovar],
+                    optional=group)
+        plans.append(plan)
+        n_opt += 1
     return plans
 
 

@@ -33,8 +33,12 @@ def _opt_plans(Q, wk):
                  unions=[[(X, Q.MEMBEROF, wk.DIR_OUT, Y)],
                          [(X, Q.UGDEGREE, wk.DIR_OUT, Y)]],
                  optional=[(X, Q.ADVISOR, wk.DIR_OUT, Z)])
+    # final ops over the optional result (once, after the rank merge)
+    dlo = wk.Plan([(Q.UGSTUDENT, Q.TYPE_ID, wk.DIR_IN, X)], 2, [Y],
+                  distinct=True, limit=9, offset=4,
+                  optional=[(X, Q.ADVISOR, wk.DIR_OUT, Y)])
     return {"k2u": k2u, "k2c": k2c, "k2k": k2k, "cfilter": cfilter,
-            "uo": uo}
+            "uo": uo, "dlo": dlo}
 
 
 def _worker(rank, world, port, results):

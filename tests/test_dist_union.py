@@ -119,6 +119,8 @@ def _worker4(rank, world, port, results):
         plans = dict(_union_plans(Q, wk))
         plans.update({f"v_{n}": p
                       for n, p in Q.versatile_plans(full).items()})
+        from tests.test_dist_optional import _opt_plans
+        plans.update({f"o_{n}": p for n, p in _opt_plans(Q, wk).items()})
         ctx = OracleCtx(wk.lubm_gen(2, seed=42, sid=rank, nsrv=world),
                         sid=rank, nsrv=world)
         out = {}
@@ -152,6 +154,8 @@ def test_dist_union_versatile_four_ranks_equal_single():
     full = OracleCtx(wk.lubm_gen(2, seed=42))
     plans = dict(_union_plans(Q, wk))
     plans.update({f"v_{n}": p for n, p in Q.versatile_plans(full).items()})
+    from tests.test_dist_optional import _opt_plans
+    plans.update({f"o_{n}": p for n, p in _opt_plans(Q, wk).items()})
     for name, plan in plans.items():
         want = sort_rows(full.run_query(plan))
         assert got[name].shape == want.shape, (name, got[name].shape,
