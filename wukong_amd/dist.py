@@ -295,6 +295,15 @@ class DistQuery:
             col = v2c[-(s + 1)]
             T, matched = self._exchange_rows_cpu(T, matched, col)
             ostat = 2 if o >= 0 else (1 if v2c[-(o + 1)] >= 0 else 0)
+            memo = {}  # per-key edge memo (the dup memo, sparql.hpp:322-345)
+
+            def edges_of(cur):
+                e_ = memo.get(cur)
+                if e_ is None:
+                    e_ = ex.get_triples(cur, p, d)
+                    memo[cur] = e_
+                return e_
+
             if ostat == 0:
                 rows, flags = [], []
                 for i in range(len(T)):
@@ -303,7 +312,7 @@ class DistQuery:
                         rows.append(np.append(T[i], B))
                         flags.append(matched[i])
                         continue
-                    edges = ex.get_triples(int(cur), p, d)
+                    edges = edges_of(int(cur))
                     if len(edges) == 0:
                         rows.append(np.append(T[i], B))
                         flags.append(True)  # deg-0 keeps the flag
@@ -322,7 +331,7 @@ class DistQuery:
                     cur = T[i, col]
                     ok = False
                     if cur != B:
-                        edges = ex.get_triples(int(cur), p, d)
+                        edges = edges_of(int(cur))
                         tgt = (np.uint32(o) if ostat == 2
                                else T[i, v2c[-(o + 1)]])
                         ok = tgt != B and bool(np.isin(tgt, edges))
