@@ -58,6 +58,22 @@ def _worker(rank, world, port, results):
                 dq.run()
                 merged = dq.gather_result()
                 out[f"{thr_name}:{name}"] = sort_rows(merged)
+        # union / optional / versatile through the DEVICE exchange
+        # (branch rebinds, host-side optional over device tables, k_vu)
+        from tests.test_dist_union import _union_plans
+        from tests.test_dist_optional import _opt_plans
+        from tests.oracle_util import OracleCtx
+        vfull = OracleCtx(wk.lubm_gen(2, seed=42))
+        extra = {f"u_{n}": p for n, p in _union_plans(Q, wk).items()}
+        extra.update({f"o_{n}": p for n, p in _opt_plans(Q, wk).items()})
+        extra.update({f"v_{n}": p
+                      for n, p in Q.versatile_plans(vfull).items()})
+        for name, plan in extra.items():
+            ex = GpuExecutor(eng, plan)
+            dq = DistQuery(ex, plan, rank, world,
+                           device=f"cuda:{rank}", threshold=0)
+            dq.run()
+            out[f"x:{name}"] = sort_rows(dq.gather_result())
         out["_peers"] = peers
         if rank == 0:
             results.put(out)
@@ -105,6 +121,16 @@ def test_rccl_exchange_two_ranks():
             g = got[f"{m}:{name}"]
             assert g.shape == want.shape, (m, name, g.shape, want.shape)
             assert np.array_equal(g, want), (m, name)
+    from tests.test_dist_union import _union_plans
+    from tests.test_dist_optional import _opt_plans
+    extra = {f"u_{n}": p for n, p in _union_plans(Q, wk).items()}
+    extra.update({f"o_{n}": p for n, p in _opt_plans(Q, wk).items()})
+    extra.update({f"v_{n}": p for n, p in Q.versatile_plans(full).items()})
+    for name, plan in extra.items():
+        want = sort_rows(full.run_query(plan))
+        g = got[f"x:{name}"]
+        assert g.shape == want.shape, (name, g.shape, want.shape)
+        assert np.array_equal(g, want), name
 
 
 @pytest.mark.timeout(600)
