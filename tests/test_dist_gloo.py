@@ -86,8 +86,17 @@ def _modifier_plans(Q, wk):
     cstart = wk.Plan([(Q.FULLPROF, Q.TYPE_ID, Q.DIR_IN, X),
                       (Q.DEPT0_UNIV0, Q.WORKSFOR, Q.DIR_IN, X)],
                      nvars=1, required_vars=[X])
+    # final-op edge values: LIMIT 0 and OFFSET past the end must give
+    # empty results once, after the rank merge
+    lim0 = wk.Plan([(Q.GRADSTUDENT, Q.TYPE_ID, Q.DIR_IN, X),
+                    (X, Q.MEMBEROF, Q.DIR_OUT, Z)],
+                   nvars=2, required_vars=[Z], distinct=True, limit=0)
+    offbig = wk.Plan([(Q.GRADSTUDENT, Q.TYPE_ID, Q.DIR_IN, X),
+                      (X, Q.MEMBEROF, Q.DIR_OUT, Z)],
+                     nvars=2, required_vars=[Z], distinct=True,
+                     offset=10**6)
     return {"distinct": distinct, "distinct_lim_off": dlo,
-            "const_mid": cstart}
+            "const_mid": cstart, "lim0": lim0, "offbig": offbig}
 
 
 def _worker_remote(rank, world, port, results):
