@@ -12,7 +12,11 @@ k_peer_step probes the OWNER rank's store in place through HIP-IPC
 xGMI mappings (the one-sided-RDMA analog, gstore.hpp:260-338).
 Mid-plan const-/index-start filters broadcast the owner's edge list;
 filter steps between exchanges launch asynchronously; the final
-DISTINCT/OFFSET/LIMIT run once after the rank merge.  Every rank
+DISTINCT/OFFSET/LIMIT run once after the rank merge.  UNION branches
+re-enter the same step loop as engine sub-query continuations;
+OPTIONAL groups run as a host-side matched-flag restatement over
+owner-exchanged rows; VERSATILE steps exchange onto the owner-local
+vp lists (details per method below).  Every rank
 executes every step on its local store: probes of non-local keys miss
 by construction, so no owner special-casing is needed (index/const
 starts are naturally local, dispatch semantics of
