@@ -331,19 +331,20 @@ class DistQuery:
                 opt_cols.append(ncols)
                 ncols += 1
             else:
-                for i in range(len(T)):
-                    cur = T[i, col]
-                    ok = False
-                    if cur != B:
-                        edges = edges_of(int(cur))
-                        tgt = (np.uint32(o) if ostat == 2
-                               else T[i, v2c[-(o + 1)]])
-                        ok = tgt != B and bool(np.isin(tgt, edges))
-                    if not ok:
-                        if matched[i]:
-                            for c in opt_cols:
-                                T[i, c] = B
-                        matched[i] = False
+                # k2k / k2c: vectorized per unique probe key
+                cur_vals = T[:, col]
+                tgts = (np.full(len(T), o, dtype=np.uint32) if ostat == 2
+                        else T[:, v2c[-(o + 1)]])
+                ok = np.zeros(len(T), dtype=bool)
+                uniq, inv = np.unique(cur_vals, return_inverse=True)
+                for u_i, u in enumerate(uniq):
+                    if u == B:
+                        continue
+                    edges = edges_of(int(u))
+                    sel = inv == u_i
+                    ok[sel] = (tgts[sel] != B) & np.isin(tgts[sel], edges)
+                blank_rows(~ok & matched)
+                matched &= ok
         self._opt_table = T
         self._opt_v2c = v2c
 
