@@ -309,24 +309,40 @@ class DistQuery:
                 return e_
 
             if ostat == 0:
-                rows, flags = [], []
-                for i in range(len(T)):
-                    cur = T[i, col]
-                    if not matched[i] or cur == B:
-                        rows.append(np.append(T[i], B))
-                        flags.append(matched[i])
+                # k2u left join, vectorized per unique probe key (row
+                # order is free — the result is a multiset)
+                cur_vals = T[:, col]
+                active = matched & (cur_vals != B)
+                uniq, inv = np.unique(cur_vals, return_inverse=True)
+                parts_t, parts_c, parts_f = [], [], []
+                ext_rows = np.zeros(len(T), dtype=bool)
+                for u_i, u in enumerate(uniq):
+                    if u == B:
                         continue
-                    edges = edges_of(int(cur))
+                    edges = np.asarray(edges_of(int(u)), dtype=np.uint32)
                     if len(edges) == 0:
-                        rows.append(np.append(T[i], B))
-                        flags.append(True)  # deg-0 keeps the flag
-                    else:
-                        for e_ in edges:
-                            rows.append(np.append(T[i], e_))
-                            flags.append(True)
-                T = (np.array(rows, dtype=np.uint32) if rows
-                     else np.empty((0, T.shape[1] + 1), dtype=np.uint32))
-                matched = np.array(flags, dtype=bool)
+                        continue
+                    sel = active & (inv == u_i)
+                    n = int(sel.sum())
+                    if n == 0:
+                        continue
+                    ext_rows |= sel
+                    parts_t.append(np.repeat(T[sel], len(edges), axis=0))
+                    parts_c.append(np.tile(edges, n))
+                    parts_f.append(np.ones(n * len(edges), dtype=bool))
+                keep = ~ext_rows  # unmatched/BLANK keep flag; deg-0
+                if keep.any():    # matched keeps flag True
+                    parts_t.append(T[keep])
+                    parts_c.append(np.full(int(keep.sum()), B, np.uint32))
+                    parts_f.append(matched[keep] | active[keep])
+                if parts_t:
+                    T = np.column_stack(
+                        [np.concatenate(parts_t, axis=0),
+                         np.concatenate(parts_c)]).astype(np.uint32)
+                    matched = np.concatenate(parts_f)
+                else:
+                    T = np.empty((0, T.shape[1] + 1), dtype=np.uint32)
+                    matched = np.zeros(0, dtype=bool)
                 v2c[-(o + 1)] = ncols
                 opt_cols.append(ncols)
                 ncols += 1
