@@ -58,3 +58,35 @@ def test_gpu_planned_text_parity(store4, oracle4):
         want = oracle4.run_query(plan)
         assert got.shape == want.shape, (name, got.shape, want.shape)
         assert np.array_equal(sort_rows(got), sort_rows(want)), name
+
+
+def test_planner_invariance_fuzz(lubm2):
+    """Random valid BGPs, patterns SHUFFLED: the planner must recover a
+    startable order whose results equal the original textual order
+    (the reference's planner only reorders; semantics are order-free,
+    planner.hpp:218)."""
+    import random
+    from tests.test_fuzz_plans import random_plan
+
+    store = wk.Store(lubm2)
+    oc = OracleCtx(lubm2)
+    rng = random.Random(31337)
+    ran = 0
+    for trial in range(40):
+        base = random_plan(rng, store)
+        try:
+            want = oc.run_query(base)
+        except RuntimeError:
+            continue
+        shuffled = list(base.patterns)
+        rng.shuffle(shuffled)
+        planned = planner.plan_patterns(store, shuffled, base.nvars,
+                                        base.required_vars)
+        got = oc.run_query(planned)
+        assert got.shape == want.shape, (trial, base.patterns,
+                                         planned.patterns,
+                                         got.shape, want.shape)
+        assert np.array_equal(sort_rows(got), sort_rows(want)), \
+            (trial, base.patterns, planned.patterns)
+        ran += 1
+    assert ran > 25

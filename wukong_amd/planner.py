@@ -31,82 +31,59 @@ class PlannerError(ValueError):
     pass
 
 
-def plan_patterns(store, patterns, nvars, required_vars, **plan_kw):
-    """patterns: (s, p, d, o) tuples as parsed (textual orientation,
-    d=DIR_OUT).  Returns a Plan with a greedy execution order."""
+def _cands(store, pat, bound, first, est_rows):
+    """Yield (cost, pat, oriented, new_rows) for every legal way to run
+    `pat` in the current state.  `first` = nothing planned yet (index /
+    const_to_unknown starts are first-pattern-only, query.hpp:660-682)."""
+    s0, p0, d0, o0 = pat
+    for (a, p, d, b) in ((s0, p0, d0, o0), (o0, p0, d0 ^ 1, s0)):
+        if isinstance(p, int) and p < 0:
+            # predicate variable: needs const or bound start
+            if p in bound:
+                continue  # predicate already bound: unsupported
+            if a >= 0 and not _is_tpid(a) and first:
+                cost = new_rows = float(
+                    len(store.get_triples(a, PREDICATE_ID, d)) or 1)
+            elif a < 0 and a in bound:
+                cost = new_rows = est_rows * 4.0
+            else:
+                continue
+            yield (cost, pat, (a, p, d, b), new_rows)
+            continue
+        if a >= 0 and _is_tpid(a):
+            # index start (type or predicate index)
+            if not first or b >= 0:
+                continue
+            n = len(store.get_index(a, DIR_IN))
+            yield (float(max(n, 1)), pat, (a, p, d, b), float(n))
+        elif a >= 0:
+            if not first and (b < 0 and b not in bound):
+                # const_to_unknown is first-pattern-only
+                continue
+            n = len(store.get_triples(a, p, d))
+            if b >= 0 or b in bound:      # membership filter
+                yield (float(max(n, 1)) * 0.01 + 1, pat,
+                       (a, p, d, b), est_rows)
+            else:
+                yield (float(max(n, 1)), pat, (a, p, d, b), float(n))
+        elif a in bound:
+            if b >= 0 or b in bound:      # k2c / k2k: prune
+                yield (est_rows * 0.1 + 1, pat, (a, p, d, b),
+                       est_rows * 0.5)
+            else:                         # k2u: grow by avg degree
+                k, e = store.seg_stats(p, d)
+                g = est_rows * max((e / k) if k else 0.0, 0.05)
+                yield (g + est_rows, pat, (a, p, d, b), g)
+
+
+def _greedy(store, patterns, first_cand):
+    """Complete the order from a forced opener; PlannerError on dead end."""
     remaining = list(patterns)
     bound = set()
     out = []
     est_rows = 1.0
-
-    def avg_deg(p, d):
-        k, e = store.seg_stats(p, d)
-        return (e / k) if k else 0.0
-
-    def candidates(pat):
-        s, p, d, o = pat
-        # both orientations of the stored direction: (start,pred,dir,end)
-        yield (s, p, d, o)
-        yield (o, p, d ^ 1, s)
-
-    while remaining:
-        best = None  # (cost, pat, oriented, new_rows)
-        for pat in remaining:
-            for (a, p, d, b) in candidates(pat):
-                if isinstance(p, int) and p < 0:
-                    # predicate variable: needs const or bound start
-                    if p in bound:
-                        continue  # predicate already bound: unsupported
-                    if a >= 0 and not _is_tpid(a) and not out:
-                        cost = new_rows = float(
-                            len(store.get_triples(a, PREDICATE_ID, d)) or 1)
-                    elif a < 0 and a in bound:
-                        cost = new_rows = est_rows * 4.0
-                    else:
-                        continue
-                    cand = (cost, pat, (a, p, d, b), new_rows)
-                    if best is None or cand[0] < best[0]:
-                        best = cand
-                    continue
-                if a >= 0 and _is_tpid(a):
-                    # index start (type or predicate index): first pattern
-                    # only (query.hpp:660-682)
-                    if out or b >= 0:
-                        continue
-                    n = len(store.get_index(a, DIR_IN))
-                    cand = (float(max(n, 1)), pat, (a, p, d, b), float(n))
-                elif a >= 0:
-                    if out and (b < 0 and b not in bound):
-                        # const_to_unknown is first-pattern-only
-                        continue
-                    n = len(store.get_triples(a, p, d))
-                    if b >= 0 or b in bound:      # membership filter
-                        cand = (float(max(n, 1)) * 0.01 + 1, pat,
-                                (a, p, d, b), est_rows)
-                    else:
-                        cand = (float(max(n, 1)), pat, (a, p, d, b), float(n))
-                elif a in bound:
-                    if b >= 0 or b in bound:      # k2c / k2k: prune
-                        cand = (est_rows * 0.1 + 1, pat, (a, p, d, b),
-                                est_rows * 0.5)
-                    else:                         # k2u: grow by avg degree
-                        g = est_rows * max(avg_deg(p, d), 0.05)
-                        cand = (g + est_rows, pat, (a, p, d, b), g)
-                else:
-                    continue
-                if best is None or cand[0] < best[0]:
-                    best = cand
-        if best is None:
-            # special case: ?X rdf:type T as the opener -> type-index
-            for pat in remaining:
-                s, p, d, o = pat
-                if (not out and p == TYPE_ID and d == DIR_OUT and s < 0
-                        and o >= 0 and _is_tpid(o)):
-                    n = len(store.get_index(o, DIR_IN))
-                    best = (float(n), pat, (o, TYPE_ID, DIR_IN, s), float(n))
-                    break
-        if best is None:
-            raise PlannerError(f"no startable pattern among {remaining!r}")
+    best = first_cand
+    while True:
         _, pat, oriented, new_rows = best
         remaining.remove(pat)
         out.append(oriented)
@@ -114,8 +91,43 @@ def plan_patterns(store, patterns, nvars, required_vars, **plan_kw):
         for t in (oriented[0], oriented[3], oriented[1]):
             if isinstance(t, int) and t < 0:
                 bound.add(t)
+        if not remaining:
+            return out
+        best = None
+        for p_ in remaining:
+            for cand in _cands(store, p_, bound, False, est_rows):
+                if best is None or cand[0] < best[0]:
+                    best = cand
+        if best is None:
+            raise PlannerError(f"no startable pattern among {remaining!r}")
 
-    return Plan(out, nvars=nvars, required_vars=required_vars, **plan_kw)
+
+def plan_patterns(store, patterns, nvars, required_vars, **plan_kw):
+    """patterns: (s, p, d, o) tuples as parsed (textual orientation,
+    d=DIR_OUT).  Returns a Plan with a greedy execution order.
+
+    The greedy can dead-end on adversarial inputs: index and
+    const_to_unknown starts are first-pattern-only, so opening with the
+    wrong one (e.g. a typeof filter flipped into a type-index scan over
+    a variable that turns out to be a predicate var) strands the rest.
+    Openers are therefore tried in ascending cost with backtracking —
+    the first choice is exactly the plain greedy's, so well-formed
+    inputs plan identically; a PlannerError only surfaces if EVERY
+    opener dead-ends."""
+    openers = []
+    for pat in patterns:
+        openers.extend(_cands(store, pat, set(), True, 1.0))
+    openers.sort(key=lambda c: c[0])
+    last = None
+    for first_cand in openers:
+        try:
+            out = _greedy(store, patterns, first_cand)
+            return Plan(out, nvars=nvars, required_vars=required_vars,
+                        **plan_kw)
+        except PlannerError as e:
+            last = e
+    raise last if last is not None else PlannerError(
+        f"no startable pattern among {list(patterns)!r}")
 
 
 def plan_text(store, text, vocab, **kw):
