@@ -90,3 +90,42 @@ def test_planner_invariance_fuzz(lubm2):
             (trial, base.patterns, planned.patterns)
         ran += 1
     assert ran > 25
+
+
+def test_planner_orients_union_and_optional_groups(lubm4, wstore4, oracle4):
+    """Group patterns parse in textual orientation too: a UNION branch
+    or OPTIONAL pattern written 'backwards' (?y pred ?x with ?x bound)
+    must be flipped to a known-start orientation — passed through raw,
+    the unknown-start branch silently returns garbage."""
+    vocab = sparql.lubm_entity_vocab(wstore4)
+    text = (
+        "PREFIX ub: <http://swat.cse.lehigh.edu/onto/univ-bench.owl#>\n"
+        "SELECT ?x ?y WHERE {\n"
+        "  ?x rdf:type ub:Department .\n"
+        "  { ?y ub:memberOf ?x . } UNION { ?y ub:worksFor ?x . }\n"
+        "}")
+    plan = planner.plan_text(wstore4, text, vocab)
+    for br in plan.unions:
+        s = br[0][0]
+        assert s == plan.patterns[0][3], "branch must start from bound ?x"
+    got = oracle4.run_query(plan)
+    hand = wk.Plan([(Q.DEPARTMENT, Q.TYPE_ID, wk.DIR_IN, -1)], 2, [-1, -2],
+                   unions=[[(-1, Q.MEMBEROF, wk.DIR_IN, -2)],
+                           [(-1, Q.WORKSFOR, wk.DIR_IN, -2)]])
+    want = oracle4.run_query(hand)
+    assert got.shape == want.shape and len(got) > 0
+    assert np.array_equal(sort_rows(got), sort_rows(want))
+
+    text2 = (
+        "PREFIX ub: <http://swat.cse.lehigh.edu/onto/univ-bench.owl#>\n"
+        "SELECT ?x ?y WHERE {\n"
+        "  ?x rdf:type ub:UndergraduateStudent .\n"
+        "  OPTIONAL { ?y ub:advisor ?x . }\n"   # reversed: ?x is bound
+        "}")
+    p2 = planner.plan_text(wstore4, text2, vocab)
+    g2 = oracle4.run_query(p2)
+    h2 = wk.Plan([(Q.UGSTUDENT, Q.TYPE_ID, wk.DIR_IN, -1)], 2, [-1, -2],
+                 optional=[(-1, Q.ADVISOR, wk.DIR_IN, -2)])
+    w2 = oracle4.run_query(h2)
+    assert g2.shape == w2.shape and len(g2) > 0
+    assert np.array_equal(sort_rows(g2), sort_rows(w2))

@@ -77,7 +77,8 @@ def _cands(store, pat, bound, first, est_rows):
 
 
 def _greedy(store, patterns, first_cand):
-    """Complete the order from a forced opener; PlannerError on dead end."""
+    """Complete the order from a forced opener; PlannerError on dead
+    end.  Returns (ordered patterns, bound vars, final row estimate)."""
     remaining = list(patterns)
     bound = set()
     out = []
@@ -92,7 +93,7 @@ def _greedy(store, patterns, first_cand):
             if isinstance(t, int) and t < 0:
                 bound.add(t)
         if not remaining:
-            return out
+            return out, bound, est_rows
         best = None
         for p_ in remaining:
             for cand in _cands(store, p_, bound, False, est_rows):
@@ -100,6 +101,52 @@ def _greedy(store, patterns, first_cand):
                     best = cand
         if best is None:
             raise PlannerError(f"no startable pattern among {remaining!r}")
+
+
+def _order_group(store, pats, bound0, est_rows):
+    """Orient + greedily order a UNION branch continuing from the main
+    BGP's bound set (BGP semantics inside a branch are order-free)."""
+    remaining = list(pats)
+    bound = set(bound0)
+    out = []
+    est = est_rows
+    while remaining:
+        best = None
+        for p_ in remaining:
+            for cand in _cands(store, p_, bound, False, est):
+                if best is None or cand[0] < best[0]:
+                    best = cand
+        if best is None:
+            raise PlannerError(f"no startable pattern in group {remaining!r}")
+        _, pat, oriented, new_rows = best
+        remaining.remove(pat)
+        out.append(oriented)
+        est = max(new_rows, 1.0)
+        for t in (oriented[0], oriented[3], oriented[1]):
+            if isinstance(t, int) and t < 0:
+                bound.add(t)
+    return out
+
+
+def _orient_group(store, pats, bound0):
+    """Orient an OPTIONAL group's patterns IN TEXTUAL ORDER (the
+    matched-flag mechanics are order-sensitive, so only the direction
+    may flip, never the sequence)."""
+    bound = set(bound0)
+    out = []
+    for pat in pats:
+        best = None
+        for cand in _cands(store, pat, bound, False, 1.0):
+            if best is None or cand[0] < best[0]:
+                best = cand
+        if best is None:
+            raise PlannerError(f"no orientation for {pat!r} in OPTIONAL")
+        oriented = best[2]
+        out.append(oriented)
+        for t in (oriented[0], oriented[3], oriented[1]):
+            if isinstance(t, int) and t < 0:
+                bound.add(t)
+    return out
 
 
 def plan_patterns(store, patterns, nvars, required_vars, **plan_kw):
@@ -121,9 +168,18 @@ def plan_patterns(store, patterns, nvars, required_vars, **plan_kw):
     last = None
     for first_cand in openers:
         try:
-            out = _greedy(store, patterns, first_cand)
-            return Plan(out, nvars=nvars, required_vars=required_vars,
-                        **plan_kw)
+            out, bound, est = _greedy(store, patterns, first_cand)
+            # groups are parsed in textual orientation too: orient
+            # union branches (and reorder — BGP semantics) and orient
+            # optional patterns (order preserved) from the main BGP's
+            # bound set
+            kw = dict(plan_kw)
+            if kw.get("unions"):
+                kw["unions"] = [_order_group(store, br, bound, est)
+                                for br in kw["unions"]]
+            if kw.get("optional"):
+                kw["optional"] = _orient_group(store, kw["optional"], bound)
+            return Plan(out, nvars=nvars, required_vars=required_vars, **kw)
         except PlannerError as e:
             last = e
     raise last if last is not None else PlannerError(
