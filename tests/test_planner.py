@@ -52,7 +52,20 @@ def test_planner_beats_textual_order(wstore4):
 def test_gpu_planned_text_parity(store4, oracle4):
     vocab = sparql.lubm_entity_vocab(store4)
     eng = wk.Engine(store4, device=0)
-    for name, text in {**QS.TEXT, **QS.TEXT_VERSATILE}.items():
+    texts = {**QS.TEXT, **QS.TEXT_VERSATILE}
+    texts["u_rev"] = (
+        "PREFIX ub: <http://swat.cse.lehigh.edu/onto/univ-bench.owl#>\n"
+        "SELECT ?x ?y WHERE {\n"
+        "  ?x rdf:type ub:Department .\n"
+        "  { ?y ub:memberOf ?x . } UNION { ?y ub:worksFor ?x . }\n"
+        "}")
+    texts["o_rev"] = (
+        "PREFIX ub: <http://swat.cse.lehigh.edu/onto/univ-bench.owl#>\n"
+        "SELECT ?x ?y WHERE {\n"
+        "  ?x rdf:type ub:UndergraduateStudent .\n"
+        "  OPTIONAL { ?y ub:advisor ?x . }\n"
+        "}")
+    for name, text in texts.items():
         plan = planner.plan_text(store4, text, vocab)
         got = eng.run_query(plan)
         want = oracle4.run_query(plan)
