@@ -141,3 +141,33 @@ def test_random_triples_vs_brute():
         assert set(st.get_index(p, OUT)) == obj[p]
         assert st.seg_stats(p, OUT) == (len(subj[p]),
                                         sum(len(out_d[k]) for k in out_d if k[1] == p))
+
+
+def test_partition_union_random_nsrv3():
+    """Random triples partitioned 3 ways: every partition holds exactly
+    its owned rows (pso by s%3, pos by o%3 — base_loader.hpp:344-352)
+    and the partitions' OUT edge lists union to the full store's."""
+    rng = np.random.default_rng(99)
+    t = np.stack([
+        rng.integers(VBASE, VBASE + 500, size=3000, dtype=np.uint32),
+        rng.integers(2, 20, size=3000, dtype=np.uint32),
+        rng.integers(VBASE, VBASE + 500, size=3000, dtype=np.uint32),
+    ], axis=1)
+    t = np.unique(t, axis=0)
+    full = wk.Store(t)
+    parts = [wk.Store(t, sid=r, nsrv=3) for r in range(3)]
+    assert all(p.check() == 0 for p in parts)
+    for v, p in {(int(r[0]), int(r[1])) for r in t[:200]}:
+        want = _edges(full, v, p, OUT)
+        got = _edges(parts[v % 3], v, p, OUT)
+        assert np.array_equal(got, want), (v, p)
+        for r in range(3):  # non-owners hold nothing for this key
+            if r != v % 3:
+                assert parts[r].get_triples(v, p, OUT).size == 0
+    # subject indexes union (with cross-partition dedupe) to the full set
+    for p in range(2, 20):
+        want = set(full.get_index(p, IN))
+        got = set()
+        for r in range(3):
+            got |= set(parts[r].get_index(p, IN))
+        assert got == want, p
