@@ -104,3 +104,68 @@ def test_parse_optional_union(lubm2):
     assert len(plan2.unions) == 2 and not plan2.optional
     got2 = ora.run_query(plan2)
     assert len(got2) > 0
+
+
+def test_parser_text_roundtrip_fuzz(lubm2):
+    """Random BGPs emitted as SPARQL text, parsed back and planned:
+    results must equal the directly-built pattern list planned the
+    same way — a parse bug (wrong var ids, wrong URI resolution, a
+    dropped triple) shows up as a result mismatch."""
+    import random
+    import wukong_amd as wk
+    from wukong_amd import planner
+    from tests.oracle_util import OracleCtx, sort_rows
+    import numpy as np
+
+    vocab = sparql.lubm_vocab()
+    by_id = {}
+    for tok, i in vocab.items():
+        by_id.setdefault(i, tok)
+    preds = [Q.SUBORG, Q.UGDEGREE, Q.MEMBEROF, Q.WORKSFOR, Q.TEACHEROF,
+             Q.ADVISOR, Q.TAKESCOURSE]
+    types = [Q.UNIVERSITY, Q.DEPARTMENT, Q.FULLPROF, Q.UGSTUDENT,
+             Q.GRADSTUDENT, Q.COURSE]
+    store = wk.Store(lubm2)
+    oc = OracleCtx(lubm2)
+    rng = random.Random(777)
+    ran = 0
+    for trial in range(30):
+        nv = rng.randint(2, 4)
+        # textual patterns (all d=OUT; reversed edges emit swapped vars)
+        pats = [(-1, Q.TYPE_ID, 1, rng.choice(types))]
+        bound = [-1]
+        free = [-v for v in range(2, nv + 1)]  # consecutive: -2..-nv
+        for _ in range(rng.randint(1, 3)):
+            a = rng.choice(bound)
+            p = rng.choice(preds)
+            if free and rng.random() < 0.7:
+                b = free.pop(0)
+                bound.append(b)
+            else:
+                b = rng.choice(bound)
+                if b == a:
+                    continue
+            if rng.random() < 0.5:
+                pats.append((a, p, 1, b))
+            else:
+                pats.append((b, p, 1, a))   # reversed in text
+        lines = []
+        for (s, p, d, o) in pats:
+            st = f"?v{-s - 1}" if s < 0 else by_id[s]
+            ot = f"?v{-o - 1}" if o < 0 else by_id[o]
+            lines.append(f"  {st} {by_id[p] if p != 1 else 'rdf:type'} {ot} .")
+        sel = " ".join(f"?v{-v - 1}" for v in bound)
+        text = ("PREFIX ub: <http://swat.cse.lehigh.edu/onto/univ-bench.owl#>\n"
+                f"SELECT {sel} WHERE {{\n" + "\n".join(lines) + "\n}")
+        try:
+            parsed = planner.plan_text(store, text, vocab)
+            direct = planner.plan_patterns(store, pats,
+                                           max(-v for v in bound), bound)
+        except planner.PlannerError:
+            continue
+        got = oc.run_query(parsed)
+        want = oc.run_query(direct)
+        assert got.shape == want.shape, (trial, text, got.shape, want.shape)
+        assert np.array_equal(sort_rows(got), sort_rows(want)), (trial, text)
+        ran += 1
+    assert ran > 20
