@@ -169,3 +169,63 @@ def test_parser_text_roundtrip_fuzz(lubm2):
         assert np.array_equal(sort_rows(got), sort_rows(want)), (trial, text)
         ran += 1
     assert ran > 20
+
+
+def test_parser_group_roundtrip_fuzz(lubm2):
+    """Same roundtrip net over OPTIONAL { } and { } UNION { } texts:
+    group brace extraction and group-var mapping must reproduce the
+    directly-built groups (both sides share the planner's orientation,
+    so mismatches isolate PARSE bugs)."""
+    import random
+    import wukong_amd as wk
+    from wukong_amd import planner
+    from tests.oracle_util import OracleCtx, sort_rows
+    import numpy as np
+
+    vocab = sparql.lubm_vocab()
+    by_id = {}
+    for tok, i in vocab.items():
+        by_id.setdefault(i, tok)
+    preds = [Q.SUBORG, Q.UGDEGREE, Q.MEMBEROF, Q.WORKSFOR, Q.ADVISOR,
+             Q.TAKESCOURSE]
+    types = [Q.DEPARTMENT, Q.FULLPROF, Q.UGSTUDENT, Q.GRADSTUDENT]
+    store = wk.Store(lubm2)
+    oc = OracleCtx(lubm2)
+    rng = random.Random(4242)
+
+    def tri_line(s, p, o):
+        st = f"?v{-s - 1}" if s < 0 else by_id[s]
+        ot = f"?v{-o - 1}" if o < 0 else by_id[o]
+        return f"{st} {by_id[p] if p != 1 else 'rdf:type'} {ot} ."
+
+    ran = 0
+    for trial in range(30):
+        main = [(-1, Q.TYPE_ID, 1, rng.choice(types))]
+        gv = -2  # the group-born var
+        a, b = rng.choice(preds), rng.choice(preds)
+        rev1, rev2 = rng.random() < 0.5, rng.random() < 0.5
+        p1 = (gv, a, 1, -1) if rev1 else (-1, a, 1, gv)
+        p2 = (gv, b, 1, -1) if rev2 else (-1, b, 1, gv)
+        if rng.random() < 0.5:
+            text = ("PREFIX ub: <x>\nSELECT ?v0 ?v1 WHERE {\n  "
+                    + tri_line(*[main[0][0], main[0][1], main[0][3]])
+                    + "\n  { " + tri_line(p1[0], p1[1], p1[3])
+                    + " } UNION { " + tri_line(p2[0], p2[1], p2[3]) + " }\n}")
+            direct = wk.Plan(main, 2, [-1, gv], unions=[[p1], [p2]])
+        else:
+            text = ("PREFIX ub: <x>\nSELECT ?v0 ?v1 WHERE {\n  "
+                    + tri_line(*[main[0][0], main[0][1], main[0][3]])
+                    + "\n  OPTIONAL { " + tri_line(p1[0], p1[1], p1[3])
+                    + " }\n}")
+            direct = wk.Plan(main, 2, [-1, gv], optional=[p1])
+        parsed = planner.plan_text(store, text, vocab)
+        want_plan = planner.plan_patterns(store, direct.patterns, 2,
+                                          [-1, gv],
+                                          optional=direct.optional,
+                                          unions=direct.unions)
+        got = oc.run_query(parsed)
+        want = oc.run_query(want_plan)
+        assert got.shape == want.shape, (trial, text, got.shape, want.shape)
+        assert np.array_equal(sort_rows(got), sort_rows(want)), (trial, text)
+        ran += 1
+    assert ran > 25
