@@ -110,3 +110,18 @@ def test_mem_usage(lubm2):
     assert slots == st.num_slots * 16
     assert edges == st.num_edges * 4
     assert side > 0  # type_of + vp CSR + fn maps + type bitmaps
+
+
+def test_watdiv_generator_partition_covers_full():
+    """WatDiv generator partitions like the LUBM one: partition r holds
+    every full-gen triple with s%n==r (pso side) and o%n==r (pos side),
+    and nothing else relevant (base_loader.hpp:344-352 contract)."""
+    full = wk.watdiv_gen(300, seed=7)
+    parts = [wk.watdiv_gen(300, seed=7, sid=r, nsrv=3) for r in range(3)]
+    for r in range(3):
+        want = full[full[:, 0] % 3 == r]
+        have = parts[r][parts[r][:, 0] % 3 == r]
+        assert set(map(tuple, want.tolist())) == set(map(tuple, have.tolist()))
+        want = full[full[:, 2] % 3 == r]
+        have = parts[r][parts[r][:, 2] % 3 == r]
+        assert set(map(tuple, want.tolist())) == set(map(tuple, have.tolist()))
