@@ -5,7 +5,7 @@
  * (NT→ID conversion).  ID scheme follows datagen/generate_data.cpp:122-123:
  * index ids (predicates AND types) from 2, normal vertex ids from 2^17.
  * Schema and cardinalities follow the LUBM ontology as used by the
- * reference's Q1-Q7 (scripts/sparql_query/lubm/basic/*): notably
+ * reference's Q1-Q7 (scripts/sparql_query/lubm/basic): notably
  * UndergraduateStudents have NO undergraduateDegreeFrom (so Q3 yields 0
  * rows, matching docs/performance/S1C24-LUBM2560-20181203.md:125), and
  * GraduateStudents draw undergraduateDegreeFrom uniformly over all
