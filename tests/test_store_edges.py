@@ -171,3 +171,15 @@ def test_partition_union_random_nsrv3():
         for r in range(3):
             got |= set(parts[r].get_index(p, IN))
         assert got == want, p
+
+
+def test_plan_rejects_out_of_range_vars():
+    """A var id below -nvars would index past the engine's v2c array on
+    the C side — the binding must reject it before it crosses the ABI."""
+    with pytest.raises(ValueError):
+        wk.Plan([(2, 1, IN, -4)], nvars=2, required_vars=[-1])
+    with pytest.raises(ValueError):
+        wk.Plan([(2, 1, IN, -1)], nvars=1, required_vars=[-2])
+    with pytest.raises(ValueError):
+        wk.Plan([(2, 1, IN, -1)], nvars=2, required_vars=[-1],
+                unions=[[(-1, 2, OUT, -5)]])

@@ -220,6 +220,19 @@ class Plan:
         self.blind = blind
         self.optional = list(optional or [])   # OPTIONAL pattern group
         self.unions = [list(u) for u in (unions or [])]  # UNION branches
+        # validate var ids HERE: an out-of-range var (< -nvars) would
+        # index past the engine's v2c array on the C side
+        all_pats = (self.patterns + self.optional
+                    + [p for u in self.unions for p in u])
+        for (s, p, d, o) in all_pats:
+            for v in (s, p, o):
+                if isinstance(v, (int, np.integer)) and v < -nvars:
+                    raise ValueError(
+                        f"variable {v} out of range for nvars={nvars}")
+        for v in self.required_vars:
+            if v < -nvars or v >= 0:
+                raise ValueError(
+                    f"required var {v} out of range for nvars={nvars}")
 
     def to_c(self):
         def mk(pat_list):
