@@ -137,7 +137,13 @@ def test_fuzz_watdiv_oracle_vs_brute():
         free = vars_[1:]
         for _ in range(rng.randint(1, 3)):
             s = rng.choice(bound)
-            if rng.random() < 0.5 and free:
+            r = rng.random()
+            if r < 0.15 and len(free) >= 2:
+                # predicate variable over the WatDiv vp lists
+                pvar, ovar = free.pop(0), free.pop(0)
+                pats.append((s, pvar, rng.choice([0, 1]), ovar))
+                bound += [pvar, ovar]
+            elif r < 0.55 and free:
                 o = free.pop(0)
                 pats.append((s, rng.choice(preds), rng.choice([0, 1]), o))
                 bound.append(o)
