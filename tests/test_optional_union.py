@@ -183,3 +183,61 @@ def test_gpu_union_vu_parity(store4, oracle4):
     want = oracle4.run_query(plan)
     assert got.shape == want.shape, (got.shape, want.shape)
     assert np.array_equal(sort_rows(got), sort_rows(want))
+
+
+def test_group_fuzz_oracle_vs_independent_numpy(lubm2):
+    """Randomized groups against restatements that share NO code with
+    the oracle: UNION == concat of the two extended plain BGPs;
+    OPTIONAL == dict-based left join over the base result."""
+    import random
+    from tests.test_fuzz_plans import random_plan, PREDS
+    from tests.oracle_util import OracleCtx
+
+    store = wk.Store(lubm2)
+    oc = OracleCtx(lubm2)
+    rng = random.Random(61)
+    checked = 0
+    while checked < 15:
+        base = random_plan(rng, store)
+        if not all(pp[1] >= 1 for pp in base.patterns):
+            continue
+        try:
+            base_res = oc.run_query(Plan(base.patterns, base.nvars,
+                                         base.required_vars))
+        except RuntimeError:
+            continue
+        bound = list(base.required_vars)
+        nv = base.nvars
+        s = rng.choice(bound)
+        ovar = -(nv + 1)
+        p1, p2 = rng.choice(PREDS), rng.choice(PREDS)
+        d1, d2 = rng.choice([0, 1]), rng.choice([0, 1])
+        up = Plan(base.patterns, nv + 1, bound + [ovar],
+                  unions=[[(s, p1, d1, ovar)], [(s, p2, d2, ovar)]])
+        got = oc.run_query(up)
+        a = oc.run_query(Plan(list(base.patterns) + [(s, p1, d1, ovar)],
+                              nv + 1, bound + [ovar]))
+        b = oc.run_query(Plan(list(base.patterns) + [(s, p2, d2, ovar)],
+                              nv + 1, bound + [ovar]))
+        want = np.vstack([a, b]) if (a.size or b.size) else a
+        assert got.shape == want.shape, (base.patterns, s, p1, p2)
+        assert np.array_equal(sort_rows(got), sort_rows(want))
+        opt = Plan(base.patterns, nv + 1, bound + [ovar],
+                   optional=[(s, p1, d1, ovar)])
+        got_o = oc.run_query(opt)
+        scol = bound.index(s)
+        rows, memo = [], {}
+        for row in base_res:
+            v = int(row[scol])
+            if v not in memo:
+                memo[v] = oc.get_triples(v, p1, d1)
+            e = memo[v]
+            if len(e):
+                rows += [list(row) + [int(x)] for x in e]
+            else:
+                rows.append(list(row) + [BLANK])
+        want_o = (np.array(rows, dtype=np.uint32) if rows
+                  else np.empty((0, len(bound) + 1), dtype=np.uint32))
+        assert got_o.shape == want_o.shape, (base.patterns, s, p1, d1)
+        assert np.array_equal(sort_rows(got_o), sort_rows(want_o))
+        checked += 1
