@@ -679,7 +679,7 @@ def main():
             dist.destroy_process_group()
         return
 
-    total_queries = 7 * args.steps
+    total_queries = len(names) * args.steps  # 7 LUBM / 5 WatDiv templates
     qps = total_queries / elapsed
 
     # roofline: dedicated measurement of the dominant kernels on the
@@ -824,12 +824,15 @@ def main():
         "dtype": "u32",
         "data": "synthetic",
         "config": {
-            "workload": "LUBM-2560 on 1xMI355X, full graph in HBM, "
-                        "Q1-Q7 mixed light/heavy" if ngpus == 1 else
-                        f"LUBM-2560 subject-hash-partitioned across "
+            "workload": (f"{dataset} on 1xMI355X, full graph in HBM, "
+                         + ("star/linear/snowflake templates"
+                            if args.watdiv else "Q1-Q7 mixed light/heavy"))
+                        if ngpus == 1 else
+                        f"{dataset} subject-hash-partitioned across "
                         f"{ngpus}xMI355X, RCCL all-to-all sub-query shipping",
             "dataset": dataset,
-            "queries": "lubm q1-q7, osdi16 plans",
+            "queries": ("watdiv templates" if args.watdiv
+                        else "lubm q1-q7, osdi16 plans"),
             "blind": True,  # reference emulator semantics, proxy.hpp:491
             "parallelism": f"graph-partitioned x{ngpus} + per-step all-to-allv",
             "inflight": inflight,  # reference emulator window (proxy.hpp -p)
