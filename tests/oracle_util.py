@@ -87,9 +87,13 @@ class OracleCtx:
             t.ctypes.data_as(ctypes.POINTER(ctypes.c_uint32)), t.shape[0], sid, nsrv)
 
     def __del__(self):
-        if getattr(self, "_h", None):
-            lib().ok_free(self._h)
-            self._h = None
+        # guard against interpreter-shutdown teardown (globals cleared)
+        try:
+            if getattr(self, "_h", None):
+                lib().ok_free(self._h)
+                self._h = None
+        except (TypeError, AttributeError):
+            pass
 
     def get_triples(self, vid, pid, direction):
         sz = ctypes.c_uint64()
@@ -191,9 +195,12 @@ class OracleExecutor:
                                        self.npat, plan.nvars)
 
     def __del__(self):
-        if getattr(self, "_h", None):
-            lib().ok_query_free(self._h)
-            self._h = None
+        try:
+            if getattr(self, "_h", None):
+                lib().ok_query_free(self._h)
+                self._h = None
+        except (TypeError, AttributeError):
+            pass
 
     def load(self, table, v2c, step):
         t = np.ascontiguousarray(table, dtype=np.uint32)

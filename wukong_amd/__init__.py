@@ -274,9 +274,11 @@ class Store:
             raise RuntimeError("wk_store_build failed")
         self.sid, self.nsrv = sid, nsrv
 
-    def __del__(self):
+    def __del__(self, _free=_store_free):
+        # default-arg capture: module globals are already cleared when
+        # destructors run at interpreter shutdown
         if getattr(self, "_h", None):
-            _store_free(self._h)
+            _free(self._h)
             self._h = None
 
     def get_triples(self, vid, pid, direction):
@@ -343,9 +345,9 @@ class GpuStore:
         if not self._h:
             raise RuntimeError("wk_gpu_store_create failed (no GPU / HBM alloc)")
 
-    def __del__(self):
+    def __del__(self, _free=_gstore_destroy):
         if getattr(self, "_h", None):
-            _gstore_destroy(self._h)
+            _free(self._h)
             self._h = None
 
     def export_blob(self):
@@ -396,9 +398,9 @@ class Engine:
             raise RuntimeError(f"submit rc={rc}")
         self._cplan = cplan
 
-    def __del__(self):
+    def __del__(self, _free=_eng_destroy):
         if getattr(self, "_h", None):
-            _eng_destroy(self._h)
+            _free(self._h)
             self._h = None
 
     def graph_build(self, plan):
