@@ -209,6 +209,31 @@ def test_dist_two_ranks_equal_single():
 
 
 @pytest.mark.timeout(600)
+def test_dist_three_ranks_equal_single():
+    """world_size=3 — odd world: vid % 3 ownership has no power-of-two
+    structure to hide modulo mistakes behind."""
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    port = 29882
+    procs = [ctx.Process(target=_worker, args=(r, 3, port, results))
+             for r in range(3)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=540)
+    for p in procs:
+        p.join(timeout=60)
+
+    import wukong_amd as wk
+    from wukong_amd import queries as Q
+    from tests.oracle_util import OracleCtx, sort_rows
+    full = OracleCtx(wk.lubm_gen(2, seed=42))
+    for name in QUERIES:
+        want = sort_rows(full.run_query(Q.ALL[name]))
+        assert got[name].shape == want.shape, (name, got[name].shape, want.shape)
+        assert np.array_equal(got[name], want), name
+
+
+@pytest.mark.timeout(600)
 def test_dist_four_ranks_equal_single():
     """world_size=4 — the shape the driver's 4/8-GPU scaling bench uses."""
     ctx = mp.get_context("spawn")
