@@ -76,55 +76,55 @@ def _cands(store, pat, bound, first, est_rows):
                 yield (g + est_rows, pat, (a, p, d, b), g)
 
 
-def _greedy(store, patterns, first_cand):
-    """Complete the order from a forced opener; PlannerError on dead
-    end.  Returns (ordered patterns, bound vars, final row estimate)."""
-    remaining = list(patterns)
-    bound = set()
-    out = []
-    est_rows = 1.0
-    best = first_cand
-    while True:
-        _, pat, oriented, new_rows = best
-        remaining.remove(pat)
-        out.append(oriented)
-        est_rows = max(new_rows, 1.0)
+def _search(store, patterns, bound0, first, est0, budget=20000):
+    """Cost-ordered depth-first search over execution orders: at each
+    level try the candidates cheapest-first and backtrack on dead ends.
+    The first complete leaf IS the plain greedy's order, so well-formed
+    inputs cost one pass; adversarial orders (e.g. a flipped k2u that
+    would bind a variable another pattern needs as its PREDICATE var)
+    backtrack instead of failing.  Returns (out, bound, est_rows)."""
+    picked = []  # (oriented, new_rows) along the accepted path
+    state = [0]  # node budget box
+
+    def dfs(remaining, bound, est, is_first):
+        if not remaining:
+            return True
+        cands = []
+        for p_ in remaining:
+            cands.extend(_cands(store, p_, bound, is_first, est))
+        cands.sort(key=lambda c: c[0])
+        for cost, pat, oriented, new_rows in cands:
+            state[0] += 1
+            if state[0] > budget:
+                raise PlannerError("planner search budget exceeded")
+            nb = set(bound)
+            for t in (oriented[0], oriented[3], oriented[1]):
+                if isinstance(t, int) and t < 0:
+                    nb.add(t)
+            picked.append((oriented, new_rows))
+            if dfs([p_ for p_ in remaining if p_ is not pat], nb,
+                   max(new_rows, 1.0), False):
+                return True
+            picked.pop()
+        return False
+
+    if not dfs(list(patterns), set(bound0), est0, first):
+        raise PlannerError(f"no startable order for {list(patterns)!r}")
+    out = [o for o, _ in picked]
+    bound = set(bound0)
+    for oriented in out:
         for t in (oriented[0], oriented[3], oriented[1]):
             if isinstance(t, int) and t < 0:
                 bound.add(t)
-        if not remaining:
-            return out, bound, est_rows
-        best = None
-        for p_ in remaining:
-            for cand in _cands(store, p_, bound, False, est_rows):
-                if best is None or cand[0] < best[0]:
-                    best = cand
-        if best is None:
-            raise PlannerError(f"no startable pattern among {remaining!r}")
+    est = max(picked[-1][1], 1.0) if picked else est0
+    return out, bound, est
 
 
 def _order_group(store, pats, bound0, est_rows):
-    """Orient + greedily order a UNION branch continuing from the main
-    BGP's bound set (BGP semantics inside a branch are order-free)."""
-    remaining = list(pats)
-    bound = set(bound0)
-    out = []
-    est = est_rows
-    while remaining:
-        best = None
-        for p_ in remaining:
-            for cand in _cands(store, p_, bound, False, est):
-                if best is None or cand[0] < best[0]:
-                    best = cand
-        if best is None:
-            raise PlannerError(f"no startable pattern in group {remaining!r}")
-        _, pat, oriented, new_rows = best
-        remaining.remove(pat)
-        out.append(oriented)
-        est = max(new_rows, 1.0)
-        for t in (oriented[0], oriented[3], oriented[1]):
-            if isinstance(t, int) and t < 0:
-                bound.add(t)
+    """Orient + order a UNION branch continuing from the main BGP's
+    bound set (BGP semantics inside a branch are order-free); same
+    backtracking search as the main BGP."""
+    out, _, _ = _search(store, pats, bound0, False, est_rows)
     return out
 
 
@@ -153,37 +153,24 @@ def plan_patterns(store, patterns, nvars, required_vars, **plan_kw):
     """patterns: (s, p, d, o) tuples as parsed (textual orientation,
     d=DIR_OUT).  Returns a Plan with a greedy execution order.
 
-    The greedy can dead-end on adversarial inputs: index and
-    const_to_unknown starts are first-pattern-only, so opening with the
-    wrong one (e.g. a typeof filter flipped into a type-index scan over
-    a variable that turns out to be a predicate var) strands the rest.
-    Openers are therefore tried in ascending cost with backtracking —
-    the first choice is exactly the plain greedy's, so well-formed
-    inputs plan identically; a PlannerError only surfaces if EVERY
-    opener dead-ends."""
-    openers = []
-    for pat in patterns:
-        openers.extend(_cands(store, pat, set(), True, 1.0))
-    openers.sort(key=lambda c: c[0])
-    last = None
-    for first_cand in openers:
-        try:
-            out, bound, est = _greedy(store, patterns, first_cand)
-            # groups are parsed in textual orientation too: orient
-            # union branches (and reorder — BGP semantics) and orient
-            # optional patterns (order preserved) from the main BGP's
-            # bound set
-            kw = dict(plan_kw)
-            if kw.get("unions"):
-                kw["unions"] = [_order_group(store, br, bound, est)
-                                for br in kw["unions"]]
-            if kw.get("optional"):
-                kw["optional"] = _orient_group(store, kw["optional"], bound)
-            return Plan(out, nvars=nvars, required_vars=required_vars, **kw)
-        except PlannerError as e:
-            last = e
-    raise last if last is not None else PlannerError(
-        f"no startable pattern among {list(patterns)!r}")
+    A plain greedy can dead-end on adversarial inputs: index and
+    const_to_unknown starts are first-pattern-only, and a flipped k2u
+    can bind a variable another pattern needs as its PREDICATE var —
+    either way the rest of the BGP strands even though a valid order
+    exists.  `_search` therefore explores candidates cheapest-first
+    with full backtracking; its first complete leaf IS the greedy
+    order, so well-formed inputs plan identically in one pass."""
+    out, bound, est = _search(store, patterns, set(), True, 1.0)
+    # groups are parsed in textual orientation too: orient union
+    # branches (and reorder — BGP semantics) and orient optional
+    # patterns (order preserved) from the main BGP's bound set
+    kw = dict(plan_kw)
+    if kw.get("unions"):
+        kw["unions"] = [_order_group(store, br, bound, est)
+                        for br in kw["unions"]]
+    if kw.get("optional"):
+        kw["optional"] = _orient_group(store, kw["optional"], bound)
+    return Plan(out, nvars=nvars, required_vars=required_vars, **kw)
 
 
 def plan_text(store, text, vocab, **kw):
