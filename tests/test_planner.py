@@ -142,3 +142,83 @@ def test_planner_orients_union_and_optional_groups(lubm4, wstore4, oracle4):
     w2 = oracle4.run_query(h2)
     assert g2.shape == w2.shape and len(g2) > 0
     assert np.array_equal(sort_rows(g2), sort_rows(w2))
+
+
+def test_planner_vu_before_known_object(lubm2):
+    """Regression: the search must never order a predicate-variable
+    pattern AFTER its object var binds — that shape
+    (known_unknown_known) is rejected by the engine and silently
+    REBINDS in the oracle, changing semantics.  Found by the vu-heavy
+    shuffled soak."""
+    from tests.oracle_util import OracleCtx
+    store = wk.Store(lubm2)
+    oc = OracleCtx(lubm2)
+    base = wk.Plan([(16, Q.TYPE_ID, wk.DIR_IN, -1),
+                    (-1, -2, wk.DIR_IN, -3),      # vu binds P, Y
+                    (-1, Q.WORKSFOR, wk.DIR_OUT, -3)],  # k2k on Y
+                   3, [-1, -2, -3])
+    want = sort_rows(oc.run_query(base))
+    for order in ([0, 1, 2], [0, 2, 1], [2, 1, 0], [1, 2, 0]):
+        shuffled = [base.patterns[i] for i in order]
+        planned = planner.plan_patterns(store, shuffled, 3, [-1, -2, -3])
+        # the vu must come before the pattern that would bind its object
+        vu_pos = next(i for i, p in enumerate(planned.patterns) if p[1] < 0)
+        k2k_pos = next(i for i, p in enumerate(planned.patterns)
+                       if p[1] == Q.WORKSFOR)
+        assert vu_pos < k2k_pos, planned.patterns
+        got = sort_rows(oc.run_query(planned))
+        assert got.shape == want.shape, (order, planned.patterns)
+        assert np.array_equal(got, want), (order, planned.patterns)
+
+
+def test_planner_invariance_fuzz_vu_heavy(lubm2):
+    """Shuffled-input invariance with MORE predicate variables and
+    cross-references than random_plan emits (the generator that found
+    the vu-after-bind bug)."""
+    import random
+    from tests.test_fuzz_plans import PREDS, TYPES
+    from tests.oracle_util import OracleCtx
+
+    store = wk.Store(lubm2)
+    oc = OracleCtx(lubm2)
+    rng = random.Random(226)
+    ran = 0
+    for trial in range(40):
+        nv = rng.randint(3, 6)
+        vars_ = [-(i + 1) for i in range(nv)]
+        t = rng.choice(TYPES)
+        pats = [(t, Q.TYPE_ID, wk.DIR_IN, vars_[0])]
+        bound = [vars_[0]]
+        free = vars_[1:]
+        for _ in range(rng.randint(2, 4)):
+            s = rng.choice(bound)
+            r = rng.random()
+            if r < 0.35 and len(free) >= 2:
+                pv, ov = free.pop(0), free.pop(0)
+                pats.append((s, pv, rng.choice([0, 1]), ov))
+                bound += [pv, ov]
+            elif r < 0.7 and free:
+                o = free.pop(0)
+                pats.append((s, rng.choice(PREDS), rng.choice([0, 1]), o))
+                bound.append(o)
+            else:
+                o = rng.choice(bound)
+                if o != s:
+                    pats.append((s, rng.choice(PREDS), rng.choice([0, 1]), o))
+                else:
+                    pats.append((s, Q.TYPE_ID, wk.DIR_OUT, rng.choice(TYPES)))
+        base = wk.Plan(pats, nvars=nv, required_vars=bound)
+        try:
+            want = oc.run_query(base)
+        except RuntimeError:
+            continue
+        shuffled = list(base.patterns)
+        rng.shuffle(shuffled)
+        planned = planner.plan_patterns(store, shuffled, nv, bound)
+        got = oc.run_query(planned)
+        assert got.shape == want.shape, (trial, base.patterns,
+                                         planned.patterns)
+        assert np.array_equal(sort_rows(got), sort_rows(want)), \
+            (trial, base.patterns, planned.patterns)
+        ran += 1
+    assert ran > 20

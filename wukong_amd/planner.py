@@ -41,6 +41,11 @@ def _cands(store, pat, bound, first, est_rows):
             # predicate variable: needs const or bound start
             if p in bound:
                 continue  # predicate already bound: unsupported
+            if b < 0 and b in bound:
+                # known_unknown_known: absent in the reference and
+                # rejected by the engine (ostat==1) — ordering a vu
+                # AFTER its object var binds would change semantics
+                continue
             if a >= 0 and not _is_tpid(a) and first:
                 cost = new_rows = float(
                     len(store.get_triples(a, PREDICATE_ID, d)) or 1)
