@@ -365,6 +365,11 @@ static void exec_pattern(const ctx &c, query &q) {
         }
         q.table.swap(out);
         q.v2c[-(p + 1)] = q.col_num;
+        // NOTE: when o is a var this REBINDS it unconditionally — the
+        // known_unknown_known shape has no operator in the reference
+        // (sparql.hpp:556-744) and the engine rejects it (WK_ERR_PLAN);
+        // the planner never emits a vu after its object var binds, so
+        // this branch only ever sees a fresh o var.
         if (o < 0) q.v2c[-(o + 1)] = q.col_num + 1;
         q.col_num += (o >= 0) ? 1 : 2;
         q.step++;
