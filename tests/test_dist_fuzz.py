@@ -231,3 +231,74 @@ def test_dist_fuzz_watdiv_two_ranks():
                                            plan.patterns)
             assert np.array_equal(g, want), (mode, j, plan.patterns)
     assert nonempty >= 8
+
+
+def _gen_planned(seed=808, n=10):
+    """Shuffled random BGPs recovered by the planner — the planner's
+    orders (filter-first, reoriented edges) through the dist driver."""
+    from wukong_amd import planner
+    store = wk.Store(wk.lubm_gen(2, seed=42))
+    rng = random.Random(seed)
+    plans = []
+    while len(plans) < n:
+        base = random_plan(rng, store)
+        shuffled = list(base.patterns)
+        rng.shuffle(shuffled)
+        try:
+            planned = planner.plan_patterns(store, shuffled, base.nvars,
+                                            base.required_vars)
+        except planner.PlannerError:
+            continue
+        plans.append((base, planned))
+    return plans
+
+
+def _worker_planned(rank, world, port, results):
+    import sys
+    sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    import torch.distributed as dist
+    from wukong_amd.dist import DistQuery
+    from tests.oracle_util import OracleExecutor
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        ctxs = [OracleCtx(wk.lubm_gen(2, seed=42, sid=r, nsrv=world),
+                          sid=r, nsrv=world) for r in range(world)]
+        out = {}
+        for j, (base, planned) in enumerate(_gen_planned()):
+            for mode, thr in (("x", 0), ("m", 300)):
+                ex = OracleExecutor(ctxs[rank], planned, peers=ctxs)
+                dq = DistQuery(ex, planned, rank, world, threshold=thr)
+                dq.run()
+                out[f"{mode}:{j}"] = sort_rows(dq.gather_result())
+        if rank == 0:
+            results.put(out)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(900)
+def test_dist_planned_plans_equal_textual():
+    """Planner output (from shuffled inputs) through the 2-rank driver
+    must equal the single oracle on the TEXTUAL order — crossing the
+    planner's reorderings with the exchange/remote machinery."""
+    ctx = mp.get_context("spawn")
+    results = ctx.Queue()
+    procs = [ctx.Process(target=_worker_planned, args=(r, 2, 29884, results))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    got = results.get(timeout=840)
+    for p in procs:
+        p.join(timeout=60)
+
+    full = OracleCtx(wk.lubm_gen(2, seed=42))
+    for j, (base, planned) in enumerate(_gen_planned()):
+        want = sort_rows(full.run_query(base))
+        for mode in ("x", "m"):
+            g = got[f"{mode}:{j}"]
+            assert g.shape == want.shape, (mode, j, base.patterns,
+                                           planned.patterns)
+            assert np.array_equal(g, want), (mode, j, planned.patterns)
