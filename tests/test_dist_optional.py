@@ -37,8 +37,15 @@ def _opt_plans(Q, wk):
     dlo = wk.Plan([(Q.UGSTUDENT, Q.TYPE_ID, wk.DIR_IN, X)], 2, [Y],
                   distinct=True, limit=9, offset=4,
                   optional=[(X, Q.ADVISOR, wk.DIR_OUT, Y)])
+    # 3-pattern group: chain over two group-born columns + typeof on
+    # the second (unmatch must blank BOTH opt columns)
+    W2 = -4
+    chain = wk.Plan([(Q.UGSTUDENT, Q.TYPE_ID, wk.DIR_IN, X)], 4, [X, Y, W2],
+                    optional=[(X, Q.ADVISOR, wk.DIR_OUT, Y),
+                              (Y, Q.WORKSFOR, wk.DIR_OUT, W2),
+                              (W2, Q.TYPE_ID, wk.DIR_OUT, Q.DEPARTMENT)])
     return {"k2u": k2u, "k2c": k2c, "k2k": k2k, "cfilter": cfilter,
-            "uo": uo, "dlo": dlo}
+            "uo": uo, "dlo": dlo, "chain": chain}
 
 
 def _worker(rank, world, port, results):
