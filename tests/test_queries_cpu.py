@@ -85,3 +85,18 @@ def test_golden_fixture(oracle4):
         t = oracle4.run_query(plans[name])
         assert t.shape[0] == rec["rows"], name
         assert fnv1a_fast(t) == rec["sha"], name
+
+
+def test_oracle_mt_slicing_equal(lubm2):
+    """cpu_baseline runs the oracle with mt=cores (index-scan slicing,
+    sparql.hpp:210-221): results must equal mt=1 for every suite query
+    at factors beyond the index sizes."""
+    from wukong_amd import queries as Q
+    from tests.oracle_util import OracleCtx
+    oc = OracleCtx(lubm2)
+    for name, plan in Q.ALL.items():
+        base = sort_rows(oc.run_query(plan, mt=1))
+        for mt in (3, 64, 1000):
+            got = sort_rows(oc.run_query(plan, mt=mt))
+            assert got.shape == base.shape, (name, mt)
+            assert np.array_equal(got, base), (name, mt)
