@@ -15,4 +15,5 @@ run python tools/soak/watdiv_dist_soak.py $((S+3)) $((P+6)) 2
 run python tools/soak/opt3_dist_soak.py $((S+4)) $((P+8)) 3
 run python tools/soak/planned_dist_soak.py $((S+5)) $((P+10))
 run python tools/soak/planned_groups_dist.py
+run python tools/soak/env_combo_soak.py
 exit $fail
