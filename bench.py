@@ -9,7 +9,9 @@ per GPU, store partitioned by vid % N, per-step RCCL all-to-allv.
 Env knobs: WK_UNIV (default 2560), WK_CPU_UNIV (cpu_baseline sample,
 default = the workload scale: SAME inputs), WK_SKIP_CPU_BASELINE=1,
 WK_SKIP_GATES=1, WK_EMU_EMBED (embedded emulator queries, 0 = off),
-WK_DIST_BACKEND=gloo (single-GPU rehearsal of the N>1 path).
+WK_DIST_BACKEND=gloo (single-GPU rehearsal of the N>1 path); the full
+knob table (WK_INFLIGHT, WK_GRAPH, WK_FN/CSR/TBM budgets, ...) is in
+README.md.
 """
 import argparse
 import json
